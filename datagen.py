@@ -1,0 +1,121 @@
+"""Synthetic input generators for the benchmark configs.
+
+All inputs numpy.random.default_rng(seed) with seed 42 by default; sizes
+scalable for tests. Shared by bench.py, tests/ and the oracle checks —
+harness code, not product code.
+"""
+from __future__ import annotations
+
+import numpy as np
+import pandas as pd
+
+SEED = 42
+
+# dictionary code maps for the synthetic TPC-H columns
+RETURNFLAG_DICT = ["A", "N", "R"]
+LINESTATUS_DICT = ["F", "O"]
+MKTSEGMENT_DICT = ["BUILDING", "AUTOMOBILE", "FURNITURE", "HOUSEHOLD", "MACHINERY"]
+
+D_EPOCH = np.datetime64("1970-01-01")
+
+
+def _days(s: str) -> int:
+    return int((np.datetime64(s) - D_EPOCH).astype(int))
+
+
+def gen_c1(n=1_000_000, n_groups=1_000, seed=SEED):
+    """C1: 1M-row 2-column frame: key in [0,1000), x uniform [0,1)."""
+    rng = np.random.default_rng(seed)
+    key = rng.integers(0, n_groups, size=n, dtype=np.int64)
+    x = rng.random(size=n)
+    return key, x
+
+
+def gen_c2(n=100_000_000, n_groups=1_000_000, seed=SEED):
+    """C2: 100M rows int64 key ~ U[0, n_groups), fp64 val ~ U[0,1)."""
+    rng = np.random.default_rng(seed)
+    key = rng.integers(0, n_groups, size=n, dtype=np.int64)
+    val = rng.random(size=n)
+    return key, val
+
+
+def gen_c3(n_build=10_000_000, n_probe=100_000_000, seed=SEED):
+    """C3: build = shuffled arange (unique), probe = choice of build keys."""
+    rng = np.random.default_rng(seed)
+    build_key = rng.permutation(n_build).astype(np.int64)
+    probe_key = rng.integers(0, n_build, size=n_probe, dtype=np.int64)  # 100% hit
+    build_val = rng.random(size=n_build)
+    probe_val = rng.random(size=n_probe)
+    return build_key, build_val, probe_key, probe_val
+
+
+def gen_lineitem_q1(n=59_986_052, seed=SEED) -> pd.DataFrame:
+    """C4: synthetic TPC-H SF10 lineitem for Q1 (SURVEY §8d distributions).
+
+    Columns: l_quantity f64 (uniform int 1-50), l_extendedprice f64 (derived),
+    l_discount f64 (uniform 0-0.1), l_tax f64 (0-0.08), l_returnflag i8 code,
+    l_linestatus i8 code, l_shipdate date32 (uniform 1992-1998).
+    """
+    rng = np.random.default_rng(seed)
+    qty = rng.integers(1, 51, size=n).astype(np.float64)
+    extprice = qty * (90000.0 + 100000.0 * rng.random(size=n)) / 50.0
+    discount = np.round(rng.random(size=n) * 0.10, 2)
+    tax = np.round(rng.random(size=n) * 0.08, 2)
+    returnflag = rng.integers(0, len(RETURNFLAG_DICT), size=n).astype(np.int8)
+    linestatus = rng.integers(0, len(LINESTATUS_DICT), size=n).astype(np.int8)
+    shipdate = rng.integers(
+        _days("1992-01-01"), _days("1998-12-01"), size=n
+    ).astype(np.int32)
+    return pd.DataFrame(
+        {
+            "l_quantity": qty,
+            "l_extendedprice": extprice,
+            "l_discount": discount,
+            "l_tax": tax,
+            "l_returnflag": returnflag,
+            "l_linestatus": linestatus,
+            "l_shipdate": shipdate,
+        }
+    )
+
+
+def gen_q3(sf_rows=(1_500_000, 15_000_000, 60_000_000), seed=SEED):
+    """C5: synthetic TPC-H SF10 customer / orders / lineitem for Q3.
+
+    customer: c_custkey i64 (unique), c_mktsegment i8 code (uniform of 5;
+      code 0 = BUILDING → 1/5 selectivity).
+    orders: o_orderkey i64 (unique), o_custkey i64 ~ U[customers],
+      o_orderdate date32 uniform 1992-1998 (≈half < 1995-03-15),
+      o_shippriority i32 = 0.
+    lineitem: l_orderkey i64 ~ U[orders], l_extendedprice / l_discount f64,
+      l_shipdate date32 uniform 1992-1998 (≈half > 1995-03-15).
+    """
+    n_cust, n_ord, n_li = sf_rows
+    rng = np.random.default_rng(seed)
+    cust = pd.DataFrame(
+        {
+            "c_custkey": np.arange(n_cust, dtype=np.int64),
+            "c_mktsegment": rng.integers(0, 5, size=n_cust).astype(np.int8),
+        }
+    )
+    orders = pd.DataFrame(
+        {
+            "o_orderkey": np.arange(n_ord, dtype=np.int64),
+            "o_custkey": rng.integers(0, n_cust, size=n_ord, dtype=np.int64),
+            "o_orderdate": rng.integers(
+                _days("1992-01-01"), _days("1998-12-01"), size=n_ord
+            ).astype(np.int32),
+            "o_shippriority": np.zeros(n_ord, dtype=np.int32),
+        }
+    )
+    li = pd.DataFrame(
+        {
+            "l_orderkey": rng.integers(0, n_ord, size=n_li, dtype=np.int64),
+            "l_extendedprice": 900.0 + 100000.0 * rng.random(size=n_li),
+            "l_discount": np.round(rng.random(size=n_li) * 0.10, 2),
+            "l_shipdate": rng.integers(
+                _days("1992-01-01"), _days("1998-12-01"), size=n_li
+            ).astype(np.int32),
+        }
+    )
+    return cust, orders, li
