@@ -1,0 +1,246 @@
+/* dsxhip — C ABI of the MI355X-native physical execution layer for dask-sql's
+ * hot path.
+ *
+ * Each entry point cites the reference interface it replaces
+ * (paths relative to /root/reference). The reference's own execution layer is
+ * pandas/Dask called from dask_sql/physical/; this library is the
+ * HIP/CDNA4 (gfx950) replacement for exactly those calls. Host bindings:
+ * ctypes (dask_sql_amd/runtime.py); a cgo/JNI/N-API-style stub is shown in
+ * INTEGRATION.md.
+ *
+ * Conventions:
+ *  - plain pointers + lengths; no torch/Arrow types in signatures. Device
+ *    pointers are HIP device memory on the context's device.
+ *  - validity masks are uint8[n] (1 = valid); NULL pointer = all valid.
+ *    (Arrow bitmaps are converted at the Python boundary.)
+ *  - outputs with data-dependent size are library-allocated (hipMalloc);
+ *    release with dsx_free. Fixed-size outputs are caller-allocated.
+ *  - return: 0 = OK, <0 = error; dsx_last_error() has the message.
+ *  - thread-safety: one DsxCtx per thread/GPU; calls on one ctx serialize on
+ *    its HIP stream. Multi-GPU collectives are the host's job (RCCL via
+ *    torch.distributed, one process per GPU) — this library is single-device.
+ */
+#ifndef DSXHIP_H
+#define DSXHIP_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef struct DsxCtx DsxCtx;
+
+/* dtype tags (subset per SURVEY.md §2 "Type mappings" row) */
+enum DsxType {
+  DSX_I64 = 0,
+  DSX_F64 = 1,
+  DSX_I32 = 2,    /* also DATE32 (days since epoch) */
+  DSX_F32 = 3,
+  DSX_I8  = 4,    /* also dictionary codes */
+  DSX_BOOL8 = 5,
+};
+
+/* expression VM opcodes — the device-compiled form of the reference's Rex
+ * operator subset (dask_sql/physical/rex/core/call.py:1047-1156: comparisons,
+ * AND/OR/NOT, +,-,*,/, IS [NOT] NULL, CASE-as-select, CAST). Programs are
+ * typed postfix; built by dask_sql_amd/physical/rex compiler. */
+enum DsxOp {
+  DSX_OP_COL = 1,       /* arg0 = column index; pushes typed value+validity  */
+  DSX_OP_LIT_F64 = 2,   /* imm = f64 bits                                    */
+  DSX_OP_LIT_I64 = 3,   /* imm = i64                                         */
+  DSX_OP_LIT_NULL = 4,
+  DSX_OP_ADD_F64 = 10, DSX_OP_SUB_F64 = 11, DSX_OP_MUL_F64 = 12, DSX_OP_DIV_F64 = 13,
+  DSX_OP_ADD_I64 = 14, DSX_OP_SUB_I64 = 15, DSX_OP_MUL_I64 = 16,
+  DSX_OP_LT_F64 = 20, DSX_OP_LE_F64 = 21, DSX_OP_GT_F64 = 22, DSX_OP_GE_F64 = 23,
+  DSX_OP_EQ_F64 = 24, DSX_OP_NE_F64 = 25,
+  DSX_OP_LT_I64 = 30, DSX_OP_LE_I64 = 31, DSX_OP_GT_I64 = 32, DSX_OP_GE_I64 = 33,
+  DSX_OP_EQ_I64 = 34, DSX_OP_NE_I64 = 35,
+  DSX_OP_AND = 40, DSX_OP_OR = 41, DSX_OP_NOT = 42,   /* SQL 3-valued logic   */
+  DSX_OP_IS_NULL = 43, DSX_OP_IS_NOT_NULL = 44,
+  DSX_OP_I64_TO_F64 = 50, DSX_OP_F64_TO_I64 = 51,     /* CAST (trunc,
+                                                         mappings.py:346-353) */
+  DSX_OP_SELECT = 60,   /* (cond, a, b) -> cond ? a : b — CASE WHEN           */
+  DSX_OP_NEG_F64 = 61, DSX_OP_NEG_I64 = 62,
+};
+
+typedef struct DsxInstr {
+  int32_t op;       /* DsxOp */
+  int32_t arg0;     /* column index for OP_COL */
+  int64_t imm;      /* literal bits */
+} DsxInstr;
+
+#define DSX_MAX_PROG 48
+#define DSX_MAX_COLS 16
+#define DSX_MAX_AGGS 8
+#define DSX_MAX_KEYS 4
+
+typedef struct DsxColumn {
+  void* data;               /* device pointer */
+  const uint8_t* validity;  /* device pointer or NULL (all valid) */
+  int64_t len;
+  int32_t dtype;            /* DsxType */
+} DsxColumn;
+
+/* ---- context / memory -------------------------------------------------- */
+
+/* replaces: process/worker setup the reference delegates to dask.distributed
+ * (SURVEY.md §5 "Distributed communication backend"). One ctx per GPU. */
+int dsx_ctx_create(int device_id, DsxCtx** out);
+void dsx_ctx_destroy(DsxCtx* ctx);
+const char* dsx_last_error(void);
+int dsx_synchronize(DsxCtx* ctx);
+
+int dsx_malloc(DsxCtx* ctx, int64_t bytes, void** out);
+int dsx_free(DsxCtx* ctx, void* ptr);
+/* replaces the host→worker data movement of dask's task shuffle for table
+ * registration (dask_sql/context.py:168 create_table / persist):
+ * pinned-host→HBM hipMemcpyAsync. */
+int dsx_upload(DsxCtx* ctx, const void* host, int64_t bytes, void** out_dev);
+int dsx_download(DsxCtx* ctx, const void* dev, void* host, int64_t bytes);
+
+/* per-kernel HIP-event timing (for bench.py roofline accounting) */
+int dsx_prof_enable(DsxCtx* ctx, int enable);
+/* fills (name, total_ms, launches) for up to cap kernels; returns count */
+int dsx_prof_get(DsxCtx* ctx, char names[][32], double* total_ms,
+                 int64_t* launches, int cap);
+int dsx_prof_reset(DsxCtx* ctx);
+
+/* ---- scan / expression / filter ---------------------------------------- */
+
+/* replaces Projection `df.assign(RexConverter result)`
+ * (dask_sql/physical/rel/logical/project.py:56-65 and rex/core/call.py ops):
+ * out = program(cols) per row. out_dtype in {DSX_I64, DSX_F64, DSX_BOOL8}.
+ * out_data caller-allocated (n * sizeof), out_validity caller-allocated u8[n]
+ * or NULL to discard. */
+int dsx_eval(DsxCtx* ctx, const DsxInstr* prog, int prog_len,
+             const DsxColumn* cols, int ncols, int64_t n,
+             void* out_data, uint8_t* out_validity, int32_t out_dtype);
+
+/* replaces Filter `cond.fillna(False); df[cond]`
+ * (dask_sql/physical/rel/logical/filter.py:20-45): evaluates the predicate
+ * program, NULL→False, and emits the ORDER-PRESERVING selection vector of
+ * matching row ids (library-allocated u32; free with dsx_free). */
+int dsx_filter(DsxCtx* ctx, const DsxInstr* prog, int prog_len,
+               const DsxColumn* cols, int ncols, int64_t n,
+               uint32_t** out_sel, int64_t* out_count);
+
+/* boolean-mask take / merge materialization: out[i] = col[sel[i]].
+ * out caller-allocated. Gathers validity too when both non-NULL. */
+int dsx_gather(DsxCtx* ctx, const DsxColumn* col, const uint32_t* sel,
+               int64_t n_sel, void* out_data, uint8_t* out_validity);
+
+/* min/max of an i64/i32/i8/date32 column ignoring NULLs (key-range probe for
+ * packing; also MIN/MAX aggregate support). */
+int dsx_minmax_i64(DsxCtx* ctx, const DsxColumn* col, int64_t* out_min,
+                   int64_t* out_max, int64_t* out_nonnull);
+
+/* pack up to 4 key columns into one u64 code column:
+ * code = Σ_k (col_k - min_k + nullable_k) * stride_k, NULL → 0 slot.
+ * Implements composite GROUP BY / join keys; NULL gets its own code, which
+ * is what gives groupby(dropna=False) (aggregate.py:575-577) for free.
+ * out caller-allocated u64[n]. */
+typedef struct DsxKeySpec {
+  int32_t col;        /* index into cols */
+  int64_t min;        /* from dsx_minmax */
+  int64_t range;      /* max-min+1 (+1 more reserved internally if nullable) */
+  int32_t nullable;   /* 0/1 */
+} DsxKeySpec;
+int dsx_keypack(DsxCtx* ctx, const DsxColumn* cols, int ncols,
+                const DsxKeySpec* keys, int nkeys, int64_t n,
+                uint64_t* out_codes);
+
+/* ---- hash join ---------------------------------------------------------- */
+
+/* replaces the per-partition pandas hash join inside
+ * `dd.merge(on=..., how=...)` (dask_sql/physical/rel/logical/join.py:241-246).
+ * Build: open-addressing multimap over u64 key codes (from dsx_keypack, or
+ * raw non-negative i64 keys). NULL keys (code with validity 0) are NOT
+ * inserted — the NULL-key drop of join.py:202-213 for the build side.
+ * Table is library-allocated; free with dsx_hash_table_free. */
+typedef struct DsxHashTable DsxHashTable;
+int dsx_hash_build(DsxCtx* ctx, const uint64_t* codes, const uint8_t* validity,
+                   int64_t n, DsxHashTable** out);
+void dsx_hash_table_free(DsxHashTable* t);
+
+enum DsxJoinType {  /* reference join.py:41-48 JOIN_TYPE_MAPPING */
+  DSX_JOIN_INNER = 0,
+  DSX_JOIN_LEFT = 1,       /* left outer: unmatched probe → rhs NULL */
+  DSX_JOIN_LEFTSEMI = 2,
+  DSX_JOIN_LEFTANTI = 3,
+};
+#define DSX_NULL_IDX 0xFFFFFFFFu
+
+/* Probe: emits (probe_rowid, build_rowid) pairs, library-allocated.
+ * Unordered (normalize by sort for comparisons); FULL OUTER is composed by
+ * the host from LEFT + unmatched-build sweep (dsx_hash_unmatched). */
+int dsx_hash_probe(DsxCtx* ctx, DsxHashTable* t, const uint64_t* codes,
+                   const uint8_t* validity, int64_t n, int join_type,
+                   uint32_t** out_probe_idx, uint32_t** out_build_idx,
+                   int64_t* out_count);
+/* build rows never matched by any probe since build (for FULL OUTER,
+ * join.py JOIN_TYPE_MAPPING "FULL" → outer). */
+int dsx_hash_unmatched(DsxCtx* ctx, DsxHashTable* t, uint32_t** out_build_idx,
+                       int64_t* out_count);
+
+/* ---- hash groupby-aggregate --------------------------------------------- */
+
+enum DsxAggOp {  /* reference AGGREGATION_MAPPING aggregate.py:117-231 subset:
+                    sum (custom_sum min_count=1, :486-493), count, min, max;
+                    avg finalized on host as sum/count. */
+  DSX_AGG_SUM_F64 = 0,
+  DSX_AGG_SUM_I64 = 1,
+  DSX_AGG_COUNT = 2,     /* counts rows where input program is non-NULL */
+  DSX_AGG_MIN_F64 = 3,
+  DSX_AGG_MAX_F64 = 4,
+  DSX_AGG_MIN_I64 = 5,
+  DSX_AGG_MAX_I64 = 6,
+};
+
+typedef struct DsxAggSpec {
+  int32_t op;                    /* DsxAggOp */
+  int32_t prog_len;
+  DsxInstr prog[DSX_MAX_PROG];   /* input expression, fused into the kernel */
+} DsxAggSpec;
+
+/* replaces `df.groupby(by, dropna=False).agg(...)`
+ * (dask_sql/physical/rel/logical/aggregate.py:575-581) with the WHERE
+ * predicate fused in (filter.py:20-45 fused into the same scan — SURVEY §3
+ * call stack (2)+(4)).
+ *
+ * keys: u64 codes from dsx_keypack with known range key_space (≤ 2^63).
+ *       key_space ≤ lds_threshold → per-CU LDS direct-indexed accumulation;
+ *       else global CAS-claim table (SURVEY §7 step 4 two-level design).
+ * pred: optional predicate program (NULL→False), pred_len 0 = no predicate.
+ * Outputs (library-allocated, compacted, one row per non-empty group):
+ *   out_codes u64[G], per-agg f64/i64 value arrays, per-agg u64 nonnull
+ *   counts (for SUM NULL semantics and COUNT), G = *out_groups.
+ *   For DSX_AGG_SUM_*: value is the sum over non-NULL inputs; nonnull count 0
+ *   ⇒ SQL NULL (custom_sum min_count=1) — host finalizes.
+ *   For MIN/MAX: same. For COUNT: value array unused, count is the result. */
+int dsx_hash_groupby(DsxCtx* ctx,
+                     const DsxColumn* cols, int ncols, int64_t n,
+                     const uint64_t* key_codes, uint64_t key_space,
+                     const DsxInstr* pred, int pred_len,
+                     const DsxAggSpec* aggs, int naggs,
+                     uint64_t** out_codes, void** out_vals /*[naggs]*/,
+                     uint64_t** out_counts /*[naggs]*/, int64_t* out_groups);
+
+/* ---- shuffle support --------------------------------------- */
+
+/* replaces dask's hash-repartition "tasks" shuffle split
+ * (dask_sql/__init__.py:16, conftest.py:17; inside dd.merge/groupby):
+ * bucket rows by mix64(code) % nbuckets, stable within bucket. Emits the
+ * per-bucket-contiguous ORDERED selection vector into caller-allocated
+ * out_sel u32[n] and bucket row offsets into out_offsets i64[nbuckets+1].
+ * The host gathers each bucket's columns into torch-allocated staging
+ * buffers and exchanges them with torch.distributed all_to_all (RCCL/xGMI). */
+int dsx_partition(DsxCtx* ctx, const uint64_t* codes, const uint8_t* validity,
+                  int64_t n, int nbuckets, uint32_t* out_sel,
+                  int64_t* out_offsets);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* DSXHIP_H */
