@@ -1,0 +1,219 @@
+"""Context — the user-facing API, kept verbatim-shaped with the reference
+(dask_sql/context.py): Context(), create_table (:168), sql (:482),
+drop_table, explain. The planner is ours (thin); execution is
+the HIP physical layer (no CPU fallback — DsxUnavailable propagates)."""
+from __future__ import annotations
+
+import logging
+
+import numpy as np
+import pandas as pd
+
+from dask_sql_amd import runtime as rt
+from dask_sql_amd.datacontainer import (ColumnContainer, DataContainer,
+                                        DeviceTable, HostDataContainer)
+from dask_sql_amd.planner.builder import Builder, Catalog
+from dask_sql_amd.physical.convert import RelConverter
+from dask_sql_amd.physical.rel_plugins import register_defaults
+
+logger = logging.getLogger(__name__)
+
+register_defaults()
+
+
+class _HostColumn:
+    def __init__(self, arr, validity, sql_type, dtype, dictionary=None):
+        self.arr = arr
+        self.validity = validity
+        self.sql_type = sql_type
+        self.dtype = dtype
+        self.dictionary = dictionary
+
+
+def _from_pandas(df: pd.DataFrame, date_columns=(), dictionaries=None):
+    """pandas → host column set (reference input_utils/convert.py:40-60
+    functionally: anything the user hands in becomes our columnar form)."""
+    dictionaries = dictionaries or {}
+    cols = {}
+    for name in df.columns:
+        s = df[name]
+        validity = None
+        dictionary = None
+        dt = s.dtype
+        if isinstance(dt, pd.CategoricalDtype):
+            codes = s.cat.codes.to_numpy()
+            validity = (codes >= 0).astype(np.uint8)
+            arr, dtype, sqlt = codes.astype(np.int32), rt.I32, "VARCHAR"
+            dictionary = list(s.cat.categories)
+            if validity.all():
+                validity = None
+        elif pd.api.types.is_extension_array_dtype(dt):
+            # nullable Int*/UInt*/Float*/boolean
+            mask = s.isna().to_numpy()
+            base = s.fillna(0).to_numpy()
+            if base.dtype == object:
+                base = base.astype(np.float64)
+            validity = (~mask).astype(np.uint8) if mask.any() else None
+            arr, dtype, sqlt = _np_map(np.asarray(base))
+        elif dt == object or pd.api.types.is_string_dtype(dt):
+            codes, uniques = pd.factorize(s)
+            validity = (codes >= 0).astype(np.uint8)
+            if validity.all():
+                validity = None
+            arr, dtype, sqlt = codes.astype(np.int32), rt.I32, "VARCHAR"
+            dictionary = list(uniques)
+        elif np.issubdtype(dt, np.datetime64):
+            vals = s.to_numpy().astype("datetime64[D]").astype(np.int32)
+            nat = s.isna().to_numpy()
+            validity = (~nat).astype(np.uint8) if nat.any() else None
+            arr, dtype, sqlt = vals, rt.I32, "DATE"
+        elif np.issubdtype(dt, np.floating):
+            arr = s.to_numpy()
+            # float NaN stays a VALUE (pandas semantics treat it as missing
+            # in aggregations; we mirror by mapping NaN → NULL validity)
+            nan = np.isnan(arr)
+            validity = (~nan).astype(np.uint8) if nan.any() else None
+            arr, dtype, sqlt = _np_map(arr)
+        else:
+            arr, dtype, sqlt = _np_map(s.to_numpy())
+        if name in date_columns:
+            sqlt = "DATE"
+        if name in dictionaries:
+            dictionary = dictionaries[name]
+            sqlt = "VARCHAR"
+        cols[str(name)] = _HostColumn(np.ascontiguousarray(arr), validity,
+                                      sqlt, dtype, dictionary)
+    return cols
+
+
+def _np_map(arr):
+    m = {
+        np.dtype("int64"): (rt.I64, "BIGINT"),
+        np.dtype("int32"): (rt.I32, "INTEGER"),
+        np.dtype("int16"): (rt.I32, "INTEGER"),
+        np.dtype("int8"): (rt.I8, "TINYINT"),
+        np.dtype("uint8"): (rt.I32, "INTEGER"),
+        np.dtype("uint16"): (rt.I32, "INTEGER"),
+        np.dtype("uint32"): (rt.I64, "BIGINT"),
+        np.dtype("uint64"): (rt.I64, "BIGINT"),
+        np.dtype("float64"): (rt.F64, "DOUBLE"),
+        np.dtype("float32"): (rt.F32, "FLOAT"),
+        np.dtype("bool"): (rt.BOOL8, "BOOLEAN"),
+    }
+    if arr.dtype in (np.dtype("int16"), np.dtype("uint16")):
+        arr = arr.astype(np.int32)
+    elif arr.dtype in (np.dtype("uint8"),):
+        arr = arr.astype(np.int32)
+    elif arr.dtype in (np.dtype("uint32"), np.dtype("uint64")):
+        arr = arr.astype(np.int64)
+    if arr.dtype not in m:
+        raise NotImplementedError(f"dtype {arr.dtype} not supported")
+    dtype, sqlt = m[arr.dtype]
+    return arr, dtype, sqlt
+
+
+class RegisteredTable:
+    def __init__(self, host_cols: dict):
+        self.host_cols = host_cols
+        self.device_table: DeviceTable | None = None
+
+    def fields(self):
+        return [(n, c.sql_type) for n, c in self.host_cols.items()]
+
+    def upload(self, runtime) -> DeviceTable:
+        if self.device_table is None:
+            cols = {}
+            for n, h in self.host_cols.items():
+                col = runtime.upload_column(h.arr, h.validity, h.dtype)
+                if h.dictionary is not None:
+                    col.dictionary = h.dictionary
+                cols[n] = col
+            self.device_table = DeviceTable(cols)
+        return self.device_table
+
+
+class ResultFrame:
+    """Shaped like the reference's lazy return of Context.sql (a dataframe
+    you .compute()); here execution already happened on the GPU and compute()
+    is the device→host materialization."""
+
+    def __init__(self, dc, rel, context):
+        self._dc = dc
+        self._rel = rel
+        self._context = context
+
+    def compute(self):
+        from dask_sql_amd.materialize import to_pandas
+        if isinstance(self._dc, HostDataContainer):
+            return self._dc.pdf
+        return to_pandas(self._dc, self._context, self._rel.getRowType())
+
+    # parity alias
+    def to_pandas(self):
+        return self.compute()
+
+    @property
+    def dc(self):
+        return self._dc
+
+
+class Context:
+    """reference dask_sql/context.py:90+ — same API surface for the hot path.
+
+    c = Context(); c.create_table("t", df); c.sql("SELECT ...").compute()
+    """
+
+    DEFAULT_SCHEMA_NAME = "root"
+
+    def __init__(self, device_id: int = 0):
+        self.schema_name = self.DEFAULT_SCHEMA_NAME
+        self.catalog = Catalog()
+        self.tables: dict[str, RegisteredTable] = {}
+        self._runtime = None
+        self._device_id = device_id
+
+    # -- reference context.py:168 create_table ----------------------------
+    def create_table(self, table_name: str, input_table, persist: bool = False,
+                     date_columns=(), dictionaries=None, gpu: bool = True,
+                     **kwargs):
+        if isinstance(input_table, dict):
+            input_table = pd.DataFrame(input_table)
+        if not isinstance(input_table, pd.DataFrame):
+            raise NotImplementedError(
+                "only pandas/dict inputs (input plugins are out of scope, "
+                "SURVEY §2)")
+        host_cols = _from_pandas(input_table, date_columns, dictionaries)
+        t = RegisteredTable(host_cols)
+        self.tables[table_name.lower()] = t
+        self.catalog.add(table_name, t.fields())
+        if persist:
+            t.upload(self._get_runtime())
+
+    def drop_table(self, table_name: str):
+        self.tables.pop(table_name.lower(), None)
+        self.catalog.drop(table_name)
+
+    # -- reference context.py:482 sql --------------------------------------
+    def sql(self, sql: str, return_futures: bool = True,
+            config_options=None) -> ResultFrame:
+        rel = self._get_ral(sql)
+        logger.debug("plan:\n%s", rel.explain())
+        dc = RelConverter.convert(rel, context=self)
+        return ResultFrame(dc, rel, self)
+
+    def explain(self, sql: str) -> str:
+        return self._get_ral(sql).explain()
+
+    # -- internals ----------------------------------------------------------
+    def _get_ral(self, sql: str):
+        """reference context.py:819 _get_ral (planner entry)."""
+        return Builder(self.catalog, self.schema_name).build(sql)
+
+    def _get_runtime(self):
+        if self._runtime is None:
+            self._runtime = rt.Runtime(self._device_id)
+        return self._runtime
+
+    def _device_table(self, table_name: str) -> DeviceTable:
+        t = self.tables[table_name.lower()]
+        return t.upload(self._get_runtime())

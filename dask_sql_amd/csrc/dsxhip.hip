@@ -271,6 +271,8 @@ __device__ bool vm_eval(const DsxInstr* prog, int len, const ColsArg& C,
       BIN_I(DSX_OP_ADD_I64, i = a + b)
       BIN_I(DSX_OP_SUB_I64, i = a - b)
       BIN_I(DSX_OP_MUL_I64, i = a * b)
+      BIN_I(DSX_OP_DIV_I64, i = b ? a / b : 0)
+      BIN_I(DSX_OP_MOD_I64, i = b ? a % b : 0)
       BIN_I(DSX_OP_LT_I64, i = (a < b) ? 1 : 0)
       BIN_I(DSX_OP_LE_I64, i = (a <= b) ? 1 : 0)
       BIN_I(DSX_OP_GT_I64, i = (a > b) ? 1 : 0)
@@ -680,6 +682,24 @@ struct KeyArg {
   int32_t nkeys;
 };
 
+
+__device__ __forceinline__ uint64_t pack_key(const KeyArg& K, const ColsArg& C,
+                                             int64_t r) {
+  uint64_t code = 0;
+  for (int j = 0; j < K.nkeys; j++) {
+    Slot v;
+    bool valid;
+    vm_load_col(C, K.k[j].col, r, v, valid);
+    uint64_t part;
+    if (K.k[j].nullable)
+      part = valid ? (uint64_t)(v.i - K.k[j].min) + 1 : 0;
+    else
+      part = (uint64_t)(v.i - K.k[j].min);
+    code += part * K.stride[j];
+  }
+  return code;
+}
+
 __global__ void k_keypack(KeyArg K, ColsArg C, int64_t n, uint64_t* out) {
   int64_t lo, hi;
   block_range(n, 1, lo, hi);
@@ -1027,7 +1047,7 @@ __device__ __forceinline__ uint64_t agg_identity(int op) {
 // LDS layout: [key_space × naggs] u64 vals, [key_space × naggs] u32 cnts,
 // [key_space] u32 gcnt. Host guarantees it fits (≤ LDS budget).
 __global__ void __launch_bounds__(BLOCK)
-k_groupby_direct(ColsArg C, int64_t n, const uint64_t* key_codes,
+k_groupby_direct(ColsArg C, int64_t n, KeyArg K,
                  int key_space, ProgArg pred, const DsxInstr* agg_progs,
                  const int32_t* agg_lens, AggArg A,
                  uint64_t* g_vals /*[naggs][key_space]*/,
@@ -1052,7 +1072,7 @@ k_groupby_direct(ColsArg C, int64_t n, const uint64_t* key_codes,
       bool pvalid = vm_eval(pred.ins, pred.len, C, r, pv);
       if (!(pvalid && pv.i != 0)) continue;  // NULL→False (filter.py:39)
     }
-    int k = (int)key_codes[r];
+    int k = (int)pack_key(K, C, r);
     atomicAdd(&s_gcnt[k], 1u);
     const DsxInstr* p = agg_progs;
     for (int a = 0; a < A.naggs; a++) {
@@ -1133,7 +1153,7 @@ k_groupby_direct(ColsArg C, int64_t n, const uint64_t* key_codes,
 }
 
 // ---- global CAS-claim path --------------------------------------------------
-__global__ void k_groupby_global(ColsArg C, int64_t n, const uint64_t* key_codes,
+__global__ void k_groupby_global(ColsArg C, int64_t n, KeyArg K,
                                  ProgArg pred, const DsxInstr* agg_progs,
                                  const int32_t* agg_lens, AggArg A,
                                  uint64_t* tkeys, int64_t mask,
@@ -1150,7 +1170,7 @@ __global__ void k_groupby_global(ColsArg C, int64_t n, const uint64_t* key_codes
       bool pvalid = vm_eval(pred.ins, pred.len, C, r, pv);
       if (!(pvalid && pv.i != 0)) continue;
     }
-    uint64_t cde = key_codes[r];
+    uint64_t cde = pack_key(K, C, r);
     int64_t s = (int64_t)(mix64(cde) & mask);
     int64_t probes = 0;
     while (true) {
@@ -1236,13 +1256,25 @@ __global__ void k_count_live(const uint64_t* tkeys,
 }
 
 extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
-                                int64_t n, const uint64_t* key_codes,
-                                uint64_t key_space, const DsxInstr* pred,
+                                int64_t n, const DsxKeySpec* keys, int nkeys,
+                                const DsxInstr* pred,
                                 int pred_len, const DsxAggSpec* aggs, int naggs,
                                 uint64_t** out_codes, void** out_vals,
                                 uint64_t** out_counts, int64_t* out_groups) {
-  if (naggs > DSX_MAX_AGGS || ncols > DSX_MAX_COLS || pred_len > DSX_MAX_PROG)
+  if (naggs > DSX_MAX_AGGS || ncols > DSX_MAX_COLS || pred_len > DSX_MAX_PROG ||
+      nkeys > DSX_MAX_KEYS)
     FAIL(-3, "groupby spec too large");
+  KeyArg K{};
+  K.nkeys = nkeys;
+  uint64_t key_space = 1;
+  for (int j = 0; j < nkeys; j++) {
+    K.k[j] = keys[j];
+    K.stride[j] = key_space;
+    uint64_t range = (uint64_t)keys[j].range + (keys[j].nullable ? 1 : 0);
+    if (range == 0) FAIL(-3, "empty key range");
+    if (key_space > (1ull << 62) / range) FAIL(-4, "key space exceeds 2^62");
+    key_space *= range;
+  }
   ColsArg C{};
   C.ncols = ncols;
   for (int i = 0; i < ncols; i++) {
@@ -1340,13 +1372,13 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
         (void)lds;
         ProfScope ps(c, "k_groupby_direct");
         hipLaunchKernelGGL(k_groupby_direct, dim3(grid), dim3(BLOCK),
-                           lds_bytes, c->stream, C, n, key_codes,
+                           lds_bytes, c->stream, C, n, K,
                            (int)key_space, P, d_progs, d_lens, A, d_vals,
                            d_cnts, d_gcnt);
       } else {
         ProfScope ps(c, "k_groupby_global");
         hipLaunchKernelGGL(k_groupby_global, dim3(grid), dim3(BLOCK), 0,
-                           c->stream, C, n, key_codes, P, d_progs, d_lens, A,
+                           c->stream, C, n, K, P, d_progs, d_lens, A,
                            d_tkeys, slots - 1, d_vals, d_cnts, d_gcnt, d_ovf);
       }
     }

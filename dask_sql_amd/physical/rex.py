@@ -1,0 +1,265 @@
+"""Rex → VM-program compiler.
+
+Replaces the reference RexConverter dispatch + RexCallPlugin evaluation
+(physical/rex/convert.py:47-76, rex/core/call.py:1047-1156): instead of
+per-operator pandas calls, an Expression tree compiles into one typed postfix
+program (include/dsxhip.h DsxOp) interpreted per row inside the HIP kernels.
+"""
+from __future__ import annotations
+
+from dask_sql_amd.planner.plan import AggCall, Call, Expression, InputRef, Literal
+from dask_sql_amd import runtime as rt
+
+# opcodes (include/dsxhip.h)
+OP_COL, OP_LIT_F64, OP_LIT_I64, OP_LIT_NULL = 1, 2, 3, 4
+OP_ADD_F64, OP_SUB_F64, OP_MUL_F64, OP_DIV_F64 = 10, 11, 12, 13
+OP_ADD_I64, OP_SUB_I64, OP_MUL_I64, OP_DIV_I64, OP_MOD_I64 = 14, 15, 16, 17, 18
+OP_LT_F64, OP_LE_F64, OP_GT_F64, OP_GE_F64, OP_EQ_F64, OP_NE_F64 = (
+    20, 21, 22, 23, 24, 25)
+OP_LT_I64, OP_LE_I64, OP_GT_I64, OP_GE_I64, OP_EQ_I64, OP_NE_I64 = (
+    30, 31, 32, 33, 34, 35)
+OP_AND, OP_OR, OP_NOT, OP_IS_NULL, OP_IS_NOT_NULL = 40, 41, 42, 43, 44
+OP_I64_TO_F64, OP_F64_TO_I64 = 50, 51
+OP_SELECT, OP_NEG_F64, OP_NEG_I64 = 60, 61, 62
+
+# VM value kinds
+KI, KF, KB = "i", "f", "b"  # int64-like, float64, boolean
+
+_SQL_TO_KIND = {
+    "BIGINT": KI, "INTEGER": KI, "SMALLINT": KI, "TINYINT": KI, "DATE": KI,
+    "TIMESTAMP": KI, "VARCHAR": KI,  # dict codes
+    "DOUBLE": KF, "FLOAT": KF, "DECIMAL": KF,
+    "BOOLEAN": KB, "NULL": KI,
+}
+
+_DSX_KIND = {rt.I64: KI, rt.I32: KI, rt.I8: KI, rt.BOOL8: KB,
+             rt.F64: KF, rt.F32: KF}
+
+_CMP = {"=": (OP_EQ_I64, OP_EQ_F64), "<>": (OP_NE_I64, OP_NE_F64),
+        "<": (OP_LT_I64, OP_LT_F64), "<=": (OP_LE_I64, OP_LE_F64),
+        ">": (OP_GT_I64, OP_GT_F64), ">=": (OP_GE_I64, OP_GE_F64)}
+_ARITH = {"+": (OP_ADD_I64, OP_ADD_F64), "-": (OP_SUB_I64, OP_SUB_F64),
+          "*": (OP_MUL_I64, OP_MUL_F64), "/": (OP_DIV_I64, OP_DIV_F64)}
+
+
+class RexCompileError(NotImplementedError):
+    pass
+
+
+class RexCompiler:
+    """Compiles an Expression over a column scope.
+
+    cols: list of DeviceColumn in frontend order (InputRef index order);
+    dictionaries: optional per-column list of string dictionaries (for
+    dict-encoded VARCHAR equality — SURVEY §8f2 dict-code compare).
+    """
+
+    def __init__(self, cols, dictionaries=None):
+        self.cols = cols
+        self.dicts = dictionaries or [None] * len(cols)
+        self.prog: list[tuple] = []
+
+    # ---- emit helpers -----------------------------------------------------
+    def _emit(self, op, arg0=0, imm=0):
+        self.prog.append((op, arg0, imm))
+
+    def _to_f(self, kind):
+        if kind == KF:
+            return KF
+        self._emit(OP_I64_TO_F64)
+        return KF
+
+    # ---- main -------------------------------------------------------------
+    def compile(self, expr: Expression) -> str:
+        """Appends code; returns result kind."""
+        if isinstance(expr, InputRef):
+            i = expr.getIndex()
+            self._emit(OP_COL, i)
+            return _DSX_KIND[self.cols[i].dtype]
+        if isinstance(expr, Literal):
+            v = expr.getValue()
+            if v is None:
+                self._emit(OP_LIT_NULL)
+                return KI
+            t = expr.getType().getSqlType() if expr.getType() else None
+            if isinstance(v, bool):
+                self._emit(OP_LIT_I64, 0, 1 if v else 0)
+                return KB
+            if isinstance(v, float):
+                self._emit(OP_LIT_F64, 0, float(v))
+                return KF
+            if isinstance(v, str):
+                raise RexCompileError(
+                    "string literal outside dict-compare context")
+            self._emit(OP_LIT_I64, 0, int(v))
+            return KI
+        if isinstance(expr, Call):
+            return self._compile_call(expr)
+        raise RexCompileError(f"cannot compile {expr!r}")
+
+    def _compile_call(self, call: Call) -> str:
+        op = call.getOperatorName()
+        ops = call.getOperands()
+        if op in _CMP:
+            return self._compile_cmp(op, ops)
+        if op in _ARITH:
+            a, b = ops
+            # pre-scan kinds to decide int vs float path
+            ka = self._peek_kind(a)
+            kb = self._peek_kind(b)
+            if KF in (ka, kb) or op == "/" and False:
+                k = self.compile(a)
+                self._to_f(k)
+                k = self.compile(b)
+                self._to_f(k)
+                self._emit(_ARITH[op][1])
+                return KF
+            self.compile(a)
+            self.compile(b)
+            self._emit(_ARITH[op][0])
+            return KI
+        if op == "AND":
+            self.compile(ops[0])
+            self.compile(ops[1])
+            self._emit(OP_AND)
+            return KB
+        if op == "OR":
+            self.compile(ops[0])
+            self.compile(ops[1])
+            self._emit(OP_OR)
+            return KB
+        if op == "NOT":
+            self.compile(ops[0])
+            self._emit(OP_NOT)
+            return KB
+        if op == "IS NULL":
+            self.compile(ops[0])
+            self._emit(OP_IS_NULL)
+            return KB
+        if op == "IS NOT NULL":
+            self.compile(ops[0])
+            self._emit(OP_IS_NOT_NULL)
+            return KB
+        if op == "NEG":
+            k = self.compile(ops[0])
+            self._emit(OP_NEG_F64 if k == KF else OP_NEG_I64)
+            return k
+        if op == "CAST":
+            target = call.getType().getSqlType()
+            k = self.compile(ops[0])
+            tk = _SQL_TO_KIND.get(target, KF)
+            if tk == KF and k != KF:
+                self._emit(OP_I64_TO_F64)
+                return KF
+            if tk in (KI, KB) and k == KF:
+                self._emit(OP_F64_TO_I64)  # trunc (mappings.py:346-353)
+                return KI
+            return k
+        if op == "CASE":
+            # operands: cond1, val1, cond2, val2, ..., else
+            return self._compile_case(ops)
+        raise RexCompileError(f"operator {op} not supported on GPU path")
+
+    def _compile_case(self, ops):
+        # rightmost-else first; build nested SELECTs. Postfix SELECT pops
+        # (cond, a, b). Emit conds/vals in order with SELECT folds from the
+        # back: CASE c1 v1 c2 v2 e == SELECT(c1, v1, SELECT(c2, v2, e))
+        *pairs, els = ops
+        assert len(pairs) % 2 == 0
+
+        def emit_chain(i):
+            if i >= len(pairs):
+                k = self.compile(els)
+                return k
+            self.compile(pairs[i])        # cond
+            kv = self.compile(pairs[i + 1])  # val
+            ke = emit_chain(i + 2)
+            if kv == KF or ke == KF:
+                pass  # mixed kinds: caller ensured same family (planner types)
+            self._emit(OP_SELECT)
+            return kv
+
+        return emit_chain(0)
+
+    def _compile_cmp(self, op, ops) -> str:
+        a, b = ops
+        # dict-encoded string compare: col vs string literal
+        lit, col = None, None
+        if isinstance(a, InputRef) and isinstance(b, Literal) \
+                and isinstance(b.getValue(), str):
+            col, lit = a, b
+        elif isinstance(b, InputRef) and isinstance(a, Literal) \
+                and isinstance(a.getValue(), str):
+            col, lit = b, a
+            op = {"<": ">", ">": "<", "<=": ">=", ">=": "<="}.get(op, op)
+        if col is not None:
+            if op not in ("=", "<>"):
+                raise RexCompileError("only =/<> on dict-encoded strings")
+            d = self.dicts[col.getIndex()]
+            if d is None:
+                raise RexCompileError("string compare on non-dict column")
+            code = d.index(lit.getValue()) if lit.getValue() in d else -2
+            self._emit(OP_COL, col.getIndex())
+            self._emit(OP_LIT_I64, 0, code)
+            self._emit(_CMP[op][0])
+            return KB
+        ka = self._peek_kind(a)
+        kb = self._peek_kind(b)
+        if KF in (ka, kb):
+            k = self.compile(a)
+            self._to_f(k)
+            k = self.compile(b)
+            self._to_f(k)
+            self._emit(_CMP[op][1])
+        else:
+            self.compile(a)
+            self.compile(b)
+            self._emit(_CMP[op][0])
+        return KB
+
+    def _peek_kind(self, expr) -> str:
+        """Result kind without emitting (cheap recursive type-check)."""
+        if isinstance(expr, InputRef):
+            return _DSX_KIND[self.cols[expr.getIndex()].dtype]
+        if isinstance(expr, Literal):
+            v = expr.getValue()
+            if isinstance(v, bool):
+                return KB
+            if isinstance(v, float):
+                return KF
+            return KI
+        if isinstance(expr, Call):
+            op = expr.getOperatorName()
+            if op in _CMP or op in ("AND", "OR", "NOT", "IS NULL",
+                                    "IS NOT NULL"):
+                return KB
+            if op in _ARITH:
+                ka = self._peek_kind(expr.getOperands()[0])
+                kb = self._peek_kind(expr.getOperands()[1])
+                return KF if KF in (ka, kb) else KI
+            if op == "CAST":
+                return _SQL_TO_KIND.get(expr.getType().getSqlType(), KF)
+            if op == "NEG":
+                return self._peek_kind(expr.getOperands()[0])
+            if op == "CASE":
+                return self._peek_kind(expr.getOperands()[1])
+        return KF
+
+
+def compile_expr(expr, cols, dictionaries=None):
+    """Returns (prog_tuple (for Runtime.make_prog), result_kind)."""
+    c = RexCompiler(cols, dictionaries)
+    kind = c.compile(expr)
+    return c.prog, kind
+
+
+def scalar_literal(expr):
+    """If expr is a scalar boolean literal, return its Python value
+    (filter_or_scalar short-circuit, reference filter.py:31-36)."""
+    if isinstance(expr, Literal):
+        v = expr.getValue()
+        if isinstance(v, bool) or v is None:
+            return bool(v) if v is not None else False
+        if isinstance(v, (int, float)):
+            return bool(v)
+    return None

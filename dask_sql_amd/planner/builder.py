@@ -1,0 +1,599 @@
+"""AST → LogicalPlan builder (name resolution, join shaping, aggregate
+extraction). Produces the plan shapes the physical plugins consume
+(SURVEY.md §8b); semantics pinned end-to-end by the golden tests.
+
+Mirrors, functionally, what the reference gets from DataFusion SqlToRel +
+the optimizer rule subset that matters on this path (PushDownFilter /
+EliminateCrossJoin — src/sql/optimizer.rs:53-98): single-table conjuncts are
+pushed below joins, comma-joins with WHERE equalities become inner joins.
+"""
+from __future__ import annotations
+
+import numpy as np
+
+from dask_sql_amd.planner.parser import SelectStmt, TableRef, parse_sql
+from dask_sql_amd.planner.plan import (
+    AggCall, AggregateNode, Call, Expression, Field, FilterNode, InputRef,
+    JoinNode, Literal, LimitNode, LogicalPlan, ProjectionNode, RelDataType,
+    SortNode, SqlType, TableScanNode,
+)
+
+_EPOCH = np.datetime64("1970-01-01")
+
+
+def _date_to_days(s: str) -> int:
+    return int((np.datetime64(s) - _EPOCH).astype(int))
+
+
+_NUMERIC_RANK = {"TINYINT": 0, "SMALLINT": 1, "INTEGER": 2, "BIGINT": 3,
+                 "DATE": 3, "FLOAT": 4, "DOUBLE": 5}
+
+
+def _is_float(t: str) -> bool:
+    return t in ("FLOAT", "DOUBLE")
+
+
+def _common_type(a: str, b: str) -> str:
+    if a == b:
+        return a
+    if a == "VARCHAR" or b == "VARCHAR":
+        return "VARCHAR"
+    ra, rb = _NUMERIC_RANK.get(a, 5), _NUMERIC_RANK.get(b, 5)
+    return a if ra >= rb else b
+
+
+class Catalog:
+    """What the builder needs from the Context's schema: table → fields."""
+
+    def __init__(self):
+        self.tables: dict[str, list[tuple[str, str]]] = {}
+
+    def add(self, name, fields):
+        self.tables[name.lower()] = fields
+
+    def drop(self, name):
+        self.tables.pop(name.lower(), None)
+
+    def get(self, name):
+        key = name.lower()
+        if key not in self.tables:
+            raise KeyError(f"Table {name!r} not registered")
+        return self.tables[key]
+
+
+def _expr_type(e: Expression) -> str:
+    t = e.getType()
+    return t.getSqlType() if t else "DOUBLE"
+
+
+class Builder:
+    def __init__(self, catalog: Catalog, schema_name: str = "root"):
+        self.catalog = catalog
+        self.schema_name = schema_name
+
+    # ------------------------------------------------------------------ API
+    def build(self, sql: str) -> LogicalPlan:
+        stmt = parse_sql(sql)
+        return self.build_stmt(stmt)
+
+    # ----------------------------------------------------------------- scans
+    def _scan(self, tr: TableRef) -> LogicalPlan:
+        fields_spec = self.catalog.get(tr.name)
+        qual = (tr.alias or tr.name).lower()
+        fields = [Field(n, SqlType(t), qualifier=qual) for n, t in fields_spec]
+        node = TableScanNode(self.schema_name, tr.name.lower())
+        return LogicalPlan("TableScan", [], RelDataType(fields), node)
+
+    # ------------------------------------------------------- expr resolution
+    def _resolve(self, ast, plan: LogicalPlan) -> Expression:
+        fields = plan.getRowType().getFieldList()
+        kind = ast[0]
+        if kind == "col":
+            _, q, n = ast
+            matches = [
+                i for i, f in enumerate(fields)
+                if f.getName().lower() == n.lower()
+                and (q is None or (f.qualifier or "").lower() == q.lower())
+            ]
+            if not matches:
+                # also match already-qualified output names like "lhs.a"
+                full = f"{q}.{n}".lower() if q else n.lower()
+                matches = [i for i, f in enumerate(fields)
+                           if f.getName().lower() == full]
+            if not matches:
+                raise KeyError(f"column {q + '.' if q else ''}{n} not found in "
+                               f"{[f.getQualifiedName() for f in fields]}")
+            if len(matches) > 1 and q is None:
+                raise KeyError(f"ambiguous column {n}")
+            i = matches[0]
+            return InputRef(i, fields[i].getType())
+        if kind == "lit":
+            _, v, t = ast
+            if t == "DATE":
+                return Literal(_date_to_days(v), SqlType("DATE"))
+            if t == "NULL":
+                return Literal(None, SqlType("NULL"))
+            return Literal(v, SqlType(t))
+        if kind == "cast":
+            _, sub, ty = ast
+            e = self._resolve(sub, plan)
+            return Call("CAST", [e], SqlType(ty))
+        if kind == "case":
+            _, whens, els = ast
+            ops = []
+            for cond, val in whens:
+                ops.append(self._resolve(cond, plan))
+                ops.append(self._resolve(val, plan))
+            ops.append(self._resolve(els, plan) if els is not None
+                       else Literal(None, SqlType("NULL")))
+            ty = _expr_type(ops[1])
+            return Call("CASE", ops, SqlType(ty))
+        if kind == "call":
+            _, op, args = ast
+            ops = [self._resolve(a, plan) for a in args]
+            if op in ("=", "<>", "<", "<=", ">", ">=", "AND", "OR", "NOT",
+                      "IS NULL", "IS NOT NULL", "LIKE"):
+                ty = "BOOLEAN"
+            elif op == "NEG":
+                ty = _expr_type(ops[0])
+            elif op == "/":
+                ty = _common_type(_expr_type(ops[0]), _expr_type(ops[1]))
+            else:
+                ty = _common_type(_expr_type(ops[0]),
+                                  _expr_type(ops[1]) if len(ops) > 1
+                                  else _expr_type(ops[0]))
+            return Call(op, ops, SqlType(ty))
+        if kind == "agg":
+            raise ValueError("aggregate in non-aggregate position")
+        raise ValueError(f"cannot resolve {ast!r}")
+
+    # -------------------------------------------------------- ast utilities
+    @staticmethod
+    def _conjuncts(ast):
+        if ast is None:
+            return []
+        if ast[0] == "call" and ast[1] == "AND":
+            out = []
+            for a in ast[2]:
+                out.extend(Builder._conjuncts(a))
+            return out
+        return [ast]
+
+    @staticmethod
+    def _tables_of(ast, out=None):
+        """qualifiers/col names referenced by an AST."""
+        if out is None:
+            out = []
+        if not isinstance(ast, tuple):
+            return out
+        if ast[0] == "col":
+            out.append((ast[1], ast[2]))
+            return out
+        if ast[0] == "call":
+            for a in ast[2]:
+                Builder._tables_of(a, out)
+        elif ast[0] == "agg":
+            for a in ast[2]:
+                Builder._tables_of(a, out)
+            if ast[4] is not None:
+                Builder._tables_of(ast[4], out)
+        elif ast[0] == "cast":
+            Builder._tables_of(ast[1], out)
+        elif ast[0] == "case":
+            for c, v in ast[1]:
+                Builder._tables_of(c, out)
+                Builder._tables_of(v, out)
+            if ast[2] is not None:
+                Builder._tables_of(ast[2], out)
+        return out
+
+    def _refs_only(self, ast, quals: set, plan) -> bool:
+        """every column in ast resolvable within plan (whose fields carry
+        qualifiers in quals)"""
+        for q, n in self._tables_of(ast):
+            found = False
+            for f in plan.getRowType().getFieldList():
+                if f.getName().lower() == n.lower() and (
+                    q is None or (f.qualifier or "").lower() == q.lower()
+                ):
+                    found = True
+                    break
+            if not found:
+                return False
+        return True
+
+    @staticmethod
+    def _has_agg(ast) -> bool:
+        if not isinstance(ast, tuple):
+            return False
+        if ast[0] == "agg":
+            return True
+        if ast[0] == "call":
+            return any(Builder._has_agg(a) for a in ast[2])
+        if ast[0] == "cast":
+            return Builder._has_agg(ast[1])
+        if ast[0] == "case":
+            return any(
+                Builder._has_agg(c) or Builder._has_agg(v) for c, v in ast[1]
+            ) or (ast[2] is not None and Builder._has_agg(ast[2]))
+        return False
+
+    # ------------------------------------------------------------- pipeline
+    def build_stmt(self, stmt: SelectStmt) -> LogicalPlan:
+        where_conjuncts = self._conjuncts(stmt.where)
+        used = [False] * len(where_conjuncts)
+
+        # 1. scans (+ pushed-down single-table filters, à la PushDownFilter)
+        def scan_with_filters(tr: TableRef) -> LogicalPlan:
+            plan = self._scan(tr)
+            quals = {(tr.alias or tr.name).lower()}
+            conds = []
+            for i, cj in enumerate(where_conjuncts):
+                if used[i] or self._has_agg(cj):
+                    continue
+                refs = self._tables_of(cj)
+                if not refs:
+                    continue  # scalar conditions stay global
+                if self._refs_only(cj, quals, plan):
+                    conds.append(cj)
+                    used[i] = True
+            for cj in conds:
+                cond = self._resolve(cj, plan)
+                plan = LogicalPlan("Filter", [plan], plan.getRowType(),
+                                   FilterNode(cond))
+            return plan
+
+        tables = list(stmt.from_tables)
+        plan = scan_with_filters(tables[0]) if tables else None
+
+        def join_plans(lhs, rhs, join_type, cond_ast_list, on_expr_ast):
+            lhs_fields = lhs.getRowType().getFieldList()
+            rhs_fields = rhs.getRowType().getFieldList()
+            combined = RelDataType(lhs_fields + rhs_fields)
+            tmp = LogicalPlan("__combined__", [], combined, None)
+            cond = None
+            if on_expr_ast is not None:
+                cond = self._resolve(on_expr_ast, tmp)
+            elif cond_ast_list:
+                ast = cond_ast_list[0]
+                for c in cond_ast_list[1:]:
+                    ast = ("call", "AND", [ast, c])
+                cond = self._resolve(ast, tmp)
+            if join_type in ("LEFTSEMI", "LEFTANTI"):
+                out_fields = lhs_fields
+            else:
+                out_fields = lhs_fields + rhs_fields
+            return LogicalPlan("Join", [lhs, rhs], RelDataType(out_fields),
+                               JoinNode(join_type, cond))
+
+        # comma-joined tables: EliminateCrossJoin — find WHERE equalities
+        for tr in tables[1:]:
+            rhs = scan_with_filters(tr)
+            lhs_fields = plan.getRowType().getFieldList()
+            combined = RelDataType(lhs_fields + rhs.getRowType().getFieldList())
+            tmp = LogicalPlan("__combined__", [], combined, None)
+            conds = []
+            for i, cj in enumerate(where_conjuncts):
+                if used[i] or self._has_agg(cj):
+                    continue
+                refs = self._tables_of(cj)
+                if not refs:
+                    continue
+                if self._refs_only(cj, set(), tmp) and not self._refs_only(
+                    cj, set(), plan
+                ):
+                    conds.append(cj)
+                    used[i] = True
+            plan = join_plans(plan, rhs, "INNER" if conds else "CROSS",
+                              conds, None)
+
+        # explicit JOIN clauses
+        for jc in stmt.joins:
+            rhs = scan_with_filters(jc.table)
+            plan = join_plans(plan, rhs, jc.join_type, [], jc.on)
+
+        # leftover WHERE conjuncts → Filter (incl. scalar TRUE/FALSE)
+        leftovers = [cj for i, cj in enumerate(where_conjuncts) if not used[i]
+                     and not self._has_agg(cj)]
+        if leftovers:
+            ast = leftovers[0]
+            for c in leftovers[1:]:
+                ast = ("call", "AND", [ast, c])
+            cond = self._resolve(ast, plan)
+            plan = LogicalPlan("Filter", [plan], plan.getRowType(),
+                               FilterNode(cond))
+
+        # 2. expand stars
+        items = []
+        for e, alias in stmt.items:
+            if e == ("star",):
+                for f in plan.getRowType().getFieldList():
+                    items.append((("col", f.qualifier, f.getName()), None))
+            else:
+                items.append((e, alias))
+
+        # 3. aggregate?
+        has_agg = any(self._has_agg(e) for e, _ in items) or bool(
+            stmt.group_by
+        ) or (stmt.having is not None and self._has_agg(stmt.having))
+        if has_agg:
+            plan, items = self._build_aggregate(stmt, plan, items)
+        # HAVING without aggregate context is just a filter
+        elif stmt.having is not None:
+            cond = self._resolve(stmt.having, plan)
+            plan = LogicalPlan("Filter", [plan], plan.getRowType(),
+                               FilterNode(cond))
+
+        # 4. final projection
+        plan = self._build_projection(plan, items)
+
+        # 5. DISTINCT
+        if stmt.distinct:
+            fields = plan.getRowType().getFieldList()
+            group_exprs = [InputRef(i, f.getType()) for i, f in enumerate(fields)]
+            node = AggregateNode(group_exprs, [], distinct_node=True,
+                                 distinct_columns=[f.getName() for f in fields])
+            plan = LogicalPlan("Distinct", [plan], plan.getRowType(), node)
+
+        # 6. ORDER BY / LIMIT
+        if stmt.order_by:
+            keys = []
+            for e, asc, nf in stmt.order_by:
+                idx = self._find_output(e, stmt, plan)
+                keys.append((idx, asc, nf))
+            plan = LogicalPlan("Sort", [plan], plan.getRowType(),
+                               SortNode(keys))
+        if stmt.limit is not None or stmt.offset:
+            plan = LogicalPlan("Limit", [plan], plan.getRowType(),
+                               LimitNode(stmt.limit, stmt.offset))
+        return plan
+
+    # ------------------------------------------------------------ aggregate
+    def _build_aggregate(self, stmt, plan, items):
+        # collect group exprs and agg calls (dedup by AST)
+        group_asts = list(stmt.group_by)
+        agg_asts = []
+
+        def collect_aggs(ast):
+            if not isinstance(ast, tuple):
+                return
+            if ast[0] == "agg":
+                if ast not in agg_asts:
+                    agg_asts.append(ast)
+                return
+            if ast[0] == "call":
+                for a in ast[2]:
+                    collect_aggs(a)
+            elif ast[0] == "cast":
+                collect_aggs(ast[1])
+            elif ast[0] == "case":
+                for c, v in ast[1]:
+                    collect_aggs(c)
+                    collect_aggs(v)
+                if ast[2] is not None:
+                    collect_aggs(ast[2])
+
+        for e, _ in items:
+            collect_aggs(e)
+        if stmt.having is not None:
+            collect_aggs(stmt.having)
+        for e, _, _ in stmt.order_by:
+            collect_aggs(e)
+
+        # GROUP BY items may be aliases of select items or positions
+        resolved_groups = []
+        for g in group_asts:
+            if g[0] == "lit" and isinstance(g[1], int):
+                e, _ = items[g[1] - 1]
+                resolved_groups.append(e)
+            elif g[0] == "col" and g[1] is None:
+                # alias of a select item?
+                hit = None
+                for e, alias in items:
+                    if alias and alias.lower() == g[2].lower():
+                        hit = e
+                        break
+                try:
+                    self._resolve(g, plan)
+                    hit = None  # real column wins
+                except KeyError:
+                    pass
+                resolved_groups.append(hit if hit is not None else g)
+            else:
+                resolved_groups.append(g)
+
+        # pre-projection: group exprs, agg args, agg filters
+        pre_named = []  # (Expression, name)
+        pre_fields = []
+        name_of = {}
+
+        def add_pre(ast, base_name):
+            key = repr(ast)
+            if key in name_of:
+                return name_of[key]
+            e = self._resolve(ast, plan)
+            name = base_name
+            k = 1
+            existing = {n for _, n in pre_named}
+            while name in existing:
+                name = f"{base_name}_{k}"
+                k += 1
+            pre_named.append((e, name))
+            src_q = None
+            if ast[0] == "col":
+                i = e.getIndex() if isinstance(e, InputRef) else None
+                if i is not None:
+                    src_q = plan.getRowType().getFieldList()[i].qualifier
+            pre_fields.append(Field(name, SqlType(_expr_type(e)),
+                                    qualifier=src_q))
+            name_of[key] = len(pre_named) - 1
+            return name_of[key]
+
+        def base_name_for(ast):
+            if ast[0] == "col":
+                return ast[2]
+            return f"expr{len(pre_named)}"
+
+        group_idx = [add_pre(g, base_name_for(g)) for g in resolved_groups]
+
+        agg_info = []  # (ast, func, arg_idx|None, filter_idx|None, distinct)
+        for ast in agg_asts:
+            _, func, args, distinct, filt = ast
+            if len(args) == 1 and args[0] == ("star",):
+                arg_idx = None  # COUNT(*)
+            elif len(args) == 1:
+                arg_idx = add_pre(args[0], base_name_for(args[0]))
+            elif len(args) == 0:
+                arg_idx = None
+            else:
+                raise NotImplementedError("multi-arg aggregates")
+            filt_idx = add_pre(filt, f"filter{len(pre_named)}") \
+                if filt is not None else None
+            agg_info.append((ast, func, arg_idx, filt_idx, distinct))
+
+        pre_plan = LogicalPlan("Projection", [plan], RelDataType(pre_fields),
+                               ProjectionNode(pre_named))
+
+        # aggregate node
+        group_exprs = [InputRef(i, pre_fields[i].getType()) for i in group_idx]
+        agg_calls = []
+        out_fields = [pre_fields[i] for i in group_idx]
+        agg_out_of = {}
+        for ast, func, arg_idx, filt_idx, distinct in agg_info:
+            if arg_idx is not None:
+                args = [InputRef(arg_idx, pre_fields[arg_idx].getType())]
+                arg_t = pre_fields[arg_idx].getType().getSqlType()
+            else:
+                args = []
+                arg_t = "BIGINT"
+            if func == "count":
+                out_t = "BIGINT"
+            elif func == "avg":
+                out_t = "DOUBLE"
+            elif func == "sum":
+                out_t = "DOUBLE" if _is_float(arg_t) else "BIGINT"
+            else:
+                out_t = arg_t
+            out_name = f"{func.upper()}({pre_fields[arg_idx].getName()})" \
+                if arg_idx is not None else f"{func.upper()}(*)"
+            k = 1
+            existing = {f.getName() for f in out_fields}
+            base = out_name
+            while out_name in existing:
+                out_name = f"{base}_{k}"
+                k += 1
+            filt_e = InputRef(filt_idx, pre_fields[filt_idx].getType()) \
+                if filt_idx is not None else None
+            agg_calls.append(AggCall(func, args, out_name, filt_e, distinct))
+            out_fields.append(Field(out_name, SqlType(out_t)))
+            agg_out_of[repr(ast)] = len(out_fields) - 1
+
+        agg_node = AggregateNode(group_exprs, agg_calls)
+        agg_plan = LogicalPlan("Aggregate", [pre_plan],
+                               RelDataType(out_fields), agg_node)
+
+        # rewrite select items over aggregate output
+        group_out_of = {repr(g): gi for gi, g in enumerate(resolved_groups)}
+
+        def rewrite(ast):
+            if repr(ast) in group_out_of:
+                i = group_out_of[repr(ast)]
+                return ("col", None, out_fields[i].getName())
+            if isinstance(ast, tuple) and ast[0] == "agg":
+                i = agg_out_of[repr(ast)]
+                return ("col", None, out_fields[i].getName())
+            if isinstance(ast, tuple) and ast[0] == "call":
+                return ("call", ast[1], [rewrite(a) for a in ast[2]])
+            if isinstance(ast, tuple) and ast[0] == "cast":
+                return ("cast", rewrite(ast[1]), ast[2])
+            if isinstance(ast, tuple) and ast[0] == "case":
+                return ("case",
+                        [(rewrite(c), rewrite(v)) for c, v in ast[1]],
+                        rewrite(ast[2]) if ast[2] is not None else None)
+            return ast
+
+        new_items = [(rewrite(e), alias) for e, alias in items]
+
+        # HAVING → filter over aggregate output
+        if stmt.having is not None:
+            cond = self._resolve(rewrite(stmt.having), agg_plan)
+            agg_plan = LogicalPlan("Filter", [agg_plan], agg_plan.getRowType(),
+                                   FilterNode(cond))
+
+        # stash rewrite for ORDER BY resolution
+        self._agg_rewrite = rewrite
+        return agg_plan, new_items
+
+    # ----------------------------------------------------------- projection
+    def _build_projection(self, plan, items):
+        named = []
+        fields = []
+        # derive names, disambiguating duplicates with qualifiers
+        # (context.py:890-898)
+        base_names = []
+        for e, alias in items:
+            if alias:
+                base_names.append(alias)
+            elif e[0] == "col":
+                base_names.append(e[2])
+            elif e[0] == "agg":
+                base_names.append(f"{e[1].upper()}()")
+            else:
+                base_names.append(None)
+        counts = {}
+        for n in base_names:
+            if n is not None:
+                counts[n.lower()] = counts.get(n.lower(), 0) + 1
+        out_names = []
+        for (e, alias), bn in zip(items, base_names):
+            if alias:
+                out_names.append(alias)
+            elif bn is not None and counts.get(bn.lower(), 0) > 1 \
+                    and e[0] == "col" and e[1] is not None:
+                out_names.append(f"{e[1]}.{e[2]}")
+            elif bn is not None:
+                out_names.append(bn)
+            else:
+                out_names.append(f"EXPR${len(out_names)}")
+        # a second pass: if still duplicated (both unqualified), qualify by
+        # source field qualifier
+        seen = {}
+        for i, n in enumerate(out_names):
+            seen.setdefault(n.lower(), []).append(i)
+        for n, idxs in seen.items():
+            if len(idxs) > 1:
+                for i in idxs:
+                    e, alias = items[i]
+                    if alias is None and e[0] == "col":
+                        ref = self._resolve(e, plan)
+                        if isinstance(ref, InputRef):
+                            f = plan.getRowType().getFieldList()[ref.getIndex()]
+                            if f.qualifier:
+                                out_names[i] = f"{f.qualifier}.{f.getName()}"
+
+        for (e, alias), name in zip(items, out_names):
+            expr = self._resolve(e, plan)
+            named.append((expr, name))
+            src_q = None
+            if isinstance(expr, InputRef):
+                src_q = plan.getRowType().getFieldList()[expr.getIndex()].qualifier
+            fields.append(Field(name, SqlType(_expr_type(expr)),
+                                qualifier=src_q))
+        return LogicalPlan("Projection", [plan], RelDataType(fields),
+                           ProjectionNode(named))
+
+    def _find_output(self, ast, stmt, plan) -> int:
+        fields = plan.getRowType().getFieldList()
+        if ast[0] == "lit" and isinstance(ast[1], int):
+            return ast[1] - 1
+        if hasattr(self, "_agg_rewrite"):
+            ast = self._agg_rewrite(ast)
+        if ast[0] == "col" and ast[1] is None:
+            for i, f in enumerate(fields):
+                if f.getName().lower() == ast[2].lower():
+                    return i
+        # structural match against select items
+        for i, (e, alias) in enumerate(stmt.items):
+            if e == ast:
+                return i
+        raise KeyError(f"ORDER BY expression not in output: {ast!r}")

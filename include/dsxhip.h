@@ -53,6 +53,7 @@ enum DsxOp {
   DSX_OP_LIT_NULL = 4,
   DSX_OP_ADD_F64 = 10, DSX_OP_SUB_F64 = 11, DSX_OP_MUL_F64 = 12, DSX_OP_DIV_F64 = 13,
   DSX_OP_ADD_I64 = 14, DSX_OP_SUB_I64 = 15, DSX_OP_MUL_I64 = 16,
+  DSX_OP_DIV_I64 = 17, DSX_OP_MOD_I64 = 18,
   DSX_OP_LT_F64 = 20, DSX_OP_LE_F64 = 21, DSX_OP_GT_F64 = 22, DSX_OP_GE_F64 = 23,
   DSX_OP_EQ_F64 = 24, DSX_OP_NE_F64 = 25,
   DSX_OP_LT_I64 = 30, DSX_OP_LE_I64 = 31, DSX_OP_GT_I64 = 32, DSX_OP_GE_I64 = 33,
@@ -207,25 +208,27 @@ typedef struct DsxAggSpec {
 /* replaces `df.groupby(by, dropna=False).agg(...)`
  * (dask_sql/physical/rel/logical/aggregate.py:575-581) with the WHERE
  * predicate fused in (filter.py:20-45 fused into the same scan — SURVEY §3
- * call stack (2)+(4)).
+ * call stack (2)+(4)) and the key pack fused in (one fused scan kernel).
  *
- * keys: u64 codes from dsx_keypack with known range key_space (≤ 2^63).
- *       key_space ≤ lds_threshold → per-CU LDS direct-indexed accumulation;
+ * keys: nkeys DsxKeySpec (ranges from dsx_minmax); the packed key space
+ *       (Π(range+nullable)) must be ≤ 2^62. nkeys 0 = full-table aggregate.
+ *       key_space ≤ LDS budget → per-CU LDS direct-indexed accumulation;
  *       else global CAS-claim table (SURVEY §7 step 4 two-level design).
  * pred: optional predicate program (NULL→False), pred_len 0 = no predicate.
  * Outputs (library-allocated, compacted, one row per non-empty group):
- *   out_codes u64[G], per-agg f64/i64 value arrays, per-agg u64 nonnull
- *   counts (for SUM NULL semantics and COUNT), G = *out_groups.
+ *   out_codes u64[G], out_vals = one buffer laid out [naggs][G] (f64/i64 per
+ *   agg op), out_counts = one buffer [naggs][G] of non-NULL input counts
+ *   (for SUM NULL semantics and COUNT), G = *out_groups.
  *   For DSX_AGG_SUM_*: value is the sum over non-NULL inputs; nonnull count 0
  *   ⇒ SQL NULL (custom_sum min_count=1) — host finalizes.
  *   For MIN/MAX: same. For COUNT: value array unused, count is the result. */
 int dsx_hash_groupby(DsxCtx* ctx,
                      const DsxColumn* cols, int ncols, int64_t n,
-                     const uint64_t* key_codes, uint64_t key_space,
+                     const DsxKeySpec* keys, int nkeys,
                      const DsxInstr* pred, int pred_len,
                      const DsxAggSpec* aggs, int naggs,
-                     uint64_t** out_codes, void** out_vals /*[naggs]*/,
-                     uint64_t** out_counts /*[naggs]*/, int64_t* out_groups);
+                     uint64_t** out_codes, void** out_vals /*[naggs][G]*/,
+                     uint64_t** out_counts /*[naggs][G]*/, int64_t* out_groups);
 
 /* ---- shuffle support --------------------------------------- */
 
