@@ -1,0 +1,209 @@
+"""CPU tests: planner shapes + duck-typed plan API + Rex
+compiler. No GPU needed."""
+import numpy as np
+import pandas as pd
+import pytest
+
+from dask_sql_amd.context import Context
+from dask_sql_amd.planner.plan import Call, InputRef, Literal
+
+
+@pytest.fixture
+def c(user_table_1, user_table_2, df_simple):
+    c = Context()
+    c.create_table("user_table_1", user_table_1)
+    c.create_table("user_table_2", user_table_2)
+    c.create_table("df_simple", df_simple)
+    return c
+
+
+def test_scan_shape(c):
+    rel = c._get_ral("SELECT * FROM user_table_1")
+    assert rel.get_current_node_type() == "Projection"
+    scan = rel.get_inputs()[0]
+    assert scan.get_current_node_type() == "TableScan"
+    assert scan.getRowType().getFieldNames() == ["user_id", "b"]
+    assert scan.table_scan().getTableName() == "user_table_1"
+
+
+def test_filter_shape(c):
+    # conjuncts may be pushed down as stacked Filter nodes (PushDownFilter
+    # analog); collect every condition on the chain
+    rel = c._get_ral("SELECT * FROM user_table_1 WHERE b < 3 AND user_id = 2")
+    node = rel.get_inputs()[0]
+    ops = []
+    while node.get_current_node_type() == "Filter":
+        cond = node.filter().getCondition()
+        assert str(cond.getRexType()) == "RexType.Call"
+        ops.append(cond.getOperatorName())
+        node = node.get_inputs()[0]
+    assert node.get_current_node_type() == "TableScan"
+    assert sorted(ops) == ["<", "="] or ops == ["AND"]
+
+
+def test_join_shape_and_condition(c):
+    rel = c._get_ral(
+        "SELECT lhs.user_id, lhs.b, rhs.c FROM user_table_1 AS lhs "
+        "JOIN user_table_2 AS rhs ON lhs.user_id = rhs.user_id"
+    )
+    join = rel.get_inputs()[0]
+    assert join.get_current_node_type() == "Join"
+    j = join.join()
+    assert str(j.getJoinType()) == "INNER"
+    cond = j.getCondition()
+    assert cond.getOperatorName() == "="
+    ops = cond.getOperands()
+    assert [o.getIndex() for o in ops] == [0, 2]  # lhs idx 0, rhs idx 2
+
+
+def test_aggregate_shape(c):
+    rel = c._get_ral(
+        "SELECT user_id, SUM(b) AS S FROM user_table_1 GROUP BY user_id"
+    )
+    agg = rel.get_inputs()[0]
+    assert agg.get_current_node_type() == "Aggregate"
+    a = agg.aggregate()
+    groups = a.getGroupSets()
+    assert len(groups) == 1 and isinstance(groups[0], InputRef)
+    calls = a.getNamedAggCalls()
+    assert len(calls) == 1
+    assert a.getAggregationFuncName(calls[0]) == "sum"
+    args = a.getArgs(calls[0])
+    assert len(args) == 1 and isinstance(args[0], InputRef)
+    assert calls[0].getFilterExpr() is None
+    assert not calls[0].isDistinctAgg()
+
+
+def test_aggregate_filter_clause(c):
+    rel = c._get_ral(
+        "SELECT SUM(b) FILTER (WHERE user_id = 2) AS S1, SUM(b) AS S2 "
+        "FROM user_table_1"
+    )
+    agg = rel.get_inputs()[0]
+    a = agg.aggregate()
+    calls = a.getNamedAggCalls()
+    assert len(calls) == 2
+    filters = [cc.getFilterExpr() for cc in calls]
+    assert sum(f is not None for f in filters) == 1
+
+
+def test_fq_disambiguation(c):
+    # context.py:890-898: duplicated output names become fully qualified
+    rel = c._get_ral(
+        "SELECT lhs.user_id, lhs.b, rhs.user_id, rhs.c FROM user_table_1 lhs "
+        "JOIN user_table_2 rhs ON lhs.user_id = rhs.user_id"
+    )
+    assert rel.getRowType().getFieldNames() == \
+        ["lhs.user_id", "b", "rhs.user_id", "c"]
+
+
+def test_implicit_join_becomes_inner(c):
+    rel = c._get_ral(
+        "SELECT lhs.b FROM user_table_1 lhs, user_table_2 rhs "
+        "WHERE lhs.user_id = rhs.user_id AND lhs.b > 1"
+    )
+    # find join node
+    node = rel
+    while node.get_current_node_type() != "Join":
+        node = node.get_inputs()[0]
+    assert node.join().getJoinType() == "INNER"
+    assert node.join().getCondition() is not None
+
+
+def test_filter_pushdown_below_join(c):
+    rel = c._get_ral(
+        "SELECT lhs.b FROM user_table_1 lhs, user_table_2 rhs "
+        "WHERE lhs.user_id = rhs.user_id AND lhs.b > 1"
+    )
+    node = rel
+    while node.get_current_node_type() != "Join":
+        node = node.get_inputs()[0]
+    lhs = node.get_inputs()[0]
+    assert lhs.get_current_node_type() == "Filter"  # b > 1 pushed down
+
+
+def test_order_limit_shape(c):
+    rel = c._get_ral(
+        "SELECT user_id, SUM(b) AS S FROM user_table_1 GROUP BY user_id "
+        "ORDER BY S DESC LIMIT 2"
+    )
+    assert rel.get_current_node_type() == "Limit"
+    srt = rel.get_inputs()[0]
+    assert srt.get_current_node_type() == "Sort"
+    keys = srt.sort().getCollation()
+    assert keys[0][0] == 1 and keys[0][1] is False
+
+
+def test_distinct_shape(c):
+    rel = c._get_ral("SELECT DISTINCT user_id FROM user_table_1")
+    assert rel.get_current_node_type() == "Distinct"
+    assert rel.aggregate().isDistinctNode()
+
+
+def test_plugin_replacement():
+    # the drop-in boundary: add_plugin_class(cls, replace=True) swaps the
+    # converter (reference physical/rel/convert.py:32-36, utils.py:61-91)
+    from dask_sql_amd.physical.convert import RelConverter
+    from dask_sql_amd.physical.rel_plugins import DaskFilterPlugin
+
+    class MyFilter(DaskFilterPlugin):
+        class_name = "Filter"
+
+    orig = RelConverter._plugins["Filter"]
+    try:
+        RelConverter.add_plugin_class(MyFilter, replace=True)
+        assert isinstance(RelConverter._plugins["Filter"], MyFilter)
+        RelConverter.add_plugin_class(DaskFilterPlugin, replace=False)
+        assert isinstance(RelConverter._plugins["Filter"], MyFilter)
+    finally:
+        RelConverter._plugins["Filter"] = orig
+
+
+def test_case_when_parse(c):
+    rel = c._get_ral(
+        "SELECT CASE WHEN b > 2 THEN 1 ELSE 0 END AS x FROM user_table_1"
+    )
+    proj = rel.projection().getNamedProjects()
+    assert proj[0][1] == "x"
+    assert proj[0][0].getOperatorName() == "CASE"
+
+
+def test_between_parse(c):
+    # BETWEEN expands to >= AND <= (reference rex/core/call.py:963 semantics)
+    rel = c._get_ral("SELECT * FROM user_table_1 WHERE b BETWEEN 1 AND 3")
+    node = rel.get_inputs()[0]
+    ops = []
+    while node.get_current_node_type() == "Filter":
+        ops.append(node.filter().getCondition().getOperatorName())
+        node = node.get_inputs()[0]
+    assert sorted(ops) in (["<=", ">="], ["AND"])
+
+
+def test_date_literal(c):
+    c.create_table("d", pd.DataFrame({"x": np.array([0, 10000], np.int32)}),
+                   date_columns=["x"])
+    rel = c._get_ral("SELECT * FROM d WHERE x < DATE '1995-03-15'")
+    cond = rel.get_inputs()[0].filter().getCondition()
+    lit = cond.getOperands()[1]
+    assert str(lit.getRexType()) == "RexType.Literal"
+    assert lit.getValue() == 9204  # days since epoch
+
+
+def test_rex_compiler_programs(c, user_table_1):
+    """Compile over fake CPU column stubs (no GPU needed)."""
+    from dask_sql_amd.physical.rex import compile_expr
+    from dask_sql_amd import runtime as rt
+
+    class FakeCol:
+        def __init__(self, dtype):
+            self.dtype = dtype
+            self.validity = None
+
+    rel = c._get_ral("SELECT * FROM user_table_1 WHERE b < 3 AND user_id = 2")
+    cond = rel.get_inputs()[0].filter().getCondition()
+    cols = [FakeCol(rt.I64), FakeCol(rt.I64)]
+    prog, kind = compile_expr(cond, cols)
+    assert kind == "b"
+    ops = [p[0] for p in prog]
+    assert 40 in ops  # AND
+    assert 30 in ops or 34 in ops  # int compare
