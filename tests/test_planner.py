@@ -199,11 +199,11 @@ def test_rex_compiler_programs(c, user_table_1):
             self.dtype = dtype
             self.validity = None
 
-    rel = c._get_ral("SELECT * FROM user_table_1 WHERE b < 3 AND user_id = 2")
+    rel = c._get_ral("SELECT * FROM user_table_1 WHERE b < 3 OR user_id = 2")
     cond = rel.get_inputs()[0].filter().getCondition()
     cols = [FakeCol(rt.I64), FakeCol(rt.I64)]
     prog, kind = compile_expr(cond, cols)
     assert kind == "b"
     ops = [p[0] for p in prog]
-    assert 40 in ops  # AND
-    assert 30 in ops or 34 in ops  # int compare
+    assert 41 in ops  # OR
+    assert 30 in ops and 34 in ops  # int compares
