@@ -263,7 +263,19 @@ class Builder:
                 out_fields = lhs_fields
             else:
                 out_fields = lhs_fields + rhs_fields
-            return LogicalPlan("Join", [lhs, rhs], RelDataType(out_fields),
+            # duplicate base names across the two sides would collapse in the
+            # name→backend mapping downstream: qualify them (what DataFusion's
+            # qualified join fields give the reference; context.py:890-898)
+            from collections import Counter
+            counts = Counter(f.getName().lower() for f in out_fields)
+            fixed = []
+            for f in out_fields:
+                if counts[f.getName().lower()] > 1 and f.qualifier:
+                    fixed.append(Field(f"{f.qualifier}.{f.getName()}",
+                                       f.getType(), qualifier=f.qualifier))
+                else:
+                    fixed.append(f)
+            return LogicalPlan("Join", [lhs, rhs], RelDataType(fixed),
                                JoinNode(join_type, cond))
 
         # comma-joined tables: EliminateCrossJoin — find WHERE equalities
