@@ -1,0 +1,173 @@
+"""Multi-GPU orchestration: one process per GPU, torch.distributed over RCCL
+(backend "nccl" IS RCCL on ROCm) across xGMI.
+
+Replaces dask's hash-repartition "tasks" shuffle (dask_sql/__init__.py:16;
+inside dd.merge / groupby.agg): rows are bucketed by key hash with
+k_part_hist/k_part_scatter, gathered into torch-owned staging buffers, and
+exchanged with all_to_all_single; each rank then runs its local HIP kernels
+on the received buckets. Partition-local aggregation (the Q1 pattern) needs
+only this partial-merge: partial group tables are exchanged by key hash so
+each rank owns a disjoint key range — strictly more parallel than the
+reference's split_out=1 funnel, results identical after a
+gather.
+
+Only communication + orchestration lives here; all compute is HIP kernels
+(no CPU fallback). The gloo branch exists so the exchange logic is covered
+by world_size-2 CPU tests.
+"""
+from __future__ import annotations
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from dask_sql_amd import runtime as rt
+
+_TORCH_DTYPE = {
+    rt.I64: torch.int64, rt.F64: torch.float64, rt.I32: torch.int32,
+    rt.F32: torch.float32, rt.I8: torch.int8, rt.BOOL8: torch.uint8,
+}
+
+
+def mix64_np(x: np.ndarray) -> np.ndarray:
+    """numpy mirror of the device mix64 (dsxhip.hip) for CPU tests."""
+    x = x.astype(np.uint64).copy()
+    with np.errstate(over="ignore"):
+        x += np.uint64(0x9E3779B97F4A7C15)
+        x ^= x >> np.uint64(30)
+        x *= np.uint64(0xBF58476D1CE4E5B9)
+        x ^= x >> np.uint64(27)
+        x *= np.uint64(0x94D049BB133111EB)
+        x ^= x >> np.uint64(31)
+    return x
+
+
+PART_SALT = np.uint64(0xA5A5A5A55A5A5A5A)
+
+
+def bucket_of_np(codes: np.ndarray, nbuckets: int) -> np.ndarray:
+    """CPU mirror of k_part_hist's bucket function (dsxhip.hip PART_SALT)."""
+    return (mix64_np(codes.astype(np.uint64) ^ PART_SALT)
+            % np.uint64(nbuckets)).astype(np.int64)
+
+
+def exchange_buckets(tensors: list[torch.Tensor], in_splits: list[int],
+                     group=None) -> tuple[list[torch.Tensor], list[int]]:
+    """All-to-all of bucket-contiguous tensors. in_splits[r] = rows this rank
+    sends to rank r; every tensor shares the same splits (one row set,
+    several columns). Returns (received tensors, recv splits).
+
+    nccl/RCCL: all_to_all_single over xGMI. gloo (CPU tests): send counts via
+    all_gather, payload via all_to_all if available else gather-broadcast."""
+    world = dist.get_world_size(group)
+    dev = tensors[0].device
+    in_t = torch.tensor(in_splits, dtype=torch.int64, device=dev)
+    counts = torch.zeros(world, dtype=torch.int64, device=dev)
+    # counts[r] after: rows rank r sends to me
+    all_counts = [torch.zeros_like(in_t) for _ in range(world)]
+    dist.all_gather(all_counts, in_t, group=group)
+    me = dist.get_rank(group)
+    out_splits = [int(all_counts[r][me].item()) for r in range(world)]
+    n_recv = sum(out_splits)
+    received = []
+    for t in tensors:
+        out = torch.empty(n_recv, dtype=t.dtype, device=t.device)
+        if dist.get_backend(group) == "nccl":
+            dist.all_to_all_single(out, t, out_splits, in_splits, group=group)
+        else:
+            # gloo has no all_to_all: emulate with all_gather of full tensors
+            gathered = [torch.empty(int(in_tot.sum().item()), dtype=t.dtype)
+                        for in_tot in all_counts]
+            # need per-rank full sizes
+            sizes = [int(c.sum().item()) for c in all_counts]
+            gathered = [torch.empty(s, dtype=t.dtype) for s in sizes]
+            dist.all_gather(gathered, t.contiguous(), group=group)
+            parts = []
+            for r in range(world):
+                ofs = [0] + list(np.cumsum(
+                    [int(all_counts[r][i].item()) for i in range(world)]))
+                parts.append(gathered[r][ofs[me]:ofs[me + 1]])
+            out = torch.cat(parts) if parts else out
+        received.append(out)
+    return received, out_splits
+
+
+def shuffle_device_columns(runtime, key_col, payload_cols, group=None):
+    """GPU path: partition rows of (key, payloads) by key hash across the
+    world, exchange over RCCL, return received columns wrapped for the local
+    kernels. key_col values must be non-negative ints (packed codes)."""
+    world = dist.get_world_size(group)
+    n = key_col.len
+    sel, offsets = runtime.partition(key_col, world)
+    in_splits = [int(offsets[b + 1] - offsets[b]) for b in range(world)]
+    dev = torch.device("cuda", runtime.device_id)
+    staged = []
+    for col in [key_col] + list(payload_cols):
+        t = torch.empty(n, dtype=_TORCH_DTYPE[col.dtype], device=dev)
+        if n:
+            runtime.gather_into(col, sel.data, n, t.data_ptr())
+        staged.append(t)
+    runtime.synchronize()  # our stream → before NCCL's stream reads
+    received, out_splits = exchange_buckets(staged, in_splits, group)
+    torch.cuda.synchronize(dev)  # NCCL writes → before our kernels read
+    cols = [runtime.wrap_devptr(t.data_ptr(), t.numel(),
+                                _dtype_of(t), keep_alive=t)
+            for t in received]
+    return cols[0], cols[1:], out_splits
+
+
+def _dtype_of(t: torch.Tensor) -> int:
+    for k, v in _TORCH_DTYPE.items():
+        if v == t.dtype:
+            return k
+    raise KeyError(t.dtype)
+
+
+def merge_groupby_partials(runtime, key_col, val_cols, val_ops, group=None):
+    """Merge per-rank partial aggregates: exchange partial rows by key hash
+    (so each rank owns a disjoint key set), then locally re-aggregate with
+    the fused kernel. val_ops: 'sum_f'|'sum_i'|'min_i'|'min_f'|'max_i'|
+    'max_f' per column (COUNT partials merge as sum_i).
+
+    This is the distributed form of the reference's tree-reduction `agg`
+    step (dd.Aggregation agg=, aggregate.py:117-231) over RCCL."""
+    rkey, rvals, _ = shuffle_device_columns(runtime, key_col, val_cols, group)
+    if rkey.len == 0:
+        return rkey, [c for c in rvals]
+    mn, mx, _ = runtime.minmax_i64(rkey)
+    keyspecs = [(0, mn, mx - mn + 1, False)]
+    op_map = {"sum_f": rt.AGG_SUM_F64, "sum_i": rt.AGG_SUM_I64,
+              "min_f": rt.AGG_MIN_F64, "min_i": rt.AGG_MIN_I64,
+              "max_f": rt.AGG_MAX_F64, "max_i": rt.AGG_MAX_I64}
+    cols = [rkey] + list(rvals)
+    specs = []
+    for i, op in enumerate(val_ops):
+        prog = runtime.make_prog([(1, i + 1, 0)])  # OP_COL i+1
+        specs.append((op_map[op], prog))
+    oc, ov, on, G = runtime.hash_groupby(cols, rkey.len, keyspecs, None,
+                                         specs)
+
+    class _H:
+        def __init__(s, ptrs):
+            s.ptrs = ptrs
+
+        def __del__(s):
+            for p in s.ptrs:
+                try:
+                    runtime._free(p)
+                except Exception:
+                    pass
+
+    h = _H([oc, ov, on])
+    # unpack codes back to keys: code = key - mn
+    codes_col = rt.DeviceColumn(runtime, oc, None, G, rt.I64, owner=False,
+                                keep_alive=h)
+    key_out = runtime.eval(
+        runtime.make_prog([(1, 0, 0), (3, 0, mn), (14, 0, 0)]),
+        [codes_col], G, rt.I64, with_validity=False)
+    out_vals = []
+    for i, op in enumerate(val_ops):
+        dtype = rt.F64 if op.endswith("_f") else rt.I64
+        out_vals.append(rt.DeviceColumn(runtime, ov + i * G * 8, None, G,
+                                        dtype, owner=False, keep_alive=h))
+    return key_out, out_vals

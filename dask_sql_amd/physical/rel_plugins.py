@@ -28,6 +28,17 @@ def _dicts_of(cols):
     return [getattr(c, "dictionary", None) for c in cols]
 
 
+def _minmax_cached(runtime, col):
+    """Column statistics memo (the reference caches table statistics too —
+    datacontainer.py Statistics / statistics.py:21). Safe: device columns are
+    immutable once built."""
+    hit = getattr(col, "_minmax", None)
+    if hit is None:
+        hit = runtime.minmax_i64(col)
+        col._minmax = hit
+    return hit
+
+
 def _gather_table(runtime, dc: DataContainer, sel_ptr, n_sel,
                   force_validity=False) -> DataContainer:
     """df[mask] / take: gather every backend column through a selection
@@ -256,8 +267,8 @@ class DaskJoinPlugin(BaseRelPlugin):
         # combined ranges over both sides so codes are comparable
         ranges = []
         for li, ri in zip(lhs_on, rhs_on):
-            lmn, lmx, lnn = runtime.minmax_i64(lcols[li])
-            rmn, rmx, rnn = runtime.minmax_i64(rcols[ri])
+            lmn, lmx, lnn = _minmax_cached(runtime, lcols[li])
+            rmn, rmx, rnn = _minmax_cached(runtime, rcols[ri])
             if lnn == 0 and rnn == 0:
                 mn, mx = 0, 0
             elif lnn == 0:
@@ -416,7 +427,7 @@ class DaskAggregatePlugin(BaseRelPlugin):
             if col.dtype not in _INT_KINDS:
                 raise RexCompileError(
                     "non-integer GROUP BY keys on GPU path (round-2)")
-            mn, mx, nn = runtime.minmax_i64(col)
+            mn, mx, nn = _minmax_cached(runtime, col)
             if nn == 0:
                 mn, mx = 0, 0
             nullable = bool(col.validity)
@@ -670,7 +681,7 @@ class DaskAggregatePlugin(BaseRelPlugin):
         codes_ref = None
         for call in calls:
             ai = agg.getArgs(call)[0].getIndex()
-            mn, mx, nn = runtime.minmax_i64(cols[ai])
+            mn, mx, nn = _minmax_cached(runtime, cols[ai])
             if nn == 0:
                 mn, mx = 0, 0
             ks2 = keyspecs + [(ai, mn, mx - mn + 1, bool(cols[ai].validity))]

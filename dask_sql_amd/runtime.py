@@ -288,6 +288,23 @@ class Runtime:
         )
         return sel.value, count.value
 
+    def gather_into(self, col: DeviceColumn, sel_ptr, n_sel, out_ptr: int):
+        """Gather into a caller-owned device buffer (e.g. a torch tensor's
+        storage, for the RCCL shuffle staging — SURVEY §8e)."""
+        _check(
+            self.lib,
+            self.lib.dsx_gather(self.ctx, ct.byref(col.c_struct()),
+                                ct.c_void_p(sel_ptr), ct.c_int64(n_sel),
+                                ct.c_void_p(out_ptr), None),
+            "dsx_gather",
+        )
+
+    def wrap_devptr(self, ptr: int, n: int, dtype: int,
+                    keep_alive=None) -> DeviceColumn:
+        """Wrap an externally-owned device buffer (e.g. torch tensor)."""
+        return DeviceColumn(self, ptr, None, n, dtype, owner=False,
+                            keep_alive=keep_alive)
+
     def gather(self, col: DeviceColumn, sel_ptr, n_sel,
                force_validity=False) -> DeviceColumn:
         need_valid = force_validity or bool(col.validity)
