@@ -973,6 +973,9 @@ extern "C" int dsx_hash_unmatched(DsxCtx* c, DsxHashTable* t,
 // ---------------------------------------------------------------------------
 struct AggArg {
   int32_t op[DSX_MAX_AGGS];
+  int32_t never_null[DSX_MAX_AGGS];  // input prog cannot yield NULL: the
+                                     // per-agg non-null count equals the
+                                     // group row count → skip its atomic
   int32_t naggs;
 };
 
@@ -991,43 +994,38 @@ __device__ __forceinline__ uint64_t i64_ordered(int64_t v) {
 }
 
 // per-agg accumulator update into (vals u64-typed, cnts)
-__device__ __forceinline__ void agg_update_global(int op, uint64_t* val,
+__device__ __forceinline__ void agg_update_global(int op, int never_null,
+                                                  uint64_t* val,
                                                   unsigned long long* cnt,
                                                   Slot v, bool valid) {
   if (!valid) return;
   switch (op) {
     case DSX_AGG_SUM_F64:
       unsafeAtomicAdd((double*)val, v.f);
-      atomicAdd(cnt, 1ull);
       break;
     case DSX_AGG_SUM_I64:
       atomicAdd((unsigned long long*)val, (unsigned long long)v.i);
-      atomicAdd(cnt, 1ull);
       break;
     case DSX_AGG_COUNT:
-      atomicAdd(cnt, 1ull);
       break;
     case DSX_AGG_MIN_F64:
       atomicMin((unsigned long long*)val,
                 (unsigned long long)f64_ordered(v.f));
-      atomicAdd(cnt, 1ull);
       break;
     case DSX_AGG_MAX_F64:
       atomicMax((unsigned long long*)val,
                 (unsigned long long)f64_ordered(v.f));
-      atomicAdd(cnt, 1ull);
       break;
     case DSX_AGG_MIN_I64:
       atomicMin((unsigned long long*)val,
                 (unsigned long long)i64_ordered(v.i));
-      atomicAdd(cnt, 1ull);
       break;
     case DSX_AGG_MAX_I64:
       atomicMax((unsigned long long*)val,
                 (unsigned long long)i64_ordered(v.i));
-      atomicAdd(cnt, 1ull);
       break;
   }
+  if (!never_null) atomicAdd(cnt, 1ull);
 }
 
 __device__ __forceinline__ uint64_t agg_identity(int op) {
@@ -1084,37 +1082,31 @@ k_groupby_direct(ColsArg C, int64_t n, KeyArg K,
       switch (A.op[a]) {
         case DSX_AGG_SUM_F64:
           unsafeAtomicAdd((double*)&s_vals[idx], v.f);
-          atomicAdd(&s_cnts[idx], 1u);
           break;
         case DSX_AGG_SUM_I64:
           atomicAdd((unsigned long long*)&s_vals[idx],
                     (unsigned long long)v.i);
-          atomicAdd(&s_cnts[idx], 1u);
           break;
         case DSX_AGG_COUNT:
-          atomicAdd(&s_cnts[idx], 1u);
           break;
         case DSX_AGG_MIN_F64:
           atomicMin((unsigned long long*)&s_vals[idx],
                     (unsigned long long)f64_ordered(v.f));
-          atomicAdd(&s_cnts[idx], 1u);
           break;
         case DSX_AGG_MAX_F64:
           atomicMax((unsigned long long*)&s_vals[idx],
                     (unsigned long long)f64_ordered(v.f));
-          atomicAdd(&s_cnts[idx], 1u);
           break;
         case DSX_AGG_MIN_I64:
           atomicMin((unsigned long long*)&s_vals[idx],
                     (unsigned long long)i64_ordered(v.i));
-          atomicAdd(&s_cnts[idx], 1u);
           break;
         case DSX_AGG_MAX_I64:
           atomicMax((unsigned long long*)&s_vals[idx],
                     (unsigned long long)i64_ordered(v.i));
-          atomicAdd(&s_cnts[idx], 1u);
           break;
       }
+      if (!A.never_null[a]) atomicAdd(&s_cnts[idx], 1u);
     }
   }
   __syncthreads();
@@ -1125,7 +1117,8 @@ k_groupby_direct(ColsArg C, int64_t n, KeyArg K,
   }
   for (int i = threadIdx.x; i < A.naggs * key_space; i += BLOCK) {
     int a = i / key_space;
-    if (s_cnts[i] == 0) continue;
+    int k = i - a * key_space;
+    if (A.never_null[a] ? (s_gcnt[k] == 0) : (s_cnts[i] == 0)) continue;
     switch (A.op[a]) {
       case DSX_AGG_SUM_F64:
         unsafeAtomicAdd((double*)&g_vals[i],
@@ -1193,7 +1186,8 @@ __global__ void k_groupby_global(ColsArg C, int64_t n, KeyArg K,
       Slot v;
       bool valid = vm_eval(p, agg_lens[a], C, r, v);
       p += DSX_MAX_PROG;
-      agg_update_global(A.op[a], &g_vals[(int64_t)a * slots + s],
+      agg_update_global(A.op[a], A.never_null[a],
+                        &g_vals[(int64_t)a * slots + s],
                         &g_cnts[(int64_t)a * slots + s], v, valid);
     }
   }
@@ -1235,8 +1229,9 @@ __global__ void k_groupby_emit(const uint64_t* tkeys, int64_t slots,
           break;
       }
       out_vals_flat[(int64_t)a * G + (int64_t)o] = raw;
-      out_counts_flat[(int64_t)a * G + (int64_t)o] =
-          (uint64_t)g_cnts[(int64_t)a * slots + s];
+      out_counts_flat[(int64_t)a * G + (int64_t)o] = A.never_null[a]
+          ? (uint64_t)g_gcnt[s]
+          : (uint64_t)g_cnts[(int64_t)a * slots + s];
     }
   }
 }
@@ -1287,7 +1282,16 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
   P.len = pred_len;
   AggArg A{};
   A.naggs = naggs;
-  for (int a = 0; a < naggs; a++) A.op[a] = aggs[a].op;
+  for (int a = 0; a < naggs; a++) {
+    A.op[a] = aggs[a].op;
+    int nn = 1;
+    for (int i = 0; i < aggs[a].prog_len; i++) {
+      const DsxInstr& in = aggs[a].prog[i];
+      if (in.op == DSX_OP_LIT_NULL) nn = 0;
+      if (in.op == DSX_OP_COL && cols[in.arg0].validity != nullptr) nn = 0;
+    }
+    A.never_null[a] = nn;
+  }
 
   // agg programs live in device memory (too big for kernel args)
   std::vector<DsxInstr> progs((size_t)naggs * DSX_MAX_PROG);

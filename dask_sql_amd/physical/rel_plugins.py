@@ -406,6 +406,17 @@ class DaskAggregatePlugin(BaseRelPlugin):
     def convert(self, rel, context):
         runtime = context._get_runtime()
         agg = rel.aggregate()
+
+        # FUSION PEEPHOLE (the north-star kernel, SURVEY §3 call stacks
+        # (2)+(4) fused): Aggregate(Projection(Filter*(X))) → ONE kernel over
+        # X's columns with the WHERE predicate as the kernel's pred program
+        # and the projection arithmetic folded into each agg program. This is
+        # what replaces the reference's filter→assign→groupby pass chain with
+        # a single HBM scan.
+        fused = self._try_fused(runtime, rel, agg, context)
+        if fused is not None:
+            return fused
+
         (dc,) = self.assert_inputs(rel, 1, context)
         cols = dc.backend_cols()
 
@@ -494,21 +505,110 @@ class DaskAggregatePlugin(BaseRelPlugin):
                                   codes_np, agg_calls, merged)
 
     # ------------------------------------------------------------------
+    def _try_fused(self, runtime, rel, agg, context):
+        """Fusion peephole; returns a DataContainer or None (fall back)."""
+        from dask_sql_amd.planner.plan import Literal as PLiteral
+        from dask_sql_amd.planner.plan import SqlType
+
+        if agg.isDistinctNode():
+            return None
+        calls = agg.getNamedAggCalls()
+        if not calls:
+            return None
+        if any(c.getFilterExpr() is not None or c.isDistinctAgg()
+               for c in calls):
+            return None
+        input_rel = rel.get_inputs()[0]
+        if input_rel.get_current_node_type() != "Projection":
+            return None
+        named = input_rel.projection().getNamedProjects()
+        node = input_rel.get_inputs()[0]
+        pred_exprs = []
+        while node.get_current_node_type() == "Filter":
+            cond = node.filter().getCondition()
+            if scalar_literal(cond) is not None:
+                return None
+            pred_exprs.append(cond)
+            node = node.get_inputs()[0]
+        base_rel = node
+        group_exprs = agg.getGroupSets()
+        if not all(isinstance(e, InputRef) for e in group_exprs):
+            return None
+        for e in group_exprs:
+            if not isinstance(named[e.getIndex()][0], InputRef):
+                return None
+        try:
+            base_dc = RelConverter.convert(base_rel, context)
+            base_cols = base_dc.backend_cols()
+            dicts = _dicts_of(base_cols)
+            keyspecs = []
+            group_meta = []
+            for e in group_exprs:
+                proj_expr, proj_name = named[e.getIndex()]
+                bi = proj_expr.getIndex()
+                col = base_cols[bi]
+                if col.dtype not in _INT_KINDS:
+                    return None
+                mn, mx, nn = _minmax_cached(runtime, col)
+                if nn == 0:
+                    mn, mx = 0, 0
+                keyspecs.append((bi, mn, mx - mn + 1, bool(col.validity)))
+                group_meta.append((proj_name, col))
+            pred_prog = None
+            if pred_exprs:
+                cond = pred_exprs[0]
+                for p in pred_exprs[1:]:
+                    cond = Call("AND", [cond, p])
+                prog, _ = compile_expr(cond, base_cols, dicts)
+                pred_prog = runtime.make_prog(prog)
+            specs, fins = [], []
+            for call in calls:
+                func = agg.getAggregationFuncName(call).lower()
+                if func not in self.AGG_OPS:
+                    return None
+                args = agg.getArgs(call)
+                if args:
+                    if not isinstance(args[0], InputRef):
+                        return None
+                    expr = named[args[0].getIndex()][0]
+                else:
+                    expr = PLiteral(1, SqlType("BIGINT"))
+                op, prog, fin = self._agg_spec_expr(func, expr, base_cols,
+                                                    dicts)
+                specs.append((op, runtime.make_prog(prog)))
+                fins.append(fin)
+        except RexCompileError:
+            return None
+        logger.debug("aggregate: fused scan over %s",
+                     base_rel.get_current_node_type())
+        return self._device_exec(runtime, rel, base_cols,
+                                 base_dc.table.num_rows, keyspecs, group_meta,
+                                 pred_prog, calls, specs, fins)
+
     def _convert_device(self, runtime, rel, dc, cols, keyspecs, group_idx,
                         filt_idx, calls, agg):
-        """Single-bucket fused kernel; everything stays in HBM."""
-        from dask_sql_amd.physical.rex import (OP_LIT_I64, OP_GT_I64,
-                                               OP_LIT_NULL, OP_SELECT,
-                                               OP_DIV_F64, OP_I64_TO_F64)
-        pred_prog = runtime.make_prog([(OP_COL, filt_idx, 0)]) \
-            if filt_idx is not None else None
+        """Single-bucket kernel over the already-converted input."""
         specs = []
         fins = []
         for call in calls:
             op, prog, fin = self._agg_spec_for(agg, call, cols)
             specs.append((op, runtime.make_prog(prog)))
             fins.append(fin)
-        oc, ov, on, G = runtime.hash_groupby(cols, dc.table.num_rows,
+        pred_prog = runtime.make_prog([(OP_COL, filt_idx, 0)]) \
+            if filt_idx is not None else None
+        cc_in = dc.column_container
+        group_meta = [(cc_in.columns[gi], cols[gi]) for gi in group_idx]
+        return self._device_exec(runtime, rel, cols, dc.table.num_rows,
+                                 keyspecs, group_meta, pred_prog, calls,
+                                 specs, fins)
+
+    def _device_exec(self, runtime, rel, cols, n_rows, keyspecs, group_meta,
+                     pred_prog, calls, specs, fins):
+        """Run the fused kernel; finalize device-resident."""
+        from dask_sql_amd.physical.rex import (OP_LIT_I64, OP_GT_I64,
+                                               OP_LIT_NULL, OP_SELECT,
+                                               OP_DIV_F64, OP_I64_TO_F64)
+        oc, ov, on, G = runtime.hash_groupby(cols, n_rows,
                                              keyspecs, pred_prog, specs)
 
         class _Holder:
@@ -529,11 +629,11 @@ class DaskAggregatePlugin(BaseRelPlugin):
 
         out_cols = {}
         order_names = []
-        cc_in = dc.column_container
 
         # group keys: unpack on device — part = (code / stride) % space
         stride = 1
-        for j, (gi, mn, rng, nullable) in enumerate(keyspecs):
+        for j, ((gi, mn, rng, nullable), (name, src)) in enumerate(
+                zip(keyspecs, group_meta)):
             space = rng + (1 if nullable else 0)
             prog = [(OP_COL, 0, 0), (OP_LIT_I64, 0, stride), (17, 0, 0),
                     (OP_LIT_I64, 0, space), (18, 0, 0)]  # DIV, MOD
@@ -553,11 +653,9 @@ class DaskAggregatePlugin(BaseRelPlugin):
             stride *= space
             col = runtime.eval(runtime.make_prog(prog), [codes_col], G,
                                rt.I64, with_validity=nullable)
-            src = cols[gi]
             if getattr(src, "dictionary", None) is not None:
                 col.dictionary = src.dictionary
             col.logical_dtype = src.dtype
-            name = cc_in.columns[gi]
             out_cols[f"g__{name}"] = col
             order_names.append((name, f"g__{name}"))
 
@@ -602,9 +700,12 @@ class DaskAggregatePlugin(BaseRelPlugin):
         """(kernel op, program, finalize) for one agg call."""
         func = agg.getAggregationFuncName(call).lower()
         args = agg.getArgs(call)
-        if args:
-            e = args[0]
-            prog, kind = compile_expr(e, cols, _dicts_of(cols))
+        return self._agg_spec_expr(func, args[0] if args else None, cols,
+                                   _dicts_of(cols))
+
+    def _agg_spec_expr(self, func, expr, cols, dicts):
+        if expr is not None:
+            prog, kind = compile_expr(expr, cols, dicts)
         else:
             prog, kind = [(3, 0, 1)], KI  # LIT_I64 1 — COUNT(*)
         if func == "count":
