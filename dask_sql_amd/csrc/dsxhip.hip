@@ -217,117 +217,168 @@ __device__ __forceinline__ void vm_load_col(const ColsArg& C, int ci,
   }
 }
 
+// VM stack lives in NAMED registers: `sp` is wave-uniform (every lane runs
+// the same program), so the switch below lowers to scalar branches and the
+// slots stay in VGPRs — a runtime-indexed local array would spill every
+// push/pop to scratch (cdna_hip_programming.md §5.4 rule 20; measured
+// 224 B/lane of scratch before this change). Host compiler enforces
+// depth ≤ 6 (runtime.py make_prog).
+struct VmStack {
+  Slot s0, s1, s2, s3, s4, s5;
+  bool v0, v1, v2, v3, v4, v5;
+  __device__ __forceinline__ void set(int i, Slot s, bool v) {
+    switch (i) {
+      case 0: s0 = s; v0 = v; break;
+      case 1: s1 = s; v1 = v; break;
+      case 2: s2 = s; v2 = v; break;
+      case 3: s3 = s; v3 = v; break;
+      case 4: s4 = s; v4 = v; break;
+      default: s5 = s; v5 = v; break;
+    }
+  }
+  __device__ __forceinline__ void get(int i, Slot& s, bool& v) const {
+    switch (i) {
+      case 0: s = s0; v = v0; break;
+      case 1: s = s1; v = v1; break;
+      case 2: s = s2; v = v2; break;
+      case 3: s = s3; v = v3; break;
+      case 4: s = s4; v = v4; break;
+      default: s = s5; v = v5; break;
+    }
+  }
+};
+
 // returns value in out, validity flag as return
-__device__ bool vm_eval(const DsxInstr* prog, int len, const ColsArg& C,
+// returns value in out, validity flag as return
+__device__ __forceinline__ bool vm_eval(const DsxInstr* prog, int len, const ColsArg& C,
                         int64_t r, Slot& out) {
-  Slot st[12];
-  bool va[12];
+  VmStack k;
   int sp = 0;
   for (int pc = 0; pc < len; pc++) {
-    const DsxInstr in = prog[pc];
-    switch (in.op) {
+    const DsxInstr raw = prog[pc];
+    // the program is wave-uniform (all lanes execute the same instruction
+    // stream) but may have been loaded through vector loads (global-memory
+    // programs); readfirstlane makes uniformity PROVABLE so the switch is a
+    // scalar branch and kernarg structs are scalar-indexed instead of being
+    // privatized to scratch (cdna_hip_programming.md T20 recipe).
+    const int op = __builtin_amdgcn_readfirstlane(raw.op);
+    const int arg0 = __builtin_amdgcn_readfirstlane(raw.arg0);
+    const int64_t imm =
+        ((int64_t)__builtin_amdgcn_readfirstlane((int)(raw.imm >> 32))
+         << 32) |
+        (uint32_t)__builtin_amdgcn_readfirstlane((int)(raw.imm & 0xFFFFFFFF));
+    Slot a, b, c, res;
+    bool av, bv, cv, rv;
+    switch (op) {
       case DSX_OP_COL:
-        vm_load_col(C, in.arg0, r, st[sp], va[sp]);
-        sp++;
+        vm_load_col(C, arg0, r, res, rv);
+        k.set(sp++, res, rv);
         break;
       case DSX_OP_LIT_F64:
       case DSX_OP_LIT_I64:
-        st[sp].i = in.imm;
-        va[sp] = true;
-        sp++;
+        res.i = imm;
+        k.set(sp++, res, true);
         break;
       case DSX_OP_LIT_NULL:
-        st[sp].i = 0;
-        va[sp] = false;
-        sp++;
+        res.i = 0;
+        k.set(sp++, res, false);
         break;
+#define POP2()                                                                 \
+  k.get(sp - 2, a, av);                                                        \
+  k.get(sp - 1, b, bv);                                                        \
+  sp--;
 #define BIN_F(OP, EXPR)                                                        \
   case OP: {                                                                   \
-    double a = st[sp - 2].f, b = st[sp - 1].f;                                 \
-    bool v = va[sp - 2] && va[sp - 1];                                         \
-    sp--;                                                                      \
-    st[sp - 1].EXPR;                                                           \
-    va[sp - 1] = v;                                                            \
+    POP2();                                                                    \
+    res.EXPR;                                                                  \
+    k.set(sp - 1, res, av && bv);                                              \
   } break;
-      BIN_F(DSX_OP_ADD_F64, f = a + b)
-      BIN_F(DSX_OP_SUB_F64, f = a - b)
-      BIN_F(DSX_OP_MUL_F64, f = a * b)
-      BIN_F(DSX_OP_DIV_F64, f = a / b)
-      BIN_F(DSX_OP_LT_F64, i = (a < b) ? 1 : 0)
-      BIN_F(DSX_OP_LE_F64, i = (a <= b) ? 1 : 0)
-      BIN_F(DSX_OP_GT_F64, i = (a > b) ? 1 : 0)
-      BIN_F(DSX_OP_GE_F64, i = (a >= b) ? 1 : 0)
-      BIN_F(DSX_OP_EQ_F64, i = (a == b) ? 1 : 0)
-      BIN_F(DSX_OP_NE_F64, i = (a != b) ? 1 : 0)
+      BIN_F(DSX_OP_ADD_F64, f = a.f + b.f)
+      BIN_F(DSX_OP_SUB_F64, f = a.f - b.f)
+      BIN_F(DSX_OP_MUL_F64, f = a.f * b.f)
+      BIN_F(DSX_OP_DIV_F64, f = a.f / b.f)
+      BIN_F(DSX_OP_LT_F64, i = (a.f < b.f) ? 1 : 0)
+      BIN_F(DSX_OP_LE_F64, i = (a.f <= b.f) ? 1 : 0)
+      BIN_F(DSX_OP_GT_F64, i = (a.f > b.f) ? 1 : 0)
+      BIN_F(DSX_OP_GE_F64, i = (a.f >= b.f) ? 1 : 0)
+      BIN_F(DSX_OP_EQ_F64, i = (a.f == b.f) ? 1 : 0)
+      BIN_F(DSX_OP_NE_F64, i = (a.f != b.f) ? 1 : 0)
+      BIN_F(DSX_OP_ADD_I64, i = a.i + b.i)
+      BIN_F(DSX_OP_SUB_I64, i = a.i - b.i)
+      BIN_F(DSX_OP_MUL_I64, i = a.i * b.i)
+      BIN_F(DSX_OP_DIV_I64, i = b.i ? a.i / b.i : 0)
+      BIN_F(DSX_OP_MOD_I64, i = b.i ? a.i % b.i : 0)
+      BIN_F(DSX_OP_LT_I64, i = (a.i < b.i) ? 1 : 0)
+      BIN_F(DSX_OP_LE_I64, i = (a.i <= b.i) ? 1 : 0)
+      BIN_F(DSX_OP_GT_I64, i = (a.i > b.i) ? 1 : 0)
+      BIN_F(DSX_OP_GE_I64, i = (a.i >= b.i) ? 1 : 0)
+      BIN_F(DSX_OP_EQ_I64, i = (a.i == b.i) ? 1 : 0)
+      BIN_F(DSX_OP_NE_I64, i = (a.i != b.i) ? 1 : 0)
 #undef BIN_F
-#define BIN_I(OP, EXPR)                                                        \
-  case OP: {                                                                   \
-    int64_t a = st[sp - 2].i, b = st[sp - 1].i;                                \
-    bool v = va[sp - 2] && va[sp - 1];                                         \
-    sp--;                                                                      \
-    st[sp - 1].EXPR;                                                           \
-    va[sp - 1] = v;                                                            \
-  } break;
-      BIN_I(DSX_OP_ADD_I64, i = a + b)
-      BIN_I(DSX_OP_SUB_I64, i = a - b)
-      BIN_I(DSX_OP_MUL_I64, i = a * b)
-      BIN_I(DSX_OP_DIV_I64, i = b ? a / b : 0)
-      BIN_I(DSX_OP_MOD_I64, i = b ? a % b : 0)
-      BIN_I(DSX_OP_LT_I64, i = (a < b) ? 1 : 0)
-      BIN_I(DSX_OP_LE_I64, i = (a <= b) ? 1 : 0)
-      BIN_I(DSX_OP_GT_I64, i = (a > b) ? 1 : 0)
-      BIN_I(DSX_OP_GE_I64, i = (a >= b) ? 1 : 0)
-      BIN_I(DSX_OP_EQ_I64, i = (a == b) ? 1 : 0)
-      BIN_I(DSX_OP_NE_I64, i = (a != b) ? 1 : 0)
-#undef BIN_I
       case DSX_OP_AND: {
         // SQL 3-valued: F if either F; NULL if any NULL else T
-        bool a = st[sp - 2].i != 0, b = st[sp - 1].i != 0;
-        bool av = va[sp - 2], bv = va[sp - 1];
-        sp--;
-        bool false_wins = (av && !a) || (bv && !b);
-        st[sp - 1].i = (!false_wins && av && bv) ? 1 : 0;
-        va[sp - 1] = false_wins || (av && bv);
+        POP2();
+        bool fa = av && a.i == 0, fb = bv && b.i == 0;
+        bool false_wins = fa || fb;
+        res.i = (!false_wins && av && bv) ? 1 : 0;
+        k.set(sp - 1, res, false_wins || (av && bv));
         break;
       }
       case DSX_OP_OR: {
-        bool a = st[sp - 2].i != 0, b = st[sp - 1].i != 0;
-        bool av = va[sp - 2], bv = va[sp - 1];
-        sp--;
-        bool true_wins = (av && a) || (bv && b);
-        st[sp - 1].i = true_wins ? 1 : 0;
-        va[sp - 1] = true_wins || (av && bv);
+        POP2();
+        bool ta = av && a.i != 0, tb = bv && b.i != 0;
+        bool true_wins = ta || tb;
+        res.i = true_wins ? 1 : 0;
+        k.set(sp - 1, res, true_wins || (av && bv));
         break;
       }
+#undef POP2
+#define UN()                                                                   \
+  k.get(sp - 1, a, av);
       case DSX_OP_NOT:
-        st[sp - 1].i = st[sp - 1].i ? 0 : 1;
-        break;  // validity unchanged
+        UN();
+        res.i = a.i ? 0 : 1;
+        k.set(sp - 1, res, av);  // validity unchanged
+        break;
       case DSX_OP_IS_NULL:
-        st[sp - 1].i = va[sp - 1] ? 0 : 1;
-        va[sp - 1] = true;
+        UN();
+        res.i = av ? 0 : 1;
+        k.set(sp - 1, res, true);
         break;
       case DSX_OP_IS_NOT_NULL:
-        st[sp - 1].i = va[sp - 1] ? 1 : 0;
-        va[sp - 1] = true;
+        UN();
+        res.i = av ? 1 : 0;
+        k.set(sp - 1, res, true);
         break;
       case DSX_OP_I64_TO_F64:
-        st[sp - 1].f = (double)st[sp - 1].i;
+        UN();
+        res.f = (double)a.i;
+        k.set(sp - 1, res, av);
         break;
       case DSX_OP_F64_TO_I64:
-        st[sp - 1].i = (int64_t)st[sp - 1].f;  // trunc, mappings.py:346-353
+        UN();
+        res.i = (int64_t)a.f;  // trunc, mappings.py:346-353
+        k.set(sp - 1, res, av);
         break;
       case DSX_OP_NEG_F64:
-        st[sp - 1].f = -st[sp - 1].f;
+        UN();
+        res.f = -a.f;
+        k.set(sp - 1, res, av);
         break;
       case DSX_OP_NEG_I64:
-        st[sp - 1].i = -st[sp - 1].i;
+        UN();
+        res.i = -a.i;
+        k.set(sp - 1, res, av);
         break;
+#undef UN
       case DSX_OP_SELECT: {
         // (cond, a, b): cond true→a, false/NULL→b (CASE WHEN semantics)
-        bool cv = va[sp - 3] && st[sp - 3].i != 0;
-        st[sp - 3] = cv ? st[sp - 2] : st[sp - 1];
-        va[sp - 3] = cv ? va[sp - 2] : va[sp - 1];
+        k.get(sp - 3, c, cv);
+        k.get(sp - 2, a, av);
+        k.get(sp - 1, b, bv);
         sp -= 2;
+        bool take = cv && c.i != 0;
+        k.set(sp - 1, take ? a : b, take ? av : bv);
         break;
       }
       default:
@@ -335,8 +386,13 @@ __device__ bool vm_eval(const DsxInstr* prog, int len, const ColsArg& C,
         return false;
     }
   }
-  out = st[0];
-  return va[0];
+  {
+    Slot s;
+    bool v;
+    k.get(0, s, v);
+    out = s;
+    return v;
+  }
 }
 
 __device__ __forceinline__ uint64_t mix64(uint64_t x) {
@@ -1045,18 +1101,18 @@ __device__ __forceinline__ uint64_t agg_identity(int op) {
 // LDS layout: [key_space × naggs] u64 vals, [key_space × naggs] u32 cnts,
 // [key_space] u32 gcnt. Host guarantees it fits (≤ LDS budget).
 __global__ void __launch_bounds__(BLOCK)
-k_groupby_direct(ColsArg C, int64_t n, KeyArg K,
+k_groupby_direct(ColsArg C, int64_t n, const KeyArg* Kp,
                  int key_space, ProgArg pred, const DsxInstr* agg_progs,
-                 const int32_t* agg_lens, AggArg A,
+                 const int32_t* agg_lens, const AggArg* Ap,
                  uint64_t* g_vals /*[naggs][key_space]*/,
                  unsigned long long* g_cnts /*[naggs][key_space]*/,
                  unsigned long long* g_gcnt /*[key_space]*/) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint64_t* s_vals = (uint64_t*)smem;                       // naggs*ks
-  uint32_t* s_cnts = (uint32_t*)(s_vals + A.naggs * key_space);  // naggs*ks
-  uint32_t* s_gcnt = (uint32_t*)(s_cnts + A.naggs * key_space); // ks
-  for (int i = threadIdx.x; i < A.naggs * key_space; i += BLOCK) {
-    s_vals[i] = agg_identity(A.op[i / key_space]);
+  uint32_t* s_cnts = (uint32_t*)(s_vals + Ap->naggs * key_space);  // naggs*ks
+  uint32_t* s_gcnt = (uint32_t*)(s_cnts + Ap->naggs * key_space); // ks
+  for (int i = threadIdx.x; i < Ap->naggs * key_space; i += BLOCK) {
+    s_vals[i] = agg_identity(Ap->op[i / key_space]);
     s_cnts[i] = 0;
   }
   for (int i = threadIdx.x; i < key_space; i += BLOCK) s_gcnt[i] = 0;
@@ -1070,16 +1126,16 @@ k_groupby_direct(ColsArg C, int64_t n, KeyArg K,
       bool pvalid = vm_eval(pred.ins, pred.len, C, r, pv);
       if (!(pvalid && pv.i != 0)) continue;  // NULL→False (filter.py:39)
     }
-    int k = (int)pack_key(K, C, r);
+    int k = (int)pack_key(*Kp, C, r);
     atomicAdd(&s_gcnt[k], 1u);
     const DsxInstr* p = agg_progs;
-    for (int a = 0; a < A.naggs; a++) {
+    for (int a = 0; a < Ap->naggs; a++) {
       Slot v;
       bool valid = vm_eval(p, agg_lens[a], C, r, v);
       p += DSX_MAX_PROG;
       if (!valid) continue;
       int idx = a * key_space + k;
-      switch (A.op[a]) {
+      switch (Ap->op[a]) {
         case DSX_AGG_SUM_F64:
           unsafeAtomicAdd((double*)&s_vals[idx], v.f);
           break;
@@ -1106,7 +1162,7 @@ k_groupby_direct(ColsArg C, int64_t n, KeyArg K,
                     (unsigned long long)i64_ordered(v.i));
           break;
       }
-      if (!A.never_null[a]) atomicAdd(&s_cnts[idx], 1u);
+      if (!Ap->never_null[a]) atomicAdd(&s_cnts[idx], 1u);
     }
   }
   __syncthreads();
@@ -1115,11 +1171,11 @@ k_groupby_direct(ColsArg C, int64_t n, KeyArg K,
     if (s_gcnt[i])
       atomicAdd(&g_gcnt[i], (unsigned long long)s_gcnt[i]);
   }
-  for (int i = threadIdx.x; i < A.naggs * key_space; i += BLOCK) {
+  for (int i = threadIdx.x; i < Ap->naggs * key_space; i += BLOCK) {
     int a = i / key_space;
     int k = i - a * key_space;
-    if (A.never_null[a] ? (s_gcnt[k] == 0) : (s_cnts[i] == 0)) continue;
-    switch (A.op[a]) {
+    if (Ap->never_null[a] ? (s_gcnt[k] == 0) : (s_cnts[i] == 0)) continue;
+    switch (Ap->op[a]) {
       case DSX_AGG_SUM_F64:
         unsafeAtomicAdd((double*)&g_vals[i],
                         __longlong_as_double((long long)s_vals[i]));
@@ -1146,9 +1202,9 @@ k_groupby_direct(ColsArg C, int64_t n, KeyArg K,
 }
 
 // ---- global CAS-claim path --------------------------------------------------
-__global__ void k_groupby_global(ColsArg C, int64_t n, KeyArg K,
+__global__ void k_groupby_global(ColsArg C, int64_t n, const KeyArg* Kp,
                                  ProgArg pred, const DsxInstr* agg_progs,
-                                 const int32_t* agg_lens, AggArg A,
+                                 const int32_t* agg_lens, const AggArg* Ap,
                                  uint64_t* tkeys, int64_t mask,
                                  uint64_t* g_vals /*[naggs][slots]*/,
                                  unsigned long long* g_cnts,
@@ -1163,7 +1219,7 @@ __global__ void k_groupby_global(ColsArg C, int64_t n, KeyArg K,
       bool pvalid = vm_eval(pred.ins, pred.len, C, r, pv);
       if (!(pvalid && pv.i != 0)) continue;
     }
-    uint64_t cde = pack_key(K, C, r);
+    uint64_t cde = pack_key(*Kp, C, r);
     int64_t s = (int64_t)(mix64(cde) & mask);
     int64_t probes = 0;
     while (true) {
@@ -1182,11 +1238,11 @@ __global__ void k_groupby_global(ColsArg C, int64_t n, KeyArg K,
     }
     atomicAdd(&g_gcnt[s], 1ull);
     const DsxInstr* p = agg_progs;
-    for (int a = 0; a < A.naggs; a++) {
+    for (int a = 0; a < Ap->naggs; a++) {
       Slot v;
       bool valid = vm_eval(p, agg_lens[a], C, r, v);
       p += DSX_MAX_PROG;
-      agg_update_global(A.op[a], A.never_null[a],
+      agg_update_global(Ap->op[a], Ap->never_null[a],
                         &g_vals[(int64_t)a * slots + s],
                         &g_cnts[(int64_t)a * slots + s], v, valid);
     }
@@ -1324,6 +1380,7 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
     // device buffers: tkeys (global path), g_vals, g_cnts, g_gcnt, programs,
     // counter, overflow flag
     int64_t need = prog_bytes + lens_bytes + 8 /*counter*/ + 8 /*ovf*/ +
+                   (int64_t)sizeof(KeyArg) + (int64_t)sizeof(AggArg) + 32 +
                    (direct ? 0 : slots * 8) + (int64_t)naggs * slots * 8 +
                    (int64_t)naggs * slots * 8 + slots * 8 + 64;
     int rc = ensure_scratch(c, need);
@@ -1337,6 +1394,10 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
     base += 8;
     int* d_ovf = (int*)base;
     base += 8;
+    KeyArg* d_K = (KeyArg*)base;
+    base += ((sizeof(KeyArg) + 15) / 16) * 16;
+    AggArg* d_A = (AggArg*)base;
+    base += ((sizeof(AggArg) + 15) / 16) * 16;
     uint64_t* d_tkeys = nullptr;
     if (!direct) {
       d_tkeys = (uint64_t*)base;
@@ -1352,6 +1413,10 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
                            hipMemcpyHostToDevice, c->stream));
     HIP_TRY(hipMemcpyAsync(d_lens, lens.data(), lens_bytes,
                            hipMemcpyHostToDevice, c->stream));
+    HIP_TRY(hipMemcpyAsync(d_K, &K, sizeof(KeyArg), hipMemcpyHostToDevice,
+                           c->stream));
+    HIP_TRY(hipMemcpyAsync(d_A, &A, sizeof(AggArg), hipMemcpyHostToDevice,
+                           c->stream));
     HIP_TRY(hipMemsetAsync(d_counter, 0, 8, c->stream));
     HIP_TRY(hipMemsetAsync(d_ovf, 0, 8, c->stream));
     HIP_TRY(hipMemsetAsync(d_cnts, 0, (size_t)naggs * slots * 8, c->stream));
@@ -1376,13 +1441,13 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
         (void)lds;
         ProfScope ps(c, "k_groupby_direct");
         hipLaunchKernelGGL(k_groupby_direct, dim3(grid), dim3(BLOCK),
-                           lds_bytes, c->stream, C, n, K,
-                           (int)key_space, P, d_progs, d_lens, A, d_vals,
+                           lds_bytes, c->stream, C, n, d_K,
+                           (int)key_space, P, d_progs, d_lens, d_A, d_vals,
                            d_cnts, d_gcnt);
       } else {
         ProfScope ps(c, "k_groupby_global");
         hipLaunchKernelGGL(k_groupby_global, dim3(grid), dim3(BLOCK), 0,
-                           c->stream, C, n, K, P, d_progs, d_lens, A,
+                           c->stream, C, n, d_K, P, d_progs, d_lens, d_A,
                            d_tkeys, slots - 1, d_vals, d_cnts, d_gcnt, d_ovf);
       }
     }
