@@ -1249,6 +1249,306 @@ __global__ void k_groupby_global(ColsArg C, int64_t n, const KeyArg* Kp,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// partition-based groupby (the large-cardinality path):
+// global atomics cap at ~23 RMW/ns on gfx950 (measured, probe_groupby.hip)
+// while streaming runs at 6.3 TB/s — so instead of one CAS table we
+//   (1) histogram rows into NB key-hash buckets per block (LDS counters),
+//   (2) scan → deterministic per-(block,bucket) bases,
+//   (3) scatter compact records (key + agg inputs, AoS) bucket-major —
+//       block-contiguous runs merge in the XCD's L2 so the scattered writes
+//       leave near-coalesced,
+//   (4) one block per bucket aggregates its contiguous run in an LDS
+//       open-addressing table (LDS atomics only), bump-appends compacted
+//       groups,
+//   (5) finalize into the [naggs][G] output slabs.
+// Requires: all agg inputs never-null (the common case; else the CAS path).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ int gb_bucket(uint64_t code, int nb_mask) {
+  return (int)((mix64(code ^ 0xC2B2AE3D27D4EB4Full) >> 32) & (uint64_t)nb_mask);
+}
+
+__global__ void k_gbpart_hist(ColsArg C, int64_t n, const KeyArg* Kp,
+                              ProgArg pred, int nb,
+                              int64_t* hist /*[grid][nb]*/) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  uint32_t* s_hist = (uint32_t*)smem;
+  for (int i = threadIdx.x; i < nb; i += BLOCK) s_hist[i] = 0;
+  __syncthreads();
+  int64_t lo, hi;
+  block_range(n, 1, lo, hi);
+  for (int64_t r = lo + threadIdx.x; r < hi; r += BLOCK) {
+    if (pred.len) {
+      Slot pv;
+      bool pvalid = vm_eval(pred.ins, pred.len, C, r, pv);
+      if (!(pvalid && pv.i != 0)) continue;
+    }
+    uint64_t code = pack_key(*Kp, C, r);
+    atomicAdd(&s_hist[gb_bucket(code, nb - 1)], 1u);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < nb; i += BLOCK)
+    hist[(int64_t)blockIdx.x * nb + i] = (int64_t)s_hist[i];
+}
+
+// one block per bucket: exclusive scan over the grid dimension; writes
+// per-(block,bucket) bases back in place and the bucket total to totals[b]
+__global__ void k_gbpart_scan(int64_t* hist, int grid, int nb,
+                              int64_t* totals) {
+  int b = blockIdx.x;
+  __shared__ int64_t s_sum[BLOCK];
+  int64_t per = (grid + BLOCK - 1) / BLOCK;
+  int64_t local = 0;
+  for (int64_t g = threadIdx.x * per; g < min((int64_t)grid,
+                                              (threadIdx.x + 1) * per); g++)
+    local += hist[g * nb + b];
+  s_sum[threadIdx.x] = local;
+  __syncthreads();
+  // block scan (Hillis-Steele) over the 256 partials
+  for (int d = 1; d < BLOCK; d <<= 1) {
+    int64_t v = (threadIdx.x >= d) ? s_sum[threadIdx.x - d] : 0;
+    __syncthreads();
+    s_sum[threadIdx.x] += v;
+    __syncthreads();
+  }
+  int64_t excl = s_sum[threadIdx.x] - local;
+  int64_t run = excl;
+  for (int64_t g = threadIdx.x * per; g < min((int64_t)grid,
+                                              (threadIdx.x + 1) * per); g++) {
+    int64_t v = hist[g * nb + b];
+    hist[g * nb + b] = run;
+    run += v;
+  }
+  if (threadIdx.x == BLOCK - 1) totals[b] = s_sum[BLOCK - 1];
+}
+
+// 1 block: exclusive scan of bucket totals → absolute bases
+__global__ void k_gbpart_bases(int64_t* totals, int nb, int64_t* bases) {
+  if (threadIdx.x == 0) {
+    int64_t run = 0;
+    for (int b = 0; b < nb; b++) {
+      bases[b] = run;
+      run += totals[b];
+    }
+    bases[nb] = run;
+  }
+}
+
+// scatter records (key u64 + nvals f64/i64 values) bucket-major
+__global__ void k_gbpart_scatter(ColsArg C, int64_t n, const KeyArg* Kp,
+                                 ProgArg pred, const DsxInstr* agg_progs,
+                                 const int32_t* agg_lens,
+                                 const int32_t* val_of /*[naggs] or -1*/,
+                                 int naggs, int nvals, int nb,
+                                 const int64_t* hist /*bases per blk*/,
+                                 const int64_t* bases, uint64_t* out_rec) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // running absolute offset per bucket for THIS block
+  int64_t* s_off = (int64_t*)smem;
+  for (int i = threadIdx.x; i < nb; i += BLOCK)
+    s_off[i] = bases[i] + hist[(int64_t)blockIdx.x * nb + i];
+  __syncthreads();
+  int64_t lo, hi;
+  block_range(n, 1, lo, hi);
+  int rec = 1 + nvals;
+  for (int64_t r = lo + threadIdx.x; r < hi; r += BLOCK) {
+    if (pred.len) {
+      Slot pv;
+      bool pvalid = vm_eval(pred.ins, pred.len, C, r, pv);
+      if (!(pvalid && pv.i != 0)) continue;
+    }
+    uint64_t code = pack_key(*Kp, C, r);
+    int b = gb_bucket(code, nb - 1);
+    int64_t o = (int64_t)atomicAdd((unsigned long long*)&s_off[b], 1ull);
+    uint64_t* dst = out_rec + o * rec;
+    dst[0] = code;
+    const DsxInstr* p = agg_progs;
+    for (int a = 0; a < naggs; a++) {
+      if (val_of[a] >= 0) {
+        Slot v;
+        vm_eval(p, agg_lens[a], C, r, v);  // never-null guaranteed by host
+        dst[1 + val_of[a]] = (uint64_t)v.i;
+      }
+      p += DSX_MAX_PROG;
+    }
+  }
+}
+
+// one block per bucket: LDS open-addressing aggregate over the bucket's
+// contiguous record run; bump-append compacted groups (AoS temporaries)
+__global__ void __launch_bounds__(BLOCK)
+k_gbpart_aggregate(const uint64_t* recs, const int64_t* bases, int nvals,
+                   const AggArg* Ap, int lds_slots,
+                   unsigned long long* g_counter, uint64_t* tmp_codes,
+                   uint64_t* tmp_vals /*[G][nvals]*/,
+                   unsigned long long* tmp_gcnt, int* overflow) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  uint64_t* l_keys = (uint64_t*)smem;                       // [slots]
+  uint64_t* l_vals = l_keys + lds_slots;                    // [slots][nvals]
+  uint32_t* l_gcnt = (uint32_t*)(l_vals + (int64_t)lds_slots * nvals);
+  int naggs = Ap->naggs;
+  for (int i = threadIdx.x; i < lds_slots; i += BLOCK) {
+    l_keys[i] = EMPTY_KEY;
+    l_gcnt[i] = 0;
+  }
+  // per-value-slot identity: value slots are the non-COUNT aggs in order
+  for (int i = threadIdx.x; i < lds_slots * nvals; i += BLOCK) {
+    int vslot = i % nvals;
+    int seen = 0;
+    uint64_t ident = 0;
+    for (int a = 0; a < naggs; a++) {
+      int op = Ap->op[a];
+      if (op == DSX_AGG_COUNT) continue;
+      if (seen == vslot) {
+        ident = agg_identity(op);
+        break;
+      }
+      seen++;
+    }
+    l_vals[i] = ident;
+  }
+  __syncthreads();
+
+  int64_t r0 = bases[blockIdx.x];
+  int64_t r1 = bases[blockIdx.x + 1];
+  int rec = 1 + nvals;
+  int mask = lds_slots - 1;
+  for (int64_t r = r0 + threadIdx.x; r < r1; r += BLOCK) {
+    const uint64_t* src = recs + r * rec;
+    uint64_t code = src[0];
+    int s = (int)(mix64(code) & (uint64_t)mask);
+    int probes = 0;
+    while (true) {
+      unsigned long long k = l_keys[s];
+      if (k == code) break;
+      if (k == EMPTY_KEY) {
+        unsigned long long old = atomicCAS(
+            (unsigned long long*)&l_keys[s], EMPTY_KEY,
+            (unsigned long long)code);
+        if (old == EMPTY_KEY || old == code) break;
+      }
+      s = (s + 1) & mask;
+      if (++probes > lds_slots) {
+        *overflow = 1;
+        return;
+      }
+    }
+    atomicAdd(&l_gcnt[s], 1u);
+    int vs = 0;
+    for (int a = 0; a < naggs; a++) {
+      int op = Ap->op[a];
+      if (op == DSX_AGG_COUNT) continue;
+      uint64_t raw = src[1 + vs];
+      uint64_t* dst = &l_vals[(int64_t)s * nvals + vs];
+      switch (op) {
+        case DSX_AGG_SUM_F64:
+          unsafeAtomicAdd((double*)dst, __longlong_as_double((long long)raw));
+          break;
+        case DSX_AGG_SUM_I64:
+          atomicAdd((unsigned long long*)dst, (unsigned long long)raw);
+          break;
+        case DSX_AGG_MIN_F64:
+          atomicMin((unsigned long long*)dst,
+                    (unsigned long long)f64_ordered(
+                        __longlong_as_double((long long)raw)));
+          break;
+        case DSX_AGG_MAX_F64:
+          atomicMax((unsigned long long*)dst,
+                    (unsigned long long)f64_ordered(
+                        __longlong_as_double((long long)raw)));
+          break;
+        case DSX_AGG_MIN_I64:
+          atomicMin((unsigned long long*)dst,
+                    (unsigned long long)i64_ordered((int64_t)raw));
+          break;
+        case DSX_AGG_MAX_I64:
+          atomicMax((unsigned long long*)dst,
+                    (unsigned long long)i64_ordered((int64_t)raw));
+          break;
+      }
+      vs++;
+    }
+  }
+  __syncthreads();
+  // count live slots, reserve output range, append
+  __shared__ unsigned long long s_base;
+  __shared__ unsigned int s_local;
+  if (threadIdx.x == 0) s_local = 0;
+  __syncthreads();
+  unsigned int my_rank = 0;
+  bool live = false;
+  // each thread owns slots i = tid, tid+256, ... — two-phase rank
+  for (int i = threadIdx.x; i < lds_slots; i += BLOCK)
+    if (l_keys[i] != EMPTY_KEY) my_rank++;  // count per thread
+  unsigned int my_count = my_rank;
+  // block scan of per-thread counts
+  __shared__ unsigned int s_scan[BLOCK];
+  s_scan[threadIdx.x] = my_count;
+  __syncthreads();
+  for (int d = 1; d < BLOCK; d <<= 1) {
+    unsigned int v = (threadIdx.x >= d) ? s_scan[threadIdx.x - d] : 0;
+    __syncthreads();
+    s_scan[threadIdx.x] += v;
+    __syncthreads();
+  }
+  unsigned int my_excl = s_scan[threadIdx.x] - my_count;
+  if (threadIdx.x == BLOCK - 1) {
+    unsigned int tot = s_scan[BLOCK - 1];
+    s_base = tot ? atomicAdd(g_counter, (unsigned long long)tot) : 0;
+    s_local = tot;
+  }
+  __syncthreads();
+  if (s_local == 0) return;
+  unsigned long long base = s_base;
+  unsigned int o = my_excl;
+  for (int i = threadIdx.x; i < lds_slots; i += BLOCK) {
+    if (l_keys[i] == EMPTY_KEY) continue;
+    int64_t dst = (int64_t)(base + o);
+    tmp_codes[dst] = l_keys[i];
+    tmp_gcnt[dst] = (unsigned long long)l_gcnt[i];
+    for (int v = 0; v < nvals; v++)
+      tmp_vals[dst * nvals + v] = l_vals[(int64_t)i * nvals + v];
+    o++;
+  }
+  (void)live;
+}
+
+// finalize: AoS temporaries → the [naggs][G] output slabs (emit format)
+__global__ void k_gbpart_finalize(const uint64_t* tmp_codes,
+                                  const uint64_t* tmp_vals,
+                                  const unsigned long long* tmp_gcnt,
+                                  int64_t G, int nvals, AggArg A,
+                                  uint64_t* out_codes, uint64_t* out_vals,
+                                  uint64_t* out_counts) {
+  int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (; i < G; i += stride) {
+    out_codes[i] = tmp_codes[i];
+    int vs = 0;
+    for (int a = 0; a < A.naggs; a++) {
+      uint64_t raw = 0;
+      if (A.op[a] != DSX_AGG_COUNT) {
+        raw = tmp_vals[i * nvals + vs];
+        switch (A.op[a]) {
+          case DSX_AGG_MIN_F64:
+          case DSX_AGG_MAX_F64:
+            raw = (uint64_t)__double_as_longlong(f64_unordered(raw));
+            break;
+          case DSX_AGG_MIN_I64:
+          case DSX_AGG_MAX_I64:
+            raw = raw ^ 0x8000000000000000ull;
+            break;
+        }
+        vs++;
+      }
+      out_vals[(int64_t)a * G + i] = raw;
+      out_counts[(int64_t)a * G + i] = (uint64_t)tmp_gcnt[i];
+    }
+  }
+}
+
 __global__ void k_init_aggs(uint64_t* g_vals, AggArg A, int64_t slots) {
   int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * BLOCK;
@@ -1304,6 +1604,151 @@ __global__ void k_count_live(const uint64_t* tkeys,
   }
   for (int d = 32; d > 0; d >>= 1) local += __shfl_down(local, d, 64);
   if ((threadIdx.x & 63) == 0 && local) atomicAdd(counter, local);
+}
+
+
+// host driver for the partition path; *fell_back=true → caller uses the CAS
+// path (LDS table overflow from key skew, or shapes it does not cover)
+static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
+                             ProgArg& P, AggArg& A,
+                             std::vector<DsxInstr>& progs,
+                             std::vector<int32_t>& lens, int naggs,
+                             int64_t g_est, uint64_t** out_codes,
+                             void** out_vals, uint64_t** out_counts,
+                             int64_t* out_groups, bool* fell_back) {
+  *fell_back = false;
+  int32_t val_of[DSX_MAX_AGGS];
+  int nvals = 0;
+  for (int a = 0; a < naggs; a++)
+    val_of[a] = (A.op[a] == DSX_AGG_COUNT) ? -1 : nvals++;
+  int nb = 64;
+  while (nb < 4096 && g_est / nb > 1024) nb <<= 1;
+  int64_t per_bucket = (g_est + nb - 1) / nb;
+  int lds_slots = 256;
+  while (lds_slots < 2 * per_bucket) lds_slots <<= 1;
+  int64_t slot_bytes = 8 + 8 * (int64_t)nvals + 4;
+  size_t lds_bytes = (size_t)lds_slots * slot_bytes;
+  if (lds_bytes > 96 * 1024) {
+    *fell_back = true;
+    return 0;
+  }
+  int grid = (int)min((int64_t)MAX_GRID, (n + BLOCK - 1) / BLOCK);
+  if (grid == 0) {  // empty input → empty output
+    HIP_TRY(hipMalloc((void**)out_codes, 8));
+    HIP_TRY(hipMalloc(out_vals, 8));
+    HIP_TRY(hipMalloc((void**)out_counts, 8));
+    *out_groups = 0;
+    return 0;
+  }
+  int64_t prog_bytes = (int64_t)progs.size() * sizeof(DsxInstr);
+  int64_t lens_bytes = naggs * 4;
+  int64_t Gcap = g_est;
+  int64_t need = prog_bytes + ((lens_bytes + 15) / 16) * 16 + 64 +
+                 sizeof(KeyArg) + sizeof(AggArg) + 32 +
+                 (int64_t)grid * nb * 8 + (nb + 2) * 8 + nb * 8 +
+                 Gcap * 8 + Gcap * (int64_t)(nvals ? nvals : 1) * 8 +
+                 Gcap * 8 + ((naggs * 4 + 15) / 16) * 16 + 256;
+  int rc = ensure_scratch(c, need);
+  if (rc) return rc;
+  char* base = (char*)c->scratch;
+  DsxInstr* d_progs = (DsxInstr*)base;
+  base += prog_bytes;
+  int32_t* d_lens = (int32_t*)base;
+  base += ((lens_bytes + 15) / 16) * 16;
+  int32_t* d_val_of = (int32_t*)base;
+  base += ((naggs * 4 + 15) / 16) * 16;
+  unsigned long long* d_counter = (unsigned long long*)base;
+  base += 16;
+  int* d_ovf = (int*)base;
+  base += 16;
+  KeyArg* d_K = (KeyArg*)base;
+  base += ((sizeof(KeyArg) + 15) / 16) * 16;
+  AggArg* d_A = (AggArg*)base;
+  base += ((sizeof(AggArg) + 15) / 16) * 16;
+  int64_t* d_hist = (int64_t*)base;
+  base += (int64_t)grid * nb * 8;
+  int64_t* d_totals = (int64_t*)base;
+  base += nb * 8;
+  int64_t* d_bases = (int64_t*)base;
+  base += (nb + 2) * 8;
+  uint64_t* d_tmp_codes = (uint64_t*)base;
+  base += Gcap * 8;
+  uint64_t* d_tmp_vals = (uint64_t*)base;
+  base += Gcap * (int64_t)(nvals ? nvals : 1) * 8;
+  unsigned long long* d_tmp_gcnt = (unsigned long long*)base;
+
+  HIP_TRY(hipMemcpyAsync(d_progs, progs.data(), prog_bytes,
+                         hipMemcpyHostToDevice, c->stream));
+  HIP_TRY(hipMemcpyAsync(d_lens, lens.data(), lens_bytes,
+                         hipMemcpyHostToDevice, c->stream));
+  HIP_TRY(hipMemcpyAsync(d_val_of, val_of, naggs * 4, hipMemcpyHostToDevice,
+                         c->stream));
+  HIP_TRY(hipMemcpyAsync(d_K, &K, sizeof(KeyArg), hipMemcpyHostToDevice,
+                         c->stream));
+  HIP_TRY(hipMemcpyAsync(d_A, &A, sizeof(AggArg), hipMemcpyHostToDevice,
+                         c->stream));
+  HIP_TRY(hipMemsetAsync(d_counter, 0, 8, c->stream));
+  HIP_TRY(hipMemsetAsync(d_ovf, 0, 4, c->stream));
+
+  {
+    ProfScope ps(c, "k_gbpart_hist");
+    hipLaunchKernelGGL(k_gbpart_hist, dim3(grid), dim3(BLOCK),
+                       (size_t)nb * 4, c->stream, C, n, d_K, P, nb, d_hist);
+  }
+  hipLaunchKernelGGL(k_gbpart_scan, dim3(nb), dim3(BLOCK), 0, c->stream,
+                     d_hist, grid, nb, d_totals);
+  hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(64), 0, c->stream,
+                     d_totals, nb, d_bases);
+  int64_t n_sel = 0;
+  HIP_TRY(hipMemcpyAsync(&n_sel, d_bases + nb, 8, hipMemcpyDeviceToHost,
+                         c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+
+  int rec = 1 + nvals;
+  uint64_t* d_recs = nullptr;
+  HIP_TRY(hipMalloc((void**)&d_recs, (n_sel > 0 ? n_sel : 1) * rec * 8));
+  if (n_sel > 0) {
+    {
+      ProfScope ps(c, "k_gbpart_scatter");
+      hipLaunchKernelGGL(k_gbpart_scatter, dim3(grid), dim3(BLOCK),
+                         (size_t)nb * 8, c->stream, C, n, d_K, P, d_progs,
+                         d_lens, d_val_of, naggs, nvals, nb, d_hist, d_bases,
+                         d_recs);
+    }
+    {
+      ProfScope ps(c, "k_gbpart_aggregate");
+      hipLaunchKernelGGL(k_gbpart_aggregate, dim3(nb), dim3(BLOCK),
+                         lds_bytes, c->stream, d_recs, d_bases,
+                         nvals ? nvals : 0, d_A, lds_slots, d_counter,
+                         d_tmp_codes, d_tmp_vals, d_tmp_gcnt, d_ovf);
+    }
+  }
+  int h_ovf = 0;
+  unsigned long long G = 0;
+  HIP_TRY(hipMemcpyAsync(&h_ovf, d_ovf, 4, hipMemcpyDeviceToHost, c->stream));
+  HIP_TRY(hipMemcpyAsync(&G, d_counter, 8, hipMemcpyDeviceToHost, c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  if (h_ovf) {
+    hipFree(d_recs);
+    *fell_back = true;
+    return 0;
+  }
+  HIP_TRY(hipMalloc((void**)out_codes, (G > 0 ? G : 1) * 8));
+  HIP_TRY(hipMalloc(out_vals, (G > 0 ? G : 1) * 8 * naggs));
+  HIP_TRY(hipMalloc((void**)out_counts, (G > 0 ? G : 1) * 8 * naggs));
+  if (G > 0) {
+    int g2 = (int)min((int64_t)MAX_GRID, ((int64_t)G + BLOCK - 1) / BLOCK);
+    ProfScope ps(c, "k_gbpart_finalize");
+    hipLaunchKernelGGL(k_gbpart_finalize, dim3(g2), dim3(BLOCK), 0, c->stream,
+                       d_tmp_codes, d_tmp_vals, d_tmp_gcnt, (int64_t)G,
+                       nvals ? nvals : 1, A, *out_codes,
+                       (uint64_t*)*out_vals, *out_counts);
+  }
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  hipFree(d_recs);
+  HIP_TRY(hipGetLastError());
+  *out_groups = (int64_t)G;
+  return 0;
 }
 
 extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
@@ -1364,13 +1809,27 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
   int64_t lds_per_slot = naggs * 12 + 4;
   bool direct = key_space > 0 && (int64_t)key_space * lds_per_slot <= 64 * 1024;
 
+  int64_t g_est = (int64_t)(key_space && key_space < (uint64_t)n
+                                ? (int64_t)key_space
+                                : n);
+  if (!direct) {
+    bool all_nn = true;
+    for (int a = 0; a < naggs; a++) all_nn &= (A.never_null[a] != 0);
+    if (all_nn && naggs <= 6 && g_est > 0) {
+      bool fell_back = false;
+      int prc = groupby_partition(c, C, n, K, P, A, progs, lens, naggs,
+                                  g_est, out_codes, out_vals, out_counts,
+                                  out_groups, &fell_back);
+      if (prc != 0) return prc;
+      if (!fell_back) return 0;
+    }
+  }
+
   int64_t slots;
   if (direct) {
     slots = (int64_t)key_space;
   } else {
-    int64_t est = (int64_t)(key_space && key_space < (uint64_t)n
-                                ? (int64_t)key_space
-                                : n);
+    int64_t est = g_est;
     if (est > (1ll << 26)) est = 1ll << 26;
     slots = 64;
     while (slots < 2 * est) slots <<= 1;
