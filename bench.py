@@ -32,8 +32,12 @@ WORKLOADS = {
         "rows": 100_000_000, "n_groups": 1_000_000,
         "sql": "SELECT key, SUM(x) AS s, COUNT(*) AS c FROM t "
                "WHERE x < 0.5 GROUP BY key",
-        # algorithmic bytes = 16 B/row in + ~24 B/group out
-        "dominant": "k_groupby_global",
+        # algorithmic bytes = 16 B/row in + ~24 B/group out.
+        # dominant = the fused-groupby kernel family (partition path:
+        # hist+scatter+aggregate+finalize; fallback: the CAS kernel)
+        "dominant": ["k_gbpart_hist", "k_gbpart_scatter",
+                     "k_gbpart_aggregate", "k_gbpart_finalize",
+                     "k_groupby_global"],
         "algo_bytes": lambda n, g: 16 * n + 24 * g,
         "scaling": "weak",
     },
@@ -41,7 +45,7 @@ WORKLOADS = {
         "rows": 100_000_000, "n_groups": 1_000,
         "sql": "SELECT key, SUM(x) AS s, COUNT(*) AS c FROM t "
                "WHERE x < 0.5 GROUP BY key",
-        "dominant": "k_groupby_direct",
+        "dominant": ["k_groupby_direct"],
         "algo_bytes": lambda n, g: 16 * n + 24 * g,
         "scaling": "weak",
     },
@@ -50,7 +54,8 @@ WORKLOADS = {
         "sql": "SELECT p.key, p.pv, b.bv FROM probe_t p JOIN build_t b "
                "ON p.key = b.key",
         # 16 B/probe-row + 16 B/build-row + 24 B/match
-        "dominant": "k_hash_probe_emit",
+        "dominant": ["k_hash_probe_emit", "k_hash_probe_count",
+                     "k_hash_build"],
         "algo_bytes": lambda n, g: 16 * n + 16 * g + 24 * n,
         "scaling": "weak",
     },
@@ -64,7 +69,7 @@ WORKLOADS = {
  AVG(l_discount) AS avg_disc, COUNT(*) AS count_order
  FROM t WHERE l_shipdate <= 10471 GROUP BY l_returnflag, l_linestatus""",
         # SURVEY §8d C4: ~38 B/row scanned
-        "dominant": "k_groupby_direct",
+        "dominant": ["k_groupby_direct"],
         "algo_bytes": lambda n, g: 38 * n,
         "scaling": "strong",
     },
@@ -77,7 +82,7 @@ WORKLOADS = {
  AND l_orderkey = o_orderkey AND o_orderdate < 9204 AND l_shipdate > 9204
  GROUP BY l_orderkey, o_orderdate, o_shippriority
  ORDER BY revenue DESC, o_orderdate LIMIT 10""",
-        "dominant": "k_hash_probe_emit",
+        "dominant": ["k_hash_probe_emit", "k_hash_probe_count"],
         # dominant scans ≈ lineitem 28 B + orders 24 B + customer 9 B per
         # their own rows; normalized per lineitem row below
         "algo_bytes": lambda n, g: 28 * n + 24 * (n // 4) + 9 * (n // 40),
@@ -309,11 +314,17 @@ def main():
         total_rows = rows_per_rank * world
     value = total_rows * args.steps / elapsed
 
-    # roofline from HIP-event per-kernel timing (events on the lib stream)
+    # roofline from HIP-event per-kernel timing (events on the lib stream).
+    # "dominant" = the kernel family one logical launch of the hot op runs;
+    # per-launch time = Σ over the family of (ms / launches).
     dom = w["dominant"]
+    if isinstance(dom, str):
+        dom = [dom]
+    present = [d for d in dom if d in prof and prof[d]["launches"] > 0]
     roofline = None
-    if dom in prof and prof[dom]["launches"] > 0:
-        per_launch_ms = prof[dom]["ms"] / prof[dom]["launches"]
+    if present:
+        per_launch_ms = sum(prof[d]["ms"] / prof[d]["launches"]
+                            for d in present)
         g = w.get("n_groups", w.get("build_rows", 16))
         algo_bytes = float(w["algo_bytes"](rows_per_rank, g))
         achieved = algo_bytes / (per_launch_ms / 1000.0) / 1e9  # GB/s

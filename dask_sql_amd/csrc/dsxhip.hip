@@ -1643,11 +1643,15 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   int64_t prog_bytes = (int64_t)progs.size() * sizeof(DsxInstr);
   int64_t lens_bytes = naggs * 4;
   int64_t Gcap = g_est;
+  int rec = 1 + nvals;
+  // records live in the persistent scratch arena: a per-call hipMalloc/Free
+  // of ~GBs costs tens of ms (measured); the arena amortizes it
+  int64_t rec_bytes = (n > 0 ? n : 1) * (int64_t)rec * 8;
   int64_t need = prog_bytes + ((lens_bytes + 15) / 16) * 16 + 64 +
                  sizeof(KeyArg) + sizeof(AggArg) + 32 +
                  (int64_t)grid * nb * 8 + (nb + 2) * 8 + nb * 8 +
                  Gcap * 8 + Gcap * (int64_t)(nvals ? nvals : 1) * 8 +
-                 Gcap * 8 + ((naggs * 4 + 15) / 16) * 16 + 256;
+                 Gcap * 8 + ((naggs * 4 + 15) / 16) * 16 + 256 + rec_bytes;
   int rc = ensure_scratch(c, need);
   if (rc) return rc;
   char* base = (char*)c->scratch;
@@ -1676,6 +1680,8 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   uint64_t* d_tmp_vals = (uint64_t*)base;
   base += Gcap * (int64_t)(nvals ? nvals : 1) * 8;
   unsigned long long* d_tmp_gcnt = (unsigned long long*)base;
+  base += Gcap * 8;
+  uint64_t* d_recs = (uint64_t*)base;
 
   HIP_TRY(hipMemcpyAsync(d_progs, progs.data(), prog_bytes,
                          hipMemcpyHostToDevice, c->stream));
@@ -1699,29 +1705,19 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
                      d_hist, grid, nb, d_totals);
   hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(64), 0, c->stream,
                      d_totals, nb, d_bases);
-  int64_t n_sel = 0;
-  HIP_TRY(hipMemcpyAsync(&n_sel, d_bases + nb, 8, hipMemcpyDeviceToHost,
-                         c->stream));
-  HIP_TRY(hipStreamSynchronize(c->stream));
-
-  int rec = 1 + nvals;
-  uint64_t* d_recs = nullptr;
-  HIP_TRY(hipMalloc((void**)&d_recs, (n_sel > 0 ? n_sel : 1) * rec * 8));
-  if (n_sel > 0) {
-    {
-      ProfScope ps(c, "k_gbpart_scatter");
-      hipLaunchKernelGGL(k_gbpart_scatter, dim3(grid), dim3(BLOCK),
-                         (size_t)nb * 8, c->stream, C, n, d_K, P, d_progs,
-                         d_lens, d_val_of, naggs, nvals, nb, d_hist, d_bases,
-                         d_recs);
-    }
-    {
-      ProfScope ps(c, "k_gbpart_aggregate");
-      hipLaunchKernelGGL(k_gbpart_aggregate, dim3(nb), dim3(BLOCK),
-                         lds_bytes, c->stream, d_recs, d_bases,
-                         nvals ? nvals : 0, d_A, lds_slots, d_counter,
-                         d_tmp_codes, d_tmp_vals, d_tmp_gcnt, d_ovf);
-    }
+  {
+    ProfScope ps(c, "k_gbpart_scatter");
+    hipLaunchKernelGGL(k_gbpart_scatter, dim3(grid), dim3(BLOCK),
+                       (size_t)nb * 8, c->stream, C, n, d_K, P, d_progs,
+                       d_lens, d_val_of, naggs, nvals, nb, d_hist, d_bases,
+                       d_recs);
+  }
+  {
+    ProfScope ps(c, "k_gbpart_aggregate");
+    hipLaunchKernelGGL(k_gbpart_aggregate, dim3(nb), dim3(BLOCK),
+                       lds_bytes, c->stream, d_recs, d_bases,
+                       nvals ? nvals : 0, d_A, lds_slots, d_counter,
+                       d_tmp_codes, d_tmp_vals, d_tmp_gcnt, d_ovf);
   }
   int h_ovf = 0;
   unsigned long long G = 0;
@@ -1729,7 +1725,6 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   HIP_TRY(hipMemcpyAsync(&G, d_counter, 8, hipMemcpyDeviceToHost, c->stream));
   HIP_TRY(hipStreamSynchronize(c->stream));
   if (h_ovf) {
-    hipFree(d_recs);
     *fell_back = true;
     return 0;
   }
@@ -1745,7 +1740,6 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
                        (uint64_t*)*out_vals, *out_counts);
   }
   HIP_TRY(hipStreamSynchronize(c->stream));
-  hipFree(d_recs);
   HIP_TRY(hipGetLastError());
   *out_groups = (int64_t)G;
   return 0;
