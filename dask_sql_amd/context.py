@@ -171,6 +171,8 @@ class Context:
         self.tables: dict[str, RegisteredTable] = {}
         self._runtime = None
         self._device_id = device_id
+        self._plan_cache: dict = {}
+        self._schema_version = 0
 
     # -- reference context.py:168 create_table ----------------------------
     def create_table(self, table_name: str, input_table, persist: bool = False,
@@ -186,12 +188,14 @@ class Context:
         t = RegisteredTable(host_cols)
         self.tables[table_name.lower()] = t
         self.catalog.add(table_name, t.fields())
+        self._schema_version += 1
         if persist:
             t.upload(self._get_runtime())
 
     def drop_table(self, table_name: str):
         self.tables.pop(table_name.lower(), None)
         self.catalog.drop(table_name)
+        self._schema_version += 1
 
     # -- reference context.py:482 sql --------------------------------------
     def sql(self, sql: str, return_futures: bool = True,
@@ -206,8 +210,16 @@ class Context:
 
     # -- internals ----------------------------------------------------------
     def _get_ral(self, sql: str):
-        """reference context.py:819 _get_ral (planner entry)."""
-        return Builder(self.catalog, self.schema_name).build(sql)
+        """reference context.py:819 _get_ral (planner entry). Plans are
+        immutable → cached per (sql, schema version)."""
+        key = (sql, self._schema_version)
+        plan = self._plan_cache.get(key)
+        if plan is None:
+            plan = Builder(self.catalog, self.schema_name).build(sql)
+            if len(self._plan_cache) > 256:
+                self._plan_cache.clear()
+            self._plan_cache[key] = plan
+        return plan
 
     def _get_runtime(self):
         if self._runtime is None:

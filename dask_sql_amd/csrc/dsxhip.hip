@@ -11,6 +11,7 @@
 #include <cstdio>
 #include <cstring>
 #include <string>
+#include <unordered_map>
 #include <vector>
 
 #include "../../include/dsxhip.h"
@@ -56,7 +57,61 @@ struct DsxCtx {
   int64_t scratch_bytes = 0;
   bool prof = false;
   std::vector<ProfRec> prof_recs;
+  // size-bucketed buffer pool: per-step outputs (group tables, selection
+  // vectors, join pairs) would otherwise pay ~0.1-1 ms of hipMalloc/Free
+  // per query step
+  std::unordered_map<int64_t, std::vector<void*>> pool_free;
+  std::unordered_map<void*, int64_t> pool_sizes;
+  int64_t pool_cached = 0;
 };
+
+static int64_t pool_round(int64_t bytes) {
+  int64_t r = 256;
+  while (r < bytes) r <<= 1;
+  return r;
+}
+
+static int pool_alloc(DsxCtx* c, int64_t bytes, void** out) {
+  int64_t r = pool_round(bytes > 0 ? bytes : 1);
+  auto it = c->pool_free.find(r);
+  if (it != c->pool_free.end() && !it->second.empty()) {
+    *out = it->second.back();
+    it->second.pop_back();
+    c->pool_cached -= r;
+    return 0;
+  }
+  hipError_t e = hipMalloc(out, r);
+  if (e != hipSuccess) {
+    // pressure: drop the pool and retry once
+    for (auto& kv : c->pool_free)
+      for (void* p : kv.second) hipFree(p);
+    c->pool_free.clear();
+    c->pool_cached = 0;
+    e = hipMalloc(out, r);
+    if (e != hipSuccess)
+      FAIL(-2, "device alloc of %lld failed: %s", (long long)r,
+           hipGetErrorString(e));
+  }
+  c->pool_sizes[*out] = r;
+  return 0;
+}
+
+static void pool_release(DsxCtx* c, void* p) {
+  if (!p) return;
+  auto it = c->pool_sizes.find(p);
+  if (it == c->pool_sizes.end()) {
+    hipFree(p);  // not pool-allocated
+    return;
+  }
+  int64_t r = it->second;
+  if (c->pool_cached + r > (int64_t)16 << 30) {
+    hipFree(p);
+    c->pool_sizes.erase(it);
+    return;
+  }
+  c->pool_free[r].push_back(p);
+  c->pool_cached += r;
+}
 
 static int ensure_scratch(DsxCtx* c, int64_t bytes) {
   if (c->scratch_bytes >= bytes) return 0;
@@ -86,6 +141,9 @@ extern "C" int dsx_ctx_create(int device_id, DsxCtx** out) {
 extern "C" void dsx_ctx_destroy(DsxCtx* c) {
   if (!c) return;
   hipStreamSynchronize(c->stream);
+  for (auto& kv : c->pool_free)
+    for (void* p : kv.second) hipFree(p);
+  for (auto& kv : c->pool_sizes) (void)kv;
   for (auto& r : c->prof_recs) {
     hipEventDestroy(r.start);
     hipEventDestroy(r.stop);
@@ -101,18 +159,16 @@ extern "C" int dsx_synchronize(DsxCtx* c) {
 }
 
 extern "C" int dsx_malloc(DsxCtx* c, int64_t bytes, void** out) {
-  (void)c;
-  HIP_TRY(hipMalloc(out, bytes > 0 ? bytes : 1));
-  return 0;
+  return pool_alloc(c, bytes, out);
 }
 extern "C" int dsx_free(DsxCtx* c, void* p) {
-  (void)c;
-  if (p) HIP_TRY(hipFree(p));
+  pool_release(c, p);
   return 0;
 }
 extern "C" int dsx_upload(DsxCtx* c, const void* host, int64_t bytes,
                           void** out_dev) {
-  HIP_TRY(hipMalloc(out_dev, bytes > 0 ? bytes : 1));
+  int rc = pool_alloc(c, bytes, out_dev);
+  if (rc) return rc;
   HIP_TRY(hipMemcpyAsync(*out_dev, host, bytes, hipMemcpyHostToDevice,
                          c->stream));
   HIP_TRY(hipStreamSynchronize(c->stream));
@@ -559,8 +615,7 @@ extern "C" int dsx_filter(DsxCtx* c, const DsxInstr* prog, int prog_len,
   *out_sel = nullptr;
   *out_count = 0;
   if (n == 0) {
-    HIP_TRY(hipMalloc((void**)out_sel, 4));
-    return 0;
+    return pool_alloc(c, 4, (void**)out_sel);
   }
   ProgArg P{};
   memcpy(P.ins, prog, prog_len * sizeof(DsxInstr));
@@ -591,7 +646,7 @@ extern "C" int dsx_filter(DsxCtx* c, const DsxInstr* prog, int prog_len,
   int64_t h_total = 0;
   HIP_TRY(hipMemcpyAsync(&h_total, total, 8, hipMemcpyDeviceToHost, c->stream));
   HIP_TRY(hipStreamSynchronize(c->stream));
-  HIP_TRY(hipMalloc((void**)out_sel, (h_total > 0 ? h_total : 1) * 4));
+  { int rc2 = pool_alloc(c, (h_total > 0 ? h_total : 1) * 4, (void**)out_sel); if (rc2) return rc2; }
   if (h_total > 0) {
     ProfScope ps(c, "k_filter_emit");
     hipLaunchKernelGGL(k_filter_emit, dim3(grid), dim3(BLOCK), 0, c->stream,
@@ -741,8 +796,12 @@ struct KeyArg {
 
 __device__ __forceinline__ uint64_t pack_key(const KeyArg& K, const ColsArg& C,
                                              int64_t r) {
+  // fully unrolled so K's fields scalarize into registers (a runtime-indexed
+  // access would re-load the spec per row / spill — §5.4 rule 20)
   uint64_t code = 0;
-  for (int j = 0; j < K.nkeys; j++) {
+#pragma unroll
+  for (int j = 0; j < DSX_MAX_KEYS; j++) {
+    if (j >= K.nkeys) break;
     Slot v;
     bool valid;
     vm_load_col(C, K.k[j].col, r, v, valid);
@@ -854,9 +913,9 @@ extern "C" int dsx_hash_build(DsxCtx* c, const uint64_t* codes,
   t->slots = slots;
   t->n_build = n;
   t->ctx = c;
-  if (hipMalloc((void**)&t->keys, slots * 8) != hipSuccess ||
-      hipMalloc((void**)&t->vals, slots * 4) != hipSuccess ||
-      hipMalloc((void**)&t->matched, slots * 4) != hipSuccess) {
+  if (pool_alloc(c, slots * 8, (void**)&t->keys) ||
+      pool_alloc(c, slots * 4, (void**)&t->vals) ||
+      pool_alloc(c, slots * 4, (void**)&t->matched)) {
     dsx_hash_table_free(t);
     FAIL(-2, "hash table alloc failed (%lld slots)", (long long)slots);
   }
@@ -875,9 +934,11 @@ extern "C" int dsx_hash_build(DsxCtx* c, const uint64_t* codes,
 
 extern "C" void dsx_hash_table_free(DsxHashTable* t) {
   if (!t) return;
-  if (t->keys) hipFree(t->keys);
-  if (t->vals) hipFree(t->vals);
-  if (t->matched) hipFree(t->matched);
+  if (t->ctx) {
+    pool_release(t->ctx, t->keys);
+    pool_release(t->ctx, t->vals);
+    pool_release(t->ctx, t->matched);
+  }
   delete t;
 }
 
@@ -968,8 +1029,8 @@ extern "C" int dsx_hash_probe(DsxCtx* c, DsxHashTable* t, const uint64_t* codes,
   unsigned long long total = 0;
   HIP_TRY(hipMemcpyAsync(&total, counter, 8, hipMemcpyDeviceToHost, c->stream));
   HIP_TRY(hipStreamSynchronize(c->stream));
-  HIP_TRY(hipMalloc((void**)out_probe_idx, (total > 0 ? total : 1) * 4));
-  HIP_TRY(hipMalloc((void**)out_build_idx, (total > 0 ? total : 1) * 4));
+  { int rc2 = pool_alloc(c, (total > 0 ? total : 1) * 4, (void**)out_probe_idx); if (rc2) return rc2; }
+  { int rc2 = pool_alloc(c, (total > 0 ? total : 1) * 4, (void**)out_build_idx); if (rc2) return rc2; }
   HIP_TRY(hipMemsetAsync(counter, 0, 8, c->stream));
   if (grid > 0 && total > 0) {
     ProfScope ps(c, "k_hash_probe_emit");
@@ -1011,7 +1072,7 @@ extern "C" int dsx_hash_unmatched(DsxCtx* c, DsxHashTable* t,
   unsigned long long total = 0;
   HIP_TRY(hipMemcpyAsync(&total, counter, 8, hipMemcpyDeviceToHost, c->stream));
   HIP_TRY(hipStreamSynchronize(c->stream));
-  HIP_TRY(hipMalloc((void**)out_build_idx, (total > 0 ? total : 1) * 4));
+  { int rc2 = pool_alloc(c, (total > 0 ? total : 1) * 4, (void**)out_build_idx); if (rc2) return rc2; }
   HIP_TRY(hipMemsetAsync(counter, 0, 8, c->stream));
   if (total > 0)
     hipLaunchKernelGGL(k_unmatched<1>, dim3(grid), dim3(BLOCK), 0, c->stream,
@@ -1634,11 +1695,11 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   }
   int grid = (int)min((int64_t)MAX_GRID, (n + BLOCK - 1) / BLOCK);
   if (grid == 0) {  // empty input → empty output
-    HIP_TRY(hipMalloc((void**)out_codes, 8));
-    HIP_TRY(hipMalloc(out_vals, 8));
-    HIP_TRY(hipMalloc((void**)out_counts, 8));
+    int rc2 = pool_alloc(c, 8, (void**)out_codes);
+    if (!rc2) rc2 = pool_alloc(c, 8, out_vals);
+    if (!rc2) rc2 = pool_alloc(c, 8, (void**)out_counts);
     *out_groups = 0;
-    return 0;
+    return rc2;
   }
   int64_t prog_bytes = (int64_t)progs.size() * sizeof(DsxInstr);
   int64_t lens_bytes = naggs * 4;
@@ -1728,9 +1789,13 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
     *fell_back = true;
     return 0;
   }
-  HIP_TRY(hipMalloc((void**)out_codes, (G > 0 ? G : 1) * 8));
-  HIP_TRY(hipMalloc(out_vals, (G > 0 ? G : 1) * 8 * naggs));
-  HIP_TRY(hipMalloc((void**)out_counts, (G > 0 ? G : 1) * 8 * naggs));
+  {
+    int rc2 = pool_alloc(c, (G > 0 ? G : 1) * 8, (void**)out_codes);
+    if (!rc2) rc2 = pool_alloc(c, (G > 0 ? G : 1) * 8 * naggs, out_vals);
+    if (!rc2) rc2 = pool_alloc(c, (G > 0 ? G : 1) * 8 * naggs,
+                               (void**)out_counts);
+    if (rc2) return rc2;
+  }
   if (G > 0) {
     int g2 = (int)min((int64_t)MAX_GRID, ((int64_t)G + BLOCK - 1) / BLOCK);
     ProfScope ps(c, "k_gbpart_finalize");
@@ -1926,9 +1991,13 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
     HIP_TRY(hipMemcpyAsync(&G, d_counter, 8, hipMemcpyDeviceToHost, c->stream));
     HIP_TRY(hipStreamSynchronize(c->stream));
 
-    HIP_TRY(hipMalloc((void**)out_codes, (G > 0 ? G : 1) * 8));
-    HIP_TRY(hipMalloc(out_vals, (G > 0 ? G : 1) * 8 * naggs));
-    HIP_TRY(hipMalloc((void**)out_counts, (G > 0 ? G : 1) * 8 * naggs));
+    {
+      int rc2 = pool_alloc(c, (G > 0 ? G : 1) * 8, (void**)out_codes);
+      if (!rc2) rc2 = pool_alloc(c, (G > 0 ? G : 1) * 8 * naggs, out_vals);
+      if (!rc2) rc2 = pool_alloc(c, (G > 0 ? G : 1) * 8 * naggs,
+                                 (void**)out_counts);
+      if (rc2) return rc2;
+    }
     HIP_TRY(hipMemsetAsync(d_counter, 0, 8, c->stream));
     if (G > 0) {
       int g = (int)min((int64_t)MAX_GRID, (slots + BLOCK - 1) / BLOCK);
