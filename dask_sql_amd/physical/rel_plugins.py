@@ -30,12 +30,22 @@ def _dicts_of(cols):
 
 def _minmax_cached(runtime, col):
     """Column statistics memo (the reference caches table statistics too —
-    datacontainer.py Statistics / statistics.py:21). Safe: device columns are
-    immutable once built."""
+    datacontainer.py Statistics / statistics.py:21). Derived columns (gather
+    outputs) chain to their source via _stats_src: a subset's range is
+    bounded by its source's, so the PERSISTENT base column caches the range
+    across query steps (a superset range only widens the key space, never
+    changes results)."""
     hit = getattr(col, "_minmax", None)
-    if hit is None:
-        hit = runtime.minmax_i64(col)
+    if hit is not None:
+        return hit
+    src = getattr(col, "_stats_src", None)
+    if src is not None and col.len > 0:
+        mn, mx, _ = _minmax_cached(runtime, src)
+        hit = (mn, mx, col.len)  # conservative: subset of source rows
         col._minmax = hit
+        return hit
+    hit = runtime.minmax_i64(col)
+    col._minmax = hit
     return hit
 
 
@@ -52,6 +62,7 @@ def _gather_table(runtime, dc: DataContainer, sel_ptr, n_sel,
             g = runtime.gather(col, sel_ptr, n_sel, force_validity)
             if getattr(col, "dictionary", None) is not None:
                 g.dictionary = col.dictionary
+            g._stats_src = col  # range(subset) ⊆ range(source)
             out_cols[backend] = g
     return DataContainer(DeviceTable(out_cols), cc)
 
@@ -196,6 +207,7 @@ class DaskJoinPlugin(BaseRelPlugin):
             g = runtime.gather(col, probe_sel.data, n_out, force_l)
             if getattr(col, "dictionary", None) is not None:
                 g.dictionary = col.dictionary
+            g._stats_src = col
             out_cols[f"l__{backend}"] = g
             mapping[frontend] = f"l__{backend}"
         keep_rhs = join_type not in ("leftanti", "leftsemi_native")
@@ -206,6 +218,7 @@ class DaskJoinPlugin(BaseRelPlugin):
                 g = runtime.gather(col, build_sel.data, n_out, force_r)
                 if getattr(col, "dictionary", None) is not None:
                     g.dictionary = col.dictionary
+                g._stats_src = col
                 out_cols[f"r__{backend}"] = g
                 mapping[frontend] = f"r__{backend}"
 
@@ -925,13 +938,26 @@ class DaskSortPlugin(BaseRelPlugin):
 
 
 class DaskLimitPlugin(BaseRelPlugin):
-    """LIMIT/OFFSET (reference rel/logical/limit.py:24-113)."""
+    """LIMIT/OFFSET (reference rel/logical/limit.py:24-113). Sort+Limit is
+    fused into a top-k (the reference's apply_sort topk optimization,
+    physical/utils/sort.py:9-34 / sql.yaml topk-nelem-limit): argpartition
+    the primary key, full-sort only the candidate set."""
 
     class_name = "Limit"
 
     def convert(self, rel, context):
-        (inp,) = self.assert_inputs(rel, 1, context)
         node = rel.limit()
+        below = rel.get_inputs()[0]
+        if (below.get_current_node_type() == "Sort"
+                and node.fetch is not None and node.offset == 0
+                and node.fetch <= 10_000):
+            (inp,) = self.assert_inputs(below, 1, context)
+            from dask_sql_amd.materialize import to_pandas
+            pdf = to_pandas(inp, context)
+            keys = below.sort().getCollation()
+            pdf = _topk(pdf, keys, node.fetch)
+            return HostDataContainer(pdf.reset_index(drop=True))
+        (inp,) = self.assert_inputs(rel, 1, context)
         if isinstance(inp, HostDataContainer):
             pdf = inp.pdf
         else:
@@ -942,6 +968,39 @@ class DaskLimitPlugin(BaseRelPlugin):
         if node.fetch is not None:
             pdf = pdf.iloc[: node.fetch]
         return HostDataContainer(pdf.reset_index(drop=True))
+
+
+def _topk(pdf, keys, k):
+    """Exact multi-key top-k: candidates by primary key via argpartition
+    (+ boundary ties), then the full mergesort ordering on candidates only."""
+    n = len(pdf)
+    if n > k:
+        idx0, asc0, _ = keys[0]
+        col0 = pdf.iloc[:, idx0].to_numpy()
+        import numpy as _np
+        v = col0 if asc0 else -_np.asarray(col0, dtype=_np.float64) \
+            if col0.dtype.kind == "f" else (col0 if asc0 else -col0)
+        v = _np.asarray(v)
+        if _np.isnan(_np.asarray(v, dtype=float)).any() if v.dtype.kind == "f" \
+                else False:
+            cand = pdf  # NaN keys: fall back to full sort
+        else:
+            part = _np.argpartition(v, min(k - 1, n - 1))[:k]
+            thresh = v[part].max()
+            cand_mask = v <= thresh  # includes boundary ties
+            if cand_mask.sum() > max(10 * k, 1000):
+                cand = pdf  # degenerate ties: full sort
+            else:
+                cand = pdf[cand_mask]
+    else:
+        cand = pdf
+    for idx, asc, nulls_first in reversed(keys):
+        col = cand.columns[idx]
+        cand = cand.sort_values(
+            col, ascending=asc,
+            na_position="first" if nulls_first else "last",
+            kind="mergesort")
+    return cand.iloc[:k]
 
 
 def register_defaults():
