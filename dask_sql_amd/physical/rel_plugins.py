@@ -292,11 +292,15 @@ class DaskJoinPlugin(BaseRelPlugin):
                 mn, mx = min(lmn, rmn), max(lmx, rmx)
             ranges.append((mn, mx - mn + 1))
 
-        swap = join_type == "right"
+        # build on the smaller side for INNER (what the reference gets from
+        # dask's merge internals + JoinReorder, src/sql/optimizer/join_reorder.rs)
+        swap = join_type == "right" or (
+            join_type == "inner"
+            and dc_lhs.table.num_rows < dc_rhs.table.num_rows)
         if swap:
             probe_dc, build_dc = dc_rhs, dc_lhs
             probe_on, build_on = rhs_on, lhs_on
-            ktype = rt.JOIN_LEFT
+            ktype = rt.JOIN_LEFT if join_type == "right" else rt.JOIN_INNER
         else:
             probe_dc, build_dc = dc_lhs, dc_rhs
             probe_on, build_on = lhs_on, rhs_on
