@@ -73,8 +73,9 @@ class Builder:
 
     # ------------------------------------------------------------------ API
     def build(self, sql: str) -> LogicalPlan:
+        from dask_sql_amd.planner.prune import prune_plan
         stmt = parse_sql(sql)
-        return self.build_stmt(stmt)
+        return prune_plan(self.build_stmt(stmt))
 
     # ----------------------------------------------------------------- scans
     def _scan(self, tr: TableRef) -> LogicalPlan:
