@@ -196,51 +196,86 @@ class DaskJoinPlugin(BaseRelPlugin):
             pairs, n_out = self._cross_join(runtime, dc_lhs, dc_rhs, join_type)
         probe_sel, build_sel = pairs
 
-        # materialize: lhs cols by probe idx, rhs cols by build idx
-        out_cols = {}
-        mapping = {}
+        # materialize only columns the plan consumes (output_indices from the
+        # pruning pass) plus residual-referenced temporaries
         force_l = join_type in ("outer",)
         force_r = join_type in ("left", "outer")
-        for frontend in cc_lhs.columns:
-            backend = cc_lhs.get_backend_by_frontend_name(frontend)
-            col = dc_lhs.table.col(backend)
-            g = runtime.gather(col, probe_sel.data, n_out, force_l)
+        combined = [("l", f) for f in cc_lhs.columns]
+        keep_rhs = join_type not in ("leftanti",)
+        if keep_rhs:
+            combined += [("r", f) for f in cc_rhs.columns]
+        out_idx = join.output_indices if getattr(join, "output_indices",
+                                                 None) is not None \
+            else list(range(len(combined)))
+        from dask_sql_amd.planner.prune import _expr_refs, _remap
+        if residual and join_type == "leftanti":
+            raise NotImplementedError(
+                "residual condition on LEFT ANTI join (round-2)")
+        res_refs = set()
+        for r in residual:
+            _expr_refs(r, res_refs)
+        mat_idx = sorted(set(out_idx) | res_refs)
+
+        gathered = {}
+        for i in mat_idx:
+            side, frontend = combined[i]
+            if side == "l":
+                col = dc_lhs.table.col(
+                    cc_lhs.get_backend_by_frontend_name(frontend))
+                g = runtime.gather(col, probe_sel.data, n_out, force_l)
+            else:
+                col = dc_rhs.table.col(
+                    cc_rhs.get_backend_by_frontend_name(frontend))
+                g = runtime.gather(col, build_sel.data, n_out, force_r)
             if getattr(col, "dictionary", None) is not None:
                 g.dictionary = col.dictionary
             g._stats_src = col
-            out_cols[f"l__{backend}"] = g
-            mapping[frontend] = f"l__{backend}"
-        keep_rhs = join_type not in ("leftanti", "leftsemi_native")
-        if keep_rhs:
-            for frontend in cc_rhs.columns:
-                backend = cc_rhs.get_backend_by_frontend_name(frontend)
-                col = dc_rhs.table.col(backend)
-                g = runtime.gather(col, build_sel.data, n_out, force_r)
-                if getattr(col, "dictionary", None) is not None:
-                    g.dictionary = col.dictionary
-                g._stats_src = col
-                out_cols[f"r__{backend}"] = g
-                mapping[frontend] = f"r__{backend}"
+            gathered[i] = g
 
-        order = list(cc_lhs.columns) + (list(cc_rhs.columns) if keep_rhs else [])
-        cc = ColumnContainer(order, mapping)
-        dc = DataContainer(DeviceTable(out_cols), cc)
-
-        # residual filter (join.py:169-181)
+        # residual filter (join.py:169-181) over the combined row, with
+        # InputRefs densified to the materialized set
         if residual:
             cond = residual[0]
             for r in residual[1:]:
                 cond = Call("AND", [cond, r])
-            dc = _apply_filter(runtime, dc, cond)
+            dense = {i: pos for pos, i in enumerate(mat_idx)}
+            cond = _remap(cond, dense)
+            cols_list = [gathered[i] for i in mat_idx]
+            s = scalar_literal(cond)
+            if s is not None:
+                if not s:
+                    empty = runtime.empty_column(0, rt.I32)
+                    gathered = {i: runtime.gather(gathered[i], empty.data, 0)
+                                for i in mat_idx}
+                    n_out = 0
+            else:
+                prog, _ = compile_expr(cond, cols_list, _dicts_of(cols_list))
+                sel_ptr, count = runtime.filter(runtime.make_prog(prog),
+                                                cols_list, n_out)
+                sel = runtime.wrap_sel(sel_ptr, count)
+                new_g = {}
+                for i in out_idx:
+                    col = gathered[i]
+                    g = runtime.gather(col, sel.data, count,
+                                       bool(col.validity))
+                    if getattr(col, "dictionary", None) is not None:
+                        g.dictionary = col.dictionary
+                    g._stats_src = col
+                    new_g[i] = g
+                gathered = new_g
+                n_out = count
 
         row_type = rel.getRowType()
-        cc = dc.column_container
         field_names = [str(f) for f in row_type.getFieldNames()]
-        if str(join.getJoinType()) in ("LEFTSEMI", "LEFTANTI"):
-            field_names = field_names[: len(cc.columns)]
-        cc = cc.rename(dict(zip(cc.columns, field_names)))
-        cc = cc.limit_to(field_names)
-        return DataContainer(dc.table, cc)
+        assert len(field_names) == len(out_idx), (field_names, out_idx)
+        out_cols = {}
+        mapping = {}
+        for name, i in zip(field_names, out_idx):
+            backend = f"j{i}__{name}"
+            out_cols[backend] = gathered[i]
+            mapping[name] = backend
+        cc = ColumnContainer(field_names, mapping)
+        return DataContainer(DeviceTable(out_cols), cc)
 
     # -- helpers ------------------------------------------------------------
     def _key_codes(self, runtime, dc, on, ranges):

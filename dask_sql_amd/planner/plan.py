@@ -260,9 +260,13 @@ class FilterNode:
 
 
 class JoinNode:
-    def __init__(self, join_type: str, condition: Expression | None):
+    def __init__(self, join_type: str, condition: Expression | None,
+                 output_indices=None):
         self._join_type = join_type  # INNER/LEFT/RIGHT/FULL/LEFTSEMI/LEFTANTI
         self._condition = condition
+        # positions into the combined (lhs ++ rhs) row this join outputs;
+        # None = all (set by the pruning pass, planner/prune.py)
+        self.output_indices = output_indices
 
     def getJoinType(self):
         return self._join_type

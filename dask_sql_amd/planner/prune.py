@@ -111,16 +111,13 @@ def _prune(plan: LogicalPlan, needed):
         new_cond = _remap(node.getCondition(), m) \
             if node.getCondition() is not None else None
         out_is_lhs_only = str(node.getJoinType()) in ("LEFTSEMI", "LEFTANTI")
-        if out_is_lhs_only:
-            new_fields = [fields[i] for i in l_refs]
-        else:
-            new_fields = [fields[i] for i in sorted(m.keys())]
+        out_old = [i for i in needed if (i < n_l or not out_is_lhs_only)]
+        new_fields = [fields[i] for i in out_old]
+        output_indices = [m[i] for i in out_old]  # child-combined positions
         new = LogicalPlan("Join", [new_l, new_r], RelDataType(new_fields),
-                          JoinNode(node.getJoinType(), new_cond))
-        # output mapping: positions follow sorted kept combined indices
-        out_m = {old: pos for pos, old in enumerate(sorted(m.keys()))}
-        # remap to NEW positions (after child renumbering the order of
-        # sorted(m.keys()) == lhs-kept then rhs-kept, matching field order)
+                          JoinNode(node.getJoinType(), new_cond,
+                                   output_indices=output_indices))
+        out_m = {old: pos for pos, old in enumerate(out_old)}
         return new, out_m
 
     if t in ("Aggregate", "Distinct"):
