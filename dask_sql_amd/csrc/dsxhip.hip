@@ -1889,7 +1889,8 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
   if (!direct) {
     bool all_nn = true;
     for (int a = 0; a < naggs; a++) all_nn &= (A.never_null[a] != 0);
-    if (all_nn && naggs <= 6 && g_est > 0) {
+    static const bool part_disabled = getenv("DSX_DISABLE_PART") != nullptr;
+    if (!part_disabled && all_nn && naggs <= 6 && g_est > 0) {
       bool fell_back = false;
       int prc = groupby_partition(c, C, n, K, P, A, progs, lens, naggs,
                                   g_est, out_codes, out_vals, out_counts,

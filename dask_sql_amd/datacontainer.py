@@ -11,13 +11,19 @@ from dask_sql_amd.runtime import DeviceColumn
 
 
 class DeviceTable:
-    """Ordered named device columns (all the same length)."""
+    """Ordered named device columns (all the same length). num_rows must be
+    given explicitly when the table has no columns (e.g. a fully pruned join
+    feeding COUNT(*))."""
 
-    def __init__(self, columns: dict[str, DeviceColumn]):
+    def __init__(self, columns: dict[str, DeviceColumn], num_rows=None):
         self.columns = dict(columns)
         lens = {c.len for c in self.columns.values()}
         assert len(lens) <= 1, f"ragged table: {lens}"
-        self.num_rows = lens.pop() if lens else 0
+        if lens:
+            self.num_rows = lens.pop()
+            assert num_rows is None or num_rows == self.num_rows
+        else:
+            self.num_rows = num_rows if num_rows is not None else 0
 
     def col(self, name) -> DeviceColumn:
         return self.columns[name]

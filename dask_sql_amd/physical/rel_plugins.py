@@ -275,7 +275,7 @@ class DaskJoinPlugin(BaseRelPlugin):
             out_cols[backend] = gathered[i]
             mapping[name] = backend
         cc = ColumnContainer(field_names, mapping)
-        return DataContainer(DeviceTable(out_cols), cc)
+        return DataContainer(DeviceTable(out_cols, num_rows=n_out), cc)
 
     # -- helpers ------------------------------------------------------------
     def _key_codes(self, runtime, dc, on, ranges):
@@ -465,7 +465,10 @@ class DaskAggregatePlugin(BaseRelPlugin):
         # and the projection arithmetic folded into each agg program. This is
         # what replaces the reference's filter→assign→groupby pass chain with
         # a single HBM scan.
-        fused = self._try_fused(runtime, rel, agg, context)
+        import os
+        fused = None
+        if not os.environ.get("DSX_DISABLE_FUSED"):
+            fused = self._try_fused(runtime, rel, agg, context)
         if fused is not None:
             return fused
 
