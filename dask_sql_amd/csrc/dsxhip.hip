@@ -63,6 +63,9 @@ struct DsxCtx {
   std::unordered_map<int64_t, std::vector<void*>> pool_free;
   std::unordered_map<void*, int64_t> pool_sizes;
   int64_t pool_cached = 0;
+  unsigned int* dbg_flag = nullptr;  // device word; bit0 gather OOB,
+                                     // bit1 probe-emit OOB (DSX_DEBUG)
+  bool debug = false;
 };
 
 static int64_t pool_round(int64_t bytes) {
@@ -134,7 +137,21 @@ extern "C" int dsx_ctx_create(int device_id, DsxCtx** out) {
     delete c;
     FAIL(-1, "stream create failed");
   }
+  c->debug = getenv("DSX_DEBUG") != nullptr;
+  HIP_TRY(hipMalloc((void**)&c->dbg_flag, 4));
+  HIP_TRY(hipMemset(c->dbg_flag, 0, 4));
   *out = c;
+  return 0;
+}
+
+static int dbg_check(DsxCtx* c, const char* what) {
+  if (!c->debug) return 0;
+  unsigned int f = 0;
+  HIP_TRY(hipMemcpyAsync(&f, c->dbg_flag, 4, hipMemcpyDeviceToHost,
+                         c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  if (f) FAIL(-9, "DEVICE BOUNDS VIOLATION (flags 0x%x) detected at %s", f,
+              what);
   return 0;
 }
 
@@ -663,7 +680,8 @@ extern "C" int dsx_filter(DsxCtx* c, const DsxInstr* prog, int prog_len,
 template <typename T>
 __global__ void k_gather(const T* in, const uint8_t* in_valid,
                          const uint32_t* sel, int64_t n, T* out,
-                         uint8_t* out_valid, T null_fill) {
+                         uint8_t* out_valid, T null_fill, int64_t n_src,
+                         unsigned int* dbg) {
   int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * BLOCK;
   for (; i < n; i += stride) {
@@ -673,6 +691,12 @@ __global__ void k_gather(const T* in, const uint8_t* in_valid,
       out[i] = null_fill;
       if (out_valid) out_valid[i] = 0;
     } else {
+      if ((int64_t)s >= n_src) {  // corrupt selection vector: report, skip
+        atomicOr(dbg, 1u);
+        out[i] = null_fill;
+        if (out_valid) out_valid[i] = 0;
+        continue;
+      }
       out[i] = in[s];
       uint8_t v = in_valid ? in_valid[s] : 1;
       if (out_valid) out_valid[i] = v;
@@ -689,37 +713,37 @@ extern "C" int dsx_gather(DsxCtx* c, const DsxColumn* col, const uint32_t* sel,
     case DSX_I64:
       hipLaunchKernelGGL(k_gather<int64_t>, dim3(grid), dim3(BLOCK), 0,
                          c->stream, (const int64_t*)col->data, col->validity,
-                         sel, n_sel, (int64_t*)out_data, out_validity, 0ll);
+                         sel, n_sel, (int64_t*)out_data, out_validity, 0ll, col->len, c->dbg_flag);
       break;
     case DSX_F64:
       hipLaunchKernelGGL(k_gather<double>, dim3(grid), dim3(BLOCK), 0,
                          c->stream, (const double*)col->data, col->validity,
                          sel, n_sel, (double*)out_data, out_validity,
-                         __builtin_nan(""));
+                         __builtin_nan(""), col->len, c->dbg_flag);
       break;
     case DSX_I32:
       hipLaunchKernelGGL(k_gather<int32_t>, dim3(grid), dim3(BLOCK), 0,
                          c->stream, (const int32_t*)col->data, col->validity,
-                         sel, n_sel, (int32_t*)out_data, out_validity, 0);
+                         sel, n_sel, (int32_t*)out_data, out_validity, 0, col->len, c->dbg_flag);
       break;
     case DSX_F32:
       hipLaunchKernelGGL(k_gather<float>, dim3(grid), dim3(BLOCK), 0,
                          c->stream, (const float*)col->data, col->validity, sel,
                          n_sel, (float*)out_data, out_validity,
-                         __builtin_nanf(""));
+                         __builtin_nanf(""), col->len, c->dbg_flag);
       break;
     case DSX_I8:
     case DSX_BOOL8:
       hipLaunchKernelGGL(k_gather<int8_t>, dim3(grid), dim3(BLOCK), 0,
                          c->stream, (const int8_t*)col->data, col->validity,
                          sel, n_sel, (int8_t*)out_data, out_validity,
-                         (int8_t)0);
+                         (int8_t)0, col->len, c->dbg_flag);
       break;
     default:
       FAIL(-3, "gather: bad dtype %d", col->dtype);
   }
   HIP_TRY(hipGetLastError());
-  return 0;
+  return dbg_check(c, "dsx_gather");
 }
 
 // ---------------------------------------------------------------------------
@@ -951,7 +975,8 @@ __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
                              const uint32_t* tvals, uint32_t* matched,
                              int64_t mask, int join_type,
                              int64_t* block_counts, uint32_t* out_p,
-                             uint32_t* out_b) {
+                             uint32_t* out_b, int64_t total,
+                             unsigned int* dbg) {
   __shared__ unsigned long long s_cnt;   // PASS0: block total; PASS1: bump
   if (threadIdx.x == 0)
     s_cnt = (PASS == 1) ? (unsigned long long)block_counts[blockIdx.x] : 0;
@@ -973,9 +998,13 @@ __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
           if (join_type == DSX_JOIN_INNER || join_type == DSX_JOIN_LEFT) {
             if (PASS == 1) {
               unsigned long long o = atomicAdd(&s_cnt, 1ull);
-              out_p[o] = (uint32_t)r;
-              out_b[o] = tvals[s];
-              matched[s] = 1;
+              if (o >= (unsigned long long)total) {
+                atomicOr(dbg, 2u);
+              } else {
+                out_p[o] = (uint32_t)r;
+                out_b[o] = tvals[s];
+                matched[s] = 1;
+              }
             }
             nmatch++;
           } else {  // SEMI / ANTI need existence only
@@ -997,12 +1026,18 @@ __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
     if (PASS == 1) {
       if (emit_nomatch) {
         unsigned long long o = atomicAdd(&s_cnt, 1ull);
-        out_p[o] = (uint32_t)r;
-        out_b[o] = DSX_NULL_IDX;
+        if (o >= (unsigned long long)total) atomicOr(dbg, 2u);
+        else {
+          out_p[o] = (uint32_t)r;
+          out_b[o] = DSX_NULL_IDX;
+        }
       } else if (emit_semi) {
         unsigned long long o = atomicAdd(&s_cnt, 1ull);
-        out_p[o] = (uint32_t)r;
-        out_b[o] = first_b;
+        if (o >= (unsigned long long)total) atomicOr(dbg, 2u);
+        else {
+          out_p[o] = (uint32_t)r;
+          out_b[o] = first_b;
+        }
       }
     } else {
       if (join_type == DSX_JOIN_INNER || join_type == DSX_JOIN_LEFT)
@@ -1037,7 +1072,7 @@ extern "C" int dsx_hash_probe(DsxCtx* c, DsxHashTable* t, const uint64_t* codes,
     hipLaunchKernelGGL(k_hash_probe<0>, dim3(grid), dim3(BLOCK), 0, c->stream,
                        codes, validity, n, t->keys, t->vals, t->matched,
                        t->slots - 1, join_type, block_counts, nullptr,
-                       nullptr);
+                       nullptr, 0, c->dbg_flag);
   }
   hipLaunchKernelGGL(k_scan_block_counts, dim3(1), dim3(64), 0, c->stream,
                      block_counts, grid, d_total);
@@ -1052,11 +1087,11 @@ extern "C" int dsx_hash_probe(DsxCtx* c, DsxHashTable* t, const uint64_t* codes,
     hipLaunchKernelGGL(k_hash_probe<1>, dim3(grid), dim3(BLOCK), 0, c->stream,
                        codes, validity, n, t->keys, t->vals, t->matched,
                        t->slots - 1, join_type, block_counts, *out_probe_idx,
-                       *out_build_idx);
+                       *out_build_idx, total, c->dbg_flag);
   }
   HIP_TRY(hipGetLastError());
   *out_count = total;
-  return 0;
+  return dbg_check(c, "dsx_hash_probe");
 }
 
 template <int PASS>
