@@ -138,20 +138,22 @@ extern "C" int dsx_ctx_create(int device_id, DsxCtx** out) {
     FAIL(-1, "stream create failed");
   }
   c->debug = getenv("DSX_DEBUG") != nullptr;
-  HIP_TRY(hipMalloc((void**)&c->dbg_flag, 4));
-  HIP_TRY(hipMemset(c->dbg_flag, 0, 4));
+  HIP_TRY(hipMalloc((void**)&c->dbg_flag, 16));
+  HIP_TRY(hipMemset(c->dbg_flag, 0, 16));
   *out = c;
   return 0;
 }
 
 static int dbg_check(DsxCtx* c, const char* what) {
   if (!c->debug) return 0;
-  unsigned int f = 0;
-  HIP_TRY(hipMemcpyAsync(&f, c->dbg_flag, 4, hipMemcpyDeviceToHost,
+  unsigned int f[4] = {0, 0, 0, 0};
+  HIP_TRY(hipMemcpyAsync(f, c->dbg_flag, 16, hipMemcpyDeviceToHost,
                          c->stream));
   HIP_TRY(hipStreamSynchronize(c->stream));
-  if (f) FAIL(-9, "DEVICE BOUNDS VIOLATION (flags 0x%x) detected at %s", f,
-              what);
+  if (f[0])
+    FAIL(-9,
+         "DEVICE BOUNDS VIOLATION (flags 0x%x, bad_sel=%u, n_src=%u, "
+         "pos=%u) at %s", f[0], f[1], f[2], f[3], what);
   return 0;
 }
 
@@ -692,7 +694,11 @@ __global__ void k_gather(const T* in, const uint8_t* in_valid,
       if (out_valid) out_valid[i] = 0;
     } else {
       if ((int64_t)s >= n_src) {  // corrupt selection vector: report, skip
-        atomicOr(dbg, 1u);
+        if (atomicOr(dbg, 1u) == 0) {
+          dbg[1] = s;
+          dbg[2] = (unsigned int)n_src;
+          dbg[3] = (unsigned int)i;
+        }
         out[i] = null_fill;
         if (out_valid) out_valid[i] = 0;
         continue;
