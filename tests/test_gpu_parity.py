@@ -367,3 +367,21 @@ def test_partition_stable_direct(ctx):
     for code in rng.choice(1000, 10):
         rows = np.nonzero(codes == code)[0]
         assert len(np.unique(got_buckets[rows])) == 1
+
+
+def test_filter_idle_block_word_alias(ctx):
+    """Regression: idle grid blocks (lo clamped to n) re-counted the last
+    partial mask word when n % 64 != 0 — filter count came out high by
+    popcount(last word) × idle blocks (found at Q3's customer scan)."""
+    rtm = ctx._get_runtime()
+    rng = np.random.default_rng(5)
+    seg = rng.integers(0, 5, 1_500_000).astype(np.int8)  # n % 64 == 32
+    col = rtm.upload_column(seg)
+    prog = rtm.make_prog([(1, 0, 0), (3, 0, 0), (34, 0, 0)])  # seg == 0
+    sel_ptr, count = rtm.filter(prog, [col], len(seg))
+    sel = rtm.wrap_sel(sel_ptr, count)
+    ids = np.empty(count, dtype=np.uint32)
+    rtm._download(sel.data, ids)
+    exp = np.nonzero(seg == 0)[0]
+    assert count == len(exp)
+    assert (ids.astype(np.int64) == exp).all()
