@@ -994,9 +994,18 @@ class DaskLimitPlugin(BaseRelPlugin):
                 and node.fetch is not None and node.offset == 0
                 and node.fetch <= 10_000):
             (inp,) = self.assert_inputs(below, 1, context)
-            from dask_sql_amd.materialize import to_pandas
-            pdf = to_pandas(inp, context)
             keys = below.sort().getCollation()
+            from dask_sql_amd.materialize import to_pandas
+            if isinstance(inp, DataContainer):
+                # device top-k: download ONLY the primary sort key, select
+                # candidates (k + boundary ties) on host, gather just those
+                # rows on device — avoids materializing the whole G-row
+                # frame (was the entire Q3 host cost)
+                pdf = _device_topk_impl(context, inp, below, keys,
+                                        node.fetch)
+                if pdf is not None:
+                    return HostDataContainer(pdf.reset_index(drop=True))
+            pdf = to_pandas(inp, context)
             pdf = _topk(pdf, keys, node.fetch)
             return HostDataContainer(pdf.reset_index(drop=True))
         (inp,) = self.assert_inputs(rel, 1, context)
@@ -1010,6 +1019,34 @@ class DaskLimitPlugin(BaseRelPlugin):
         if node.fetch is not None:
             pdf = pdf.iloc[: node.fetch]
         return HostDataContainer(pdf.reset_index(drop=True))
+
+
+def _device_topk_impl(context, inp, below, keys, k):
+    runtime = context._get_runtime()
+    cc = inp.column_container
+    n = inp.table.num_rows
+    if n <= max(4 * k, 4096):
+        return None  # small: plain path is fine
+    idx0, asc0, _ = keys[0]
+    col0 = inp.table.col(cc.get_backend_by_frontend_name(cc.columns[idx0]))
+    v, valid = col0.to_numpy()
+    if valid is not None and not valid.all():
+        return None  # NULL keys: fall back (NULLS FIRST/LAST ordering)
+    v = np.asarray(v)
+    if v.dtype.kind == "f" and np.isnan(v).any():
+        return None
+    key_arr = v if asc0 else -v.astype(np.float64 if v.dtype.kind == "f"
+                                       else np.int64)
+    part = np.argpartition(key_arr, k - 1)[:k]
+    thresh = key_arr[part].max()
+    cand = np.nonzero(key_arr <= thresh)[0]
+    if len(cand) > max(20 * k, 20_000):
+        return None  # degenerate ties
+    sel = runtime.upload_column(cand.astype(np.uint32), dtype=rt.I32)
+    cand_dc = _gather_table(runtime, inp, sel.data, len(cand))
+    from dask_sql_amd.materialize import to_pandas
+    pdf = to_pandas(cand_dc, context, below.getRowType())
+    return _topk(pdf, keys, k)
 
 
 def _topk(pdf, keys, k):
