@@ -905,30 +905,34 @@ extern "C" int dsx_keypack(DsxCtx* c, const DsxColumn* cols, int ncols,
 // duplicates occupy separate slots; probe scans to first EMPTY.
 // ---------------------------------------------------------------------------
 struct DsxHashTable {
-  uint64_t* keys = nullptr;   // EMPTY_KEY = empty
-  uint32_t* vals = nullptr;   // build row id
+  uint64_t* keys = nullptr;   // EMPTY_KEY = empty; packed: (code<<32)|rowid
+  uint32_t* vals = nullptr;   // build row id (unpacked layout only)
   uint32_t* matched = nullptr;
   int64_t slots = 0;
   int64_t n_build = 0;
+  int packed = 0;             // codes < 2^32-1 → one 8-B entry per slot,
+                              // single random read per probe
   DsxCtx* ctx = nullptr;
 };
 
 __global__ void k_hash_build(const uint64_t* codes, const uint8_t* validity,
                              int64_t n, uint64_t* tkeys, uint32_t* tvals,
-                             int64_t mask) {
+                             int64_t mask, int packed) {
   int64_t lo, hi;
   block_range(n, 1, lo, hi);
   for (int64_t r = lo + threadIdx.x; r < hi; r += BLOCK) {
     if (validity && !validity[r]) continue;  // NULL-key drop (join.py:202-213)
     uint64_t cde = codes[r];
+    uint64_t entry = packed ? ((cde << 32) | (uint64_t)(uint32_t)r) : cde;
     int64_t s = (int64_t)(mix64(cde) & mask);
     while (true) {
       unsigned long long old = atomicCAS((unsigned long long*)&tkeys[s],
-                                         EMPTY_KEY, (unsigned long long)cde);
+                                         EMPTY_KEY,
+                                         (unsigned long long)entry);
       if (old == EMPTY_KEY) {
-        tvals[s] = (uint32_t)r;  // visible to later kernels via end-of-kernel
-                                 // release (build and probe are separate
-                                 // dispatches on one stream)
+        if (!packed)
+          tvals[s] = (uint32_t)r;  // visible to later kernels via
+                                   // end-of-kernel release (same stream)
         break;
       }
       s = (s + 1) & mask;
@@ -938,7 +942,7 @@ __global__ void k_hash_build(const uint64_t* codes, const uint8_t* validity,
 
 extern "C" int dsx_hash_build(DsxCtx* c, const uint64_t* codes,
                               const uint8_t* validity, int64_t n,
-                              DsxHashTable** out) {
+                              uint64_t code_max, DsxHashTable** out) {
   if (n > 0xFFFFFFFEll) FAIL(-3, "build side too large for u32 row ids");
   int64_t slots = 64;
   while (slots < 2 * n) slots <<= 1;
@@ -946,8 +950,11 @@ extern "C" int dsx_hash_build(DsxCtx* c, const uint64_t* codes,
   t->slots = slots;
   t->n_build = n;
   t->ctx = c;
+  // codes bounded below 2^32-1 → pack (code,rowid) into the claim word:
+  // one random 8-B read per probe instead of two, half the table bytes
+  t->packed = (code_max != 0 && code_max < 0xFFFFFFFEull) ? 1 : 0;
   if (pool_alloc(c, slots * 8, (void**)&t->keys) ||
-      pool_alloc(c, slots * 4, (void**)&t->vals) ||
+      (!t->packed && pool_alloc(c, slots * 4, (void**)&t->vals)) ||
       pool_alloc(c, slots * 4, (void**)&t->matched)) {
     dsx_hash_table_free(t);
     FAIL(-2, "hash table alloc failed (%lld slots)", (long long)slots);
@@ -958,7 +965,8 @@ extern "C" int dsx_hash_build(DsxCtx* c, const uint64_t* codes,
   if (grid > 0) {
     ProfScope ps(c, "k_hash_build");
     hipLaunchKernelGGL(k_hash_build, dim3(grid), dim3(BLOCK), 0, c->stream,
-                       codes, validity, n, t->keys, t->vals, slots - 1);
+                       codes, validity, n, t->keys, t->vals, slots - 1,
+                       t->packed);
   }
   HIP_TRY(hipGetLastError());
   *out = t;
@@ -982,9 +990,9 @@ template <int PASS>
 __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
                              int64_t n, const uint64_t* tkeys,
                              const uint32_t* tvals, uint32_t* matched,
-                             int64_t mask, int join_type,
-                             int64_t* block_counts, uint32_t* out_p,
-                             uint32_t* out_b, int64_t total,
+                             int64_t mask, int join_type, int packed,
+                             int mark_matched, int64_t* block_counts,
+                             uint32_t* out_p, uint32_t* out_b, int64_t total,
                              unsigned int* dbg) {
   __shared__ unsigned long long s_cnt;   // PASS0: block total; PASS1: bump
   if (threadIdx.x == 0)
@@ -1003,7 +1011,9 @@ __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
       while (true) {
         uint64_t k = tkeys[s];
         if (k == EMPTY_KEY) break;
-        if (k == cde) {
+        uint64_t kc = packed ? (k >> 32) : k;
+        if (kc == cde) {
+          uint32_t bid = packed ? (uint32_t)k : tvals[s];
           if (join_type == DSX_JOIN_INNER || join_type == DSX_JOIN_LEFT) {
             if (PASS == 1) {
               unsigned long long o = atomicAdd(&s_cnt, 1ull);
@@ -1011,14 +1021,14 @@ __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
                 atomicOr(dbg, 2u);
               } else {
                 out_p[o] = (uint32_t)r;
-                out_b[o] = tvals[s];
-                matched[s] = 1;
+                out_b[o] = bid;
+                if (mark_matched) matched[s] = 1;  // FULL OUTER sweep only
               }
             }
             nmatch++;
           } else {  // SEMI / ANTI need existence only
             nmatch++;
-            first_b = tvals[s];
+            first_b = bid;
             break;
           }
         }
@@ -1065,6 +1075,7 @@ __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
 
 extern "C" int dsx_hash_probe(DsxCtx* c, DsxHashTable* t, const uint64_t* codes,
                               const uint8_t* validity, int64_t n, int join_type,
+                              int mark_matched,
                               uint32_t** out_probe_idx, uint32_t** out_build_idx,
                               int64_t* out_count) {
   if (n > 0xFFFFFFFEll) FAIL(-3, "probe side too large for u32 row ids");
@@ -1080,8 +1091,8 @@ extern "C" int dsx_hash_probe(DsxCtx* c, DsxHashTable* t, const uint64_t* codes,
     ProfScope ps(c, "k_hash_probe_count");
     hipLaunchKernelGGL(k_hash_probe<0>, dim3(grid), dim3(BLOCK), 0, c->stream,
                        codes, validity, n, t->keys, t->vals, t->matched,
-                       t->slots - 1, join_type, block_counts, nullptr,
-                       nullptr, 0, c->dbg_flag);
+                       t->slots - 1, join_type, t->packed, mark_matched,
+                       block_counts, nullptr, nullptr, 0, c->dbg_flag);
   }
   hipLaunchKernelGGL(k_scan_block_counts, dim3(1), dim3(64), 0, c->stream,
                      block_counts, grid, d_total);
@@ -1095,7 +1106,8 @@ extern "C" int dsx_hash_probe(DsxCtx* c, DsxHashTable* t, const uint64_t* codes,
     ProfScope ps(c, "k_hash_probe_emit");
     hipLaunchKernelGGL(k_hash_probe<1>, dim3(grid), dim3(BLOCK), 0, c->stream,
                        codes, validity, n, t->keys, t->vals, t->matched,
-                       t->slots - 1, join_type, block_counts, *out_probe_idx,
+                       t->slots - 1, join_type, t->packed, mark_matched,
+                       block_counts, *out_probe_idx,
                        *out_build_idx, total, c->dbg_flag);
   }
   HIP_TRY(hipGetLastError());
@@ -1106,13 +1118,15 @@ extern "C" int dsx_hash_probe(DsxCtx* c, DsxHashTable* t, const uint64_t* codes,
 template <int PASS>
 __global__ void k_unmatched(const uint64_t* tkeys, const uint32_t* tvals,
                             const uint32_t* matched, int64_t slots,
-                            unsigned long long* counter, uint32_t* out_b) {
+                            int packed, unsigned long long* counter,
+                            uint32_t* out_b) {
   int64_t lo, hi;
   block_range(slots, 1, lo, hi);
   for (int64_t s = lo + threadIdx.x; s < hi; s += BLOCK) {
     if (tkeys[s] != EMPTY_KEY && !matched[s]) {
       unsigned long long o = atomicAdd(counter, 1ull);
-      if (PASS == 1) out_b[o] = tvals[s];
+      if (PASS == 1)
+        out_b[o] = packed ? (uint32_t)tkeys[s] : tvals[s];
     }
   }
 }
@@ -1127,7 +1141,8 @@ extern "C" int dsx_hash_unmatched(DsxCtx* c, DsxHashTable* t,
   HIP_TRY(hipMemsetAsync(counter, 0, 8, c->stream));
   int grid = (int)min((int64_t)MAX_GRID, (t->slots + BLOCK - 1) / BLOCK);
   hipLaunchKernelGGL(k_unmatched<0>, dim3(grid), dim3(BLOCK), 0, c->stream,
-                     t->keys, t->vals, t->matched, t->slots, counter, nullptr);
+                     t->keys, t->vals, t->matched, t->slots, t->packed,
+                     counter, nullptr);
   unsigned long long total = 0;
   HIP_TRY(hipMemcpyAsync(&total, counter, 8, hipMemcpyDeviceToHost, c->stream));
   HIP_TRY(hipStreamSynchronize(c->stream));
@@ -1135,8 +1150,8 @@ extern "C" int dsx_hash_unmatched(DsxCtx* c, DsxHashTable* t,
   HIP_TRY(hipMemsetAsync(counter, 0, 8, c->stream));
   if (total > 0)
     hipLaunchKernelGGL(k_unmatched<1>, dim3(grid), dim3(BLOCK), 0, c->stream,
-                       t->keys, t->vals, t->matched, t->slots, counter,
-                       *out_build_idx);
+                       t->keys, t->vals, t->matched, t->slots, t->packed,
+                       counter, *out_build_idx);
   HIP_TRY(hipGetLastError());
   *out_count = (int64_t)total;
   return 0;

@@ -348,10 +348,14 @@ class DaskJoinPlugin(BaseRelPlugin):
                                               ranges)
         pcodes, pval, pkeep = self._key_codes(runtime, probe_dc, probe_on,
                                               ranges)
-        table = runtime.hash_build(bcodes, bval)
+        space = 1
+        for (_, rng) in ranges:
+            space *= rng
+        table = runtime.hash_build(bcodes, bval, code_max=space - 1)
         try:
-            p_ptr, b_ptr, count = runtime.hash_probe(table, pcodes, ktype,
-                                                     pval)
+            p_ptr, b_ptr, count = runtime.hash_probe(
+                table, pcodes, ktype, pval,
+                mark_matched=(join_type == "outer"))
             probe_sel = runtime.wrap_sel(p_ptr, count)
             build_sel = runtime.wrap_sel(b_ptr, count)
             if join_type == "outer":

@@ -352,19 +352,20 @@ class Runtime:
         )
         return out, space
 
-    def hash_build(self, codes: DeviceColumn, validity_ptr=None):
+    def hash_build(self, codes: DeviceColumn, validity_ptr=None, code_max=0):
         t = ct.c_void_p()
         _check(
             self.lib,
             self.lib.dsx_hash_build(self.ctx, ct.c_void_p(codes.data),
                                     ct.c_void_p(validity_ptr) if validity_ptr else None,
-                                    ct.c_int64(codes.len), ct.byref(t)),
+                                    ct.c_int64(codes.len),
+                                    ct.c_uint64(code_max), ct.byref(t)),
             "dsx_hash_build",
         )
         return t
 
     def hash_probe(self, table, codes: DeviceColumn, join_type,
-                   validity_ptr=None):
+                   validity_ptr=None, mark_matched=False):
         p = ct.c_void_p()
         b = ct.c_void_p()
         count = ct.c_int64()
@@ -373,6 +374,7 @@ class Runtime:
             self.lib.dsx_hash_probe(self.ctx, table, ct.c_void_p(codes.data),
                                     ct.c_void_p(validity_ptr) if validity_ptr else None,
                                     ct.c_int64(codes.len), ct.c_int(join_type),
+                                    ct.c_int(1 if mark_matched else 0),
                                     ct.byref(p), ct.byref(b), ct.byref(count)),
             "dsx_hash_probe",
         )

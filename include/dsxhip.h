@@ -161,8 +161,10 @@ int dsx_keypack(DsxCtx* ctx, const DsxColumn* cols, int ncols,
  * inserted — the NULL-key drop of join.py:202-213 for the build side.
  * Table is library-allocated; free with dsx_hash_table_free. */
 typedef struct DsxHashTable DsxHashTable;
+/* code_max: max possible key code (0 = unknown). codes below 2^32-1 pack
+ * (code,rowid) into one 8-byte slot — one random read per probe. */
 int dsx_hash_build(DsxCtx* ctx, const uint64_t* codes, const uint8_t* validity,
-                   int64_t n, DsxHashTable** out);
+                   int64_t n, uint64_t code_max, DsxHashTable** out);
 void dsx_hash_table_free(DsxHashTable* t);
 
 enum DsxJoinType {  /* reference join.py:41-48 JOIN_TYPE_MAPPING */
@@ -176,8 +178,11 @@ enum DsxJoinType {  /* reference join.py:41-48 JOIN_TYPE_MAPPING */
 /* Probe: emits (probe_rowid, build_rowid) pairs, library-allocated.
  * Unordered (normalize by sort for comparisons); FULL OUTER is composed by
  * the host from LEFT + unmatched-build sweep (dsx_hash_unmatched). */
+/* mark_matched: record matched build slots (needed only when
+ * dsx_hash_unmatched will run, i.e. FULL OUTER). */
 int dsx_hash_probe(DsxCtx* ctx, DsxHashTable* t, const uint64_t* codes,
                    const uint8_t* validity, int64_t n, int join_type,
+                   int mark_matched,
                    uint32_t** out_probe_idx, uint32_t** out_build_idx,
                    int64_t* out_count);
 /* build rows never matched by any probe since build (for FULL OUTER,
