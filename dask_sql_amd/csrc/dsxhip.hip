@@ -66,6 +66,7 @@ struct DsxCtx {
   unsigned int* dbg_flag = nullptr;  // device word; bit0 gather OOB,
                                      // bit1 probe-emit OOB (DSX_DEBUG)
   bool debug = false;
+  void* jit_cache = nullptr;  // JitCacheMap (jit.inc)
 };
 
 static int64_t pool_round(int64_t bytes) {
@@ -157,9 +158,12 @@ static int dbg_check(DsxCtx* c, const char* what) {
   return 0;
 }
 
+void jit_cache_destroy(DsxCtx* c);  // defined after jit includes
+
 extern "C" void dsx_ctx_destroy(DsxCtx* c) {
   if (!c) return;
   hipStreamSynchronize(c->stream);
+  jit_cache_destroy(c);
   for (auto& kv : c->pool_free)
     for (void* p : kv.second) hipFree(p);
   for (auto& kv : c->pool_sizes) (void)kv;
@@ -1219,7 +1223,7 @@ __device__ __forceinline__ void agg_update_global(int op, int never_null,
   if (!never_null) atomicAdd(cnt, 1ull);
 }
 
-__device__ __forceinline__ uint64_t agg_identity(int op) {
+__host__ __device__ __forceinline__ uint64_t agg_identity(int op) {
   switch (op) {
     case DSX_AGG_MIN_F64:
     case DSX_AGG_MIN_I64:
@@ -1231,6 +1235,9 @@ __device__ __forceinline__ uint64_t agg_identity(int op) {
       return 0;  // sums
   }
 }
+
+#include "jit.inc"
+#include "jit_kernels.inc"
 
 // ---- LDS direct-index path -------------------------------------------------
 // LDS layout: [key_space × naggs] u64 vals, [key_space × naggs] u32 cnts,
@@ -1745,7 +1752,7 @@ __global__ void k_count_live(const uint64_t* tkeys,
 // host driver for the partition path; *fell_back=true → caller uses the CAS
 // path (LDS table overflow from key skew, or shapes it does not cover)
 static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
-                             ProgArg& P, AggArg& A,
+                             ProgArg& P, AggArg& A, const DsxAggSpec* aggs_arr,
                              std::vector<DsxInstr>& progs,
                              std::vector<int32_t>& lens, int naggs,
                              int64_t g_est, uint64_t** out_codes,
@@ -1831,21 +1838,50 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   HIP_TRY(hipMemsetAsync(d_counter, 0, 8, c->stream));
   HIP_TRY(hipMemsetAsync(d_ovf, 0, 4, c->stream));
 
-  {
-    ProfScope ps(c, "k_gbpart_hist");
-    hipLaunchKernelGGL(k_gbpart_hist, dim3(grid), dim3(BLOCK),
-                       (size_t)nb * 4, c->stream, C, n, d_K, P, nb, d_hist);
-  }
-  hipLaunchKernelGGL(k_gbpart_scan, dim3(nb), dim3(BLOCK), 0, c->stream,
-                     d_hist, grid, nb, d_totals);
-  hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(64), 0, c->stream,
-                     d_totals, nb, d_bases);
-  {
-    ProfScope ps(c, "k_gbpart_scatter");
-    hipLaunchKernelGGL(k_gbpart_scatter, dim3(grid), dim3(BLOCK),
-                       (size_t)nb * 8, c->stream, C, n, d_K, P, d_progs,
-                       d_lens, d_val_of, naggs, nvals, nb, d_hist, d_bases,
-                       d_recs);
+  JitEntry* je = jit_source_entry(
+      c, jit_gbpart_source(C, K, P, aggs_arr, val_of, naggs, nvals));
+  hipFunction_t f_hist = je ? jit_fn(c, je, "j_hist") : nullptr;
+  hipFunction_t f_scat = je ? jit_fn(c, je, "j_scatter") : nullptr;
+  if (f_hist && f_scat) {
+    {
+      ProfScope ps(c, "k_gbpart_hist");
+      struct { ColsArg C; int64_t n; int nb; int64_t* hist; } a1{C, n, nb,
+                                                                 d_hist};
+      void* args[] = {&a1.C, &a1.n, &a1.nb, &a1.hist};
+      hipModuleLaunchKernel(f_hist, grid, 1, 1, BLOCK, 1, 1,
+                            (unsigned)(nb * 4), c->stream, args, nullptr);
+    }
+    hipLaunchKernelGGL(k_gbpart_scan, dim3(nb), dim3(BLOCK), 0, c->stream,
+                       d_hist, grid, nb, d_totals);
+    hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(64), 0, c->stream,
+                       d_totals, nb, d_bases);
+    {
+      ProfScope ps(c, "k_gbpart_scatter");
+      struct {
+        ColsArg C; int64_t n; int nb; const int64_t* hist;
+        const int64_t* bases; uint64_t* out;
+      } a2{C, n, nb, d_hist, d_bases, d_recs};
+      void* args[] = {&a2.C, &a2.n, &a2.nb, &a2.hist, &a2.bases, &a2.out};
+      hipModuleLaunchKernel(f_scat, grid, 1, 1, BLOCK, 1, 1,
+                            (unsigned)(nb * 8), c->stream, args, nullptr);
+    }
+  } else {
+    {
+      ProfScope ps(c, "k_gbpart_hist");
+      hipLaunchKernelGGL(k_gbpart_hist, dim3(grid), dim3(BLOCK),
+                         (size_t)nb * 4, c->stream, C, n, d_K, P, nb, d_hist);
+    }
+    hipLaunchKernelGGL(k_gbpart_scan, dim3(nb), dim3(BLOCK), 0, c->stream,
+                       d_hist, grid, nb, d_totals);
+    hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(64), 0, c->stream,
+                       d_totals, nb, d_bases);
+    {
+      ProfScope ps(c, "k_gbpart_scatter");
+      hipLaunchKernelGGL(k_gbpart_scatter, dim3(grid), dim3(BLOCK),
+                         (size_t)nb * 8, c->stream, C, n, d_K, P, d_progs,
+                         d_lens, d_val_of, naggs, nvals, nb, d_hist, d_bases,
+                         d_recs);
+    }
   }
   {
     ProfScope ps(c, "k_gbpart_aggregate");
@@ -1951,7 +1987,7 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
     static const bool part_disabled = getenv("DSX_DISABLE_PART") != nullptr;
     if (!part_disabled && all_nn && naggs <= 6 && g_est > 0) {
       bool fell_back = false;
-      int prc = groupby_partition(c, C, n, K, P, A, progs, lens, naggs,
+      int prc = groupby_partition(c, C, n, K, P, A, aggs, progs, lens, naggs,
                                   g_est, out_codes, out_vals, out_counts,
                                   out_groups, &fell_back);
       if (prc != 0) return prc;
@@ -2026,17 +2062,28 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
     int grid = (int)min((int64_t)MAX_GRID, (n + BLOCK - 1) / BLOCK);
     if (grid > 0) {
       if (direct) {
-        int64_t lds = key_space * lds_per_slot;
-        // round LDS arrays: vals 8B aligned first, then u32 arrays
         size_t lds_bytes =
             (size_t)(naggs * (int64_t)key_space * 8 +
                      naggs * (int64_t)key_space * 4 + (int64_t)key_space * 4);
-        (void)lds;
+        JitEntry* je = jit_source_entry(
+            c, jit_direct_source(C, K, P, aggs, A, naggs, (int)key_space));
+        hipFunction_t fd = je ? jit_fn(c, je, "j_direct") : nullptr;
         ProfScope ps(c, "k_groupby_direct");
-        hipLaunchKernelGGL(k_groupby_direct, dim3(grid), dim3(BLOCK),
-                           lds_bytes, c->stream, C, n, d_K,
-                           (int)key_space, P, d_progs, d_lens, d_A, d_vals,
-                           d_cnts, d_gcnt);
+        if (fd) {
+          struct {
+            ColsArg C; int64_t n; uint64_t* v; unsigned long long* cn;
+            unsigned long long* gc;
+          } a0{C, n, d_vals, d_cnts, d_gcnt};
+          void* args[] = {&a0.C, &a0.n, &a0.v, &a0.cn, &a0.gc};
+          hipModuleLaunchKernel(fd, grid, 1, 1, BLOCK, 1, 1,
+                                (unsigned)lds_bytes, c->stream, args,
+                                nullptr);
+        } else {
+          hipLaunchKernelGGL(k_groupby_direct, dim3(grid), dim3(BLOCK),
+                             lds_bytes, c->stream, C, n, d_K,
+                             (int)key_space, P, d_progs, d_lens, d_A, d_vals,
+                             d_cnts, d_gcnt);
+        }
       } else {
         ProfScope ps(c, "k_groupby_global");
         hipLaunchKernelGGL(k_groupby_global, dim3(grid), dim3(BLOCK), 0,
@@ -2215,4 +2262,12 @@ extern "C" int dsx_partition(DsxCtx* c, const uint64_t* codes,
   HIP_TRY(hipStreamSynchronize(c->stream));
   HIP_TRY(hipGetLastError());
   return 0;
+}
+
+void jit_cache_destroy(DsxCtx* c) {
+  if (!c->jit_cache) return;
+  for (auto& kv : ((JitCacheMap*)c->jit_cache)->m)
+    if (kv.second.mod) hipModuleUnload(kv.second.mod);
+  delete (JitCacheMap*)c->jit_cache;
+  c->jit_cache = nullptr;
 }
