@@ -1839,9 +1839,11 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   HIP_TRY(hipMemsetAsync(d_ovf, 0, 4, c->stream));
 
   JitEntry* je = jit_source_entry(
-      c, jit_gbpart_source(C, K, P, aggs_arr, val_of, naggs, nvals));
+      c, jit_gbpart_source(C, K, P, aggs_arr, A, val_of, naggs, nvals,
+                           lds_slots));
   hipFunction_t f_hist = je ? jit_fn(c, je, "j_hist") : nullptr;
   hipFunction_t f_scat = je ? jit_fn(c, je, "j_scatter") : nullptr;
+  hipFunction_t f_aggr = je ? jit_fn(c, je, "j_aggregate") : nullptr;
   if (f_hist && f_scat) {
     {
       ProfScope ps(c, "k_gbpart_hist");
@@ -1885,10 +1887,23 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   }
   {
     ProfScope ps(c, "k_gbpart_aggregate");
-    hipLaunchKernelGGL(k_gbpart_aggregate, dim3(nb), dim3(BLOCK),
-                       lds_bytes, c->stream, d_recs, d_bases,
-                       nvals ? nvals : 0, d_A, lds_slots, d_counter,
-                       d_tmp_codes, d_tmp_vals, d_tmp_gcnt, d_ovf);
+    if (f_aggr) {
+      struct {
+        const uint64_t* recs; const int64_t* bases;
+        unsigned long long* counter; uint64_t* tc; uint64_t* tv;
+        unsigned long long* tg; int* ovf;
+      } a3{d_recs, d_bases, d_counter, d_tmp_codes,
+           (uint64_t*)d_tmp_vals, d_tmp_gcnt, d_ovf};
+      void* args[] = {&a3.recs, &a3.bases, &a3.counter, &a3.tc, &a3.tv,
+                      &a3.tg, &a3.ovf};
+      hipModuleLaunchKernel(f_aggr, nb, 1, 1, BLOCK, 1, 1,
+                            (unsigned)lds_bytes, c->stream, args, nullptr);
+    } else {
+      hipLaunchKernelGGL(k_gbpart_aggregate, dim3(nb), dim3(BLOCK),
+                         lds_bytes, c->stream, d_recs, d_bases,
+                         nvals ? nvals : 0, d_A, lds_slots, d_counter,
+                         d_tmp_codes, d_tmp_vals, d_tmp_gcnt, d_ovf);
+    }
   }
   int h_ovf = 0;
   unsigned long long G = 0;
