@@ -1763,13 +1763,12 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   int nvals = 0;
   for (int a = 0; a < naggs; a++)
     val_of[a] = (A.op[a] == DSX_AGG_COUNT) ? -1 : nvals++;
-  // fewer buckets → longer per-(block,bucket) write runs → the scattered
-  // 16-B record stores merge into full lines in the XCD L2 before eviction
-  // (measured 3.3× write amplification at 1024 buckets / ~24-row runs);
-  // bounded by the aggregate kernel's LDS table (slots ≤ 4096 ⇒ ≤ 2048
-  // groups per bucket at load factor 0.5)
+  // bucket-count tradeoff (measured): fewer buckets lengthen write runs
+  // (less 16-B-store line-merge loss in L2) but shrink the aggregate grid
+  // and grow its LDS table; 1024 targets the best total at C2 shape
+  // (512: scatter −0.17 ms but aggregate +0.81 ms)
   int nb = 64;
-  while (nb < 4096 && g_est / nb > 2048) nb <<= 1;
+  while (nb < 4096 && g_est / nb > 1024) nb <<= 1;
   int64_t per_bucket = (g_est + nb - 1) / nb;
   int lds_slots = 256;
   while (lds_slots < 2 * per_bucket) lds_slots <<= 1;
