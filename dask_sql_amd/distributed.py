@@ -75,18 +75,15 @@ def exchange_buckets(tensors: list[torch.Tensor], in_splits: list[int],
         if dist.get_backend(group) == "nccl":
             dist.all_to_all_single(out, t, out_splits, in_splits, group=group)
         else:
-            # gloo has no all_to_all: emulate with all_gather of full tensors
-            gathered = [torch.empty(int(in_tot.sum().item()), dtype=t.dtype)
-                        for in_tot in all_counts]
-            # need per-rank full sizes
-            sizes = [int(c.sum().item()) for c in all_counts]
-            gathered = [torch.empty(s, dtype=t.dtype) for s in sizes]
-            dist.all_gather(gathered, t.contiguous(), group=group)
+            # gloo has no all_to_all (and all_gather needs equal sizes):
+            # object-gather the full tensors (CPU test path only)
+            objs = [None] * world
+            dist.all_gather_object(objs, t.contiguous().cpu(), group=group)
             parts = []
             for r in range(world):
                 ofs = [0] + list(np.cumsum(
                     [int(all_counts[r][i].item()) for i in range(world)]))
-                parts.append(gathered[r][ofs[me]:ofs[me + 1]])
+                parts.append(objs[r][ofs[me]:ofs[me + 1]])
             out = torch.cat(parts) if parts else out
         received.append(out)
     return received, out_splits
