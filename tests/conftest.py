@@ -3,6 +3,10 @@ import os
 import sys
 from pathlib import Path
 
+# GPU tests run with device-side bounds validation active (runtime reads the
+# env at context creation; production default is off)
+os.environ.setdefault("DSX_DEBUG", "1")
+
 import numpy as np
 import pandas as pd
 import pytest

@@ -1,0 +1,211 @@
+"""GPU semantics suite (-m gpu): broader operator coverage, product path vs
+the oracle restatement (checker only), including a randomized differential
+sweep. Complements the golden suite in test_gpu_parity.py."""
+import numpy as np
+import pandas as pd
+import pytest
+
+from oracle.frame import oracle_filter, oracle_groupby, oracle_join
+from tests.conftest import assert_frame_close
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def ctx():
+    import torch
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from dask_sql_amd.context import Context
+    return Context()
+
+
+def _rand_frame(rng, n, with_nulls=True):
+    k = rng.integers(0, 7, n).astype(np.int64)
+    v = np.round(rng.random(n) * 100, 3)
+    w = rng.integers(-5, 20, n).astype(np.int64)
+    df = pd.DataFrame({"k": k, "v": v, "w": w})
+    if with_nulls:
+        mask = rng.random(n) < 0.15
+        df.loc[mask, "v"] = np.nan
+        kk = pd.array(df["k"], dtype="Int64")
+        kk[rng.random(n) < 0.1] = pd.NA
+        df["k"] = kk
+    return df
+
+
+def test_having(ctx, ):
+    rng = np.random.default_rng(1)
+    df = _rand_frame(rng, 5000, with_nulls=False)
+    ctx.create_table("th", df)
+    out = ctx.sql("SELECT k, SUM(w) AS s FROM th GROUP BY k "
+                  "HAVING SUM(w) > 100").compute()
+    exp = oracle_groupby(df, ["k"], [("w", "s", "sum", None, False)])
+    exp = exp[exp["s"] > 100]
+    assert_frame_close(out.sort_values("k").reset_index(drop=True),
+                       exp.sort_values("k").reset_index(drop=True))
+
+
+def test_in_list_and_between_and_not(ctx):
+    rng = np.random.default_rng(2)
+    df = _rand_frame(rng, 5000, with_nulls=False)
+    ctx.create_table("tibn", df)
+    out = ctx.sql("SELECT w FROM tibn WHERE k IN (1, 3, 5) "
+                  "AND w BETWEEN 0 AND 10 AND NOT (w = 7)").compute()
+    m = df["k"].isin([1, 3, 5]) & df["w"].between(0, 10) & (df["w"] != 7)
+    exp = df[m][["w"]]
+    assert sorted(out["w"].astype(np.int64).tolist()) == \
+        sorted(exp["w"].tolist())
+
+
+def test_cast_trunc(ctx):
+    df = pd.DataFrame({"x": [1.9, -1.9, 2.5, 0.1]})
+    ctx.create_table("tc", df)
+    out = ctx.sql("SELECT CAST(x AS BIGINT) AS i FROM tc").compute()
+    # mappings.py:346-353: float→int truncates
+    assert out["i"].astype(np.int64).tolist() == [1, -1, 2, 0]
+
+
+def test_date_column_materializes(ctx):
+    days = np.array([9000, 9500, 10000], dtype=np.int32)
+    ctx.create_table("td", pd.DataFrame({"d": days}), date_columns=["d"])
+    out = ctx.sql("SELECT d FROM td WHERE d >= DATE '1996-01-01'").compute()
+    assert str(out["d"].dtype).startswith("datetime64")
+    exp = pd.to_datetime(days[days >= 9497], unit="D")
+    assert list(out["d"]) == list(exp)
+
+
+def test_right_join(ctx, user_table_1, user_table_2):
+    ctx.create_table("u1", user_table_1)
+    ctx.create_table("u2", user_table_2)
+    out = ctx.sql("SELECT lhs.user_id, lhs.b, rhs.c FROM u1 lhs "
+                  "RIGHT JOIN u2 rhs ON lhs.user_id = rhs.user_id").compute()
+    exp = oracle_join(user_table_1, user_table_2, [0], [0], "RIGHT")
+    exp = exp[["lhs_0", "lhs_1", "rhs_1"]]
+    exp.columns = ["user_id", "b", "c"]
+    assert_frame_close(out, exp, sort_by=["c", "user_id"])
+
+
+def test_left_anti_join(ctx):
+    df1 = pd.DataFrame({"id": [1, 1, 2, 4], "a": [10, 11, 12, 13]})
+    df2 = pd.DataFrame({"id": [2, 1, 2, 3], "b": [7, 7, 8, 7]})
+    ctx.create_table("la1", df1)
+    ctx.create_table("la2", df2)
+    out = ctx.sql("SELECT lhs.id, lhs.a FROM la1 lhs LEFT ANTI JOIN la2 rhs "
+                  "ON lhs.id = rhs.id").compute()
+    assert out["id"].astype(np.int64).tolist() == [4]
+    assert out["a"].astype(np.int64).tolist() == [13]
+
+
+def test_multikey_groupby_with_nulls(ctx):
+    rng = np.random.default_rng(3)
+    df = _rand_frame(rng, 8000, with_nulls=True)
+    df["k2"] = rng.integers(0, 3, len(df)).astype(np.int64)
+    ctx.create_table("tmk", df)
+    out = ctx.sql("SELECT k, k2, SUM(v) AS s, COUNT(v) AS c, MIN(w) AS mn, "
+                  "MAX(w) AS mx, AVG(v) AS a FROM tmk GROUP BY k, k2"
+                  ).compute()
+    exp = oracle_groupby(df, ["k", "k2"], [
+        ("v", "s", "sum", None, False), ("v", "c", "count", None, False),
+        ("w", "mn", "min", None, False), ("w", "mx", "max", None, False),
+        ("v", "a", "avg", None, False)])
+    key = ["k", "k2"]
+    out = out.sort_values(key, na_position="last").reset_index(drop=True)
+    exp = exp.sort_values(key, na_position="last").reset_index(drop=True)
+    assert len(out) == len(exp)
+    for col in ("c", "mn", "mx"):
+        g = out[col].to_numpy(dtype=np.float64)
+        e = exp[col].to_numpy(dtype=np.float64)
+        assert ((g == e) | (np.isnan(g) & np.isnan(e))).all(), col
+    for col in ("s", "a"):
+        g = out[col].to_numpy(dtype=np.float64)
+        e = exp[col].to_numpy(dtype=np.float64)
+        ok = np.isclose(g, e, rtol=1e-6) | (np.isnan(g) & np.isnan(e))
+        assert ok.all(), col
+
+
+def test_distinct_multicol(ctx):
+    df = pd.DataFrame({"a": [1, 1, 2, 2, 1], "b": [5, 5, 6, 6, 7]})
+    ctx.create_table("tdm", df)
+    out = ctx.sql("SELECT DISTINCT a, b FROM tdm").compute()
+    exp = df.drop_duplicates()
+    assert len(out) == len(exp)
+    got = set(map(tuple, out.astype(np.int64).to_numpy().tolist()))
+    assert got == set(map(tuple, exp.to_numpy().tolist()))
+
+
+def test_projection_arithmetic(ctx):
+    rng = np.random.default_rng(4)
+    df = pd.DataFrame({"p": rng.random(1000) * 100,
+                       "d": np.round(rng.random(1000) * 0.1, 2),
+                       "t": np.round(rng.random(1000) * 0.08, 2)})
+    ctx.create_table("tpa", df)
+    out = ctx.sql("SELECT p * (1 - d) AS disc, p * (1 - d) * (1 + t) AS chg "
+                  "FROM tpa").compute()
+    assert np.allclose(out["disc"], df["p"] * (1 - df["d"]), rtol=1e-12)
+    assert np.allclose(out["chg"],
+                       df["p"] * (1 - df["d"]) * (1 + df["t"]), rtol=1e-12)
+
+
+def test_case_with_null(ctx):
+    df = pd.DataFrame({"x": pd.array([1, None, 3], dtype="Int64")})
+    ctx.create_table("tcn", df)
+    out = ctx.sql("SELECT CASE WHEN x > 1 THEN 100 ELSE 0 END AS y "
+                  "FROM tcn").compute()
+    # NULL condition → ELSE branch (CASE WHEN semantics, call.py:217-253)
+    assert out["y"].astype(np.int64).tolist() == [0, 0, 100]
+
+
+QUERY_TEMPLATES = [
+    "SELECT k, SUM(v) AS s, COUNT(*) AS c FROM {t} GROUP BY k",
+    "SELECT k, SUM(v) AS s FROM {t} WHERE w > 3 GROUP BY k",
+    "SELECT w, MIN(v) AS mn, MAX(v) AS mx FROM {t} WHERE k IS NOT NULL "
+    "GROUP BY w",
+    "SELECT SUM(w) AS s, AVG(v) AS a FROM {t} WHERE v < 50",
+    "SELECT k, COUNT(*) AS c FROM {t} WHERE v IS NULL GROUP BY k",
+]
+
+
+@pytest.mark.parametrize("seed", [11, 12, 13])
+@pytest.mark.parametrize("qi", range(len(QUERY_TEMPLATES)))
+def test_differential_random(ctx, seed, qi):
+    """Randomized differential: product path vs oracle on NULL-bearing
+    frames."""
+    rng = np.random.default_rng(seed)
+    df = _rand_frame(rng, 4000, with_nulls=True)
+    name = f"dr_{seed}"
+    ctx.create_table(name, df)
+    q = QUERY_TEMPLATES[qi].format(t=name)
+    out = ctx.sql(q).compute()
+
+    # oracle evaluation of the same query (hand-mapped per template)
+    w = df.copy()
+    if qi == 0:
+        # w is never NULL in these frames → COUNT(w) == COUNT(*)
+        exp = oracle_groupby(w, ["k"], [("v", "s", "sum", None, False),
+                                        ("w", "c", "count", None, False)])
+    elif qi == 1:
+        w2 = oracle_filter(w, w["w"] > 3)
+        exp = oracle_groupby(w2, ["k"], [("v", "s", "sum", None, False)])
+    elif qi == 2:
+        w2 = oracle_filter(w, w["k"].notna())
+        exp = oracle_groupby(w2, ["w"], [("v", "mn", "min", None, False),
+                                         ("v", "mx", "max", None, False)])
+    elif qi == 3:
+        w2 = oracle_filter(w, w["v"] < 50)
+        exp = oracle_groupby(w2, [], [("w", "s", "sum", None, False),
+                                      ("v", "a", "avg", None, False)])
+    else:
+        w2 = oracle_filter(w, w["v"].isna())
+        exp = oracle_groupby(w2, ["k"], [("w", "c", "count", None, False)])
+
+    keys = [c for c in ("k", "w") if c in out.columns]
+    if keys:
+        out = out.sort_values(keys, na_position="last").reset_index(drop=True)
+        exp = exp.sort_values(keys, na_position="last").reset_index(drop=True)
+    assert len(out) == len(exp), (q, len(out), len(exp))
+    for col in out.columns:
+        g = out[col].to_numpy(dtype=np.float64)
+        e = exp[col].to_numpy(dtype=np.float64)
+        ok = np.isclose(g, e, rtol=1e-6, equal_nan=True)
+        assert ok.all(), (q, col, g[~ok][:4], e[~ok][:4])
