@@ -198,7 +198,7 @@ class DaskJoinPlugin(BaseRelPlugin):
 
         # materialize only columns the plan consumes (output_indices from the
         # pruning pass) plus residual-referenced temporaries
-        force_l = join_type in ("outer",)
+        force_l = join_type in ("outer", "right")
         force_r = join_type in ("left", "outer")
         combined = [("l", f) for f in cc_lhs.columns]
         keep_rhs = join_type not in ("leftanti",)
