@@ -990,12 +990,16 @@ extern "C" void dsx_hash_table_free(DsxHashTable* t) {
 // PASS=0: per-block match counts (LDS reduce — a single global counter
 // would serialize at ~88 wave-atomics/µs, measured 19-67 ms at C3 scale).
 // PASS=1: emit at block bases from the scanned counts + LDS bump.
+// PASS 0 probes the table once per row and CACHES (first match slot, match
+// count) so PASS 1 only re-walks the chain for multi-match rows — for
+// PK-FK joins (the common case) the emit pass does zero table probes.
 template <int PASS>
 __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
                              int64_t n, const uint64_t* tkeys,
                              const uint32_t* tvals, uint32_t* matched,
                              int64_t mask, int join_type, int packed,
                              int mark_matched, int64_t* block_counts,
+                             uint32_t* cache_slot, uint32_t* cache_cnt,
                              uint32_t* out_p, uint32_t* out_b, int64_t total,
                              unsigned int* dbg) {
   __shared__ unsigned long long s_cnt;   // PASS0: block total; PASS1: bump
@@ -1007,65 +1011,83 @@ __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
   unsigned long long local = 0;
   for (int64_t r = lo + threadIdx.x; r < hi; r += BLOCK) {
     bool key_valid = !(validity && !validity[r]);
-    int nmatch = 0;
-    uint32_t first_b = DSX_NULL_IDX;
-    if (key_valid) {
-      uint64_t cde = codes[r];
-      int64_t s = (int64_t)(mix64(cde) & mask);
-      while (true) {
-        uint64_t k = tkeys[s];
-        if (k == EMPTY_KEY) break;
-        uint64_t kc = packed ? (k >> 32) : k;
-        if (kc == cde) {
-          uint32_t bid = packed ? (uint32_t)k : tvals[s];
-          if (join_type == DSX_JOIN_INNER || join_type == DSX_JOIN_LEFT) {
-            if (PASS == 1) {
-              unsigned long long o = atomicAdd(&s_cnt, 1ull);
-              if (o >= (unsigned long long)total) {
-                atomicOr(dbg, 2u);
-              } else {
-                out_p[o] = (uint32_t)r;
-                out_b[o] = bid;
-                if (mark_matched) matched[s] = 1;  // FULL OUTER sweep only
-              }
+    if (PASS == 0) {
+      uint32_t nmatch = 0;
+      uint32_t first_s = DSX_NULL_IDX;
+      if (key_valid) {
+        uint64_t cde = codes[r];
+        int64_t s = (int64_t)(mix64(cde) & mask);
+        while (true) {
+          uint64_t k = tkeys[s];
+          if (k == EMPTY_KEY) break;
+          uint64_t kc = packed ? (k >> 32) : k;
+          if (kc == cde) {
+            if (first_s == DSX_NULL_IDX) first_s = (uint32_t)s;
+            nmatch++;
+            if (join_type == DSX_JOIN_LEFTSEMI ||
+                join_type == DSX_JOIN_LEFTANTI)
+              break;  // existence only
+          }
+          s = (s + 1) & mask;
+        }
+      }
+      cache_slot[r] = first_s;
+      cache_cnt[r] = nmatch;
+      if (join_type == DSX_JOIN_INNER || join_type == DSX_JOIN_LEFT)
+        local += (unsigned long long)nmatch;
+      if (nmatch == 0 && (join_type == DSX_JOIN_LEFT ||
+                          join_type == DSX_JOIN_LEFTANTI))
+        local += 1;  // NULL-fill / anti row
+      if (nmatch > 0 && join_type == DSX_JOIN_LEFTSEMI) local += 1;
+    } else {  // PASS 1: emit from the cache
+      uint32_t nmatch = cache_cnt[r];
+      uint32_t first_s = cache_slot[r];
+      if (nmatch > 0 &&
+          (join_type == DSX_JOIN_INNER || join_type == DSX_JOIN_LEFT)) {
+        unsigned long long o =
+            atomicAdd(&s_cnt, (unsigned long long)nmatch);
+        if (o + nmatch > (unsigned long long)total) {
+          atomicOr(dbg, 2u);
+        } else if (nmatch == 1) {
+          uint64_t k = tkeys[first_s];
+          out_p[o] = (uint32_t)r;
+          out_b[o] = packed ? (uint32_t)k : tvals[first_s];
+          if (mark_matched) matched[first_s] = 1;  // FULL OUTER sweep only
+        } else {
+          // multi-match: walk the chain from the first cached slot
+          uint64_t cde = codes[r];
+          int64_t s = (int64_t)first_s;
+          uint32_t emitted = 0;
+          while (emitted < nmatch) {
+            uint64_t k = tkeys[s];
+            uint64_t kc = packed ? (k >> 32) : k;
+            if (k == EMPTY_KEY) break;  // cannot happen if cache consistent
+            if (kc == cde) {
+              out_p[o + emitted] = (uint32_t)r;
+              out_b[o + emitted] = packed ? (uint32_t)k : tvals[s];
+              if (mark_matched) matched[s] = 1;
+              emitted++;
             }
-            nmatch++;
-          } else {  // SEMI / ANTI need existence only
-            nmatch++;
-            first_b = bid;
-            break;
+            s = (s + 1) & mask;
           }
         }
-        s = (s + 1) & mask;
-      }
-    }
-    // NULL probe key: for LEFT/ANTI behaves as no-match row (the reference
-    // drops NULL keys only on sides noted in join.py:202-213; LEFT keeps
-    // lhs rows with NULL key and fills rhs with NULL)
-    bool emit_nomatch =
-        (nmatch == 0) &&
-        (join_type == DSX_JOIN_LEFT || join_type == DSX_JOIN_LEFTANTI);
-    bool emit_semi = (nmatch > 0) && (join_type == DSX_JOIN_LEFTSEMI);
-    if (PASS == 1) {
-      if (emit_nomatch) {
+      } else if (nmatch == 0 && (join_type == DSX_JOIN_LEFT ||
+                                 join_type == DSX_JOIN_LEFTANTI)) {
         unsigned long long o = atomicAdd(&s_cnt, 1ull);
         if (o >= (unsigned long long)total) atomicOr(dbg, 2u);
         else {
           out_p[o] = (uint32_t)r;
           out_b[o] = DSX_NULL_IDX;
         }
-      } else if (emit_semi) {
+      } else if (nmatch > 0 && join_type == DSX_JOIN_LEFTSEMI) {
         unsigned long long o = atomicAdd(&s_cnt, 1ull);
         if (o >= (unsigned long long)total) atomicOr(dbg, 2u);
         else {
+          uint64_t k = tkeys[first_s];
           out_p[o] = (uint32_t)r;
-          out_b[o] = first_b;
+          out_b[o] = packed ? (uint32_t)k : tvals[first_s];
         }
       }
-    } else {
-      if (join_type == DSX_JOIN_INNER || join_type == DSX_JOIN_LEFT)
-        local += (unsigned long long)nmatch;
-      if (emit_nomatch || emit_semi) local += 1;
     }
   }
   if (PASS == 0) {
@@ -1087,16 +1109,19 @@ extern "C" int dsx_hash_probe(DsxCtx* c, DsxHashTable* t, const uint64_t* codes,
   *out_build_idx = nullptr;
   *out_count = 0;
   int grid = (int)min((int64_t)MAX_GRID, (n + BLOCK - 1) / BLOCK);
-  int rc = ensure_scratch(c, (grid + 2) * 8);
+  int rc = ensure_scratch(c, (grid + 2) * 8 + n * 8 + 64);
   if (rc) return rc;
   int64_t* block_counts = (int64_t*)c->scratch;
   int64_t* d_total = block_counts + grid;
+  uint32_t* cache_slot = (uint32_t*)(d_total + 2);
+  uint32_t* cache_cnt = cache_slot + n;
   if (grid > 0) {
     ProfScope ps(c, "k_hash_probe_count");
     hipLaunchKernelGGL(k_hash_probe<0>, dim3(grid), dim3(BLOCK), 0, c->stream,
                        codes, validity, n, t->keys, t->vals, t->matched,
                        t->slots - 1, join_type, t->packed, mark_matched,
-                       block_counts, nullptr, nullptr, 0, c->dbg_flag);
+                       block_counts, cache_slot, cache_cnt, nullptr, nullptr,
+                       0, c->dbg_flag);
   }
   hipLaunchKernelGGL(k_scan_block_counts, dim3(1), dim3(64), 0, c->stream,
                      block_counts, grid, d_total);
@@ -1111,7 +1136,7 @@ extern "C" int dsx_hash_probe(DsxCtx* c, DsxHashTable* t, const uint64_t* codes,
     hipLaunchKernelGGL(k_hash_probe<1>, dim3(grid), dim3(BLOCK), 0, c->stream,
                        codes, validity, n, t->keys, t->vals, t->matched,
                        t->slots - 1, join_type, t->packed, mark_matched,
-                       block_counts, *out_probe_idx,
+                       block_counts, cache_slot, cache_cnt, *out_probe_idx,
                        *out_build_idx, total, c->dbg_flag);
   }
   HIP_TRY(hipGetLastError());
