@@ -1014,6 +1014,7 @@ __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
     if (PASS == 0) {
       uint32_t nmatch = 0;
       uint32_t first_s = DSX_NULL_IDX;
+      uint32_t first_b = DSX_NULL_IDX;
       if (key_valid) {
         uint64_t cde = codes[r];
         int64_t s = (int64_t)(mix64(cde) & mask);
@@ -1022,7 +1023,10 @@ __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
           if (k == EMPTY_KEY) break;
           uint64_t kc = packed ? (k >> 32) : k;
           if (kc == cde) {
-            if (first_s == DSX_NULL_IDX) first_s = (uint32_t)s;
+            if (first_s == DSX_NULL_IDX) {
+              first_s = (uint32_t)s;
+              first_b = packed ? (uint32_t)k : tvals[s];
+            }
             nmatch++;
             if (join_type == DSX_JOIN_LEFTSEMI ||
                 join_type == DSX_JOIN_LEFTANTI)
@@ -1031,7 +1035,9 @@ __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
           s = (s + 1) & mask;
         }
       }
-      cache_slot[r] = first_s;
+      // single-match rows with no FULL-OUTER marking need only the build id
+      // in the emit pass — cache it instead of the slot (zero table reads)
+      cache_slot[r] = (nmatch == 1 && !mark_matched) ? first_b : first_s;
       cache_cnt[r] = nmatch;
       if (join_type == DSX_JOIN_INNER || join_type == DSX_JOIN_LEFT)
         local += (unsigned long long)nmatch;
@@ -1049,10 +1055,14 @@ __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
         if (o + nmatch > (unsigned long long)total) {
           atomicOr(dbg, 2u);
         } else if (nmatch == 1) {
-          uint64_t k = tkeys[first_s];
           out_p[o] = (uint32_t)r;
-          out_b[o] = packed ? (uint32_t)k : tvals[first_s];
-          if (mark_matched) matched[first_s] = 1;  // FULL OUTER sweep only
+          if (mark_matched) {  // cache holds the SLOT in this mode
+            uint64_t k = tkeys[first_s];
+            out_b[o] = packed ? (uint32_t)k : tvals[first_s];
+            matched[first_s] = 1;  // FULL OUTER sweep only
+          } else {             // cache holds the BUILD ID directly
+            out_b[o] = first_s;
+          }
         } else {
           // multi-match: walk the chain from the first cached slot
           uint64_t cde = codes[r];
@@ -1083,9 +1093,8 @@ __global__ void k_hash_probe(const uint64_t* codes, const uint8_t* validity,
         unsigned long long o = atomicAdd(&s_cnt, 1ull);
         if (o >= (unsigned long long)total) atomicOr(dbg, 2u);
         else {
-          uint64_t k = tkeys[first_s];
           out_p[o] = (uint32_t)r;
-          out_b[o] = packed ? (uint32_t)k : tvals[first_s];
+          out_b[o] = first_s;  // SEMI: nmatch==1 ⇒ cache holds the build id
         }
       }
     }
