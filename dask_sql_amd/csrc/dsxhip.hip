@@ -1897,12 +1897,18 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
                        d_totals, nb, d_bases);
     {
       ProfScope ps(c, "k_gbpart_scatter");
+      static const int sthreads = [] {
+        const char* e = getenv("DSX_SCATTER_THREADS");
+        return e ? atoi(e) : BLOCK;
+      }();
+      // NB: grid MUST match the hist pass (per-block bucket bases are keyed
+      // by blockIdx); only the thread count may vary.
       struct {
         ColsArg C; int64_t n; int nb; const int64_t* hist;
         const int64_t* bases; uint64_t* out;
       } a2{C, n, nb, d_hist, d_bases, d_recs};
       void* args[] = {&a2.C, &a2.n, &a2.nb, &a2.hist, &a2.bases, &a2.out};
-      hipModuleLaunchKernel(f_scat, grid, 1, 1, BLOCK, 1, 1,
+      hipModuleLaunchKernel(f_scat, grid, 1, 1, sthreads, 1, 1,
                             (unsigned)(nb * 8), c->stream, args, nullptr);
     }
   } else {
