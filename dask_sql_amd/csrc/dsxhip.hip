@@ -1942,7 +1942,7 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
                       &a3.tg, &a3.ovf};
       static const int athreads = [] {
         const char* e = getenv("DSX_AGG_THREADS");
-        return e ? atoi(e) : 256;
+        return e ? atoi(e) : 1024;  // measured best at C2 (0.84→0.46 ms)
       }();
       hipModuleLaunchKernel(f_aggr, nb, 1, 1, athreads, 1, 1,
                             (unsigned)lds_bytes, c->stream, args, nullptr);
