@@ -1801,8 +1801,12 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   // (less 16-B-store line-merge loss in L2) but shrink the aggregate grid
   // and grow its LDS table; 1024 targets the best total at C2 shape
   // (512: scatter −0.17 ms but aggregate +0.81 ms)
+  static const int nb_target = [] {
+    const char* e = getenv("DSX_GB_NB_TARGET");
+    return e ? atoi(e) : 1024;  // groups per bucket (tradeoff note above)
+  }();
   int nb = 64;
-  while (nb < 4096 && g_est / nb > 1024) nb <<= 1;
+  while (nb < 4096 && g_est / nb > nb_target) nb <<= 1;
   int64_t per_bucket = (g_est + nb - 1) / nb;
   int lds_slots = 256;
   while (lds_slots < 2 * per_bucket) lds_slots <<= 1;
