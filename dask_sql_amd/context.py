@@ -10,8 +10,7 @@ import numpy as np
 import pandas as pd
 
 from dask_sql_amd import runtime as rt
-from dask_sql_amd.datacontainer import (ColumnContainer, DataContainer,
-                                        DeviceTable, HostDataContainer)
+from dask_sql_amd.datacontainer import DeviceTable, HostDataContainer
 from dask_sql_amd.planner.builder import Builder, Catalog
 from dask_sql_amd.physical.convert import RelConverter
 from dask_sql_amd.physical.rel_plugins import register_defaults

@@ -17,7 +17,7 @@ from dask_sql_amd.physical.convert import BaseRelPlugin, RelConverter
 from dask_sql_amd.physical.rex import (KB, KF, KI, OP_AND, OP_COL,
                                        OP_IS_NOT_NULL, RexCompileError,
                                        compile_expr, scalar_literal)
-from dask_sql_amd.planner.plan import Call, InputRef, Literal
+from dask_sql_amd.planner.plan import Call, InputRef
 
 logger = logging.getLogger(__name__)
 

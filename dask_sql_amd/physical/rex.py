@@ -7,7 +7,7 @@ program (include/dsxhip.h DsxOp) interpreted per row inside the HIP kernels.
 """
 from __future__ import annotations
 
-from dask_sql_amd.planner.plan import AggCall, Call, Expression, InputRef, Literal
+from dask_sql_amd.planner.plan import Call, Expression, InputRef, Literal
 from dask_sql_amd import runtime as rt
 
 # opcodes (include/dsxhip.h)
@@ -107,7 +107,7 @@ class RexCompiler:
             # pre-scan kinds to decide int vs float path
             ka = self._peek_kind(a)
             kb = self._peek_kind(b)
-            if KF in (ka, kb) or op == "/" and False:
+            if KF in (ka, kb):
                 k = self.compile(a)
                 self._to_f(k)
                 k = self.compile(b)

@@ -9,8 +9,7 @@ from __future__ import annotations
 
 from dask_sql_amd.planner.plan import (AggCall, AggregateNode, Call,
                                        Expression, FilterNode, InputRef,
-                                       JoinNode, LimitNode, Literal,
-                                       LogicalPlan, ProjectionNode,
+                                       JoinNode, LogicalPlan, ProjectionNode,
                                        RelDataType, SortNode, TableScanNode)
 
 

@@ -7,7 +7,6 @@ path.
 from __future__ import annotations
 
 import ctypes as ct
-import os
 from pathlib import Path
 
 import numpy as np

@@ -209,3 +209,41 @@ def test_differential_random(ctx, seed, qi):
         e = exp[col].to_numpy(dtype=np.float64)
         ok = np.isclose(g, e, rtol=1e-6, equal_nan=True)
         assert ok.all(), (q, col, g[~ok][:4], e[~ok][:4])
+
+
+def test_empty_table(ctx):
+    ctx.create_table("tempty", pd.DataFrame({"k": np.array([], np.int64),
+                                             "v": np.array([], np.float64)}))
+    out = ctx.sql("SELECT k, SUM(v) AS s FROM tempty GROUP BY k").compute()
+    assert len(out) == 0
+    out = ctx.sql("SELECT * FROM tempty WHERE v > 1").compute()
+    assert len(out) == 0
+
+
+def test_filter_selects_nothing(ctx):
+    ctx.create_table("tnone", pd.DataFrame({"k": [1, 2, 3],
+                                            "v": [1.0, 2.0, 3.0]}))
+    out = ctx.sql("SELECT k, SUM(v) AS s FROM tnone WHERE v > 99 "
+                  "GROUP BY k").compute()
+    assert len(out) == 0
+
+
+def test_single_row(ctx):
+    ctx.create_table("tone", pd.DataFrame({"k": [7], "v": [3.5]}))
+    out = ctx.sql("SELECT k, SUM(v) AS s, COUNT(*) AS c FROM tone "
+                  "GROUP BY k").compute()
+    assert out["k"].astype(np.int64).tolist() == [7]
+    assert out["c"].astype(np.int64).tolist() == [1]
+    assert abs(out["s"].iloc[0] - 3.5) < 1e-12
+
+
+def test_join_empty_side(ctx):
+    ctx.create_table("je1", pd.DataFrame({"k": [1, 2], "a": [1.0, 2.0]}))
+    ctx.create_table("je2", pd.DataFrame({"k": np.array([], np.int64),
+                                          "b": np.array([], np.float64)}))
+    out = ctx.sql("SELECT l.k, l.a, r.b FROM je1 l JOIN je2 r "
+                  "ON l.k = r.k").compute()
+    assert len(out) == 0
+    out = ctx.sql("SELECT l.k, l.a, r.b FROM je1 l LEFT JOIN je2 r "
+                  "ON l.k = r.k").compute()
+    assert len(out) == 2 and out["b"].isna().all()
