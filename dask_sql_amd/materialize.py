@@ -48,11 +48,10 @@ def _convert(arr, valid, col, sql_t):
         out[in_range] = lut[codes[in_range]]
         return pd.Series(out)
     if sql_t == "DATE":
-        vals = arr.astype("int64")
-        s = pd.Series(vals.view("datetime64[D]" if arr.dtype == np.int64
-                                else "datetime64[D]")
-                      if False else
-                      pd.to_datetime(vals, unit="D", errors="coerce"))
+        # date32 day-ints → datetime64[ns] (mappings.py:78-80: DATE columns
+        # are datetime64[ns] on the reference path)
+        s = pd.Series(pd.to_datetime(arr.astype("int64"), unit="D",
+                                     errors="coerce"))
         if valid is not None:
             s[~valid] = pd.NaT
         return s

@@ -234,11 +234,37 @@ class Runtime:
         }
 
     # ---- helpers ---------------------------------------------------------
+    # stack effect per opcode family (VM slots are 6 named registers —
+    # dsxhip.hip VmStack; exceeding them would silently alias slot 5)
+    _PUSH_OPS = {1, 2, 3, 4}          # COL, LIT_*
+    _BIN_OPS = set(range(10, 42))     # arith/cmp/AND/OR
+    _SELECT_OP = 60
+    _MAX_DEPTH = 6
+
+    @classmethod
+    def _check_depth(cls, instrs):
+        depth = 0
+        peak = 0
+        for op, _, _ in instrs:
+            if op in cls._PUSH_OPS:
+                depth += 1
+            elif op in cls._BIN_OPS and op != 42:  # NOT(42) is unary
+                depth -= 1
+            elif op == cls._SELECT_OP:
+                depth -= 2
+            peak = max(peak, depth)
+        if peak > cls._MAX_DEPTH:
+            raise DsxError(
+                f"expression too deep for the VM register stack "
+                f"({peak} > {cls._MAX_DEPTH}); split the expression "
+                "(e.g. nested CASE) into projection steps")
+
     @staticmethod
     def make_prog(instrs) -> tuple:
         """instrs: list of (op, arg0, imm) where imm may be float/int."""
         if len(instrs) > MAX_PROG:
             raise DsxError(f"program too long ({len(instrs)})")
+        Runtime._check_depth(instrs)
         arr = (_Instr * max(len(instrs), 1))()
         for i, (op, arg0, imm) in enumerate(instrs):
             arr[i].op = op
