@@ -948,8 +948,12 @@ extern "C" int dsx_hash_build(DsxCtx* c, const uint64_t* codes,
                               const uint8_t* validity, int64_t n,
                               uint64_t code_max, DsxHashTable** out) {
   if (n > 0xFFFFFFFEll) FAIL(-3, "build side too large for u32 row ids");
+  static const double slots_mult = [] {
+    const char* e = getenv("DSX_JOIN_SLOTS_MULT");
+    return e ? atof(e) : 2.0;
+  }();
   int64_t slots = 64;
-  while (slots < 2 * n) slots <<= 1;
+  while (slots < (int64_t)(slots_mult * n)) slots <<= 1;
   DsxHashTable* t = new DsxHashTable();
   t->slots = slots;
   t->n_build = n;
