@@ -62,8 +62,7 @@ def exchange_buckets(tensors: list[torch.Tensor], in_splits: list[int],
     world = dist.get_world_size(group)
     dev = tensors[0].device
     in_t = torch.tensor(in_splits, dtype=torch.int64, device=dev)
-    counts = torch.zeros(world, dtype=torch.int64, device=dev)
-    # counts[r] after: rows rank r sends to me
+    # all_counts[r][b] = rows rank r sends to rank b
     all_counts = [torch.zeros_like(in_t) for _ in range(world)]
     dist.all_gather(all_counts, in_t, group=group)
     me = dist.get_rank(group)

@@ -300,3 +300,32 @@ def test_multimatch_join(ctx):
     cnt = collections.Counter(pk.tolist())
     for k, c in list(cnt.items())[:50]:
         assert s[k] == 4 * c
+
+
+def test_count_col_vs_count_star(ctx):
+    """COUNT(v) skips NULLs, COUNT(*) does not (pandas 'count' vs size —
+    aggregate.py AGGREGATION_MAPPING)."""
+    df = pd.DataFrame({"k": [1, 1, 2, 2],
+                       "v": pd.array([1.0, None, None, None],
+                                     dtype="Float64")})
+    ctx.create_table("tcc", df)
+    out = ctx.sql("SELECT k, COUNT(v) AS cv, COUNT(*) AS cs FROM tcc "
+                  "GROUP BY k").compute()
+    out = out.sort_values("k").reset_index(drop=True)
+    assert out["cv"].astype(np.int64).tolist() == [1, 0]
+    assert out["cs"].astype(np.int64).tolist() == [2, 2]
+
+
+def test_orderby_nulls_last(ctx):
+    df = pd.DataFrame({"k": pd.array([3, None, 1], dtype="Int64"),
+                       "v": [1.0, 2.0, 3.0]})
+    ctx.create_table("tnl", df)
+    out = ctx.sql("SELECT k, v FROM tnl ORDER BY k LIMIT 3").compute()
+    ks = out["k"].tolist()
+    assert ks[0] == 1 and ks[1] == 3 and pd.isna(ks[2])  # ASC → NULLS LAST
+
+
+def test_limit_offset(ctx):
+    ctx.create_table("tlo", pd.DataFrame({"x": np.arange(100, dtype=np.int64)}))
+    out = ctx.sql("SELECT x FROM tlo ORDER BY x LIMIT 10 OFFSET 5").compute()
+    assert out["x"].astype(np.int64).tolist() == list(range(5, 15))
