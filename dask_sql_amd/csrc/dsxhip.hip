@@ -1820,7 +1820,11 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
     *fell_back = true;
     return 0;
   }
-  int grid = (int)min((int64_t)MAX_GRID, (n + BLOCK - 1) / BLOCK);
+  static const int max_grid = [] {
+    const char* e = getenv("DSX_GB_GRID");
+    return e ? atoi(e) : MAX_GRID;
+  }();
+  int grid = (int)min((int64_t)max_grid, (n + BLOCK - 1) / BLOCK);
   if (grid == 0) {  // empty input → empty output
     int rc2 = pool_alloc(c, 8, (void**)out_codes);
     if (!rc2) rc2 = pool_alloc(c, 8, out_vals);
