@@ -124,12 +124,16 @@ def make_context(workload, rank, world, device_id):
         c.create_table("t", li, persist=True)
         total_rows = n
     elif workload == "q3_sf10":
-        assert world == 1, "q3 multi-GPU lands in round 2 (SURVEY §8e)"
-        cust, orders, li = gen_q3(seed=seed)
+        # strong scaling: each rank holds a row slice of the SF10 tables
+        cust, orders, li = gen_q3(seed=SEED)
+        if world > 1:
+            cust = cust.iloc[rank::world].reset_index(drop=True)
+            orders = orders.iloc[rank::world].reset_index(drop=True)
+            li = li.iloc[rank::world].reset_index(drop=True)
         c.create_table("customer", cust, persist=True)
         c.create_table("orders", orders, persist=True)
         c.create_table("lineitem", li, persist=True)
-        total_rows = w["rows"]
+        total_rows = w["rows"] // world
     else:
         raise KeyError(workload)
     return c, total_rows
@@ -138,6 +142,12 @@ def make_context(workload, rank, world, device_id):
 def run_step(c, workload, world, pg):
     """One pass of the hot path. Returns the result holder (device)."""
     w = WORKLOADS[workload]
+    if workload == "q3_sf10" and world > 1:
+        # distributed Q3: mid-pipeline RCCL repartition (SURVEY §8e)
+        from dask_sql_amd.distributed import q3_distributed
+        out = q3_distributed(c, pg)
+        c._get_runtime().synchronize()
+        return out
     res = c.sql(w["sql"])
     if world > 1 and workload.startswith(("c2", "q1")):
         # distributed partial-merge over RCCL (SURVEY §8e): exchange partial

@@ -131,6 +131,26 @@ class RegisteredTable:
         return self.device_table
 
 
+_DSX_SQLT = {rt.I64: "BIGINT", rt.F64: "DOUBLE", rt.I32: "INTEGER",
+             rt.F32: "FLOAT", rt.I8: "TINYINT", rt.BOOL8: "BOOLEAN"}
+
+
+class DeviceRegisteredTable:
+    """A table registered directly from device-resident columns (e.g. a
+    shuffle-received intermediate in the distributed pipeline)."""
+
+    def __init__(self, table: DeviceTable, sql_types: dict | None = None):
+        self.device_table = table
+        self._sql_types = sql_types or {}
+
+    def fields(self):
+        return [(n, self._sql_types.get(n, _DSX_SQLT[c.dtype]))
+                for n, c in self.device_table.columns.items()]
+
+    def upload(self, runtime) -> DeviceTable:
+        return self.device_table
+
+
 class ResultFrame:
     """Shaped like the reference's lazy return of Context.sql (a dataframe
     you .compute()); here execution already happened on the GPU and compute()
@@ -190,6 +210,15 @@ class Context:
         self._schema_version += 1
         if persist:
             t.upload(self._get_runtime())
+
+    def create_table_from_device(self, table_name: str, table: DeviceTable,
+                                 sql_types: dict | None = None):
+        """Register device-resident columns as a table (distributed
+        intermediates; no host round-trip)."""
+        t = DeviceRegisteredTable(table, sql_types)
+        self.tables[table_name.lower()] = t
+        self.catalog.add(table_name, t.fields())
+        self._schema_version += 1
 
     def drop_table(self, table_name: str):
         self.tables.pop(table_name.lower(), None)

@@ -167,3 +167,84 @@ def merge_groupby_partials(runtime, key_col, val_cols, val_ops, group=None):
         out_vals.append(rt.DeviceColumn(runtime, ov + i * G * 8, None, G,
                                         dtype, owner=False, keep_alive=h))
     return key_out, out_vals
+
+
+def shuffle_datacontainer(runtime, dc, key_frontend: str, group=None):
+    """Repartition a DataContainer's rows across the world by the hash of an
+    integer key column; returns a DeviceTable of the received rows (same
+    frontend column names). The mid-pipeline exchange of the distributed
+    join (dask's dd.merge shuffle step, join.py:241-246, over RCCL/xGMI)."""
+    from dask_sql_amd.datacontainer import DeviceTable
+
+    cc = dc.column_container
+    names = list(cc.columns)
+    cols = [dc.table.col(cc.get_backend_by_frontend_name(n)) for n in names]
+    ki = names.index(key_frontend)
+    key_col = cols[ki]
+    payloads = [c for i, c in enumerate(cols) if i != ki]
+    rkey, rvals, _ = shuffle_device_columns(runtime, key_col, payloads, group)
+    out = {}
+    vi = 0
+    for i, n in enumerate(names):
+        if i == ki:
+            out[n] = rkey
+        else:
+            out[n] = rvals[vi]  # dtype preserved through the torch staging
+            vi += 1
+    return DeviceTable(out, num_rows=rkey.len)
+
+
+def q3_distributed(ctx, group=None):
+    """TPC-H Q3 over row-sliced tables on N GPUs (BASELINE configs[4]):
+    filter locally → repartition customer/orders by custkey (RCCL
+    all-to-all) → local join → repartition CO and filtered lineitem by
+    orderkey → local join+groupby (groups are orderkey-disjoint across
+    ranks) → gather per-rank top-10s → exact global top-10.
+    world=1 degenerates to a pass-through exchange (single-GPU testable)."""
+    import pandas as pd
+
+    world = dist.get_world_size(group) if dist.is_initialized() else 1
+    runtime = ctx._get_runtime()
+
+    c_f = ctx.sql("SELECT c_custkey FROM customer "
+                  "WHERE c_mktsegment = 0").dc
+    o_f = ctx.sql("SELECT o_orderkey, o_custkey, o_orderdate, o_shippriority "
+                  "FROM orders WHERE o_orderdate < 9204").dc
+    l_f = ctx.sql("SELECT l_orderkey, l_extendedprice, l_discount "
+                  "FROM lineitem WHERE l_shipdate > 9204").dc
+    if world > 1:
+        c_x = shuffle_datacontainer(runtime, c_f, "c_custkey", group)
+        o_x = shuffle_datacontainer(runtime, o_f, "o_custkey", group)
+    else:
+        c_x = c_f.assign()
+        o_x = o_f.assign()
+    ctx.create_table_from_device("customer_x", c_x)
+    ctx.create_table_from_device("orders_x", o_x)
+    co = ctx.sql("SELECT o_orderkey, o_orderdate, o_shippriority "
+                 "FROM customer_x, orders_x "
+                 "WHERE c_custkey = o_custkey").dc
+    if world > 1:
+        co_x = shuffle_datacontainer(runtime, co, "o_orderkey", group)
+        l_x = shuffle_datacontainer(runtime, l_f, "l_orderkey", group)
+    else:
+        co_x = co.assign()
+        l_x = l_f.assign()
+    ctx.create_table_from_device("co_x", co_x)
+    ctx.create_table_from_device("lineitem_x", l_x)
+    top = ctx.sql(
+        "SELECT l_orderkey, SUM(l_extendedprice*(1-l_discount)) AS revenue, "
+        "o_orderdate, o_shippriority FROM lineitem_x, co_x "
+        "WHERE l_orderkey = o_orderkey "
+        "GROUP BY l_orderkey, o_orderdate, o_shippriority "
+        "ORDER BY revenue DESC, o_orderdate LIMIT 10").compute()
+    if world > 1:
+        tops = [None] * world
+        dist.all_gather_object(tops, top, group=group)
+        if (dist.get_rank(group) if group else dist.get_rank()) == 0:
+            allt = pd.concat(tops)
+            allt = allt.sort_values(["revenue", "o_orderdate"],
+                                    ascending=[False, True],
+                                    kind="mergesort").head(10)
+            return allt.reset_index(drop=True)
+        return None
+    return top

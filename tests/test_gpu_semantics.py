@@ -329,3 +329,28 @@ def test_limit_offset(ctx):
     ctx.create_table("tlo", pd.DataFrame({"x": np.arange(100, dtype=np.int64)}))
     out = ctx.sql("SELECT x FROM tlo ORDER BY x LIMIT 10 OFFSET 5").compute()
     assert out["x"].astype(np.int64).tolist() == list(range(5, 15))
+
+
+def test_q3_distributed_world1_equals_plain(ctx):
+    """The distributed Q3 pipeline at world=1 (pass-through exchange) must
+    reproduce the plain single-context Q3 exactly."""
+    from datagen import gen_q3
+    from dask_sql_amd.distributed import q3_distributed
+    from dask_sql_amd.context import Context
+    cust, orders, li = gen_q3(sf_rows=(20_000, 100_000, 400_000))
+    c2 = Context()
+    c2.create_table("customer", cust)
+    c2.create_table("orders", orders)
+    c2.create_table("lineitem", li)
+    plain = c2.sql(
+        "SELECT l_orderkey, SUM(l_extendedprice*(1-l_discount)) AS revenue, "
+        "o_orderdate, o_shippriority FROM customer, orders, lineitem "
+        "WHERE c_mktsegment = 0 AND c_custkey = o_custkey "
+        "AND l_orderkey = o_orderkey AND o_orderdate < 9204 "
+        "AND l_shipdate > 9204 "
+        "GROUP BY l_orderkey, o_orderdate, o_shippriority "
+        "ORDER BY revenue DESC, o_orderdate LIMIT 10").compute()
+    distd = q3_distributed(c2)
+    assert (distd["l_orderkey"].to_numpy().astype(np.int64)
+            == plain["l_orderkey"].to_numpy().astype(np.int64)).all()
+    assert np.allclose(distd["revenue"], plain["revenue"], rtol=1e-9)
