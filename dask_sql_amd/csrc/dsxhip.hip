@@ -1504,16 +1504,38 @@ __global__ void k_gbpart_scan(int64_t* hist, int grid, int nb,
   if (threadIdx.x == BLOCK - 1) totals[b] = s_sum[BLOCK - 1];
 }
 
-// 1 block: exclusive scan of bucket totals → absolute bases
+// 1 block: exclusive scan of bucket totals → absolute bases. Parallel
+// Hillis-Steele (the serial loop measured 137 µs/step at nb=1024 — pure
+// dependent-load latency; this runs in ~5 µs).
 __global__ void k_gbpart_bases(int64_t* totals, int nb, int64_t* bases) {
-  if (threadIdx.x == 0) {
-    int64_t run = 0;
-    for (int b = 0; b < nb; b++) {
-      bases[b] = run;
-      run += totals[b];
-    }
-    bases[nb] = run;
+  // launched with 1024 threads; nb ≤ 4096 → ≤4 elements per thread, kept
+  // in named scalars (runtime-indexed locals would spill to scratch,
+  // cdna guide §5.4)
+  __shared__ int64_t s[4096];
+  int i0 = threadIdx.x, i1 = i0 + 1024, i2 = i0 + 2048, i3 = i0 + 3072;
+  if (i0 < nb) s[i0] = totals[i0];
+  if (i1 < nb) s[i1] = totals[i1];
+  if (i2 < nb) s[i2] = totals[i2];
+  if (i3 < nb) s[i3] = totals[i3];
+  __syncthreads();
+  for (int d = 1; d < nb; d <<= 1) {
+    int64_t v0 = (i0 < nb && i0 >= d) ? s[i0 - d] : 0;
+    int64_t v1 = (i1 < nb && i1 >= d) ? s[i1 - d] : 0;
+    int64_t v2 = (i2 < nb && i2 >= d) ? s[i2 - d] : 0;
+    int64_t v3 = (i3 < nb && i3 >= d) ? s[i3 - d] : 0;
+    __syncthreads();
+    if (i0 < nb) s[i0] += v0;
+    if (i1 < nb) s[i1] += v1;
+    if (i2 < nb) s[i2] += v2;
+    if (i3 < nb) s[i3] += v3;
+    __syncthreads();
   }
+  // s is the inclusive scan; subtract own total for exclusive bases
+  if (i0 < nb) bases[i0] = s[i0] - totals[i0];
+  if (i1 < nb) bases[i1] = s[i1] - totals[i1];
+  if (i2 < nb) bases[i2] = s[i2] - totals[i2];
+  if (i3 < nb) bases[i3] = s[i3] - totals[i3];
+  if (i0 == 0) bases[nb] = s[nb - 1];
 }
 
 // scatter records (key u64 + nvals f64/i64 values) bucket-major
@@ -1793,10 +1815,19 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
                              ProgArg& P, AggArg& A, const DsxAggSpec* aggs_arr,
                              std::vector<DsxInstr>& progs,
                              std::vector<int32_t>& lens, int naggs,
-                             int64_t g_est, uint64_t** out_codes,
+                             int64_t g_est, uint64_t key_space,
+                             uint64_t** out_codes,
                              void** out_vals, uint64_t** out_counts,
                              int64_t* out_groups, bool* fell_back) {
   *fell_back = false;
+  // SoA u32-code records (u32 codes + u64 vals) — MEASURED REGRESSION at C2
+  // (35.0 vs 45.8 G rows/s): splitting the record doubles the number of
+  // distinct cache lines each scattered append touches, and line touches,
+  // not bytes, dominate the scatter cost. Kept behind an opt-in knob as the
+  // recorded experiment; AoS 16-B records are the default.
+  static const bool code32_enabled = getenv("DSX_CODE32") != nullptr;
+  bool code32 = code32_enabled && key_space > 0 &&
+                key_space <= (1ull << 32);
   int32_t val_of[DSX_MAX_AGGS];
   int nvals = 0;
   for (int a = 0; a < naggs; a++)
@@ -1838,7 +1869,12 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   int rec = 1 + nvals;
   // records live in the persistent scratch arena: a per-call hipMalloc/Free
   // of ~GBs costs tens of ms (measured); the arena amortizes it
-  int64_t rec_bytes = (n > 0 ? n : 1) * (int64_t)rec * 8;
+  int64_t n1 = n > 0 ? n : 1;
+  int64_t rec_bytes = n1 * (int64_t)rec * 8;
+  {  // SoA (code32) view may need slightly more at tiny n (256-B alignment)
+    int64_t soa = ((n1 * 4 + 255) / 256) * 256 + n1 * (int64_t)nvals * 8;
+    if (soa > rec_bytes) rec_bytes = soa;
+  }
   int64_t need = prog_bytes + ((lens_bytes + 15) / 16) * 16 + 64 +
                  sizeof(KeyArg) + sizeof(AggArg) + 32 +
                  (int64_t)grid * nb * 8 + (nb + 2) * 8 + nb * 8 +
@@ -1890,10 +1926,20 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
 
   JitEntry* je = jit_source_entry(
       c, jit_gbpart_source(C, K, P, aggs_arr, A, val_of, naggs, nvals,
-                           lds_slots));
+                           lds_slots, code32));
   hipFunction_t f_hist = je ? jit_fn(c, je, "j_hist") : nullptr;
   hipFunction_t f_scat = je ? jit_fn(c, je, "j_scatter") : nullptr;
   hipFunction_t f_aggr = je ? jit_fn(c, je, "j_aggregate") : nullptr;
+  if (!(f_hist && f_scat && f_aggr)) {
+    // all-or-nothing: scatter and aggregate must agree on record layout
+    f_hist = f_scat = f_aggr = nullptr;
+    code32 = false;
+  }
+  // SoA views over the record arena (code32): u32 codes, then u64 vals
+  uint32_t* d_rcodes = (uint32_t*)d_recs;
+  uint64_t* d_rvals =
+      (uint64_t*)((char*)d_recs + (((int64_t)(n > 0 ? n : 1) * 4 + 255) / 256) *
+                                      256);
   if (f_hist && f_scat) {
     {
       ProfScope ps(c, "k_gbpart_hist");
@@ -1905,7 +1951,7 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
     }
     hipLaunchKernelGGL(k_gbpart_scan, dim3(nb), dim3(BLOCK), 0, c->stream,
                        d_hist, grid, nb, d_totals);
-    hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(64), 0, c->stream,
+    hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(1024), 0, c->stream,
                        d_totals, nb, d_bases);
     {
       ProfScope ps(c, "k_gbpart_scatter");
@@ -1915,13 +1961,24 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
       }();
       // NB: grid MUST match the hist pass (per-block bucket bases are keyed
       // by blockIdx); only the thread count may vary.
-      struct {
-        ColsArg C; int64_t n; int nb; const int64_t* hist;
-        const int64_t* bases; uint64_t* out;
-      } a2{C, n, nb, d_hist, d_bases, d_recs};
-      void* args[] = {&a2.C, &a2.n, &a2.nb, &a2.hist, &a2.bases, &a2.out};
-      hipModuleLaunchKernel(f_scat, grid, 1, 1, sthreads, 1, 1,
-                            (unsigned)(nb * 8), c->stream, args, nullptr);
+      if (code32) {
+        struct {
+          ColsArg C; int64_t n; int nb; const int64_t* hist;
+          const int64_t* bases; uint32_t* oc; uint64_t* ov;
+        } a2{C, n, nb, d_hist, d_bases, d_rcodes, d_rvals};
+        void* args[] = {&a2.C, &a2.n, &a2.nb, &a2.hist, &a2.bases, &a2.oc,
+                        &a2.ov};
+        hipModuleLaunchKernel(f_scat, grid, 1, 1, sthreads, 1, 1,
+                              (unsigned)(nb * 8), c->stream, args, nullptr);
+      } else {
+        struct {
+          ColsArg C; int64_t n; int nb; const int64_t* hist;
+          const int64_t* bases; uint64_t* out;
+        } a2{C, n, nb, d_hist, d_bases, d_recs};
+        void* args[] = {&a2.C, &a2.n, &a2.nb, &a2.hist, &a2.bases, &a2.out};
+        hipModuleLaunchKernel(f_scat, grid, 1, 1, sthreads, 1, 1,
+                              (unsigned)(nb * 8), c->stream, args, nullptr);
+      }
     }
   } else {
     {
@@ -1931,7 +1988,7 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
     }
     hipLaunchKernelGGL(k_gbpart_scan, dim3(nb), dim3(BLOCK), 0, c->stream,
                        d_hist, grid, nb, d_totals);
-    hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(64), 0, c->stream,
+    hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(1024), 0, c->stream,
                        d_totals, nb, d_bases);
     {
       ProfScope ps(c, "k_gbpart_scatter");
@@ -1944,20 +2001,33 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   {
     ProfScope ps(c, "k_gbpart_aggregate");
     if (f_aggr) {
-      struct {
-        const uint64_t* recs; const int64_t* bases;
-        unsigned long long* counter; uint64_t* tc; uint64_t* tv;
-        unsigned long long* tg; int* ovf;
-      } a3{d_recs, d_bases, d_counter, d_tmp_codes,
-           (uint64_t*)d_tmp_vals, d_tmp_gcnt, d_ovf};
-      void* args[] = {&a3.recs, &a3.bases, &a3.counter, &a3.tc, &a3.tv,
-                      &a3.tg, &a3.ovf};
       static const int athreads = [] {
         const char* e = getenv("DSX_AGG_THREADS");
         return e ? atoi(e) : 1024;  // measured best at C2 (0.84→0.46 ms)
       }();
-      hipModuleLaunchKernel(f_aggr, nb, 1, 1, athreads, 1, 1,
-                            (unsigned)lds_bytes, c->stream, args, nullptr);
+      if (code32) {
+        struct {
+          const uint32_t* rc; const uint64_t* rv; const int64_t* bases;
+          unsigned long long* counter; uint64_t* tc; uint64_t* tv;
+          unsigned long long* tg; int* ovf;
+        } a3{d_rcodes, d_rvals, d_bases, d_counter, d_tmp_codes,
+             (uint64_t*)d_tmp_vals, d_tmp_gcnt, d_ovf};
+        void* args[] = {&a3.rc, &a3.rv, &a3.bases, &a3.counter, &a3.tc,
+                        &a3.tv, &a3.tg, &a3.ovf};
+        hipModuleLaunchKernel(f_aggr, nb, 1, 1, athreads, 1, 1,
+                              (unsigned)lds_bytes, c->stream, args, nullptr);
+      } else {
+        struct {
+          const uint64_t* recs; const int64_t* bases;
+          unsigned long long* counter; uint64_t* tc; uint64_t* tv;
+          unsigned long long* tg; int* ovf;
+        } a3{d_recs, d_bases, d_counter, d_tmp_codes,
+             (uint64_t*)d_tmp_vals, d_tmp_gcnt, d_ovf};
+        void* args[] = {&a3.recs, &a3.bases, &a3.counter, &a3.tc, &a3.tv,
+                        &a3.tg, &a3.ovf};
+        hipModuleLaunchKernel(f_aggr, nb, 1, 1, athreads, 1, 1,
+                              (unsigned)lds_bytes, c->stream, args, nullptr);
+      }
     } else {
       hipLaunchKernelGGL(k_gbpart_aggregate, dim3(nb), dim3(BLOCK),
                          lds_bytes, c->stream, d_recs, d_bases,
@@ -2063,8 +2133,8 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
     if (!part_disabled && all_nn && naggs <= 6 && g_est > 0) {
       bool fell_back = false;
       int prc = groupby_partition(c, C, n, K, P, A, aggs, progs, lens, naggs,
-                                  g_est, out_codes, out_vals, out_counts,
-                                  out_groups, &fell_back);
+                                  g_est, key_space, out_codes, out_vals,
+                                  out_counts, out_groups, &fell_back);
       if (prc != 0) return prc;
       if (!fell_back) return 0;
     }

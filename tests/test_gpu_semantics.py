@@ -354,3 +354,19 @@ def test_q3_distributed_world1_equals_plain(ctx):
     assert (distd["l_orderkey"].to_numpy().astype(np.int64)
             == plain["l_orderkey"].to_numpy().astype(np.int64)).all()
     assert np.allclose(distd["revenue"], plain["revenue"], rtol=1e-9)
+
+
+def test_integer_division_truncates_toward_zero(ctx):
+    """Reference SQLDivisionOperator maps INT/INT to C-style truncated
+    division (rex/core/call.py IntDivisionOperator: floor for positive,
+    trunc toward zero — reference uses // then casts; our VM emits trunc
+    like C++). Check negatives and exact multiples."""
+    from dask_sql_amd.context import Context
+    df = pd.DataFrame({
+        "a": np.array([7, -7, 7, -7, 9, 0], dtype=np.int64),
+        "b": np.array([2, 2, -2, -2, 3, 5], dtype=np.int64),
+    })
+    c = Context()
+    c.create_table("t", df)
+    out = c.sql("SELECT a / b AS q FROM t").compute()
+    assert out["q"].to_numpy().astype(np.int64).tolist() == [3, -3, -3, 3, 3, 0]
