@@ -444,6 +444,11 @@ __device__ __forceinline__ bool vm_eval(const DsxInstr* prog, int len, const Col
         res.f = -a.f;
         k.set(sp - 1, res, av);
         break;
+      case DSX_OP_SQRT_F64:
+        UN();
+        res.f = sqrt(a.f);
+        k.set(sp - 1, res, av);
+        break;
       case DSX_OP_NEG_I64:
         UN();
         res.i = -a.i;

@@ -457,7 +457,15 @@ class DaskAggregatePlugin(BaseRelPlugin):
     class_name = ["Aggregate", "Distinct"]
 
     AGG_OPS = {"sum", "count", "avg", "min", "max", "any_value",
-               "single_value"}
+               "single_value", "stddev", "stddev_samp", "stddev_pop",
+               "var_samp", "var_pop", "variance"}
+
+    # stddev/variance family: ONE call decomposes into TWO kernel slots
+    # (SUM(x), SUM(x*x)); finalize composes m2 = Σx² − (Σx)²/n like the
+    # reference's dask std aggregation (aggregate.py AGGREGATION_MAPPING
+    # "stddev" → dd.Aggregation over sum/count moments), ddof=1 for the
+    # sample forms (pandas default), ddof=0 for *_POP.
+    STD_FINS = {"std_samp", "std_pop", "var_samp", "var_pop"}
 
     def convert(self, rel, context):
         runtime = context._get_runtime()
@@ -553,12 +561,20 @@ class DaskAggregatePlugin(BaseRelPlugin):
                 idx = np.array([pos.get(c, -1) for c in codes_np.tolist()],
                                dtype=np.int64)
                 for call, (vals, cnts) in zip(calls, percall):
-                    av = np.zeros(len(codes_np), dtype=vals.dtype)
-                    ac = np.zeros(len(codes_np), dtype=np.uint64)
                     hit = idx >= 0
-                    av[hit] = vals[idx[hit]]
+                    ac = np.zeros(len(codes_np), dtype=np.uint64)
                     ac[hit] = cnts[idx[hit]]
-                    merged[call.toString()] = (av, ac)
+
+                    def align(v):
+                        av = np.zeros(len(codes_np), dtype=v.dtype)
+                        av[hit] = v[idx[hit]]
+                        return av
+
+                    if isinstance(vals, tuple):  # stddev family moments
+                        merged[call.toString()] = (
+                            tuple(align(v) for v in vals), ac)
+                    else:
+                        merged[call.toString()] = (align(vals), ac)
 
         return self._build_output(runtime, rel, dc, keyspecs, group_idx,
                                   codes_np, agg_calls, merged)
@@ -620,7 +636,7 @@ class DaskAggregatePlugin(BaseRelPlugin):
                     cond = Call("AND", [cond, p])
                 prog, _ = compile_expr(cond, base_cols, dicts)
                 pred_prog = runtime.make_prog(prog)
-            specs, fins = [], []
+            specs, fins, slab = [], [], []
             for call in calls:
                 func = agg.getAggregationFuncName(call).lower()
                 if func not in self.AGG_OPS:
@@ -632,9 +648,11 @@ class DaskAggregatePlugin(BaseRelPlugin):
                     expr = named[args[0].getIndex()][0]
                 else:
                     expr = PLiteral(1, SqlType("BIGINT"))
-                op, prog, fin = self._agg_spec_expr(func, expr, base_cols,
+                speclist, fin = self._agg_spec_expr(func, expr, base_cols,
                                                     dicts)
-                specs.append((op, runtime.make_prog(prog)))
+                slab.append(len(specs))
+                for op, prog in speclist:
+                    specs.append((op, runtime.make_prog(prog)))
                 fins.append(fin)
         except RexCompileError:
             return None
@@ -642,16 +660,19 @@ class DaskAggregatePlugin(BaseRelPlugin):
                      base_rel.get_current_node_type())
         return self._device_exec(runtime, rel, base_cols,
                                  base_dc.table.num_rows, keyspecs, group_meta,
-                                 pred_prog, calls, specs, fins)
+                                 pred_prog, calls, specs, fins, slab)
 
     def _convert_device(self, runtime, rel, dc, cols, keyspecs, group_idx,
                         filt_idx, calls, agg):
         """Single-bucket kernel over the already-converted input."""
         specs = []
         fins = []
+        slab = []
         for call in calls:
-            op, prog, fin = self._agg_spec_for(agg, call, cols)
-            specs.append((op, runtime.make_prog(prog)))
+            speclist, fin = self._agg_spec_for(agg, call, cols)
+            slab.append(len(specs))
+            for op, prog in speclist:
+                specs.append((op, runtime.make_prog(prog)))
             fins.append(fin)
         pred_prog = runtime.make_prog([(OP_COL, filt_idx, 0)]) \
             if filt_idx is not None else None
@@ -659,14 +680,17 @@ class DaskAggregatePlugin(BaseRelPlugin):
         group_meta = [(cc_in.columns[gi], cols[gi]) for gi in group_idx]
         return self._device_exec(runtime, rel, cols, dc.table.num_rows,
                                  keyspecs, group_meta, pred_prog, calls,
-                                 specs, fins)
+                                 specs, fins, slab)
 
     def _device_exec(self, runtime, rel, cols, n_rows, keyspecs, group_meta,
-                     pred_prog, calls, specs, fins):
-        """Run the fused kernel; finalize device-resident."""
+                     pred_prog, calls, specs, fins, slab=None):
+        """Run the fused kernel; finalize device-resident.
+        slab[i] = first kernel-slot index of call i (stddev family spans 2)."""
         from dask_sql_amd.physical.rex import (OP_LIT_I64, OP_GT_I64,
                                                OP_LIT_NULL, OP_SELECT,
                                                OP_DIV_F64, OP_I64_TO_F64)
+        if slab is None:
+            slab = list(range(len(calls)))
         oc, ov, on, G = runtime.hash_groupby(cols, n_rows,
                                              keyspecs, pred_prog, specs)
 
@@ -719,17 +743,57 @@ class DaskAggregatePlugin(BaseRelPlugin):
             order_names.append((name, f"g__{name}"))
 
         # agg finalize on device over (vals_a, cnts_a) slabs
-        for a, (call, fin) in enumerate(zip(calls, fins)):
+        from dask_sql_amd.physical.rex import (OP_LIT_F64, OP_GE_I64,
+                                               OP_SUB_I64, OP_SUB_F64,
+                                               OP_MUL_F64, OP_GT_F64,
+                                               OP_SQRT_F64)
+        for call, fin, s in zip(calls, fins, slab):
             name = call.toString()
-            val_col = rt.DeviceColumn(runtime, ov + a * G * 8, None, G,
+            val_col = rt.DeviceColumn(runtime, ov + s * G * 8, None, G,
                                       rt.F64 if fin in ("avg", "sum_f",
                                                         "min_f", "max_f")
+                                      or fin in self.STD_FINS
                                       else rt.I64,
                                       owner=False, keep_alive=holder)
-            cnt_col = rt.DeviceColumn(runtime, on + a * G * 8, None, G,
+            cnt_col = rt.DeviceColumn(runtime, on + s * G * 8, None, G,
                                       rt.I64, owner=False, keep_alive=holder)
             if fin == "count":
                 out = cnt_col  # always valid
+            elif fin in self.STD_FINS:
+                # moments: slot s = Σx (f64), slot s+1 = Σx², count = non-NULL
+                # x per group. m2 = Σx² − (Σx)²/n, clamped at 0 (fp rounding),
+                # ÷ (n−ddof); NULL below the minimum count (pandas ddof rules)
+                sq_col = rt.DeviceColumn(runtime, ov + (s + 1) * G * 8, None,
+                                         G, rt.F64, owner=False,
+                                         keep_alive=holder)
+                pop = fin.endswith("_pop")
+                prog_var = [(OP_COL, 1, 0),                       # Σx²
+                            (OP_COL, 0, 0), (OP_COL, 0, 0),
+                            (OP_MUL_F64, 0, 0),                   # (Σx)²
+                            (OP_COL, 2, 0), (OP_I64_TO_F64, 0, 0),
+                            (OP_DIV_F64, 0, 0), (OP_SUB_F64, 0, 0)]  # m2
+                if pop:
+                    prog_var += [(OP_COL, 2, 0), (OP_I64_TO_F64, 0, 0),
+                                 (OP_DIV_F64, 0, 0)]
+                else:
+                    prog_var += [(OP_COL, 2, 0), (OP_LIT_I64, 0, 1),
+                                 (OP_SUB_I64, 0, 0), (OP_I64_TO_F64, 0, 0),
+                                 (OP_DIV_F64, 0, 0)]
+                var_col = runtime.eval(runtime.make_prog(prog_var),
+                                       [val_col, sq_col, cnt_col], G, rt.F64,
+                                       with_validity=False)
+                mink = 1 if pop else 2
+                prog_fin = [(OP_COL, 1, 0), (OP_LIT_I64, 0, mink),
+                            (OP_GE_I64, 0, 0),
+                            (OP_COL, 0, 0), (OP_LIT_F64, 0, 0.0),
+                            (OP_GT_F64, 0, 0),
+                            (OP_COL, 0, 0), (OP_LIT_F64, 0, 0.0),
+                            (OP_SELECT, 0, 0)]                   # max(var, 0)
+                if fin.startswith("std"):
+                    prog_fin += [(OP_SQRT_F64, 0, 0)]
+                prog_fin += [(OP_LIT_NULL, 0, 0), (OP_SELECT, 0, 0)]
+                out = runtime.eval(runtime.make_prog(prog_fin),
+                                   [var_col, cnt_col], G, rt.F64)
             elif fin == "avg":
                 prog = [(OP_COL, 1, 0), (OP_LIT_I64, 0, 0), (OP_GT_I64, 0, 0),
                         (OP_COL, 0, 0),
@@ -756,7 +820,7 @@ class DaskAggregatePlugin(BaseRelPlugin):
 
     # ------------------------------------------------------------------
     def _agg_spec_for(self, agg, call, cols):
-        """(kernel op, program, finalize) for one agg call."""
+        """([(kernel op, program), ...], finalize) for one agg call."""
         func = agg.getAggregationFuncName(call).lower()
         args = agg.getArgs(call)
         return self._agg_spec_expr(func, args[0] if args else None, cols,
@@ -768,43 +832,59 @@ class DaskAggregatePlugin(BaseRelPlugin):
         else:
             prog, kind = [(3, 0, 1)], KI  # LIT_I64 1 — COUNT(*)
         if func == "count":
-            return rt.AGG_COUNT, prog, "count"
+            return [(rt.AGG_COUNT, prog)], "count"
         if func == "avg":
             if kind != KF:
                 prog = prog + [(50, 0, 0)]  # I64_TO_F64
-            return rt.AGG_SUM_F64, prog, "avg"
+            return [(rt.AGG_SUM_F64, prog)], "avg"
         if func == "sum":
             if kind == KF:
-                return rt.AGG_SUM_F64, prog, "sum_f"
-            return rt.AGG_SUM_I64, prog, "sum_i"
+                return [(rt.AGG_SUM_F64, prog)], "sum_f"
+            return [(rt.AGG_SUM_I64, prog)], "sum_i"
         if func in ("min", "any_value", "single_value"):
-            return (rt.AGG_MIN_F64 if kind == KF else rt.AGG_MIN_I64), prog, \
-                ("min_f" if kind == KF else "min_i")
+            return ([((rt.AGG_MIN_F64 if kind == KF else rt.AGG_MIN_I64),
+                      prog)],
+                    ("min_f" if kind == KF else "min_i"))
         if func == "max":
-            return (rt.AGG_MAX_F64 if kind == KF else rt.AGG_MAX_I64), prog, \
-                ("max_f" if kind == KF else "max_i")
+            return ([((rt.AGG_MAX_F64 if kind == KF else rt.AGG_MAX_I64),
+                      prog)],
+                    ("max_f" if kind == KF else "max_i"))
+        if func in ("stddev", "stddev_samp", "stddev_pop", "var_samp",
+                    "var_pop", "variance"):
+            if kind != KF:
+                prog = prog + [(50, 0, 0)]  # I64_TO_F64
+            sq = prog + prog + [(12, 0, 0)]  # MUL_F64 (x·x, NULL iff x NULL)
+            fin = {"stddev": "std_samp", "stddev_samp": "std_samp",
+                   "stddev_pop": "std_pop", "var_samp": "var_samp",
+                   "variance": "var_samp", "var_pop": "var_pop"}[func]
+            return [(rt.AGG_SUM_F64, prog), (rt.AGG_SUM_F64, sq)], fin
         raise RexCompileError(f"aggregate {func}")
 
     def _run_bucket(self, runtime, agg, dc, cols, keyspecs, filt_idx, calls):
         """One fused kernel pass. Returns (codes np.uint64 sorted,
-        [(vals np, cnts np) per call])."""
+        [(vals np | tuple of np, cnts np) per call]) — stddev-family calls
+        carry a (Σx, Σx²) tuple."""
         pred_prog = None
         if filt_idx is not None:
             pred_prog = runtime.make_prog([(OP_COL, filt_idx, 0)])
         specs = []
         fins = []
+        slab = []
         for call in calls:
-            op, prog, fin = self._agg_spec_for(agg, call, cols)
-            specs.append((op, runtime.make_prog(prog)))
+            speclist, fin = self._agg_spec_for(agg, call, cols)
+            slab.append(len(specs))
+            for op, prog in speclist:
+                specs.append((op, runtime.make_prog(prog)))
             fins.append(fin)
+        nspec = len(specs)
         oc, ov, on, G = runtime.hash_groupby(cols, dc.table.num_rows,
                                              keyspecs, pred_prog, specs)
         codes = np.empty(G, dtype=np.uint64)
         if G:
             runtime._download(oc, codes)
-        vals_all = np.empty(G * max(len(calls), 1), dtype=np.uint64)
-        cnts_all = np.empty(G * max(len(calls), 1), dtype=np.uint64)
-        if G and calls:
+        vals_all = np.empty(G * max(nspec, 1), dtype=np.uint64)
+        cnts_all = np.empty(G * max(nspec, 1), dtype=np.uint64)
+        if G and specs:
             runtime._download(ov, vals_all)
             runtime._download(on, cnts_all)
         runtime._free(oc)
@@ -813,16 +893,22 @@ class DaskAggregatePlugin(BaseRelPlugin):
         # sort by code for deterministic output & merging
         order = np.argsort(codes, kind="stable")
         codes = codes[order]
+
+        def slot(a, as_f64):
+            u = vals_all[a * G:(a + 1) * G][order] if G else \
+                np.empty(0, dtype=np.uint64)
+            return u.view(np.float64) if as_f64 else u.view(np.int64)
+
         percall = []
-        for a, fin in enumerate(fins):
-            vals_u64 = vals_all[a * G:(a + 1) * G][order] if G else \
+        for fin, s in zip(fins, slab):
+            cnts = cnts_all[s * G:(s + 1) * G][order] if G else \
                 np.empty(0, dtype=np.uint64)
-            cnts = cnts_all[a * G:(a + 1) * G][order] if G else \
-                np.empty(0, dtype=np.uint64)
-            if fin in ("avg", "sum_f", "min_f", "max_f"):
-                vals = vals_u64.view(np.float64)
+            if fin in self.STD_FINS:
+                vals = (slot(s, True), slot(s + 1, True))
+            elif fin in ("avg", "sum_f", "min_f", "max_f"):
+                vals = slot(s, True)
             else:
-                vals = vals_u64.view(np.int64)
+                vals = slot(s, False)
             percall.append((vals, cnts))
         return codes, percall
 
@@ -938,6 +1024,20 @@ class DaskAggregatePlugin(BaseRelPlugin):
             has_null = (cnts == 0).any()
             if func == "count":
                 col = runtime.upload_column(cnts.astype(np.int64))
+            elif func in ("stddev", "stddev_samp", "stddev_pop", "var_samp",
+                          "var_pop", "variance"):
+                s, ss = vals  # (Σx, Σx²) moments from the two kernel slots
+                pop = func.endswith("_pop")
+                n = cnts.astype(np.float64)
+                with np.errstate(invalid="ignore", divide="ignore"):
+                    m2 = np.maximum(ss - s * s / n, 0.0)
+                    v = m2 / (n if pop else n - 1.0)
+                if func.startswith("stddev"):
+                    v = np.sqrt(v)
+                ok = cnts >= (1 if pop else 2)
+                v = np.where(ok, v, 0.0)
+                col = runtime.upload_column(
+                    v, validity=ok.astype(np.uint8) if not ok.all() else None)
             elif func == "avg":
                 with np.errstate(invalid="ignore", divide="ignore"):
                     a = vals.astype(np.float64) / cnts.astype(np.float64)

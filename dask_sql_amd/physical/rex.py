@@ -20,7 +20,7 @@ OP_LT_I64, OP_LE_I64, OP_GT_I64, OP_GE_I64, OP_EQ_I64, OP_NE_I64 = (
     30, 31, 32, 33, 34, 35)
 OP_AND, OP_OR, OP_NOT, OP_IS_NULL, OP_IS_NOT_NULL = 40, 41, 42, 43, 44
 OP_I64_TO_F64, OP_F64_TO_I64 = 50, 51
-OP_SELECT, OP_NEG_F64, OP_NEG_I64 = 60, 61, 62
+OP_SELECT, OP_NEG_F64, OP_NEG_I64, OP_SQRT_F64 = 60, 61, 62, 63
 
 # VM value kinds
 KI, KF, KB = "i", "f", "b"  # int64-like, float64, boolean
@@ -44,6 +44,21 @@ _ARITH = {"+": (OP_ADD_I64, OP_ADD_F64), "-": (OP_SUB_I64, OP_SUB_F64),
 
 class RexCompileError(NotImplementedError):
     pass
+
+
+def _like_regex(pattern: str):
+    """SQL LIKE pattern → compiled regex: % = any run, _ = any single char,
+    everything else literal (reference rex/core/call.py LIKE lowering)."""
+    import re
+    out = []
+    for ch in pattern:
+        if ch == "%":
+            out.append(".*")
+        elif ch == "_":
+            out.append(".")
+        else:
+            out.append(re.escape(ch))
+    return re.compile("".join(out), re.DOTALL)
 
 
 class RexCompiler:
@@ -158,7 +173,44 @@ class RexCompiler:
         if op == "CASE":
             # operands: cond1, val1, cond2, val2, ..., else
             return self._compile_case(ops)
+        if op == "LIKE":
+            return self._compile_like(ops)
         raise RexCompileError(f"operator {op} not supported on GPU path")
+
+    def _compile_like(self, ops):
+        """LIKE on a dict-encoded column: the SQL pattern (%/_ wildcards,
+        reference rex/core/call.py SargPythonImplementation / re-based LIKE
+        lowering) is matched against the (small, host-resident) dictionary
+        once at compile time; the kernel-side predicate is an OR-chain of
+        integer code equalities, so NULL → NULL falls out of EQ validity."""
+        col, pat = ops
+        if not (isinstance(col, InputRef) and isinstance(pat, Literal)
+                and isinstance(pat.getValue(), str)):
+            raise RexCompileError("LIKE needs <column> LIKE '<pattern>'")
+        d = self.dicts[col.getIndex()]
+        if d is None:
+            raise RexCompileError("LIKE on non-dict-encoded column")
+        rx = _like_regex(pat.getValue())
+        matched = [i for i, s in enumerate(d)
+                   if s is not None and rx.fullmatch(s)]
+        ci = col.getIndex()
+        if len(matched) > 10:
+            raise RexCompileError(
+                f"LIKE matches {len(matched)} dictionary entries "
+                "(> VM program budget)")
+        if not matched:
+            # FALSE for every present code, NULL stays NULL (-2 never a code)
+            self._emit(OP_COL, ci)
+            self._emit(OP_LIT_I64, 0, -2)
+            self._emit(OP_EQ_I64)
+            return KB
+        for j, code in enumerate(matched):
+            self._emit(OP_COL, ci)
+            self._emit(OP_LIT_I64, 0, code)
+            self._emit(OP_EQ_I64)
+            if j:
+                self._emit(OP_OR)
+        return KB
 
     def _compile_case(self, ops):
         # rightmost-else first; build nested SELECTs. Postfix SELECT pops
@@ -231,7 +283,7 @@ class RexCompiler:
         if isinstance(expr, Call):
             op = expr.getOperatorName()
             if op in _CMP or op in ("AND", "OR", "NOT", "IS NULL",
-                                    "IS NOT NULL"):
+                                    "IS NOT NULL", "LIKE"):
                 return KB
             if op in _ARITH:
                 ka = self._peek_kind(expr.getOperands()[0])

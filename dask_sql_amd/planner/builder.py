@@ -481,7 +481,8 @@ class Builder:
                 arg_t = "BIGINT"
             if func == "count":
                 out_t = "BIGINT"
-            elif func == "avg":
+            elif func in ("avg", "stddev", "stddev_samp", "stddev_pop",
+                          "var_samp", "var_pop", "variance"):
                 out_t = "DOUBLE"
             elif func == "sum":
                 out_t = "DOUBLE" if _is_float(arg_t) else "BIGINT"

@@ -90,7 +90,8 @@ def tokenize(sql: str):
 
 
 AGG_FUNCS = {"SUM", "COUNT", "AVG", "MIN", "MAX", "ANY_VALUE", "STDDEV",
-             "STDDEV_POP", "SINGLE_VALUE", "EVERY", "BIT_AND", "BIT_OR"}
+             "STDDEV_POP", "STDDEV_SAMP", "VAR_SAMP", "VAR_POP", "VARIANCE",
+             "SINGLE_VALUE", "EVERY", "BIT_AND", "BIT_OR"}
 
 
 class Parser:
@@ -290,9 +291,17 @@ class Parser:
                 e = ("call", "AND",
                      [("call", ">=", [e, lo]), ("call", "<=", [e, hi])])
                 continue
+            if self.accept_kw("LIKE"):
+                pat = self.add_expr()
+                e = ("call", "LIKE", [e, pat])
+                continue
             if self.accept_kw("NOT"):
                 if self.accept_kw("IN"):
                     e = ("call", "NOT", [self._in_list(e)])
+                    continue
+                if self.accept_kw("LIKE"):
+                    pat = self.add_expr()
+                    e = ("call", "NOT", [("call", "LIKE", [e, pat])])
                     continue
                 if self.accept_kw("BETWEEN"):
                     lo = self.add_expr()

@@ -63,7 +63,7 @@ enum DsxOp {
   DSX_OP_I64_TO_F64 = 50, DSX_OP_F64_TO_I64 = 51,     /* CAST (trunc,
                                                          mappings.py:346-353) */
   DSX_OP_SELECT = 60,   /* (cond, a, b) -> cond ? a : b — CASE WHEN           */
-  DSX_OP_NEG_F64 = 61, DSX_OP_NEG_I64 = 62,
+  DSX_OP_NEG_F64 = 61, DSX_OP_NEG_I64 = 62, DSX_OP_SQRT_F64 = 63,
 };
 
 typedef struct DsxInstr {

@@ -158,6 +158,14 @@ _AGG_FUNCS = {
     "max": "max",
     "any_value": "first",
     "single_value": "first",
+    # stddev family: pandas std/var, ddof=1 for sample forms (the
+    # reference's dask "std" aggregation default), ddof=0 for *_POP
+    "stddev": "std",
+    "stddev_samp": "std",
+    "stddev_pop": lambda s: s.std(ddof=0),
+    "var_samp": "var",
+    "variance": "var",
+    "var_pop": lambda s: s.var(ddof=0),
 }
 
 

@@ -370,3 +370,87 @@ def test_integer_division_truncates_toward_zero(ctx):
     c.create_table("t", df)
     out = c.sql("SELECT a / b AS q FROM t").compute()
     assert out["q"].to_numpy().astype(np.int64).tolist() == [3, -3, -3, 3, 3, 0]
+
+
+def test_stddev_variance_family(ctx):
+    """STDDEV/VAR family vs the oracle (pandas std/var, ddof rules):
+    reference aggregate.py AGGREGATION_MAPPING "stddev" custom aggregation.
+    Covers NULLs (skipped like pandas), single-row groups (samp -> NULL,
+    pop -> 0), and the fused filter path."""
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(77)
+    n = 50_000
+    df = pd.DataFrame({
+        "k": rng.integers(0, 500, n).astype(np.int64),
+        "x": np.round(rng.normal(10, 3, n), 4),
+    })
+    df.loc[rng.random(n) < 0.1, "x"] = np.nan
+    # a few forced single-row groups
+    extra = pd.DataFrame({"k": [900, 901], "x": [5.0, np.nan]})
+    df = pd.concat([df, extra], ignore_index=True)
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql(
+        "SELECT k, STDDEV(x) AS sd, STDDEV_POP(x) AS sdp, VAR_SAMP(x) AS vs, "
+        "VAR_POP(x) AS vp, COUNT(x) AS c FROM t GROUP BY k").compute()
+    got = got.sort_values("k").reset_index(drop=True)
+    exp = oracle_groupby(df, ["k"], [
+        ("x", "sd", "stddev", None, False),
+        ("x", "sdp", "stddev_pop", None, False),
+        ("x", "vs", "var_samp", None, False),
+        ("x", "vp", "var_pop", None, False),
+        ("x", "c", "count", None, False),
+    ]).sort_values("k").reset_index(drop=True)
+    assert (got["k"].to_numpy() == exp["k"].to_numpy()).all()
+    for col in ("sd", "sdp", "vs", "vp"):
+        g = got[col].to_numpy(dtype=np.float64)
+        e = exp[col].to_numpy(dtype=np.float64)
+        assert np.isnan(g).tolist() == np.isnan(e).tolist(), col
+        m = ~np.isnan(e)
+        np.testing.assert_allclose(g[m], e[m], rtol=1e-9, atol=1e-12,
+                                   err_msg=col)
+    assert (got["c"].to_numpy() == exp["c"].to_numpy()).all()
+
+
+def test_stddev_fused_with_where(ctx):
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(5)
+    df = pd.DataFrame({
+        "k": rng.integers(0, 50, 20_000).astype(np.int64),
+        "x": rng.random(20_000) * 10,
+    })
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT k, STDDEV(x) AS sd FROM t WHERE x > 2.5 "
+                "GROUP BY k").compute().sort_values("k").reset_index(drop=True)
+    pdf = df[df.x > 2.5]
+    exp = pdf.groupby("k")["x"].std().reset_index()
+    np.testing.assert_allclose(got["sd"].to_numpy(),
+                               exp["x"].to_numpy(), rtol=1e-9)
+    assert (got["k"].to_numpy() == exp["k"].to_numpy()).all()
+
+
+def test_like_on_dict_strings(ctx):
+    """LIKE over dict-encoded VARCHAR: pattern evaluated against the
+    dictionary at compile time -> OR-chain of code equalities in the kernel
+    (reference rex/core/call.py LIKE lowering via re)."""
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(9)
+    cats = ["BUILDING", "AUTOMOBILE", "MACHINERY", "HOUSEHOLD", "FURNITURE"]
+    n = 30_000
+    seg = pd.Series(rng.choice(cats, n)).astype("category")
+    df = pd.DataFrame({"seg": seg, "v": rng.integers(0, 100, n)})
+    df.loc[rng.random(n) < 0.05, "seg"] = None
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT COUNT(*) AS c, SUM(v) AS s FROM t "
+                "WHERE seg LIKE '%U%LD%'").compute()
+    m = df["seg"].astype(object).str.fullmatch(".*U.*LD.*", na=False)
+    assert int(got["c"][0]) == int(m.sum())
+    assert int(got["s"][0]) == int(df.loc[m, "v"].sum())
+    got2 = c.sql("SELECT COUNT(*) AS c FROM t "
+                 "WHERE seg NOT LIKE 'M_CHINERY'").compute()
+    notm = (~(df["seg"] == "MACHINERY")) & df["seg"].notna()
+    assert int(got2["c"][0]) == int(notm.sum())
+    got3 = c.sql("SELECT COUNT(*) AS c FROM t WHERE seg LIKE 'ZZZ%'").compute()
+    assert int(got3["c"][0]) == 0

@@ -207,3 +207,19 @@ def test_rex_compiler_programs(c, user_table_1):
     ops = [p[0] for p in prog]
     assert 41 in ops  # OR
     assert 30 in ops and 34 in ops  # int compares
+
+
+def test_like_parses_and_types():
+    from dask_sql_amd.planner.parser import Parser
+    ast = Parser("SELECT a FROM t WHERE b LIKE 'x%' AND c NOT LIKE '_y'"
+                 ).parse()
+    s = repr(ast)
+    assert "LIKE" in s
+
+
+def test_like_regex_translation():
+    from dask_sql_amd.physical.rex import _like_regex
+    rx = _like_regex("AB%c_d.e")
+    assert rx.fullmatch("ABzzzcXd.e")
+    assert not rx.fullmatch("ABcXdYe")  # literal dot must match
+    assert rx.fullmatch("AB%c_d.e".replace("%", "").replace("_", "Q"))
