@@ -694,6 +694,32 @@ class DaskAggregatePlugin(BaseRelPlugin):
         oc, ov, on, G = runtime.hash_groupby(cols, n_rows,
                                              keyspecs, pred_prog, specs)
 
+        if G == 0 and not group_meta:
+            # global aggregate over zero selected rows: SQL mandates ONE row
+            # (COUNT = 0, everything else NULL) — reference aggregate.py:251
+            # agg-on-whole-frame path
+            runtime._free(oc)
+            runtime._free(ov)
+            runtime._free(on)
+            out_cols = {}
+            order_names = []
+            for call, fin in zip(calls, fins):
+                name = call.toString()
+                if fin == "count":
+                    col = runtime.upload_column(np.zeros(1, dtype=np.int64))
+                else:
+                    f64 = fin in ("avg", "sum_f", "min_f", "max_f") \
+                        or fin in self.STD_FINS
+                    col = runtime.upload_column(
+                        np.zeros(1, dtype=np.float64 if f64 else np.int64),
+                        validity=np.zeros(1, dtype=np.uint8))
+                out_cols[f"a__{name}"] = col
+                order_names.append((name, f"a__{name}"))
+            cc = ColumnContainer([n for n, _ in order_names],
+                                 dict(order_names))
+            cc = self.fix_column_to_row_type(cc, rel.getRowType())
+            return DataContainer(DeviceTable(out_cols, num_rows=1), cc)
+
         class _Holder:
             def __init__(self, runtime, ptrs):
                 self.runtime = runtime
@@ -984,6 +1010,19 @@ class DaskAggregatePlugin(BaseRelPlugin):
         """Unpack group codes → key columns; finalize agg columns
         (SUM min_count=1 → NULL on zero count; AVG = sum/count)."""
         G = len(codes_np)
+        if G == 0 and not keyspecs and agg_calls:
+            # global aggregate over zero rows → ONE row (COUNT=0, rest NULL);
+            # zero-count synthesis lets the normal finalize below produce it
+            codes_np = np.zeros(1, dtype=np.uint64)
+            fixed = {}
+            for k, (vals, cnts) in merged.items():
+                if isinstance(vals, tuple):
+                    vals = tuple(np.zeros(1, dtype=v.dtype) for v in vals)
+                else:
+                    vals = np.zeros(1, dtype=vals.dtype)
+                fixed[k] = (vals, np.zeros(1, dtype=np.uint64))
+            merged = fixed
+            G = 1
         out_cols = {}
         order_names = []
         cols = dc.backend_cols()

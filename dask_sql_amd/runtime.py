@@ -35,7 +35,7 @@ _DSX_SIZE = {I64: 8, F64: 8, I32: 4, F32: 4, I8: 1, BOOL8: 1}
 
 MAX_PROG = 48
 MAX_COLS = 16
-MAX_AGGS = 8
+MAX_AGGS = 16
 
 # agg ops (include/dsxhip.h DsxAggOp)
 AGG_SUM_F64, AGG_SUM_I64, AGG_COUNT = 0, 1, 2

@@ -74,7 +74,7 @@ typedef struct DsxInstr {
 
 #define DSX_MAX_PROG 48
 #define DSX_MAX_COLS 16
-#define DSX_MAX_AGGS 8
+#define DSX_MAX_AGGS 16
 #define DSX_MAX_KEYS 4
 
 typedef struct DsxColumn {

@@ -215,9 +215,20 @@ def oracle_groupby(
         }
         if group_cols:
             g = tmp.groupby(list(group_cols), dropna=False)
-        else:
-            g = tmp.groupby([const_col])
-        return g.agg(**named)
+            return g.agg(**named)
+        g = tmp.groupby([const_col])
+        res = g.agg(**named)
+        if len(res) == 0:
+            # global aggregate over zero rows: SQL yields ONE row — count 0,
+            # everything else NULL (reference aggregate.py:251 whole-frame agg)
+            def _apply(ic, f):
+                fn = _AGG_FUNCS[f]
+                s = pd.Series(tmp[ic])
+                return fn(s) if callable(fn) else s.agg(fn)
+
+            row = {out_name: [_apply(ic, f)] for ic, out_name, f in items}
+            res = pd.DataFrame(row, index=pd.Index([1], name=const_col))
+        return res
 
     # non-filtered, non-distinct bucket first (aggregate.py:336-350)
     df_result = None

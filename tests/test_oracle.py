@@ -232,3 +232,18 @@ def test_q3_small():
     assert list(out.columns) == ["l_orderkey", "revenue", "o_orderdate", "o_shippriority"]
     # revenue sorted descending
     assert (np.diff(out["revenue"].to_numpy()) <= 1e-9).all()
+
+
+def test_oracle_global_agg_empty_selection():
+    """SQL global aggregate over zero rows -> one row: COUNT 0, SUM/AVG NULL
+    (reference aggregate.py:251 whole-frame agg; test_groupby.py empty-filter
+    cases)."""
+    df = pd.DataFrame({"v": [1.0, 2.0], "f": [False, False]})
+    out = oracle_groupby(df[df.f], [], [
+        ("v", "c", "count", None, False),
+        ("v", "s", "sum", None, False),
+        ("v", "a", "avg", None, False),
+    ])
+    assert len(out) == 1
+    assert out["c"].iloc[0] == 0
+    assert np.isnan(out["s"].iloc[0]) and np.isnan(out["a"].iloc[0])
