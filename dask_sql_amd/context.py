@@ -42,7 +42,8 @@ def _from_pandas(df: pd.DataFrame, date_columns=(), dictionaries=None):
         if isinstance(dt, pd.CategoricalDtype):
             codes = s.cat.codes.to_numpy()
             validity = (codes >= 0).astype(np.uint8)
-            arr, dtype, sqlt = codes.astype(np.int32), rt.I32, "VARCHAR"
+            arr = np.where(codes >= 0, codes, 0).astype(np.int32)
+            dtype, sqlt = rt.I32, "VARCHAR"
             dictionary = list(s.cat.categories)
             if validity.all():
                 validity = None
@@ -59,7 +60,8 @@ def _from_pandas(df: pd.DataFrame, date_columns=(), dictionaries=None):
             validity = (codes >= 0).astype(np.uint8)
             if validity.all():
                 validity = None
-            arr, dtype, sqlt = codes.astype(np.int32), rt.I32, "VARCHAR"
+            arr = np.where(codes >= 0, codes, 0).astype(np.int32)
+            dtype, sqlt = rt.I32, "VARCHAR"
             dictionary = list(uniques)
         elif np.issubdtype(dt, np.datetime64):
             vals = s.to_numpy().astype("datetime64[D]").astype(np.int32)
@@ -199,10 +201,27 @@ class Context:
                      **kwargs):
         if isinstance(input_table, dict):
             input_table = pd.DataFrame(input_table)
+        if isinstance(input_table, str):
+            # file-path inputs (reference input_utils/location.py:22-60:
+            # format inferred from the extension; parquet via pyarrow,
+            # csv via pandas — the same engines the reference delegates to)
+            fmt = kwargs.get("format")
+            low = input_table.lower()
+            if fmt == "parquet" or low.endswith(".parquet") \
+                    or low.endswith(".parq"):
+                input_table = pd.read_parquet(input_table)
+            elif fmt == "csv" or low.endswith(".csv"):
+                input_table = pd.read_csv(input_table)
+            elif fmt == "json" or low.endswith(".json"):
+                input_table = pd.read_json(input_table)
+            else:
+                raise NotImplementedError(
+                    f"cannot infer input format of {input_table!r} "
+                    "(pass format='parquet'|'csv'|'json')")
         if not isinstance(input_table, pd.DataFrame):
             raise NotImplementedError(
-                "only pandas/dict inputs (input plugins are out of scope, "
-                "SURVEY §2)")
+                "only pandas/dict/path inputs (other input plugins are out "
+                "of scope, SURVEY §2)")
         host_cols = _from_pandas(input_table, date_columns, dictionaries)
         t = RegisteredTable(host_cols)
         self.tables[table_name.lower()] = t

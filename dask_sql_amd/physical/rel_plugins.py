@@ -278,6 +278,48 @@ class DaskJoinPlugin(BaseRelPlugin):
         return DataContainer(DeviceTable(out_cols, num_rows=n_out), cc)
 
     # -- helpers ------------------------------------------------------------
+    def _reconcile_dict_keys(self, runtime, dc_lhs, dc_rhs, lhs_on, rhs_on):
+        """String join keys: the reference merges on the STRINGS
+        (join.py:241-246 dd.merge on object columns); our per-table
+        factorization makes raw dict codes incomparable across tables, so
+        rhs key codes are remapped into the lhs dictionary space on device
+        (host builds the small code→code map, one gather applies it;
+        strings absent from the lhs dictionary get a sentinel code that can
+        never match). Returns a shadow (dc_rhs, rcols) for key building only
+        — materialization keeps the original codes + dictionary."""
+        lcols = dc_lhs.backend_cols()
+        rcols = dc_rhs.backend_cols()
+        sub = {}
+        for li, ri in zip(lhs_on, rhs_on):
+            dl = getattr(lcols[li], "dictionary", None)
+            dr = getattr(rcols[ri], "dictionary", None)
+            if dl is None and dr is None:
+                continue
+            if (dl is None) != (dr is None):
+                raise RexCompileError(
+                    "join key pairs a string column with a non-string column")
+            if dl is dr or list(dl) == list(dr):
+                continue  # same dictionary → codes already comparable
+            index = {s: c for c, s in enumerate(dl)}
+            miss = len(dl)  # sentinel: never present on the lhs side
+            m = np.array([index.get(s, miss) for s in dr], dtype=np.int64)
+            map_col = runtime.upload_column(m)
+            g = runtime.gather(map_col, rcols[ri].data,
+                               dc_rhs.table.num_rows)
+            rem = rt.DeviceColumn(runtime, g.data, rcols[ri].validity,
+                                  g.len, rt.I64, owner=False,
+                                  keep_alive=(g, rcols[ri], map_col))
+            rem.dictionary = dl
+            sub[ri] = rem
+        if not sub:
+            return dc_rhs, rcols
+        cc = dc_rhs.column_container
+        cols2 = dict(dc_rhs.table.columns)
+        for ri, col in sub.items():
+            cols2[cc.get_backend_by_frontend_name(cc.columns[ri])] = col
+        dc2 = DataContainer(DeviceTable(cols2), cc)
+        return dc2, dc2.backend_cols()
+
     def _key_codes(self, runtime, dc, on, ranges):
         """Build i64 code column + optional validity for join keys."""
         cols = dc.backend_cols()
@@ -312,6 +354,8 @@ class DaskJoinPlugin(BaseRelPlugin):
         for i in rhs_on:
             if rcols[i].dtype not in _INT_KINDS:
                 raise RexCompileError("non-integer join keys (round-2)")
+        dc_rhs, rcols = self._reconcile_dict_keys(runtime, dc_lhs, dc_rhs,
+                                                  lhs_on, rhs_on)
         # combined ranges over both sides so codes are comparable
         ranges = []
         for li, ri in zip(lhs_on, rhs_on):

@@ -454,3 +454,36 @@ def test_like_on_dict_strings(ctx):
     assert int(got2["c"][0]) == int(notm.sum())
     got3 = c.sql("SELECT COUNT(*) AS c FROM t WHERE seg LIKE 'ZZZ%'").compute()
     assert int(got3["c"][0]) == 0
+
+
+def test_string_join_keys_across_dictionaries(ctx):
+    """Join on string columns factorized independently per table: codes are
+    remapped into one dictionary space on device, so the join compares the
+    STRINGS like the reference's dd.merge (join.py:241-246)."""
+    from dask_sql_amd.context import Context
+    left = pd.DataFrame({
+        "name": ["apple", "pear", "plum", "apple", None, "kiwi"],
+        "a": [1, 2, 3, 4, 5, 6],
+    })
+    right = pd.DataFrame({
+        "name": ["plum", "mango", "apple", None, "apple"],
+        "b": [10, 20, 30, 40, 50],
+    })
+    c = Context()
+    c.create_table("l", left)
+    c.create_table("r", right)
+    got = c.sql("SELECT lhs.name, lhs.a, rhs.b FROM l lhs "
+                "JOIN r rhs ON lhs.name = rhs.name").compute()
+    exp = left.merge(right, on="name")
+    assert sorted(map(tuple, got[["a", "b"]].astype(int).to_numpy())) == \
+        sorted(map(tuple, exp[["a", "b"]].to_numpy()))
+    assert set(got["name"]) == set(exp["name"])
+    # LEFT join keeps unmatched + NULL-key lhs rows, rhs NULL-filled
+    got2 = c.sql("SELECT lhs.a, rhs.b FROM l lhs "
+                 "LEFT JOIN r rhs ON lhs.name = rhs.name").compute()
+    exp2 = left.merge(right, on="name", how="left")
+    g = sorted(map(tuple, np.nan_to_num(
+        got2[["a", "b"]].astype(float).to_numpy(), nan=-1)))
+    e = sorted(map(tuple, np.nan_to_num(
+        exp2[["a", "b"]].astype(float).to_numpy(), nan=-1)))
+    assert g == e

@@ -223,3 +223,26 @@ def test_like_regex_translation():
     assert rx.fullmatch("ABzzzcXd.e")
     assert not rx.fullmatch("ABcXdYe")  # literal dot must match
     assert rx.fullmatch("AB%c_d.e".replace("%", "").replace("_", "Q"))
+
+
+def test_create_table_from_files(tmp_path):
+    """File-path create_table (reference context.py:168 +
+    input_utils/location.py extension dispatch)."""
+    import pandas as pd
+    import numpy as np
+    from dask_sql_amd.context import Context
+    df = pd.DataFrame({"a": np.arange(5, dtype=np.int64),
+                       "b": np.linspace(0, 1, 5)})
+    pq = tmp_path / "t.parquet"
+    cv = tmp_path / "t.csv"
+    df.to_parquet(pq)
+    df.to_csv(cv, index=False)
+    c = Context()
+    c.create_table("tp", str(pq))
+    c.create_table("tc", str(cv))
+    for t in ("tp", "tc"):
+        fields = dict(c.tables[t].fields())
+        assert fields["a"] == "BIGINT" and fields["b"] == "DOUBLE"
+    import pytest
+    with pytest.raises(NotImplementedError):
+        c.create_table("tx", "no_such.xyz")
