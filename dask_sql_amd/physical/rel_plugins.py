@@ -190,8 +190,18 @@ class DaskJoinPlugin(BaseRelPlugin):
                 condition, n_lhs_cols)
 
         if lhs_on:
-            pairs, n_out = self._equi_join(runtime, dc_lhs, dc_rhs, lhs_on,
-                                           rhs_on, join_type)
+            if residual and join_type == "leftanti":
+                # ANTI with extra condition: keep lhs rows with NO rhs match
+                # satisfying key AND residual — inner pairs → residual filter
+                # → complement of surviving probe ids (join.py:169-181
+                # composed with the anti semantics of :78-90)
+                pairs, n_out = self._anti_residual(
+                    runtime, dc_lhs, dc_rhs, lhs_on, rhs_on, residual,
+                    cc_lhs, cc_rhs, n_lhs_cols)
+                residual = []
+            else:
+                pairs, n_out = self._equi_join(runtime, dc_lhs, dc_rhs,
+                                               lhs_on, rhs_on, join_type)
         else:
             pairs, n_out = self._cross_join(runtime, dc_lhs, dc_rhs, join_type)
         probe_sel, build_sel = pairs
@@ -208,9 +218,6 @@ class DaskJoinPlugin(BaseRelPlugin):
                                                  None) is not None \
             else list(range(len(combined)))
         from dask_sql_amd.planner.prune import _expr_refs, _remap
-        if residual and join_type == "leftanti":
-            raise NotImplementedError(
-                "residual condition on LEFT ANTI join (round-2)")
         res_refs = set()
         for r in residual:
             _expr_refs(r, res_refs)
@@ -278,6 +285,58 @@ class DaskJoinPlugin(BaseRelPlugin):
         return DataContainer(DeviceTable(out_cols, num_rows=n_out), cc)
 
     # -- helpers ------------------------------------------------------------
+    def _anti_residual(self, runtime, dc_lhs, dc_rhs, lhs_on, rhs_on,
+                       residual, cc_lhs, cc_rhs, n_lhs_cols):
+        """LEFT ANTI with a residual condition: inner pairs, residual filter
+        on the pairs, then the complement of the surviving probe ids (host
+        complement — this shape is rare and the surviving-id set is small)."""
+        from dask_sql_amd.planner.prune import _expr_refs, _remap
+        (pi, bi), n_i = self._equi_join(runtime, dc_lhs, dc_rhs, lhs_on,
+                                        rhs_on, "inner")
+        n_l = dc_lhs.table.num_rows
+        matched = np.empty(0, dtype=np.uint32)
+        if n_i:
+            refs = set()
+            for r in residual:
+                _expr_refs(r, refs)
+            refs = sorted(refs)
+            cols_list = []
+            for i in refs:
+                if i < n_lhs_cols:
+                    col = dc_lhs.table.col(cc_lhs.get_backend_by_frontend_name(
+                        cc_lhs.columns[i]))
+                    g = runtime.gather(col, pi.data, n_i)
+                else:
+                    col = dc_rhs.table.col(cc_rhs.get_backend_by_frontend_name(
+                        cc_rhs.columns[i - n_lhs_cols]))
+                    g = runtime.gather(col, bi.data, n_i)
+                if getattr(col, "dictionary", None) is not None:
+                    g.dictionary = col.dictionary
+                cols_list.append(g)
+            cond = residual[0]
+            for r in residual[1:]:
+                cond = Call("AND", [cond, r])
+            cond = _remap(cond, {i: pos for pos, i in enumerate(refs)})
+            s = scalar_literal(cond)
+            if s is None:
+                prog, _ = compile_expr(cond, cols_list, _dicts_of(cols_list))
+                sel_ptr, cnt = runtime.filter(runtime.make_prog(prog),
+                                              cols_list, n_i)
+                surv = runtime.wrap_sel(sel_ptr, cnt)
+                if cnt:
+                    pid = runtime.gather(pi, surv.data, cnt)
+                    matched = np.empty(cnt, dtype=np.uint32)
+                    runtime._download(pid.data, matched)
+            elif s:
+                matched = np.empty(n_i, dtype=np.uint32)
+                runtime._download(pi.data, matched)
+        mask = np.ones(n_l, dtype=bool)
+        if len(matched):
+            mask[matched.astype(np.int64)] = False
+        anti = np.flatnonzero(mask).astype(np.uint32)
+        sel = runtime.upload_column(anti, dtype=rt.I32)
+        return (sel, sel), len(anti)
+
     def _reconcile_dict_keys(self, runtime, dc_lhs, dc_rhs, lhs_on, rhs_on):
         """String join keys: the reference merges on the STRINGS
         (join.py:241-246 dd.merge on object columns); our per-table
