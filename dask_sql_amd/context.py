@@ -233,11 +233,16 @@ class Context:
     def create_table_from_device(self, table_name: str, table: DeviceTable,
                                  sql_types: dict | None = None):
         """Register device-resident columns as a table (distributed
-        intermediates; no host round-trip)."""
+        intermediates; no host round-trip). Re-registering with an identical
+        schema keeps cached plans valid (plans are data-independent), so the
+        per-step re-registration in the distributed pipeline stays cheap."""
         t = DeviceRegisteredTable(table, sql_types)
-        self.tables[table_name.lower()] = t
-        self.catalog.add(table_name, t.fields())
-        self._schema_version += 1
+        key = table_name.lower()
+        old = self.tables.get(key)
+        self.tables[key] = t
+        if old is None or old.fields() != t.fields():
+            self.catalog.add(table_name, t.fields())
+            self._schema_version += 1
 
     def drop_table(self, table_name: str):
         self.tables.pop(table_name.lower(), None)
