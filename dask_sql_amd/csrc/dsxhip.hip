@@ -1478,35 +1478,40 @@ __global__ void k_gbpart_hist(ColsArg C, int64_t n, const KeyArg* Kp,
     hist[(int64_t)blockIdx.x * nb + i] = (int64_t)s_hist[i];
 }
 
-// one block per bucket: exclusive scan over the grid dimension; writes
-// per-(block,bucket) bases back in place and the bucket total to totals[b]
+// 8 buckets per block, transposed: lanes j=0..7 of one chunk read
+// hist[g][8b..8b+7] — a full 64-B line — instead of the one-bucket-per-block
+// version's 8-B strided walk (which the PMC counters showed fetching 143 MB
+// per launch, 8.5× the algorithmic bytes). Exclusive scan over the grid
+// dimension per bucket; bases written back in place, totals[b] per bucket.
 __global__ void k_gbpart_scan(int64_t* hist, int grid, int nb,
                               int64_t* totals) {
-  int b = blockIdx.x;
-  __shared__ int64_t s_sum[BLOCK];
-  int64_t per = (grid + BLOCK - 1) / BLOCK;
+  const int J = 8;            // buckets per block (one 64-B line)
+  const int C = BLOCK / J;    // grid-chunks per bucket
+  int j = threadIdx.x % J;
+  int cidx = threadIdx.x / J;
+  int b = blockIdx.x * J + j;
+  __shared__ int64_t s_sum[BLOCK];  // [chunk][bucket-lane]
+  int64_t per = ((int64_t)grid + C - 1) / C;
+  int64_t g0 = cidx * per;
+  int64_t g1 = min((int64_t)grid, g0 + per);
   int64_t local = 0;
-  for (int64_t g = threadIdx.x * per; g < min((int64_t)grid,
-                                              (threadIdx.x + 1) * per); g++)
-    local += hist[g * nb + b];
-  s_sum[threadIdx.x] = local;
+  for (int64_t g = g0; g < g1; g++) local += hist[g * nb + b];
+  s_sum[cidx * J + j] = local;
   __syncthreads();
-  // block scan (Hillis-Steele) over the 256 partials
-  for (int d = 1; d < BLOCK; d <<= 1) {
-    int64_t v = (threadIdx.x >= d) ? s_sum[threadIdx.x - d] : 0;
+  // Hillis-Steele over the chunk dimension, per bucket lane
+  for (int d = 1; d < C; d <<= 1) {
+    int64_t v = (cidx >= d) ? s_sum[(cidx - d) * J + j] : 0;
     __syncthreads();
-    s_sum[threadIdx.x] += v;
+    s_sum[cidx * J + j] += v;
     __syncthreads();
   }
-  int64_t excl = s_sum[threadIdx.x] - local;
-  int64_t run = excl;
-  for (int64_t g = threadIdx.x * per; g < min((int64_t)grid,
-                                              (threadIdx.x + 1) * per); g++) {
+  int64_t run = s_sum[cidx * J + j] - local;
+  for (int64_t g = g0; g < g1; g++) {
     int64_t v = hist[g * nb + b];
     hist[g * nb + b] = run;
     run += v;
   }
-  if (threadIdx.x == BLOCK - 1) totals[b] = s_sum[BLOCK - 1];
+  if (cidx == C - 1) totals[b] = s_sum[(C - 1) * J + j];
 }
 
 // 1 block: exclusive scan of bucket totals → absolute bases. Parallel
@@ -1954,7 +1959,7 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
       hipModuleLaunchKernel(f_hist, grid, 1, 1, BLOCK, 1, 1,
                             (unsigned)(nb * 4), c->stream, args, nullptr);
     }
-    hipLaunchKernelGGL(k_gbpart_scan, dim3(nb), dim3(BLOCK), 0, c->stream,
+    hipLaunchKernelGGL(k_gbpart_scan, dim3(nb / 8), dim3(BLOCK), 0, c->stream,
                        d_hist, grid, nb, d_totals);
     hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(1024), 0, c->stream,
                        d_totals, nb, d_bases);
@@ -1963,6 +1968,14 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
       static const int sthreads = [] {
         const char* e = getenv("DSX_SCATTER_THREADS");
         return e ? atoi(e) : 1024;  // measured best at C2 (1.24→1.04 ms)
+      }();
+      // occupancy throttle: padding the dynamic LDS ask caps resident
+      // blocks per CU, shrinking the (resident blocks × nb × 64 B) set of
+      // open write lines toward the 4 MB/XCD L2 — knob for measuring the
+      // scattered-write eviction amplification
+      static const unsigned lds_pad = [] {
+        const char* e = getenv("DSX_SCATTER_LDS_PAD");
+        return e ? (unsigned)atoi(e) : 0u;
       }();
       // NB: grid MUST match the hist pass (per-block bucket bases are keyed
       // by blockIdx); only the thread count may vary.
@@ -1974,7 +1987,8 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
         void* args[] = {&a2.C, &a2.n, &a2.nb, &a2.hist, &a2.bases, &a2.oc,
                         &a2.ov};
         hipModuleLaunchKernel(f_scat, grid, 1, 1, sthreads, 1, 1,
-                              (unsigned)(nb * 8), c->stream, args, nullptr);
+                              (unsigned)(nb * 8) + lds_pad, c->stream, args,
+                              nullptr);
       } else {
         struct {
           ColsArg C; int64_t n; int nb; const int64_t* hist;
@@ -1982,7 +1996,8 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
         } a2{C, n, nb, d_hist, d_bases, d_recs};
         void* args[] = {&a2.C, &a2.n, &a2.nb, &a2.hist, &a2.bases, &a2.out};
         hipModuleLaunchKernel(f_scat, grid, 1, 1, sthreads, 1, 1,
-                              (unsigned)(nb * 8), c->stream, args, nullptr);
+                              (unsigned)(nb * 8) + lds_pad, c->stream, args,
+                              nullptr);
       }
     }
   } else {
@@ -1991,7 +2006,7 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
       hipLaunchKernelGGL(k_gbpart_hist, dim3(grid), dim3(BLOCK),
                          (size_t)nb * 4, c->stream, C, n, d_K, P, nb, d_hist);
     }
-    hipLaunchKernelGGL(k_gbpart_scan, dim3(nb), dim3(BLOCK), 0, c->stream,
+    hipLaunchKernelGGL(k_gbpart_scan, dim3(nb / 8), dim3(BLOCK), 0, c->stream,
                        d_hist, grid, nb, d_totals);
     hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(1024), 0, c->stream,
                        d_totals, nb, d_bases);
