@@ -1,0 +1,84 @@
+#!/usr/bin/env python3
+"""Collect HBM traffic for a bench workload on the GPU box (run via gpurun).
+
+Two separate rocprofv3 --pmc passes (FETCH_SIZE, WRITE_SIZE — per gpurun
+policy never combined with trace domains), per-dispatch CSVs aggregated to
+mean KB per kernel, then the groupby-family bytes per step with the gfx950
+FETCH_SIZE x2 correction (MI355X_MICROARCH.md §HBM: the counter reports half
+the bytes of wide coalesced reads).
+
+Writes gpurun_out/traffic_<workload>.json and
+gpurun_out/<tag>_pmc_per_dispatch.json — copy the judged ones to profiles/.
+"""
+import csv
+import glob
+import json
+import os
+import subprocess
+import sys
+from collections import defaultdict
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent.parent
+WORKLOAD = sys.argv[1] if len(sys.argv) > 1 else "c2_filter_groupby_100m"
+STEPS = int(sys.argv[2]) if len(sys.argv) > 2 else 4
+TAG = sys.argv[3] if len(sys.argv) > 3 else "c2"
+
+# kernel family whose traffic makes up one step of the hot path
+FAMILY = {
+    "c2": ["j_hist", "j_scatter", "j_aggregate", "k_gbpart_scan",
+           "k_gbpart_bases", "k_gbpart_finalize", "k_gbpart_hist",
+           "k_gbpart_scatter", "k_gbpart_aggregate"],
+}.get(TAG.split("_")[0], [])
+
+os.environ.setdefault("TMPDIR", "/tmp")
+out_dir = REPO / "gpurun_out"
+out_dir.mkdir(exist_ok=True)
+
+
+def one_pass(counter):
+    d = f"/tmp/pmc_{counter}"
+    subprocess.run(
+        ["rocprofv3", "--pmc", counter, "-d", d, "--output-format", "csv",
+         "--", "python", str(REPO / "bench.py"), "--workload", WORKLOAD,
+         "--steps", str(STEPS), "--warmup", "2", "--no-cpu-baseline"],
+        cwd="/tmp", check=True, stdout=subprocess.DEVNULL,
+        stderr=subprocess.DEVNULL, timeout=280)
+    acc, cnt = defaultdict(float), defaultdict(int)
+    for f in glob.glob(f"{d}/**/*counter_collection.csv", recursive=True):
+        with open(f) as fh:
+            for row in csv.DictReader(fh):
+                if row.get("Counter_Name") != counter:
+                    continue
+                name = row["Kernel_Name"].split("(")[0].split(".")[0]
+                acc[name] += float(row["Counter_Value"])
+                cnt[name] += 1
+    return {k: acc[k] / cnt[k] for k in acc}  # mean KB per dispatch
+
+
+fetch = one_pass("FETCH_SIZE")
+write = one_pass("WRITE_SIZE")
+per_dispatch = {k: {"fetch_kb": fetch.get(k, 0.0), "write_kb": write.get(k)}
+                for k in sorted(set(fetch) | set(write))}
+(out_dir / f"{TAG}_pmc_per_dispatch.json").write_text(
+    json.dumps(per_dispatch, indent=1))
+
+if FAMILY:
+    bytes_per_launch = 0.0
+    rf = rw = 0.0
+    for k in FAMILY:
+        f_kb = fetch.get(k, 0.0) or 0.0
+        w_kb = write.get(k, 0.0) or 0.0
+        rf += f_kb * 1024
+        rw += w_kb * 1024
+    bytes_per_launch = 2 * rf + rw  # gfx950 FETCH x2 correction
+    (out_dir / f"traffic_{WORKLOAD}.json").write_text(json.dumps({
+        "bytes_per_launch": bytes_per_launch,
+        "method": "rocprofv3 --pmc FETCH_SIZE / WRITE_SIZE separate passes; "
+                  "FETCH doubled per MI355X_MICROARCH.md §HBM gfx950 "
+                  "correction; summed over the groupby family per step",
+        "raw_fetch_bytes": rf,
+        "raw_write_bytes": rw,
+    }, indent=1))
+    print("traffic bytes/step:", round(bytes_per_launch / 1e9, 3), "GB")
+print("kernels:", len(per_dispatch))
