@@ -764,6 +764,82 @@ extern "C" int dsx_gather(DsxCtx* c, const DsxColumn* col, const uint32_t* sel,
   return dbg_check(c, "dsx_gather");
 }
 
+extern "C" int dsx_memset(DsxCtx* c, void* dev, int value, int64_t bytes) {
+  HIP_TRY(hipMemsetAsync(dev, value, (size_t)bytes, c->stream));
+  return 0;
+}
+
+// row scatter — inverse of gather: out[sel[i]] = in[i]. Places join-back /
+// window columns into original row order (sel must be a permutation or a
+// subset of [0, n_out); untouched out rows keep their init value).
+template <typename T>
+__global__ void k_scatter_rows(const T* in, const uint8_t* in_valid,
+                               const uint32_t* sel, int64_t n, T* out,
+                               uint8_t* out_valid, int64_t n_out,
+                               unsigned int* dbg) {
+  int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (; i < n; i += stride) {
+    uint32_t s = sel[i];
+    if ((int64_t)s >= n_out) {
+      if (atomicOr(dbg, 1u) == 0) {
+        dbg[1] = s;
+        dbg[2] = (unsigned int)n_out;
+        dbg[3] = (unsigned int)i;
+      }
+      continue;
+    }
+    out[s] = in[i];
+    if (out_valid) out_valid[s] = in_valid ? in_valid[i] : 1;
+  }
+}
+
+extern "C" int dsx_scatter_rows(DsxCtx* c, const DsxColumn* col,
+                                const uint32_t* sel, int64_t n_sel,
+                                int64_t n_out, void* out_data,
+                                uint8_t* out_validity) {
+  int grid = (int)min((int64_t)MAX_GRID, (n_sel + BLOCK - 1) / BLOCK);
+  if (grid == 0) return 0;
+  ProfScope ps(c, "k_scatter_rows");
+  switch (col->dtype) {
+    case DSX_I64:
+      hipLaunchKernelGGL(k_scatter_rows<int64_t>, dim3(grid), dim3(BLOCK), 0,
+                         c->stream, (const int64_t*)col->data, col->validity,
+                         sel, n_sel, (int64_t*)out_data, out_validity, n_out,
+                         c->dbg_flag);
+      break;
+    case DSX_F64:
+      hipLaunchKernelGGL(k_scatter_rows<double>, dim3(grid), dim3(BLOCK), 0,
+                         c->stream, (const double*)col->data, col->validity,
+                         sel, n_sel, (double*)out_data, out_validity, n_out,
+                         c->dbg_flag);
+      break;
+    case DSX_I32:
+      hipLaunchKernelGGL(k_scatter_rows<int32_t>, dim3(grid), dim3(BLOCK), 0,
+                         c->stream, (const int32_t*)col->data, col->validity,
+                         sel, n_sel, (int32_t*)out_data, out_validity, n_out,
+                         c->dbg_flag);
+      break;
+    case DSX_F32:
+      hipLaunchKernelGGL(k_scatter_rows<float>, dim3(grid), dim3(BLOCK), 0,
+                         c->stream, (const float*)col->data, col->validity,
+                         sel, n_sel, (float*)out_data, out_validity, n_out,
+                         c->dbg_flag);
+      break;
+    case DSX_I8:
+    case DSX_BOOL8:
+      hipLaunchKernelGGL(k_scatter_rows<int8_t>, dim3(grid), dim3(BLOCK), 0,
+                         c->stream, (const int8_t*)col->data, col->validity,
+                         sel, n_sel, (int8_t*)out_data, out_validity, n_out,
+                         c->dbg_flag);
+      break;
+    default:
+      FAIL(-3, "scatter_rows: bad dtype %d", col->dtype);
+  }
+  HIP_TRY(hipGetLastError());
+  return dbg_check(c, "dsx_scatter_rows");
+}
+
 // ---------------------------------------------------------------------------
 // dsx_minmax_i64
 // ---------------------------------------------------------------------------

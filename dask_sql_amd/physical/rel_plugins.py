@@ -1328,8 +1328,211 @@ def _topk(pdf, keys, k):
     return cand.iloc[:k]
 
 
+class DaskWindowPlugin(BaseRelPlugin):
+    """reference rel/logical/window.py:212-428 (OverOperation dispatch +
+    map_on_each_group over sorted partitions). Two paths:
+
+    - unordered partition aggregates (SUM/COUNT/AVG/MIN/MAX OVER
+      (PARTITION BY …)): all-device — fused groupby, hash join-back of the
+      per-group value, and a row scatter into original order;
+    - ranking and ordered (running, RANGE-peers default frame) aggregates:
+      the REFERENCE computes these in pandas per partition
+      (window.py:266-427 map_on_each_group); we restate the same pandas
+      computation on host over the key columns, then upload the one result
+      column. Parity-first; a device sort is a round-2+ widening."""
+
+    class_name = "Window"
+
+    def convert(self, rel, context):
+        runtime = context._get_runtime()
+        (dc,) = self.assert_inputs(rel, 1, context)
+        cols = dc.backend_cols()
+        cc_in = dc.column_container
+        n = dc.table.num_rows
+        out_cols = dict(dc.table.columns)
+        order_names = [(f, cc_in.get_backend_by_frontend_name(f))
+                       for f in cc_in.columns]
+        for spec in rel.window().getWindowSpecs():
+            col = self._one(runtime, cols, n, spec)
+            bname = f"win__{spec.out_name}"
+            out_cols[bname] = col
+            order_names.append((spec.out_name, bname))
+        cc = ColumnContainer([nm for nm, _ in order_names],
+                             dict(order_names))
+        cc = self.fix_column_to_row_type(cc, rel.getRowType())
+        return DataContainer(DeviceTable(out_cols, num_rows=n), cc)
+
+    _AGGS = {"sum", "count", "avg", "min", "max"}
+
+    def _one(self, runtime, cols, n, spec):
+        device_ok = (spec.func in self._AGGS and not spec.order_idx
+                     and all(cols[i].dtype in _INT_KINDS
+                             for i in spec.part_idx))
+        if device_ok:
+            return self._device_agg(runtime, cols, n, spec)
+        return self._host_ordered(runtime, cols, n, spec)
+
+    def _device_agg(self, runtime, cols, n, spec):
+        """groupby → per-group value → hash join-back → row scatter."""
+        keyspecs = []
+        for gi in spec.part_idx:
+            mn, mx, nn = _minmax_cached(runtime, cols[gi])
+            if nn == 0:
+                mn, mx = 0, 0
+            keyspecs.append((gi, mn, mx - mn + 1, bool(cols[gi].validity)))
+        arg = InputRef(spec.arg_idx, None) if spec.arg_idx is not None \
+            else None
+        speclist, fin = DaskAggregatePlugin()._agg_spec_expr(
+            spec.func, arg, cols, _dicts_of(cols))
+        specs = [(op, runtime.make_prog(p)) for op, p in speclist]
+        oc, ov, on, G = runtime.hash_groupby(cols, n, keyspecs, None, specs)
+
+        class _H:
+            def __init__(s, ptrs):
+                s.ptrs = ptrs
+
+            def __del__(s):
+                for p in s.ptrs:
+                    try:
+                        runtime._free(p)
+                    except Exception:
+                        pass
+
+        h = _H([oc, ov, on])
+        from dask_sql_amd.physical.rex import (OP_DIV_F64, OP_GT_I64,
+                                               OP_I64_TO_F64, OP_LIT_I64,
+                                               OP_LIT_NULL, OP_SELECT)
+        val_col = rt.DeviceColumn(runtime, ov, None, G,
+                                  rt.F64 if fin in ("avg", "sum_f", "min_f",
+                                                    "max_f") else rt.I64,
+                                  owner=False, keep_alive=h)
+        cnt_col = rt.DeviceColumn(runtime, on, None, G, rt.I64, owner=False,
+                                  keep_alive=h)
+        if fin == "count":
+            gval = cnt_col
+        elif fin == "avg":
+            prog = [(OP_COL, 1, 0), (OP_LIT_I64, 0, 0), (OP_GT_I64, 0, 0),
+                    (OP_COL, 0, 0), (OP_COL, 1, 0), (OP_I64_TO_F64, 0, 0),
+                    (OP_DIV_F64, 0, 0), (OP_LIT_NULL, 0, 0),
+                    (OP_SELECT, 0, 0)]
+            gval = runtime.eval(runtime.make_prog(prog), [val_col, cnt_col],
+                                G, rt.F64)
+        else:
+            prog = [(OP_COL, 1, 0), (OP_LIT_I64, 0, 0), (OP_GT_I64, 0, 0),
+                    (OP_COL, 0, 0), (OP_LIT_NULL, 0, 0), (OP_SELECT, 0, 0)]
+            gval = runtime.eval(runtime.make_prog(prog), [val_col, cnt_col],
+                                G, val_col.dtype)
+        # join-back: every row's partition code is in the group table
+        bcodes = rt.DeviceColumn(runtime, oc, None, G, rt.I64, owner=False,
+                                 keep_alive=h)
+        pcodes, space = runtime.keypack(cols, keyspecs, n)
+        table = runtime.hash_build(bcodes, None, code_max=space - 1)
+        try:
+            p_ptr, b_ptr, count = runtime.hash_probe(table, pcodes,
+                                                     rt.JOIN_INNER, None)
+            p_sel = runtime.wrap_sel(p_ptr, count)
+            b_sel = runtime.wrap_sel(b_ptr, count)
+            assert count == n, (count, n)
+            pair_val = runtime.gather(gval, b_sel.data, count,
+                                      bool(gval.validity))
+            out = runtime.scatter_rows(pair_val, p_sel.data, count, n,
+                                       with_validity=bool(pair_val.validity))
+        finally:
+            runtime.hash_table_free(table)
+        return out
+
+    def _host_ordered(self, runtime, cols, n, spec):
+        import pandas as pd
+
+        def np_col(i):
+            arr, valid = cols[i].to_numpy()
+            if valid is not None and not valid.all():
+                arr = arr.astype(np.float64)
+                arr[~valid] = np.nan
+            return arr
+
+        df = pd.DataFrame({f"p{j}": np_col(i)
+                           for j, i in enumerate(spec.part_idx)})
+        pnames = list(df.columns)
+        onames = []
+        asc = []
+        for j, (i, desc) in enumerate(spec.order_idx):
+            df[f"o{j}"] = np_col(i)
+            onames.append(f"o{j}")
+            asc.append(not desc)
+        if spec.arg_idx is not None:
+            df["v"] = np_col(spec.arg_idx)
+        if not pnames:
+            df["p0"] = 0
+            pnames = ["p0"]
+        if onames:
+            df = df.sort_values(onames, ascending=asc, na_position="last",
+                                kind="mergesort")
+        df = df.sort_values(pnames, na_position="last", kind="mergesort")
+        grp = df.groupby(pnames, dropna=False, sort=False)
+        f = spec.func
+        if f == "row_number":
+            res = grp.cumcount() + 1
+        elif f in ("rank", "dense_rank"):
+            rn = grp.cumcount() + 1
+            df["_rn"] = rn
+            tie = df.groupby(pnames + onames, dropna=False, sort=False)
+            if f == "rank":
+                res = tie["_rn"].transform("first")
+            else:
+                tid = tie.ngroup()
+                df["_tid"] = tid
+                res = tid - grp["_tid"].transform("first") + 1
+        else:
+            # running aggregates, default RANGE UNBOUNDED..CURRENT frame:
+            # cumulative then broadcast the tie-group's last value (peers)
+            if f == "count" and spec.arg_idx is None:
+                cum = grp.cumcount() + 1
+            elif f == "count":
+                df["_nn"] = df["v"].notna().astype(np.int64)
+                cum = grp["_nn"].cumsum()
+            elif f == "sum":
+                cum = grp["v"].cumsum()
+                cum = cum.groupby(
+                    [df[c] for c in pnames], dropna=False).ffill()
+            elif f == "min":
+                cum = grp["v"].cummin()
+                cum = cum.groupby(
+                    [df[c] for c in pnames], dropna=False).ffill()
+            elif f == "max":
+                cum = grp["v"].cummax()
+                cum = cum.groupby(
+                    [df[c] for c in pnames], dropna=False).ffill()
+            elif f == "avg":
+                df["_nn"] = df["v"].notna().astype(np.int64)
+                s = grp["v"].cumsum().groupby(
+                    [df[c] for c in pnames], dropna=False).ffill()
+                c = grp["_nn"].cumsum()
+                cum = s / c.replace(0, np.nan)
+            else:
+                raise RexCompileError(f"window function {f}")
+            if onames:
+                df["_cum"] = cum
+                res = df.groupby(pnames + onames, dropna=False,
+                                 sort=False)["_cum"].transform("last")
+            else:
+                res = cum
+        vals = np.asarray(res, dtype=np.float64)
+        out = np.empty(n, dtype=np.float64)
+        out[df.index.to_numpy()] = vals
+        nanmask = np.isnan(out)
+        if not nanmask.any() and spec.func in ("row_number", "rank",
+                                               "dense_rank", "count") \
+                or (not nanmask.any()
+                    and spec.out_type.getSqlType() == "BIGINT"):
+            return runtime.upload_column(out.astype(np.int64))
+        return runtime.upload_column(
+            out, validity=(~nanmask).astype(np.uint8)
+            if nanmask.any() else None)
+
+
 def register_defaults():
     for cls in (DaskTableScanPlugin, DaskFilterPlugin, DaskProjectPlugin,
                 DaskJoinPlugin, DaskAggregatePlugin, DaskSortPlugin,
-                DaskLimitPlugin):
+                DaskLimitPlugin, DaskWindowPlugin):
         RelConverter.add_plugin_class(cls, replace=False)

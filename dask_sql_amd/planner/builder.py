@@ -325,6 +325,37 @@ class Builder:
             else:
                 items.append((e, alias))
 
+        # 2.5 window functions (reference rel/logical/window.py): compute
+        # window columns over the joined/filtered input, rewrite select items
+        win_asts = []
+
+        def collect_windows(ast):
+            if not isinstance(ast, tuple):
+                return
+            if ast[0] == "window":
+                if ast not in win_asts:
+                    win_asts.append(ast)
+                return
+            if ast[0] == "call":
+                for a in ast[2]:
+                    collect_windows(a)
+            elif ast[0] == "cast":
+                collect_windows(ast[1])
+            elif ast[0] == "case":
+                for cnd, v in ast[1]:
+                    collect_windows(cnd)
+                    collect_windows(v)
+                if ast[2] is not None:
+                    collect_windows(ast[2])
+
+        for e, _ in items:
+            collect_windows(e)
+        if win_asts:
+            if stmt.group_by or any(self._has_agg(e) for e, _ in items):
+                raise NotImplementedError(
+                    "window functions combined with GROUP BY/aggregates")
+            plan, items = self._build_window(plan, items, win_asts)
+
         # 3. aggregate?
         has_agg = any(self._has_agg(e) for e, _ in items) or bool(
             stmt.group_by
@@ -360,6 +391,87 @@ class Builder:
             plan = LogicalPlan("Limit", [plan], plan.getRowType(),
                                LimitNode(stmt.limit, stmt.offset))
         return plan
+
+    # ------------------------------------------------------------ window
+    def _build_window(self, plan, items, win_asts):
+        """Insert Projection(inputs + window arg/key exprs) → Window(specs)
+        and rewrite select items to reference the window output columns
+        (reference rel/logical/window.py:212-428)."""
+        from dask_sql_amd.planner.plan import (ProjectionNode, WindowNode,
+                                               WindowSpec)
+        in_fields = plan.getRowType().getFieldList()
+        pre_named = [(InputRef(i, f.getType()), f.getName())
+                     for i, f in enumerate(in_fields)]
+        pre_fields = [Field(f.getName(), f.getType(), qualifier=f.qualifier)
+                      for f in in_fields]
+        key_of = {}
+
+        def idx_of(ast):
+            key = repr(ast)
+            if key in key_of:
+                return key_of[key]
+            e = self._resolve(ast, plan)
+            if isinstance(e, InputRef):
+                key_of[key] = e.getIndex()
+                return e.getIndex()
+            name = f"w_in{len(pre_named)}"
+            pre_named.append((e, name))
+            pre_fields.append(Field(name, SqlType(_expr_type(e))))
+            key_of[key] = len(pre_named) - 1
+            return key_of[key]
+
+        specs = []
+        win_out = {}
+        for ast in win_asts:
+            _, func, args, part, order = ast
+            arg_idx = None
+            arg_t = None
+            if args and args[0] != ("star",):
+                arg_idx = idx_of(args[0])
+                arg_t = pre_fields[arg_idx].getType().getSqlType()
+            part_idx = [idx_of(p) for p in part]
+            order_idx = [(idx_of(o), desc) for o, desc in order]
+            if func in ("row_number", "rank", "dense_rank") and not order_idx:
+                raise ValueError(f"{func.upper()} requires ORDER BY in OVER")
+            if func in ("row_number", "rank", "dense_rank", "count"):
+                ty = "BIGINT"
+            elif func == "avg":
+                ty = "DOUBLE"
+            elif func == "sum":
+                ty = "DOUBLE" if _is_float(arg_t or "BIGINT") else "BIGINT"
+            else:  # min/max keep the arg type
+                ty = arg_t or "BIGINT"
+            name = f"w{len(specs)}__{func}"
+            specs.append(WindowSpec(func, arg_idx, part_idx, order_idx,
+                                    name, SqlType(ty)))
+            win_out[repr(ast)] = name
+        if len(pre_named) > len(in_fields):
+            plan = LogicalPlan("Projection", [plan], RelDataType(pre_fields),
+                               ProjectionNode(pre_named))
+        wfields = pre_fields + [Field(s.out_name, s.out_type) for s in specs]
+        plan = LogicalPlan("Window", [plan], RelDataType(wfields),
+                           WindowNode(specs))
+
+        def rewrite(ast):
+            if isinstance(ast, tuple):
+                if ast[0] == "window":
+                    return ("col", None, win_out[repr(ast)])
+                if ast[0] == "call":
+                    return ("call", ast[1], [rewrite(a) for a in ast[2]])
+                if ast[0] == "cast":
+                    return ("cast", rewrite(ast[1]), ast[2])
+                if ast[0] == "case":
+                    return ("case",
+                            [(rewrite(c), rewrite(v)) for c, v in ast[1]],
+                            rewrite(ast[2]) if ast[2] is not None else None)
+            return ast
+
+        items2 = []
+        for e, alias in items:
+            if alias is None and isinstance(e, tuple) and e[0] == "window":
+                alias = f"{e[1].upper()}() OVER"
+            items2.append((rewrite(e), alias))
+        return plan, items2
 
     # ------------------------------------------------------------ aggregate
     def _build_aggregate(self, stmt, plan, items):

@@ -46,8 +46,11 @@ KEYWORDS = {
     "RIGHT", "FULL", "OUTER", "SEMI", "ANTI", "CROSS", "ON", "TRUE", "FALSE",
     "NULL", "IS", "IN", "BETWEEN", "LIKE", "CASE", "WHEN", "THEN", "ELSE",
     "END", "CAST", "DATE", "ASC", "DESC", "NULLS", "FIRST", "LAST", "FILTER",
-    "TIMESTAMP", "INTERVAL", "UNION", "ALL",
+    "TIMESTAMP", "INTERVAL", "UNION", "ALL", "OVER", "PARTITION",
 }
+
+WINDOW_FUNCS = {"ROW_NUMBER", "RANK", "DENSE_RANK", "SUM", "COUNT", "AVG",
+                "MIN", "MAX"}
 
 _TOKEN_RE = re.compile(
     r"""
@@ -446,6 +449,29 @@ class Parser:
             self.expect_kw("WHERE")
             filter_expr = self.expr()
             self.expect_op(")")
+        if self.peek() == ("kw", "OVER"):
+            if fname not in WINDOW_FUNCS:
+                raise ValueError(f"{fname} is not a supported window function")
+            self.next()
+            self.expect_op("(")
+            part, order = [], []
+            if self.accept_kw("PARTITION"):
+                self.expect_kw("BY")
+                part.append(self.expr())
+                while self.accept_op(","):
+                    part.append(self.expr())
+            if self.accept_kw("ORDER"):
+                self.expect_kw("BY")
+                while True:
+                    e = self.expr()
+                    desc = bool(self.accept_kw("DESC"))
+                    if not desc:
+                        self.accept_kw("ASC")
+                    order.append((e, desc))
+                    if not self.accept_op(","):
+                        break
+            self.expect_op(")")
+            return ("window", fname.lower(), args, tuple(part), tuple(order))
         if fname in AGG_FUNCS:
             return ("agg", fname.lower(), args, distinct, filter_expr)
         return ("call", fname, args)

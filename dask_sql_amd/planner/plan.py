@@ -210,6 +210,9 @@ class LogicalPlan:
     def sort(self):
         return self._get("Sort")
 
+    def window(self):
+        return self._get("Window")
+
     def limit(self):
         return self._get("Limit")
 
@@ -301,6 +304,29 @@ class AggregateNode:
 
     def getDistinctColumns(self):
         return list(self._distinct_columns)
+
+
+class WindowSpec:
+    """One window column (reference rel/logical/window.py:212-428 lowering).
+    arg/partition/order are input field indices; out column appended after
+    the input fields."""
+
+    def __init__(self, func: str, arg_idx, part_idx: list,
+                 order_idx: list, out_name: str, out_type):
+        self.func = func                  # row_number|rank|dense_rank|sum|...
+        self.arg_idx = arg_idx            # int | None (ranking / COUNT(*))
+        self.part_idx = list(part_idx)
+        self.order_idx = list(order_idx)  # [(field index, desc bool)]
+        self.out_name = out_name
+        self.out_type = out_type
+
+
+class WindowNode:
+    def __init__(self, specs: list):
+        self.specs = list(specs)
+
+    def getWindowSpecs(self):
+        return list(self.specs)
 
 
 class ProjectionNode:

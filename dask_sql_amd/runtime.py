@@ -330,6 +330,28 @@ class Runtime:
         return DeviceColumn(self, ptr, None, n, dtype, owner=False,
                             keep_alive=keep_alive)
 
+    def scatter_rows(self, col: DeviceColumn, sel_ptr, n_sel, n_out,
+                     with_validity=True) -> DeviceColumn:
+        """out[sel[i]] = col[i] (inverse of gather; window/join-back
+        placement). Rows not covered by sel come out NULL."""
+        out = self.empty_column(n_out, col.dtype, with_validity)
+        if out.validity:
+            _check(self.lib,
+                   self.lib.dsx_memset(self.ctx, ct.c_void_p(out.validity),
+                                       ct.c_int(0), ct.c_int64(n_out)),
+                   "dsx_memset")
+        _check(
+            self.lib,
+            self.lib.dsx_scatter_rows(self.ctx, ct.byref(col.c_struct()),
+                                      ct.c_void_p(sel_ptr),
+                                      ct.c_int64(n_sel), ct.c_int64(n_out),
+                                      ct.c_void_p(out.data),
+                                      ct.c_void_p(out.validity)
+                                      if out.validity else None),
+            "dsx_scatter_rows",
+        )
+        return out
+
     def gather(self, col: DeviceColumn, sel_ptr, n_sel,
                force_validity=False) -> DeviceColumn:
         need_valid = force_validity or bool(col.validity)

@@ -100,6 +100,7 @@ int dsx_free(DsxCtx* ctx, void* ptr);
  * pinned-host→HBM hipMemcpyAsync. */
 int dsx_upload(DsxCtx* ctx, const void* host, int64_t bytes, void** out_dev);
 int dsx_download(DsxCtx* ctx, const void* dev, void* host, int64_t bytes);
+int dsx_memset(DsxCtx* ctx, void* dev, int value, int64_t bytes);
 
 /* per-kernel HIP-event timing (for bench.py roofline accounting) */
 int dsx_prof_enable(DsxCtx* ctx, int enable);
@@ -131,6 +132,14 @@ int dsx_filter(DsxCtx* ctx, const DsxInstr* prog, int prog_len,
  * out caller-allocated. Gathers validity too when both non-NULL. */
 int dsx_gather(DsxCtx* ctx, const DsxColumn* col, const uint32_t* sel,
                int64_t n_sel, void* out_data, uint8_t* out_validity);
+
+/* inverse of dsx_gather: out[sel[i]] = col[i] — places join-back / window
+ * columns into original row order (reference window.py:212-428 assigns the
+ * computed window column back onto the frame's index). out caller-allocated
+ * and pre-initialized (untouched rows keep their init). */
+int dsx_scatter_rows(DsxCtx* ctx, const DsxColumn* col, const uint32_t* sel,
+                     int64_t n_sel, int64_t n_out, void* out_data,
+                     uint8_t* out_validity);
 
 /* min/max of an i64/i32/i8/date32 column ignoring NULLs (key-range probe for
  * packing; also MIN/MAX aggregate support). */

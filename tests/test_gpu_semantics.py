@@ -511,3 +511,78 @@ def test_left_anti_join_with_residual(ctx):
     got2 = c.sql("SELECT lhs.id FROM x1 lhs LEFT ANTI JOIN x2 rhs "
                  "ON lhs.id = rhs.id AND rhs.b > 1000").compute()
     assert sorted(got2["id"].astype(int).tolist()) == [1, 1, 2, 4, 5]
+
+
+def test_window_partition_aggregates_device(ctx):
+    """SUM/COUNT/AVG/MIN/MAX OVER (PARTITION BY k) — device path (groupby +
+    join-back + row scatter); reference window.py:212-428."""
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(21)
+    n = 30_000
+    df = pd.DataFrame({"k": rng.integers(0, 97, n).astype(np.int64),
+                       "v": np.round(rng.random(n) * 10, 3)})
+    df.loc[rng.random(n) < 0.1, "v"] = np.nan
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT k, v, SUM(v) OVER (PARTITION BY k) AS s, "
+                "COUNT(v) OVER (PARTITION BY k) AS cv, "
+                "COUNT(*) OVER (PARTITION BY k) AS ca, "
+                "AVG(v) OVER (PARTITION BY k) AS a, "
+                "MIN(v) OVER (PARTITION BY k) AS mn, "
+                "MAX(v) OVER (PARTITION BY k) AS mx FROM t").compute()
+    g = df.groupby("k")["v"]
+    exp = df.assign(s=g.transform("sum"), cv=g.transform("count"),
+                    ca=df.groupby("k")["k"].transform("size"),
+                    a=g.transform("mean"), mn=g.transform("min"),
+                    mx=g.transform("max"))
+    # row order preserved by the scatter
+    assert (got["k"].to_numpy() == exp["k"].to_numpy()).all()
+    for col in ("s", "a", "mn", "mx"):
+        np.testing.assert_allclose(got[col].to_numpy(dtype=np.float64),
+                                   exp[col].to_numpy(), rtol=1e-9,
+                                   err_msg=col)
+    assert (got["cv"].to_numpy(np.int64) == exp["cv"].to_numpy()).all()
+    assert (got["ca"].to_numpy(np.int64) == exp["ca"].to_numpy()).all()
+
+
+def test_window_ranking_and_running(ctx):
+    """ROW_NUMBER/RANK/DENSE_RANK + running SUM (RANGE-peers default frame)
+    — host path, restating the reference's per-partition pandas
+    (window.py:266-427)."""
+    from dask_sql_amd.context import Context
+    df = pd.DataFrame({
+        "k": [1, 1, 1, 1, 2, 2, 2],
+        "t": [10, 10, 20, 30, 5, 5, 5],
+        "v": [1.0, 2.0, 3.0, 4.0, 10.0, 20.0, 30.0],
+    })
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql(
+        "SELECT k, t, v, ROW_NUMBER() OVER (PARTITION BY k ORDER BY t) AS rn,"
+        " RANK() OVER (PARTITION BY k ORDER BY t) AS rk,"
+        " DENSE_RANK() OVER (PARTITION BY k ORDER BY t) AS dr,"
+        " SUM(v) OVER (PARTITION BY k ORDER BY t) AS rs FROM t").compute()
+    got = got.sort_values(["k", "t", "v"]).reset_index(drop=True)
+    # hand-computed SQL semantics (RANGE peers share the running value)
+    exp_rk = [1, 1, 3, 4, 1, 1, 1]
+    exp_dr = [1, 1, 2, 3, 1, 1, 1]
+    exp_rs = [3.0, 3.0, 6.0, 10.0, 60.0, 60.0, 60.0]
+    assert got["rk"].astype(int).tolist() == exp_rk
+    assert got["dr"].astype(int).tolist() == exp_dr
+    assert np.allclose(got["rs"], exp_rs)
+    # row_number within ties is arbitrary but must be a permutation 1..n
+    for k, grp in got.groupby("k"):
+        assert sorted(grp["rn"].astype(int).tolist()) == \
+            list(range(1, len(grp) + 1))
+
+
+def test_window_desc_order_and_global(ctx):
+    from dask_sql_amd.context import Context
+    df = pd.DataFrame({"t": [3, 1, 2], "v": [30.0, 10.0, 20.0]})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT t, ROW_NUMBER() OVER (ORDER BY t DESC) AS rn, "
+                "SUM(v) OVER (ORDER BY t) AS rs FROM t").compute()
+    got = got.sort_values("t").reset_index(drop=True)
+    assert got["rn"].astype(int).tolist() == [3, 2, 1]
+    assert np.allclose(got["rs"], [10.0, 30.0, 60.0])
