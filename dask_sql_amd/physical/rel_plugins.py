@@ -133,11 +133,26 @@ class DaskProjectPlugin(BaseRelPlugin):
         named = rel.projection().getNamedProjects()
         out_cols = {}
         new_names = []
+        from dask_sql_amd.physical.rex import dict_string_fn
         for i, (expr, name) in enumerate(named):
             backend_name = f"p{i}__{name}"
+            sfn = None if isinstance(expr, InputRef) \
+                else dict_string_fn(expr, dicts)
             if isinstance(expr, InputRef):
                 src = cols[expr.getIndex()]
                 out_cols[backend_name] = src  # zero-copy reuse
+            elif sfn is not None:
+                # string function over a dict column: same codes, the
+                # transform runs once over the dictionary
+                # (rex/core/call.py:1069-1135 string ops)
+                ci, f = sfn
+                src = cols[ci]
+                col = rt.DeviceColumn(runtime, src.data, src.validity,
+                                      src.len, src.dtype, owner=False,
+                                      keep_alive=src)
+                col.dictionary = [f(s) if s is not None else None
+                                  for s in src.dictionary]
+                out_cols[backend_name] = col
             else:
                 prog, kind = compile_expr(expr, cols, dicts)
                 out_dtype = rt.F64 if kind == KF else (

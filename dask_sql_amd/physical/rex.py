@@ -46,6 +46,46 @@ class RexCompileError(NotImplementedError):
     pass
 
 
+def dict_string_fn(expr, dicts):
+    """If expr is a chain of string functions (UPPER/LOWER/SUBSTRING) over a
+    dict-encoded column, return (col_index, python_fn) computing the string
+    result per dictionary entry — the whole transform runs once over the
+    (small, host-resident) dictionary, never per row (reference
+    rex/core/call.py:1069-1135 string operations). None if not such a
+    chain."""
+    if isinstance(expr, InputRef):
+        i = expr.getIndex()
+        if i < len(dicts) and dicts[i] is not None:
+            return i, (lambda s: s)
+        return None
+    if isinstance(expr, Call):
+        op = expr.getOperatorName().upper()
+        ops_ = expr.getOperands()
+        sub = dict_string_fn(ops_[0], dicts) if ops_ else None
+        if sub is None:
+            return None
+        i, f = sub
+        if op in ("UPPER", "LOWER") and len(ops_) == 1:
+            if op == "UPPER":
+                return i, (lambda s, f=f: f(s).upper())
+            return i, (lambda s, f=f: f(s).lower())
+        if op in ("SUBSTRING", "SUBSTR") and len(ops_) in (2, 3):
+            if not all(isinstance(o, Literal) for o in ops_[1:]):
+                return None
+            a = int(ops_[1].getValue())
+            ln = int(ops_[2].getValue()) if len(ops_) == 3 else None
+            if a < 1 or (ln is not None and ln < 0):
+                return None  # Calcite negative-start edge: not covered
+            start = a - 1
+
+            def g(s, f=f, start=start, ln=ln):
+                t = f(s)
+                return t[start:start + ln] if ln is not None else t[start:]
+
+            return i, g
+    return None
+
+
 def _like_regex(pattern: str):
     """SQL LIKE pattern → compiled regex: % = any run, _ = any single char,
     everything else literal (reference rex/core/call.py LIKE lowering)."""
@@ -184,27 +224,32 @@ class RexCompiler:
         once at compile time; the kernel-side predicate is an OR-chain of
         integer code equalities, so NULL → NULL falls out of EQ validity."""
         col, pat = ops
-        if not (isinstance(col, InputRef) and isinstance(pat, Literal)
-                and isinstance(pat.getValue(), str)):
-            raise RexCompileError("LIKE needs <column> LIKE '<pattern>'")
-        d = self.dicts[col.getIndex()]
-        if d is None:
-            raise RexCompileError("LIKE on non-dict-encoded column")
+        fn = dict_string_fn(col, self.dicts)
+        if fn is None or not (isinstance(pat, Literal)
+                              and isinstance(pat.getValue(), str)):
+            raise RexCompileError(
+                "LIKE needs <dict string expr> LIKE '<pattern>'")
+        ci, f = fn
+        d = self.dicts[ci]
         rx = _like_regex(pat.getValue())
         matched = [i for i, s in enumerate(d)
-                   if s is not None and rx.fullmatch(s)]
-        ci = col.getIndex()
-        if len(matched) > 10:
+                   if s is not None and rx.fullmatch(f(s))]
+        return self._emit_code_in(ci, matched)
+
+    def _emit_code_in(self, ci, codes):
+        """predicate: column's dict code ∈ codes (OR-chain of EQ; NULL→NULL
+        via EQ validity). Empty set → always-FALSE via a never-present
+        code."""
+        if len(codes) > 10:
             raise RexCompileError(
-                f"LIKE matches {len(matched)} dictionary entries "
+                f"predicate matches {len(codes)} dictionary entries "
                 "(> VM program budget)")
-        if not matched:
-            # FALSE for every present code, NULL stays NULL (-2 never a code)
+        if not codes:
             self._emit(OP_COL, ci)
             self._emit(OP_LIT_I64, 0, -2)
             self._emit(OP_EQ_I64)
             return KB
-        for j, code in enumerate(matched):
+        for j, code in enumerate(codes):
             self._emit(OP_COL, ci)
             self._emit(OP_LIT_I64, 0, code)
             self._emit(OP_EQ_I64)
@@ -235,25 +280,33 @@ class RexCompiler:
 
     def _compile_cmp(self, op, ops) -> str:
         a, b = ops
-        # dict-encoded string compare: col vs string literal
-        lit, col = None, None
-        if isinstance(a, InputRef) and isinstance(b, Literal) \
-                and isinstance(b.getValue(), str):
-            col, lit = a, b
-        elif isinstance(b, InputRef) and isinstance(a, Literal) \
-                and isinstance(a.getValue(), str):
-            col, lit = b, a
+        # dict-encoded string compare: string expr chain vs string literal
+        lit, sexpr = None, None
+        if isinstance(b, Literal) and isinstance(b.getValue(), str):
+            sexpr, lit = a, b
+        elif isinstance(a, Literal) and isinstance(a.getValue(), str):
+            sexpr, lit = b, a
             op = {"<": ">", ">": "<", "<=": ">=", ">=": "<="}.get(op, op)
-        if col is not None:
+        if sexpr is not None:
+            fn = dict_string_fn(sexpr, self.dicts)
+            if fn is None:
+                raise RexCompileError("string compare on non-dict column")
             if op not in ("=", "<>"):
                 raise RexCompileError("only =/<> on dict-encoded strings")
-            d = self.dicts[col.getIndex()]
-            if d is None:
-                raise RexCompileError("string compare on non-dict column")
-            code = d.index(lit.getValue()) if lit.getValue() in d else -2
-            self._emit(OP_COL, col.getIndex())
-            self._emit(OP_LIT_I64, 0, code)
-            self._emit(_CMP[op][0])
+            ci, f = fn
+            d = self.dicts[ci]
+            if isinstance(sexpr, InputRef):
+                # plain column: single-code compare keeps <> 3VL cheap
+                code = d.index(lit.getValue()) if lit.getValue() in d else -2
+                self._emit(OP_COL, ci)
+                self._emit(OP_LIT_I64, 0, code)
+                self._emit(_CMP[op][0])
+                return KB
+            matched = [c for c, s in enumerate(d)
+                       if s is not None and f(s) == lit.getValue()]
+            self._emit_code_in(ci, matched)
+            if op == "<>":
+                self._emit(OP_NOT)
             return KB
         ka = self._peek_kind(a)
         kb = self._peek_kind(b)

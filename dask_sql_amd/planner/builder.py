@@ -137,6 +137,8 @@ class Builder:
                 ty = "BOOLEAN"
             elif op == "NEG":
                 ty = _expr_type(ops[0])
+            elif op in ("UPPER", "LOWER", "SUBSTRING", "SUBSTR"):
+                ty = "VARCHAR"
             elif op == "/":
                 ty = _common_type(_expr_type(ops[0]), _expr_type(ops[1]))
             else:

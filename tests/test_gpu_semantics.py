@@ -586,3 +586,28 @@ def test_window_desc_order_and_global(ctx):
     got = got.sort_values("t").reset_index(drop=True)
     assert got["rn"].astype(int).tolist() == [3, 2, 1]
     assert np.allclose(got["rs"], [10.0, 30.0, 60.0])
+
+
+def test_dict_string_functions(ctx):
+    """UPPER/LOWER/SUBSTRING on dict-encoded strings: projection = dictionary
+    transform (same codes); predicates = compile-time code sets
+    (rex/core/call.py:1069-1135)."""
+    from dask_sql_amd.context import Context
+    df = pd.DataFrame({
+        "s": ["Apple", "banana", "Cherry", None, "apple"],
+        "v": [1, 2, 3, 4, 5],
+    })
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT UPPER(s) AS u, LOWER(s) AS l, "
+                "SUBSTRING(s, 2, 3) AS m, v FROM t").compute()
+    assert got["u"].tolist() == ["APPLE", "BANANA", "CHERRY", None, "APPLE"]
+    assert got["l"].tolist() == ["apple", "banana", "cherry", None, "apple"]
+    assert got["m"].tolist() == ["ppl", "ana", "her", None, "ppl"]
+    got2 = c.sql("SELECT v FROM t WHERE UPPER(s) = 'APPLE'").compute()
+    assert sorted(got2["v"].astype(int).tolist()) == [1, 5]
+    got3 = c.sql("SELECT v FROM t WHERE LOWER(s) <> 'apple'").compute()
+    # NULL <> ... -> NULL -> filtered (3VL)
+    assert sorted(got3["v"].astype(int).tolist()) == [2, 3]
+    got4 = c.sql("SELECT v FROM t WHERE UPPER(s) LIKE 'A%'").compute()
+    assert sorted(got4["v"].astype(int).tolist()) == [1, 5]
