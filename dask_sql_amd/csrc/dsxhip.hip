@@ -439,6 +439,11 @@ __device__ __forceinline__ bool vm_eval(const DsxInstr* prog, int len, const Col
         res.i = (int64_t)a.f;  // trunc, mappings.py:346-353
         k.set(sp - 1, res, av);
         break;
+      case DSX_OP_BITS_F64:
+        UN();
+        res.f = __longlong_as_double(a.i);
+        k.set(sp - 1, res, av);
+        break;
       case DSX_OP_NEG_F64:
         UN();
         res.f = -a.f;
@@ -924,10 +929,18 @@ __device__ __forceinline__ uint64_t pack_key(const KeyArg& K, const ColsArg& C,
     bool valid;
     vm_load_col(C, K.k[j].col, r, v, valid);
     uint64_t part;
-    if (K.k[j].nullable)
+    if (K.k[j].mode == 1) {
+      // f64 bit-pattern key (single key): exact equality incl. -0.0==0.0;
+      // NaN and NULL share code 0 (pandas NaN key group, dropna=False)
+      double x = v.f;
+      part = (!valid || x != x)
+                 ? 0ull
+                 : (uint64_t)__double_as_longlong(x == 0.0 ? 0.0 : x) + 1;
+    } else if (K.k[j].nullable) {
       part = valid ? (uint64_t)(v.i - K.k[j].min) + 1 : 0;
-    else
+    } else {
       part = (uint64_t)(v.i - K.k[j].min);
+    }
     code += part * K.stride[j];
   }
   return code;
@@ -937,19 +950,7 @@ __global__ void k_keypack(KeyArg K, ColsArg C, int64_t n, uint64_t* out) {
   int64_t lo, hi;
   block_range(n, 1, lo, hi);
   for (int64_t r = lo + threadIdx.x; r < hi; r += BLOCK) {
-    uint64_t code = 0;
-    for (int j = 0; j < K.nkeys; j++) {
-      Slot v;
-      bool valid;
-      vm_load_col(C, K.k[j].col, r, v, valid);
-      uint64_t part;
-      if (K.k[j].nullable)
-        part = valid ? (uint64_t)(v.i - K.k[j].min) + 1 : 0;
-      else
-        part = (uint64_t)(v.i - K.k[j].min);
-      code += part * K.stride[j];
-    }
-    out[r] = code;
+    out[r] = pack_key(K, C, r);
   }
 }
 
@@ -963,6 +964,10 @@ extern "C" int dsx_keypack(DsxCtx* c, const DsxColumn* cols, int ncols,
   for (int j = 0; j < nkeys; j++) {
     K.k[j] = keys[j];
     K.stride[j] = stride;
+    if (keys[j].mode == 1) {
+      if (nkeys != 1) FAIL(-3, "f64-bits key must be the only key");
+      continue;  // full u64 code space; stride stays 1
+    }
     uint64_t range = (uint64_t)keys[j].range + (keys[j].nullable ? 1 : 0);
     if (range == 0) FAIL(-3, "empty key range");
     if (stride > (1ull << 62) / range) FAIL(-4, "key space exceeds 2^62");
@@ -2173,14 +2178,21 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
   KeyArg K{};
   K.nkeys = nkeys;
   uint64_t key_space = 1;
+  bool bits_key = false;
   for (int j = 0; j < nkeys; j++) {
     K.k[j] = keys[j];
     K.stride[j] = key_space;
+    if (keys[j].mode == 1) {
+      if (nkeys != 1) FAIL(-3, "f64-bits key must be the only key");
+      bits_key = true;
+      continue;
+    }
     uint64_t range = (uint64_t)keys[j].range + (keys[j].nullable ? 1 : 0);
     if (range == 0) FAIL(-3, "empty key range");
     if (key_space > (1ull << 62) / range) FAIL(-4, "key space exceeds 2^62");
     key_space *= range;
   }
+  if (bits_key) key_space = 0;  // unbounded: CAS hash path only
   ColsArg C{};
   C.ncols = ncols;
   for (int i = 0; i < ncols; i++) {
@@ -2226,7 +2238,8 @@ extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
     bool all_nn = true;
     for (int a = 0; a < naggs; a++) all_nn &= (A.never_null[a] != 0);
     static const bool part_disabled = getenv("DSX_DISABLE_PART") != nullptr;
-    if (!part_disabled && all_nn && naggs <= 6 && g_est > 0) {
+    if (!part_disabled && all_nn && naggs <= 6 && g_est > 0 &&
+        key_space > 0) {
       bool fell_back = false;
       int prc = groupby_partition(c, C, n, K, P, A, aggs, progs, lens, naggs,
                                   g_est, key_space, out_codes, out_vals,

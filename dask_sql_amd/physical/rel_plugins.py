@@ -616,18 +616,8 @@ class DaskAggregatePlugin(BaseRelPlugin):
                 group_idx.append(e.getIndex())
             agg_calls = agg.getNamedAggCalls()
 
-        # key specs from minmax
-        keyspecs = []
-        for gi in group_idx:
-            col = cols[gi]
-            if col.dtype not in _INT_KINDS:
-                raise RexCompileError(
-                    "non-integer GROUP BY keys on GPU path (round-2)")
-            mn, mx, nn = _minmax_cached(runtime, col)
-            if nn == 0:
-                mn, mx = 0, 0
-            nullable = bool(col.validity)
-            keyspecs.append((gi, mn, mx - mn + 1, nullable))
+        # key specs from minmax (float key: bit-pattern mode, single key)
+        keyspecs = self._keyspecs_for(runtime, cols, group_idx)
 
         # bucket aggs by (filter_col_index, distinct) — aggregate.py:377-520
         from collections import OrderedDict
@@ -734,19 +724,14 @@ class DaskAggregatePlugin(BaseRelPlugin):
             base_dc = RelConverter.convert(base_rel, context)
             base_cols = base_dc.backend_cols()
             dicts = _dicts_of(base_cols)
-            keyspecs = []
             group_meta = []
+            gidx = []
             for e in group_exprs:
                 proj_expr, proj_name = named[e.getIndex()]
                 bi = proj_expr.getIndex()
-                col = base_cols[bi]
-                if col.dtype not in _INT_KINDS:
-                    return None
-                mn, mx, nn = _minmax_cached(runtime, col)
-                if nn == 0:
-                    mn, mx = 0, 0
-                keyspecs.append((bi, mn, mx - mn + 1, bool(col.validity)))
-                group_meta.append((proj_name, col))
+                gidx.append(bi)
+                group_meta.append((proj_name, base_cols[bi]))
+            keyspecs = self._keyspecs_for(runtime, base_cols, gidx)
             pred_prog = None
             if pred_exprs:
                 cond = pred_exprs[0]
@@ -779,6 +764,31 @@ class DaskAggregatePlugin(BaseRelPlugin):
         return self._device_exec(runtime, rel, base_cols,
                                  base_dc.table.num_rows, keyspecs, group_meta,
                                  pred_prog, calls, specs, fins, slab)
+
+    @staticmethod
+    def _keyspecs_for(runtime, cols, group_idx):
+        """(idx, min, range, nullable[, mode]) per key. Float keys group by
+        canonical f64 bit pattern (mode 1) through the CAS hash path —
+        pandas float group keys compare exactly, NaN is one group under
+        dropna=False (aggregate.py:575-577)."""
+        keyspecs = []
+        for gi in group_idx:
+            col = cols[gi]
+            if col.dtype in (rt.F64, rt.F32):
+                if len(group_idx) != 1:
+                    raise RexCompileError(
+                        "a float GROUP BY key must be the only key")
+                keyspecs.append((gi, 0, 0, True, 1))
+                continue
+            if col.dtype not in _INT_KINDS:
+                raise RexCompileError(
+                    "unsupported GROUP BY key dtype on GPU path")
+            mn, mx, nn = _minmax_cached(runtime, col)
+            if nn == 0:
+                mn, mx = 0, 0
+            nullable = bool(col.validity)
+            keyspecs.append((gi, mn, mx - mn + 1, nullable))
+        return keyspecs
 
     def _convert_device(self, runtime, rel, dc, cols, keyspecs, group_idx,
                         filt_idx, calls, agg):
@@ -858,9 +868,23 @@ class DaskAggregatePlugin(BaseRelPlugin):
         order_names = []
 
         # group keys: unpack on device — part = (code / stride) % space
+        from dask_sql_amd.physical.rex import OP_BITS_F64, OP_SUB_I64
         stride = 1
-        for j, ((gi, mn, rng, nullable), (name, src)) in enumerate(
-                zip(keyspecs, group_meta)):
+        for j, (ks, (name, src)) in enumerate(zip(keyspecs, group_meta)):
+            gi, mn, rng, nullable = ks[:4]
+            if len(ks) > 4 and ks[4] == 1:
+                # f64-bits key: value = bitcast(code-1); code 0 = NaN/NULL
+                prog = [(OP_COL, 0, 0), (OP_LIT_I64, 0, 0),
+                        (OP_GT_I64, 0, 0),
+                        (OP_COL, 0, 0), (OP_LIT_I64, 0, 1), (OP_SUB_I64, 0, 0),
+                        (OP_BITS_F64, 0, 0),
+                        (OP_LIT_NULL, 0, 0), (OP_SELECT, 0, 0)]
+                col = runtime.eval(runtime.make_prog(prog), [codes_col], G,
+                                   rt.F64, with_validity=True)
+                col.logical_dtype = src.dtype
+                out_cols[f"g__{name}"] = col
+                order_names.append((name, f"g__{name}"))
+                continue
             space = rng + (1 if nullable else 0)
             prog = [(OP_COL, 0, 0), (OP_LIT_I64, 0, stride), (17, 0, 0),
                     (OP_LIT_I64, 0, space), (18, 0, 0)]  # DIV, MOD
@@ -1060,6 +1084,9 @@ class DaskAggregatePlugin(BaseRelPlugin):
                          calls):
         """SUM/AVG/COUNT(DISTINCT x): two-level groupby — first
         (group, x) distinct pairs, then aggregate (aggregate.py:562-565)."""
+        if any(len(k) > 4 and k[4] == 1 for k in keyspecs):
+            raise RexCompileError(
+                "DISTINCT aggregate with a float GROUP BY key")
         for call in calls:
             args = agg.getArgs(call)
             if not args or not isinstance(args[0], InputRef):
@@ -1087,7 +1114,11 @@ class DaskAggregatePlugin(BaseRelPlugin):
             runtime._free(on)
             # unpack: group code = pairs % group_space; value part on top
             group_space = 1
-            for (_, _, rng, nullable) in keyspecs:
+            for ks in keyspecs:
+                _, _, rng, nullable = ks[:4]
+                if len(ks) > 4 and ks[4] == 1:
+                    raise RexCompileError(
+                        "DISTINCT aggregate with a float GROUP BY key")
                 group_space *= rng + (1 if nullable else 0)
             gcodes = pairs % group_space
             vpart = pairs // group_space
@@ -1148,7 +1179,20 @@ class DaskAggregatePlugin(BaseRelPlugin):
 
         # group key columns
         stride = 1
-        for j, (gi, mn, rng, nullable) in enumerate(keyspecs):
+        for j, ks in enumerate(keyspecs):
+            gi, mn, rng, nullable = ks[:4]
+            if len(ks) > 4 and ks[4] == 1:
+                # f64-bits key (host path): bitcast(code-1), code 0 → NaN
+                src = cols[gi]
+                vals = np.where(codes_np > 0,
+                                (codes_np - 1).astype(np.uint64),
+                                np.uint64(0)).view(np.float64)
+                vals = np.where(codes_np > 0, vals, np.nan)
+                col = runtime.upload_column(vals.astype(np.float64))
+                name = cc_in.columns[gi]
+                out_cols[f"g__{name}"] = col
+                order_names.append((name, f"g__{name}"))
+                continue
             space = rng + (1 if nullable else 0)
             part = (codes_np // stride) % space
             stride *= space

@@ -19,7 +19,7 @@ OP_LT_F64, OP_LE_F64, OP_GT_F64, OP_GE_F64, OP_EQ_F64, OP_NE_F64 = (
 OP_LT_I64, OP_LE_I64, OP_GT_I64, OP_GE_I64, OP_EQ_I64, OP_NE_I64 = (
     30, 31, 32, 33, 34, 35)
 OP_AND, OP_OR, OP_NOT, OP_IS_NULL, OP_IS_NOT_NULL = 40, 41, 42, 43, 44
-OP_I64_TO_F64, OP_F64_TO_I64 = 50, 51
+OP_I64_TO_F64, OP_F64_TO_I64, OP_BITS_F64 = 50, 51, 52
 OP_SELECT, OP_NEG_F64, OP_NEG_I64, OP_SQRT_F64 = 60, 61, 62, 63
 
 # VM value kinds

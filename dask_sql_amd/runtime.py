@@ -73,6 +73,7 @@ class _KeySpec(ct.Structure):
         ("min", ct.c_int64),
         ("range", ct.c_int64),
         ("nullable", ct.c_int32),
+        ("mode", ct.c_int32),
     ]
 
 
@@ -383,7 +384,9 @@ class Runtime:
         Returns (codes DeviceColumn(u64-as-i64), key_space)."""
         ks = (_KeySpec * len(keyspecs))()
         space = 1
-        for i, (ci, mn, rng, nullable) in enumerate(keyspecs):
+        for i, spec in enumerate(keyspecs):
+            ci, mn, rng, nullable = spec[:4]
+            ks[i].mode = spec[4] if len(spec) > 4 else 0
             ks[i].col = ci
             ks[i].min = mn
             ks[i].range = rng
@@ -446,7 +449,9 @@ class Runtime:
         in-kernel. agg_specs: list of (agg_op, prog). Returns device pointers
         (out_codes, out_vals [naggs][G], out_counts [naggs][G], n_groups)."""
         ks = (_KeySpec * max(len(keyspecs), 1))()
-        for i, (ci, mn, rng, nullable) in enumerate(keyspecs):
+        for i, spec in enumerate(keyspecs):
+            ci, mn, rng, nullable = spec[:4]
+            ks[i].mode = spec[4] if len(spec) > 4 else 0
             ks[i].col = ci
             ks[i].min = mn
             ks[i].range = rng

@@ -60,8 +60,9 @@ enum DsxOp {
   DSX_OP_EQ_I64 = 34, DSX_OP_NE_I64 = 35,
   DSX_OP_AND = 40, DSX_OP_OR = 41, DSX_OP_NOT = 42,   /* SQL 3-valued logic   */
   DSX_OP_IS_NULL = 43, DSX_OP_IS_NOT_NULL = 44,
-  DSX_OP_I64_TO_F64 = 50, DSX_OP_F64_TO_I64 = 51,     /* CAST (trunc,
-                                                         mappings.py:346-353) */
+  DSX_OP_I64_TO_F64 = 50, DSX_OP_F64_TO_I64 = 51,  /* CAST (trunc,
+                                                      mappings.py:346-353) */
+  DSX_OP_BITS_F64 = 52,  /* reinterpret i64 bits as f64 (float key unpack) */
   DSX_OP_SELECT = 60,   /* (cond, a, b) -> cond ? a : b — CASE WHEN           */
   DSX_OP_NEG_F64 = 61, DSX_OP_NEG_I64 = 62, DSX_OP_SQRT_F64 = 63,
 };
@@ -156,6 +157,10 @@ typedef struct DsxKeySpec {
   int64_t min;        /* from dsx_minmax */
   int64_t range;      /* max-min+1 (+1 more reserved internally if nullable) */
   int32_t nullable;   /* 0/1 */
+  int32_t mode;       /* 0 = radix pack; 1 = f64 bit-pattern (must be the
+                         ONLY key): code = canonical_bits(x)+1, NaN and NULL
+                         share code 0 — pandas groups NaN keys together under
+                         dropna=False (aggregate.py:575-577) */
 } DsxKeySpec;
 int dsx_keypack(DsxCtx* ctx, const DsxColumn* cols, int ncols,
                 const DsxKeySpec* keys, int nkeys, int64_t n,

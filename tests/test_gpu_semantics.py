@@ -611,3 +611,48 @@ def test_dict_string_functions(ctx):
     assert sorted(got3["v"].astype(int).tolist()) == [2, 3]
     got4 = c.sql("SELECT v FROM t WHERE UPPER(s) LIKE 'A%'").compute()
     assert sorted(got4["v"].astype(int).tolist()) == [1, 5]
+
+
+def test_float_group_by_key(ctx):
+    """GROUP BY on a float column: exact bit-pattern grouping (-0.0 == 0.0,
+    one NaN group, NULL joins the NaN group like pandas dropna=False over a
+    NaN-bearing float key — aggregate.py:575-577)."""
+    from dask_sql_amd.context import Context
+    vals = np.array([1.5, 2.5, 1.5, -0.0, 0.0, np.nan, 2.5, np.nan, 1.5])
+    df = pd.DataFrame({"f": vals, "v": np.arange(9, dtype=np.int64)})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT f, COUNT(*) AS c, SUM(v) AS s FROM t GROUP BY f"
+                ).compute()
+    exp = df.groupby("f", dropna=False).agg(
+        c=("v", "size"), s=("v", "sum")).reset_index()
+    got = got.sort_values("f", na_position="last").reset_index(drop=True)
+    exp = exp.sort_values("f", na_position="last").reset_index(drop=True)
+    assert len(got) == len(exp) == 4  # 1.5, 0.0(-0.0 merged), 2.5, NaN
+    gf = got["f"].to_numpy(dtype=np.float64)
+    ef = exp["f"].to_numpy(dtype=np.float64)
+    assert ((gf == ef) | (np.isnan(gf) & np.isnan(ef))).all()
+    assert got["c"].astype(int).tolist() == exp["c"].astype(int).tolist()
+    assert got["s"].astype(int).tolist() == exp["s"].astype(int).tolist()
+
+
+def test_float_group_by_key_large(ctx):
+    """Float key at scale through the CAS hash path (unbounded key space)."""
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(31)
+    f = np.round(rng.random(200_000) * 50, 1)  # ~501 distinct values
+    df = pd.DataFrame({"f": f, "v": rng.random(200_000)})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT f, SUM(v) AS s, COUNT(*) AS c FROM t WHERE v > 0.25 "
+                "GROUP BY f").compute()
+    pdf = df[df.v > 0.25]
+    exp = pdf.groupby("f").agg(s=("v", "sum"), c=("v", "size")).reset_index()
+    got = got.sort_values("f").reset_index(drop=True)
+    exp = exp.sort_values("f").reset_index(drop=True)
+    assert len(got) == len(exp)
+    np.testing.assert_array_equal(got["f"].to_numpy(np.float64),
+                                  exp["f"].to_numpy())
+    np.testing.assert_allclose(got["s"].to_numpy(np.float64),
+                               exp["s"].to_numpy(), rtol=1e-9)
+    assert (got["c"].to_numpy(np.int64) == exp["c"].to_numpy()).all()
