@@ -296,6 +296,23 @@ __device__ __forceinline__ void vm_load_col(const ColsArg& C, int ci,
   }
 }
 
+// civil date from days-since-epoch (Howard Hinnant's algorithm; the
+// reference gets this from pandas datetime64[D] → .dt.year/month/day,
+// rex/core/call.py date extraction)
+__device__ __forceinline__ void civil_from_days(int64_t days, int& y, int& m,
+                                                int& d) {
+  int64_t z = days + 719468;
+  int64_t era = (z >= 0 ? z : z - 146096) / 146097;
+  int64_t doe = z - era * 146097;
+  int64_t yoe = (doe - doe / 1460 + doe / 36524 - doe / 146096) / 365;
+  int64_t yy = yoe + era * 400;
+  int64_t doy = doe - (365 * yoe + yoe / 4 - yoe / 100);
+  int64_t mp = (5 * doy + 2) / 153;
+  d = (int)(doy - (153 * mp + 2) / 5 + 1);
+  m = (int)(mp < 10 ? mp + 3 : mp - 9);
+  y = (int)(yy + (m <= 2));
+}
+
 // VM stack lives in NAMED registers: `sp` is wave-uniform (every lane runs
 // the same program), so the switch below lowers to scalar branches and the
 // slots stay in VGPRs — a runtime-indexed local array would spill every
@@ -444,6 +461,57 @@ __device__ __forceinline__ bool vm_eval(const DsxInstr* prog, int len, const Col
         res.f = __longlong_as_double(a.i);
         k.set(sp - 1, res, av);
         break;
+      case DSX_OP_ABS_I64:
+        UN();
+        res.i = a.i < 0 ? -a.i : a.i;
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_ABS_F64:
+        UN();
+        res.f = fabs(a.f);
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_FLOOR_F64:
+        UN();
+        res.f = floor(a.f);
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_CEIL_F64:
+        UN();
+        res.f = ceil(a.f);
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_RINT_F64:  // ties-to-even, numpy round (call.py round op)
+        UN();
+        res.f = rint(a.f);
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_EXP_F64:
+        UN();
+        res.f = exp(a.f);
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_LN_F64:
+        UN();
+        res.f = log(a.f);
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_POW_F64:
+        k.get(--sp, b, bv);
+        k.get(sp - 1, a, av);
+        res.f = pow(a.f, b.f);
+        k.set(sp - 1, res, av && bv);
+        break;
+      case DSX_OP_YEAR:
+      case DSX_OP_MONTH:
+      case DSX_OP_DAY: {
+        UN();
+        int y, m, d;
+        civil_from_days(a.i, y, m, d);
+        res.i = op == DSX_OP_YEAR ? y : op == DSX_OP_MONTH ? m : d;
+        k.set(sp - 1, res, av);
+        break;
+      }
       case DSX_OP_NEG_F64:
         UN();
         res.f = -a.f;

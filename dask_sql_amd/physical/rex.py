@@ -21,6 +21,9 @@ OP_LT_I64, OP_LE_I64, OP_GT_I64, OP_GE_I64, OP_EQ_I64, OP_NE_I64 = (
 OP_AND, OP_OR, OP_NOT, OP_IS_NULL, OP_IS_NOT_NULL = 40, 41, 42, 43, 44
 OP_I64_TO_F64, OP_F64_TO_I64, OP_BITS_F64 = 50, 51, 52
 OP_SELECT, OP_NEG_F64, OP_NEG_I64, OP_SQRT_F64 = 60, 61, 62, 63
+OP_ABS_I64, OP_ABS_F64, OP_FLOOR_F64, OP_CEIL_F64 = 64, 65, 66, 67
+OP_RINT_F64, OP_EXP_F64, OP_LN_F64, OP_POW_F64 = 68, 69, 70, 71
+OP_YEAR, OP_MONTH, OP_DAY = 72, 73, 74
 
 # VM value kinds
 KI, KF, KB = "i", "f", "b"  # int64-like, float64, boolean
@@ -215,6 +218,53 @@ class RexCompiler:
             return self._compile_case(ops)
         if op == "LIKE":
             return self._compile_like(ops)
+        # scalar math + date extraction (rex/core/call.py scalar operations:
+        # abs/floor/ceil/round via numpy, exp/log/power, year/month/day)
+        if op == "ABS":
+            k = self.compile(ops[0])
+            self._emit(OP_ABS_F64 if k == KF else OP_ABS_I64)
+            return k
+        _f1 = {"FLOOR": OP_FLOOR_F64, "CEIL": OP_CEIL_F64, "CEILING":
+               OP_CEIL_F64, "EXP": OP_EXP_F64, "LN": OP_LN_F64,
+               "LOG": OP_LN_F64, "SQRT": OP_SQRT_F64}
+        if op in _f1:
+            self._to_f(self.compile(ops[0]))
+            self._emit(_f1[op])
+            return KF
+        if op == "ROUND":
+            self._to_f(self.compile(ops[0]))
+            if len(ops) == 2:
+                if not isinstance(ops[1], Literal):
+                    raise RexCompileError("ROUND digits must be a literal")
+                p = 10.0 ** int(ops[1].getValue())
+                self._emit(OP_LIT_F64, 0, p)
+                self._emit(OP_MUL_F64)
+                self._emit(OP_RINT_F64)  # ties-to-even like numpy round
+                self._emit(OP_LIT_F64, 0, p)
+                self._emit(OP_DIV_F64)
+            else:
+                self._emit(OP_RINT_F64)
+            return KF
+        if op in ("POWER", "POW"):
+            self._to_f(self.compile(ops[0]))
+            self._to_f(self.compile(ops[1]))
+            self._emit(OP_POW_F64)
+            return KF
+        if op == "MOD":
+            self.compile(ops[0])
+            self.compile(ops[1])
+            self._emit(OP_MOD_I64)
+            return KI
+        _dx = {"EXTRACT_YEAR": OP_YEAR, "YEAR": OP_YEAR,
+               "EXTRACT_MONTH": OP_MONTH, "MONTH": OP_MONTH,
+               "EXTRACT_DAY": OP_DAY, "DAY": OP_DAY,
+               "DAYOFMONTH": OP_DAY}
+        if op in _dx:
+            k = self.compile(ops[0])
+            if k != KI:
+                raise RexCompileError(f"{op} needs a DATE (day-int) operand")
+            self._emit(_dx[op])
+            return KI
         raise RexCompileError(f"operator {op} not supported on GPU path")
 
     def _compile_like(self, ops):
@@ -344,10 +394,13 @@ class RexCompiler:
                 return KF if KF in (ka, kb) else KI
             if op == "CAST":
                 return _SQL_TO_KIND.get(expr.getType().getSqlType(), KF)
-            if op == "NEG":
+            if op in ("NEG", "ABS"):
                 return self._peek_kind(expr.getOperands()[0])
             if op == "CASE":
                 return self._peek_kind(expr.getOperands()[1])
+            if op in ("MOD", "EXTRACT_YEAR", "EXTRACT_MONTH", "EXTRACT_DAY",
+                      "YEAR", "MONTH", "DAY", "DAYOFMONTH"):
+                return KI
         return KF
 
 

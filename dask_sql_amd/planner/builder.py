@@ -139,6 +139,14 @@ class Builder:
                 ty = _expr_type(ops[0])
             elif op in ("UPPER", "LOWER", "SUBSTRING", "SUBSTR"):
                 ty = "VARCHAR"
+            elif op in ("FLOOR", "CEIL", "CEILING", "ROUND", "EXP", "LN",
+                        "LOG", "POWER", "POW", "SQRT"):
+                ty = "DOUBLE"
+            elif op in ("EXTRACT_YEAR", "EXTRACT_MONTH", "EXTRACT_DAY",
+                        "MOD", "YEAR", "MONTH", "DAY", "DAYOFMONTH"):
+                ty = "BIGINT"
+            elif op == "ABS":
+                ty = _expr_type(ops[0])
             elif op == "/":
                 ty = _common_type(_expr_type(ops[0]), _expr_type(ops[1]))
             else:

@@ -397,6 +397,17 @@ class Parser:
                 return ("cast", e, ty)
         if t[0] == "id":
             name = self._name()
+            if name.upper() == "EXTRACT" and self.peek() == ("op", "("):
+                # EXTRACT(field FROM expr)
+                self.next()
+                ft = self.next()
+                field = ft[1].upper()
+                if field not in ("YEAR", "MONTH", "DAY"):
+                    raise ValueError(f"EXTRACT({field}) not supported")
+                self.expect_kw("FROM")
+                e = self.expr()
+                self.expect_op(")")
+                return ("call", f"EXTRACT_{field}", [e])
             if self.peek() == ("op", "("):
                 return self._func_call(name)
             if self.accept_op("."):

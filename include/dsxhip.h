@@ -65,6 +65,12 @@ enum DsxOp {
   DSX_OP_BITS_F64 = 52,  /* reinterpret i64 bits as f64 (float key unpack) */
   DSX_OP_SELECT = 60,   /* (cond, a, b) -> cond ? a : b — CASE WHEN           */
   DSX_OP_NEG_F64 = 61, DSX_OP_NEG_I64 = 62, DSX_OP_SQRT_F64 = 63,
+  /* scalar math + date extraction (rex/core/call.py scalar operations) */
+  DSX_OP_ABS_I64 = 64, DSX_OP_ABS_F64 = 65,
+  DSX_OP_FLOOR_F64 = 66, DSX_OP_CEIL_F64 = 67,
+  DSX_OP_RINT_F64 = 68,  /* ties-to-even like numpy round */
+  DSX_OP_EXP_F64 = 69, DSX_OP_LN_F64 = 70, DSX_OP_POW_F64 = 71,
+  DSX_OP_YEAR = 72, DSX_OP_MONTH = 73, DSX_OP_DAY = 74, /* date32 day-int */
 };
 
 typedef struct DsxInstr {

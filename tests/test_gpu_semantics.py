@@ -656,3 +656,58 @@ def test_float_group_by_key_large(ctx):
     np.testing.assert_allclose(got["s"].to_numpy(np.float64),
                                exp["s"].to_numpy(), rtol=1e-9)
     assert (got["c"].to_numpy(np.int64) == exp["c"].to_numpy()).all()
+
+
+def test_scalar_math_functions(ctx):
+    """ABS/FLOOR/CEIL/ROUND/EXP/LN/POWER/MOD/SQRT — rex/core/call.py scalar
+    operations; ROUND is ties-to-even like the reference's numpy round."""
+    from dask_sql_amd.context import Context
+    df = pd.DataFrame({
+        "x": [-2.7, -0.5, 0.5, 1.5, 2.345],
+        "i": np.array([-7, -1, 0, 5, 12], dtype=np.int64),
+    })
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql(
+        "SELECT ABS(x) AS ax, ABS(i) AS ai, FLOOR(x) AS fl, CEIL(x) AS ce, "
+        "ROUND(x) AS r0, ROUND(x, 1) AS r1, EXP(x) AS ex, "
+        "POWER(ABS(x), 2) AS p2, MOD(i, 3) AS md, SQRT(ABS(i)) AS sq "
+        "FROM t").compute()
+    np.testing.assert_allclose(got["ax"], np.abs(df.x), rtol=1e-12)
+    assert got["ai"].astype(int).tolist() == [7, 1, 0, 5, 12]
+    np.testing.assert_allclose(got["fl"], np.floor(df.x))
+    np.testing.assert_allclose(got["ce"], np.ceil(df.x))
+    np.testing.assert_allclose(got["r0"], np.round(df.x))  # ties-to-even
+    np.testing.assert_allclose(got["r1"], np.round(df.x, 1), atol=1e-12)
+    np.testing.assert_allclose(got["ex"], np.exp(df.x), rtol=1e-12)
+    np.testing.assert_allclose(got["p2"], df.x ** 2, rtol=1e-12)
+    # C-style % (truncated): matches our integer division semantics
+    assert got["md"].astype(int).tolist() == [
+        int(np.fmod(v, 3)) for v in df.i]
+    np.testing.assert_allclose(got["sq"], np.sqrt(np.abs(df.i)), rtol=1e-12)
+
+
+def test_extract_date_parts(ctx):
+    """EXTRACT(YEAR/MONTH/DAY FROM date) and YEAR()/MONTH()/DAY() —
+    rex/core/call.py date extraction (pandas .dt accessors)."""
+    from dask_sql_amd.context import Context
+    dates = pd.to_datetime(["1995-03-15", "2000-02-29", "1970-01-01",
+                            "2023-12-31", "1969-07-20"])
+    df = pd.DataFrame({"d": dates, "v": np.arange(5)})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT EXTRACT(YEAR FROM d) AS y, "
+                "EXTRACT(MONTH FROM d) AS m, EXTRACT(DAY FROM d) AS dd, "
+                "YEAR(d) AS y2 FROM t").compute()
+    assert got["y"].astype(int).tolist() == dates.year.tolist()
+    assert got["m"].astype(int).tolist() == dates.month.tolist()
+    assert got["dd"].astype(int).tolist() == dates.day.tolist()
+    assert got["y2"].astype(int).tolist() == dates.year.tolist()
+    # in WHERE and GROUP BY positions too
+    got2 = c.sql("SELECT YEAR(d) AS y, COUNT(*) AS c FROM t "
+                 "WHERE EXTRACT(YEAR FROM d) >= 1970 GROUP BY YEAR(d)"
+                 ).compute()
+    exp = df[dates.year >= 1970].groupby(dates.year[dates.year >= 1970]
+                                         ).size()
+    assert sorted(got2["y"].astype(int).tolist()) == sorted(
+        exp.index.tolist())
