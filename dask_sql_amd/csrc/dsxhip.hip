@@ -842,6 +842,13 @@ extern "C" int dsx_memset(DsxCtx* c, void* dev, int value, int64_t bytes) {
   return 0;
 }
 
+extern "C" int dsx_copy(DsxCtx* c, void* dst, const void* src,
+                        int64_t bytes) {
+  HIP_TRY(hipMemcpyAsync(dst, src, (size_t)bytes, hipMemcpyDeviceToDevice,
+                         c->stream));
+  return 0;
+}
+
 // row scatter — inverse of gather: out[sel[i]] = in[i]. Places join-back /
 // window columns into original row order (sel must be a permutation or a
 // subset of [0, n_out); untouched out rows keep their init value).

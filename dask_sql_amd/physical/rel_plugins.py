@@ -1590,8 +1590,84 @@ class DaskWindowPlugin(BaseRelPlugin):
             if nanmask.any() else None)
 
 
+class DaskUnionPlugin(BaseRelPlugin):
+    """Positional UNION ALL: device concatenation of the branch columns
+    (reference Union rel → dd.concat of the branch frames). Dict string
+    columns merge dictionaries (rhs codes remapped on device); mixed
+    numeric positions promote to f64."""
+
+    class_name = "Union"
+
+    def convert(self, rel, context):
+        runtime = context._get_runtime()
+        dcs = self.assert_inputs(rel, 2, context)
+        fields = rel.getRowType().getFieldList()
+        per_input = []
+        for dc in dcs:
+            cc = dc.column_container
+            per_input.append([dc.table.col(cc.get_backend_by_frontend_name(f))
+                              for f in cc.columns])
+        n_total = sum(dc.table.num_rows for dc in dcs)
+        out_cols = {}
+        order_names = []
+        for i, f in enumerate(fields):
+            col = self._concat(runtime, [cols[i] for cols in per_input])
+            bname = f"u{i}__{f.getName()}"
+            out_cols[bname] = col
+            order_names.append((f.getName(), bname))
+        cc = ColumnContainer([nm for nm, _ in order_names],
+                             dict(order_names))
+        cc = self.fix_column_to_row_type(cc, rel.getRowType())
+        return DataContainer(DeviceTable(out_cols, num_rows=n_total), cc)
+
+    def _concat(self, runtime, cols):
+        dicts = [getattr(c, "dictionary", None) for c in cols]
+        if any(d is not None for d in dicts):
+            if not all(d is not None for d in dicts):
+                raise RexCompileError(
+                    "UNION position mixes string and non-string")
+            base = list(dicts[0])
+            index = {s: j for j, s in enumerate(base)}
+            parts = [cols[0]]
+            for c, d in zip(cols[1:], dicts[1:]):
+                if d is dicts[0] or list(d) == base[:len(d)]:
+                    parts.append(c)
+                    continue
+                m = []
+                for s in d:
+                    if s is not None and s not in index:
+                        index[s] = len(base)
+                        base.append(s)
+                    m.append(index.get(s, 0))
+                map_col = runtime.upload_column(np.array(m, dtype=np.int32),
+                                                dtype=rt.I32)
+                g = runtime.gather(map_col, c.data, c.len)
+                parts.append(rt.DeviceColumn(runtime, g.data, c.validity,
+                                             c.len, rt.I32, owner=False,
+                                             keep_alive=(g, c, map_col)))
+            out = runtime.concat_columns(parts, rt.I32)
+            out.dictionary = base
+            return out
+        target = cols[0].dtype
+        if any(c.dtype != target for c in cols):
+            target = rt.F64 if any(c.dtype in (rt.F64, rt.F32)
+                                   for c in cols) else rt.I64
+        parts = []
+        for c in cols:
+            if c.dtype == target:
+                parts.append(c)
+            else:
+                prog = [(OP_COL, 0, 0)]
+                if target == rt.F64 and c.dtype not in (rt.F64, rt.F32):
+                    prog.append((50, 0, 0))  # I64_TO_F64
+                parts.append(runtime.eval(runtime.make_prog(prog), [c],
+                                          c.len, target,
+                                          with_validity=bool(c.validity)))
+        return runtime.concat_columns(parts, target)
+
+
 def register_defaults():
     for cls in (DaskTableScanPlugin, DaskFilterPlugin, DaskProjectPlugin,
                 DaskJoinPlugin, DaskAggregatePlugin, DaskSortPlugin,
-                DaskLimitPlugin, DaskWindowPlugin):
+                DaskLimitPlugin, DaskWindowPlugin, DaskUnionPlugin):
         RelConverter.add_plugin_class(cls, replace=False)

@@ -29,6 +29,7 @@ class JoinClause:
 @dataclass
 class SelectStmt:
     items: list  # [(expr_ast, alias|None)] ; expr_ast ('star',) allowed
+    # (UnionStmt defined below wraps several SelectStmts)
     distinct: bool = False
     from_tables: list = field(default_factory=list)  # [TableRef]
     joins: list = field(default_factory=list)  # [JoinClause]
@@ -36,6 +37,15 @@ class SelectStmt:
     group_by: list = field(default_factory=list)
     having: tuple | None = None
     order_by: list = field(default_factory=list)  # [(expr, asc, nulls_first)]
+    limit: int | None = None
+    offset: int = 0
+
+
+@dataclass
+class UnionStmt:
+    branches: list  # [SelectStmt]
+    alls: list      # [bool] per UNION step (False = UNION DISTINCT)
+    order_by: list = field(default_factory=list)
     limit: int | None = None
     offset: int = 0
 
@@ -136,11 +146,25 @@ class Parser:
             raise ValueError(f"expected {op!r}, got {t}")
 
     # -- entry --------------------------------------------------------------
-    def parse(self) -> SelectStmt:
+    def parse(self):
         stmt = self.select_stmt()
+        branches = [stmt]
+        alls = []
+        while self.accept_kw("UNION"):
+            alls.append(bool(self.accept_kw("ALL")))
+            branches.append(self.select_stmt())
         if self.peek()[0] != "eof":
             raise ValueError(f"trailing tokens: {self.peek()}")
-        return stmt
+        if len(branches) == 1:
+            return stmt
+        # trailing ORDER BY / LIMIT bind to the whole union, not the last
+        # branch (standard SQL)
+        last = branches[-1]
+        u = UnionStmt(branches=branches, alls=alls,
+                      order_by=last.order_by, limit=last.limit,
+                      offset=last.offset)
+        last.order_by, last.limit, last.offset = [], None, 0
+        return u
 
     def select_stmt(self) -> SelectStmt:
         self.expect_kw("SELECT")

@@ -306,6 +306,11 @@ class AggregateNode:
         return list(self._distinct_columns)
 
 
+class UnionNode:
+    """Positional UNION ALL of the inputs (reference: the Union rel lowered
+    to dd.concat of the branch frames)."""
+
+
 class WindowSpec:
     """One window column (reference rel/logical/window.py:212-428 lowering).
     arg/partition/order are input field indices; out column appended after

@@ -331,6 +331,35 @@ class Runtime:
         return DeviceColumn(self, ptr, None, n, dtype, owner=False,
                             keep_alive=keep_alive)
 
+    def concat_columns(self, cols, dtype) -> DeviceColumn:
+        """Concatenate device columns of one dtype (UNION ALL — reference
+        dd.concat). Validity kept if any input has one (absent = all-1)."""
+        total = sum(c.len for c in cols)
+        need_valid = any(c.validity for c in cols)
+        out = self.empty_column(total, dtype, need_valid)
+        sz = _DSX_SIZE[dtype]
+        off = 0
+        for c in cols:
+            assert c.dtype == dtype
+            if c.len:
+                _check(self.lib,
+                       self.lib.dsx_copy(self.ctx,
+                                         ct.c_void_p(out.data + off * sz),
+                                         ct.c_void_p(c.data),
+                                         ct.c_int64(c.len * sz)), "dsx_copy")
+                if need_valid:
+                    if c.validity:
+                        _check(self.lib, self.lib.dsx_copy(
+                            self.ctx, ct.c_void_p(out.validity + off),
+                            ct.c_void_p(c.validity), ct.c_int64(c.len)),
+                            "dsx_copy")
+                    else:
+                        _check(self.lib, self.lib.dsx_memset(
+                            self.ctx, ct.c_void_p(out.validity + off),
+                            ct.c_int(1), ct.c_int64(c.len)), "dsx_memset")
+            off += c.len
+        return out
+
     def scatter_rows(self, col: DeviceColumn, sel_ptr, n_sel, n_out,
                      with_validity=True) -> DeviceColumn:
         """out[sel[i]] = col[i] (inverse of gather; window/join-back

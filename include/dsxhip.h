@@ -108,6 +108,9 @@ int dsx_free(DsxCtx* ctx, void* ptr);
 int dsx_upload(DsxCtx* ctx, const void* host, int64_t bytes, void** out_dev);
 int dsx_download(DsxCtx* ctx, const void* dev, void* host, int64_t bytes);
 int dsx_memset(DsxCtx* ctx, void* dev, int value, int64_t bytes);
+/* device-to-device copy (UNION ALL concatenation — the reference's
+ * dd.concat of the union branches) */
+int dsx_copy(DsxCtx* ctx, void* dst, const void* src, int64_t bytes);
 
 /* per-kernel HIP-event timing (for bench.py roofline accounting) */
 int dsx_prof_enable(DsxCtx* ctx, int enable);

@@ -711,3 +711,26 @@ def test_extract_date_parts(ctx):
                                          ).size()
     assert sorted(got2["y"].astype(int).tolist()) == sorted(
         exp.index.tolist())
+
+
+def test_union_all_and_distinct(ctx):
+    """UNION ALL = device concat; UNION = concat + Distinct; dictionary
+    merge for string positions; mixed numeric positions promote to f64
+    (reference Union rel -> dd.concat [+ drop_duplicates])."""
+    from dask_sql_amd.context import Context
+    a = pd.DataFrame({"s": ["x", "y", "x"], "v": [1, 2, 3]})
+    b = pd.DataFrame({"s": ["z", "y"], "v": [2, 9]})
+    c = Context()
+    c.create_table("a", a)
+    c.create_table("b", b)
+    got = c.sql("SELECT s, v FROM a UNION ALL SELECT s, v FROM b").compute()
+    exp = pd.concat([a, b], ignore_index=True)
+    assert sorted(zip(got["s"], got["v"].astype(int))) == \
+        sorted(zip(exp["s"], exp["v"]))
+    got2 = c.sql("SELECT v FROM a UNION SELECT v FROM b").compute()
+    assert sorted(got2["v"].astype(int).tolist()) == [1, 2, 3, 9]
+    # mixed int/float position promotes; ORDER BY applies to the whole union
+    got3 = c.sql("SELECT v AS x FROM a UNION ALL "
+                 "SELECT v * 0.5 AS x FROM b ORDER BY x").compute()
+    exp3 = sorted([1.0, 2.0, 3.0, 1.0, 4.5])
+    np.testing.assert_allclose(got3["x"].to_numpy(np.float64), exp3)

@@ -246,3 +246,13 @@ def test_create_table_from_files(tmp_path):
     import pytest
     with pytest.raises(NotImplementedError):
         c.create_table("tx", "no_such.xyz")
+
+
+def test_union_parses_and_folds():
+    from dask_sql_amd.planner.parser import Parser, UnionStmt
+    u = Parser("SELECT a FROM t UNION ALL SELECT b FROM u "
+               "UNION SELECT c FROM w ORDER BY a LIMIT 5").parse()
+    assert isinstance(u, UnionStmt)
+    assert u.alls == [True, False]
+    assert u.limit == 5 and len(u.order_by) == 1
+    assert u.branches[2].order_by == [] and u.branches[2].limit is None
