@@ -218,6 +218,45 @@ class RexCompiler:
             return self._compile_case(ops)
         if op == "LIKE":
             return self._compile_like(ops)
+        if op == "COALESCE":
+            # right-fold of SELECT(arg IS NOT NULL, arg, rest)
+            # (rex/core/call.py CoalesceOperation)
+            target = KF if any(self._peek_kind(o) == KF for o in ops) \
+                else self._peek_kind(ops[-1])
+
+            def emit(i):
+                if i == len(ops) - 1:
+                    k = self.compile(ops[i])
+                    if target == KF:
+                        self._to_f(k)
+                    return
+                self.compile(ops[i])
+                self._emit(OP_IS_NOT_NULL)
+                k = self.compile(ops[i])
+                if target == KF:
+                    self._to_f(k)
+                emit(i + 1)
+                self._emit(OP_SELECT)
+
+            emit(0)
+            return target
+        if op == "NULLIF":
+            # SELECT(a = b, NULL, a); NULL condition falls through to a —
+            # exactly SQL NULLIF (rex/core/call.py NullIf)
+            a, b = ops
+            ka, kb = self._peek_kind(a), self._peek_kind(b)
+            use_f = KF in (ka, kb)
+            k = self.compile(a)
+            if use_f:
+                self._to_f(k)
+            k = self.compile(b)
+            if use_f:
+                self._to_f(k)
+            self._emit(OP_EQ_F64 if use_f else OP_EQ_I64)
+            self._emit(OP_LIT_NULL)
+            k = self.compile(a)
+            self._emit(OP_SELECT)
+            return k
         # scalar math + date extraction (rex/core/call.py scalar operations:
         # abs/floor/ceil/round via numpy, exp/log/power, year/month/day)
         if op == "ABS":
@@ -401,6 +440,11 @@ class RexCompiler:
             if op in ("MOD", "EXTRACT_YEAR", "EXTRACT_MONTH", "EXTRACT_DAY",
                       "YEAR", "MONTH", "DAY", "DAYOFMONTH"):
                 return KI
+            if op == "COALESCE":
+                kids = [self._peek_kind(o) for o in expr.getOperands()]
+                return KF if KF in kids else kids[-1]
+            if op == "NULLIF":
+                return self._peek_kind(expr.getOperands()[0])
         return KF
 
 

@@ -442,9 +442,15 @@ class Parser:
 
     def _case(self):
         self.expect_kw("CASE")
+        # simple CASE (CASE x WHEN v THEN ...) rewrites to searched form
+        operand = None
+        if self.peek() != ("kw", "WHEN"):
+            operand = self.expr()
         whens = []
         while self.accept_kw("WHEN"):
             cond = self.expr()
+            if operand is not None:
+                cond = ("call", "=", [operand, cond])
             self.expect_kw("THEN")
             val = self.expr()
             whens.append((cond, val))

@@ -734,3 +734,25 @@ def test_union_all_and_distinct(ctx):
                  "SELECT v * 0.5 AS x FROM b ORDER BY x").compute()
     exp3 = sorted([1.0, 2.0, 3.0, 1.0, 4.5])
     np.testing.assert_allclose(got3["x"].to_numpy(np.float64), exp3)
+
+
+def test_coalesce_nullif_simple_case(ctx):
+    """COALESCE / NULLIF (rex/core/call.py CoalesceOperation, NullIf) and
+    simple-form CASE (rewritten to searched)."""
+    from dask_sql_amd.context import Context
+    df = pd.DataFrame({
+        "a": pd.array([1, None, 3, None], dtype="Int64"),
+        "b": pd.array([None, 20, 30, None], dtype="Int64"),
+        "x": [1.5, np.nan, 2.5, 0.5],
+    })
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT COALESCE(a, b, 0) AS co, NULLIF(a, 3) AS nf, "
+                "COALESCE(x, 0.0) AS cx, "
+                "CASE a WHEN 1 THEN 100 WHEN 3 THEN 300 ELSE -1 END AS sc "
+                "FROM t").compute()
+    assert got["co"].astype(int).tolist() == [1, 20, 3, 0]
+    nf = got["nf"].tolist()
+    assert nf[0] == 1 and pd.isna(nf[1]) and pd.isna(nf[2]) and pd.isna(nf[3])
+    np.testing.assert_allclose(got["cx"], [1.5, 0.0, 2.5, 0.5])
+    assert got["sc"].astype(int).tolist() == [100, -1, 300, -1]
