@@ -79,6 +79,17 @@ class Builder:
 
     # ----------------------------------------------------------------- scans
     def _scan(self, tr: TableRef) -> LogicalPlan:
+        if getattr(tr, "subquery", None) is not None:
+            # derived table: build the sub-select, requalify its output with
+            # the alias (reference: DataFusion subquery alias rel)
+            sub = self.build_stmt(tr.subquery)
+            qual = tr.alias.lower()
+            fields = [Field(f.getName(), f.getType(), qualifier=qual)
+                      for f in sub.getRowType().getFieldList()]
+            named = [(InputRef(i, f.getType()), f.getName())
+                     for i, f in enumerate(fields)]
+            return LogicalPlan("Projection", [sub], RelDataType(fields),
+                               ProjectionNode(named))
         fields_spec = self.catalog.get(tr.name)
         qual = (tr.alias or tr.name).lower()
         fields = [Field(n, SqlType(t), qualifier=qual) for n, t in fields_spec]

@@ -15,8 +15,9 @@ from dataclasses import dataclass, field
 # ---- AST ------------------------------------------------------------------
 @dataclass
 class TableRef:
-    name: str
+    name: str | None
     alias: str | None = None
+    subquery: object = None  # SelectStmt for derived tables
 
 
 @dataclass
@@ -263,6 +264,22 @@ class Parser:
         return None
 
     def table_ref(self) -> TableRef:
+        if self.peek() == ("op", "("):
+            # derived table: FROM (SELECT ...) [AS] alias
+            self.next()
+            sub = self.select_stmt()
+            while self.accept_kw("UNION"):
+                raise ValueError("UNION inside a derived table: wrap each "
+                                 "branch in its own derived table")
+            self.expect_op(")")
+            alias = None
+            if self.accept_kw("AS"):
+                alias = self._name()
+            elif self.peek()[0] == "id":
+                alias = self._name()
+            if alias is None:
+                raise ValueError("derived table needs an alias")
+            return TableRef(name=None, alias=alias, subquery=sub)
         name = self._name()
         alias = None
         if self.accept_kw("AS"):

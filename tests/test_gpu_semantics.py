@@ -756,3 +756,35 @@ def test_coalesce_nullif_simple_case(ctx):
     assert nf[0] == 1 and pd.isna(nf[1]) and pd.isna(nf[2]) and pd.isna(nf[3])
     np.testing.assert_allclose(got["cx"], [1.5, 0.0, 2.5, 0.5])
     assert got["sc"].astype(int).tolist() == [100, -1, 300, -1]
+
+
+def test_derived_table_subquery(ctx):
+    """FROM (SELECT ...) alias — derived tables (DataFusion subquery-alias
+    rel on the reference side)."""
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(41)
+    df = pd.DataFrame({"k": rng.integers(0, 10, 5000).astype(np.int64),
+                       "v": rng.random(5000)})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql(
+        "SELECT k2, SUM(s) AS ss FROM "
+        "(SELECT k % 3 AS k2, SUM(v) AS s FROM t GROUP BY k) sub "
+        "GROUP BY k2").compute()
+    inner = df.groupby("k")["v"].sum().reset_index()
+    inner["k2"] = inner["k"] % 3
+    exp = inner.groupby("k2")["v"].sum().reset_index()
+    got = got.sort_values("k2").reset_index(drop=True)
+    np.testing.assert_allclose(got["ss"].to_numpy(np.float64),
+                               exp["v"].to_numpy(), rtol=1e-9)
+    # derived table joined with a base table
+    got2 = c.sql(
+        "SELECT t.k, t.v, sub.s FROM t JOIN "
+        "(SELECT k, SUM(v) AS s FROM t GROUP BY k) sub ON t.k = sub.k "
+        "WHERE t.v > 0.9").compute()
+    ksum = df.groupby("k")["v"].sum()
+    pdf = df[df.v > 0.9]
+    np.testing.assert_allclose(
+        got2.sort_values(["k", "v"])["s"].to_numpy(np.float64),
+        pdf.assign(s=pdf.k.map(ksum)).sort_values(["k", "v"])["s"].to_numpy(),
+        rtol=1e-9)
