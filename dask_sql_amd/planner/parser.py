@@ -70,7 +70,7 @@ _TOKEN_RE = re.compile(
   | (?P<str>'(?:[^']|'')*')
   | (?P<qid>"[^"]+")
   | (?P<id>[A-Za-z_][A-Za-z_0-9]*)
-  | (?P<op><>|!=|>=|<=|=|<|>|\+|-|\*|/|\(|\)|,|\.)
+  | (?P<op><>|!=|>=|<=|=|<|>|\+|-|\*|/|%|\(|\)|,|\.)
     """,
     re.VERBOSE,
 )
@@ -385,10 +385,13 @@ class Parser:
     def mul_expr(self):
         e = self.unary_expr()
         while True:
-            op = self.accept_op("*", "/")
+            op = self.accept_op("*", "/", "%")
             if not op:
                 break
-            e = ("call", op, [e, self.unary_expr()])
+            if op == "%":
+                e = ("call", "MOD", [e, self.unary_expr()])
+            else:
+                e = ("call", op, [e, self.unary_expr()])
         return e
 
     def unary_expr(self):
