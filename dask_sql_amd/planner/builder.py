@@ -246,6 +246,20 @@ class Builder:
         if isinstance(stmt, UnionStmt):
             return self._build_union(stmt)
         where_conjuncts = self._conjuncts(stmt.where)
+        # IN (SELECT ...) conjuncts → SEMI/ANTI joins (DataFusion's subquery
+        # decorrelation on the reference side). Pulled out before pushdown.
+        in_subs = []
+        rest = []
+        for cj in where_conjuncts:
+            if cj[0] == "in_sub":
+                in_subs.append((cj[1], cj[2], False))
+            elif (cj[0] == "call" and cj[1] == "NOT"
+                  and isinstance(cj[2][0], tuple)
+                  and cj[2][0][0] == "in_sub"):
+                in_subs.append((cj[2][0][1], cj[2][0][2], True))
+            else:
+                rest.append(cj)
+        where_conjuncts = rest
         used = [False] * len(where_conjuncts)
 
         # 1. scans (+ pushed-down single-table filters, à la PushDownFilter)
@@ -328,6 +342,30 @@ class Builder:
         for jc in stmt.joins:
             rhs = scan_with_filters(jc.table)
             plan = join_plans(plan, rhs, jc.join_type, [], jc.on)
+
+        # IN-subquery joins: x IN (SELECT c FROM ...) ≡ SEMI join on x = c
+        # over the DISTINCT subquery output; NOT IN ≡ ANTI (NULL-in-subquery
+        # divergence documented in DESIGN.md)
+        for e_ast, substmt, negated in in_subs:
+            subplan = self.build_stmt(substmt)
+            sfields = subplan.getRowType().getFieldList()
+            if len(sfields) != 1:
+                raise ValueError("IN subquery must select exactly one column")
+            node = AggregateNode(
+                [InputRef(0, sfields[0].getType())], [], distinct_node=True,
+                distinct_columns=[sfields[0].getName()])
+            subplan = LogicalPlan("Distinct", [subplan],
+                                  subplan.getRowType(), node)
+            lhs_fields = plan.getRowType().getFieldList()
+            combined = RelDataType(lhs_fields + sfields)
+            tmp = LogicalPlan("__combined__", [], combined, None)
+            cond = Call("=", [self._resolve(e_ast, tmp),
+                              InputRef(len(lhs_fields),
+                                       sfields[0].getType())],
+                        SqlType("BOOLEAN"))
+            jt = "LEFTANTI" if negated else "LEFTSEMI"
+            plan = LogicalPlan("Join", [plan, subplan],
+                               RelDataType(lhs_fields), JoinNode(jt, cond))
 
         # leftover WHERE conjuncts → Filter (incl. scalar TRUE/FALSE)
         leftovers = [cj for i, cj in enumerate(where_conjuncts) if not used[i]

@@ -364,6 +364,13 @@ class Parser:
 
     def _in_list(self, e):
         self.expect_op("(")
+        if self.peek() == ("kw", "SELECT"):
+            # IN (SELECT ...) — decorrelated to a SEMI/ANTI join by the
+            # builder (what DataFusion's subquery rewriting gives the
+            # reference)
+            sub = self.select_stmt()
+            self.expect_op(")")
+            return ("in_sub", e, sub)
         items = [self.expr()]
         while self.accept_op(","):
             items.append(self.expr())

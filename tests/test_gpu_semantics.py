@@ -788,3 +788,27 @@ def test_derived_table_subquery(ctx):
         got2.sort_values(["k", "v"])["s"].to_numpy(np.float64),
         pdf.assign(s=pdf.k.map(ksum)).sort_values(["k", "v"])["s"].to_numpy(),
         rtol=1e-9)
+
+
+def test_in_subquery_semi_anti(ctx):
+    """x IN (SELECT ...) -> SEMI join over DISTINCT sub output; NOT IN ->
+    ANTI (DataFusion decorrelation on the reference side)."""
+    from dask_sql_amd.context import Context
+    big = pd.DataFrame({"k": [1, 2, 3, 4, 5, 2], "v": [10, 20, 30, 40, 50,
+                                                       21]})
+    small = pd.DataFrame({"id": [2, 4, 4, 9], "w": [0.1, 0.9, 0.5, 0.7]})
+    c = Context()
+    c.create_table("big", big)
+    c.create_table("small", small)
+    got = c.sql("SELECT k, v FROM big WHERE k IN "
+                "(SELECT id FROM small WHERE w > 0.4)").compute()
+    # ids with w>0.4: {4, 9} -> k=4 rows
+    assert sorted(zip(got["k"].astype(int), got["v"].astype(int))) == \
+        [(4, 40)]
+    got2 = c.sql("SELECT k FROM big WHERE k NOT IN (SELECT id FROM small)"
+                 ).compute()
+    assert sorted(got2["k"].astype(int).tolist()) == [1, 3, 5]
+    # no duplication through multi-match subqueries (DISTINCT before SEMI)
+    got3 = c.sql("SELECT v FROM big WHERE k IN (SELECT id FROM small)"
+                 ).compute()
+    assert sorted(got3["v"].astype(int).tolist()) == [20, 21, 40]
