@@ -1318,6 +1318,177 @@ extern "C" int dsx_hash_probe(DsxCtx* c, DsxHashTable* t, const uint64_t* codes,
   return dbg_check(c, "dsx_hash_probe");
 }
 
+// ---------------------------------------------------------------------------
+// fused probe-emit + materialization (INNER/LEFT/SEMI/ANTI, no residual):
+// instead of (probe, build) pair vectors + one gather per output column,
+// the emit pass writes the output columns directly — one pass, no pair
+// traffic (join.py:241-246 dd.merge column copy, fused).
+// ---------------------------------------------------------------------------
+struct JoinMatArg {
+  const void* src[16];
+  const uint8_t* srcv[16];
+  void* dst[16];
+  uint8_t* dstv[16];
+  int32_t dtype[16];
+  int32_t side[16];  // 0 = probe row index, 1 = build row id
+  int32_t ncols;
+};
+
+__device__ __forceinline__ void jm_write(const JoinMatArg& M, int64_t o,
+                                         int64_t r, uint32_t bid) {
+  for (int ci = 0; ci < M.ncols; ci++) {
+    int64_t idx = M.side[ci] ? (int64_t)bid : r;
+    bool missing = M.side[ci] && bid == DSX_NULL_IDX;  // LEFT NULL-fill
+    switch (M.dtype[ci]) {
+      case DSX_I64:
+        ((int64_t*)M.dst[ci])[o] = missing ? 0 : ((const int64_t*)M.src[ci])[idx];
+        break;
+      case DSX_F64:
+        ((double*)M.dst[ci])[o] =
+            missing ? __builtin_nan("") : ((const double*)M.src[ci])[idx];
+        break;
+      case DSX_I32:
+        ((int32_t*)M.dst[ci])[o] = missing ? 0 : ((const int32_t*)M.src[ci])[idx];
+        break;
+      case DSX_F32:
+        ((float*)M.dst[ci])[o] =
+            missing ? __builtin_nanf("") : ((const float*)M.src[ci])[idx];
+        break;
+      default:  // I8 / BOOL8
+        ((int8_t*)M.dst[ci])[o] = missing ? 0 : ((const int8_t*)M.src[ci])[idx];
+    }
+    if (M.dstv[ci])
+      M.dstv[ci][o] = missing ? 0 : (M.srcv[ci] ? M.srcv[ci][idx] : 1);
+  }
+}
+
+__global__ void k_hash_probe_mat(const uint64_t* codes,
+                                 const uint8_t* validity, int64_t n,
+                                 const uint64_t* tkeys, const uint32_t* tvals,
+                                 int64_t mask, int join_type, int packed,
+                                 const int64_t* block_counts,
+                                 const uint32_t* cache_slot,
+                                 const uint32_t* cache_cnt,
+                                 const JoinMatArg* Mp, int64_t total,
+                                 unsigned int* dbg) {
+  __shared__ unsigned long long s_cnt;
+  if (threadIdx.x == 0) s_cnt = (unsigned long long)block_counts[blockIdx.x];
+  __syncthreads();
+  const JoinMatArg& M = *Mp;
+  int64_t lo, hi;
+  block_range(n, 1, lo, hi);
+  for (int64_t r = lo + threadIdx.x; r < hi; r += BLOCK) {
+    uint32_t nmatch = cache_cnt[r];
+    uint32_t first_s = cache_slot[r];
+    if (nmatch > 0 &&
+        (join_type == DSX_JOIN_INNER || join_type == DSX_JOIN_LEFT)) {
+      unsigned long long o = atomicAdd(&s_cnt, (unsigned long long)nmatch);
+      if (o + nmatch > (unsigned long long)total) {
+        atomicOr(dbg, 2u);
+      } else if (nmatch == 1) {
+        jm_write(M, (int64_t)o, r, first_s);  // cache holds the build id
+      } else {
+        uint64_t cde = codes[r];
+        int64_t s = (int64_t)first_s;
+        uint32_t emitted = 0;
+        while (emitted < nmatch) {
+          uint64_t k = tkeys[s];
+          uint64_t kc = packed ? (k >> 32) : k;
+          if (k == EMPTY_KEY) break;
+          if (kc == cde) {
+            jm_write(M, (int64_t)(o + emitted), r,
+                     packed ? (uint32_t)k : tvals[s]);
+            emitted++;
+          }
+          s = (s + 1) & mask;
+        }
+      }
+    } else if (nmatch == 0 && (join_type == DSX_JOIN_LEFT ||
+                               join_type == DSX_JOIN_LEFTANTI)) {
+      unsigned long long o = atomicAdd(&s_cnt, 1ull);
+      if (o >= (unsigned long long)total) atomicOr(dbg, 2u);
+      else jm_write(M, (int64_t)o, r, DSX_NULL_IDX);
+    } else if (nmatch > 0 && join_type == DSX_JOIN_LEFTSEMI) {
+      unsigned long long o = atomicAdd(&s_cnt, 1ull);
+      if (o >= (unsigned long long)total) atomicOr(dbg, 2u);
+      else jm_write(M, (int64_t)o, r, first_s);
+    }
+  }
+}
+
+extern "C" int dsx_hash_probe_cols(
+    DsxCtx* c, DsxHashTable* t, const uint64_t* codes,
+    const uint8_t* validity, int64_t n, int join_type,
+    const DsxColumn* pcols, int n_pcols, const DsxColumn* bcols, int n_bcols,
+    int force_build_validity, void** out_datas, uint8_t** out_valids,
+    int64_t* out_count) {
+  if (n > 0xFFFFFFFEll) FAIL(-3, "probe side too large for u32 row ids");
+  int ncols = n_pcols + n_bcols;
+  if (ncols > 16) FAIL(-3, "too many join output columns for fused emit");
+  *out_count = 0;
+  int grid = (int)min((int64_t)MAX_GRID, (n + BLOCK - 1) / BLOCK);
+  int64_t arg_off = (grid + 2) * 8 + n * 8 + 64;
+  int rc = ensure_scratch(c, arg_off + (int64_t)sizeof(JoinMatArg) + 64);
+  if (rc) return rc;
+  int64_t* block_counts = (int64_t*)c->scratch;
+  int64_t* d_total = block_counts + grid;
+  uint32_t* cache_slot = (uint32_t*)(d_total + 2);
+  uint32_t* cache_cnt = cache_slot + n;
+  JoinMatArg* d_M = (JoinMatArg*)((char*)c->scratch + ((arg_off + 63) / 64) * 64);
+  if (grid > 0) {
+    ProfScope ps(c, "k_hash_probe_count");
+    hipLaunchKernelGGL(k_hash_probe<0>, dim3(grid), dim3(BLOCK), 0, c->stream,
+                       codes, validity, n, t->keys, t->vals, t->matched,
+                       t->slots - 1, join_type, t->packed, 0,
+                       block_counts, cache_slot, cache_cnt, nullptr, nullptr,
+                       0, c->dbg_flag);
+  }
+  hipLaunchKernelGGL(k_scan_block_counts, dim3(1), dim3(64), 0, c->stream,
+                     block_counts, grid, d_total);
+  int64_t total = 0;
+  HIP_TRY(hipMemcpyAsync(&total, d_total, 8, hipMemcpyDeviceToHost,
+                         c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  JoinMatArg M{};
+  M.ncols = ncols;
+  int64_t tsz = total > 0 ? total : 1;
+  for (int i = 0; i < ncols; i++) {
+    const DsxColumn* src = i < n_pcols ? &pcols[i] : &bcols[i - n_pcols];
+    int side = i < n_pcols ? 0 : 1;
+    int esz = src->dtype == DSX_I64 || src->dtype == DSX_F64 ? 8
+              : src->dtype == DSX_I32 || src->dtype == DSX_F32 ? 4 : 1;
+    int rc2 = pool_alloc(c, tsz * esz, &out_datas[i]);
+    if (rc2) return rc2;
+    bool want_valid = src->validity != nullptr ||
+                      (side == 1 && force_build_validity);
+    out_valids[i] = nullptr;
+    if (want_valid) {
+      void* vp = nullptr;
+      rc2 = pool_alloc(c, tsz, &vp);
+      if (rc2) return rc2;
+      out_valids[i] = (uint8_t*)vp;
+    }
+    M.src[i] = src->data;
+    M.srcv[i] = src->validity;
+    M.dst[i] = out_datas[i];
+    M.dstv[i] = out_valids[i];
+    M.dtype[i] = src->dtype;
+    M.side[i] = side;
+  }
+  HIP_TRY(hipMemcpyAsync(d_M, &M, sizeof(JoinMatArg), hipMemcpyHostToDevice,
+                         c->stream));
+  if (grid > 0 && total > 0) {
+    ProfScope ps(c, "k_hash_probe_mat");
+    hipLaunchKernelGGL(k_hash_probe_mat, dim3(grid), dim3(BLOCK), 0,
+                       c->stream, codes, validity, n, t->keys, t->vals,
+                       t->slots - 1, join_type, t->packed, block_counts,
+                       cache_slot, cache_cnt, d_M, total, c->dbg_flag);
+  }
+  HIP_TRY(hipGetLastError());
+  *out_count = total;
+  return dbg_check(c, "dsx_hash_probe_cols");
+}
+
 template <int PASS>
 __global__ void k_unmatched(const uint64_t* tkeys, const uint32_t* tvals,
                             const uint32_t* matched, int64_t slots,

@@ -204,23 +204,6 @@ class DaskJoinPlugin(BaseRelPlugin):
             lhs_on, rhs_on, residual = self._split_join_condition(
                 condition, n_lhs_cols)
 
-        if lhs_on:
-            if residual and join_type == "leftanti":
-                # ANTI with extra condition: keep lhs rows with NO rhs match
-                # satisfying key AND residual — inner pairs → residual filter
-                # → complement of surviving probe ids (join.py:169-181
-                # composed with the anti semantics of :78-90)
-                pairs, n_out = self._anti_residual(
-                    runtime, dc_lhs, dc_rhs, lhs_on, rhs_on, residual,
-                    cc_lhs, cc_rhs, n_lhs_cols)
-                residual = []
-            else:
-                pairs, n_out = self._equi_join(runtime, dc_lhs, dc_rhs,
-                                               lhs_on, rhs_on, join_type)
-        else:
-            pairs, n_out = self._cross_join(runtime, dc_lhs, dc_rhs, join_type)
-        probe_sel, build_sel = pairs
-
         # materialize only columns the plan consumes (output_indices from the
         # pruning pass) plus residual-referenced temporaries
         force_l = join_type in ("outer", "right")
@@ -238,21 +221,48 @@ class DaskJoinPlugin(BaseRelPlugin):
             _expr_refs(r, res_refs)
         mat_idx = sorted(set(out_idx) | res_refs)
 
-        gathered = {}
-        for i in mat_idx:
-            side, frontend = combined[i]
-            if side == "l":
-                col = dc_lhs.table.col(
-                    cc_lhs.get_backend_by_frontend_name(frontend))
-                g = runtime.gather(col, probe_sel.data, n_out, force_l)
+        # FUSED PATH: emit writes the output columns directly — no pair
+        # vectors, no per-column gathers (k_hash_probe_mat)
+        gathered = None
+        if (lhs_on and not residual and len(mat_idx) <= 16
+                and join_type in ("inner", "left", "right", "leftanti")):
+            gathered, n_out = self._equi_join_fused(
+                runtime, dc_lhs, dc_rhs, lhs_on, rhs_on, join_type,
+                combined, mat_idx, cc_lhs, cc_rhs)
+
+        if gathered is None:
+            if lhs_on:
+                if residual and join_type == "leftanti":
+                    # ANTI with extra condition: keep lhs rows with NO rhs
+                    # match satisfying key AND residual (join.py:169-181
+                    # composed with the anti semantics of :78-90)
+                    pairs, n_out = self._anti_residual(
+                        runtime, dc_lhs, dc_rhs, lhs_on, rhs_on, residual,
+                        cc_lhs, cc_rhs, n_lhs_cols)
+                    residual = []
+                else:
+                    pairs, n_out = self._equi_join(runtime, dc_lhs, dc_rhs,
+                                                   lhs_on, rhs_on, join_type)
             else:
-                col = dc_rhs.table.col(
-                    cc_rhs.get_backend_by_frontend_name(frontend))
-                g = runtime.gather(col, build_sel.data, n_out, force_r)
-            if getattr(col, "dictionary", None) is not None:
-                g.dictionary = col.dictionary
-            g._stats_src = col
-            gathered[i] = g
+                pairs, n_out = self._cross_join(runtime, dc_lhs, dc_rhs,
+                                                join_type)
+            probe_sel, build_sel = pairs
+
+            gathered = {}
+            for i in mat_idx:
+                side, frontend = combined[i]
+                if side == "l":
+                    col = dc_lhs.table.col(
+                        cc_lhs.get_backend_by_frontend_name(frontend))
+                    g = runtime.gather(col, probe_sel.data, n_out, force_l)
+                else:
+                    col = dc_rhs.table.col(
+                        cc_rhs.get_backend_by_frontend_name(frontend))
+                    g = runtime.gather(col, build_sel.data, n_out, force_r)
+                if getattr(col, "dictionary", None) is not None:
+                    g.dictionary = col.dictionary
+                g._stats_src = col
+                gathered[i] = g
 
         # residual filter (join.py:169-181) over the combined row, with
         # InputRefs densified to the materialized set
@@ -502,6 +512,87 @@ class DaskJoinPlugin(BaseRelPlugin):
         if swap:
             probe_sel, build_sel = build_sel, probe_sel
         return (probe_sel, build_sel), count
+
+    def _equi_join_fused(self, runtime, dc_lhs, dc_rhs, lhs_on, rhs_on,
+                         join_type, combined, mat_idx, cc_lhs, cc_rhs):
+        """Equi join with fused emit+materialization (dsx_hash_probe_cols):
+        the probe's emit pass writes the output columns directly. Returns
+        ({combined_idx: DeviceColumn}, n_out) or (None, 0) on unsupported
+        shapes (caller falls back to the pair path)."""
+        lcols = dc_lhs.backend_cols()
+        rcols = dc_rhs.backend_cols()
+        for i in lhs_on:
+            if lcols[i].dtype not in _INT_KINDS:
+                raise RexCompileError("non-integer join keys (round-2)")
+        for i in rhs_on:
+            if rcols[i].dtype not in _INT_KINDS:
+                raise RexCompileError("non-integer join keys (round-2)")
+        kdc_rhs, krcols = self._reconcile_dict_keys(runtime, dc_lhs, dc_rhs,
+                                                    lhs_on, rhs_on)
+        ranges = []
+        for li, ri in zip(lhs_on, rhs_on):
+            lmn, lmx, lnn = _minmax_cached(runtime, lcols[li])
+            rmn, rmx, rnn = _minmax_cached(runtime, krcols[ri])
+            if lnn == 0 and rnn == 0:
+                mn, mx = 0, 0
+            elif lnn == 0:
+                mn, mx = rmn, rmx
+            elif rnn == 0:
+                mn, mx = lmn, lmx
+            else:
+                mn, mx = min(lmn, rmn), max(lmx, rmx)
+            ranges.append((mn, mx - mn + 1))
+
+        swap = join_type == "right" or (
+            join_type == "inner"
+            and dc_lhs.table.num_rows < kdc_rhs.table.num_rows)
+        if swap:
+            probe_kdc, build_kdc = kdc_rhs, dc_lhs
+            probe_on, build_on = rhs_on, lhs_on
+            ktype = rt.JOIN_LEFT if join_type == "right" else rt.JOIN_INNER
+        else:
+            probe_kdc, build_kdc = dc_lhs, kdc_rhs
+            probe_on, build_on = lhs_on, rhs_on
+            ktype = {
+                "inner": rt.JOIN_INNER, "left": rt.JOIN_LEFT,
+                "leftanti": rt.JOIN_LEFTANTI,
+            }[join_type]
+
+        bcodes, bval, bkeep = self._key_codes(runtime, build_kdc, build_on,
+                                              ranges)
+        pcodes, pval, pkeep = self._key_codes(runtime, probe_kdc, probe_on,
+                                              ranges)
+        space = 1
+        for (_, rng) in ranges:
+            space *= rng
+        # assemble per-side materialization lists (sources: ORIGINAL tables)
+        probe_is_l = not swap
+        p_items, b_items = [], []
+        for i in mat_idx:
+            side, frontend = combined[i]
+            if side == "l":
+                col = dc_lhs.table.col(
+                    cc_lhs.get_backend_by_frontend_name(frontend))
+            else:
+                col = dc_rhs.table.col(
+                    cc_rhs.get_backend_by_frontend_name(frontend))
+            on_probe = (side == "l") == probe_is_l
+            (p_items if on_probe else b_items).append((i, col))
+        table = runtime.hash_build(bcodes, bval, code_max=space - 1)
+        try:
+            cols_out, n_out = runtime.hash_probe_cols(
+                table, pcodes, ktype, pval,
+                [c for _, c in p_items], [c for _, c in b_items],
+                force_build_validity=(ktype == rt.JOIN_LEFT))
+        finally:
+            runtime.hash_table_free(table)
+        gathered = {}
+        for (i, src), col in zip(p_items + b_items, cols_out):
+            if getattr(src, "dictionary", None) is not None:
+                col.dictionary = src.dictionary
+            col._stats_src = src
+            gathered[i] = col
+        return gathered, n_out
 
     def _cross_join(self, runtime, dc_lhs, dc_rhs, join_type):
         # reference join.py:133-140 (merge on constant); tiny-only guard

@@ -470,6 +470,33 @@ class Runtime:
         )
         return b.value, count.value
 
+    def hash_probe_cols(self, table, codes: DeviceColumn, join_type,
+                        validity_ptr, pcols, bcols, force_build_validity):
+        """Fused probe-emit + materialization (dsx_hash_probe_cols): returns
+        ([DeviceColumn] ordered pcols then bcols, n_out)."""
+        ncols = len(pcols) + len(bcols)
+        datas = (ct.c_void_p * max(ncols, 1))()
+        valids = (ct.c_void_p * max(ncols, 1))()
+        count = ct.c_int64()
+        _check(
+            self.lib,
+            self.lib.dsx_hash_probe_cols(
+                self.ctx, table, ct.c_void_p(codes.data),
+                ct.c_void_p(validity_ptr) if validity_ptr else None,
+                ct.c_int64(codes.len), ct.c_int(join_type),
+                self._cols_array(pcols), ct.c_int(len(pcols)),
+                self._cols_array(bcols), ct.c_int(len(bcols)),
+                ct.c_int(1 if force_build_validity else 0),
+                datas, valids, ct.byref(count)),
+            "dsx_hash_probe_cols",
+        )
+        n = count.value
+        out = []
+        for i, src in enumerate(list(pcols) + list(bcols)):
+            out.append(DeviceColumn(self, datas[i], valids[i] or None, n,
+                                    src.dtype, owner=True))
+        return out, n
+
     def hash_table_free(self, table):
         self.lib.dsx_hash_table_free(table)
 

@@ -210,6 +210,20 @@ int dsx_hash_probe(DsxCtx* ctx, DsxHashTable* t, const uint64_t* codes,
                    int64_t* out_count);
 /* build rows never matched by any probe since build (for FULL OUTER,
  * join.py JOIN_TYPE_MAPPING "FULL" → outer). */
+/* fused probe-emit + materialization (INNER/LEFT/SEMI/ANTI, no residual):
+ * the emit pass writes the join's output columns directly — one pass, no
+ * (probe,build) pair vectors and no per-column gathers (the fusion of
+ * join.py:241-246 dd.merge's column copy). out_datas/out_valids are caller
+ * arrays of n_pcols+n_bcols slots, filled with pool allocations (free with
+ * dsx_free); out_valids[i] NULL when the source has no validity and no
+ * LEFT NULL-fill applies. */
+int dsx_hash_probe_cols(DsxCtx* ctx, DsxHashTable* t, const uint64_t* codes,
+                        const uint8_t* validity, int64_t n, int join_type,
+                        const DsxColumn* pcols, int n_pcols,
+                        const DsxColumn* bcols, int n_bcols,
+                        int force_build_validity, void** out_datas,
+                        uint8_t** out_valids, int64_t* out_count);
+
 int dsx_hash_unmatched(DsxCtx* ctx, DsxHashTable* t, uint32_t** out_build_idx,
                        int64_t* out_count);
 
