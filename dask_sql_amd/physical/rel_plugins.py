@@ -216,14 +216,11 @@ class DaskJoinPlugin(BaseRelPlugin):
                                                  None) is not None \
             else list(range(len(combined)))
         from dask_sql_amd.planner.prune import _expr_refs, _remap
-        res_refs = set()
-        for r in residual:
-            _expr_refs(r, res_refs)
-        mat_idx = sorted(set(out_idx) | res_refs)
 
         # FUSED PATH: emit writes the output columns directly — no pair
         # vectors, no per-column gathers (k_hash_probe_mat)
         gathered = None
+        mat_idx = sorted(set(out_idx))
         if (lhs_on and not residual and len(mat_idx) <= 16
                 and join_type in ("inner", "left", "right", "leftanti")):
             gathered, n_out = self._equi_join_fused(
@@ -248,6 +245,10 @@ class DaskJoinPlugin(BaseRelPlugin):
                                                 join_type)
             probe_sel, build_sel = pairs
 
+            res_refs = set()
+            for r in residual:
+                _expr_refs(r, res_refs)
+            mat_idx = sorted(set(out_idx) | res_refs)
             gathered = {}
             for i in mat_idx:
                 side, frontend = combined[i]
