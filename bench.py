@@ -54,8 +54,8 @@ WORKLOADS = {
         "sql": "SELECT p.key, p.pv, b.bv FROM probe_t p JOIN build_t b "
                "ON p.key = b.key",
         # 16 B/probe-row + 16 B/build-row + 24 B/match
-        "dominant": ["k_hash_probe_emit", "k_hash_probe_count",
-                     "k_hash_build"],
+        "dominant": ["k_hash_probe_mat", "k_hash_probe_emit",
+                     "k_hash_probe_count", "k_hash_build"],
         "algo_bytes": lambda n, g: 16 * n + 16 * g + 24 * n,
         "scaling": "weak",
     },
@@ -82,7 +82,8 @@ WORKLOADS = {
  AND l_orderkey = o_orderkey AND o_orderdate < 9204 AND l_shipdate > 9204
  GROUP BY l_orderkey, o_orderdate, o_shippriority
  ORDER BY revenue DESC, o_orderdate LIMIT 10""",
-        "dominant": ["k_hash_probe_emit", "k_hash_probe_count"],
+        "dominant": ["k_hash_probe_mat", "k_hash_probe_emit",
+                     "k_hash_probe_count"],
         # dominant scans ≈ lineitem 28 B + orders 24 B + customer 9 B per
         # their own rows; normalized per lineitem row below
         "algo_bytes": lambda n, g: 28 * n + 24 * (n // 4) + 9 * (n // 40),
