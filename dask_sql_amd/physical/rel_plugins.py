@@ -219,9 +219,11 @@ class DaskJoinPlugin(BaseRelPlugin):
 
         # FUSED PATH: emit writes the output columns directly — no pair
         # vectors, no per-column gathers (k_hash_probe_mat)
+        import os as _os
         gathered = None
         mat_idx = sorted(set(out_idx))
         if (lhs_on and not residual and len(mat_idx) <= 16
+                and not _os.environ.get("DSX_DISABLE_JOINFUSE")
                 and join_type in ("inner", "left", "right", "leftanti")):
             gathered, n_out = self._equi_join_fused(
                 runtime, dc_lhs, dc_rhs, lhs_on, rhs_on, join_type,
