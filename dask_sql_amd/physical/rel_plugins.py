@@ -64,7 +64,7 @@ def _gather_table(runtime, dc: DataContainer, sel_ptr, n_sel,
                 g.dictionary = col.dictionary
             g._stats_src = col  # range(subset) ⊆ range(source)
             out_cols[backend] = g
-    return DataContainer(DeviceTable(out_cols), cc)
+    return DataContainer(DeviceTable(out_cols, num_rows=n_sel), cc)
 
 
 def _empty_like(runtime, dc: DataContainer) -> DataContainer:
@@ -165,7 +165,8 @@ class DaskProjectPlugin(BaseRelPlugin):
         cc = ColumnContainer([n for n, _ in new_names],
                              dict(new_names))
         cc = self.fix_column_to_row_type(cc, rel.getRowType())
-        return DataContainer(DeviceTable(out_cols), cc)
+        return DataContainer(DeviceTable(out_cols,
+                                         num_rows=dc.table.num_rows), cc)
 
 
 class DaskJoinPlugin(BaseRelPlugin):
