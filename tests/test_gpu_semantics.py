@@ -812,3 +812,16 @@ def test_in_subquery_semi_anti(ctx):
     got3 = c.sql("SELECT v FROM big WHERE k IN (SELECT id FROM small)"
                  ).compute()
     assert sorted(got3["v"].astype(int).tolist()) == [20, 21, 40]
+
+
+def test_count_star_unfused_path(ctx, monkeypatch):
+    """Regression: COUNT(*) over a fully-pruned (zero-column) projection on
+    the unfused path must keep the filtered row count."""
+    from dask_sql_amd.context import Context
+    monkeypatch.setenv("DSX_DISABLE_FUSED", "1")
+    df = pd.DataFrame({"s": ["a", "b", "a", None, "c"] * 100})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT COUNT(*) AS c FROM t WHERE s NOT LIKE 'a'").compute()
+    exp = ((df["s"] != "a") & df["s"].notna()).sum()
+    assert int(got["c"].iloc[0]) == int(exp)
