@@ -825,3 +825,43 @@ def test_count_star_unfused_path(ctx, monkeypatch):
     got = c.sql("SELECT COUNT(*) AS c FROM t WHERE s NOT LIKE 'a'").compute()
     exp = ((df["s"] != "a") & df["s"].notna()).sum()
     assert int(got["c"].iloc[0]) == int(exp)
+
+
+@pytest.mark.parametrize("seed", [21, 22, 23, 24])
+def test_differential_random_extended(ctx, seed):
+    """Differential sweep over the newer surface: stddev family, window
+    partition aggregates, union, derived tables — vs pandas restatements."""
+    rng = np.random.default_rng(seed)
+    df = _rand_frame(rng, 3000, with_nulls=True)
+    name = f"dx_{seed}"
+    ctx.create_table(name, df)
+
+    # stddev family over groups
+    got = ctx.sql(f"SELECT w, STDDEV(v) AS sd, VAR_POP(v) AS vp FROM {name} "
+                  "GROUP BY w").compute().sort_values("w").reset_index(
+                      drop=True)
+    exp = df.groupby("w").agg(sd=("v", "std"),
+                              vp=("v", lambda s: s.var(ddof=0))
+                              ).reset_index()
+    for col in ("sd", "vp"):
+        g = got[col].to_numpy(np.float64)
+        e = exp[col].to_numpy(np.float64)
+        ok = np.isclose(g, e, rtol=1e-8, equal_nan=True) | (
+            np.isnan(g) & np.isnan(e))
+        assert ok.all(), (seed, col)
+
+    # window partition aggregate == groupby transform
+    got2 = ctx.sql(f"SELECT w, v, AVG(v) OVER (PARTITION BY w) AS a "
+                   f"FROM {name}").compute()
+    exp2 = df.groupby("w")["v"].transform("mean").to_numpy()
+    g2 = got2["a"].to_numpy(np.float64)
+    ok = np.isclose(g2, exp2, rtol=1e-9) | (np.isnan(g2) & np.isnan(exp2))
+    assert ok.all(), seed
+
+    # union all with itself doubles every group count
+    got3 = ctx.sql(
+        f"SELECT w, COUNT(*) AS c FROM "
+        f"(SELECT w FROM {name} UNION ALL SELECT w FROM {name}) u "
+        "GROUP BY w").compute().sort_values("w").reset_index(drop=True)
+    exp3 = df.groupby("w").size() * 2
+    assert got3["c"].astype(int).tolist() == exp3.tolist()
