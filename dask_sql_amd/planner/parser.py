@@ -268,9 +268,12 @@ class Parser:
             # derived table: FROM (SELECT ...) [AS] alias
             self.next()
             sub = self.select_stmt()
+            branches, alls = [sub], []
             while self.accept_kw("UNION"):
-                raise ValueError("UNION inside a derived table: wrap each "
-                                 "branch in its own derived table")
+                alls.append(bool(self.accept_kw("ALL")))
+                branches.append(self.select_stmt())
+            if len(branches) > 1:
+                sub = UnionStmt(branches=branches, alls=alls)
             self.expect_op(")")
             alias = None
             if self.accept_kw("AS"):
