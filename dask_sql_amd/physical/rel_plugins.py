@@ -1422,6 +1422,14 @@ class DaskLimitPlugin(BaseRelPlugin):
 
 
 def _device_topk_impl(context, inp, below, keys, k):
+    """ORDER BY + LIMIT over a large result: device sampled-threshold
+    selection — a strided device sample picks an approximate k-th key,
+    a device filter keeps only rows at or beyond it, and the (tiny)
+    candidate set is sorted exactly on host with full tie-breaking.
+    Replaces the reference's topk_sort nsmallest/nlargest
+    (physical/utils/sort.py:9-34) without downloading the column."""
+    from dask_sql_amd.physical.rex import (OP_GE_F64, OP_GE_I64, OP_LE_F64,
+                                           OP_LE_I64, OP_LIT_F64, OP_NE_F64)
     runtime = context._get_runtime()
     cc = inp.column_container
     n = inp.table.num_rows
@@ -1429,21 +1437,47 @@ def _device_topk_impl(context, inp, below, keys, k):
         return None  # small: plain path is fine
     idx0, asc0, _ = keys[0]
     col0 = inp.table.col(cc.get_backend_by_frontend_name(cc.columns[idx0]))
-    v, valid = col0.to_numpy()
-    if valid is not None and not valid.all():
-        return None  # NULL keys: fall back (NULLS FIRST/LAST ordering)
-    v = np.asarray(v)
-    if v.dtype.kind == "f" and np.isnan(v).any():
-        return None
-    key_arr = v if asc0 else -v.astype(np.float64 if v.dtype.kind == "f"
-                                       else np.int64)
-    part = np.argpartition(key_arr, k - 1)[:k]
-    thresh = key_arr[part].max()
-    cand = np.nonzero(key_arr <= thresh)[0]
-    if len(cand) > max(20 * k, 20_000):
-        return None  # degenerate ties
-    sel = runtime.upload_column(cand.astype(np.uint32), dtype=rt.I32)
-    cand_dc = _gather_table(runtime, inp, sel.data, len(cand))
+    if col0.validity or col0.dtype not in (rt.F64, rt.I64, rt.I32, rt.I8):
+        return None  # NULL keys / unsupported dtype: host fallback
+    isf = col0.dtype == rt.F64
+    if isf:
+        # NaN keys break threshold comparison: cheap device count, fall back
+        pnan = runtime.make_prog([(OP_COL, 0, 0), (OP_COL, 0, 0),
+                                  (OP_NE_F64, 0, 0)])
+        p, cnt = runtime.filter(pnan, [col0], n)
+        runtime.wrap_sel(p, cnt)
+        if cnt:
+            return None
+    # strided device sample → approximate k-th order statistic
+    S = int(min(65536, n))
+    sel = runtime.upload_column(
+        np.linspace(0, n - 1, S).astype(np.uint32), dtype=rt.I32)
+    sv, _ = runtime.gather(col0, sel.data, S).to_numpy()
+    key_s = sv if asc0 else -sv.astype(np.float64 if isf else np.int64)
+    cand_ptr = None
+    cnt = 0
+    r = min(S - 1, max(int(np.ceil(k * S / n * 4)) + 8, k))
+    for _attempt in range(2):
+        thr_key = np.partition(key_s, r)[r]  # native dtype (int precision)
+        thr = thr_key if asc0 else -thr_key
+        if isf:
+            op = OP_LE_F64 if asc0 else OP_GE_F64
+            prog = [(OP_COL, 0, 0), (OP_LIT_F64, 0, float(thr)), (op, 0, 0)]
+        else:
+            op = OP_LE_I64 if asc0 else OP_GE_I64
+            prog = [(OP_COL, 0, 0), (OP_LIT_I64, 0, int(thr)), (op, 0, 0)]
+        cand_ptr, cnt = runtime.filter(runtime.make_prog(prog), [col0], n)
+        if cnt >= k:
+            break
+        runtime.wrap_sel(cand_ptr, cnt)  # free; loosen and retry
+        cand_ptr = None
+        r = min(S - 1, r * 8)
+    if cand_ptr is None or cnt < k or cnt > max(40 * k, 40_000):
+        if cand_ptr is not None:
+            runtime.wrap_sel(cand_ptr, cnt)
+        return None  # sample missed or degenerate ties: host fallback
+    sel2 = runtime.wrap_sel(cand_ptr, cnt)
+    cand_dc = _gather_table(runtime, inp, sel2.data, cnt)
     from dask_sql_amd.materialize import to_pandas
     pdf = to_pandas(cand_dc, context, below.getRowType())
     return _topk(pdf, keys, k)
