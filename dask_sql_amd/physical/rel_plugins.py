@@ -1437,14 +1437,18 @@ def _device_topk_impl(context, inp, below, keys, k):
         return None  # small: plain path is fine
     idx0, asc0, _ = keys[0]
     col0 = inp.table.col(cc.get_backend_by_frontend_name(cc.columns[idx0]))
-    if col0.validity or col0.dtype not in (rt.F64, rt.I64, rt.I32, rt.I8):
-        return None  # NULL keys / unsupported dtype: host fallback
+    if col0.dtype not in (rt.F64, rt.I64, rt.I32, rt.I8):
+        return None  # unsupported key dtype: host fallback
     isf = col0.dtype == rt.F64
-    if isf:
-        # NaN keys break threshold comparison: cheap device count, fall back
-        pnan = runtime.make_prog([(OP_COL, 0, 0), (OP_COL, 0, 0),
-                                  (OP_NE_F64, 0, 0)])
-        p, cnt = runtime.filter(pnan, [col0], n)
+    if isf or col0.validity:
+        # NULL or NaN keys break the threshold compare: one device count,
+        # fall back if any exist (NULLS FIRST/LAST ordering on host)
+        from dask_sql_amd.physical.rex import OP_IS_NULL, OP_OR
+        bad = [(OP_COL, 0, 0), (OP_IS_NULL, 0, 0)]
+        if isf:
+            bad += [(OP_COL, 0, 0), (OP_COL, 0, 0), (OP_NE_F64, 0, 0),
+                    (OP_OR, 0, 0)]
+        p, cnt = runtime.filter(runtime.make_prog(bad), [col0], n)
         runtime.wrap_sel(p, cnt)
         if cnt:
             return None
