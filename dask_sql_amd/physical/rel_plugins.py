@@ -1397,10 +1397,8 @@ class DaskLimitPlugin(BaseRelPlugin):
             keys = below.sort().getCollation()
             from dask_sql_amd.materialize import to_pandas
             if isinstance(inp, DataContainer):
-                # device top-k: download ONLY the primary sort key, select
-                # candidates (k + boundary ties) on host, gather just those
-                # rows on device — avoids materializing the whole G-row
-                # frame (was the entire Q3 host cost)
+                # device top-k: sampled threshold + device filter keep only
+                # ~k candidate rows; the full frame never leaves the GPU
                 pdf = _device_topk_impl(context, inp, below, keys,
                                         node.fetch)
                 if pdf is not None:

@@ -865,3 +865,26 @@ def test_differential_random_extended(ctx, seed):
         "GROUP BY w").compute().sort_values("w").reset_index(drop=True)
     exp3 = df.groupby("w").size() * 2
     assert got3["c"].astype(int).tolist() == exp3.tolist()
+
+
+def test_topk_sampled_at_scale(ctx):
+    """ORDER BY + LIMIT over a large frame takes the sampled-threshold
+    device path; exact against numpy, including boundary ties and DESC."""
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(55)
+    n = 500_000
+    df = pd.DataFrame({
+        "v": np.round(rng.random(n) * 1000, 1),  # ties at 0.1 granularity
+        "t": rng.integers(0, 1_000_000, n).astype(np.int64),
+    })
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT v, t FROM t ORDER BY v DESC, t LIMIT 25").compute()
+    exp = df.sort_values(["v", "t"], ascending=[False, True],
+                         kind="mergesort").head(25)
+    np.testing.assert_allclose(got["v"].to_numpy(np.float64),
+                               exp["v"].to_numpy())
+    assert got["t"].astype(np.int64).tolist() == exp["t"].tolist()
+    got2 = c.sql("SELECT t FROM t ORDER BY t LIMIT 7").compute()
+    assert got2["t"].astype(np.int64).tolist() == sorted(
+        df["t"].tolist())[:7]
