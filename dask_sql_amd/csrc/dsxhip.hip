@@ -727,6 +727,156 @@ __global__ void k_filter_emit(const uint64_t* mask, int64_t n,
   }
 }
 
+// fused filter emit + materialization: like k_filter_emit but writes the
+// output columns directly (FilterMatArg reuses JoinMatArg with side=0) —
+// replaces the per-column gathers of `df[cond]` (filter.py:40) with one
+// order-preserving pass.
+struct FilterMatArg {
+  const void* src[16];
+  const uint8_t* srcv[16];
+  void* dst[16];
+  uint8_t* dstv[16];
+  int32_t dtype[16];
+  int32_t ncols;
+};
+
+__device__ __forceinline__ void fm_write(const FilterMatArg& M, int64_t o,
+                                         int64_t r) {
+  for (int ci = 0; ci < M.ncols; ci++) {
+    switch (M.dtype[ci]) {
+      case DSX_I64:
+        ((int64_t*)M.dst[ci])[o] = ((const int64_t*)M.src[ci])[r];
+        break;
+      case DSX_F64:
+        ((double*)M.dst[ci])[o] = ((const double*)M.src[ci])[r];
+        break;
+      case DSX_I32:
+        ((int32_t*)M.dst[ci])[o] = ((const int32_t*)M.src[ci])[r];
+        break;
+      case DSX_F32:
+        ((float*)M.dst[ci])[o] = ((const float*)M.src[ci])[r];
+        break;
+      default:
+        ((int8_t*)M.dst[ci])[o] = ((const int8_t*)M.src[ci])[r];
+    }
+    if (M.dstv[ci]) M.dstv[ci][o] = M.srcv[ci] ? M.srcv[ci][r] : 1;
+  }
+}
+
+__global__ void k_filter_emit_cols(const uint64_t* mask, int64_t n,
+                                   const int64_t* block_offsets,
+                                   const FilterMatArg* Mp) {
+  int64_t lo, hi;
+  block_range(n, 64, lo, hi);
+  __shared__ int64_t s_prefix[BLOCK];
+  __shared__ int64_t s_running;
+  if (threadIdx.x == 0) s_running = block_offsets[blockIdx.x];
+  __syncthreads();
+  const FilterMatArg& M = *Mp;
+  int64_t w0 = lo / 64;
+  int64_t nw = (hi - lo + 63) / 64;
+  for (int64_t base = 0; base < nw; base += BLOCK) {
+    int64_t w = w0 + base + threadIdx.x;
+    uint64_t m = (base + threadIdx.x < nw) ? mask[w] : 0;
+    int cnt = __popcll(m);
+    s_prefix[threadIdx.x] = cnt;
+    __syncthreads();
+    for (int d = 1; d < BLOCK; d <<= 1) {
+      int64_t v = (threadIdx.x >= d) ? s_prefix[threadIdx.x - d] : 0;
+      __syncthreads();
+      s_prefix[threadIdx.x] += v;
+      __syncthreads();
+    }
+    int64_t excl = s_prefix[threadIdx.x] - cnt;
+    int64_t off = s_running + excl;
+    while (m) {
+      int b = __ffsll((unsigned long long)m) - 1;
+      fm_write(M, off++, w * 64 + b);
+      m &= m - 1;
+    }
+    __syncthreads();
+    if (threadIdx.x == BLOCK - 1) s_running += s_prefix[threadIdx.x];
+    __syncthreads();
+  }
+}
+
+/* fused filter + materialization: evaluate the predicate, then write the
+ * selected rows of the given columns directly (no selection vector, no
+ * per-column gathers). Outputs pool-allocated like dsx_hash_probe_cols. */
+extern "C" int dsx_filter_cols(DsxCtx* c, const DsxInstr* prog, int prog_len,
+                               const DsxColumn* cols, int ncols, int64_t n,
+                               const DsxColumn* mats, int nmats,
+                               void** out_datas, uint8_t** out_valids,
+                               int64_t* out_count) {
+  if (prog_len > DSX_MAX_PROG || ncols > DSX_MAX_COLS)
+    FAIL(-3, "filter spec too large");
+  if (nmats > 16) FAIL(-3, "too many filter output columns");
+  *out_count = 0;
+  ProgArg P{};
+  memcpy(P.ins, prog, prog_len * sizeof(DsxInstr));
+  P.len = prog_len;
+  ColsArg C{};
+  C.ncols = ncols;
+  for (int i = 0; i < ncols; i++) {
+    C.data[i] = cols[i].data;
+    C.validity[i] = cols[i].validity;
+    C.dtype[i] = cols[i].dtype;
+  }
+  int64_t nwords = (n + 63) / 64;
+  int grid = (int)min((int64_t)MAX_GRID, (nwords + WAVES_PER_BLOCK - 1) /
+                                             WAVES_PER_BLOCK);
+  int64_t arg_off = nwords * 8 + (grid + 2) * 8 + 64;
+  int rc = ensure_scratch(c, arg_off + (int64_t)sizeof(FilterMatArg) + 64);
+  if (rc) return rc;
+  uint64_t* mask = (uint64_t*)c->scratch;
+  int64_t* block_counts = (int64_t*)(mask + nwords);
+  int64_t* total = block_counts + grid;
+  FilterMatArg* d_M =
+      (FilterMatArg*)((char*)c->scratch + ((arg_off + 63) / 64) * 64);
+  if (grid > 0) {
+    ProfScope ps(c, "k_filter_mask");
+    hipLaunchKernelGGL(k_filter_mask, dim3(grid), dim3(BLOCK), 0, c->stream,
+                       P, C, n, mask, block_counts);
+  }
+  hipLaunchKernelGGL(k_scan_block_counts, dim3(1), dim3(1024), 0, c->stream,
+                     block_counts, grid, total);
+  int64_t h_total = 0;
+  HIP_TRY(hipMemcpyAsync(&h_total, total, 8, hipMemcpyDeviceToHost,
+                         c->stream));
+  HIP_TRY(hipStreamSynchronize(c->stream));
+  FilterMatArg M{};
+  M.ncols = nmats;
+  int64_t tsz = h_total > 0 ? h_total : 1;
+  for (int i = 0; i < nmats; i++) {
+    int esz = mats[i].dtype == DSX_I64 || mats[i].dtype == DSX_F64 ? 8
+              : mats[i].dtype == DSX_I32 || mats[i].dtype == DSX_F32 ? 4 : 1;
+    int rc2 = pool_alloc(c, tsz * esz, &out_datas[i]);
+    if (rc2) return rc2;
+    out_valids[i] = nullptr;
+    if (mats[i].validity) {
+      void* vp = nullptr;
+      rc2 = pool_alloc(c, tsz, &vp);
+      if (rc2) return rc2;
+      out_valids[i] = (uint8_t*)vp;
+    }
+    M.src[i] = mats[i].data;
+    M.srcv[i] = mats[i].validity;
+    M.dst[i] = out_datas[i];
+    M.dstv[i] = out_valids[i];
+    M.dtype[i] = mats[i].dtype;
+  }
+  HIP_TRY(hipMemcpyAsync(d_M, &M, sizeof(FilterMatArg), hipMemcpyHostToDevice,
+                         c->stream));
+  if (h_total > 0 && grid > 0) {
+    ProfScope ps(c, "k_filter_emit_cols");
+    hipLaunchKernelGGL(k_filter_emit_cols, dim3(grid), dim3(BLOCK), 0,
+                       c->stream, mask, n, block_counts, d_M);
+  }
+  HIP_TRY(hipGetLastError());
+  *out_count = h_total;
+  return dbg_check(c, "dsx_filter_cols");
+}
+
 extern "C" int dsx_filter(DsxCtx* c, const DsxInstr* prog, int prog_len,
                           const DsxColumn* cols, int ncols, int64_t n,
                           uint32_t** out_sel, int64_t* out_count) {

@@ -73,12 +73,32 @@ def _empty_like(runtime, dc: DataContainer) -> DataContainer:
 
 
 def _apply_filter(runtime, dc: DataContainer, condition) -> DataContainer:
-    """filter_or_scalar semantics (reference filter.py:20-45)."""
+    """filter_or_scalar semantics (reference filter.py:20-45). Fused path:
+    the filter's emit pass writes the surviving rows of every column
+    directly (dsx_filter_cols) — no selection vector, no per-column
+    gathers."""
     s = scalar_literal(condition)
     if s is not None:
         return dc if s else _empty_like(runtime, dc)
     cols = dc.backend_cols()
     prog, kind = compile_expr(condition, cols, _dicts_of(cols))
+    cc = dc.column_container
+    backends = []
+    for f in cc.columns:
+        b = cc.get_backend_by_frontend_name(f)
+        if b not in backends:
+            backends.append(b)
+    mats = [dc.table.col(b) for b in backends]
+    if len(mats) <= 16:
+        out_list, count = runtime.filter_cols(runtime.make_prog(prog), cols,
+                                              dc.table.num_rows, mats)
+        out_cols = {}
+        for b, src, col in zip(backends, mats, out_list):
+            if getattr(src, "dictionary", None) is not None:
+                col.dictionary = src.dictionary
+            col._stats_src = src
+            out_cols[b] = col
+        return DataContainer(DeviceTable(out_cols, num_rows=count), cc)
     sel_ptr, count = runtime.filter(runtime.make_prog(prog), cols,
                                     dc.table.num_rows)
     sel = runtime.wrap_sel(sel_ptr, count)  # owns the library buffer

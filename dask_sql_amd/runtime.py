@@ -470,6 +470,28 @@ class Runtime:
         )
         return b.value, count.value
 
+    def filter_cols(self, prog, cols, n, mats):
+        """Fused filter + materialization (dsx_filter_cols): returns
+        ([DeviceColumn per mat], n_out)."""
+        parr, plen = prog
+        nm = len(mats)
+        datas = (ct.c_void_p * max(nm, 1))()
+        valids = (ct.c_void_p * max(nm, 1))()
+        count = ct.c_int64()
+        _check(
+            self.lib,
+            self.lib.dsx_filter_cols(
+                self.ctx, parr, ct.c_int(plen), self._cols_array(cols),
+                ct.c_int(len(cols)), ct.c_int64(n),
+                self._cols_array(mats), ct.c_int(nm), datas, valids,
+                ct.byref(count)),
+            "dsx_filter_cols",
+        )
+        nr = count.value
+        return [DeviceColumn(self, datas[i], valids[i] or None, nr,
+                             m.dtype, owner=True)
+                for i, m in enumerate(mats)], nr
+
     def hash_probe_cols(self, table, codes: DeviceColumn, join_type,
                         validity_ptr, pcols, bcols, force_build_validity):
         """Fused probe-emit + materialization (dsx_hash_probe_cols): returns

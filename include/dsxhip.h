@@ -138,6 +138,16 @@ int dsx_filter(DsxCtx* ctx, const DsxInstr* prog, int prog_len,
                const DsxColumn* cols, int ncols, int64_t n,
                uint32_t** out_sel, int64_t* out_count);
 
+/* fused filter + materialization: evaluate the predicate and write the
+ * selected rows of `mats` directly, order-preserving — `df[cond]`
+ * (filter.py:40) in one pass, with no selection vector and no per-column
+ * gathers. out_datas/out_valids are caller arrays of nmats slots, filled
+ * with pool allocations (free with dsx_free). */
+int dsx_filter_cols(DsxCtx* ctx, const DsxInstr* prog, int prog_len,
+                    const DsxColumn* cols, int ncols, int64_t n,
+                    const DsxColumn* mats, int nmats, void** out_datas,
+                    uint8_t** out_valids, int64_t* out_count);
+
 /* boolean-mask take / merge materialization: out[i] = col[sel[i]].
  * out caller-allocated. Gathers validity too when both non-NULL. */
 int dsx_gather(DsxCtx* ctx, const DsxColumn* col, const uint32_t* sel,
