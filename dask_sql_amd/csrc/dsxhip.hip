@@ -770,9 +770,13 @@ __global__ void k_filter_emit_cols(const uint64_t* mask, int64_t n,
   block_range(n, 64, lo, hi);
   __shared__ int64_t s_prefix[BLOCK];
   __shared__ int64_t s_running;
-  if (threadIdx.x == 0) s_running = block_offsets[blockIdx.x];
+  __shared__ FilterMatArg s_M;  // LDS copy (see k_hash_probe_mat note)
+  if (threadIdx.x == 0) {
+    s_running = block_offsets[blockIdx.x];
+    s_M = *Mp;
+  }
   __syncthreads();
-  const FilterMatArg& M = *Mp;
+  const FilterMatArg& M = s_M;
   int64_t w0 = lo / 64;
   int64_t nw = (hi - lo + 63) / 64;
   for (int64_t base = 0; base < nw; base += BLOCK) {
@@ -1539,9 +1543,16 @@ __global__ void k_hash_probe_mat(const uint64_t* codes,
                                  const JoinMatArg* Mp, int64_t total,
                                  unsigned int* dbg) {
   __shared__ unsigned long long s_cnt;
-  if (threadIdx.x == 0) s_cnt = (unsigned long long)block_counts[blockIdx.x];
+  // LDS copy: through the global pointer the compiler must assume the
+  // output writes alias the arg block and re-loads every pointer per
+  // write; staged in LDS they load once (measured 4×)
+  __shared__ JoinMatArg s_M;
+  if (threadIdx.x == 0) {
+    s_cnt = (unsigned long long)block_counts[blockIdx.x];
+    s_M = *Mp;
+  }
   __syncthreads();
-  const JoinMatArg& M = *Mp;
+  const JoinMatArg& M = s_M;
   int64_t lo, hi;
   block_range(n, 1, lo, hi);
   for (int64_t r = lo + threadIdx.x; r < hi; r += BLOCK) {

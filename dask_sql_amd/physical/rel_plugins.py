@@ -1447,7 +1447,8 @@ def _device_topk_impl(context, inp, below, keys, k):
     Replaces the reference's topk_sort nsmallest/nlargest
     (physical/utils/sort.py:9-34) without downloading the column."""
     from dask_sql_amd.physical.rex import (OP_GE_F64, OP_GE_I64, OP_LE_F64,
-                                           OP_LE_I64, OP_LIT_F64, OP_NE_F64)
+                                           OP_LE_I64, OP_LIT_F64, OP_LIT_I64,
+                                           OP_NE_F64)
     runtime = context._get_runtime()
     cc = inp.column_container
     n = inp.table.num_rows
