@@ -766,10 +766,14 @@ __device__ __forceinline__ void fm_write(const FilterMatArg& M, int64_t o,
 __global__ void k_filter_emit_cols(const uint64_t* mask, int64_t n,
                                    const int64_t* block_offsets,
                                    const FilterMatArg* Mp) {
+  // one row per lane, wave-ballot compaction: selected lanes of a wave
+  // write CONSECUTIVE output slots in one store instruction (the
+  // bit-serial per-thread walk scattered every wave store across ~64
+  // cache lines — measured 1.7 ms vs this version on the Q3 filters)
   int64_t lo, hi;
-  block_range(n, 64, lo, hi);
-  __shared__ int64_t s_prefix[BLOCK];
+  block_range(n, 64, lo, hi);  // 64-aligned lo: a wave shares one word
   __shared__ int64_t s_running;
+  __shared__ int s_wave[BLOCK / 64 + 1];
   __shared__ FilterMatArg s_M;  // LDS copy (see k_hash_probe_mat note)
   if (threadIdx.x == 0) {
     s_running = block_offsets[blockIdx.x];
@@ -777,29 +781,32 @@ __global__ void k_filter_emit_cols(const uint64_t* mask, int64_t n,
   }
   __syncthreads();
   const FilterMatArg& M = s_M;
-  int64_t w0 = lo / 64;
-  int64_t nw = (hi - lo + 63) / 64;
-  for (int64_t base = 0; base < nw; base += BLOCK) {
-    int64_t w = w0 + base + threadIdx.x;
-    uint64_t m = (base + threadIdx.x < nw) ? mask[w] : 0;
-    int cnt = __popcll(m);
-    s_prefix[threadIdx.x] = cnt;
+  int lane = threadIdx.x & 63;
+  int wid = threadIdx.x >> 6;
+  for (int64_t base = lo; base < hi; base += BLOCK) {
+    int64_t r = base + threadIdx.x;
+    bool sel = false;
+    if (r < hi) sel = (mask[r >> 6] >> (r & 63)) & 1;
+    uint64_t ball = __ballot(sel);
+    if (lane == 0) s_wave[wid] = __popcll(ball);
     __syncthreads();
-    for (int d = 1; d < BLOCK; d <<= 1) {
-      int64_t v = (threadIdx.x >= d) ? s_prefix[threadIdx.x - d] : 0;
-      __syncthreads();
-      s_prefix[threadIdx.x] += v;
-      __syncthreads();
-    }
-    int64_t excl = s_prefix[threadIdx.x] - cnt;
-    int64_t off = s_running + excl;
-    while (m) {
-      int b = __ffsll((unsigned long long)m) - 1;
-      fm_write(M, off++, w * 64 + b);
-      m &= m - 1;
+    if (threadIdx.x == 0) {
+      int run = 0;
+      for (int i = 0; i < BLOCK / 64; i++) {
+        int v = s_wave[i];
+        s_wave[i] = run;
+        run += v;
+      }
+      s_wave[BLOCK / 64] = run;
     }
     __syncthreads();
-    if (threadIdx.x == BLOCK - 1) s_running += s_prefix[threadIdx.x];
+    if (sel) {
+      int64_t o = s_running + s_wave[wid] +
+                  __popcll(ball & ((1ull << lane) - 1));
+      fm_write(M, o, r);
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) s_running += s_wave[BLOCK / 64];
     __syncthreads();
   }
 }
