@@ -1255,14 +1255,16 @@ struct DsxHashTable {
   int64_t n_build = 0;
   int packed = 0;             // codes < 2^32-1 → one 8-B entry per slot,
                               // single random read per probe
+  unsigned int* dup = nullptr;  // device flag: any duplicate build key
   DsxCtx* ctx = nullptr;
 };
 
 __global__ void k_hash_build(const uint64_t* codes, const uint8_t* validity,
                              int64_t n, uint64_t* tkeys, uint32_t* tvals,
-                             int64_t mask, int packed) {
+                             int64_t mask, int packed, unsigned int* dup) {
   int64_t lo, hi;
   block_range(n, 1, lo, hi);
+  bool saw_dup = false;
   for (int64_t r = lo + threadIdx.x; r < hi; r += BLOCK) {
     if (validity && !validity[r]) continue;  // NULL-key drop (join.py:202-213)
     uint64_t cde = codes[r];
@@ -1278,9 +1280,12 @@ __global__ void k_hash_build(const uint64_t* codes, const uint8_t* validity,
                                    // end-of-kernel release (same stream)
         break;
       }
+      // duplicate build keys disable the single-pass probe (PK-FK check)
+      if ((packed ? (old >> 32) : old) == cde) saw_dup = true;
       s = (s + 1) & mask;
     }
   }
+  if (saw_dup) atomicOr(dup, 1u);
 }
 
 extern "C" int dsx_hash_build(DsxCtx* c, const uint64_t* codes,
@@ -1302,18 +1307,20 @@ extern "C" int dsx_hash_build(DsxCtx* c, const uint64_t* codes,
   t->packed = (code_max != 0 && code_max < 0xFFFFFFFEull) ? 1 : 0;
   if (pool_alloc(c, slots * 8, (void**)&t->keys) ||
       (!t->packed && pool_alloc(c, slots * 4, (void**)&t->vals)) ||
-      pool_alloc(c, slots * 4, (void**)&t->matched)) {
+      pool_alloc(c, slots * 4, (void**)&t->matched) ||
+      pool_alloc(c, 4, (void**)&t->dup)) {
     dsx_hash_table_free(t);
     FAIL(-2, "hash table alloc failed (%lld slots)", (long long)slots);
   }
   HIP_TRY(hipMemsetAsync(t->keys, 0xFF, slots * 8, c->stream));
   HIP_TRY(hipMemsetAsync(t->matched, 0, slots * 4, c->stream));
+  HIP_TRY(hipMemsetAsync(t->dup, 0, 4, c->stream));
   int grid = (int)min((int64_t)MAX_GRID, (n + BLOCK - 1) / BLOCK);
   if (grid > 0) {
     ProfScope ps(c, "k_hash_build");
     hipLaunchKernelGGL(k_hash_build, dim3(grid), dim3(BLOCK), 0, c->stream,
                        codes, validity, n, t->keys, t->vals, slots - 1,
-                       t->packed);
+                       t->packed, t->dup);
   }
   HIP_TRY(hipGetLastError());
   *out = t;
@@ -1326,6 +1333,7 @@ extern "C" void dsx_hash_table_free(DsxHashTable* t) {
     pool_release(t->ctx, t->keys);
     pool_release(t->ctx, t->vals);
     pool_release(t->ctx, t->matched);
+    pool_release(t->ctx, t->dup);
   }
   delete t;
 }
@@ -1601,6 +1609,64 @@ __global__ void k_hash_probe_mat(const uint64_t* codes,
   }
 }
 
+__global__ void k_hash_probe_mat_1pass(
+    const uint64_t* codes, const uint8_t* validity, int64_t n,
+    const uint64_t* tkeys, const uint32_t* tvals, int64_t mask, int packed,
+    const JoinMatArg* Mp, unsigned long long* counter) {
+  // unique build keys + INNER: probe and emit in ONE pass — no count
+  // pass, no slot caches. Wave-ballot compaction gives dense, coalesced
+  // writes; output order is nondeterministic across blocks (SQL imposes
+  // none; the pair path stays available under DSX_DISABLE_1PASS).
+  __shared__ JoinMatArg s_M;
+  __shared__ int s_wave[BLOCK / 64 + 1];
+  __shared__ unsigned long long s_base;
+  if (threadIdx.x == 0) s_M = *Mp;
+  __syncthreads();
+  const JoinMatArg& M = s_M;
+  int lane = threadIdx.x & 63;
+  int wid = threadIdx.x >> 6;
+  int64_t lo, hi;
+  block_range(n, 1, lo, hi);
+  for (int64_t base = lo; base < hi; base += BLOCK) {
+    int64_t r = base + threadIdx.x;
+    bool found = false;
+    uint32_t bid = 0;
+    if (r < hi && !(validity && !validity[r])) {
+      uint64_t cde = codes[r];
+      int64_t s = (int64_t)(mix64(cde) & mask);
+      while (true) {
+        uint64_t k = tkeys[s];
+        if (k == EMPTY_KEY) break;
+        if ((packed ? (k >> 32) : k) == cde) {
+          found = true;
+          bid = packed ? (uint32_t)k : tvals[s];
+          break;
+        }
+        s = (s + 1) & mask;
+      }
+    }
+    uint64_t ball = __ballot(found);
+    if (lane == 0) s_wave[wid] = __popcll(ball);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      int run = 0;
+      for (int i = 0; i < BLOCK / 64; i++) {
+        int v = s_wave[i];
+        s_wave[i] = run;
+        run += v;
+      }
+      s_base = run ? atomicAdd(counter, (unsigned long long)run) : 0;
+    }
+    __syncthreads();
+    if (found) {
+      int64_t o = (int64_t)(s_base + s_wave[wid] +
+                            __popcll(ball & ((1ull << lane) - 1)));
+      jm_write(M, o, r, bid);
+    }
+    __syncthreads();
+  }
+}
+
 extern "C" int dsx_hash_probe_cols(
     DsxCtx* c, DsxHashTable* t, const uint64_t* codes,
     const uint8_t* validity, int64_t n, int join_type,
@@ -1620,6 +1686,56 @@ extern "C" int dsx_hash_probe_cols(
   uint32_t* cache_slot = (uint32_t*)(d_total + 2);
   uint32_t* cache_cnt = cache_slot + n;
   JoinMatArg* d_M = (JoinMatArg*)((char*)c->scratch + ((arg_off + 63) / 64) * 64);
+  // unique build keys + INNER → single-pass probe+emit (no count pass)
+  static const bool no_1pass = getenv("DSX_DISABLE_1PASS") != nullptr;
+  unsigned int h_dup = 1;
+  if (!no_1pass && join_type == DSX_JOIN_INNER) {
+    HIP_TRY(hipMemcpyAsync(&h_dup, t->dup, 4, hipMemcpyDeviceToHost,
+                           c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+  }
+  if (!no_1pass && join_type == DSX_JOIN_INNER && h_dup == 0) {
+    JoinMatArg M{};
+    M.ncols = ncols;
+    for (int i = 0; i < ncols; i++) {
+      const DsxColumn* src = i < n_pcols ? &pcols[i] : &bcols[i - n_pcols];
+      int side = i < n_pcols ? 0 : 1;
+      int esz = src->dtype == DSX_I64 || src->dtype == DSX_F64 ? 8
+                : src->dtype == DSX_I32 || src->dtype == DSX_F32 ? 4 : 1;
+      int rc2 = pool_alloc(c, (n > 0 ? n : 1) * esz, &out_datas[i]);
+      if (rc2) return rc2;
+      out_valids[i] = nullptr;
+      if (src->validity) {
+        void* vp = nullptr;
+        rc2 = pool_alloc(c, n > 0 ? n : 1, &vp);
+        if (rc2) return rc2;
+        out_valids[i] = (uint8_t*)vp;
+      }
+      M.src[i] = src->data;
+      M.srcv[i] = src->validity;
+      M.dst[i] = out_datas[i];
+      M.dstv[i] = out_valids[i];
+      M.dtype[i] = src->dtype;
+      M.side[i] = side;
+    }
+    HIP_TRY(hipMemcpyAsync(d_M, &M, sizeof(JoinMatArg),
+                           hipMemcpyHostToDevice, c->stream));
+    HIP_TRY(hipMemsetAsync(d_total, 0, 8, c->stream));
+    if (grid > 0) {
+      ProfScope ps(c, "k_hash_probe_mat");
+      hipLaunchKernelGGL(k_hash_probe_mat_1pass, dim3(grid), dim3(BLOCK), 0,
+                         c->stream, codes, validity, n, t->keys, t->vals,
+                         t->slots - 1, t->packed, d_M,
+                         (unsigned long long*)d_total);
+    }
+    unsigned long long tot = 0;
+    HIP_TRY(hipMemcpyAsync(&tot, d_total, 8, hipMemcpyDeviceToHost,
+                           c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipGetLastError());
+    *out_count = (int64_t)tot;
+    return dbg_check(c, "dsx_hash_probe_cols");
+  }
   if (grid > 0) {
     ProfScope ps(c, "k_hash_probe_count");
     hipLaunchKernelGGL(k_hash_probe<0>, dim3(grid), dim3(BLOCK), 0, c->stream,
