@@ -1686,15 +1686,20 @@ extern "C" int dsx_hash_probe_cols(
   uint32_t* cache_slot = (uint32_t*)(d_total + 2);
   uint32_t* cache_cnt = cache_slot + n;
   JoinMatArg* d_M = (JoinMatArg*)((char*)c->scratch + ((arg_off + 63) / 64) * 64);
-  // unique build keys + INNER → single-pass probe+emit (no count pass)
-  static const bool no_1pass = getenv("DSX_DISABLE_1PASS") != nullptr;
+  // unique build keys + INNER → single-pass probe+emit (no count pass).
+  // MEASURED SLOWER than the 2-pass fused path (C3 5.59 vs 5.29 ms of
+  // kernel time; Q3 1.04 vs 0.59): the chunk-synchronous ballot emission
+  // stalls every 256 rows on the slowest probe chain, losing the
+  // memory-level parallelism of free-running probes. Kept opt-in as the
+  // recorded experiment.
+  static const bool use_1pass = getenv("DSX_ENABLE_1PASS") != nullptr;
   unsigned int h_dup = 1;
-  if (!no_1pass && join_type == DSX_JOIN_INNER) {
+  if (use_1pass && join_type == DSX_JOIN_INNER) {
     HIP_TRY(hipMemcpyAsync(&h_dup, t->dup, 4, hipMemcpyDeviceToHost,
                            c->stream));
     HIP_TRY(hipStreamSynchronize(c->stream));
   }
-  if (!no_1pass && join_type == DSX_JOIN_INNER && h_dup == 0) {
+  if (use_1pass && join_type == DSX_JOIN_INNER && h_dup == 0) {
     JoinMatArg M{};
     M.ncols = ncols;
     for (int i = 0; i < ncols; i++) {
