@@ -72,6 +72,36 @@ def _empty_like(runtime, dc: DataContainer) -> DataContainer:
     return _gather_table(runtime, dc, sel.data, 0)
 
 
+def _resolve_scalar_subs(expr, context):
+    """Replace ScalarSub nodes with the Literal their (eagerly executed)
+    subplan yields — ≤1 row enforced; 0 rows → NULL (SQL scalar subquery)."""
+    from dask_sql_amd.planner.plan import ScalarSub
+    if isinstance(expr, ScalarSub):
+        dc = RelConverter.convert(expr.plan, context=context)
+        from dask_sql_amd.materialize import to_pandas
+        pdf = to_pandas(dc, context, expr.plan.getRowType())
+        if len(pdf) > 1:
+            raise RuntimeError(
+                f"scalar subquery returned {len(pdf)} rows")
+        import pandas as pd
+        v = None if len(pdf) == 0 else pdf.iloc[0, 0]
+        if v is not None and pd.isna(v):
+            v = None
+        if isinstance(v, np.integer):
+            v = int(v)
+        elif isinstance(v, np.floating):
+            v = float(v)
+        elif isinstance(v, np.bool_):
+            v = bool(v)
+        from dask_sql_amd.planner.plan import Literal as PLit
+        return PLit(v, expr.getType())
+    if isinstance(expr, Call):
+        return Call(expr.getOperatorName(),
+                    [_resolve_scalar_subs(o, context)
+                     for o in expr.getOperands()], expr.getType())
+    return expr
+
+
 def _apply_filter(runtime, dc: DataContainer, condition) -> DataContainer:
     """filter_or_scalar semantics (reference filter.py:20-45). Fused path:
     the filter's emit pass writes the surviving rows of every column
@@ -133,7 +163,7 @@ class DaskFilterPlugin(BaseRelPlugin):
 
     def convert(self, rel, context):
         (dc,) = self.assert_inputs(rel, 1, context)
-        condition = rel.filter().getCondition()
+        condition = _resolve_scalar_subs(rel.filter().getCondition(), context)
         dc = _apply_filter(context._get_runtime(), dc, condition)
         cc = self.fix_column_to_row_type(dc.column_container, rel.getRowType())
         return DataContainer(dc.table, cc)
@@ -156,6 +186,8 @@ class DaskProjectPlugin(BaseRelPlugin):
         from dask_sql_amd.physical.rex import dict_string_fn
         for i, (expr, name) in enumerate(named):
             backend_name = f"p{i}__{name}"
+            if not isinstance(expr, InputRef):
+                expr = _resolve_scalar_subs(expr, context)
             sfn = None if isinstance(expr, InputRef) \
                 else dict_string_fn(expr, dicts)
             if isinstance(expr, InputRef):
@@ -852,6 +884,7 @@ class DaskAggregatePlugin(BaseRelPlugin):
                 cond = pred_exprs[0]
                 for p in pred_exprs[1:]:
                     cond = Call("AND", [cond, p])
+                cond = _resolve_scalar_subs(cond, context)
                 prog, _ = compile_expr(cond, base_cols, dicts)
                 pred_prog = runtime.make_prog(prog)
             specs, fins, slab = [], [], []

@@ -165,6 +165,12 @@ class Builder:
                                   _expr_type(ops[1]) if len(ops) > 1
                                   else _expr_type(ops[0]))
             return Call(op, ops, SqlType(ty))
+        if kind == "scalar_sub":
+            from dask_sql_amd.planner.plan import ScalarSub
+            sub = self.build_stmt(ast[1])
+            if len(sub.getRowType().getFieldList()) != 1:
+                raise ValueError("scalar subquery must select one column")
+            return ScalarSub(sub)
         if kind == "agg":
             raise ValueError("aggregate in non-aggregate position")
         raise ValueError(f"cannot resolve {ast!r}")

@@ -414,6 +414,10 @@ class Parser:
         t = self.peek()
         if t == ("op", "("):
             self.next()
+            if self.peek() == ("kw", "SELECT"):
+                sub = self.select_stmt()
+                self.expect_op(")")
+                return ("scalar_sub", sub)
             e = self.expr()
             self.expect_op(")")
             return e

@@ -95,6 +95,20 @@ class InputRef(Expression):
         return f"InputRef({self.index})"
 
 
+class ScalarSub(Expression):
+    """Uncorrelated scalar subquery — resolved to a Literal at convert time
+    (execution is eager; DataFusion folds these on the reference side)."""
+
+    def __init__(self, plan):
+        self.plan = plan
+
+    def getRexType(self):
+        return "RexType.ScalarSubquery"
+
+    def getType(self):
+        return self.plan.getRowType().getFieldList()[0].getType()
+
+
 class Literal(Expression):
     def __init__(self, value, sql_type: SqlType):
         self.value = value

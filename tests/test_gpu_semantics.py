@@ -888,3 +888,25 @@ def test_topk_sampled_at_scale(ctx):
     got2 = c.sql("SELECT t FROM t ORDER BY t LIMIT 7").compute()
     assert got2["t"].astype(np.int64).tolist() == sorted(
         df["t"].tolist())[:7]
+
+
+def test_scalar_subquery(ctx):
+    """Uncorrelated scalar subqueries in WHERE and SELECT, resolved at
+    convert time (DataFusion folds these for the reference)."""
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(61)
+    df = pd.DataFrame({"k": rng.integers(0, 5, 2000).astype(np.int64),
+                       "v": rng.random(2000) * 100})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT COUNT(*) AS c FROM t "
+                "WHERE v > (SELECT AVG(v) FROM t)").compute()
+    assert int(got["c"].iloc[0]) == int((df.v > df.v.mean()).sum())
+    got2 = c.sql("SELECT k, v - (SELECT MIN(v) FROM t) AS dv FROM t "
+                 "LIMIT 5").compute()
+    np.testing.assert_allclose(got2["dv"].to_numpy(np.float64),
+                               (df.v - df.v.min()).to_numpy()[:5], rtol=1e-12)
+    got3 = c.sql("SELECT COUNT(*) AS c FROM t WHERE v < "
+                 "(SELECT AVG(v) FROM t WHERE k = 99)").compute()
+    # empty subquery -> NULL -> comparison NULL -> no rows
+    assert int(got3["c"].iloc[0]) == 0
