@@ -318,10 +318,10 @@ __device__ __forceinline__ void civil_from_days(int64_t days, int& y, int& m,
 // slots stay in VGPRs — a runtime-indexed local array would spill every
 // push/pop to scratch (cdna_hip_programming.md §5.4 rule 20; measured
 // 224 B/lane of scratch before this change). Host compiler enforces
-// depth ≤ 6 (runtime.py make_prog).
+// depth ≤ 8 (runtime.py make_prog).
 struct VmStack {
-  Slot s0, s1, s2, s3, s4, s5;
-  bool v0, v1, v2, v3, v4, v5;
+  Slot s0, s1, s2, s3, s4, s5, s6, s7;
+  bool v0, v1, v2, v3, v4, v5, v6, v7;
   __device__ __forceinline__ void set(int i, Slot s, bool v) {
     switch (i) {
       case 0: s0 = s; v0 = v; break;
@@ -329,7 +329,9 @@ struct VmStack {
       case 2: s2 = s; v2 = v; break;
       case 3: s3 = s; v3 = v; break;
       case 4: s4 = s; v4 = v; break;
-      default: s5 = s; v5 = v; break;
+      case 5: s5 = s; v5 = v; break;
+      case 6: s6 = s; v6 = v; break;
+      default: s7 = s; v7 = v; break;
     }
   }
   __device__ __forceinline__ void get(int i, Slot& s, bool& v) const {
@@ -339,7 +341,9 @@ struct VmStack {
       case 2: s = s2; v = v2; break;
       case 3: s = s3; v = v3; break;
       case 4: s = s4; v = v4; break;
-      default: s = s5; v = v5; break;
+      case 5: s = s5; v = v5; break;
+      case 6: s = s6; v = v6; break;
+      default: s = s7; v = v7; break;
     }
   }
 };

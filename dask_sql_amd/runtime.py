@@ -240,7 +240,7 @@ class Runtime:
     _PUSH_OPS = {1, 2, 3, 4}          # COL, LIT_*
     _BIN_OPS = set(range(10, 42))     # arith/cmp/AND/OR
     _SELECT_OP = 60
-    _MAX_DEPTH = 6
+    _MAX_DEPTH = 8
 
     @classmethod
     def _check_depth(cls, instrs):

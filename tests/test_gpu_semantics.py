@@ -910,3 +910,21 @@ def test_scalar_subquery(ctx):
                  "(SELECT AVG(v) FROM t WHERE k = 99)").compute()
     # empty subquery -> NULL -> comparison NULL -> no rows
     assert int(got3["c"].iloc[0]) == 0
+
+
+def test_deep_expression_vm(ctx):
+    """8-slot VM stack: nested CASE + 4-arg COALESCE compile and evaluate."""
+    from dask_sql_amd.context import Context
+    df = pd.DataFrame({
+        "a": pd.array([1, None, None, None], dtype="Int64"),
+        "b": pd.array([None, 2, None, None], dtype="Int64"),
+        "c": pd.array([None, None, 3, None], dtype="Int64"),
+    })
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT COALESCE(a, b, c, -1) AS x, "
+                "CASE WHEN a IS NULL THEN CASE WHEN b IS NULL THEN "
+                "CASE WHEN c IS NULL THEN 0 ELSE 3 END ELSE 2 END "
+                "ELSE 1 END AS y FROM t").compute()
+    assert got["x"].astype(int).tolist() == [1, 2, 3, -1]
+    assert got["y"].astype(int).tolist() == [1, 2, 3, 0]
