@@ -136,6 +136,8 @@ class RexCompiler:
             return _DSX_KIND[self.cols[i].dtype]
         if isinstance(expr, Literal):
             v = expr.getValue()
+            if isinstance(v, tuple):
+                raise RexCompileError("unfolded INTERVAL literal")
             if v is None:
                 self._emit(OP_LIT_NULL)
                 return KI

@@ -126,6 +126,8 @@ class Builder:
             if t == "NULL":
                 return Literal(None, SqlType("NULL"))
             return Literal(v, SqlType(t))
+        if kind == "interval":
+            return Literal((ast[1], ast[2]), SqlType("INTERVAL"))
         if kind == "cast":
             _, sub, ty = ast
             e = self._resolve(sub, plan)
@@ -143,6 +145,10 @@ class Builder:
         if kind == "call":
             _, op, args = ast
             ops = [self._resolve(a, plan) for a in args]
+            if op in ("+", "-") and any(
+                    isinstance(o, Literal)
+                    and o.getType().getSqlType() == "INTERVAL" for o in ops):
+                return self._date_interval(op, ops)
             if op in ("=", "<>", "<", "<=", ">", ">=", "AND", "OR", "NOT",
                       "IS NULL", "IS NOT NULL", "LIKE"):
                 ty = "BOOLEAN"
@@ -174,6 +180,40 @@ class Builder:
         if kind == "agg":
             raise ValueError("aggregate in non-aggregate position")
         raise ValueError(f"cannot resolve {ast!r}")
+
+    @staticmethod
+    def _date_interval(op, ops):
+        """date ± INTERVAL: DAY/WEEK fold to day-int arithmetic; MONTH/YEAR
+        use exact calendar math on literal dates (the reference gets this
+        from pandas Timestamp + DateOffset, rex/core/call.py datetime ops)."""
+        iv = next(o for o in ops
+                  if isinstance(o, Literal)
+                  and o.getType().getSqlType() == "INTERVAL")
+        other = ops[0] if ops[1] is iv else ops[1]
+        if ops[0] is iv:
+            raise NotImplementedError("INTERVAL on the left of +/-")
+        n_, unit = iv.getValue()
+        sign = 1 if op == "+" else -1
+        if unit in ("DAY", "WEEK"):
+            days = n_ * (7 if unit == "WEEK" else 1)
+            if isinstance(other, Literal):
+                return Literal(int(other.getValue()) + sign * days,
+                               SqlType("DATE"))
+            return Call(op, [other, Literal(days, SqlType("BIGINT"))],
+                        SqlType("DATE"))
+        if not isinstance(other, Literal):
+            raise NotImplementedError(
+                "MONTH/YEAR interval arithmetic on a column")
+        import calendar
+        import datetime
+        d = datetime.date(1970, 1, 1) + datetime.timedelta(
+            days=int(other.getValue()))
+        months = sign * n_ * (12 if unit == "YEAR" else 1)
+        y = d.year + (d.month - 1 + months) // 12
+        m = (d.month - 1 + months) % 12 + 1
+        dd = min(d.day, calendar.monthrange(y, m)[1])
+        nd = datetime.date(y, m, dd)
+        return Literal((nd - datetime.date(1970, 1, 1)).days, SqlType("DATE"))
 
     # -------------------------------------------------------- ast utilities
     @staticmethod

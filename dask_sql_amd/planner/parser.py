@@ -443,6 +443,17 @@ class Parser:
                 if s[0] != "str":
                     raise ValueError("DATE needs a string literal")
                 return ("lit", s[1], "DATE")
+            if t[1] == "INTERVAL":
+                # INTERVAL '<n>' DAY|WEEK|MONTH|YEAR (TPC-H date arithmetic)
+                self.next()
+                s = self.next()
+                if s[0] not in ("str", "num"):
+                    raise ValueError("INTERVAL needs a quantity literal")
+                u = self.next()
+                unit = str(u[1]).upper().rstrip("S")
+                if unit not in ("DAY", "WEEK", "MONTH", "YEAR"):
+                    raise ValueError(f"INTERVAL unit {u[1]!r} not supported")
+                return ("interval", int(str(s[1]).strip()), unit)
             if t[1] == "CASE":
                 return self._case()
             if t[1] == "CAST":

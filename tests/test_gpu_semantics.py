@@ -928,3 +928,27 @@ def test_deep_expression_vm(ctx):
                 "ELSE 1 END AS y FROM t").compute()
     assert got["x"].astype(int).tolist() == [1, 2, 3, -1]
     assert got["y"].astype(int).tolist() == [1, 2, 3, 0]
+
+
+def test_interval_date_arithmetic(ctx):
+    """date ± INTERVAL (TPC-H predicates): DAY folds to day-int arithmetic,
+    MONTH/YEAR use exact calendar math on literal dates."""
+    from dask_sql_amd.context import Context
+    dates = pd.to_datetime(["1998-08-28", "1998-09-03", "1998-12-01",
+                            "1996-02-29"])
+    df = pd.DataFrame({"d": dates, "v": [1, 2, 3, 4]})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT v FROM t WHERE d <= DATE '1998-12-01' - "
+                "INTERVAL '90' DAY").compute()
+    cutoff = pd.Timestamp("1998-12-01") - pd.Timedelta(days=90)
+    assert sorted(got["v"].astype(int).tolist()) == sorted(
+        df[df.d <= cutoff]["v"].tolist())
+    got2 = c.sql("SELECT d + INTERVAL '7' DAY AS d2 FROM t").compute()
+    assert list(got2["d2"]) == list(dates + pd.Timedelta(days=7))
+    got3 = c.sql("SELECT v FROM t WHERE d = DATE '1996-01-31' + "
+                 "INTERVAL '1' MONTH").compute()
+    assert got3["v"].astype(int).tolist() == [4]  # 1996-02-29 (leap clamp)
+    got4 = c.sql("SELECT v FROM t WHERE d >= DATE '1999-09-01' - "
+                 "INTERVAL '1' YEAR").compute()
+    assert sorted(got4["v"].astype(int).tolist()) == [2, 3]
