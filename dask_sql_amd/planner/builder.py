@@ -181,6 +181,58 @@ class Builder:
             raise ValueError("aggregate in non-aggregate position")
         raise ValueError(f"cannot resolve {ast!r}")
 
+    def _decorrelate_exists(self, sub):
+        """Split the EXISTS subquery's WHERE into local conjuncts and
+        equality correlations on OUTER columns (qualified names not bound
+        by the sub's own FROM). Returns (subplan selecting the inner keys,
+        [outer key asts]), or (None, None) when uncorrelated. Only
+        equality correlation with qualified outer references is supported
+        (what DataFusion's decorrelation handles for the reference)."""
+        sub_quals = {(t.alias or t.name).lower() for t in sub.from_tables
+                     if (t.alias or t.name)}
+        sub_quals |= {(j.table.alias or j.table.name).lower()
+                      for j in sub.joins if (j.table.alias or j.table.name)}
+        conjs = self._conjuncts(sub.where)
+
+        def outer_col(x):
+            return (isinstance(x, tuple) and x[0] == "col"
+                    and x[1] is not None and x[1].lower() not in sub_quals)
+
+        def has_outer(ast):
+            return any(q is not None and q.lower() not in sub_quals
+                       for q, _ in self._tables_of(ast))
+
+        local, inner_keys, outer_keys = [], [], []
+        for cj in conjs:
+            if not has_outer(cj):
+                local.append(cj)
+                continue
+            if not (cj[0] == "call" and cj[1] == "=" and len(cj[2]) == 2):
+                raise NotImplementedError(
+                    "EXISTS correlation must be qualified equality "
+                    "(outer.col = inner.col)")
+            a, b = cj[2]
+            if outer_col(a) and not has_outer(b):
+                outer_keys.append(a)
+                inner_keys.append(b)
+            elif outer_col(b) and not has_outer(a):
+                outer_keys.append(b)
+                inner_keys.append(a)
+            else:
+                raise NotImplementedError(
+                    "EXISTS correlation must pair one outer column with an "
+                    "inner expression")
+        if not outer_keys:
+            return None, None
+        where = None
+        for cj in local:
+            where = cj if where is None else ("call", "AND", [where, cj])
+        s2 = SelectStmt(items=[(k, f"ck{i}")
+                               for i, k in enumerate(inner_keys)],
+                        from_tables=sub.from_tables, joins=sub.joins,
+                        where=where)
+        return self.build_stmt(s2), outer_keys
+
     @staticmethod
     def _date_interval(op, ops):
         """date ± INTERVAL: DAY/WEEK fold to day-int arithmetic; MONTH/YEAR
@@ -295,14 +347,16 @@ class Builder:
         # IN (SELECT ...) conjuncts → SEMI/ANTI joins (DataFusion's subquery
         # decorrelation on the reference side). Pulled out before pushdown.
         in_subs = []
+        exists_subs = []
         rest = []
         for cj in where_conjuncts:
-            if cj[0] == "in_sub":
-                in_subs.append((cj[1], cj[2], False))
-            elif (cj[0] == "call" and cj[1] == "NOT"
-                  and isinstance(cj[2][0], tuple)
-                  and cj[2][0][0] == "in_sub"):
-                in_subs.append((cj[2][0][1], cj[2][0][2], True))
+            neg = (cj[0] == "call" and cj[1] == "NOT"
+                   and isinstance(cj[2][0], tuple))
+            base = cj[2][0] if neg else cj
+            if base[0] == "in_sub":
+                in_subs.append((base[1], base[2], neg))
+            elif base[0] == "exists":
+                exists_subs.append((base[1], neg))
             else:
                 rest.append(cj)
         where_conjuncts = rest
@@ -388,6 +442,46 @@ class Builder:
         for jc in stmt.joins:
             rhs = scan_with_filters(jc.table)
             plan = join_plans(plan, rhs, jc.join_type, [], jc.on)
+
+        # EXISTS: equality-correlated → SEMI/ANTI join over the DISTINCT
+        # correlation keys (DataFusion's decorrelation on the reference
+        # side); uncorrelated → COUNT(*) scalar subquery > 0
+        for substmt, negated in exists_subs:
+            subplan, outer_keys = self._decorrelate_exists(substmt)
+            if subplan is None:
+                # uncorrelated EXISTS → scalar COUNT(*) comparison
+                cnt = SelectStmt(
+                    items=[(("agg", "count", [("star",)], False, None),
+                            "c")],
+                    from_tables=substmt.from_tables, joins=substmt.joins,
+                    where=substmt.where)
+                cmp_ast = ("call", "=" if negated else ">",
+                           [("scalar_sub", cnt), ("lit", 0, "BIGINT")])
+                cond = self._resolve(cmp_ast, plan)
+                plan = LogicalPlan("Filter", [plan], plan.getRowType(),
+                                   FilterNode(cond))
+                continue
+            sfields = subplan.getRowType().getFieldList()
+            node = AggregateNode(
+                [InputRef(i, f.getType()) for i, f in enumerate(sfields)],
+                [], distinct_node=True,
+                distinct_columns=[f.getName() for f in sfields])
+            subplan = LogicalPlan("Distinct", [subplan],
+                                  subplan.getRowType(), node)
+            lhs_fields = plan.getRowType().getFieldList()
+            combined = RelDataType(lhs_fields + sfields)
+            tmp = LogicalPlan("__combined__", [], combined, None)
+            cond = None
+            for i, ok_ast in enumerate(outer_keys):
+                eq = Call("=", [self._resolve(ok_ast, tmp),
+                                InputRef(len(lhs_fields) + i,
+                                         sfields[i].getType())],
+                          SqlType("BOOLEAN"))
+                cond = eq if cond is None else Call("AND", [cond, eq],
+                                                    SqlType("BOOLEAN"))
+            jt = "LEFTANTI" if negated else "LEFTSEMI"
+            plan = LogicalPlan("Join", [plan, subplan],
+                               RelDataType(lhs_fields), JoinNode(jt, cond))
 
         # IN-subquery joins: x IN (SELECT c FROM ...) ≡ SEMI join on x = c
         # over the DISTINCT subquery output; NOT IN ≡ ANTI (NULL-in-subquery

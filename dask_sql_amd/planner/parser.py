@@ -57,7 +57,7 @@ KEYWORDS = {
     "RIGHT", "FULL", "OUTER", "SEMI", "ANTI", "CROSS", "ON", "TRUE", "FALSE",
     "NULL", "IS", "IN", "BETWEEN", "LIKE", "CASE", "WHEN", "THEN", "ELSE",
     "END", "CAST", "DATE", "ASC", "DESC", "NULLS", "FIRST", "LAST", "FILTER",
-    "TIMESTAMP", "INTERVAL", "UNION", "ALL", "OVER", "PARTITION",
+    "TIMESTAMP", "INTERVAL", "UNION", "ALL", "OVER", "PARTITION", "EXISTS",
 }
 
 WINDOW_FUNCS = {"ROW_NUMBER", "RANK", "DENSE_RANK", "SUM", "COUNT", "AVG",
@@ -454,6 +454,12 @@ class Parser:
                 if unit not in ("DAY", "WEEK", "MONTH", "YEAR"):
                     raise ValueError(f"INTERVAL unit {u[1]!r} not supported")
                 return ("interval", int(str(s[1]).strip()), unit)
+            if t[1] == "EXISTS":
+                self.next()
+                self.expect_op("(")
+                sub = self.select_stmt()
+                self.expect_op(")")
+                return ("exists", sub)
             if t[1] == "CASE":
                 return self._case()
             if t[1] == "CAST":

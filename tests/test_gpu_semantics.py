@@ -952,3 +952,28 @@ def test_interval_date_arithmetic(ctx):
     got4 = c.sql("SELECT v FROM t WHERE d >= DATE '1999-09-01' - "
                  "INTERVAL '1' YEAR").compute()
     assert sorted(got4["v"].astype(int).tolist()) == [2, 3]
+
+
+def test_exists_correlated_and_not(ctx):
+    """EXISTS / NOT EXISTS with equality correlation -> SEMI/ANTI joins;
+    uncorrelated EXISTS -> scalar COUNT(*) comparison (DataFusion
+    decorrelation on the reference side)."""
+    from dask_sql_amd.context import Context
+    orders = pd.DataFrame({"o_id": [1, 2, 3, 4], "cust": [10, 20, 30, 40]})
+    li = pd.DataFrame({"oid": [1, 1, 3, 3], "qty": [5, 6, 200, 7]})
+    c = Context()
+    c.create_table("orders", orders)
+    c.create_table("li", li)
+    got = c.sql("SELECT o_id FROM orders o WHERE EXISTS "
+                "(SELECT 1 FROM li l WHERE l.oid = o.o_id AND l.qty > 100)"
+                ).compute()
+    assert sorted(got["o_id"].astype(int).tolist()) == [3]
+    got2 = c.sql("SELECT o_id FROM orders o WHERE NOT EXISTS "
+                 "(SELECT 1 FROM li l WHERE l.oid = o.o_id)").compute()
+    assert sorted(got2["o_id"].astype(int).tolist()) == [2, 4]
+    got3 = c.sql("SELECT COUNT(*) AS c FROM orders WHERE EXISTS "
+                 "(SELECT 1 FROM li WHERE qty > 1000)").compute()
+    assert int(got3["c"].iloc[0]) == 0
+    got4 = c.sql("SELECT COUNT(*) AS c FROM orders WHERE NOT EXISTS "
+                 "(SELECT 1 FROM li WHERE qty > 1000)").compute()
+    assert int(got4["c"].iloc[0]) == 4
