@@ -181,7 +181,33 @@ class Builder:
             raise ValueError("aggregate in non-aggregate position")
         raise ValueError(f"cannot resolve {ast!r}")
 
-    def _decorrelate_exists(self, sub):
+    def _semi_anti_join(self, plan, subplan, outer_asts, negated):
+        """SEMI/ANTI join `plan` against the DISTINCT subplan output on
+        positional key equalities (outer_asts resolve against the combined
+        row; sub keys are the subplan's columns in order)."""
+        sfields = subplan.getRowType().getFieldList()
+        node = AggregateNode(
+            [InputRef(i, f.getType()) for i, f in enumerate(sfields)],
+            [], distinct_node=True,
+            distinct_columns=[f.getName() for f in sfields])
+        subplan = LogicalPlan("Distinct", [subplan], subplan.getRowType(),
+                              node)
+        lhs_fields = plan.getRowType().getFieldList()
+        tmp = LogicalPlan("__combined__", [],
+                          RelDataType(lhs_fields + sfields), None)
+        cond = None
+        for i, ast in enumerate(outer_asts):
+            eq = Call("=", [self._resolve(ast, tmp),
+                            InputRef(len(lhs_fields) + i,
+                                     sfields[i].getType())],
+                      SqlType("BOOLEAN"))
+            cond = eq if cond is None else Call("AND", [cond, eq],
+                                                SqlType("BOOLEAN"))
+        jt = "LEFTANTI" if negated else "LEFTSEMI"
+        return LogicalPlan("Join", [plan, subplan], RelDataType(lhs_fields),
+                           JoinNode(jt, cond))
+
+    def _decorrelate_exists(self, sub, lead_items=None):
         """Split the EXISTS subquery's WHERE into local conjuncts and
         equality correlations on OUTER columns (qualified names not bound
         by the sub's own FROM). Returns (subplan selecting the inner keys,
@@ -227,10 +253,10 @@ class Builder:
         where = None
         for cj in local:
             where = cj if where is None else ("call", "AND", [where, cj])
-        s2 = SelectStmt(items=[(k, f"ck{i}")
-                               for i, k in enumerate(inner_keys)],
-                        from_tables=sub.from_tables, joins=sub.joins,
-                        where=where)
+        items = list(lead_items or [])
+        items += [(k, f"ck{i}") for i, k in enumerate(inner_keys)]
+        s2 = SelectStmt(items=items, from_tables=sub.from_tables,
+                        joins=sub.joins, where=where)
         return self.build_stmt(s2), outer_keys
 
     @staticmethod
@@ -461,51 +487,22 @@ class Builder:
                 plan = LogicalPlan("Filter", [plan], plan.getRowType(),
                                    FilterNode(cond))
                 continue
-            sfields = subplan.getRowType().getFieldList()
-            node = AggregateNode(
-                [InputRef(i, f.getType()) for i, f in enumerate(sfields)],
-                [], distinct_node=True,
-                distinct_columns=[f.getName() for f in sfields])
-            subplan = LogicalPlan("Distinct", [subplan],
-                                  subplan.getRowType(), node)
-            lhs_fields = plan.getRowType().getFieldList()
-            combined = RelDataType(lhs_fields + sfields)
-            tmp = LogicalPlan("__combined__", [], combined, None)
-            cond = None
-            for i, ok_ast in enumerate(outer_keys):
-                eq = Call("=", [self._resolve(ok_ast, tmp),
-                                InputRef(len(lhs_fields) + i,
-                                         sfields[i].getType())],
-                          SqlType("BOOLEAN"))
-                cond = eq if cond is None else Call("AND", [cond, eq],
-                                                    SqlType("BOOLEAN"))
-            jt = "LEFTANTI" if negated else "LEFTSEMI"
-            plan = LogicalPlan("Join", [plan, subplan],
-                               RelDataType(lhs_fields), JoinNode(jt, cond))
+            plan = self._semi_anti_join(plan, subplan, outer_keys, negated)
 
         # IN-subquery joins: x IN (SELECT c FROM ...) ≡ SEMI join on x = c
         # over the DISTINCT subquery output; NOT IN ≡ ANTI (NULL-in-subquery
-        # divergence documented in DESIGN.md)
+        # divergence documented in DESIGN.md). Equality-correlated subqueries
+        # add their correlation keys to the join.
         for e_ast, substmt, negated in in_subs:
-            subplan = self.build_stmt(substmt)
-            sfields = subplan.getRowType().getFieldList()
-            if len(sfields) != 1:
+            if len(substmt.items) != 1:
                 raise ValueError("IN subquery must select exactly one column")
-            node = AggregateNode(
-                [InputRef(0, sfields[0].getType())], [], distinct_node=True,
-                distinct_columns=[sfields[0].getName()])
-            subplan = LogicalPlan("Distinct", [subplan],
-                                  subplan.getRowType(), node)
-            lhs_fields = plan.getRowType().getFieldList()
-            combined = RelDataType(lhs_fields + sfields)
-            tmp = LogicalPlan("__combined__", [], combined, None)
-            cond = Call("=", [self._resolve(e_ast, tmp),
-                              InputRef(len(lhs_fields),
-                                       sfields[0].getType())],
-                        SqlType("BOOLEAN"))
-            jt = "LEFTANTI" if negated else "LEFTSEMI"
-            plan = LogicalPlan("Join", [plan, subplan],
-                               RelDataType(lhs_fields), JoinNode(jt, cond))
+            subplan, outer_keys = self._decorrelate_exists(
+                substmt, lead_items=list(substmt.items))
+            if subplan is None:
+                subplan = self.build_stmt(substmt)
+                outer_keys = []
+            plan = self._semi_anti_join(plan, subplan,
+                                        [e_ast] + outer_keys, negated)
 
         # leftover WHERE conjuncts → Filter (incl. scalar TRUE/FALSE)
         leftovers = [cj for i, cj in enumerate(where_conjuncts) if not used[i]

@@ -977,3 +977,20 @@ def test_exists_correlated_and_not(ctx):
     got4 = c.sql("SELECT COUNT(*) AS c FROM orders WHERE NOT EXISTS "
                  "(SELECT 1 FROM li WHERE qty > 1000)").compute()
     assert int(got4["c"].iloc[0]) == 4
+
+
+def test_in_subquery_correlated(ctx):
+    """Correlated IN: x IN (SELECT c FROM s WHERE s.k = outer.k) — the
+    correlation keys join alongside the IN key."""
+    from dask_sql_amd.context import Context
+    t = pd.DataFrame({"k": [1, 1, 2, 2], "x": [5, 6, 5, 9]})
+    s = pd.DataFrame({"k": [1, 1, 2], "c": [5, 7, 9]})
+    c = Context()
+    c.create_table("t", t)
+    c.create_table("s", s)
+    got = c.sql("SELECT t.k, t.x FROM t WHERE t.x IN "
+                "(SELECT s.c FROM s WHERE s.k = t.k)").compute()
+    # (1,5): s has (1,5) -> in; (1,6): no; (2,5): s k=2 has only 9 -> no;
+    # (2,9): yes
+    assert sorted(zip(got["k"].astype(int), got["x"].astype(int))) == \
+        [(1, 5), (2, 9)]
