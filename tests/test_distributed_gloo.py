@@ -103,7 +103,7 @@ def _run_partial_merge(rank, world, port, results):
         dist.destroy_process_group()
 
 
-def _spawn(fn):
+def _spawn(fn, world=2):
     import socket
     with socket.socket() as s:
         s.bind(("127.0.0.1", 0))
@@ -111,18 +111,28 @@ def _spawn(fn):
     ctx = mp.get_context("spawn")
     with ctx.Manager() as mgr:
         results = mgr.dict()
-        procs = [ctx.Process(target=fn, args=(r, 2, port, results))
-                 for r in range(2)]
+        procs = [ctx.Process(target=fn, args=(r, world, port, results))
+                 for r in range(world)]
         for p in procs:
             p.start()
         for p in procs:
             p.join(120)
-        assert dict(results) == {0: "ok", 1: "ok"}, dict(results)
+        assert dict(results) == {r: "ok" for r in range(world)}, \
+            dict(results)
 
 
 def test_exchange_buckets_gloo():
     _spawn(_run_exchange)
 
 
+def test_exchange_buckets_gloo_world3():
+    # odd world size: uneven splits, non-power-of-two bucket hash
+    _spawn(_run_exchange, world=3)
+
+
 def test_partial_merge_algebra_gloo():
     _spawn(_run_partial_merge)
+
+
+def test_partial_merge_algebra_gloo_world3():
+    _spawn(_run_partial_merge, world=3)
