@@ -1717,6 +1717,17 @@ class DaskWindowPlugin(BaseRelPlugin):
         f = spec.func
         if f == "row_number":
             res = grp.cumcount() + 1
+        elif f in ("lag", "lead"):
+            # row-based shift within the partition; boundary rows get the
+            # default (NULL unless given) — reference window.py lag/lead
+            off = spec.offset if f == "lag" else -spec.offset
+            res = grp["v"].shift(off)
+            if spec.default is not None:
+                pos = grp.cumcount()
+                size = grp["v"].transform("size")
+                bm = pos < spec.offset if f == "lag" \
+                    else pos >= size - spec.offset
+                res = res.where(~bm, spec.default)
         elif f in ("rank", "dense_rank"):
             rn = grp.cumcount() + 1
             df["_rn"] = rn

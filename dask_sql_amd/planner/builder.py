@@ -679,19 +679,31 @@ class Builder:
                 arg_t = pre_fields[arg_idx].getType().getSqlType()
             part_idx = [idx_of(p) for p in part]
             order_idx = [(idx_of(o), desc) for o, desc in order]
-            if func in ("row_number", "rank", "dense_rank") and not order_idx:
+            if func in ("row_number", "rank", "dense_rank", "lag",
+                        "lead") and not order_idx:
                 raise ValueError(f"{func.upper()} requires ORDER BY in OVER")
+            offset, default = 1, None
+            if func in ("lag", "lead"):
+                if len(args) > 1:
+                    if args[1][0] != "lit":
+                        raise ValueError("LAG/LEAD offset must be a literal")
+                    offset = int(args[1][1])
+                if len(args) > 2:
+                    if args[2][0] != "lit":
+                        raise ValueError("LAG/LEAD default must be a literal")
+                    default = args[2][1]
             if func in ("row_number", "rank", "dense_rank", "count"):
                 ty = "BIGINT"
             elif func == "avg":
                 ty = "DOUBLE"
             elif func == "sum":
                 ty = "DOUBLE" if _is_float(arg_t or "BIGINT") else "BIGINT"
-            else:  # min/max keep the arg type
+            else:  # min/max/lag/lead keep the arg type
                 ty = arg_t or "BIGINT"
             name = f"w{len(specs)}__{func}"
             specs.append(WindowSpec(func, arg_idx, part_idx, order_idx,
-                                    name, SqlType(ty)))
+                                    name, SqlType(ty), offset=offset,
+                                    default=default))
             win_out[repr(ast)] = name
         if len(pre_named) > len(in_fields):
             plan = LogicalPlan("Projection", [plan], RelDataType(pre_fields),

@@ -61,7 +61,7 @@ KEYWORDS = {
 }
 
 WINDOW_FUNCS = {"ROW_NUMBER", "RANK", "DENSE_RANK", "SUM", "COUNT", "AVG",
-                "MIN", "MAX"}
+                "MIN", "MAX", "LAG", "LEAD"}
 
 _TOKEN_RE = re.compile(
     r"""

@@ -331,13 +331,16 @@ class WindowSpec:
     the input fields."""
 
     def __init__(self, func: str, arg_idx, part_idx: list,
-                 order_idx: list, out_name: str, out_type):
-        self.func = func                  # row_number|rank|dense_rank|sum|...
+                 order_idx: list, out_name: str, out_type,
+                 offset: int = 1, default=None):
+        self.func = func                  # row_number|rank|...|lag|lead
         self.arg_idx = arg_idx            # int | None (ranking / COUNT(*))
         self.part_idx = list(part_idx)
         self.order_idx = list(order_idx)  # [(field index, desc bool)]
         self.out_name = out_name
         self.out_type = out_type
+        self.offset = offset              # LAG/LEAD row offset
+        self.default = default            # LAG/LEAD boundary default
 
 
 class WindowNode:

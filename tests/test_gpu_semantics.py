@@ -994,3 +994,21 @@ def test_in_subquery_correlated(ctx):
     # (2,9): yes
     assert sorted(zip(got["k"].astype(int), got["x"].astype(int))) == \
         [(1, 5), (2, 9)]
+
+
+def test_window_lag_lead(ctx):
+    from dask_sql_amd.context import Context
+    df = pd.DataFrame({"k": [1, 1, 1, 2, 2], "t": [1, 2, 3, 1, 2],
+                       "v": [10.0, 20.0, 30.0, 5.0, 6.0]})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql(
+        "SELECT k, t, LAG(v) OVER (PARTITION BY k ORDER BY t) AS pv, "
+        "LEAD(v, 1, -1.0) OVER (PARTITION BY k ORDER BY t) AS nv, "
+        "LAG(v, 2, 0.0) OVER (PARTITION BY k ORDER BY t) AS p2 "
+        "FROM t").compute().sort_values(["k", "t"]).reset_index(drop=True)
+    pv = got["pv"].tolist()
+    assert pd.isna(pv[0]) and pv[1] == 10.0 and pv[2] == 20.0
+    assert pd.isna(pv[3]) and pv[4] == 5.0
+    assert got["nv"].tolist() == [20.0, 30.0, -1.0, 6.0, -1.0]
+    assert got["p2"].tolist() == [0.0, 0.0, 10.0, 0.0, 0.0]
