@@ -684,14 +684,19 @@ class Builder:
                 raise ValueError(f"{func.upper()} requires ORDER BY in OVER")
             offset, default = 1, None
             if func in ("lag", "lead"):
+
+                def lit_of(a, what):
+                    if a[0] == "lit":
+                        return a[1]
+                    if (a[0] == "call" and a[1] == "NEG"
+                            and a[2][0][0] == "lit"):
+                        return -a[2][0][1]
+                    raise ValueError(f"LAG/LEAD {what} must be a literal")
+
                 if len(args) > 1:
-                    if args[1][0] != "lit":
-                        raise ValueError("LAG/LEAD offset must be a literal")
-                    offset = int(args[1][1])
+                    offset = int(lit_of(args[1], "offset"))
                 if len(args) > 2:
-                    if args[2][0] != "lit":
-                        raise ValueError("LAG/LEAD default must be a literal")
-                    default = args[2][1]
+                    default = lit_of(args[2], "default")
             if func in ("row_number", "rank", "dense_rank", "count"):
                 ty = "BIGINT"
             elif func == "avg":
