@@ -582,10 +582,13 @@ def test_window_desc_order_and_global(ctx):
     c = Context()
     c.create_table("t", df)
     got = c.sql("SELECT t, ROW_NUMBER() OVER (ORDER BY t DESC) AS rn, "
-                "SUM(v) OVER (ORDER BY t) AS rs FROM t").compute()
+                "SUM(v) OVER (ORDER BY t) AS rs, "
+                "SUM(v) OVER (ORDER BY t DESC) AS rd FROM t").compute()
     got = got.sort_values("t").reset_index(drop=True)
     assert got["rn"].astype(int).tolist() == [3, 2, 1]
     assert np.allclose(got["rs"], [10.0, 30.0, 60.0])
+    # DESC frame: rows with t >= current
+    assert np.allclose(got["rd"], [60.0, 50.0, 30.0])
 
 
 def test_dict_string_functions(ctx):
