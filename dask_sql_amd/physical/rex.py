@@ -451,9 +451,20 @@ class RexCompiler:
 
 
 def compile_expr(expr, cols, dictionaries=None):
-    """Returns (prog_tuple (for Runtime.make_prog), result_kind)."""
+    """Returns (prog_tuple (for Runtime.make_prog), result_kind). Cached on
+    the expression object — programs depend only on the column dtypes and
+    dictionaries, which are stable across the steps of a cached plan."""
+    key = (tuple(c.dtype for c in cols),
+           tuple(id(d) for d in (dictionaries or [])))
+    hit = getattr(expr, "_dsx_compiled", None)
+    if hit is not None and hit[0] == key:
+        return hit[1], hit[2]
     c = RexCompiler(cols, dictionaries)
     kind = c.compile(expr)
+    try:
+        expr._dsx_compiled = (key, c.prog, kind)
+    except Exception:
+        pass
     return c.prog, kind
 
 

@@ -262,7 +262,17 @@ class Runtime:
 
     @staticmethod
     def make_prog(instrs) -> tuple:
-        """instrs: list of (op, arg0, imm) where imm may be float/int."""
+        """instrs: list of (op, arg0, imm) where imm may be float/int.
+        Cached by content — the ctypes arrays are immutable after build and
+        the C side copies them, so reuse across steps is safe."""
+        return Runtime._make_prog_cached(tuple(instrs))
+
+    @staticmethod
+    def _make_prog_cached(instrs) -> tuple:
+        cache = Runtime._prog_cache
+        hit = cache.get(instrs)
+        if hit is not None:
+            return hit
         if len(instrs) > MAX_PROG:
             raise DsxError(f"program too long ({len(instrs)})")
         Runtime._check_depth(instrs)
@@ -276,7 +286,12 @@ class Runtime:
                 )
             else:
                 arr[i].imm = int(imm)
+        if len(cache) > 4096:
+            cache.clear()
+        cache[instrs] = (arr, len(instrs))
         return arr, len(instrs)
+
+    _prog_cache: dict = {}
 
     @staticmethod
     def _cols_array(cols):
