@@ -975,6 +975,7 @@ class Builder:
 
     def _find_output(self, ast, stmt, plan) -> int:
         fields = plan.getRowType().getFieldList()
+        orig = ast
         if ast[0] == "lit" and isinstance(ast[1], int):
             return ast[1] - 1
         if hasattr(self, "_agg_rewrite"):
@@ -983,8 +984,10 @@ class Builder:
             for i, f in enumerate(fields):
                 if f.getName().lower() == ast[2].lower():
                     return i
-        # structural match against select items
+        # structural match against select items (ORDER BY SUM(v) where the
+        # same aggregate appears in the SELECT list, possibly aliased)
         for i, (e, alias) in enumerate(stmt.items):
-            if e == ast:
+            if e == orig or e == ast:
                 return i
-        raise KeyError(f"ORDER BY expression not in output: {ast!r}")
+        raise KeyError(f"ORDER BY expression not in output: {ast!r} "
+                       "(an ORDER BY aggregate must also appear in SELECT)")
