@@ -5,6 +5,7 @@ the HIP physical layer (no CPU fallback — DsxUnavailable propagates)."""
 from __future__ import annotations
 
 import logging
+import re
 
 import numpy as np
 import pandas as pd
@@ -201,6 +202,28 @@ class Context:
                      **kwargs):
         if isinstance(input_table, dict):
             input_table = pd.DataFrame(input_table)
+        if isinstance(input_table, ResultFrame):
+            # register a query result device-resident (CTAS; reference
+            # context.py create_table accepts dask frames — ours are device
+            # tables, no host round-trip)
+            dc = input_table.dc
+            if isinstance(dc, HostDataContainer):
+                input_table = dc.pdf
+            else:
+                cc = dc.column_container
+                cols = {}
+                sqlts = {}
+                fields = input_table._rel.getRowType().getFieldList()
+                for i, f in enumerate(cc.columns):
+                    col = dc.table.col(cc.get_backend_by_frontend_name(f))
+                    cols[f] = col
+                    if i < len(fields):
+                        sqlts[f] = fields[i].getType().getSqlType()
+                self.create_table_from_device(
+                    table_name, DeviceTable(cols,
+                                            num_rows=dc.table.num_rows),
+                    sql_types=sqlts)
+                return
         if isinstance(input_table, str):
             # file-path inputs (reference input_utils/location.py:22-60:
             # format inferred from the extension; parquet via pyarrow,
@@ -252,6 +275,14 @@ class Context:
     # -- reference context.py:482 sql --------------------------------------
     def sql(self, sql: str, return_futures: bool = True,
             config_options=None) -> ResultFrame:
+        # CREATE TABLE <name> AS <select> (reference DDL create_table.py)
+        m = re.match(r"\s*CREATE\s+(?:OR\s+REPLACE\s+)?TABLE\s+(\w+)\s+AS\s*"
+                     r"\(?\s*(SELECT.*?)\)?\s*;?\s*$", sql,
+                     re.IGNORECASE | re.DOTALL)
+        if m:
+            res = self.sql(m.group(2))
+            self.create_table(m.group(1), res)
+            return res
         rel = self._get_ral(sql)
         logger.debug("plan:\n%s", rel.explain())
         dc = RelConverter.convert(rel, context=self)

@@ -1015,3 +1015,23 @@ def test_window_lag_lead(ctx):
     assert pd.isna(pv[3]) and pv[4] == 5.0
     assert got["nv"].tolist() == [20.0, 30.0, -1.0, 6.0, -1.0]
     assert got["p2"].tolist() == [0.0, 0.0, 10.0, 0.0, 0.0]
+
+
+def test_ctas_and_order_by_aggregate(ctx):
+    """CREATE TABLE AS (device-resident registration) + ORDER BY on an
+    aggregate expression appearing in SELECT."""
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(71)
+    df = pd.DataFrame({"k": rng.integers(0, 20, 5000).astype(np.int64),
+                       "v": rng.random(5000)})
+    c = Context()
+    c.create_table("t", df)
+    c.sql("CREATE TABLE sums AS SELECT k, SUM(v) AS s FROM t GROUP BY k")
+    got = c.sql("SELECT k, s FROM sums ORDER BY s DESC LIMIT 3").compute()
+    exp = df.groupby("k")["v"].sum().sort_values(ascending=False).head(3)
+    assert got["k"].astype(int).tolist() == exp.index.tolist()
+    np.testing.assert_allclose(got["s"].to_numpy(np.float64),
+                               exp.to_numpy(), rtol=1e-9)
+    got2 = c.sql("SELECT k, SUM(v) AS s FROM t GROUP BY k "
+                 "ORDER BY SUM(v) DESC LIMIT 3").compute()
+    assert got2["k"].astype(int).tolist() == exp.index.tolist()
