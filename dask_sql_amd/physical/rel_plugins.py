@@ -723,7 +723,8 @@ class DaskAggregatePlugin(BaseRelPlugin):
 
     AGG_OPS = {"sum", "count", "avg", "min", "max", "any_value",
                "single_value", "stddev", "stddev_samp", "stddev_pop",
-               "var_samp", "var_pop", "variance"}
+               "var_samp", "var_pop", "variance", "every", "bool_and",
+               "bool_or"}
 
     # stddev/variance family: ONE call decomposes into TWO kernel slots
     # (SUM(x), SUM(x*x)); finalize composes m2 = Σx² − (Σx)²/n like the
@@ -1157,6 +1158,14 @@ class DaskAggregatePlugin(BaseRelPlugin):
             if kind == KF:
                 return [(rt.AGG_SUM_F64, prog)], "sum_f"
             return [(rt.AGG_SUM_I64, prog)], "sum_i"
+        if func in ("every", "bool_and", "bool_or"):
+            # boolean aggregates = MIN/MAX over {0,1} ignoring NULLs
+            # (reference aggregate.py AGGREGATION_MAPPING every/bool ops)
+            if kind != KB:
+                raise RexCompileError(f"{func} needs a boolean argument")
+            op_ = rt.AGG_MIN_I64 if func in ("every", "bool_and") \
+                else rt.AGG_MAX_I64
+            return [(op_, prog)], "min_i" if func != "bool_or" else "max_i"
         if func in ("min", "any_value", "single_value"):
             return ([((rt.AGG_MIN_F64 if kind == KF else rt.AGG_MIN_I64),
                       prog)],

@@ -858,6 +858,8 @@ class Builder:
                 arg_t = "BIGINT"
             if func == "count":
                 out_t = "BIGINT"
+            elif func in ("every", "bool_and", "bool_or"):
+                out_t = "BOOLEAN"
             elif func in ("avg", "stddev", "stddev_samp", "stddev_pop",
                           "var_samp", "var_pop", "variance"):
                 out_t = "DOUBLE"

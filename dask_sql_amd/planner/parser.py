@@ -105,7 +105,7 @@ def tokenize(sql: str):
 
 AGG_FUNCS = {"SUM", "COUNT", "AVG", "MIN", "MAX", "ANY_VALUE", "STDDEV",
              "STDDEV_POP", "STDDEV_SAMP", "VAR_SAMP", "VAR_POP", "VARIANCE",
-             "SINGLE_VALUE", "EVERY", "BIT_AND", "BIT_OR"}
+             "SINGLE_VALUE", "EVERY", "BOOL_AND", "BOOL_OR"}
 
 
 class Parser:

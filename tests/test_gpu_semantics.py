@@ -1035,3 +1035,16 @@ def test_ctas_and_order_by_aggregate(ctx):
     got2 = c.sql("SELECT k, SUM(v) AS s FROM t GROUP BY k "
                  "ORDER BY SUM(v) DESC LIMIT 3").compute()
     assert got2["k"].astype(int).tolist() == exp.index.tolist()
+
+
+def test_boolean_aggregates(ctx):
+    """EVERY / BOOL_AND / BOOL_OR (min/max over {0,1}, NULLs skipped)."""
+    from dask_sql_amd.context import Context
+    df = pd.DataFrame({"k": [1, 1, 2, 2, 3],
+                       "v": [5, 10, 10, 20, 7]})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT k, EVERY(v >= 10) AS e, BOOL_OR(v >= 10) AS o "
+                "FROM t GROUP BY k ORDER BY k").compute()
+    assert got["e"].astype(bool).tolist() == [False, True, False]
+    assert got["o"].astype(bool).tolist() == [True, True, False]
