@@ -183,16 +183,41 @@ class DaskProjectPlugin(BaseRelPlugin):
         named = rel.projection().getNamedProjects()
         out_cols = {}
         new_names = []
-        from dask_sql_amd.physical.rex import dict_string_fn
+        from dask_sql_amd.physical.rex import (dict_int_fn, dict_string_fn,
+                                               fold_string_literal)
         for i, (expr, name) in enumerate(named):
             backend_name = f"p{i}__{name}"
             if not isinstance(expr, InputRef):
                 expr = _resolve_scalar_subs(expr, context)
             sfn = None if isinstance(expr, InputRef) \
                 else dict_string_fn(expr, dicts)
+            ifn = None if isinstance(expr, InputRef) or sfn is not None \
+                else dict_int_fn(expr, dicts)
+            const_s = None
+            if not isinstance(expr, InputRef) and sfn is None and ifn is None:
+                const_s = fold_string_literal(expr)
             if isinstance(expr, InputRef):
                 src = cols[expr.getIndex()]
                 out_cols[backend_name] = src  # zero-copy reuse
+            elif ifn is not None:
+                # int-valued string function (CHAR_LENGTH): per-dictionary
+                # LUT gathered by the code column (test_rex.py:603)
+                ci, f = ifn
+                src = cols[ci]
+                lut = np.array([f(s_) if s_ is not None else 0
+                                for s_ in src.dictionary], dtype=np.int64)
+                lut_col = runtime.upload_column(lut)
+                g = runtime.gather(lut_col, src.data, src.len)
+                out_cols[backend_name] = rt.DeviceColumn(
+                    runtime, g.data, src.validity, src.len, rt.I64,
+                    owner=False, keep_alive=(g, src, lut_col))
+            elif const_s is not None:
+                # literal-only string chain → constant dict column
+                n_ = dc.table.num_rows
+                col = runtime.upload_column(
+                    np.zeros(n_, dtype=np.int32), dtype=rt.I32)
+                col.dictionary = [const_s]
+                out_cols[backend_name] = col
             elif sfn is not None:
                 # string function over a dict column: same codes, the
                 # transform runs once over the dictionary

@@ -49,6 +49,18 @@ class RexCompileError(NotImplementedError):
     pass
 
 
+def _lit_int(o):
+    if isinstance(o, Literal):
+        try:
+            return int(o.getValue())
+        except (TypeError, ValueError):
+            return None
+    if isinstance(o, Call) and o.getOperatorName() == "NEG" \
+            and isinstance(o.getOperands()[0], Literal):
+        return -int(o.getOperands()[0].getValue())
+    return None
+
+
 def dict_string_fn(expr, dicts):
     """If expr is a chain of string functions (UPPER/LOWER/SUBSTRING) over a
     dict-encoded column, return (col_index, python_fn) computing the string
@@ -73,19 +85,124 @@ def dict_string_fn(expr, dicts):
                 return i, (lambda s, f=f: f(s).upper())
             return i, (lambda s, f=f: f(s).lower())
         if op in ("SUBSTRING", "SUBSTR") and len(ops_) in (2, 3):
-            if not all(isinstance(o, Literal) for o in ops_[1:]):
+            a = _lit_int(ops_[1])
+            ln = _lit_int(ops_[2]) if len(ops_) == 3 else None
+            if a is None or (len(ops_) == 3 and ln is None):
                 return None
-            a = int(ops_[1].getValue())
-            ln = int(ops_[2].getValue()) if len(ops_) == 3 else None
-            if a < 1 or (ln is not None and ln < 0):
-                return None  # Calcite negative-start edge: not covered
-            start = a - 1
+            if ln is not None and ln < 0:
+                return None
+            # Calcite: FROM 0/negative counts from the start without
+            # shifting the window (test_rex.py:614 SUBSTRING(a FROM -1)
+            # yields the whole string)
+            start = max(a - 1, 0)
 
             def g(s, f=f, start=start, ln=ln):
                 t = f(s)
                 return t[start:start + ln] if ln is not None else t[start:]
 
             return i, g
+        if op == "INITCAP" and len(ops_) == 1:
+            return i, (lambda s, f=f: f(s).title())
+        if op == "TRIM" and len(ops_) == 3:
+            if not all(isinstance(o, Literal) for o in ops_[1:]):
+                return None
+            mode = str(ops_[1].getValue()).upper()
+            ch = str(ops_[2].getValue())
+
+            def g(s, f=f, mode=mode, ch=ch):
+                t = f(s)
+                if mode == "LEADING":
+                    return t.lstrip(ch)
+                if mode == "TRAILING":
+                    return t.rstrip(ch)
+                return t.strip(ch)
+
+            return i, g
+        if op == "REPLACE" and len(ops_) == 3:
+            if not all(isinstance(o, Literal) for o in ops_[1:]):
+                return None
+            s1 = str(ops_[1].getValue())
+            s2 = str(ops_[2].getValue())
+            return i, (lambda s, f=f, s1=s1, s2=s2: f(s).replace(s1, s2))
+        if op == "CONCAT":
+            parts = []
+            ci = None
+            for o in ops_:
+                if isinstance(o, Literal) and isinstance(o.getValue(), str):
+                    parts.append(("lit", o.getValue()))
+                    continue
+                sub2 = dict_string_fn(o, dicts)
+                if sub2 is None:
+                    return None
+                if ci is None:
+                    ci = sub2[0]
+                elif sub2[0] != ci:
+                    return None  # concat across two dict columns: per-row
+                parts.append(("fn", sub2[1]))
+            if ci is None:
+                return None
+
+            def g(s, parts=parts):
+                return "".join(p[1] if p[0] == "lit" else p[1](s)
+                               for p in parts)
+
+            return ci, g
+    return None
+
+
+def fold_string_literal(expr):
+    """Evaluate a literal-only string-function chain to a constant str
+    (REPLACE('Another String', ...) etc. — test_rex.py:624)."""
+    if isinstance(expr, Literal) and isinstance(expr.getValue(), str):
+        return expr.getValue()
+    if not isinstance(expr, Call):
+        return None
+    op = expr.getOperatorName().upper()
+    ops_ = expr.getOperands()
+    vals = [fold_string_literal(o) if i == 0 or op == "CONCAT"
+            else (o.getValue() if isinstance(o, Literal) else _lit_int(o))
+            for i, o in enumerate(ops_)]
+    if any(v is None for v in vals):
+        return None
+    s = vals[0]
+    if op == "UPPER":
+        return s.upper()
+    if op == "LOWER":
+        return s.lower()
+    if op == "INITCAP":
+        return s.title()
+    if op == "REPLACE":
+        return s.replace(str(vals[1]), str(vals[2]))
+    if op == "TRIM":
+        mode, ch = str(vals[1]).upper(), str(vals[2])
+        return s.lstrip(ch) if mode == "LEADING" else \
+            s.rstrip(ch) if mode == "TRAILING" else s.strip(ch)
+    if op in ("SUBSTRING", "SUBSTR"):
+        a0 = _lit_int(ops_[1])
+        if a0 is None:
+            return None
+        a = max(a0 - 1, 0)
+        ln = _lit_int(ops_[2]) if len(ops_) > 2 else None
+        return s[a:a + ln] if ln is not None else s[a:]
+    if op == "CONCAT":
+        return "".join(str(v) for v in vals)
+    return None
+
+
+def dict_int_fn(expr, dicts):
+    """String function with an INTEGER result over a dict column
+    (CHAR_LENGTH): returns (col_index, per-entry int fn) — evaluated once
+    over the dictionary, applied per row via a LUT gather."""
+    if not isinstance(expr, Call):
+        return None
+    op = expr.getOperatorName().upper()
+    if op in ("CHAR_LENGTH", "CHARACTER_LENGTH", "LENGTH") \
+            and len(expr.getOperands()) == 1:
+        sub = dict_string_fn(expr.getOperands()[0], dicts)
+        if sub is None:
+            return None
+        i, f = sub
+        return i, (lambda s, f=f: len(f(s)))
     return None
 
 

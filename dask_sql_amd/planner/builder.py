@@ -154,8 +154,11 @@ class Builder:
                 ty = "BOOLEAN"
             elif op == "NEG":
                 ty = _expr_type(ops[0])
-            elif op in ("UPPER", "LOWER", "SUBSTRING", "SUBSTR"):
+            elif op in ("UPPER", "LOWER", "SUBSTRING", "SUBSTR", "CONCAT",
+                        "TRIM", "REPLACE", "INITCAP"):
                 ty = "VARCHAR"
+            elif op in ("CHAR_LENGTH", "CHARACTER_LENGTH", "LENGTH"):
+                ty = "BIGINT"
             elif op in ("FLOOR", "CEIL", "CEILING", "ROUND", "EXP", "LN",
                         "LOG", "POWER", "POW", "SQRT"):
                 ty = "DOUBLE"

@@ -70,7 +70,7 @@ _TOKEN_RE = re.compile(
   | (?P<str>'(?:[^']|'')*')
   | (?P<qid>"[^"]+")
   | (?P<id>[A-Za-z_][A-Za-z_0-9]*)
-  | (?P<op><>|!=|>=|<=|=|<|>|\+|-|\*|/|%|\(|\)|,|\.)
+  | (?P<op>\|\||<>|!=|>=|<=|=|<|>|\+|-|\*|/|%|\(|\)|,|\.)
     """,
     re.VERBOSE,
 )
@@ -386,10 +386,14 @@ class Parser:
     def add_expr(self):
         e = self.mul_expr()
         while True:
-            op = self.accept_op("+", "-")
+            op = self.accept_op("+", "-", "||")
             if not op:
                 break
-            e = ("call", op, [e, self.mul_expr()])
+            rhs = self.mul_expr()
+            if op == "||":
+                e = ("call", "CONCAT", [e, rhs])
+            else:
+                e = ("call", op, [e, rhs])
         return e
 
     def mul_expr(self):
@@ -472,6 +476,43 @@ class Parser:
                 return ("cast", e, ty)
         if t[0] == "id":
             name = self._name()
+            if name.upper() == "TRIM" and self.peek() == ("op", "("):
+                # TRIM([LEADING|TRAILING|BOTH] ['ch'] FROM x) | TRIM(x)
+                self.next()
+                mode = "BOTH"
+                if self.peek()[0] == "id" and self.peek()[1].upper() in (
+                        "LEADING", "TRAILING", "BOTH"):
+                    mode = self.next()[1].upper()
+                ch = " "
+                if self.peek()[0] == "str":
+                    ch = self.next()[1]
+                if self.accept_kw("FROM"):
+                    e = self.expr()
+                else:
+                    e = self.expr()
+                self.expect_op(")")
+                return ("call", "TRIM",
+                        [e, ("lit", mode, "VARCHAR"),
+                         ("lit", ch, "VARCHAR")])
+            if name.upper() in ("SUBSTRING", "SUBSTR") \
+                    and self.peek() == ("op", "("):
+                # SUBSTRING(x FROM a [FOR n]) — Calcite form
+                save = self.i
+                self.next()
+                e = self.expr()
+                if self.accept_kw("FROM"):
+                    start = self.expr()
+                    args = [e, start]
+                    if self.peek() == ("id", "FOR") or \
+                            self.accept_kw("FOR") or \
+                            (self.peek()[0] == "id"
+                             and self.peek()[1].upper() == "FOR"):
+                        if self.peek()[0] == "id":
+                            self.next()
+                        args.append(self.expr())
+                    self.expect_op(")")
+                    return ("call", "SUBSTRING", args)
+                self.i = save  # comma form: reparse generically
             if name.upper() == "EXTRACT" and self.peek() == ("op", "("):
                 # EXTRACT(field FROM expr)
                 self.next()

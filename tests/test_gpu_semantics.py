@@ -1048,3 +1048,84 @@ def test_boolean_aggregates(ctx):
                 "FROM t GROUP BY k ORDER BY k").compute()
     assert got["e"].astype(bool).tolist() == [False, True, False]
     assert got["o"].astype(bool).tolist() == [True, True, False]
+
+
+def test_over_golden_reference(ctx, user_table_1):
+    """Window results pinned against the reference's own expected frames
+    (tests/integration/test_over.py:56-76 test_over_with_different and
+    :81-117 test_over_calls — the aggregate columns our subset covers)."""
+    from dask_sql_amd.context import Context
+    c = Context()
+    c.create_table("user_table_1", user_table_1)
+    got = c.sql(
+        'SELECT user_id, b, '
+        'ROW_NUMBER() OVER (PARTITION BY user_id ORDER BY b) AS "R1", '
+        'ROW_NUMBER() OVER (ORDER BY user_id, b) AS "R2" '
+        'FROM user_table_1').compute()
+    assert got["R1"].astype(int).tolist() == [2, 1, 1, 1]
+    assert got["R2"].astype(int).tolist() == [3, 1, 2, 4]
+    got2 = c.sql(
+        'SELECT user_id, b, '
+        'SUM(user_id) OVER (PARTITION BY user_id ORDER BY b) AS "O5", '
+        'AVG(user_id) OVER (PARTITION BY user_id ORDER BY b) AS "O6", '
+        'COUNT(*) OVER (PARTITION BY user_id ORDER BY b) AS "O7", '
+        'COUNT(b) OVER (PARTITION BY user_id ORDER BY b) AS "O7b", '
+        'MAX(b) OVER (PARTITION BY user_id ORDER BY b) AS "O8", '
+        'MIN(b) OVER (PARTITION BY user_id ORDER BY b) AS "O9" '
+        'FROM user_table_1').compute()
+    assert got2["O5"].astype(int).tolist() == [4, 1, 2, 3]
+    assert got2["O6"].astype(float).tolist() == [2.0, 1.0, 2.0, 3.0]
+    assert got2["O7"].astype(int).tolist() == [2, 1, 1, 1]
+    assert got2["O7b"].astype(int).tolist() == [2, 1, 1, 1]
+    assert got2["O8"].astype(int).tolist() == [3, 3, 1, 3]
+    assert got2["O9"].astype(int).tolist() == [1, 3, 1, 3]
+
+
+def test_string_functions_golden_reference(ctx):
+    """String-function outputs pinned against the reference's own expected
+    frame (tests/integration/test_rex.py:591-660 test_string_functions) on
+    the same 'a normal string' input."""
+    from dask_sql_amd.context import Context
+    c = Context()
+    c.create_table("string_table", pd.DataFrame({"a": ["a normal string"]}))
+    got = c.sql("""
+        SELECT
+            a || 'hello' || a AS a2,
+            CONCAT(a, 'hello', a) AS b,
+            CHAR_LENGTH(a) AS c,
+            UPPER(a) AS d,
+            LOWER(a) AS e,
+            TRIM('a' FROM a) AS h,
+            TRIM(BOTH 'a' FROM a) AS i,
+            TRIM(LEADING 'a' FROM a) AS j,
+            TRIM(TRAILING 'a' FROM a) AS k,
+            SUBSTRING(a FROM -1) AS o,
+            SUBSTRING(a FROM 10) AS p,
+            SUBSTRING(a FROM 2) AS q,
+            SUBSTRING(a FROM 2 FOR 2) AS r,
+            SUBSTR(a, 3, 6) AS s,
+            INITCAP(a) AS t,
+            INITCAP(UPPER(a)) AS u,
+            INITCAP(LOWER(a)) AS v,
+            REPLACE(a, 'r', 'l') AS w,
+            REPLACE('Another String', 'th', 'b') AS x
+        FROM string_table""").compute()
+    exp = {  # reference test_rex.py:632-660 expected_df, verbatim
+        "a2": "a normal stringhelloa normal string",
+        "b": "a normal stringhelloa normal string",
+        "c": 15,
+        "d": "A NORMAL STRING", "e": "a normal string",
+        "h": " normal string", "i": " normal string",
+        "j": " normal string", "k": "a normal string",
+        "o": "a normal string", "p": "string",
+        "q": " normal string", "r": " n", "s": "normal",
+        "t": "A Normal String", "u": "A Normal String",
+        "v": "A Normal String",
+        "w": "a nolmal stling", "x": "Anober String",
+    }
+    for k_, v_ in exp.items():
+        got_v = got[k_].iloc[0]
+        if k_ == "c":
+            assert int(got_v) == v_, k_
+        else:
+            assert got_v == v_, (k_, got_v, v_)
