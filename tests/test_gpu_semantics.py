@@ -1129,3 +1129,40 @@ def test_string_functions_golden_reference(ctx):
             assert int(got_v) == v_, k_
         else:
             assert got_v == v_, (k_, got_v, v_)
+
+
+def test_sort_nulls_golden_reference(ctx):
+    """ORDER BY NULLS FIRST/LAST pinned against the reference's expected
+    frames (tests/integration/test_sort.py:100-235 test_sort_with_nan +
+    test_sort_with_nan_more_columns)."""
+    from dask_sql_amd.context import Context
+    nan, inf = float("nan"), float("inf")
+
+    def col(got, name):
+        return [None if pd.isna(v) else v for v in got[name]]
+
+    c = Context()
+    c.create_table("df", pd.DataFrame(
+        {"a": [1, 2, nan, 2], "b": [4, nan, 5, inf]}))
+    got = c.sql("SELECT * FROM df ORDER BY a").compute()
+    assert col(got, "a") == [1, 2, 2, None]
+    assert col(got, "b") == [4, None, inf, 5]
+    got = c.sql("SELECT * FROM df ORDER BY a NULLS FIRST").compute()
+    assert col(got, "a") == [None, 1, 2, 2]
+    assert col(got, "b") == [5, 4, None, inf]
+
+    c2 = Context()
+    c2.create_table("df", pd.DataFrame({
+        "a": [1, 1, 2, 2, nan, nan],
+        "b": [1, 1, 2, nan, inf, 5],
+        "c": [1, nan, 3, 4, 5, 6]}))
+    got = c2.sql("SELECT * FROM df ORDER BY a ASC NULLS FIRST, "
+                 "b DESC NULLS LAST, c ASC NULLS FIRST").compute()
+    assert col(got, "a") == [None, None, 1, 1, 2, 2]
+    assert col(got, "b") == [inf, 5, 1, 1, 2, None]
+    assert col(got, "c") == [5, 6, None, 1, 3, 4]
+    got = c2.sql("SELECT * FROM df ORDER BY a ASC NULLS LAST, "
+                 "b DESC NULLS FIRST, c DESC NULLS LAST").compute()
+    assert col(got, "a") == [1, 1, 2, 2, None, None]
+    assert col(got, "b") == [1, 1, None, 2, inf, 5]
+    assert col(got, "c") == [1, None, 4, 3, 5, 6]
