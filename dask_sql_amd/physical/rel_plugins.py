@@ -1896,8 +1896,19 @@ class DaskUnionPlugin(BaseRelPlugin):
         return runtime.concat_columns(parts, target)
 
 
+class DaskValuesPlugin(BaseRelPlugin):
+    """One-row empty relation for FROM-less SELECTs (SELECT 1 + 1)."""
+
+    class_name = "Values"
+
+    def convert(self, rel, context):
+        context._get_runtime()  # GPU required like everything else
+        return DataContainer(DeviceTable({}, num_rows=1), ColumnContainer([]))
+
+
 def register_defaults():
     for cls in (DaskTableScanPlugin, DaskFilterPlugin, DaskProjectPlugin,
                 DaskJoinPlugin, DaskAggregatePlugin, DaskSortPlugin,
-                DaskLimitPlugin, DaskWindowPlugin, DaskUnionPlugin):
+                DaskLimitPlugin, DaskWindowPlugin, DaskUnionPlugin,
+                DaskValuesPlugin):
         RelConverter.add_plugin_class(cls, replace=False)

@@ -412,7 +412,12 @@ class Builder:
             return plan
 
         tables = list(stmt.from_tables)
-        plan = scan_with_filters(tables[0]) if tables else None
+        if tables:
+            plan = scan_with_filters(tables[0])
+        else:
+            # FROM-less SELECT (constants only, e.g. SELECT 1 + 1 —
+            # reference supports via a one-row relation)
+            plan = LogicalPlan("Values", [], RelDataType([]), None)
 
         def join_plans(lhs, rhs, join_type, cond_ast_list, on_expr_ast):
             lhs_fields = lhs.getRowType().getFieldList()

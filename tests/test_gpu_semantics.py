@@ -1166,3 +1166,15 @@ def test_sort_nulls_golden_reference(ctx):
     assert col(got, "a") == [1, 1, 2, 2, None, None]
     assert col(got, "b") == [1, 1, None, 2, inf, 5]
     assert col(got, "c") == [1, None, 4, 3, 5, 6]
+
+
+def test_fromless_select(ctx):
+    """SELECT without FROM (reference test_jdbc.py:33 SELECT 1 + 1)."""
+    from dask_sql_amd.context import Context
+    c = Context()
+    got = c.sql("SELECT 1 + 1 AS two, 3.5 AS f, 'hi' AS s, "
+                "UPPER('ab') AS u").compute()
+    assert int(got["two"].iloc[0]) == 2
+    assert float(got["f"].iloc[0]) == 3.5
+    assert got["s"].iloc[0] == "hi"
+    assert got["u"].iloc[0] == "AB"
