@@ -256,3 +256,23 @@ def test_union_parses_and_folds():
     assert u.alls == [True, False]
     assert u.limit == 5 and len(u.order_by) == 1
     assert u.branches[2].order_by == [] and u.branches[2].limit is None
+
+
+def test_show_commands():
+    """SHOW SCHEMAS/TABLES/COLUMNS — expected frames per reference
+    tests/integration/test_show.py:9-62."""
+    import pandas as pd
+    from dask_sql_amd.context import Context
+    c = Context()
+    c.create_table("user_table_1", pd.DataFrame(
+        {"user_id": [1], "b": [2]}))
+    s = c.sql("SHOW SCHEMAS").compute()
+    assert s["Schema"].tolist() == ["root", "information_schema"]
+    t = c.sql('SHOW TABLES FROM "root"').compute()
+    assert t["Table"].tolist() == ["user_table_1"]
+    cols = c.sql('SHOW COLUMNS FROM "root"."user_table_1"').compute()
+    assert cols["Column"].tolist() == ["user_id", "b"]
+    assert cols["Type"].tolist() == ["bigint", "bigint"]
+    import pytest
+    with pytest.raises(KeyError):
+        c.sql('SHOW COLUMNS FROM "root"."missing"')
