@@ -300,6 +300,34 @@ class Context:
                 })
             from dask_sql_amd.datacontainer import HostDataContainer
             return ResultFrame(HostDataContainer(pdf), None, self)
+        # DROP TABLE (reference rel/custom/drop.py)
+        dm = re.match(r"\s*DROP\s+TABLE\s+(?:IF\s+EXISTS\s+)?(\w+)\s*;?\s*$",
+                      sql, re.IGNORECASE)
+        if dm:
+            self.drop_table(dm.group(1))
+            from dask_sql_amd.datacontainer import HostDataContainer
+            return ResultFrame(HostDataContainer(pd.DataFrame()), None, self)
+        # ANALYZE TABLE ... COMPUTE STATISTICS (reference rel/custom/
+        # analyze_table.py; frame shape pinned by test_analyze.py:8-33:
+        # describe() rows + data_type + col_name)
+        am = re.match(r"\s*ANALYZE\s+TABLE\s+(\w+)\s+COMPUTE\s+STATISTICS"
+                      r"\s+FOR\s+(ALL\s+COLUMNS|COLUMNS\s+(.+?))\s*;?\s*$",
+                      sql, re.IGNORECASE)
+        if am:
+            pdf = self.sql(f"SELECT * FROM {am.group(1)}").compute()
+            if am.group(3):
+                cols = [c.strip() for c in am.group(3).split(",")]
+                pdf = pdf[cols]
+            stats = pdf.describe()
+            extra = pd.DataFrame(
+                {c: [str(t).lower(), c] for c, t in zip(
+                    pdf.columns,
+                    [dict(self.tables[am.group(1).lower()].fields()).get(
+                        c, "double") for c in pdf.columns])},
+                index=["data_type", "col_name"])
+            out = pd.concat([stats, extra])
+            from dask_sql_amd.datacontainer import HostDataContainer
+            return ResultFrame(HostDataContainer(out), None, self)
         # CREATE TABLE <name> AS <select> (reference DDL create_table.py)
         m = re.match(r"\s*CREATE\s+(?:OR\s+REPLACE\s+)?TABLE\s+(\w+)\s+AS\s*"
                      r"\(?\s*(SELECT.*?)\)?\s*;?\s*$", sql,

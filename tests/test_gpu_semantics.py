@@ -1178,3 +1178,26 @@ def test_fromless_select(ctx):
     assert float(got["f"].iloc[0]) == 3.5
     assert got["s"].iloc[0] == "hi"
     assert got["u"].iloc[0] == "AB"
+
+
+def test_drop_and_analyze_table(ctx):
+    """DROP TABLE + ANALYZE TABLE ... COMPUTE STATISTICS (frame shape per
+    reference test_analyze.py:8-33: describe() + data_type + col_name)."""
+    from dask_sql_amd.context import Context
+    c = Context()
+    df = pd.DataFrame({"a": [1.0, 2.0, 3.0], "b": [4, 5, 6]})
+    c.create_table("t", df)
+    res = c.sql("ANALYZE TABLE t COMPUTE STATISTICS FOR ALL COLUMNS"
+                ).compute()
+    assert list(res.columns) == ["a", "b"]
+    assert "mean" in res.index and "data_type" in res.index \
+        and "col_name" in res.index
+    assert float(res.loc["mean", "a"]) == 2.0
+    assert res.loc["col_name", "b"] == "b"
+    res2 = c.sql("ANALYZE TABLE t COMPUTE STATISTICS FOR COLUMNS a"
+                 ).compute()
+    assert list(res2.columns) == ["a"]
+    c.sql("DROP TABLE t")
+    import pytest
+    with pytest.raises(KeyError):
+        c.sql("SELECT * FROM t")
