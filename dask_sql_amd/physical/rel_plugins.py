@@ -1751,6 +1751,9 @@ class DaskWindowPlugin(BaseRelPlugin):
         f = spec.func
         if f == "row_number":
             res = grp.cumcount() + 1
+        elif f == "first_value":
+            # default frame starts at the partition head (test_over.py:90)
+            res = grp["v"].transform("first")
         elif f in ("lag", "lead"):
             # row-based shift within the partition; boundary rows get the
             # default (NULL unless given) — reference window.py lag/lead

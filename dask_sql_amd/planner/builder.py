@@ -688,7 +688,7 @@ class Builder:
             part_idx = [idx_of(p) for p in part]
             order_idx = [(idx_of(o), desc) for o, desc in order]
             if func in ("row_number", "rank", "dense_rank", "lag",
-                        "lead") and not order_idx:
+                        "lead", "first_value") and not order_idx:
                 raise ValueError(f"{func.upper()} requires ORDER BY in OVER")
             offset, default = 1, None
             if func in ("lag", "lead"):
@@ -711,7 +711,7 @@ class Builder:
                 ty = "DOUBLE"
             elif func == "sum":
                 ty = "DOUBLE" if _is_float(arg_t or "BIGINT") else "BIGINT"
-            else:  # min/max/lag/lead keep the arg type
+            else:  # min/max/lag/lead/first_value keep the arg type
                 ty = arg_t or "BIGINT"
             name = f"w{len(specs)}__{func}"
             specs.append(WindowSpec(func, arg_idx, part_idx, order_idx,

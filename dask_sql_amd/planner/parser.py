@@ -61,7 +61,7 @@ KEYWORDS = {
 }
 
 WINDOW_FUNCS = {"ROW_NUMBER", "RANK", "DENSE_RANK", "SUM", "COUNT", "AVG",
-                "MIN", "MAX", "LAG", "LEAD"}
+                "MIN", "MAX", "LAG", "LEAD", "FIRST_VALUE"}
 
 _TOKEN_RE = re.compile(
     r"""

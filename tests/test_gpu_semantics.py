@@ -1066,6 +1066,8 @@ def test_over_golden_reference(ctx, user_table_1):
     assert got["R2"].astype(int).tolist() == [3, 1, 2, 4]
     got2 = c.sql(
         'SELECT user_id, b, '
+        'FIRST_VALUE(user_id*10 - b) OVER '
+        '(PARTITION BY user_id ORDER BY b) AS "O2", '
         'SUM(user_id) OVER (PARTITION BY user_id ORDER BY b) AS "O5", '
         'AVG(user_id) OVER (PARTITION BY user_id ORDER BY b) AS "O6", '
         'COUNT(*) OVER (PARTITION BY user_id ORDER BY b) AS "O7", '
@@ -1073,6 +1075,7 @@ def test_over_golden_reference(ctx, user_table_1):
         'MAX(b) OVER (PARTITION BY user_id ORDER BY b) AS "O8", '
         'MIN(b) OVER (PARTITION BY user_id ORDER BY b) AS "O9" '
         'FROM user_table_1').compute()
+    assert got2["O2"].astype(int).tolist() == [19, 7, 19, 27]
     assert got2["O5"].astype(int).tolist() == [4, 1, 2, 3]
     assert got2["O6"].astype(float).tolist() == [2.0, 1.0, 2.0, 3.0]
     assert got2["O7"].astype(int).tolist() == [2, 1, 1, 1]
