@@ -1204,3 +1204,14 @@ def test_drop_and_analyze_table(ctx):
     import pytest
     with pytest.raises(KeyError):
         c.sql("SELECT * FROM t")
+
+
+def test_order_limit_offset_at_scale(ctx):
+    """LIMIT + OFFSET through the sampled top-k path."""
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(81)
+    df = pd.DataFrame({"t": rng.permutation(200_000).astype(np.int64)})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT t FROM t ORDER BY t LIMIT 5 OFFSET 7").compute()
+    assert got["t"].astype(int).tolist() == [7, 8, 9, 10, 11]
