@@ -65,10 +65,20 @@ def _from_pandas(df: pd.DataFrame, date_columns=(), dictionaries=None):
             dtype, sqlt = rt.I32, "VARCHAR"
             dictionary = list(uniques)
         elif np.issubdtype(dt, np.datetime64):
-            vals = s.to_numpy().astype("datetime64[D]").astype(np.int32)
             nat = s.isna().to_numpy()
             validity = (~nat).astype(np.uint8) if nat.any() else None
-            arr, dtype, sqlt = vals, rt.I32, "DATE"
+            ns = s.to_numpy().astype("datetime64[ns]").astype(np.int64)
+            day_ns = 86_400_000_000_000
+            if ((ns % day_ns == 0) | nat).all():
+                # midnight-only → day-int DATE (compact; reference
+                # mappings.py:78-80 semantics are identical)
+                vals = s.to_numpy().astype("datetime64[D]").astype(np.int32)
+                arr, dtype, sqlt = vals, rt.I32, "DATE"
+            else:
+                # sub-day precision → ns-resolution TIMESTAMP (i64, exactly
+                # the reference's datetime64[ns])
+                arr = np.where(nat, 0, ns)
+                dtype, sqlt = rt.I64, "TIMESTAMP"
         elif np.issubdtype(dt, np.floating):
             arr = s.to_numpy()
             # float NaN stays a VALUE (pandas semantics treat it as missing

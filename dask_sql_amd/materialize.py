@@ -55,6 +55,12 @@ def _convert(arr, valid, col, sql_t):
         if valid is not None:
             s[~valid] = pd.NaT
         return s
+    if sql_t == "TIMESTAMP":
+        s = pd.Series(pd.to_datetime(arr.astype("int64"), unit="ns",
+                                     errors="coerce"))
+        if valid is not None:
+            s[~valid] = pd.NaT
+        return s
     if valid is not None and not valid.all():
         # NULL-bearing numeric → float64 with NaN (pandas upcast semantics)
         out = arr.astype(np.float64)

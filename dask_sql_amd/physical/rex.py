@@ -417,10 +417,32 @@ class RexCompiler:
                "EXTRACT_MONTH": OP_MONTH, "MONTH": OP_MONTH,
                "EXTRACT_DAY": OP_DAY, "DAY": OP_DAY,
                "DAYOFMONTH": OP_DAY}
-        if op in _dx:
+        _tx = {"EXTRACT_HOUR": (3_600_000_000_000, 24),
+               "HOUR": (3_600_000_000_000, 24),
+               "EXTRACT_MINUTE": (60_000_000_000, 60),
+               "MINUTE": (60_000_000_000, 60),
+               "EXTRACT_SECOND": (1_000_000_000, 60),
+               "SECOND": (1_000_000_000, 60)}
+        if op in _dx or op in _tx:
+            t0 = getattr(ops[0], "getType", lambda: None)()
+            is_ts = t0 is not None and t0.getSqlType() == "TIMESTAMP"
             k = self.compile(ops[0])
             if k != KI:
-                raise RexCompileError(f"{op} needs a DATE (day-int) operand")
+                raise RexCompileError(f"{op} needs a DATE/TIMESTAMP operand")
+            if op in _tx:
+                if not is_ts:
+                    raise RexCompileError(f"{op} needs a TIMESTAMP operand")
+                unit_ns, modulus = _tx[op]
+                self._emit(OP_LIT_I64, 0, unit_ns)
+                self._emit(OP_DIV_I64)
+                self._emit(OP_LIT_I64, 0, modulus)
+                self._emit(OP_MOD_I64)
+                return KI
+            if is_ts:
+                # ns → days (trunc; pre-1970 sub-day values shift a day —
+                # documented)
+                self._emit(OP_LIT_I64, 0, 86_400_000_000_000)
+                self._emit(OP_DIV_I64)
             self._emit(_dx[op])
             return KI
         raise RexCompileError(f"operator {op} not supported on GPU path")

@@ -123,6 +123,9 @@ class Builder:
             _, v, t = ast
             if t == "DATE":
                 return Literal(_date_to_days(v), SqlType("DATE"))
+            if t == "TIMESTAMP":
+                ns = int(np.datetime64(v, "ns").astype("int64"))
+                return Literal(ns, SqlType("TIMESTAMP"))
             if t == "NULL":
                 return Literal(None, SqlType("NULL"))
             return Literal(v, SqlType(t))
@@ -149,6 +152,21 @@ class Builder:
                     isinstance(o, Literal)
                     and o.getType().getSqlType() == "INTERVAL" for o in ops):
                 return self._date_interval(op, ops)
+            if op in ("=", "<>", "<", "<=", ">", ">=") and len(ops) == 2:
+                ta, tb = _expr_type(ops[0]), _expr_type(ops[1])
+                if {ta, tb} == {"TIMESTAMP", "DATE"}:
+                    # promote the DATE side to ns so the compare is exact
+                    def _prom(o):
+                        if _expr_type(o) != "DATE":
+                            return o
+                        day_ns = 86_400_000_000_000
+                        if isinstance(o, Literal):
+                            return Literal(int(o.getValue()) * day_ns,
+                                           SqlType("TIMESTAMP"))
+                        return Call("*", [o, Literal(day_ns,
+                                                     SqlType("BIGINT"))],
+                                    SqlType("TIMESTAMP"))
+                    ops = [_prom(o) for o in ops]
             if op in ("=", "<>", "<", "<=", ">", ">=", "AND", "OR", "NOT",
                       "IS NULL", "IS NOT NULL", "LIKE"):
                 ty = "BOOLEAN"
@@ -163,6 +181,7 @@ class Builder:
                         "LOG", "POWER", "POW", "SQRT"):
                 ty = "DOUBLE"
             elif op in ("EXTRACT_YEAR", "EXTRACT_MONTH", "EXTRACT_DAY",
+                        "EXTRACT_HOUR", "EXTRACT_MINUTE", "EXTRACT_SECOND",
                         "MOD", "YEAR", "MONTH", "DAY", "DAYOFMONTH"):
                 ty = "BIGINT"
             elif op == "ABS":
@@ -275,16 +294,25 @@ class Builder:
             raise NotImplementedError("INTERVAL on the left of +/-")
         n_, unit = iv.getValue()
         sign = 1 if op == "+" else -1
+        is_ts = _expr_type(other) == "TIMESTAMP"
         if unit in ("DAY", "WEEK"):
             days = n_ * (7 if unit == "WEEK" else 1)
+            step = days * (86_400_000_000_000 if is_ts else 1)
+            out_t = "TIMESTAMP" if is_ts else "DATE"
             if isinstance(other, Literal):
-                return Literal(int(other.getValue()) + sign * days,
-                               SqlType("DATE"))
-            return Call(op, [other, Literal(days, SqlType("BIGINT"))],
-                        SqlType("DATE"))
+                return Literal(int(other.getValue()) + sign * step,
+                               SqlType(out_t))
+            return Call(op, [other, Literal(step, SqlType("BIGINT"))],
+                        SqlType(out_t))
         if not isinstance(other, Literal):
             raise NotImplementedError(
                 "MONTH/YEAR interval arithmetic on a column")
+        if is_ts:
+            import pandas as pd
+            months = sign * n_ * (12 if unit == "YEAR" else 1)
+            nd = pd.Timestamp(int(other.getValue())) + \
+                pd.DateOffset(months=months)
+            return Literal(int(nd.value), SqlType("TIMESTAMP"))
         import calendar
         import datetime
         d = datetime.date(1970, 1, 1) + datetime.timedelta(

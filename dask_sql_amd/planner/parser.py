@@ -447,6 +447,12 @@ class Parser:
                 if s[0] != "str":
                     raise ValueError("DATE needs a string literal")
                 return ("lit", s[1], "DATE")
+            if t[1] == "TIMESTAMP":
+                self.next()
+                s = self.next()
+                if s[0] != "str":
+                    raise ValueError("TIMESTAMP needs a string literal")
+                return ("lit", s[1], "TIMESTAMP")
             if t[1] == "INTERVAL":
                 # INTERVAL '<n>' DAY|WEEK|MONTH|YEAR (TPC-H date arithmetic)
                 self.next()
@@ -518,7 +524,8 @@ class Parser:
                 self.next()
                 ft = self.next()
                 field = ft[1].upper()
-                if field not in ("YEAR", "MONTH", "DAY"):
+                if field not in ("YEAR", "MONTH", "DAY", "HOUR", "MINUTE",
+                                 "SECOND"):
                     raise ValueError(f"EXTRACT({field}) not supported")
                 self.expect_kw("FROM")
                 e = self.expr()

@@ -1215,3 +1215,34 @@ def test_order_limit_offset_at_scale(ctx):
     c.create_table("t", df)
     got = c.sql("SELECT t FROM t ORDER BY t LIMIT 5 OFFSET 7").compute()
     assert got["t"].astype(int).tolist() == [7, 8, 9, 10, 11]
+
+
+def test_timestamp_columns(ctx):
+    """Sub-day datetimes keep ns precision (TIMESTAMP i64) — comparisons
+    vs TIMESTAMP/DATE literals, intervals, EXTRACT time units, grouping."""
+    from dask_sql_amd.context import Context
+    ts = pd.to_datetime(["1994-01-01 08:30:15", "1994-01-01 17:45:00",
+                         "1994-01-02 00:00:01", "1995-06-15 12:00:00"])
+    df = pd.DataFrame({"ts": ts, "v": [1, 2, 3, 4]})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT ts, v FROM t WHERE ts > "
+                "TIMESTAMP '1994-01-01 12:00:00'").compute()
+    assert sorted(got["v"].astype(int).tolist()) == [2, 3, 4]
+    assert list(got.sort_values("v")["ts"]) == list(ts[1:])  # ns preserved
+    got2 = c.sql("SELECT v FROM t WHERE ts < DATE '1994-01-02'").compute()
+    assert sorted(got2["v"].astype(int).tolist()) == [1, 2]
+    got3 = c.sql("SELECT EXTRACT(HOUR FROM ts) AS h, "
+                 "EXTRACT(MINUTE FROM ts) AS m, "
+                 "EXTRACT(SECOND FROM ts) AS s, "
+                 "EXTRACT(YEAR FROM ts) AS y FROM t").compute()
+    assert got3["h"].astype(int).tolist() == [8, 17, 0, 12]
+    assert got3["m"].astype(int).tolist() == [30, 45, 0, 0]
+    assert got3["s"].astype(int).tolist() == [15, 0, 1, 0]
+    assert got3["y"].astype(int).tolist() == [1994, 1994, 1994, 1995]
+    got4 = c.sql("SELECT v FROM t WHERE ts >= TIMESTAMP '1994-01-01 00:00:00'"
+                 " + INTERVAL '1' DAY").compute()
+    assert sorted(got4["v"].astype(int).tolist()) == [3, 4]
+    got5 = c.sql("SELECT EXTRACT(YEAR FROM ts) AS y, COUNT(*) AS c FROM t "
+                 "GROUP BY EXTRACT(YEAR FROM ts) ORDER BY y").compute()
+    assert got5["c"].astype(int).tolist() == [3, 1]
