@@ -1246,3 +1246,27 @@ def test_timestamp_columns(ctx):
     got5 = c.sql("SELECT EXTRACT(YEAR FROM ts) AS y, COUNT(*) AS c FROM t "
                  "GROUP BY EXTRACT(YEAR FROM ts) ORDER BY y").compute()
     assert got5["c"].astype(int).tolist() == [3, 1]
+
+
+def test_stddev_with_filter_clause(ctx):
+    """STDDEV FILTER (WHERE ...) + plain aggs: the multi-bucket host merge
+    must align the (Σx, Σx²) moment tuple across group sets
+    (aggregate.py:336-374 bucket merge)."""
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(91)
+    df = pd.DataFrame({"k": rng.integers(0, 8, 4000).astype(np.int64),
+                       "v": np.round(rng.random(4000) * 10, 3),
+                       "w": rng.integers(0, 10, 4000).astype(np.int64)})
+    c = Context()
+    c.create_table("t", df)
+    got = c.sql("SELECT k, SUM(v) AS s, "
+                "STDDEV(v) FILTER (WHERE w > 5) AS sd FROM t GROUP BY k"
+                ).compute().sort_values("k").reset_index(drop=True)
+    exp_s = df.groupby("k")["v"].sum()
+    exp_sd = df[df.w > 5].groupby("k")["v"].std()
+    np.testing.assert_allclose(got["s"].to_numpy(np.float64),
+                               exp_s.to_numpy(), rtol=1e-9)
+    g = got["sd"].to_numpy(np.float64)
+    e = exp_sd.reindex(exp_s.index).to_numpy()
+    ok = np.isclose(g, e, rtol=1e-8) | (np.isnan(g) & np.isnan(e))
+    assert ok.all()
