@@ -276,3 +276,88 @@ def test_show_commands():
     import pytest
     with pytest.raises(KeyError):
         c.sql('SHOW COLUMNS FROM "root"."missing"')
+
+
+PLAN_BATTERY = [
+    # every supported construct must at least PLAN (conversion is gpu-only)
+    "SELECT a, b FROM t WHERE a > 1 AND b < 2.5",
+    "SELECT a + b * 2 - 1 AS x, a % 3 AS m FROM t",
+    "SELECT COUNT(*), SUM(a), AVG(b), MIN(a), MAX(b) FROM t",
+    "SELECT a, STDDEV(b), VAR_POP(b) FROM t GROUP BY a HAVING COUNT(*) > 2",
+    "SELECT a, SUM(b) FILTER (WHERE a > 0) FROM t GROUP BY a",
+    "SELECT COUNT(DISTINCT a) FROM t",
+    "SELECT DISTINCT a, b FROM t",
+    "SELECT t.a, u.d FROM t JOIN u ON t.a = u.c",
+    "SELECT t.a FROM t LEFT JOIN u ON t.a = u.c AND u.d > 1",
+    "SELECT t.a FROM t LEFT SEMI JOIN u ON t.a = u.c",
+    "SELECT t.a FROM t LEFT ANTI JOIN u ON t.a = u.c",
+    "SELECT * FROM t, u WHERE t.a = u.c",
+    "SELECT a FROM t WHERE a IN (1, 2, 3) OR b BETWEEN 0 AND 1",
+    "SELECT a FROM t WHERE a IN (SELECT c FROM u WHERE d > 0)",
+    "SELECT a FROM t WHERE NOT EXISTS (SELECT 1 FROM u WHERE u.c = t.a)",
+    "SELECT a FROM t WHERE b > (SELECT AVG(b) FROM t)",
+    "SELECT x.a2 FROM (SELECT a AS a2 FROM t WHERE b > 0) x",
+    "SELECT a FROM t UNION ALL SELECT c FROM u ORDER BY a LIMIT 3",
+    "SELECT a, ROW_NUMBER() OVER (PARTITION BY a ORDER BY b DESC) FROM t",
+    "SELECT SUM(b) OVER (PARTITION BY a) AS s FROM t",
+    "SELECT LAG(b, 2, 0.0) OVER (PARTITION BY a ORDER BY b) FROM t",
+    "SELECT FIRST_VALUE(b) OVER (PARTITION BY a ORDER BY b) FROM t",
+    "SELECT CASE WHEN a > 1 THEN 'x' ELSE 'y' END FROM t",
+    "SELECT CASE a WHEN 1 THEN 10 ELSE 0 END FROM t",
+    "SELECT COALESCE(a, 0), NULLIF(a, 1) FROM t",
+    "SELECT ABS(a), FLOOR(b), CEIL(b), ROUND(b, 2), EXP(b), LN(b + 1), "
+    "POWER(b, 2), SQRT(ABS(b)), MOD(a, 2) FROM t",
+    "SELECT CAST(b AS BIGINT), CAST(a AS DOUBLE) FROM t",
+    "SELECT 1 + 1 AS two",
+    "SELECT a FROM t ORDER BY a DESC NULLS LAST LIMIT 10 OFFSET 5",
+    "SELECT a, SUM(b) AS s FROM t GROUP BY a ORDER BY SUM(b) DESC LIMIT 5",
+]
+
+
+def test_plan_battery():
+    import pandas as pd
+    from dask_sql_amd.context import Context
+    c = Context()
+    c.create_table("t", pd.DataFrame({"a": [1, 2], "b": [0.5, 1.5]}))
+    c.create_table("u", pd.DataFrame({"c": [1], "d": [2]}))
+    for q in PLAN_BATTERY:
+        c.explain(q)  # must not raise
+
+
+def test_plan_battery_dates():
+    import pandas as pd
+    from dask_sql_amd.context import Context
+    c = Context()
+    c.create_table("t", pd.DataFrame(
+        {"d": pd.to_datetime(["2020-01-01"]),
+         "ts": pd.to_datetime(["2020-01-01 10:30:00"]),
+         "s": pd.Series(["x"]).astype("category"), "v": [1.0]}))
+    for q in [
+        "SELECT v FROM t WHERE d >= DATE '2020-01-01' - INTERVAL '90' DAY",
+        "SELECT v FROM t WHERE d < DATE '2020-01-01' + INTERVAL '2' MONTH",
+        "SELECT EXTRACT(YEAR FROM d), YEAR(d), MONTH(d), DAY(d) FROM t",
+        "SELECT EXTRACT(HOUR FROM ts), EXTRACT(MINUTE FROM ts) FROM t",
+        "SELECT v FROM t WHERE ts > TIMESTAMP '2020-01-01 09:00:00'",
+        "SELECT v FROM t WHERE ts > DATE '2020-01-01'",
+        "SELECT UPPER(s), LOWER(s), SUBSTRING(s, 1, 1), s || '!' FROM t",
+        "SELECT TRIM('x' FROM s), INITCAP(s), REPLACE(s, 'x', 'y'), "
+        "CHAR_LENGTH(s) FROM t",
+        "SELECT v FROM t WHERE s LIKE 'x%' AND UPPER(s) = 'X'",
+    ]:
+        c.explain(q)
+
+
+def test_plan_errors_are_loud():
+    import pandas as pd
+    import pytest
+    from dask_sql_amd.context import Context
+    c = Context()
+    c.create_table("t", pd.DataFrame({"a": [1], "b": [2]}))
+    with pytest.raises(KeyError):
+        c.explain("SELECT missing FROM t")
+    with pytest.raises(KeyError):
+        c.explain("SELECT a FROM missing_table")
+    with pytest.raises(ValueError):
+        c.explain("SELECT ROW_NUMBER() OVER (PARTITION BY a) FROM t")
+    with pytest.raises((ValueError, NotImplementedError)):
+        c.explain("SELECT a FROM t WHERE a IN (SELECT a, b FROM t)")
