@@ -789,7 +789,11 @@ class DaskAggregatePlugin(BaseRelPlugin):
                 group_idx.append(e.getIndex())
             agg_calls = agg.getNamedAggCalls()
 
-        # key specs from minmax (float key: bit-pattern mode, single key)
+        # key specs from minmax (float key: bit-pattern mode, single key;
+        # floats inside composite key sets are densified to integer ids)
+        name_idx = list(group_idx)
+        cols, group_idx, key_restore = self._densify_float_keys(
+            runtime, cols, group_idx)
         keyspecs = self._keyspecs_for(runtime, cols, group_idx)
 
         # bucket aggs by (filter_col_index, distinct) — aggregate.py:377-520
@@ -812,7 +816,8 @@ class DaskAggregatePlugin(BaseRelPlugin):
             calls = list(agg_calls)
             filt_idx = next(iter(buckets.keys()))[0] if buckets else None
             return self._convert_device(runtime, rel, dc, cols, keyspecs,
-                                        group_idx, filt_idx, calls, agg)
+                                        group_idx, filt_idx, calls, agg,
+                                        name_idx, key_restore)
 
         # GENERAL PATH: several (filter, distinct) buckets → merge on host
         # like the reference's multi-pass _do_aggregations (aggregate.py:336+)
@@ -858,7 +863,8 @@ class DaskAggregatePlugin(BaseRelPlugin):
                         merged[call.toString()] = (align(vals), ac)
 
         return self._build_output(runtime, rel, dc, keyspecs, group_idx,
-                                  codes_np, agg_calls, merged)
+                                  codes_np, agg_calls, merged,
+                                  name_idx=name_idx, key_restore=key_restore)
 
     # ------------------------------------------------------------------
     def _try_fused(self, runtime, rel, agg, context):
@@ -941,17 +947,19 @@ class DaskAggregatePlugin(BaseRelPlugin):
 
     @staticmethod
     def _keyspecs_for(runtime, cols, group_idx):
-        """(idx, min, range, nullable[, mode]) per key. Float keys group by
-        canonical f64 bit pattern (mode 1) through the CAS hash path —
-        pandas float group keys compare exactly, NaN is one group under
-        dropna=False (aggregate.py:575-577)."""
+        """(idx, min, range, nullable[, mode]) per key. A single float key
+        groups by canonical f64 bit pattern (mode 1) through the CAS hash
+        path; float keys in COMPOSITE key sets are densified first
+        (_densify_float_keys) — pandas float group keys compare exactly,
+        NaN is one group under dropna=False (aggregate.py:575-577)."""
         keyspecs = []
         for gi in group_idx:
             col = cols[gi]
             if col.dtype in (rt.F64, rt.F32):
                 if len(group_idx) != 1:
                     raise RexCompileError(
-                        "a float GROUP BY key must be the only key")
+                        "a float GROUP BY key must be the only key "
+                        "(densify composites first)")
                 keyspecs.append((gi, 0, 0, True, 1))
                 continue
             if col.dtype not in _INT_KINDS:
@@ -964,8 +972,71 @@ class DaskAggregatePlugin(BaseRelPlugin):
             keyspecs.append((gi, mn, mx - mn + 1, nullable))
         return keyspecs
 
+    @staticmethod
+    def _densify_float_keys(runtime, cols, group_idx):
+        """Composite key sets with float columns: each float key becomes a
+        dense integer id (distinct floats via the bits-mode groupby, ids
+        assigned by table position, joined back per row) so radix packing
+        applies; the output restores the float via a host LUT. Returns
+        (cols2, group_idx2, restore {key_pos: np.float64 LUT})."""
+        if len(group_idx) <= 1 or not any(
+                cols[gi].dtype in (rt.F64, rt.F32) for gi in group_idx):
+            return cols, group_idx, {}
+        cols = list(cols)
+        group_idx = list(group_idx)
+        restore = {}
+        for j, gi in enumerate(group_idx):
+            col = cols[gi]
+            if col.dtype not in (rt.F64, rt.F32):
+                continue
+            oc, ov, on, G = runtime.hash_groupby(
+                [col], col.len, [(0, 0, 0, True, 1)], None, [])
+            runtime._free(ov)
+            runtime._free(on)
+            codes = np.empty(max(G, 1), dtype=np.uint64)
+            if G:
+                runtime._download(oc, codes)
+            codes = codes[:G]
+            lut = np.where(codes > 0, codes - 1, 0).astype(
+                np.uint64).view(np.float64)
+            lut = np.where(codes > 0, lut, np.nan)
+
+            class _H:
+                def __init__(s, rt_, ptrs):
+                    s.rt = rt_
+                    s.ptrs = ptrs
+
+                def __del__(s):
+                    for p in s.ptrs:
+                        try:
+                            s.rt._free(p)
+                        except Exception:
+                            pass
+
+            h = _H(runtime, [oc])
+            bcol = rt.DeviceColumn(runtime, oc, None, G, rt.I64,
+                                   owner=False, keep_alive=h)
+            pcodes, _ = runtime.keypack([col], [(0, 0, 0, True, 1)],
+                                        col.len)
+            table = runtime.hash_build(bcol, None, code_max=0)
+            try:
+                p, b, cnt = runtime.hash_probe(table, pcodes,
+                                               rt.JOIN_INNER, None)
+                psel = runtime.wrap_sel(p, cnt)
+                bsel = runtime.wrap_sel(b, cnt)
+                assert cnt == col.len, (cnt, col.len)
+                fid = runtime.scatter_rows(bsel, psel.data, cnt, col.len,
+                                           with_validity=False)
+            finally:
+                runtime.hash_table_free(table)
+            cols.append(fid)
+            restore[j] = lut
+            group_idx[j] = len(cols) - 1
+        return cols, group_idx, restore
+
     def _convert_device(self, runtime, rel, dc, cols, keyspecs, group_idx,
-                        filt_idx, calls, agg):
+                        filt_idx, calls, agg, name_idx=None,
+                        key_restore=None):
         """Single-bucket kernel over the already-converted input."""
         specs = []
         fins = []
@@ -979,13 +1050,17 @@ class DaskAggregatePlugin(BaseRelPlugin):
         pred_prog = runtime.make_prog([(OP_COL, filt_idx, 0)]) \
             if filt_idx is not None else None
         cc_in = dc.column_container
-        group_meta = [(cc_in.columns[gi], cols[gi]) for gi in group_idx]
+        names = name_idx if name_idx is not None else group_idx
+        group_meta = [(cc_in.columns[ni], cols[gi])
+                      for ni, gi in zip(names, group_idx)]
         return self._device_exec(runtime, rel, cols, dc.table.num_rows,
                                  keyspecs, group_meta, pred_prog, calls,
-                                 specs, fins, slab)
+                                 specs, fins, slab,
+                                 key_restore=key_restore or {})
 
     def _device_exec(self, runtime, rel, cols, n_rows, keyspecs, group_meta,
-                     pred_prog, calls, specs, fins, slab=None):
+                     pred_prog, calls, specs, fins, slab=None,
+                     key_restore=None):
         """Run the fused kernel; finalize device-resident.
         slab[i] = first kernel-slot index of call i (stddev family spans 2)."""
         from dask_sql_amd.physical.rex import (OP_LIT_I64, OP_GT_I64,
@@ -1044,8 +1119,33 @@ class DaskAggregatePlugin(BaseRelPlugin):
         # group keys: unpack on device — part = (code / stride) % space
         from dask_sql_amd.physical.rex import OP_BITS_F64, OP_SUB_I64
         stride = 1
+        key_restore = key_restore or {}
         for j, (ks, (name, src)) in enumerate(zip(keyspecs, group_meta)):
             gi, mn, rng, nullable = ks[:4]
+            if j in key_restore:
+                # densified float key: unpack the dense id, restore the
+                # float through the host LUT (small: one value per distinct)
+                space = rng + (1 if nullable else 0)
+                prog = [(OP_COL, 0, 0), (OP_LIT_I64, 0, stride), (17, 0, 0),
+                        (OP_LIT_I64, 0, space), (18, 0, 0),
+                        (OP_LIT_I64, 0, mn), (14, 0, 0)]
+                stride *= space
+                fid_col = runtime.eval(runtime.make_prog(prog), [codes_col],
+                                       G, rt.I64, with_validity=False)
+                fid = np.empty(G, dtype=np.int64)
+                if G:
+                    runtime._download(fid_col.data, fid)
+                lut = key_restore[j]
+                vals = lut[np.clip(fid, 0, max(len(lut) - 1, 0))] \
+                    if len(lut) else np.full(G, np.nan)
+                nanm = np.isnan(vals)
+                col = runtime.upload_column(
+                    np.where(nanm, 0.0, vals),
+                    validity=(~nanm).astype(np.uint8) if nanm.any()
+                    else None)
+                out_cols[f"g__{name}"] = col
+                order_names.append((name, f"g__{name}"))
+                continue
             if len(ks) > 4 and ks[4] == 1:
                 # f64-bits key: value = bitcast(code-1); code 0 = NaN/NULL
                 prog = [(OP_COL, 0, 0), (OP_LIT_I64, 0, 0),
@@ -1337,7 +1437,7 @@ class DaskAggregatePlugin(BaseRelPlugin):
 
     # ------------------------------------------------------------------
     def _build_output(self, runtime, rel, dc, keyspecs, group_idx, codes_np,
-                      agg_calls, merged):
+                      agg_calls, merged, name_idx=None, key_restore=None):
         """Unpack group codes → key columns; finalize agg columns
         (SUM min_count=1 → NULL on zero count; AVG = sum/count)."""
         G = len(codes_np)
@@ -1361,8 +1461,27 @@ class DaskAggregatePlugin(BaseRelPlugin):
 
         # group key columns
         stride = 1
+        key_restore = key_restore or {}
+        names = name_idx if name_idx is not None else group_idx
         for j, ks in enumerate(keyspecs):
             gi, mn, rng, nullable = ks[:4]
+            ni = names[j]
+            if j in key_restore:
+                space = rng + (1 if nullable else 0)
+                fid = ((codes_np // stride) % space).astype(np.int64) + mn
+                stride *= space
+                lut = key_restore[j]
+                vals = lut[np.clip(fid, 0, max(len(lut) - 1, 0))] \
+                    if len(lut) else np.full(G, np.nan)
+                nanm = np.isnan(vals)
+                col = runtime.upload_column(
+                    np.where(nanm, 0.0, vals),
+                    validity=(~nanm).astype(np.uint8) if nanm.any()
+                    else None)
+                name = cc_in.columns[ni]
+                out_cols[f"g__{name}"] = col
+                order_names.append((name, f"g__{name}"))
+                continue
             if len(ks) > 4 and ks[4] == 1:
                 # f64-bits key (host path): bitcast(code-1), code 0 → NaN
                 src = cols[gi]
@@ -1371,7 +1490,7 @@ class DaskAggregatePlugin(BaseRelPlugin):
                                 np.uint64(0)).view(np.float64)
                 vals = np.where(codes_np > 0, vals, np.nan)
                 col = runtime.upload_column(vals.astype(np.float64))
-                name = cc_in.columns[gi]
+                name = cc_in.columns[ni]
                 out_cols[f"g__{name}"] = col
                 order_names.append((name, f"g__{name}"))
                 continue
@@ -1394,7 +1513,7 @@ class DaskAggregatePlugin(BaseRelPlugin):
                 dtype=src.dtype)
             if getattr(src, "dictionary", None) is not None:
                 col.dictionary = src.dictionary
-            name = cc_in.columns[gi]
+            name = cc_in.columns[ni]
             out_cols[f"g__{name}"] = col
             order_names.append((name, f"g__{name}"))
 
