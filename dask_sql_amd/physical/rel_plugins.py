@@ -72,6 +72,40 @@ def _empty_like(runtime, dc: DataContainer) -> DataContainer:
     return _gather_table(runtime, dc, sel.data, 0)
 
 
+def _case_string_rewrite(expr):
+    """CASE whose value branches are string literals (or NULL) → the same
+    CASE over dictionary codes + the dictionary (rex/core/call.py CASE with
+    string outputs). Returns (rewritten Call, dictionary) or None."""
+    from dask_sql_amd.planner.plan import Literal as PLit
+    from dask_sql_amd.planner.plan import SqlType
+    if not (isinstance(expr, Call) and expr.getOperatorName() == "CASE"):
+        return None
+    ops = expr.getOperands()
+    vpos = list(range(1, len(ops) - 1, 2)) + [len(ops) - 1]
+    has_str = False
+    for p in vpos:
+        o = ops[p]
+        if not isinstance(o, Literal):
+            return None
+        v = o.getValue()
+        if isinstance(v, str):
+            has_str = True
+        elif v is not None:
+            return None
+    if not has_str:
+        return None
+    d = []
+    new_ops = list(ops)
+    for p in vpos:
+        v = ops[p].getValue()
+        if v is None:
+            continue
+        if v not in d:
+            d.append(v)
+        new_ops[p] = PLit(d.index(v), SqlType("BIGINT"))
+    return Call("CASE", new_ops, SqlType("BIGINT")), d
+
+
 def _resolve_scalar_subs(expr, context):
     """Replace ScalarSub nodes with the Literal their (eagerly executed)
     subplan yields — ≤1 row enforced; 0 rows → NULL (SQL scalar subquery)."""
@@ -194,8 +228,11 @@ class DaskProjectPlugin(BaseRelPlugin):
             ifn = None if isinstance(expr, InputRef) or sfn is not None \
                 else dict_int_fn(expr, dicts)
             const_s = None
+            csr = None
             if not isinstance(expr, InputRef) and sfn is None and ifn is None:
                 const_s = fold_string_literal(expr)
+                if const_s is None:
+                    csr = _case_string_rewrite(expr)
             if isinstance(expr, InputRef):
                 src = cols[expr.getIndex()]
                 out_cols[backend_name] = src  # zero-copy reuse
@@ -217,6 +254,15 @@ class DaskProjectPlugin(BaseRelPlugin):
                 col = runtime.upload_column(
                     np.zeros(n_, dtype=np.int32), dtype=rt.I32)
                 col.dictionary = [const_s]
+                out_cols[backend_name] = col
+            elif csr is not None:
+                # string-valued CASE → integer-code CASE + dictionary
+                e2, d = csr
+                prog, _k = compile_expr(e2, cols, dicts)
+                col = runtime.eval(runtime.make_prog(prog), cols,
+                                   dc.table.num_rows, rt.I64,
+                                   with_validity=True)
+                col.dictionary = d
                 out_cols[backend_name] = col
             elif sfn is not None:
                 # string function over a dict column: same codes, the
