@@ -85,7 +85,7 @@ def _case_string_rewrite(expr):
     has_str = False
     for p in vpos:
         o = ops[p]
-        if not isinstance(o, Literal):
+        if not isinstance(o, PLit):
             return None
         v = o.getValue()
         if isinstance(v, str):
