@@ -361,3 +361,50 @@ def test_plan_errors_are_loud():
         c.explain("SELECT ROW_NUMBER() OVER (PARTITION BY a) FROM t")
     with pytest.raises((ValueError, NotImplementedError)):
         c.explain("SELECT a FROM t WHERE a IN (SELECT a, b FROM t)")
+
+
+def test_interval_folding_unit():
+    """_date_interval folds literal date arithmetic exactly (day-int and ns
+    forms; month-end clamping like pandas DateOffset)."""
+    import numpy as np
+    from dask_sql_amd.planner.builder import Builder, _date_to_days
+    from dask_sql_amd.planner.plan import Literal, SqlType
+
+    def date_lit(s):
+        return Literal(_date_to_days(s), SqlType("DATE"))
+
+    def iv(n, u):
+        return Literal((n, u), SqlType("INTERVAL"))
+
+    r = Builder._date_interval("-", [date_lit("1998-12-01"), iv(90, "DAY")])
+    assert r.getValue() == _date_to_days("1998-09-02")
+    r = Builder._date_interval("+", [date_lit("1996-01-31"), iv(1, "MONTH")])
+    assert r.getValue() == _date_to_days("1996-02-29")  # leap clamp
+    r = Builder._date_interval("+", [date_lit("1995-03-15"), iv(2, "WEEK")])
+    assert r.getValue() == _date_to_days("1995-03-29")
+    r = Builder._date_interval("-", [date_lit("2000-03-31"), iv(1, "YEAR")])
+    assert r.getValue() == _date_to_days("1999-03-31")
+    ns = int(np.datetime64("2020-01-01T06:00:00", "ns").astype("int64"))
+    r = Builder._date_interval(
+        "+", [Literal(ns, SqlType("TIMESTAMP")), iv(1, "DAY")])
+    assert r.getValue() == ns + 86_400_000_000_000
+    assert r.getType().getSqlType() == "TIMESTAMP"
+
+
+def test_fold_string_literal_unit():
+    from dask_sql_amd.physical.rex import fold_string_literal
+    from dask_sql_amd.planner.plan import Call, Literal, SqlType
+
+    def lit(v, t="VARCHAR"):
+        return Literal(v, SqlType(t))
+
+    assert fold_string_literal(
+        Call("REPLACE", [lit("Another String"), lit("th"), lit("b")])
+    ) == "Anober String"  # reference test_rex.py:660 "x"
+    assert fold_string_literal(
+        Call("UPPER", [Call("SUBSTRING",
+                            [lit("abcdef"), lit(2, "BIGINT"),
+                             lit(3, "BIGINT")])])) == "BCD"
+    assert fold_string_literal(
+        Call("CONCAT", [lit("a"), lit("b"), lit("c")])) == "abc"
+    assert fold_string_literal(Call("+", [lit("a"), lit("b")])) is None
