@@ -176,20 +176,15 @@ def run_step(c, workload, world, pg):
 
 def _q1_allgather_merge(c, res, pg):
     """Q1 partials are ≤ 12 rows — a host gather is the cheap correct merge
-    (Q1 needs only a trivially small reduce)."""
+    (Q1 needs only a trivially small reduce). AVG columns
+    recombine count-weighted (distributed.q1_merge_partials)."""
     import torch.distributed as dist
+
+    from dask_sql_amd.distributed import q1_merge_partials
     pdf = res.compute()
     gathered = [None] * dist.get_world_size(pg)
     dist.all_gather_object(gathered, pdf, group=pg)
-    import pandas as pd
-    allp = pd.concat(gathered)
-    g = allp.groupby(["l_returnflag", "l_linestatus"], dropna=False)
-    merged = g.agg(
-        sum_qty=("sum_qty", "sum"), sum_base_price=("sum_base_price", "sum"),
-        sum_disc_price=("sum_disc_price", "sum"),
-        sum_charge=("sum_charge", "sum"),
-        count_order=("count_order", "sum"))
-    return merged
+    return q1_merge_partials(gathered)
 
 
 def cpu_baseline_leg(workload):

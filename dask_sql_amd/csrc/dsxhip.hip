@@ -408,6 +408,7 @@ __device__ __forceinline__ bool vm_eval(const DsxInstr* prog, int len, const Col
       BIN_F(DSX_OP_MUL_I64, i = a.i * b.i)
       BIN_F(DSX_OP_DIV_I64, i = b.i ? a.i / b.i : 0)
       BIN_F(DSX_OP_MOD_I64, i = b.i ? a.i % b.i : 0)
+      BIN_F(DSX_OP_FLOORMOD_I64, i = b.i ? ((a.i % b.i) + b.i) % b.i : 0)
       BIN_F(DSX_OP_LT_I64, i = (a.i < b.i) ? 1 : 0)
       BIN_F(DSX_OP_LE_I64, i = (a.i <= b.i) ? 1 : 0)
       BIN_F(DSX_OP_GT_I64, i = (a.i > b.i) ? 1 : 0)

@@ -15,8 +15,9 @@ from dask_sql_amd.datacontainer import (ColumnContainer, DataContainer,
                                         DeviceTable, HostDataContainer)
 from dask_sql_amd.physical.convert import BaseRelPlugin, RelConverter
 from dask_sql_amd.physical.rex import (KB, KF, KI, OP_AND, OP_COL,
-                                       OP_IS_NOT_NULL, RexCompileError,
-                                       compile_expr, scalar_literal)
+                                       OP_IS_NOT_NULL, OP_NOT,
+                                       RexCompileError, compile_expr,
+                                       scalar_literal)
 from dask_sql_amd.planner.plan import Call, InputRef
 
 logger = logging.getLogger(__name__)
@@ -531,9 +532,21 @@ class DaskJoinPlugin(BaseRelPlugin):
         dc2 = DataContainer(DeviceTable(cols2), cc)
         return dc2, dc2.backend_cols()
 
-    def _key_codes(self, runtime, dc, on, ranges):
-        """Build i64 code column + optional validity for join keys."""
+    def _key_codes(self, runtime, dc, on, ranges, null_flags=None):
+        """Build i64 code column + optional validity for join keys.
+
+        null_flags (FULL OUTER only): per-key bools — NULL packs as its own
+        code slot so NULL keys MATCH each other, which is what the reference
+        produces (pandas merge how="outer" matches NA keys; join.py:202-213
+        drops NULL keys only for inner/left/right/semi). Returns
+        (codes, validity_ptr, keep, key_space)."""
         cols = dc.backend_cols()
+        if null_flags is not None:
+            keyspecs = [(idx, mn, rng, nf) for (idx, (mn, rng)), nf
+                        in zip(zip(on, ranges), null_flags)]
+            codes, space = runtime.keypack(cols, keyspecs,
+                                           dc.table.num_rows)
+            return codes, None, None, space
         keyspecs = []
         for (idx, (mn, rng)) in zip(on, ranges):
             keyspecs.append((idx, mn, rng, False))
@@ -554,7 +567,7 @@ class DaskJoinPlugin(BaseRelPlugin):
                                 with_validity=False)
             validity_ptr = vcol.data
             keep = vcol
-        return codes, validity_ptr, keep
+        return codes, validity_ptr, keep, space
 
     def _equi_join(self, runtime, dc_lhs, dc_rhs, lhs_on, rhs_on, join_type):
         lcols = dc_lhs.backend_cols()
@@ -599,14 +612,21 @@ class DaskJoinPlugin(BaseRelPlugin):
                 "outer": rt.JOIN_LEFT, "leftanti": rt.JOIN_LEFTANTI,
             }[join_type]
 
-        bcodes, bval, bkeep = self._key_codes(runtime, build_dc, build_on,
-                                              ranges)
-        pcodes, pval, pkeep = self._key_codes(runtime, probe_dc, probe_on,
-                                              ranges)
-        space = 1
-        for (_, rng) in ranges:
-            space *= rng
-        table = runtime.hash_build(bcodes, bval, code_max=space - 1)
+        # FULL OUTER: NULL keys get their own code slot so they MATCH each
+        # other — the reference keeps both sides' NULL-key rows and pandas
+        # merge how="outer" matches NA keys (join.py:202-213 drops NULL keys
+        # only for inner/left/right/semi). ADVICE r1 (medium).
+        null_flags = None
+        if join_type == "outer":
+            pkc, bkc = probe_dc.backend_cols(), build_dc.backend_cols()
+            null_flags = [
+                bool(pkc[pi].validity) or bool(bkc[bi].validity)
+                for pi, bi in zip(probe_on, build_on)]
+        bcodes, bval, bkeep, bspace = self._key_codes(
+            runtime, build_dc, build_on, ranges, null_flags)
+        pcodes, pval, pkeep, _ = self._key_codes(
+            runtime, probe_dc, probe_on, ranges, null_flags)
+        table = runtime.hash_build(bcodes, bval, code_max=bspace - 1)
         try:
             p_ptr, b_ptr, count = runtime.hash_probe(
                 table, pcodes, ktype, pval,
@@ -685,13 +705,10 @@ class DaskJoinPlugin(BaseRelPlugin):
                 "leftanti": rt.JOIN_LEFTANTI,
             }[join_type]
 
-        bcodes, bval, bkeep = self._key_codes(runtime, build_kdc, build_on,
-                                              ranges)
-        pcodes, pval, pkeep = self._key_codes(runtime, probe_kdc, probe_on,
-                                              ranges)
-        space = 1
-        for (_, rng) in ranges:
-            space *= rng
+        bcodes, bval, bkeep, space = self._key_codes(
+            runtime, build_kdc, build_on, ranges)
+        pcodes, pval, pkeep, _ = self._key_codes(
+            runtime, probe_kdc, probe_on, ranges)
         # assemble per-side materialization lists (sources: ORIGINAL tables)
         probe_is_l = not swap
         p_items, b_items = [], []

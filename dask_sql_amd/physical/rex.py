@@ -24,6 +24,7 @@ OP_SELECT, OP_NEG_F64, OP_NEG_I64, OP_SQRT_F64 = 60, 61, 62, 63
 OP_ABS_I64, OP_ABS_F64, OP_FLOOR_F64, OP_CEIL_F64 = 64, 65, 66, 67
 OP_RINT_F64, OP_EXP_F64, OP_LN_F64, OP_POW_F64 = 68, 69, 70, 71
 OP_YEAR, OP_MONTH, OP_DAY = 72, 73, 74
+OP_FLOORMOD_I64 = 75
 
 # VM value kinds
 KI, KF, KB = "i", "f", "b"  # int64-like, float64, boolean
@@ -409,9 +410,11 @@ class RexCompiler:
             self._emit(OP_POW_F64)
             return KF
         if op == "MOD":
+            # reference evaluates operator.mod on pandas = FLOOR-mod
+            # (MOD(-5,3) = 1), not C truncated remainder
             self.compile(ops[0])
             self.compile(ops[1])
-            self._emit(OP_MOD_I64)
+            self._emit(OP_FLOORMOD_I64)
             return KI
         _dx = {"EXTRACT_YEAR": OP_YEAR, "YEAR": OP_YEAR,
                "EXTRACT_MONTH": OP_MONTH, "MONTH": OP_MONTH,

@@ -419,9 +419,19 @@ class Builder:
         where_conjuncts = rest
         used = [False] * len(where_conjuncts)
 
+        # Pushing a WHERE conjunct below a scan is only valid when that table
+        # sits on the PRESERVED side of every join above it (DataFusion's
+        # PushDownFilter rule pushes only to preserved sides; a predicate on
+        # a null-supplying side must stay a post-join Filter, else
+        # NULL-extended rows that SQL excludes would be returned).
+        join_types = [jc.join_type for jc in stmt.joins]
+        lhs_preserved = not any(t in ("RIGHT", "FULL") for t in join_types)
+
         # 1. scans (+ pushed-down single-table filters, à la PushDownFilter)
-        def scan_with_filters(tr: TableRef) -> LogicalPlan:
+        def scan_with_filters(tr: TableRef, push: bool = True) -> LogicalPlan:
             plan = self._scan(tr)
+            if not push:
+                return plan
             quals = {(tr.alias or tr.name).lower()}
             conds = []
             for i, cj in enumerate(where_conjuncts):
@@ -441,7 +451,7 @@ class Builder:
 
         tables = list(stmt.from_tables)
         if tables:
-            plan = scan_with_filters(tables[0])
+            plan = scan_with_filters(tables[0], push=lhs_preserved)
         else:
             # FROM-less SELECT (constants only, e.g. SELECT 1 + 1 —
             # reference supports via a one-row relation)
@@ -481,28 +491,36 @@ class Builder:
 
         # comma-joined tables: EliminateCrossJoin — find WHERE equalities
         for tr in tables[1:]:
-            rhs = scan_with_filters(tr)
+            rhs = scan_with_filters(tr, push=lhs_preserved)
             lhs_fields = plan.getRowType().getFieldList()
             combined = RelDataType(lhs_fields + rhs.getRowType().getFieldList())
             tmp = LogicalPlan("__combined__", [], combined, None)
             conds = []
-            for i, cj in enumerate(where_conjuncts):
-                if used[i] or self._has_agg(cj):
-                    continue
-                refs = self._tables_of(cj)
-                if not refs:
-                    continue
-                if self._refs_only(cj, set(), tmp) and not self._refs_only(
-                    cj, set(), plan
-                ):
-                    conds.append(cj)
-                    used[i] = True
+            if lhs_preserved:
+                for i, cj in enumerate(where_conjuncts):
+                    if used[i] or self._has_agg(cj):
+                        continue
+                    refs = self._tables_of(cj)
+                    if not refs:
+                        continue
+                    if self._refs_only(cj, set(), tmp) and not self._refs_only(
+                        cj, set(), plan
+                    ):
+                        conds.append(cj)
+                        used[i] = True
             plan = join_plans(plan, rhs, "INNER" if conds else "CROSS",
                               conds, None)
 
         # explicit JOIN clauses
-        for jc in stmt.joins:
-            rhs = scan_with_filters(jc.table)
+        for k, jc in enumerate(stmt.joins):
+            # rhs is preserved for INNER/RIGHT/CROSS; null-supplying under
+            # LEFT/FULL (and semi/anti rhs columns are not in scope for
+            # WHERE at all) — and a later RIGHT/FULL join null-supplies the
+            # whole accumulated lhs, rhs included.
+            rhs_push = jc.join_type not in (
+                "LEFT", "FULL", "LEFTSEMI", "LEFTANTI"
+            ) and not any(t in ("RIGHT", "FULL") for t in join_types[k + 1:])
+            rhs = scan_with_filters(jc.table, push=rhs_push)
             plan = join_plans(plan, rhs, jc.join_type, [], jc.on)
 
         # EXISTS: equality-correlated → SEMI/ANTI join over the DISTINCT

@@ -71,6 +71,9 @@ enum DsxOp {
   DSX_OP_RINT_F64 = 68,  /* ties-to-even like numpy round */
   DSX_OP_EXP_F64 = 69, DSX_OP_LN_F64 = 70, DSX_OP_POW_F64 = 71,
   DSX_OP_YEAR = 72, DSX_OP_MONTH = 73, DSX_OP_DAY = 74, /* date32 day-int */
+  DSX_OP_FLOORMOD_I64 = 75, /* Python/pandas floor-mod: MOD(-5,3) = 1
+                               (reference evaluates operator.mod on pandas,
+                               rex/core/call.py:1047-1156) */
 };
 
 typedef struct DsxInstr {

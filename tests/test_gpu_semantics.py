@@ -1288,3 +1288,62 @@ def test_full_surface_executes(ctx):
     for q in PLAN_BATTERY:
         pdf = c.sql(q).compute()
         assert pdf is not None, q
+
+
+def test_full_outer_null_keys_match(ctx):
+    """FULL OUTER keeps NULL-key rows on BOTH sides and NULL keys match
+    each other (pandas merge how="outer" NA-match; join.py:202-213 drops
+    NULL keys only for inner/left/right/semi). ADVICE r1 (medium)."""
+    lhs = pd.DataFrame({"k": pd.array([1, None, 2, None], dtype="Int64"),
+                        "a": [10.0, 20.0, 30.0, 40.0]})
+    rhs = pd.DataFrame({"k": pd.array([1, None, 3], dtype="Int64"),
+                        "b": [100.0, 200.0, 300.0]})
+    ctx.create_table("fon_l", lhs)
+    ctx.create_table("fon_r", rhs)
+    out = ctx.sql("SELECT l.k, l.a, r.b FROM fon_l l FULL JOIN fon_r r "
+                  "ON l.k = r.k").compute()
+    from oracle.frame import oracle_join
+    exp = oracle_join(lhs, rhs, [0], [0], "FULL")
+    # exp columns: lhs_0 lhs_1 rhs_0 rhs_1; output k = lhs k (NULL where
+    # rhs-only). Compare as sorted (a, b) multisets + row count.
+    assert len(out) == len(exp)
+    got = out[["a", "b"]].astype("float64").fillna(-1).sort_values(
+        ["a", "b"]).to_numpy()
+    expv = exp[["lhs_1", "rhs_1"]].astype("float64").fillna(-1)
+    expv.columns = ["a", "b"]
+    expv = expv.sort_values(["a", "b"]).to_numpy()
+    assert (got == expv).all()
+    # the two NULL-key lhs rows each matched the one NULL-key rhs row
+    nullk = out[out["a"].notna() & out["b"].notna()
+                & out["k"].isna()]
+    assert sorted(nullk["a"].tolist()) == [20.0, 40.0]
+    assert nullk["b"].tolist() == [200.0, 200.0]
+
+
+def test_where_on_null_supplying_side(ctx):
+    """WHERE on the rhs of a LEFT join must filter POST-join (NULL-extended
+    rows where r.x IS NULL are excluded by r.x = 5). ADVICE r1 (high)."""
+    lhs = pd.DataFrame({"k": np.array([1, 2, 3, 4], dtype=np.int64)})
+    rhs = pd.DataFrame({"k": np.array([1, 2], dtype=np.int64),
+                        "x": np.array([5, 6], dtype=np.int64)})
+    ctx.create_table("wns_l", lhs)
+    ctx.create_table("wns_r", rhs)
+    out = ctx.sql("SELECT l.k FROM wns_l l LEFT JOIN wns_r r "
+                  "ON l.k = r.k WHERE r.x = 5").compute()
+    assert out["k"].astype(np.int64).tolist() == [1]
+    # and IS NULL on the rhs still sees the NULL-extended rows
+    out2 = ctx.sql("SELECT l.k FROM wns_l l LEFT JOIN wns_r r "
+                   "ON l.k = r.k WHERE r.x IS NULL").compute()
+    assert sorted(out2["k"].astype(np.int64).tolist()) == [3, 4]
+
+
+def test_mod_floor_semantics(ctx):
+    """MOD matches the reference's Python/pandas floor-mod for negative
+    operands: MOD(-5,3) = 1, not C's -2 (operator.mod on pandas,
+    rex/core/call.py OPERATION_MAPPING). ADVICE r1 (low)."""
+    a = np.array([-5, 5, -5, 5, -7, 0], dtype=np.int64)
+    b = np.array([3, -3, -3, 3, 2, 5], dtype=np.int64)
+    ctx.create_table("tmod", pd.DataFrame({"a": a, "b": b}))
+    out = ctx.sql("SELECT MOD(a, b) AS m FROM tmod").compute()
+    exp = np.mod(a, b)  # numpy mod IS floor-mod, same as pandas
+    assert out["m"].to_numpy(dtype=np.int64).tolist() == exp.tolist()

@@ -408,3 +408,47 @@ def test_fold_string_literal_unit():
     assert fold_string_literal(
         Call("CONCAT", [lit("a"), lit("b"), lit("c")])) == "abc"
     assert fold_string_literal(Call("+", [lit("a"), lit("b")])) is None
+
+
+def _join_chain(rel):
+    """Walk down from the root to the first Join; return (join, nodes_above)."""
+    node = rel
+    above = []
+    while node.get_current_node_type() != "Join":
+        above.append(node.get_current_node_type())
+        node = node.get_inputs()[0]
+    return node, above
+
+
+def test_no_pushdown_below_null_supplying_side():
+    # ADVICE r1 (high): WHERE on the null-supplying side of an outer join
+    # must NOT be pushed below the join (DataFusion PushDownFilter pushes
+    # only to preserved sides). `SELECT ... LEFT JOIN r ... WHERE r.x = 5`
+    # keeps the predicate as a post-join Filter.
+    c = Context()
+    c.create_table("l", pd.DataFrame({"k": [1, 2, 3]}))
+    c.create_table("r", pd.DataFrame({"k": [1, 2], "x": [5, 6]}))
+
+    rel = c._get_ral(
+        "SELECT l.k FROM l LEFT JOIN r ON l.k = r.k WHERE r.x = 5")
+    join, above = _join_chain(rel)
+    assert "Filter" in above, "WHERE r.x=5 must stay above the LEFT join"
+    rhs = join.get_inputs()[1]
+    assert rhs.get_current_node_type() == "TableScan", \
+        "no Filter below the null-supplying rhs of a LEFT join"
+
+    # RIGHT join: lhs is null-supplying — WHERE l.* stays post-join,
+    # WHERE r.* (preserved side) is pushed below.
+    rel = c._get_ral(
+        "SELECT r.k FROM l RIGHT JOIN r ON l.k = r.k WHERE l.k = 1 AND r.x = 5")
+    join, above = _join_chain(rel)
+    assert "Filter" in above
+    assert join.get_inputs()[0].get_current_node_type() == "TableScan"
+    assert join.get_inputs()[1].get_current_node_type() == "Filter"
+
+    # INNER join still pushes both sides down (no post-join Filter).
+    rel = c._get_ral(
+        "SELECT l.k FROM l JOIN r ON l.k = r.k WHERE r.x = 5")
+    join, above = _join_chain(rel)
+    assert "Filter" not in above
+    assert join.get_inputs()[1].get_current_node_type() == "Filter"
