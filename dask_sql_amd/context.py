@@ -285,6 +285,13 @@ class Context:
     # -- reference context.py:482 sql --------------------------------------
     def sql(self, sql: str, return_futures: bool = True,
             config_options=None) -> ResultFrame:
+        if config_options:
+            # per-query overrides, scoped like the reference's
+            # dask.config.set(config_options) (context.py:519); unknown or
+            # unsupported keys raise (round-1 silently ignored them)
+            from dask_sql_amd import config
+            with config.set(config_options):
+                return self.sql(sql, return_futures=return_futures)
         # SHOW SCHEMAS / TABLES / COLUMNS (reference rel/custom/*.py,
         # expected frames pinned by tests/integration/test_show.py)
         sm = re.match(r'\s*SHOW\s+(SCHEMAS|TABLES|COLUMNS)'
@@ -358,7 +365,8 @@ class Context:
     def _get_ral(self, sql: str):
         """reference context.py:819 _get_ral (planner entry). Plans are
         immutable → cached per (sql, schema version)."""
-        key = (sql, self._schema_version)
+        from dask_sql_amd import config
+        key = (sql, self._schema_version, config.plan_fingerprint())
         plan = self._plan_cache.get(key)
         if plan is None:
             plan = Builder(self.catalog, self.schema_name).build(sql)

@@ -200,6 +200,66 @@ def merge_groupby_partials(runtime, key_col, val_cols, val_ops, group=None):
     return key_out, out_vals
 
 
+_DSX_SIZE = {rt.I64: 8, rt.F64: 8, rt.I32: 4, rt.F32: 4, rt.I8: 1,
+             rt.BOOL8: 1}
+
+
+def allgather_device_columns(runtime, cols, n, group=None):
+    """Broadcast-join build side: every rank receives the concatenation of
+    ALL ranks' rows — the RCCL analog of the reference's broadcast join
+    (sql.join.broadcast, join.py:228-246: ship the small side everywhere,
+    skip the big side's shuffle). Cheap only when the table is small,
+    which is exactly the broadcast-join precondition.
+
+    Implemented over the same exchange as the shuffle (send the full local
+    table to every rank), so the gloo CPU tests cover the identical path."""
+    world = dist.get_world_size(group)
+    dev = torch.device("cuda", runtime.device_id)
+    staged, has_validity = [], []
+    for col in cols:
+        if getattr(col, "dictionary", None) is not None:
+            raise NotImplementedError(
+                "broadcasting dictionary-encoded columns needs a global "
+                "dictionary merge — decode or remap before the exchange")
+        t = torch.empty(n, dtype=_TORCH_DTYPE[col.dtype], device=dev)
+        runtime.copy_raw(t.data_ptr(), col.data, n * _DSX_SIZE[col.dtype])
+        staged.append(t.repeat(world) if world > 1 else t)
+        has_validity.append(bool(col.validity))
+        if col.validity:
+            vt = torch.empty(n, dtype=torch.uint8, device=dev)
+            runtime.copy_raw(vt.data_ptr(), col.validity, n)
+            staged.append(vt.repeat(world) if world > 1 else vt)
+    runtime.synchronize()
+    received, _ = exchange_buckets(staged, [n] * world, group)
+    torch.cuda.synchronize(dev)
+    out = []
+    it = iter(received)
+    for hv in has_validity:
+        t = next(it)
+        vptr, keep = None, t
+        if hv:
+            vt = next(it)
+            vptr, keep = vt.data_ptr(), (t, vt)
+        out.append(rt.DeviceColumn(runtime, t.data_ptr(), vptr, t.numel(),
+                                   _dtype_of(t), owner=False,
+                                   keep_alive=keep))
+    return out
+
+
+def allgather_datacontainer(runtime, dc, group=None):
+    """Replicate a DataContainer's rows on every rank (broadcast-join build
+    side). Returns a DeviceTable with the same frontend column names."""
+    from dask_sql_amd.datacontainer import DeviceTable
+
+    cc = dc.column_container
+    names = list(cc.columns)
+    cols = [dc.table.col(cc.get_backend_by_frontend_name(n)) for n in names]
+    out_cols = allgather_device_columns(runtime, cols, dc.table.num_rows,
+                                        group)
+    return DeviceTable(dict(zip(names, out_cols)),
+                       num_rows=out_cols[0].len if out_cols else 0)
+
+
 def q1_merge_partials(partials):
     """Merge per-rank TPC-H Q1 partial frames into the global Q1 frame.
 
@@ -276,7 +336,15 @@ def q3_distributed(ctx, group=None):
                   "FROM orders WHERE o_orderdate < 9204").dc
     l_f = ctx.sql("SELECT l_orderkey, l_extendedprice, l_discount "
                   "FROM lineitem WHERE l_shipdate > 9204").dc
-    if world > 1:
+    from dask_sql_amd import config
+    broadcast = bool(config.get("sql.join.broadcast", None))
+    if world > 1 and broadcast:
+        # sql.join.broadcast: replicate the (small, filtered) customer
+        # build side on every rank and keep orders local — the reference's
+        # broadcast-join variant (join.py:228-246) over RCCL
+        c_x = allgather_datacontainer(runtime, c_f, group)
+        o_x = o_f.assign()
+    elif world > 1:
         c_x = shuffle_datacontainer(runtime, c_f, "c_custkey", group)
         o_x = shuffle_datacontainer(runtime, o_f, "o_custkey", group)
     else:

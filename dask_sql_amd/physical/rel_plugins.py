@@ -1659,9 +1659,12 @@ class DaskLimitPlugin(BaseRelPlugin):
     def convert(self, rel, context):
         node = rel.limit()
         below = rel.get_inputs()[0]
+        ncols_in = max(1, len(below.getRowType().getFieldList()))
+        from dask_sql_amd import config as _config
+        nelem_limit = _config.get("sql.sort.topk-nelem-limit", 1_000_000)
         if (below.get_current_node_type() == "Sort"
                 and node.fetch is not None
-                and node.fetch + node.offset <= 10_000):
+                and (node.fetch + node.offset) * ncols_in <= nelem_limit):
             (inp,) = self.assert_inputs(below, 1, context)
             keys = below.sort().getCollation()
             k = node.fetch + node.offset

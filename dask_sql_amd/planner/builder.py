@@ -426,6 +426,12 @@ class Builder:
         # NULL-extended rows that SQL excludes would be returned).
         join_types = [jc.join_type for jc in stmt.joins]
         lhs_preserved = not any(t in ("RIGHT", "FULL") for t in join_types)
+        # sql.predicate_pushdown=False disables the PushDownFilter analog
+        # entirely (reference sql.yaml:predicate_pushdown; the WHERE then
+        # applies as one post-join Filter — results identical)
+        from dask_sql_amd import config as _config
+        _pushdown = bool(_config.get("sql.predicate_pushdown", True))
+        lhs_preserved = lhs_preserved and _pushdown
 
         # 1. scans (+ pushed-down single-table filters, à la PushDownFilter)
         def scan_with_filters(tr: TableRef, push: bool = True) -> LogicalPlan:
@@ -517,7 +523,7 @@ class Builder:
             # LEFT/FULL (and semi/anti rhs columns are not in scope for
             # WHERE at all) — and a later RIGHT/FULL join null-supplies the
             # whole accumulated lhs, rhs included.
-            rhs_push = jc.join_type not in (
+            rhs_push = _pushdown and jc.join_type not in (
                 "LEFT", "FULL", "LEFTSEMI", "LEFTANTI"
             ) and not any(t in ("RIGHT", "FULL") for t in join_types[k + 1:])
             rhs = scan_with_filters(jc.table, push=rhs_push)

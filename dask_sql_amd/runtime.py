@@ -189,6 +189,16 @@ class Runtime:
         )
         return p.value
 
+    def copy_raw(self, dst_ptr: int, src_ptr: int, nbytes: int):
+        """Device→device copy (dsx_copy / hipMemcpyAsync on the lib
+        stream); staging into externally-owned buffers (e.g. torch tensors
+        for RCCL collectives)."""
+        if nbytes:
+            _check(self.lib,
+                   self.lib.dsx_copy(self.ctx, ct.c_void_p(dst_ptr),
+                                     ct.c_void_p(src_ptr),
+                                     ct.c_int64(nbytes)), "dsx_copy")
+
     def _download(self, dev_ptr, out: np.ndarray):
         _check(
             self.lib,

@@ -281,3 +281,36 @@ def test_q1_merge_gloo():
 
 def test_q3_pipeline_merge_gloo():
     _spawn(_run_q3_pipeline)
+
+
+def _run_broadcast_exchange(rank, world, port, results):
+    """Broadcast-join build-side replication (sql.join.broadcast →
+    allgather_device_columns' exchange pattern): every rank must end up
+    with the concatenation of all ranks' rows."""
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from dask_sql_amd.distributed import exchange_buckets
+
+        n = 5 + rank
+        t = torch.arange(n, dtype=torch.int64) + rank * 100
+        received, out_splits = exchange_buckets([t.repeat(world)],
+                                                [n] * world)
+        exp = []
+        for s in range(world):
+            exp += [s * 100 + i for i in range(5 + s)]
+        assert received[0].tolist() == exp, (rank, received[0].tolist())
+        assert out_splits == [5 + s for s in range(world)]
+        results[rank] = "ok"
+    finally:
+        dist.destroy_process_group()
+
+
+def test_broadcast_exchange_gloo():
+    _spawn(_run_broadcast_exchange)
+
+
+def test_broadcast_exchange_gloo_world3():
+    _spawn(_run_broadcast_exchange, world=3)
