@@ -331,11 +331,13 @@ def q3_distributed(ctx, group=None):
     runtime = ctx._get_runtime()
 
     c_f = ctx.sql("SELECT c_custkey FROM customer "
-                  "WHERE c_mktsegment = 0").dc
+                  "WHERE c_mktsegment = 'BUILDING'").dc
     o_f = ctx.sql("SELECT o_orderkey, o_custkey, o_orderdate, o_shippriority "
-                  "FROM orders WHERE o_orderdate < 9204").dc
+                  "FROM orders "
+                  "WHERE o_orderdate < DATE '1995-03-15'").dc
     l_f = ctx.sql("SELECT l_orderkey, l_extendedprice, l_discount "
-                  "FROM lineitem WHERE l_shipdate > 9204").dc
+                  "FROM lineitem "
+                  "WHERE l_shipdate > DATE '1995-03-15'").dc
     from dask_sql_amd import config
     broadcast = bool(config.get("sql.join.broadcast", None))
     if world > 1 and broadcast:
@@ -351,7 +353,8 @@ def q3_distributed(ctx, group=None):
         c_x = c_f.assign()
         o_x = o_f.assign()
     ctx.create_table_from_device("customer_x", c_x)
-    ctx.create_table_from_device("orders_x", o_x)
+    ctx.create_table_from_device("orders_x", o_x,
+                                 sql_types={"o_orderdate": "DATE"})
     co = ctx.sql("SELECT o_orderkey, o_orderdate, o_shippriority "
                  "FROM customer_x, orders_x "
                  "WHERE c_custkey = o_custkey").dc
@@ -361,7 +364,8 @@ def q3_distributed(ctx, group=None):
     else:
         co_x = co.assign()
         l_x = l_f.assign()
-    ctx.create_table_from_device("co_x", co_x)
+    ctx.create_table_from_device("co_x", co_x,
+                                 sql_types={"o_orderdate": "DATE"})
     ctx.create_table_from_device("lineitem_x", l_x)
     top = ctx.sql(
         "SELECT l_orderkey, SUM(l_extendedprice*(1-l_discount)) AS revenue, "

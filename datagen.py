@@ -119,3 +119,48 @@ def gen_q3(sf_rows=(1_500_000, 15_000_000, 60_000_000), seed=SEED):
         }
     )
     return cust, orders, li
+
+
+# The REAL TPC-H Q3 text (the timed query must be the
+# reference's own literals — 'BUILDING' and DATE '1995-03-15' — not
+# pre-encoded codes/day-ints; literal encoding is the planner's job).
+Q3_SQL = """SELECT l_orderkey, SUM(l_extendedprice*(1-l_discount)) AS revenue,
+ o_orderdate, o_shippriority
+ FROM customer, orders, lineitem
+ WHERE c_mktsegment = 'BUILDING' AND c_custkey = o_custkey
+ AND l_orderkey = o_orderkey AND o_orderdate < DATE '1995-03-15'
+ AND l_shipdate > DATE '1995-03-15'
+ GROUP BY l_orderkey, o_orderdate, o_shippriority
+ ORDER BY revenue DESC, o_orderdate LIMIT 10"""
+
+# Q1 with the reference literals (DATE '1998-12-01' - INTERVAL '90' DAY
+# folds to 1998-09-02 at plan time, like DataFusion's SimplifyExpressions)
+Q1_SQL = """SELECT l_returnflag, l_linestatus, SUM(l_quantity) AS sum_qty,
+ SUM(l_extendedprice) AS sum_base_price,
+ SUM(l_extendedprice*(1-l_discount)) AS sum_disc_price,
+ SUM(l_extendedprice*(1-l_discount)*(1+l_tax)) AS sum_charge,
+ AVG(l_quantity) AS avg_qty, AVG(l_extendedprice) AS avg_price,
+ AVG(l_discount) AS avg_disc, COUNT(*) AS count_order
+ FROM lineitem WHERE l_shipdate <= DATE '1998-12-01' - INTERVAL '90' DAY
+ GROUP BY l_returnflag, l_linestatus"""
+
+
+def register_q3_tables(ctx, cust, orders, li, persist=False):
+    """Register the Q3 tables with their REAL types: c_mktsegment as a
+    dictionary-encoded string column, o_orderdate/l_shipdate as DATE — so
+    Q3_SQL's 'BUILDING' / DATE literals plan exactly as on the reference."""
+    ctx.create_table("customer", cust, persist=persist,
+                     dictionaries={"c_mktsegment": MKTSEGMENT_DICT})
+    ctx.create_table("orders", orders, persist=persist,
+                     date_columns=("o_orderdate",))
+    ctx.create_table("lineitem", li, persist=persist,
+                     date_columns=("l_shipdate",))
+
+
+def register_q1_table(ctx, li, persist=False):
+    """Register the Q1 lineitem with real types: returnflag/linestatus as
+    dictionary strings, l_shipdate as DATE."""
+    ctx.create_table("lineitem", li, persist=persist,
+                     date_columns=("l_shipdate",),
+                     dictionaries={"l_returnflag": RETURNFLAG_DICT,
+                                   "l_linestatus": LINESTATUS_DICT})

@@ -385,3 +385,54 @@ def test_filter_idle_block_word_alias(ctx):
     exp = np.nonzero(seg == 0)[0]
     assert count == len(exp)
     assert (ids.astype(np.int64) == exp).all()
+
+
+def test_q3_real_text_parity(ctx):
+    """The bench headline path: REAL Q3 text ('BUILDING',
+    DATE '1995-03-15') over dictionary/DATE-typed tables must match the
+    oracle (which computes on codes/day-ints — monotone-equivalent,
+    SURVEY §8 appendix)."""
+    from datagen import Q3_SQL, gen_q3, register_q3_tables
+    from dask_sql_amd.context import Context
+    from oracle.tpch import oracle_q3
+    cust, orders, li = gen_q3(sf_rows=(20_000, 100_000, 400_000))
+    c = Context()
+    register_q3_tables(c, cust, orders, li)
+    out = c.sql(Q3_SQL).compute()
+    exp = oracle_q3(cust, orders, li)
+    assert len(out) == len(exp)
+    assert (out["l_orderkey"].to_numpy().astype(np.int64)
+            == exp["l_orderkey"].to_numpy()).all()
+    assert np.allclose(out["revenue"], exp["revenue"], rtol=REL_TOL)
+    # o_orderdate comes back as datetime64 (DATE column); oracle holds
+    # day-ints
+    got_days = out["o_orderdate"].to_numpy().astype(
+        "datetime64[D]").astype(np.int64)
+    assert (got_days == exp["o_orderdate"].to_numpy()).all()
+
+
+def test_q1_real_text_parity(ctx):
+    """Q1 with reference literals (DATE - INTERVAL folding) over
+    dictionary returnflag/linestatus and DATE shipdate."""
+    from datagen import (Q1_SQL, RETURNFLAG_DICT, LINESTATUS_DICT,
+                         gen_lineitem_q1, register_q1_table)
+    from dask_sql_amd.context import Context
+    from oracle.tpch import oracle_q1
+    li = gen_lineitem_q1(n=500_000, seed=5)
+    c = Context()
+    register_q1_table(c, li)
+    out = c.sql(Q1_SQL).compute()
+    exp = oracle_q1(li)  # keys are codes, sorted
+    out = out.assign(
+        rf=out["l_returnflag"].map(
+            {s: i for i, s in enumerate(RETURNFLAG_DICT)}),
+        ls=out["l_linestatus"].map(
+            {s: i for i, s in enumerate(LINESTATUS_DICT)}),
+    ).sort_values(["rf", "ls"]).reset_index(drop=True)
+    assert out["rf"].tolist() == exp["l_returnflag"].tolist()
+    assert out["ls"].tolist() == exp["l_linestatus"].tolist()
+    assert (out["count_order"].to_numpy().astype(np.int64)
+            == exp["count_order"].to_numpy()).all()
+    for col in ["sum_qty", "sum_base_price", "sum_disc_price", "sum_charge",
+                "avg_qty", "avg_price", "avg_disc"]:
+        assert np.allclose(out[col], exp[col], rtol=REL_TOL), col

@@ -337,19 +337,11 @@ def test_q3_distributed_world1_equals_plain(ctx):
     from datagen import gen_q3
     from dask_sql_amd.distributed import q3_distributed
     from dask_sql_amd.context import Context
+    from datagen import Q3_SQL, register_q3_tables
     cust, orders, li = gen_q3(sf_rows=(20_000, 100_000, 400_000))
     c2 = Context()
-    c2.create_table("customer", cust)
-    c2.create_table("orders", orders)
-    c2.create_table("lineitem", li)
-    plain = c2.sql(
-        "SELECT l_orderkey, SUM(l_extendedprice*(1-l_discount)) AS revenue, "
-        "o_orderdate, o_shippriority FROM customer, orders, lineitem "
-        "WHERE c_mktsegment = 0 AND c_custkey = o_custkey "
-        "AND l_orderkey = o_orderkey AND o_orderdate < 9204 "
-        "AND l_shipdate > 9204 "
-        "GROUP BY l_orderkey, o_orderdate, o_shippriority "
-        "ORDER BY revenue DESC, o_orderdate LIMIT 10").compute()
+    register_q3_tables(c2, cust, orders, li)
+    plain = c2.sql(Q3_SQL).compute()
     distd = q3_distributed(c2)
     assert (distd["l_orderkey"].to_numpy().astype(np.int64)
             == plain["l_orderkey"].to_numpy().astype(np.int64)).all()
