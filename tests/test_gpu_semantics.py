@@ -676,9 +676,10 @@ def test_scalar_math_functions(ctx):
     np.testing.assert_allclose(got["r1"], np.round(df.x, 1), atol=1e-12)
     np.testing.assert_allclose(got["ex"], np.exp(df.x), rtol=1e-12)
     np.testing.assert_allclose(got["p2"], df.x ** 2, rtol=1e-12)
-    # C-style % (truncated): matches our integer division semantics
+    # floor-mod like the reference's operator.mod on pandas:
+    # MOD(-5,3) = 1, numpy mod is the same floor-mod
     assert got["md"].astype(int).tolist() == [
-        int(np.fmod(v, 3)) for v in df.i]
+        int(np.mod(v, 3)) for v in df.i]
     np.testing.assert_allclose(got["sq"], np.sqrt(np.abs(df.i)), rtol=1e-12)
 
 
