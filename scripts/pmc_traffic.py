@@ -26,7 +26,8 @@ TAG = sys.argv[3] if len(sys.argv) > 3 else "c2"
 
 # kernel family whose traffic makes up one step of the hot path
 FAMILY = {
-    "c2": ["j_hist", "j_scatter", "j_aggregate", "k_gbpart_scan",
+    "c2": ["j_hist", "j_scatter", "j_scatter_staged", "j_aggregate",
+           "k_gbpart_scan",
            "k_gbpart_bases", "k_gbpart_finalize", "k_gbpart_hist",
            "k_gbpart_scatter", "k_gbpart_aggregate"],
 }.get(TAG.split("_")[0], [])
