@@ -2459,14 +2459,31 @@ __global__ void k_count_live(const uint64_t* tkeys,
 
 // host driver for the partition path; *fell_back=true → caller uses the CAS
 // path (LDS table overflow from key skew, or shapes it does not cover)
-static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
-                             ProgArg& P, AggArg& A, const DsxAggSpec* aggs_arr,
-                             std::vector<DsxInstr>& progs,
-                             std::vector<int32_t>& lens, int naggs,
-                             int64_t g_est, uint64_t key_space,
-                             uint64_t** out_codes,
-                             void** out_vals, uint64_t** out_counts,
-                             int64_t* out_groups, bool* fell_back) {
+// direct_shift ≥ 0 selects RANGE bucketing + the direct-index aggregate
+// (no hash table in LDS); pick_direct_shift decides, the wrapper retries
+// with hashing when the range histogram turns out skewed (*redo).
+static int pick_direct_shift(uint64_t key_space, int nvals) {
+  static const bool off = getenv("DSX_GB_NO_DIRECT") != nullptr;
+  if (off || !jit_enabled() || key_space == 0) return -1;
+  if (key_space > (4ull << 20)) return -1;  // nb would exceed 2048
+  int64_t slot_b = 8 * (int64_t)(nvals ? nvals : 0) + 4;
+  int ds = 11;  // 2048 slots default
+  while (ds > 6 && ((1ll << ds) * slot_b) > 88 * 1024) ds--;
+  // keep the aggregate grid >=128 blocks when the space allows
+  while (ds > 6 &&
+         (int64_t)((key_space + (1ull << ds) - 1) >> ds) < 128)
+    ds--;
+  int64_t nbd = (int64_t)((key_space + (1ull << ds) - 1) >> ds);
+  if (nbd > 4096) return -1;
+  return ds;
+}
+
+static int groupby_partition_impl(
+    DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K, ProgArg& P, AggArg& A,
+    const DsxAggSpec* aggs_arr, std::vector<DsxInstr>& progs,
+    std::vector<int32_t>& lens, int naggs, int64_t g_est, uint64_t key_space,
+    uint64_t** out_codes, void** out_vals, uint64_t** out_counts,
+    int64_t* out_groups, bool* fell_back, int direct_shift, bool* redo) {
   *fell_back = false;
   // SoA u32-code records (u32 codes + u64 vals) — MEASURED REGRESSION at C2
   // (35.0 vs 45.8 G rows/s): splitting the record doubles the number of
@@ -2488,13 +2505,26 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
     const char* e = getenv("DSX_GB_NB_TARGET");
     return e ? atoi(e) : 1024;  // groups per bucket (tradeoff note above)
   }();
-  int nb = 64;
-  while (nb < 4096 && g_est / nb > nb_target) nb <<= 1;
-  int64_t per_bucket = (g_est + nb - 1) / nb;
-  int lds_slots = 256;
-  while (lds_slots < 2 * per_bucket) lds_slots <<= 1;
-  int64_t slot_bytes = 8 + 8 * (int64_t)nvals + 4;
-  size_t lds_bytes = (size_t)lds_slots * slot_bytes;
+  int nb;
+  int lds_slots;
+  size_t lds_bytes;
+  if (direct_shift >= 0) {
+    // one bucket covers 2^shift consecutive codes; aggregate LDS =
+    // [slots][nvals] vals + [slots] gcnt, direct-indexed by code low bits
+    int64_t nbd =
+        (int64_t)((key_space + (1ull << direct_shift) - 1) >> direct_shift);
+    nb = (int)((nbd + 7) & ~7ll);  // k_gbpart_scan wants a multiple of 8
+    lds_slots = 1 << direct_shift;
+    lds_bytes = (size_t)lds_slots * (8 * (size_t)nvals + 4);
+  } else {
+    nb = 64;
+    while (nb < 4096 && g_est / nb > nb_target) nb <<= 1;
+    int64_t per_bucket = (g_est + nb - 1) / nb;
+    lds_slots = 256;
+    while (lds_slots < 2 * per_bucket) lds_slots <<= 1;
+    int64_t slot_bytes = 8 + 8 * (int64_t)nvals + 4;
+    lds_bytes = (size_t)lds_slots * slot_bytes;
+  }
   if (lds_bytes > 96 * 1024) {
     *fell_back = true;
     return 0;
@@ -2595,13 +2625,18 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   }
   JitEntry* je = jit_source_entry(
       c, jit_gbpart_source(C, K, P, aggs_arr, A, val_of, naggs, nvals,
-                           lds_slots, code32, tile));
+                           lds_slots, code32, tile, direct_shift));
   hipFunction_t f_hist = je ? jit_fn(c, je, "j_hist") : nullptr;
   hipFunction_t f_scat = je ? jit_fn(c, je, "j_scatter") : nullptr;
   hipFunction_t f_scat_staged =
       (je && tile > 0) ? jit_fn(c, je, "j_scatter_staged") : nullptr;
   hipFunction_t f_aggr = je ? jit_fn(c, je, "j_aggregate") : nullptr;
   if (!(f_hist && f_scat && f_aggr)) {
+    if (direct_shift >= 0) {
+      // direct mode exists only as JIT; retry hashed (static fallback ok)
+      *redo = true;
+      return 0;
+    }
     // all-or-nothing: scatter and aggregate must agree on record layout
     f_hist = f_scat = f_scat_staged = f_aggr = nullptr;
     code32 = false;
@@ -2624,6 +2659,23 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
                        d_hist, grid, nb, d_totals);
     hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(1024), 0, c->stream,
                        d_totals, nb, d_bases);
+    if (direct_shift >= 0) {
+      // balance guard: a skewed key distribution overloads range buckets
+      // (one aggregate block per bucket) — fall back to hash bucketing
+      std::vector<int64_t> h_tot((size_t)nb);
+      HIP_TRY(hipMemcpyAsync(h_tot.data(), d_totals, (size_t)nb * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+      HIP_TRY(hipStreamSynchronize(c->stream));
+      int64_t mx = 0, sum = 0;
+      for (int i = 0; i < nb; i++) {
+        if (h_tot[i] > mx) mx = h_tot[i];
+        sum += h_tot[i];
+      }
+      if (mx > 8 * (sum / nb) + 65536) {
+        *redo = true;
+        return 0;
+      }
+    }
     {
       ProfScope ps(c, "k_gbpart_scatter");
       static const int sthreads = [] {
@@ -2748,6 +2800,32 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   HIP_TRY(hipGetLastError());
   *out_groups = (int64_t)G;
   return 0;
+}
+
+static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
+                             ProgArg& P, AggArg& A, const DsxAggSpec* aggs_arr,
+                             std::vector<DsxInstr>& progs,
+                             std::vector<int32_t>& lens, int naggs,
+                             int64_t g_est, uint64_t key_space,
+                             uint64_t** out_codes,
+                             void** out_vals, uint64_t** out_counts,
+                             int64_t* out_groups, bool* fell_back) {
+  int nvals_probe = 0;
+  for (int a = 0; a < naggs; a++)
+    if (A.op[a] != DSX_AGG_COUNT) nvals_probe++;
+  int ds = pick_direct_shift(key_space, nvals_probe);
+  if (ds >= 0) {
+    bool redo = false;
+    int rc = groupby_partition_impl(c, C, n, K, P, A, aggs_arr, progs, lens,
+                                    naggs, g_est, key_space, out_codes,
+                                    out_vals, out_counts, out_groups,
+                                    fell_back, ds, &redo);
+    if (rc != 0 || !redo) return rc;
+  }
+  bool redo = false;
+  return groupby_partition_impl(c, C, n, K, P, A, aggs_arr, progs, lens,
+                                naggs, g_est, key_space, out_codes, out_vals,
+                                out_counts, out_groups, fell_back, -1, &redo);
 }
 
 extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,
@@ -3151,8 +3229,10 @@ extern "C" int dsx_jit_selftest(void) {
   int32_t val_of[2] = {0, -1};
   for (int tile : {0, 4096}) {
     for (int code32 : {0, 1}) {
+     for (int ds : {-1, 11}) {
       std::string src = jit_gbpart_source(C, K, P, aggs, A, val_of, 2, 1,
-                                          2048, code32 != 0, tile);
+                                          ds >= 0 ? (1 << ds) : 2048,
+                                          code32 != 0, tile, ds);
       if (src.empty()) {
         fprintf(stderr, "[selftest] empty source (tile=%d code32=%d)\n",
                 tile, code32);
@@ -3171,12 +3251,13 @@ extern "C" int dsx_jit_selftest(void) {
         std::string log(lsz, '\0');
         if (lsz) hiprtcGetProgramLog(prog, &log[0]);
         fprintf(stderr,
-                "[selftest] JIT compile FAILED (tile=%d code32=%d):\n%s\n",
-                tile, code32, log.c_str());
+                "[selftest] JIT compile FAILED (tile=%d code32=%d ds=%d):"
+                "\n%s\n", tile, code32, ds, log.c_str());
         hiprtcDestroyProgram(&prog);
         return 3;
       }
       hiprtcDestroyProgram(&prog);
+     }
     }
   }
   return 0;
