@@ -2467,7 +2467,12 @@ static int pick_direct_shift(uint64_t key_space, int nvals) {
   if (off || !jit_enabled() || key_space == 0) return -1;
   if (key_space > (4ull << 20)) return -1;  // nb would exceed 2048
   int64_t slot_b = 8 * (int64_t)(nvals ? nvals : 0) + 4;
-  int ds = 11;  // 2048 slots default
+  static const int ds_cfg = [] {
+    const char* e = getenv("DSX_GB_DS");
+    return e ? atoi(e) : 12;  // measured best at C2 (scatter likes fewer,
+                              // longer bucket runs; aggregate flat 10..12)
+  }();
+  int ds = ds_cfg;  // 4096 slots default
   while (ds > 6 && ((1ll << ds) * slot_b) > 88 * 1024) ds--;
   // keep the aggregate grid >=128 blocks when the space allows
   while (ds > 6 &&
@@ -2611,7 +2616,10 @@ static int groupby_partition_impl(
   }();
   static const long tile_cfg = [] {
     const char* e = getenv("DSX_SCATTER_TILE");
-    return e ? atol(e) : -1;  // -1 = auto, 0 = disable staged
+    // default 0 = plain scatter: the staged variant measured SLOWER at
+    // every feasible tile (LDS cost + double-eval beat the write-merge
+    // gain; see DESIGN.md negative results) — kept as an opt-in knob
+    return e ? atol(e) : 0;
   }();
   int tile = 0;
   if (tile_cfg != 0 && nb <= sthreads_cfg) {
