@@ -1712,23 +1712,24 @@ def _device_topk_impl(context, inp, below, keys, k):
     if col0.dtype not in (rt.F64, rt.I64, rt.I32, rt.I8):
         return None  # unsupported key dtype: host fallback
     isf = col0.dtype == rt.F64
-    if isf or col0.validity:
-        # NULL or NaN keys break the threshold compare: one device count,
-        # fall back if any exist (NULLS FIRST/LAST ordering on host)
-        from dask_sql_amd.physical.rex import OP_IS_NULL, OP_OR
-        bad = [(OP_COL, 0, 0), (OP_IS_NULL, 0, 0)]
-        if isf:
-            bad += [(OP_COL, 0, 0), (OP_COL, 0, 0), (OP_NE_F64, 0, 0),
-                    (OP_OR, 0, 0)]
-        p, cnt = runtime.filter(runtime.make_prog(bad), [col0], n)
-        runtime.wrap_sel(p, cnt)
-        if cnt:
-            return None
-    # strided device sample → approximate k-th order statistic
-    S = int(min(65536, n))
-    sel = runtime.upload_column(
-        np.linspace(0, n - 1, S).astype(np.uint32), dtype=rt.I32)
+    # NULL/NaN keys are folded into the candidate filter below (they all
+    # pass it and surface on host); no separate full scan + sync needed
+    # strided device sample → approximate k-th order statistic. The index
+    # vector depends only on (n, S): cached on the runtime across steps.
+    S = int(min(16384, n))
+    cache = getattr(runtime, "_topk_sample_cache", None)
+    if cache is None:
+        cache = runtime._topk_sample_cache = {}
+    sel = cache.get((n, S))
+    if sel is None:
+        if len(cache) > 32:
+            cache.clear()
+        sel = runtime.upload_column(
+            np.linspace(0, n - 1, S).astype(np.uint32), dtype=rt.I32)
+        cache[(n, S)] = sel
     sv, _ = runtime.gather(col0, sel.data, S).to_numpy()
+    if isf and np.isnan(sv).any():
+        return None  # NaN in sample: host fallback (ordering on host)
     key_s = sv if asc0 else -sv.astype(np.float64 if isf else np.int64)
     cand_ptr = None
     cnt = 0
@@ -1742,6 +1743,14 @@ def _device_topk_impl(context, inp, below, keys, k):
         else:
             op = OP_LE_I64 if asc0 else OP_GE_I64
             prog = [(OP_COL, 0, 0), (OP_LIT_I64, 0, int(thr)), (op, 0, 0)]
+        # NULL/NaN keys also become candidates (no separate scan); if any
+        # survive to the host frame we fall back (see below)
+        from dask_sql_amd.physical.rex import OP_IS_NULL, OP_OR
+        if col0.validity:
+            prog += [(OP_COL, 0, 0), (OP_IS_NULL, 0, 0), (OP_OR, 0, 0)]
+        if isf:
+            prog += [(OP_COL, 0, 0), (OP_COL, 0, 0), (OP_NE_F64, 0, 0),
+                     (OP_OR, 0, 0)]
         cand_ptr, cnt = runtime.filter(runtime.make_prog(prog), [col0], n)
         if cnt >= k:
             break
@@ -1756,6 +1765,9 @@ def _device_topk_impl(context, inp, below, keys, k):
     cand_dc = _gather_table(runtime, inp, sel2.data, cnt)
     from dask_sql_amd.materialize import to_pandas
     pdf = to_pandas(cand_dc, context, below.getRowType())
+    kcol = pdf.iloc[:, idx0]
+    if kcol.isna().to_numpy().any():
+        return None  # NULL/NaN keys present: exact NULLS ordering on host
     return _topk(pdf, keys, k)
 
 

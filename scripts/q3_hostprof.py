@@ -26,15 +26,13 @@ for m in ("filter", "gather", "eval", "hash_build", "hash_probe",
     wrap(m)
 
 from dask_sql_amd.context import Context
-from datagen import gen_q3
+from datagen import gen_q3, register_q3_tables
 from bench import WORKLOADS
 
 cust, orders, li = gen_q3()
 c = Context()
-c.create_table("customer", cust, persist=True)
-c.create_table("orders", orders, persist=True)
-c.create_table("lineitem", li, persist=True)
-Q = WORKLOADS["q3_sf10"]["sql"].replace("lineitem3", "lineitem")
+register_q3_tables(c, cust, orders, li, persist=True)
+Q = WORKLOADS["q3_sf10"]["sql"]
 for _ in range(2):
     c.sql(Q).compute()
 acc.clear(); cnt.clear()
@@ -49,3 +47,16 @@ for name, t in acc.most_common():
     print(f"  {name:14s} {t/N*1000:8.3f} ms/step  ({cnt[name]//N} calls)")
     tot += t / N
 print(f"  accounted: {tot*1000:.2f} ms")
+
+
+# python-level profile of one step (where the non-runtime host ms goes)
+import cProfile, pstats, io
+pr = cProfile.Profile()
+pr.enable()
+for _ in range(3):
+    c.sql(Q).compute()
+pr.disable()
+sio = io.StringIO()
+ps = pstats.Stats(pr, stream=sio).sort_stats("cumulative")
+ps.print_stats(30)
+print(sio.getvalue())
