@@ -3216,6 +3216,8 @@ void jit_cache_destroy(DsxCtx* c) {
   c->jit_cache = nullptr;
 }
 
+#include "radix_join.inc"
+
 // ---------------------------------------------------------------------------
 // dsx_jit_selftest — hiprtc-compile a representative C2-shaped partition
 // groupby source (incl. j_scatter_staged) WITHOUT a GPU. Test harness only:

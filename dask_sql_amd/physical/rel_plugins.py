@@ -347,7 +347,17 @@ class DaskJoinPlugin(BaseRelPlugin):
         import os as _os
         gathered = None
         mat_idx = sorted(set(out_idx))
+        # RADIX PATH first: once the flat probe table
+        # outgrows the XCD L2, partition both sides and probe LDS-resident
+        # bucket tables instead (dsx_radix_join)
         if (lhs_on and not residual and len(mat_idx) <= 16
+                and not _os.environ.get("DSX_DISABLE_RADIX")
+                and join_type in ("inner", "left", "right", "leftanti")):
+            gathered, n_out = self._radix_join(
+                runtime, dc_lhs, dc_rhs, lhs_on, rhs_on, join_type,
+                combined, mat_idx, cc_lhs, cc_rhs)
+        if (gathered is None
+                and lhs_on and not residual and len(mat_idx) <= 16
                 and not _os.environ.get("DSX_DISABLE_JOINFUSE")
                 and join_type in ("inner", "left", "right", "leftanti")):
             gathered, n_out = self._equi_join_fused(
@@ -659,6 +669,119 @@ class DaskJoinPlugin(BaseRelPlugin):
         if swap:
             probe_sel, build_sel = build_sel, probe_sel
         return (probe_sel, build_sel), count
+
+    def _radix_join(self, runtime, dc_lhs, dc_rhs, lhs_on, rhs_on,
+                    join_type, combined, mat_idx, cc_lhs, cc_rhs):
+        """Radix-partitioned equijoin (dsx_radix_join — VERDICT r1 #3, the
+        north star's LDS-staged build/probe). Worth the partition passes
+        only when the flat probe table would spill the XCD L2: gated on
+        build/probe sizes (DSX_RADIX_MIN_BUILD/_MIN_PROBE). Returns
+        ({combined_idx: DeviceColumn}, n_out) or (None, 0) → caller tries
+        the flat fused path."""
+        import os as _os
+        n_l, n_r = dc_lhs.table.num_rows, dc_rhs.table.num_rows
+        min_build = int(_os.environ.get("DSX_RADIX_MIN_BUILD", 1_000_000))
+        min_probe = int(_os.environ.get("DSX_RADIX_MIN_PROBE", 4_000_000))
+        # swap exactly like the fused path: build on rhs, except RIGHT
+        # (probe = rhs) and INNER on the smaller side
+        swap = join_type == "right" or (
+            join_type == "inner" and n_l < n_r)
+        nb_, np_ = (n_l, n_r) if swap else (n_r, n_l)
+        if nb_ < min_build or np_ < min_probe:
+            return None, 0
+        lcols = dc_lhs.backend_cols()
+        rcols = dc_rhs.backend_cols()
+        for i in lhs_on:
+            if lcols[i].dtype not in _INT_KINDS:
+                raise RexCompileError("non-integer join keys (round-2)")
+        for i in rhs_on:
+            if rcols[i].dtype not in _INT_KINDS:
+                raise RexCompileError("non-integer join keys (round-2)")
+        kdc_rhs, krcols = self._reconcile_dict_keys(runtime, dc_lhs, dc_rhs,
+                                                    lhs_on, rhs_on)
+        ranges = []
+        for li, ri in zip(lhs_on, rhs_on):
+            lmn, lmx, lnn = _minmax_cached(runtime, lcols[li])
+            rmn, rmx, rnn = _minmax_cached(runtime, krcols[ri])
+            if lnn == 0 and rnn == 0:
+                mn, mx = 0, 0
+            elif lnn == 0:
+                mn, mx = rmn, rmx
+            elif rnn == 0:
+                mn, mx = lmn, lmx
+            else:
+                mn, mx = min(lmn, rmn), max(lmx, rmx)
+            ranges.append((mn, mx - mn + 1))
+        if swap:
+            probe_dc, build_dc = kdc_rhs, dc_lhs
+            probe_on, build_on = rhs_on, lhs_on
+            ktype = rt.JOIN_LEFT if join_type == "right" else rt.JOIN_INNER
+        else:
+            probe_dc, build_dc = dc_lhs, kdc_rhs
+            probe_on, build_on = lhs_on, rhs_on
+            ktype = {"inner": rt.JOIN_INNER, "left": rt.JOIN_LEFT,
+                     "leftanti": rt.JOIN_LEFTANTI}[join_type]
+        bcols = build_dc.backend_cols()
+        pcols = probe_dc.backend_cols()
+        # NULL keys: pack them into their own code slot (nullable flag,
+        # both sides identically) and DROP build-side NULL keys with a
+        # predicate — code 0 then never matches, so INNER skips NULL-key
+        # probe rows and LEFT null-extends them, exactly join.py:202-213
+        null_flags = [bool(pcols[pi].validity) or bool(bcols[bi].validity)
+                      for pi, bi in zip(probe_on, build_on)]
+        keyspecs_b = [(bi, mn, rng, nf) for bi, (mn, rng), nf
+                      in zip(build_on, ranges, null_flags)]
+        keyspecs_p = [(pi, mn, rng, nf) for pi, (mn, rng), nf
+                      in zip(probe_on, ranges, null_flags)]
+        bpred = []
+        for bi in build_on:
+            if bcols[bi].validity:
+                bpred += [(OP_COL, bi, 0), (OP_IS_NOT_NULL, 0, 0)]
+                if len(bpred) > 2:
+                    bpred.append((OP_AND, 0, 0))
+        # output spec: mat_idx over `combined` → (side, col-in-side-array).
+        # The arrays passed to C start as the (possibly dict-remapped) key
+        # views; payloads whose original column was substituted by the
+        # remap are APPENDED so outputs keep the original codes.
+        probe_is_l = not swap
+        b_arr = list(bcols)
+        p_arr = list(pcols)
+        out_specs = []
+        srcs = []
+        force_bv = ktype == rt.JOIN_LEFT
+        for i in mat_idx:
+            side, frontend = combined[i]
+            if side == "l":
+                col = dc_lhs.table.col(
+                    cc_lhs.get_backend_by_frontend_name(frontend))
+                on_probe = probe_is_l
+            else:
+                col = dc_rhs.table.col(
+                    cc_rhs.get_backend_by_frontend_name(frontend))
+                on_probe = not probe_is_l
+            arr = p_arr if on_probe else b_arr
+            ci = next((j for j, c in enumerate(arr) if c is col), None)
+            if ci is None:
+                arr.append(col)
+                ci = len(arr) - 1
+            need_valid = bool(col.validity) or (not on_probe and force_bv)
+            out_specs.append((0 if on_probe else 1, ci, need_valid))
+            srcs.append(col)
+        if len(b_arr) > 16 or len(p_arr) > 16:
+            return None, 0  # DSX_MAX_COLS — flat path handles wide tables
+        res = runtime.radix_join(
+            b_arr, build_dc.table.num_rows, keyspecs_b, keyspecs_p, bpred,
+            p_arr, probe_dc.table.num_rows, ktype, out_specs)
+        if res is None:
+            return None, 0
+        cols_out, n_out = res
+        gathered = {}
+        for i, src, col in zip(mat_idx, srcs, cols_out):
+            if getattr(src, "dictionary", None) is not None:
+                col.dictionary = src.dictionary
+            col._stats_src = src
+            gathered[i] = col
+        return gathered, n_out
 
     def _equi_join_fused(self, runtime, dc_lhs, dc_rhs, lhs_on, rhs_on,
                          join_type, combined, mat_idx, cc_lhs, cc_rhs):

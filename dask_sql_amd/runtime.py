@@ -544,6 +544,53 @@ class Runtime:
                                     src.dtype, owner=True))
         return out, n
 
+    def radix_join(self, bcols, n_build, keyspecs_b, keyspecs_p, bpred_prog,
+                   pcols, n_probe, join_type, out_specs):
+        """Radix-partitioned equijoin (dsx_radix_join): out_specs =
+        [(side, col_idx, need_valid)] with side 0=probe/1=build and col_idx
+        into that side's cols array. Returns ([DeviceColumn], n) or None on
+        bucket overflow / unsupported shape (caller falls back to the
+        flat-table join)."""
+        nk = len(keyspecs_b)
+        kb = (_KeySpec * nk)()
+        kp = (_KeySpec * nk)()
+        for i, (sb, sp) in enumerate(zip(keyspecs_b, keyspecs_p)):
+            for arr, spec in ((kb, sb), (kp, sp)):
+                arr[i].col = spec[0]
+                arr[i].min = spec[1]
+                arr[i].range = spec[2]
+                arr[i].nullable = 1 if spec[3] else 0
+                arr[i].mode = spec[4] if len(spec) > 4 else 0
+        n_out = len(out_specs)
+        side_a = (ct.c_int32 * n_out)(*[s[0] for s in out_specs])
+        col_a = (ct.c_int32 * n_out)(*[s[1] for s in out_specs])
+        nv_a = (ct.c_int32 * n_out)(*[1 if s[2] else 0 for s in out_specs])
+        datas = (ct.c_void_p * n_out)()
+        valids = (ct.c_void_p * n_out)()
+        count = ct.c_int64()
+        bp = None
+        bl = 0
+        if bpred_prog:
+            bp, bl = self.make_prog(bpred_prog)
+        rc = self.lib.dsx_radix_join(
+            self.ctx, self._cols_array(bcols), ct.c_int(len(bcols)),
+            ct.c_int64(n_build), kb, kp, ct.c_int(nk),
+            bp, ct.c_int(bl),
+            self._cols_array(pcols), ct.c_int(len(pcols)),
+            ct.c_int64(n_probe), ct.c_int(join_type),
+            side_a, col_a, nv_a, ct.c_int(n_out),
+            datas, valids, ct.byref(count))
+        if rc in (-5, -6):
+            return None  # skew overflow / degenerate: flat-table fallback
+        _check(self.lib, rc, "dsx_radix_join")
+        n = count.value
+        out = []
+        for i, (side, ci, _) in enumerate(out_specs):
+            src = bcols[ci] if side else pcols[ci]
+            out.append(DeviceColumn(self, datas[i], valids[i] or None, n,
+                                    src.dtype, owner=True))
+        return out, n
+
     def hash_table_free(self, table):
         self.lib.dsx_hash_table_free(table)
 

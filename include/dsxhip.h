@@ -240,6 +240,27 @@ int dsx_hash_probe_cols(DsxCtx* ctx, DsxHashTable* t, const uint64_t* codes,
 int dsx_hash_unmatched(DsxCtx* ctx, DsxHashTable* t, uint32_t** out_build_idx,
                        int64_t* out_count);
 
+/* Radix-partitioned equijoin (INNER/LEFT/LEFTANTI): both sides hash-
+ * partitioned into bucket-major records, per-bucket LDS build+probe, fused
+ * column emit. Replaces dd.merge's per-partition hash join
+ * (dask_sql/physical/rel/logical/join.py:241-246) at sizes where the flat
+ * probe table spills the XCD L2. keys_b/keys_p: per-key specs (col index
+ * into the side's column array; min/range/nullable MUST match pairwise).
+ * bpred: build-side row predicate (e.g. key IS NOT NULL). out_side[i]:
+ * 0 = probe column, 1 = build column; out_col[i]: column index in that
+ * side's array; out_need_valid[i]: allocate a validity mask for output i.
+ * Returns -6 on bucket overflow (key skew) — caller falls back to the
+ * flat-table join. */
+int dsx_radix_join(DsxCtx* ctx, const DsxColumn* build_cols, int n_build_cols,
+                   int64_t n_build, const DsxKeySpec* keys_b,
+                   const DsxKeySpec* keys_p, int nkeys,
+                   const DsxInstr* bpred, int bpred_len,
+                   const DsxColumn* probe_cols, int n_probe_cols,
+                   int64_t n_probe, int join_type, const int32_t* out_side,
+                   const int32_t* out_col, const int32_t* out_need_valid,
+                   int n_out, void** out_datas, uint8_t** out_valids,
+                   int64_t* out_count);
+
 /* ---- hash groupby-aggregate --------------------------------------------- */
 
 enum DsxAggOp {  /* reference AGGREGATION_MAPPING aggregate.py:117-231 subset:

@@ -1340,3 +1340,103 @@ def test_mod_floor_semantics(ctx):
     out = ctx.sql("SELECT MOD(a, b) AS m FROM tmod").compute()
     exp = np.mod(a, b)  # numpy mod IS floor-mod, same as pandas
     assert out["m"].to_numpy(dtype=np.int64).tolist() == exp.tolist()
+
+
+def _radix_env(monkeypatch, on=True):
+    import os
+    if on:
+        monkeypatch.setenv("DSX_RADIX_MIN_BUILD", "1000")
+        monkeypatch.setenv("DSX_RADIX_MIN_PROBE", "1000")
+    else:
+        monkeypatch.setenv("DSX_DISABLE_RADIX", "1")
+
+
+def _join_both_paths(ctx, monkeypatch, sql, sort_cols):
+    """Run `sql` with the radix join forced on and with it disabled; both
+    frames must be identical up to row order."""
+    _radix_env(monkeypatch, on=True)
+    a = ctx.sql(sql).compute()
+    monkeypatch.delenv("DSX_RADIX_MIN_BUILD")
+    monkeypatch.delenv("DSX_RADIX_MIN_PROBE")
+    _radix_env(monkeypatch, on=False)
+    b = ctx.sql(sql).compute()
+    monkeypatch.delenv("DSX_DISABLE_RADIX")
+    a = a.sort_values(sort_cols).reset_index(drop=True)
+    b = b.sort_values(sort_cols).reset_index(drop=True)
+    assert len(a) == len(b)
+    for c in a.columns:
+        av, bv = a[c], b[c]
+        if av.dtype.kind == "f" or bv.dtype.kind == "f":
+            av = av.astype("float64")
+            bv = bv.astype("float64")
+            assert ((av.isna() == bv.isna()).all()
+                    and np.allclose(av.fillna(0), bv.fillna(0))), c
+        else:
+            assert av.tolist() == bv.tolist(), c
+
+
+def test_radix_join_inner_multimatch(ctx, monkeypatch):
+    """Radix inner join vs the flat-table join on the same data: duplicate
+    build keys (every pair emitted), misses, 2M⋈200k (dsx_radix_join)."""
+    rng = np.random.default_rng(31)
+    bk = np.concatenate([np.arange(100_000, dtype=np.int64),
+                         np.arange(50_000, dtype=np.int64)])  # dups
+    rng.shuffle(bk)
+    pk = rng.integers(0, 200_000, 2_000_000).astype(np.int64)  # ~25% miss
+    ctx.create_table("rj_b", pd.DataFrame({"k": bk,
+                                           "bv": np.arange(len(bk),
+                                                           dtype=np.int64)}))
+    ctx.create_table("rj_p", pd.DataFrame({"k": pk,
+                                           "pv": rng.random(len(pk))}))
+    _join_both_paths(
+        ctx, monkeypatch,
+        "SELECT p.k, p.pv, b.bv FROM rj_p p JOIN rj_b b ON p.k = b.k",
+        ["k", "pv", "bv"])
+
+
+def test_radix_join_left_and_anti(ctx, monkeypatch):
+    rng = np.random.default_rng(32)
+    bk = rng.choice(300_000, 150_000, replace=False).astype(np.int64)
+    pk = rng.integers(0, 300_000, 1_500_000).astype(np.int64)
+    ctx.create_table("rjl_b", pd.DataFrame({"k": bk, "bv": bk * 3}))
+    ctx.create_table("rjl_p", pd.DataFrame({"k": pk}))
+    _join_both_paths(
+        ctx, monkeypatch,
+        "SELECT p.k, b.bv FROM rjl_p p LEFT JOIN rjl_b b ON p.k = b.k",
+        ["k", "bv"])
+    _join_both_paths(
+        ctx, monkeypatch,
+        "SELECT p.k FROM rjl_p p LEFT ANTI JOIN rjl_b b ON p.k = b.k",
+        ["k"])
+
+
+def test_radix_join_null_keys(ctx, monkeypatch):
+    """Nullable keys through the radix path: INNER drops them, LEFT
+    null-extends them (join.py:202-213)."""
+    rng = np.random.default_rng(33)
+    n = 400_000
+    pk = pd.array(rng.integers(0, 50_000, n), dtype="Int64")
+    pk[rng.choice(n, 1000, replace=False)] = None
+    bk = pd.array(np.arange(50_000), dtype="Int64")
+    ctx.create_table("rjn_b", pd.DataFrame({"k": bk, "bv": np.arange(50_000)}))
+    ctx.create_table("rjn_p", pd.DataFrame({"k": pk}))
+    _join_both_paths(
+        ctx, monkeypatch,
+        "SELECT p.k, b.bv FROM rjn_p p JOIN rjn_b b ON p.k = b.k",
+        ["k", "bv"])
+    _join_both_paths(
+        ctx, monkeypatch,
+        "SELECT p.k, b.bv FROM rjn_p p LEFT JOIN rjn_b b ON p.k = b.k",
+        ["k", "bv"])
+
+
+def test_radix_join_right(ctx, monkeypatch):
+    rng = np.random.default_rng(34)
+    lk = rng.choice(200_000, 120_000, replace=False).astype(np.int64)
+    rk = rng.integers(0, 200_000, 1_200_000).astype(np.int64)
+    ctx.create_table("rjr_l", pd.DataFrame({"k": lk, "lv": lk + 7}))
+    ctx.create_table("rjr_r", pd.DataFrame({"k": rk}))
+    _join_both_paths(
+        ctx, monkeypatch,
+        "SELECT l.k, l.lv FROM rjr_l l RIGHT JOIN rjr_r r ON l.k = r.k",
+        ["k", "lv"])
