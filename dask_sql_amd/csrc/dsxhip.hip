@@ -3290,5 +3290,50 @@ extern "C" int dsx_jit_selftest(void) {
      }
     }
   }
+  // radix-join JIT shapes (C3 inner; LEFT with nullable build payload)
+  for (int jt : {DSX_JOIN_INNER, DSX_JOIN_LEFT, DSX_JOIN_LEFTANTI}) {
+    ColsArg CB2{}, CP2{};
+    CB2.ncols = 2;
+    CB2.dtype[0] = DSX_I64;
+    CB2.dtype[1] = DSX_F64;
+    CB2.validity[1] = (const uint8_t*)1;  // metadata-only probe
+    CP2.ncols = 2;
+    CP2.dtype[0] = DSX_I64;
+    CP2.dtype[1] = DSX_F64;
+    KeyArg KB2{}, KP2{};
+    KB2.nkeys = KP2.nkeys = 1;
+    KB2.k[0].col = KP2.k[0].col = 0;
+    KB2.k[0].range = KP2.k[0].range = 10'000'000;
+    KB2.stride[0] = KP2.stride[0] = 1;
+    DsxInstr bp2[2] = {{DSX_OP_COL, 0, 0}, {DSX_OP_IS_NOT_NULL, 0, 0}};
+    int32_t bpay[1] = {1}, ppay[2] = {0, 1};
+    int32_t oside[3] = {0, 0, 1}, oslot2[3] = {0, 1, 0};
+    int32_t odt[3] = {DSX_I64, DSX_F64, DSX_F64};
+    int32_t onv[3] = {0, 0, jt == DSX_JOIN_LEFT ? 1 : 0};
+    std::string src = jit_radix_source(CB2, KB2, bp2, 2, bpay, 1, 1, CP2,
+                                       KP2, ppay, 2, 0, 8192, jt, oside,
+                                       oslot2, odt, onv, 3);
+    if (src.empty()) {
+      fprintf(stderr, "[selftest] empty radix source (jt=%d)\n", jt);
+      return 4;
+    }
+    hiprtcProgram prog;
+    if (hiprtcCreateProgram(&prog, src.c_str(), "dsx_rj.cu", 0, nullptr,
+                            nullptr) != HIPRTC_SUCCESS)
+      return 4;
+    const char* opts[] = {"-O3", "--offload-arch=gfx950", "-std=c++17",
+                          "-munsafe-fp-atomics"};
+    if (hiprtcCompileProgram(prog, 4, opts) != HIPRTC_SUCCESS) {
+      size_t lsz = 0;
+      hiprtcGetProgramLogSize(prog, &lsz);
+      std::string log(lsz, '\0');
+      if (lsz) hiprtcGetProgramLog(prog, &log[0]);
+      fprintf(stderr, "[selftest] radix JIT FAILED (jt=%d):\n%s\n", jt,
+              log.c_str());
+      hiprtcDestroyProgram(&prog);
+      return 5;
+    }
+    hiprtcDestroyProgram(&prog);
+  }
   return 0;
 }
