@@ -680,7 +680,7 @@ class DaskJoinPlugin(BaseRelPlugin):
         the flat fused path."""
         import os as _os
         n_l, n_r = dc_lhs.table.num_rows, dc_rhs.table.num_rows
-        min_build = int(_os.environ.get("DSX_RADIX_MIN_BUILD", 1_000_000))
+        min_build = int(_os.environ.get("DSX_RADIX_MIN_BUILD", 8_000_000))
         min_probe = int(_os.environ.get("DSX_RADIX_MIN_PROBE", 4_000_000))
         # swap exactly like the fused path: build on rhs, except RIGHT
         # (probe = rhs) and INNER on the smaller side
