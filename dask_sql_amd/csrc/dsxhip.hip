@@ -3313,6 +3313,12 @@ extern "C" int dsx_jit_selftest(void) {
     std::string src = jit_radix_source(CB2, KB2, bp2, 2, bpay, 1, 1, CP2,
                                        KP2, ppay, 2, 0, 8192, jt, oside,
                                        oslot2, odt, onv, 3);
+    if (const char* dump = getenv("DSX_JIT_DUMP")) {
+      char fn[512];
+      snprintf(fn, sizeof fn, "%s/selftest_rj_%d.cu", dump, jt);
+      FILE* f = fopen(fn, "w");
+      if (f) { fwrite(src.data(), 1, src.size(), f); fclose(f); }
+    }
     if (src.empty()) {
       fprintf(stderr, "[selftest] empty radix source (jt=%d)\n", jt);
       return 4;
