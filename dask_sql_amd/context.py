@@ -277,6 +277,60 @@ class Context:
             self.catalog.add(table_name, t.fields())
             self._schema_version += 1
 
+    # -- reference context.py:324/:415 UDF registration --------------------
+    @staticmethod
+    def _np_to_sql(t) -> str:
+        m = {"int64": "BIGINT", "int32": "INTEGER", "int16": "INTEGER",
+             "int8": "TINYINT", "uint64": "BIGINT", "uint32": "BIGINT",
+             "uint16": "INTEGER", "uint8": "INTEGER",
+             "float64": "DOUBLE", "float32": "FLOAT", "bool": "BOOLEAN",
+             "object": "VARCHAR", "str": "VARCHAR"}
+        name = np.dtype(t).name if t is not str else "str"
+        if name not in m:
+            raise NotImplementedError(f"UDF return type {t!r} unsupported")
+        return m[name]
+
+    def register_function(self, f, name: str, parameters, return_type,
+                          replace: bool = False, schema_name: str = None,
+                          row_udf: bool = False):
+        """reference context.py:324 — register a scalar function usable in
+        SQL. Executed exactly as on the reference: the Python callable runs
+        on host column data (a UDF IS Python — the operand columns round-
+        trip device→host→device on the explicit slow path)."""
+        self._register_callable(f, name, False, parameters, return_type,
+                                replace, row_udf)
+
+    def register_aggregation(self, f, name: str, parameters, return_type,
+                             replace: bool = False, schema_name: str = None):
+        """reference context.py:415 — register a custom aggregation (a
+        dask.dataframe.Aggregation-like object with .chunk/.agg[/.finalize],
+        or a plain callable Series→scalar)."""
+        self._register_callable(f, name, True, parameters, return_type,
+                                replace, False)
+
+    def _register_callable(self, f, name, aggregation, parameters,
+                           return_type, replace, row_udf):
+        key = name.lower()
+        cat = self.catalog
+        existing = cat.functions.get(key) or cat.aggregations.get(key)
+        if existing is not None and not replace and existing[0] is not f:
+            # reference _register_callable: one namespace for both kinds;
+            # re-registering the SAME callable (type overloads,
+            # test_function.py:180-188) is fine, a different one needs
+            # replace=True
+            raise ValueError(
+                f"A function with the name {name} is already present; "
+                "use replace=True to overwrite it")
+        cat.functions.pop(key, None)
+        cat.aggregations.pop(key, None)
+        ret = self._np_to_sql(return_type)
+        if aggregation:
+            cat.aggregations[key] = (f, ret)
+        else:
+            cat.functions[key] = (f, ret, bool(row_udf),
+                                  list(parameters or []))
+        self._schema_version += 1
+
     def drop_table(self, table_name: str):
         self.tables.pop(table_name.lower(), None)
         self.catalog.drop(table_name)

@@ -137,7 +137,68 @@ def _resolve_scalar_subs(expr, context):
     return expr
 
 
-def _apply_filter(runtime, dc: DataContainer, condition) -> DataContainer:
+def _eval_udf_nodes(expr, cols, runtime, context):
+    """Replace UDF:<name> Calls with InputRefs to freshly computed columns.
+
+    The reference executes registered Python functions on pandas partition
+    data (context.py:324 register_function → rex call) — we do EXACTLY the
+    same: operand columns round-trip device→host, the Python callable runs
+    vectorized, the result uploads back. This is the explicit UDF slow
+    path, not a silent fallback. `cols` is extended in
+    place; returns the rewritten expression."""
+    if isinstance(expr, InputRef) or not hasattr(expr, "getOperands"):
+        return expr
+    ops = [_eval_udf_nodes(o, cols, runtime, context) for o in
+           expr.getOperands()]
+    name = expr.getOperatorName()
+    if not (isinstance(name, str) and name.startswith("UDF:")):
+        if ops != list(expr.getOperands()):
+            return Call(name, ops, expr.getType())
+        return expr
+    f, ret_sql, row_udf, params = context.catalog.functions[name[4:]]
+    import pandas as pd
+    args = []
+    n = cols[0].len if cols else 0
+    for o in ops:
+        prog, kind = compile_expr(o, cols, _dicts_of(cols))
+        col = runtime.eval(runtime.make_prog(prog), cols, n,
+                           rt.F64 if kind == KF else rt.I64,
+                           with_validity=True)
+        arr, valid = col.to_numpy()
+        ser = pd.Series(arr)
+        if valid is not None:
+            ser = ser.where(valid.astype(bool))
+        args.append(ser)
+    if row_udf:
+        # reference row_udf: f(row) with row[param_name]
+        # (test_function.py:24-33)
+        frame = pd.DataFrame({pn: a for (pn, _), a in zip(params, args)})
+        out = frame.apply(f, axis=1)
+    else:
+        out = f(*args)
+    out = pd.Series(out)
+    res = out.to_numpy()
+    validity = None
+    if res.dtype == object:
+        res = res.astype(np.float64)
+    if res.dtype.kind == "f":
+        nan = np.isnan(res)
+        if nan.any():
+            validity = (~nan).astype(np.uint8)
+    if res.dtype.kind == "b":
+        res = res.astype(np.uint8)
+        new_col = runtime.upload_column(res, validity, rt.BOOL8)
+    elif res.dtype.kind in "iu":
+        new_col = runtime.upload_column(res.astype(np.int64), validity)
+    else:
+        new_col = runtime.upload_column(res.astype(np.float64), validity)
+    cols.append(new_col)
+    from dask_sql_amd.planner.plan import SqlType
+    return InputRef(len(cols) - 1, SqlType(ret_sql))
+
+
+def _apply_filter(runtime, dc: DataContainer, condition,
+                  context=None) -> DataContainer:
     """filter_or_scalar semantics (reference filter.py:20-45). Fused path:
     the filter's emit pass writes the surviving rows of every column
     directly (dsx_filter_cols) — no selection vector, no per-column
@@ -146,6 +207,8 @@ def _apply_filter(runtime, dc: DataContainer, condition) -> DataContainer:
     if s is not None:
         return dc if s else _empty_like(runtime, dc)
     cols = dc.backend_cols()
+    if context is not None and context.catalog.functions:
+        condition = _eval_udf_nodes(condition, cols, runtime, context)
     prog, kind = compile_expr(condition, cols, _dicts_of(cols))
     cc = dc.column_container
     backends = []
@@ -199,7 +262,7 @@ class DaskFilterPlugin(BaseRelPlugin):
     def convert(self, rel, context):
         (dc,) = self.assert_inputs(rel, 1, context)
         condition = _resolve_scalar_subs(rel.filter().getCondition(), context)
-        dc = _apply_filter(context._get_runtime(), dc, condition)
+        dc = _apply_filter(context._get_runtime(), dc, condition, context)
         cc = self.fix_column_to_row_type(dc.column_container, rel.getRowType())
         return DataContainer(dc.table, cc)
 
@@ -224,6 +287,14 @@ class DaskProjectPlugin(BaseRelPlugin):
             backend_name = f"p{i}__{name}"
             if not isinstance(expr, InputRef):
                 expr = _resolve_scalar_subs(expr, context)
+            if context.catalog.functions and not isinstance(expr, InputRef):
+                expr = _eval_udf_nodes(expr, cols, runtime, context)
+            if isinstance(expr, InputRef) and expr.getIndex() >= len(
+                    dc.column_container.columns):
+                # UDF result column: materialize it directly
+                out_cols[backend_name] = cols[expr.getIndex()]
+                new_names.append((name, backend_name))
+                continue
             sfn = None if isinstance(expr, InputRef) \
                 else dict_string_fn(expr, dicts)
             ifn = None if isinstance(expr, InputRef) or sfn is not None \
@@ -944,6 +1015,81 @@ class DaskAggregatePlugin(BaseRelPlugin):
     # sample forms (pandas default), ddof=0 for *_POP.
     STD_FINS = {"std_samp", "std_pop", "var_samp", "var_pop"}
 
+    def _host_udf_aggregate(self, rel, agg, context):
+        """Registered aggregate UDFs run as Python on host frames — the
+        reference's own execution model for register_aggregation
+        (dd.Aggregation chunk/agg[/finalize], context.py:415; executed by
+        pandas either way). Single partition: chunk over the groupby, agg
+        over the chunk results, optional finalize. Built-in aggs mixed
+        into the same node run with their pandas equivalents
+        (aggregate.py:117-231 semantics)."""
+        import pandas as pd
+
+        from dask_sql_amd.context import _from_pandas
+        from dask_sql_amd.materialize import to_pandas
+
+        (dc,) = self.assert_inputs(rel, 1, context)
+        pdf = to_pandas(dc, context)
+        group_idx = [e.getIndex() for e in agg.getGroupSets()]
+        keys = [pdf.columns[i] for i in group_idx]
+        work = pdf.assign(__const_1__=1)
+        gb_keys = keys if keys else ["__const_1__"]
+        field_names = [str(f) for f in rel.getRowType().getFieldNames()]
+        results = {}
+        builtin = {"sum": lambda sg: sg.sum(min_count=1),
+                   "count": lambda sg: sg.count(),
+                   "avg": lambda sg: sg.mean(),
+                   "min": lambda sg: sg.min(), "max": lambda sg: sg.max()}
+        for pos, call in enumerate(agg.getNamedAggCalls()):
+            func = agg.getAggregationFuncName(call).lower()
+            args = agg.getArgs(call)
+            if args:
+                if not isinstance(args[0], InputRef):
+                    raise RexCompileError(
+                        "UDF aggregate over an expression: project it to a "
+                        "column first")
+                in_col = pdf.columns[args[0].getIndex()]
+            else:
+                in_col = "__const_1__"
+            sg = work.groupby(gb_keys, dropna=False)[in_col]
+            if func.startswith("udf:"):
+                obj = context.catalog.aggregations[func[4:]][0]
+                if hasattr(obj, "chunk") and hasattr(obj, "agg"):
+                    chunked = obj.chunk(sg)
+                    # single partition: the tree-reduce step runs once
+                    r = obj.agg(chunked.groupby(
+                        level=list(range(len(gb_keys)))))
+                    fin = getattr(obj, "finalize", None)
+                    if fin is not None:
+                        r = fin(r.to_frame().groupby(
+                            level=list(range(len(gb_keys)))))
+                        r = pd.Series(r)
+                else:
+                    r = sg.agg(obj)
+            elif func in builtin:
+                r = builtin[func](sg)
+            else:
+                raise RexCompileError(
+                    f"aggregate {func} beside a UDF aggregate (round-3)")
+            results[field_names[len(keys) + pos]] = r
+        out = pd.DataFrame(results).reset_index()
+        if not keys:
+            out = out.drop(columns=["__const_1__"], errors="ignore")
+        out.columns = field_names[len(keys):] if not keys else (
+            field_names[:len(keys)] + list(out.columns[len(keys):]))
+        out = out[[c for c in field_names if c in out.columns]]
+        runtime = context._get_runtime()
+        host_cols = _from_pandas(out)
+        dev = {}
+        for n_, h in host_cols.items():
+            col = runtime.upload_column(h.arr, h.validity, h.dtype)
+            if h.dictionary is not None:
+                col.dictionary = h.dictionary
+            dev[n_] = col
+        cc = ColumnContainer(list(out.columns))
+        cc = self.fix_column_to_row_type(cc, rel.getRowType())
+        return DataContainer(DeviceTable(dev, num_rows=len(out)), cc)
+
     def convert(self, rel, context):
         runtime = context._get_runtime()
         agg = rel.aggregate()
@@ -954,6 +1100,12 @@ class DaskAggregatePlugin(BaseRelPlugin):
         # and the projection arithmetic folded into each agg program. This is
         # what replaces the reference's filter→assign→groupby pass chain with
         # a single HBM scan.
+        if context.catalog.aggregations and not agg.isDistinctNode():
+            calls = agg.getNamedAggCalls()
+            if any(agg.getAggregationFuncName(c).lower().startswith("udf:")
+                   for c in calls):
+                return self._host_udf_aggregate(rel, agg, context)
+
         import os
         fused = None
         if not os.environ.get("DSX_DISABLE_FUSED"):

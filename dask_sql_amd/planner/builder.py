@@ -43,10 +43,15 @@ def _common_type(a: str, b: str) -> str:
 
 
 class Catalog:
-    """What the builder needs from the Context's schema: table → fields."""
+    """What the builder needs from the Context's schema: table → fields,
+    plus the registered scalar/aggregate UDFs (reference
+    context.py:324/:415 register_function/register_aggregation — one
+    shared namespace for both kinds)."""
 
     def __init__(self):
         self.tables: dict[str, list[tuple[str, str]]] = {}
+        self.functions: dict[str, tuple] = {}     # name → (f, ret_sql, row_udf)
+        self.aggregations: dict[str, tuple] = {}  # name → (obj, ret_sql)
 
     def add(self, name, fields):
         self.tables[name.lower()] = fields
@@ -148,6 +153,9 @@ class Builder:
         if kind == "call":
             _, op, args = ast
             ops = [self._resolve(a, plan) for a in args]
+            fn = self.catalog.functions.get(op.lower())
+            if fn is not None:
+                return Call(f"UDF:{op.lower()}", ops, SqlType(fn[1]))
             if op in ("+", "-") and any(
                     isinstance(o, Literal)
                     and o.getType().getSqlType() == "INTERVAL" for o in ops):
@@ -396,10 +404,28 @@ class Builder:
         return False
 
     # ------------------------------------------------------------- pipeline
+    def _rewrite_udf_aggs(self, ast):
+        """Registered aggregate UDFs parse as plain calls (the parser's
+        AGG_FUNCS set is static) — rewrite them into agg nodes so they go
+        through the Aggregate plan node like the reference's
+        register_aggregation does."""
+        if not isinstance(ast, tuple):
+            return ast
+        if ast[0] == "call" and isinstance(ast[1], str)                 and ast[1].lower() in self.catalog.aggregations:
+            return ("agg", f"udf:{ast[1].lower()}",
+                    [self._rewrite_udf_aggs(a) for a in ast[2]], False, None)
+        if ast[0] == "call":
+            return (ast[0], ast[1],
+                    [self._rewrite_udf_aggs(a) for a in ast[2]])
+        return ast
+
     def build_stmt(self, stmt) -> LogicalPlan:
         from dask_sql_amd.planner.parser import UnionStmt
         if isinstance(stmt, UnionStmt):
             return self._build_union(stmt)
+        if self.catalog.aggregations:
+            stmt.items = [(self._rewrite_udf_aggs(e), a)
+                          for e, a in stmt.items]
         where_conjuncts = self._conjuncts(stmt.where)
         # IN (SELECT ...) conjuncts → SEMI/ANTI joins (DataFusion's subquery
         # decorrelation on the reference side). Pulled out before pushdown.

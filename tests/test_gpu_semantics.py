@@ -1440,3 +1440,90 @@ def test_radix_join_right(ctx, monkeypatch):
         ctx, monkeypatch,
         "SELECT l.k, l.lv FROM rjr_l l RIGHT JOIN rjr_r r ON l.k = r.k",
         ["k", "lv"])
+
+
+class _FakeAggregation:
+    """dd.Aggregation-shaped object (name, chunk, agg[, finalize]) — dask
+    is not installed here; the contract is the attribute triple."""
+
+    def __init__(self, name, chunk, agg, finalize=None):
+        self.name = name
+        self.chunk = chunk
+        self.agg = agg
+        if finalize is not None:
+            self.finalize = finalize
+
+
+def test_register_function_scalar(ctx):
+    """reference test_function.py:13-21: SELECT F(a) runs the registered
+    Python callable (UDF = Python on the reference too)."""
+    from dask_sql_amd.context import Context
+    c = Context()
+    df = pd.DataFrame({"a": np.random.default_rng(3).random(1000)})
+    c.create_table("df", df)
+
+    def f(x):
+        return x ** 2
+
+    c.register_function(f, "f", [("x", np.float64)], np.float64)
+    out = c.sql("SELECT F(a) AS a FROM df").compute()
+    np.testing.assert_allclose(out["a"], df["a"] ** 2, rtol=1e-12)
+    # and inside WHERE (rex path)
+    out2 = c.sql("SELECT a FROM df WHERE f(a) > 0.25").compute()
+    assert len(out2) == int((df["a"] ** 2 > 0.25).sum())
+
+
+def test_register_function_row_udf(ctx):
+    """reference test_function.py:24-33: row_udf f(row) with row[name]."""
+    from dask_sql_amd.context import Context
+    c = Context()
+    df = pd.DataFrame({"a": np.arange(100, dtype=np.int64),
+                       "b": np.arange(100, dtype=np.int64) * 3})
+    c.create_table("dfw", df)
+
+    def f(row):
+        return row["x"] + row["y"]
+
+    c.register_function(f, "f", [("x", np.int64), ("y", np.int64)],
+                        np.int64, row_udf=True)
+    out = c.sql("SELECT F(a, b) AS s FROM dfw").compute()
+    assert out["s"].astype(np.int64).tolist() == (df["a"] + df["b"]).tolist()
+
+
+def test_register_function_reregistration(ctx):
+    """reference test_function.py:180-207: same callable ok, different one
+    raises unless replace=True; one namespace with aggregations."""
+    from dask_sql_amd.context import Context
+    c = Context()
+
+    def f(x):
+        return x ** 2
+
+    c.register_function(f, "f", [("x", np.float64)], np.float64)
+    c.register_function(f, "f", [("x", np.int64)], np.int64)
+
+    def g(x):
+        return x ** 3
+
+    with pytest.raises(ValueError):
+        c.register_function(g, "f", [("x", np.float64)], np.float64)
+    c.register_function(g, "f", [("x", np.float64)], np.float64,
+                        replace=True)
+
+
+def test_register_aggregation(ctx):
+    """reference test_function.py:166-177: FAGG(b) == SUM(b) for a
+    sum/sum Aggregation."""
+    from dask_sql_amd.context import Context
+    c = Context()
+    np.random.seed(42)
+    df = pd.DataFrame({"k": np.random.randint(0, 5, 700),
+                       "b": 10 * np.random.rand(700)})
+    c.create_table("df", df)
+    fagg = _FakeAggregation("f", lambda x: x.sum(), lambda x: x.sum())
+    c.register_aggregation(fagg, "fagg", [("x", np.float64)], np.float64)
+    out = c.sql('SELECT FAGG(b) AS test, SUM(b) AS "S" FROM df').compute()
+    np.testing.assert_allclose(out["test"], out["S"], rtol=1e-12)
+    out2 = c.sql('SELECT k, FAGG(b) AS test, SUM(b) AS "S" FROM df '
+                 "GROUP BY k").compute()
+    np.testing.assert_allclose(out2["test"], out2["S"], rtol=1e-12)
