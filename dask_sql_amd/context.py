@@ -98,6 +98,62 @@ def _from_pandas(df: pd.DataFrame, date_columns=(), dictionaries=None):
     return cols
 
 
+def _from_arrow(table, date_columns=(), dictionaries=None):
+    """pyarrow.Table → host column set with NO pandas round-trip (SURVEY
+    §8f3 direct parquet ingest): primitive buffers are zero-copy numpy
+    views, validity bitmaps unpack to byte masks, arrow dictionary columns
+    map 1:1 onto our code+dictionary form, strings dictionary-encode in
+    arrow. Upload then streams pinned→HBM (dsx_upload_pinned)."""
+    import pyarrow as pa
+    import pyarrow.compute as pc
+
+    dictionaries = dictionaries or {}
+    cols = {}
+    for name, chunked in zip(table.column_names, table.columns):
+        arr = chunked.combine_chunks() if chunked.num_chunks != 1             else chunked.chunk(0)
+        if isinstance(arr, pa.ChunkedArray):
+            arr = arr.combine_chunks()
+        t = arr.type
+        validity = None
+        dictionary = None
+        if arr.null_count:
+            validity = pc.is_valid(arr).to_numpy(
+                zero_copy_only=False).astype(np.uint8)
+        if pa.types.is_string(t) or pa.types.is_large_string(t):
+            arr = arr.dictionary_encode()
+            t = arr.type
+        if pa.types.is_dictionary(t):
+            dictionary = arr.dictionary.to_pylist()
+            codes = arr.indices
+            vals = codes.fill_null(0).to_numpy(
+                zero_copy_only=False).astype(np.int32)
+            data, dtype, sqlt = vals, rt.I32, "VARCHAR"
+        elif pa.types.is_date32(t):
+            data = arr.view(pa.int32()).to_numpy(zero_copy_only=False)
+            dtype, sqlt = rt.I32, "DATE"
+        elif pa.types.is_timestamp(t):
+            ns = arr.cast(pa.timestamp("ns")).view(pa.int64()).to_numpy(
+                zero_copy_only=False)
+            data, dtype, sqlt = ns, rt.I64, "TIMESTAMP"
+        elif pa.types.is_boolean(t):
+            data = arr.fill_null(False).to_numpy(
+                zero_copy_only=False).astype(np.uint8)
+            dtype, sqlt = rt.BOOL8, "BOOLEAN"
+        else:
+            if arr.null_count:
+                arr = arr.fill_null(0)
+            npv = arr.to_numpy(zero_copy_only=arr.null_count == 0)
+            data, dtype, sqlt = _np_map(np.ascontiguousarray(npv))
+        if name in date_columns:
+            sqlt = "DATE"
+        if name in dictionaries:
+            dictionary = dictionaries[name]
+            sqlt = "VARCHAR"
+        cols[str(name)] = _HostColumn(np.ascontiguousarray(data), validity,
+                                      sqlt, dtype, dictionary)
+    return cols
+
+
 def _np_map(arr):
     m = {
         np.dtype("int64"): (rt.I64, "BIGINT"),
@@ -242,7 +298,17 @@ class Context:
             low = input_table.lower()
             if fmt == "parquet" or low.endswith(".parquet") \
                     or low.endswith(".parq"):
-                input_table = pd.read_parquet(input_table)
+                # direct arrow → HBM, no pandas round-trip
+                import pyarrow.parquet as pq
+                host_cols = _from_arrow(pq.read_table(input_table),
+                                        date_columns, dictionaries)
+                t = RegisteredTable(host_cols)
+                self.tables[table_name.lower()] = t
+                self.catalog.add(table_name, t.fields())
+                self._schema_version += 1
+                if persist:
+                    t.upload(self._get_runtime())
+                return
             elif fmt == "csv" or low.endswith(".csv"):
                 input_table = pd.read_csv(input_table)
             elif fmt == "json" or low.endswith(".json"):

@@ -208,15 +208,34 @@ class Runtime:
             "dsx_download",
         )
 
+    _PINNED_MIN = 8 << 20  # below this the pageable path is fine
+
     def upload_column(self, arr: np.ndarray, validity: np.ndarray | None = None,
                       dtype: int | None = None) -> DeviceColumn:
         if dtype is None:
             dtype = _NP_TO_DSX[arr.dtype]
-        data = self._upload_raw(arr)
+        data = self._upload_raw_auto(arr)
         vptr = None
         if validity is not None:
-            vptr = self._upload_raw(validity.astype(np.uint8))
+            vptr = self._upload_raw_auto(validity.astype(np.uint8))
         return DeviceColumn(self, data, vptr, len(arr), dtype)
+
+    def _upload_raw_auto(self, arr: np.ndarray) -> int:
+        """Large buffers go through the pinned staging arena
+        (dsx_upload_pinned — chunked memcpy + hipMemcpyAsync overlap);
+        small ones take the plain path."""
+        arr = np.ascontiguousarray(arr)
+        if arr.nbytes < self._PINNED_MIN:
+            return self._upload_raw(arr)
+        p = ct.c_void_p()
+        _check(
+            self.lib,
+            self.lib.dsx_upload_pinned(self.ctx,
+                                       arr.ctypes.data_as(ct.c_void_p),
+                                       ct.c_int64(arr.nbytes), ct.byref(p)),
+            "dsx_upload_pinned",
+        )
+        return p.value
 
     def empty_column(self, n: int, dtype: int, with_validity=False) -> DeviceColumn:
         data = self._malloc(n * _DSX_SIZE[dtype])

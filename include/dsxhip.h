@@ -109,6 +109,13 @@ int dsx_free(DsxCtx* ctx, void* ptr);
  * registration (dask_sql/context.py:168 create_table / persist):
  * pinned-host→HBM hipMemcpyAsync. */
 int dsx_upload(DsxCtx* ctx, const void* host, int64_t bytes, void** out_dev);
+/* host→HBM through a persistent pinned staging arena (chunked, overlapped
+ * memcpy + hipMemcpyAsync) — the parquet-ingest fast path (reference reads
+ * via dask IO, physical/utils/filter.py:17; here: pyarrow column buffers →
+ * pinned → HBM with no pandas round-trip). */
+int dsx_upload_pinned(DsxCtx* ctx, const void* host_data, int64_t nbytes,
+                      void** out_device_ptr);
+
 int dsx_download(DsxCtx* ctx, const void* dev, void* host, int64_t bytes);
 int dsx_memset(DsxCtx* ctx, void* dev, int value, int64_t bytes);
 /* device-to-device copy (UNION ALL concatenation — the reference's

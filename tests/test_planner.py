@@ -485,3 +485,39 @@ def test_q1_real_text_plans():
     assert rel is not None
     txt = rel.explain()
     assert "Aggregate" in txt
+
+
+def test_from_arrow_ingest(tmp_path):
+    """Direct parquet→columnar ingest: _from_arrow bypasses
+    pandas; dtypes/validity/dictionary/date mapping checked host-side."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from dask_sql_amd.context import _from_arrow
+    import dask_sql_amd.runtime as rt
+
+    n = 1000
+    t = pa.table({
+        "i": pa.array(list(range(n)), type=pa.int64()),
+        "f": pa.array([float(x) / 7 for x in range(n)], type=pa.float64()),
+        "ni": pa.array([None if x % 10 == 0 else x for x in range(n)],
+                       type=pa.int64()),
+        "s": pa.array(["BUILDING", "AUTO", "FURN"][x % 3] for x in range(n)),
+        "d": pa.array([x % 3000 for x in range(n)], type=pa.date32()),
+        "b": pa.array([x % 2 == 0 for x in range(n)]),
+    })
+    f = tmp_path / "t.parquet"
+    pq.write_table(t, f)
+    cols = _from_arrow(pq.read_table(f))
+    assert cols["i"].dtype == rt.I64 and cols["i"].validity is None
+    assert cols["f"].dtype == rt.F64
+    assert cols["ni"].validity is not None
+    assert cols["ni"].validity.sum() == n - 100
+    assert cols["s"].dictionary is not None and cols["s"].sql_type == "VARCHAR"
+    import numpy as np
+    dec = [cols["s"].dictionary[c] for c in cols["s"].arr[:6]]
+    assert dec == ["BUILDING", "AUTO", "FURN", "BUILDING", "AUTO", "FURN"]
+    assert cols["d"].sql_type == "DATE" and cols["d"].arr.dtype == np.int32
+    assert cols["b"].dtype == rt.BOOL8
+    assert cols["i"].arr[-1] == n - 1
+    assert abs(cols["f"].arr[7] - 1.0) < 1e-12
