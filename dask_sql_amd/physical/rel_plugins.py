@@ -146,7 +146,8 @@ def _eval_udf_nodes(expr, cols, runtime, context):
     vectorized, the result uploads back. This is the explicit UDF slow
     path, not a silent fallback. `cols` is extended in
     place; returns the rewritten expression."""
-    if isinstance(expr, InputRef) or not hasattr(expr, "getOperands"):
+    if isinstance(expr, InputRef) or not hasattr(expr, "getOperands") \
+            or not hasattr(expr, "getOperatorName"):
         return expr
     ops = [_eval_udf_nodes(o, cols, runtime, context) for o in
            expr.getOperands()]
