@@ -1240,6 +1240,16 @@ __device__ __forceinline__ uint64_t pack_key(const KeyArg& K, const ColsArg& C,
       part = (!valid || x != x)
                  ? 0ull
                  : (uint64_t)__double_as_longlong(x == 0.0 ? 0.0 : x) + 1;
+    } else if (K.k[j].mode & 2 || K.k[j].mode & 4) {
+      // ORDER BY packing (dsx_sort_perm): mode bit1 = DESC (flip value
+      // part), bit2 = NULLS LAST (NULL takes the TOP slot, else slot 0)
+      uint64_t vp = (uint64_t)(v.i - K.k[j].min);
+      if (K.k[j].mode & 2) vp = (uint64_t)(K.k[j].range - 1) - vp;
+      if (K.k[j].nullable)
+        part = (K.k[j].mode & 4) ? (valid ? vp : (uint64_t)K.k[j].range)
+                                 : (valid ? vp + 1 : 0);
+      else
+        part = vp;
     } else if (K.k[j].nullable) {
       part = valid ? (uint64_t)(v.i - K.k[j].min) + 1 : 0;
     } else {
@@ -3258,6 +3268,7 @@ void jit_cache_destroy(DsxCtx* c) {
 }
 
 #include "radix_join.inc"
+#include "sort.inc"
 
 // ---------------------------------------------------------------------------
 // dsx_jit_selftest — hiprtc-compile a representative C2-shaped partition

@@ -1906,13 +1906,20 @@ class DaskAggregatePlugin(BaseRelPlugin):
 
 
 class DaskSortPlugin(BaseRelPlugin):
-    """ORDER BY on the (≤G-row) result — host-side top-k per SURVEY §8f1
-    (reference physical/utils/sort.py:9-60)."""
+    """ORDER BY (reference physical/utils/sort.py:9-60, pandas mergesort
+    per partition). Device path: order-preserving packed
+    codes → range partition → per-bucket LDS bitonic (dsx_sort_perm,
+    stable via rowid tiebreak), then one gather per column. Host fallback
+    for float keys / key spaces > 2^62 / skew-overloaded buckets."""
 
     class_name = "Sort"
 
     def convert(self, rel, context):
         (dc,) = self.assert_inputs(rel, 1, context)
+        if isinstance(dc, DataContainer) and dc.table.num_rows >= 65536:
+            out = self._device_sort(context, dc, rel.sort().getCollation())
+            if out is not None:
+                return out
         from dask_sql_amd.materialize import to_pandas
         pdf = to_pandas(dc, context)
         for idx, asc, nulls_first in reversed(rel.sort().getCollation()):
@@ -1922,6 +1929,59 @@ class DaskSortPlugin(BaseRelPlugin):
                 na_position="first" if nulls_first else "last",
                 kind="mergesort")
         return HostDataContainer(pdf.reset_index(drop=True))
+
+    def _device_sort(self, context, dc, collation):
+        runtime = context._get_runtime()
+        cols = dc.backend_cols()
+        work = list(cols)
+        specs = []
+        n = dc.table.num_rows
+        for idx, asc, nulls_first in collation:
+            col = cols[idx]
+            use_idx = idx
+            if getattr(col, "dictionary", None) is not None:
+                # dictionary codes are unordered — remap through the
+                # alphabetic rank LUT (None ranks with NULL handling)
+                order = sorted(
+                    (i for i, s_ in enumerate(col.dictionary)
+                     if s_ is not None),
+                    key=lambda i: col.dictionary[i])
+                rank = np.zeros(len(col.dictionary), dtype=np.int64)
+                for r_, i in enumerate(order):
+                    rank[i] = r_
+                lut = runtime.upload_column(rank)
+                g = runtime.gather(lut, col.data, n)
+                ranked = rt.DeviceColumn(runtime, g.data, col.validity,
+                                         n, rt.I64, owner=False,
+                                         keep_alive=(g, col, lut))
+                work.append(ranked)
+                use_idx = len(work) - 1
+                col = ranked
+            elif col.dtype not in _INT_KINDS:
+                return None  # float keys: exact NULLS/NaN order on host
+            mn, mx, nn = _minmax_cached(runtime, col)
+            if nn == 0:
+                mn, mx = 0, 0
+            mode = (2 if not asc else 0) | (0 if nulls_first else 4)
+            specs.append((use_idx, mn, mx - mn + 1,
+                          bool(col.validity), mode))
+        # first ORDER BY key → highest stride: pass reversed
+        perm = runtime.sort_perm(work, list(reversed(specs)), n)
+        if perm is None:
+            return None
+        cc = dc.column_container
+        out_cols = {}
+        for f in cc.columns:
+            b = cc.get_backend_by_frontend_name(f)
+            if b in out_cols:
+                continue
+            src = dc.table.col(b)
+            g = runtime.gather(src, perm.data, n, bool(src.validity))
+            if getattr(src, "dictionary", None) is not None:
+                g.dictionary = src.dictionary
+            g._stats_src = src
+            out_cols[b] = g
+        return DataContainer(DeviceTable(out_cols, num_rows=n), cc)
 
 
 class DaskLimitPlugin(BaseRelPlugin):

@@ -610,6 +610,28 @@ class Runtime:
                                     src.dtype, owner=True))
         return out, n
 
+    def sort_perm(self, cols, keyspecs, n):
+        """Device ORDER BY permutation (dsx_sort_perm): keyspecs =
+        (col, min, range, nullable, mode) with mode bit1=DESC,
+        bit2=NULLS LAST, passed in REVERSED significance order. Returns a
+        DeviceColumn of u32 row ids or None (unsupported/skew → host)."""
+        nk = len(keyspecs)
+        ks = (_KeySpec * nk)()
+        for i, spec in enumerate(keyspecs):
+            ks[i].col = spec[0]
+            ks[i].min = spec[1]
+            ks[i].range = spec[2]
+            ks[i].nullable = 1 if spec[3] else 0
+            ks[i].mode = spec[4]
+        perm = ct.c_void_p()
+        rc = self.lib.dsx_sort_perm(self.ctx, self._cols_array(cols),
+                                    ct.c_int(len(cols)), ks, ct.c_int(nk),
+                                    ct.c_int64(n), ct.byref(perm))
+        if rc in (-3, -4, -6):
+            return None
+        _check(self.lib, rc, "dsx_sort_perm")
+        return DeviceColumn(self, perm.value, None, n, I32, owner=True)
+
     def hash_table_free(self, table):
         self.lib.dsx_hash_table_free(table)
 

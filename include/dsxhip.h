@@ -268,6 +268,16 @@ int dsx_radix_join(DsxCtx* ctx, const DsxColumn* build_cols, int n_build_cols,
                    int n_out, void** out_datas, uint8_t** out_valids,
                    int64_t* out_count);
 
+/* Device general sort for ORDER BY without LIMIT (reference
+ * physical/utils/sort.py:9-60): order-preserving packed codes (DsxKeySpec
+ * mode bit1 = DESC, bit2 = NULLS LAST; caller passes keys REVERSED so the
+ * first ORDER BY key takes the highest stride) → range partition →
+ * per-bucket LDS bitonic (stable via rowid tiebreak). Returns the sorted
+ * row permutation; -6 on skew (caller sorts on host). */
+int dsx_sort_perm(DsxCtx* ctx, const DsxColumn* cols, int ncols,
+                  const DsxKeySpec* keys, int nkeys, int64_t n,
+                  uint32_t** out_perm);
+
 /* ---- hash groupby-aggregate --------------------------------------------- */
 
 enum DsxAggOp {  /* reference AGGREGATION_MAPPING aggregate.py:117-231 subset:
