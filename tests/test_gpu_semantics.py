@@ -1601,7 +1601,9 @@ def test_device_general_sort_at_scale(ctx):
     runtime.prof_enable(False)
     assert "k_sort_bucket" in prof and prof["k_sort_bucket"]["launches"] > 0, \
         "device sort did not run (host fallback?)"
-    exp = df.sort_values("k2", ascending=False, na_position="last",
+    # planner default: NULLS FIRST for DESC (DataFusion convention,
+    # parser.py:221)
+    exp = df.sort_values("k2", ascending=False, na_position="first",
                          kind="mergesort")
     exp = exp.sort_values("k1", kind="mergesort").reset_index(drop=True)
     assert out["p"].astype(np.int64).tolist() == exp["p"].tolist()
