@@ -3269,6 +3269,7 @@ void jit_cache_destroy(DsxCtx* c) {
 
 #include "radix_join.inc"
 #include "sort.inc"
+#include "window.inc"
 
 // ---------------------------------------------------------------------------
 // dsx_jit_selftest — hiprtc-compile a representative C2-shaped partition

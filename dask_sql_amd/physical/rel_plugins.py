@@ -2182,7 +2182,138 @@ class DaskWindowPlugin(BaseRelPlugin):
                              for i in spec.part_idx))
         if device_ok:
             return self._device_agg(runtime, cols, n, spec)
+        col = self._device_ordered(runtime, cols, n, spec)
+        if col is not None:
+            return col
         return self._host_ordered(runtime, cols, n, spec)
+
+    def _device_ordered(self, runtime, cols, n, spec):
+        """Device ordered window frames: sort permutation
+        over (partition, order) keys → boundary starts → dsx_window_ordered
+        (binary-search ranks / shifts; per-partition running scans with
+        peer broadcast). Mirrors _host_ordered's pandas semantics exactly
+        (na_position="last" on order keys, partitions keep NULL groups).
+        Returns None on unsupported shapes → host fallback."""
+        import os as _os
+        if _os.environ.get("DSX_DISABLE_DEVWINDOW") or n == 0:
+            return None
+        if spec.func not in runtime.WIN_FUNCS:
+            return None
+        work = list(cols)
+
+        def key_of(i, want_rank):
+            col = cols[i]
+            if getattr(col, "dictionary", None) is not None and want_rank:
+                order = sorted(
+                    (j for j, s_ in enumerate(col.dictionary)
+                     if s_ is not None),
+                    key=lambda j: col.dictionary[j])
+                rank = np.zeros(len(col.dictionary), dtype=np.int64)
+                for r_, j in enumerate(order):
+                    rank[j] = r_
+                lut = runtime.upload_column(rank)
+                g = runtime.gather(lut, col.data, n)
+                ranked = rt.DeviceColumn(runtime, g.data, col.validity, n,
+                                         rt.I64, owner=False,
+                                         keep_alive=(g, col, lut))
+                work.append(ranked)
+                return len(work) - 1, ranked
+            return i, col
+
+        part, order = [], []
+        for i in spec.part_idx:
+            ki, col = key_of(i, False)
+            if col.dtype not in _INT_KINDS:
+                return None
+            part.append((ki, col, False))
+        for i, desc in spec.order_idx:
+            ki, col = key_of(i, True)
+            if col.dtype not in _INT_KINDS:
+                return None
+            order.append((ki, col, desc))
+
+        def spec_of(ki, col, desc, order_key):
+            mn, mx, nn = _minmax_cached(runtime, col)
+            if nn == 0:
+                mn, mx = 0, 0
+            mode = 0
+            if order_key:
+                # host sorts order keys na_position="last" for BOTH
+                # directions (_host_ordered)
+                mode = (2 if desc else 0) | 4
+            return (ki, mn, mx - mn + 1, bool(col.validity), mode)
+
+        sort_specs = [spec_of(*p, False) for p in part] +                      [spec_of(*o, True) for o in order]
+        perm = runtime.sort_perm(work, list(reversed(sort_specs)), n)
+        if perm is None:
+            return None
+
+        def packed(specs):
+            if not specs:
+                z = runtime.upload_column(np.zeros(1, dtype=np.int64))
+                # constant zero code for "no partition": broadcastless —
+                # use an eval of literal 0 over n rows instead
+                from dask_sql_amd.physical.rex import OP_LIT_I64
+                return runtime.eval(
+                    runtime.make_prog([(OP_LIT_I64, 0, 0)]), [z], n,
+                    rt.I64, with_validity=False)
+            codes, _sp = runtime.keypack(
+                work, [(ki, mn, rng, nf) for (ki, mn, rng, nf, _m)
+                       in specs], n)
+            return codes
+
+        pspecs = [spec_of(*p, False) for p in part]
+        fspecs = pspecs + [spec_of(*o, True) for o in order]
+        try:
+            pcodes = packed(pspecs)
+            fcodes = packed(fspecs) if order else pcodes
+        except Exception:
+            return None  # key space overflow etc → host
+        ps = runtime.gather(pcodes, perm.data, n)
+        fs = runtime.gather(fcodes, perm.data, n) if order else ps
+
+        v_col = cols[spec.arg_idx] if spec.arg_idx is not None else None
+        func = spec.func
+        ranking = func in ("row_number", "rank", "dense_rank")
+        if ranking:
+            out_dtype, want_valid = rt.I64, False
+            default_bits, has_def = 0, False
+        elif func in ("lag", "lead", "first_value"):
+            if v_col is None or v_col.dictionary is not None:
+                return None  # dict value shift: host (string payload)
+            out_dtype = v_col.dtype
+            want_valid = True
+            has_def = spec.default is not None and func != "first_value"
+            default_bits = 0
+            if has_def:
+                d = spec.default
+                if out_dtype in (rt.F64,):
+                    default_bits = int(np.float64(d).view(np.int64))
+                elif out_dtype == rt.F32:
+                    default_bits = int(np.float32(d).view(np.int32))
+                else:
+                    default_bits = int(d)
+        else:
+            if func != "count" and v_col is None:
+                return None
+            if v_col is not None and v_col.dictionary is not None:
+                return None
+            # host emits int64 when the plan type is BIGINT and nothing is
+            # NULL; float64 (+validity) otherwise (_host_ordered tail)
+            is_big = spec.out_type is not None and                 spec.out_type.getSqlType() == "BIGINT"
+            nonnull_v = v_col is None or not v_col.validity
+            if func == "count":
+                out_dtype, want_valid = rt.I64, False
+            elif is_big and nonnull_v and func in ("sum", "min", "max"):
+                out_dtype, want_valid = rt.I64, False
+            else:
+                out_dtype, want_valid = rt.F64, True
+            default_bits, has_def = 0, False
+        col = runtime.window_ordered(perm, n, ps, fs, func, v_col,
+                                     spec.offset, default_bits, has_def,
+                                     out_dtype, want_valid)
+        col._keep_alive = (perm, ps, fs, v_col)
+        return col
 
     def _device_agg(self, runtime, cols, n, spec):
         """groupby → per-group value → hash join-back → row scatter."""

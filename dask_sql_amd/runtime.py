@@ -632,6 +632,37 @@ class Runtime:
         _check(self.lib, rc, "dsx_sort_perm")
         return DeviceColumn(self, perm.value, None, n, I32, owner=True)
 
+    WIN_FUNCS = {"row_number": 0, "rank": 1, "dense_rank": 2, "lag": 3,
+                 "lead": 4, "first_value": 5, "sum": 6, "count": 7,
+                 "min": 8, "max": 9, "avg": 10}
+
+    def window_ordered(self, perm, n, pcode_sorted, fcode_sorted, func,
+                       v_col, offset, default_bits, has_default, out_dtype,
+                       want_valid):
+        """Device ordered window frame (dsx_window_ordered)."""
+        d = ct.c_void_p()
+        vv = ct.c_void_p()
+        vstruct = None
+        if v_col is not None:
+            vstruct = v_col.c_struct()
+        _check(
+            self.lib,
+            self.lib.dsx_window_ordered(
+                self.ctx, ct.c_void_p(perm.data), ct.c_int64(n),
+                ct.c_void_p(pcode_sorted.data),
+                ct.c_void_p(fcode_sorted.data) if fcode_sorted is not None
+                else None,
+                ct.c_int(self.WIN_FUNCS[func]),
+                ct.byref(vstruct) if vstruct is not None else None,
+                ct.c_int64(offset), ct.c_int64(default_bits),
+                ct.c_int(1 if has_default else 0), ct.c_int(out_dtype),
+                ct.byref(d), ct.byref(vv),
+                ct.c_int(1 if want_valid else 0)),
+            "dsx_window_ordered",
+        )
+        return DeviceColumn(self, d.value, vv.value or None, n, out_dtype,
+                            owner=True)
+
     def hash_table_free(self, table):
         self.lib.dsx_hash_table_free(table)
 

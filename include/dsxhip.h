@@ -278,6 +278,19 @@ int dsx_sort_perm(DsxCtx* ctx, const DsxColumn* cols, int ncols,
                   const DsxKeySpec* keys, int nkeys, int64_t n,
                   uint32_t** out_perm);
 
+/* Device ordered window frames (reference rel/logical/window.py:212-428):
+ * caller provides the sort permutation over (partition, order) keys and the
+ * partition / partition+order codes gathered into sorted order; computes
+ * ROW_NUMBER/RANK/DENSE_RANK/LAG/LEAD/FIRST_VALUE and the running
+ * SUM/COUNT/MIN/MAX (RANGE UNBOUNDED..CURRENT with peer broadcast),
+ * scattered back to original row order. func = DsxWinFunc (0..9). */
+int dsx_window_ordered(DsxCtx* ctx, const uint32_t* perm, int64_t n,
+                       const uint64_t* pcode_sorted,
+                       const uint64_t* fcode_sorted, int func,
+                       const DsxColumn* value_or_null, int64_t offset,
+                       int64_t default_bits, int has_default, int out_dtype,
+                       void** out_data, uint8_t** out_valid, int want_valid);
+
 /* ---- hash groupby-aggregate --------------------------------------------- */
 
 enum DsxAggOp {  /* reference AGGREGATION_MAPPING aggregate.py:117-231 subset:

@@ -1628,3 +1628,53 @@ def test_device_sort_strings_and_desc(ctx):
     exp = exp.sort_values("s", ascending=False,
                           kind="mergesort").reset_index(drop=True)
     assert out["p"].astype(np.int64).tolist() == exp["p"].tolist()
+
+
+def test_device_window_ordered_at_scale(ctx):
+    """Ordered window frames run device-side at scale (dsx_window_ordered;
+    VERDICT r1 #5) and match pandas exactly: ROW_NUMBER, RANK, LAG and a
+    running SUM over 2M rows / 50k partitions."""
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(51)
+    n = 2_000_000
+    p = rng.integers(0, 50_000, n).astype(np.int64)
+    o = rng.integers(0, 10_000, n).astype(np.int64)
+    v = rng.random(n)
+    df = pd.DataFrame({"p": p, "o": o, "v": v})
+    c = Context()
+    c.create_table("tw", df)
+    runtime = c._get_runtime()
+    runtime.prof_enable(True)
+    runtime.prof_reset()
+    out = c.sql(
+        "SELECT p, o, v, "
+        "ROW_NUMBER() OVER (PARTITION BY p ORDER BY o) AS rn, "
+        "RANK() OVER (PARTITION BY p ORDER BY o) AS rk, "
+        "LAG(v) OVER (PARTITION BY p ORDER BY o) AS lg, "
+        "SUM(v) OVER (PARTITION BY p ORDER BY o) AS rs "
+        "FROM tw").compute()
+    prof = runtime.prof_get()
+    runtime.prof_enable(False)
+    assert prof.get("k_win_pos", {}).get("launches", 0) > 0
+    assert prof.get("k_win_scan", {}).get("launches", 0) > 0
+
+    out = out.sort_values(["p", "o", "v"]).reset_index(drop=True)
+    pdf = df.sort_values("o", kind="mergesort")
+    pdf = pdf.sort_values("p", kind="mergesort")
+    g = pdf.groupby("p", sort=False)
+    pdf = pdf.assign(rn=g.cumcount() + 1)
+    tie = pdf.groupby(["p", "o"], sort=False)
+    pdf = pdf.assign(rk=tie["rn"].transform("first"),
+                     lg=g["v"].shift(1))
+    cum = g["v"].cumsum()
+    pdf = pdf.assign(rs=cum.groupby(pdf["p"]).ffill())
+    pdf = pdf.assign(
+        rs=pdf.groupby(["p", "o"], sort=False)["rs"].transform("last"))
+    exp = pdf.sort_values(["p", "o", "v"]).reset_index(drop=True)
+    assert out["rn"].astype(np.int64).tolist() == exp["rn"].tolist()
+    assert out["rk"].astype(np.int64).tolist() == exp["rk"].tolist()
+    assert (out["lg"].isna() == exp["lg"].isna()).all()
+    np.testing.assert_allclose(out["lg"].fillna(0), exp["lg"].fillna(0),
+                               rtol=1e-9)
+    np.testing.assert_allclose(out["rs"].astype(np.float64),
+                               exp["rs"].to_numpy(), rtol=1e-9)
