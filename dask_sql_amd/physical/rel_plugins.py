@@ -2279,7 +2279,8 @@ class DaskWindowPlugin(BaseRelPlugin):
             out_dtype, want_valid = rt.I64, False
             default_bits, has_def = 0, False
         elif func in ("lag", "lead", "first_value"):
-            if v_col is None or v_col.dictionary is not None:
+            if v_col is None or getattr(v_col, "dictionary",
+                                        None) is not None:
                 return None  # dict value shift: host (string payload)
             out_dtype = v_col.dtype
             want_valid = True
@@ -2296,7 +2297,8 @@ class DaskWindowPlugin(BaseRelPlugin):
         else:
             if func != "count" and v_col is None:
                 return None
-            if v_col is not None and v_col.dictionary is not None:
+            if v_col is not None and getattr(v_col, "dictionary",
+                                             None) is not None:
                 return None
             # host emits int64 when the plan type is BIGINT and nothing is
             # NULL; float64 (+validity) otherwise (_host_ordered tail)
