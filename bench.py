@@ -266,8 +266,9 @@ def run_workload(workload, args, world, rank, local_rank, pg,
     if torch.cuda.is_available():
         torch.cuda.synchronize(local_rank)
 
-    runtime.prof_enable(True)
-    runtime.prof_reset()
+    # timed region runs WITHOUT per-kernel profiling (HIP event
+    # create/record cost ~0.2 ms/step at Q3's ~44 launches); a separate
+    # profiled pass afterwards provides the kernel breakdown
     if world > 1:
         import torch.distributed as dist
         dist.barrier()
@@ -287,6 +288,14 @@ def run_workload(workload, args, world, rank, local_rank, pg,
                          device=f"cuda:{local_rank}")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
+    # profiled pass (same step count → same per-step kernel accounting)
+    runtime.prof_enable(True)
+    runtime.prof_reset()
+    for _ in range(args.steps):
+        run_step(c, workload, world, pg)
+    runtime.synchronize()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize(local_rank)
     prof = runtime.prof_get()
     runtime.prof_enable(False)
 

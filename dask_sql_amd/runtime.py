@@ -125,7 +125,15 @@ class DeviceColumn:
         self._keep_alive = keep_alive  # e.g. torch tensor backing the ptrs
 
     def c_struct(self):
-        return _Column(self.data, self.validity or None, self.len, self.dtype)
+        # cached: data/validity/len are immutable after construction and
+        # ~40 ctypes struct builds per query step measured on the Q3 host
+        # overhead profile
+        c = getattr(self, "_c_struct", None)
+        if c is None:
+            c = _Column(self.data, self.validity or None, self.len,
+                        self.dtype)
+            self._c_struct = c
+        return c
 
     def __del__(self):
         if getattr(self, "_owner", False) and self.rt and self.rt.lib:
