@@ -2104,7 +2104,15 @@ def _device_topk_impl(context, inp, below, keys, k):
     kcol = pdf.iloc[:, idx0]
     if kcol.isna().to_numpy().any():
         return None  # NULL/NaN keys present: exact NULLS ordering on host
-    return _topk(pdf, keys, k)
+    # candidates are NaN-free: ONE stable multi-key sort replaces the
+    # argpartition + per-key mergesort chain (~0.5 ms of pandas overhead
+    # per step on the Q3 headline; same ordering — pandas multi-column
+    # stable sort is lexicographic with tie order preserved)
+    if len(set(pdf.columns)) != len(pdf.columns):
+        return _topk(pdf, keys, k)  # duplicate names: positional path
+    by = [pdf.columns[i] for i, _a, _nf in keys]
+    asc = [a for _i, a, _nf in keys]
+    return pdf.sort_values(by, ascending=asc, kind="stable").iloc[:k]
 
 
 def _topk(pdf, keys, k):
