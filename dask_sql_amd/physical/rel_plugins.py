@@ -2122,12 +2122,31 @@ def _device_topk_impl(context, inp, below, keys, k):
         if np.asarray(raws[i]).dtype.kind == "f" and                 np.isnan(raws[i]).any():
             break
     else:
-        import pandas as pd
-        raw_pdf = pd.DataFrame({j: raws[j] for j in range(len(raws))})
-        by = [i for i, _a, _nf in keys]
-        asc = [a for _i, a, _nf in keys]
-        top = raw_pdf.sort_values(by, ascending=asc,
-                                  kind="stable").index[:k].to_numpy()
+        # np.lexsort: last key = primary; DESC via exact negation (f64 and
+        # sub-64-bit ints negate exactly in i64/f64; i64 at INT64_MIN bails)
+        lex = []
+        ok = True
+        for i, a, _nf in keys:
+            arr = np.asarray(raws[i])
+            if a:
+                lex.append(arr)
+            elif arr.dtype.kind == "f":
+                lex.append(-arr)
+            else:
+                a64 = arr.astype(np.int64)
+                if a64.size and a64.min() == np.iinfo(np.int64).min:
+                    ok = False
+                    break
+                lex.append(-a64)
+        if ok:
+            top = np.lexsort(tuple(reversed(lex)))[:k]
+        else:
+            import pandas as pd
+            raw_pdf = pd.DataFrame({j: raws[j] for j in range(len(raws))})
+            by = [i for i, _a, _nf in keys]
+            asc = [a for _i, a, _nf in keys]
+            top = raw_pdf.sort_values(by, ascending=asc,
+                                      kind="stable").index[:k].to_numpy()
         from dask_sql_amd.materialize import _convert
         fields = below.getRowType().getFieldList()
         data = {}
