@@ -2122,6 +2122,8 @@ def _device_topk_impl(context, inp, below, keys, k):
         if np.asarray(raws[i]).dtype.kind == "f" and                 np.isnan(raws[i]).any():
             break
     else:
+        import pandas as pd
+
         # np.lexsort: last key = primary; DESC via exact negation (f64 and
         # sub-64-bit ints negate exactly in i64/f64; i64 at INT64_MIN bails)
         lex = []
@@ -2141,7 +2143,6 @@ def _device_topk_impl(context, inp, below, keys, k):
         if ok:
             top = np.lexsort(tuple(reversed(lex)))[:k]
         else:
-            import pandas as pd
             raw_pdf = pd.DataFrame({j: raws[j] for j in range(len(raws))})
             by = [i for i, _a, _nf in keys]
             asc = [a for _i, a, _nf in keys]
