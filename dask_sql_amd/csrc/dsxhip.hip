@@ -77,7 +77,23 @@ struct DsxCtx {
   // of the data, so the mid-pipeline totals sync runs once per table, not
   // every step (repeat steps of the same query pay no guard stall)
   std::unordered_map<uint64_t, int> gb_guard_cache;
+  // per-table (hist, bases) cache for the partition groupby: the histogram
+  // is predicate-free (pure function of the key column), so repeat steps
+  // skip the hist+scan passes. Sig = key col ptrs + n + space + layout +
+  // per-key min/range (aliasing a recycled device ptr additionally needs
+  // identical n/min/max — accepted residual risk, like the stats cache).
+  struct GbHist {
+    uint64_t sig;
+    void* hist;
+    void* bases;
+    const void* srcs[DSX_MAX_KEYS];  // key column data ptrs: a free of any
+    int nsrc;                        // of these evicts the entry (stale-
+                                     // pointer-reuse hazard)
+  };
+  std::vector<GbHist> gb_hist_cache;
 };
+
+static void gb_hist_evict_ptr(DsxCtx* c, void* p);
 
 static int64_t pool_round(int64_t bytes) {
   int64_t r = 256;
@@ -111,6 +127,7 @@ static int pool_alloc(DsxCtx* c, int64_t bytes, void** out) {
 }
 
 static void pool_release(DsxCtx* c, void* p) {
+  gb_hist_evict_ptr(c, p);
   if (!p) return;
   auto it = c->pool_sizes.find(p);
   if (it == c->pool_sizes.end()) {
@@ -180,6 +197,10 @@ extern "C" void dsx_ctx_destroy(DsxCtx* c) {
   for (auto& r : c->prof_recs) {
     hipEventDestroy(r.start);
     hipEventDestroy(r.stop);
+  }
+  for (auto& h : c->gb_hist_cache) {
+    pool_release(c, h.hist);
+    pool_release(c, h.bases);
   }
   if (c->scratch) hipFree(c->scratch);
   if (c->pinned) hipHostFree(c->pinned);
@@ -2518,6 +2539,25 @@ __global__ void k_count_live(const uint64_t* tkeys,
 // direct_shift ≥ 0 selects RANGE bucketing + the direct-index aggregate
 // (no hash table in LDS); pick_direct_shift decides, the wrapper retries
 // with hashing when the range histogram turns out skewed (*redo).
+static void gb_hist_evict_ptr(DsxCtx* c, void* p) {
+  if (c->gb_hist_cache.empty()) return;
+  for (size_t i = 0; i < c->gb_hist_cache.size();) {
+    bool dead = false;
+    for (int j = 0; j < c->gb_hist_cache[i].nsrc; j++)
+      if (c->gb_hist_cache[i].srcs[j] == p) dead = true;
+    if (dead) {
+      // NOTE: release the entry's own buffers WITHOUT re-entering the
+      // eviction hook (they are never key sources)
+      DsxCtx::GbHist h = c->gb_hist_cache[i];
+      c->gb_hist_cache.erase(c->gb_hist_cache.begin() + i);
+      pool_release(c, h.hist);
+      pool_release(c, h.bases);
+    } else {
+      i++;
+    }
+  }
+}
+
 static int pick_direct_shift(uint64_t key_space, int nvals) {
   static const bool off = getenv("DSX_GB_NO_DIRECT") != nullptr;
   if (off || !jit_enabled() || key_space == 0) return -1;
@@ -2617,6 +2657,7 @@ static int groupby_partition_impl(
   int64_t need = prog_bytes + ((lens_bytes + 15) / 16) * 16 + 64 +
                  sizeof(KeyArg) + sizeof(AggArg) + 32 +
                  (int64_t)grid * nb * 8 + (nb + 2) * 8 + nb * 8 +
+                 (int64_t)grid * nb * 4 + 16 +
                  Gcap * 8 + Gcap * (int64_t)(nvals ? nvals : 1) * 8 +
                  Gcap * 8 + ((naggs * 4 + 15) / 16) * 16 + 256 + rec_bytes;
   int rc = ensure_scratch(c, need);
@@ -2642,6 +2683,8 @@ static int groupby_partition_impl(
   base += nb * 8;
   int64_t* d_bases = (int64_t*)base;
   base += (nb + 2) * 8;
+  uint32_t* d_selcnt = (uint32_t*)base;
+  base += (((int64_t)grid * nb * 4 + 15) / 16) * 16;
   uint64_t* d_tmp_codes = (uint64_t*)base;
   base += Gcap * 8;
   uint64_t* d_tmp_vals = (uint64_t*)base;
@@ -2711,19 +2754,58 @@ static int groupby_partition_impl(
       (uint64_t*)((char*)d_recs + (((int64_t)(n > 0 ? n : 1) * 4 + 255) / 256) *
                                       256);
   if (f_hist && f_scat) {
-    {
-      ProfScope ps(c, "k_gbpart_hist");
-      struct { ColsArg C; int64_t n; int nb; int64_t* hist; } a1{C, n, nb,
-                                                                 d_hist};
-      void* args[] = {&a1.C, &a1.n, &a1.nb, &a1.hist};
-      hipModuleLaunchKernel(f_hist, grid, 1, 1, BLOCK, 1, 1,
-                            (unsigned)(nb * 4), c->stream, args, nullptr);
+    // per-table (hist, bases) cache: the JIT histogram is predicate-free,
+    // a pure function of the key column — repeat steps skip hist+scan
+    uint64_t hsig = ((uint64_t)n * 0x9E3779B97F4A7C15ull) ^ key_space ^
+                    ((uint64_t)nb << 1) ^ ((uint64_t)grid << 17) ^
+                    (uint64_t)(direct_shift + 2);
+    auto hmix = [](uint64_t x) {
+      x += 0x9E3779B97F4A7C15ull;
+      x ^= x >> 30; x *= 0xBF58476D1CE4E5B9ull;
+      x ^= x >> 27; x *= 0x94D049BB133111EBull;
+      return x ^ (x >> 31);
+    };
+    for (int j = 0; j < K.nkeys; j++) {
+      hsig = hmix(hsig ^ (uint64_t)(uintptr_t)C.data[K.k[j].col]);
+      hsig = hmix(hsig ^ (uint64_t)K.k[j].min ^
+                  ((uint64_t)K.k[j].range << 8));
     }
-    hipLaunchKernelGGL(k_gbpart_scan, dim3(nb / 8), dim3(BLOCK), 0, c->stream,
-                       d_hist, grid, nb, d_totals);
-    hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(1024), 0, c->stream,
-                       d_totals, nb, d_bases);
-    if (direct_shift >= 0) {
+    DsxCtx::GbHist* hit = nullptr;
+    for (auto& h : c->gb_hist_cache)
+      if (h.sig == hsig) hit = &h;
+    if (hit != nullptr) {
+      d_hist = (int64_t*)hit->hist;
+      d_bases = (int64_t*)hit->bases;
+    } else {
+      DsxCtx::GbHist ch{};
+      ch.sig = hsig;
+      ch.nsrc = K.nkeys;
+      for (int j = 0; j < K.nkeys; j++) ch.srcs[j] = C.data[K.k[j].col];
+      int rc2 = pool_alloc(c, (int64_t)grid * nb * 8, &ch.hist);
+      if (!rc2) rc2 = pool_alloc(c, ((int64_t)nb + 2) * 8, &ch.bases);
+      if (rc2) return rc2;
+      d_hist = (int64_t*)ch.hist;
+      d_bases = (int64_t*)ch.bases;
+      {
+        ProfScope ps(c, "k_gbpart_hist");
+        struct { ColsArg C; int64_t n; int nb; int64_t* hist; } a1{
+            C, n, nb, d_hist};
+        void* args[] = {&a1.C, &a1.n, &a1.nb, &a1.hist};
+        hipModuleLaunchKernel(f_hist, grid, 1, 1, BLOCK, 1, 1,
+                              (unsigned)(nb * 4), c->stream, args, nullptr);
+      }
+      hipLaunchKernelGGL(k_gbpart_scan, dim3(nb / 8), dim3(BLOCK), 0,
+                         c->stream, d_hist, grid, nb, d_totals);
+      hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(1024), 0, c->stream,
+                         d_totals, nb, d_bases);
+      if (c->gb_hist_cache.size() >= 6) {
+        pool_release(c, c->gb_hist_cache.front().hist);
+        pool_release(c, c->gb_hist_cache.front().bases);
+        c->gb_hist_cache.erase(c->gb_hist_cache.begin());
+      }
+      c->gb_hist_cache.push_back(ch);
+    }
+    if (hit == nullptr && direct_shift >= 0) {
       // balance guard: a skewed key distribution overloads range buckets
       // (one aggregate block per bucket) — fall back to hash bucketing.
       // Verdict cached per (key data ptr, n, space): repeat steps of the
@@ -2780,18 +2862,20 @@ static int groupby_partition_impl(
       if (code32) {
         struct {
           ColsArg C; int64_t n; int nb; const int64_t* hist;
-          const int64_t* bases; uint32_t* oc; uint64_t* ov;
-        } a2{C, n, nb, d_hist, d_bases, d_rcodes, d_rvals};
-        void* args[] = {&a2.C, &a2.n, &a2.nb, &a2.hist, &a2.bases, &a2.oc,
-                        &a2.ov};
+          const int64_t* bases; uint32_t* selcnt; uint32_t* oc;
+          uint64_t* ov;
+        } a2{C, n, nb, d_hist, d_bases, d_selcnt, d_rcodes, d_rvals};
+        void* args[] = {&a2.C, &a2.n, &a2.nb, &a2.hist, &a2.bases,
+                        &a2.selcnt, &a2.oc, &a2.ov};
         hipModuleLaunchKernel(f_use, grid, 1, 1, sthreads, 1, 1,
                               smem, c->stream, args, nullptr);
       } else {
         struct {
           ColsArg C; int64_t n; int nb; const int64_t* hist;
-          const int64_t* bases; uint64_t* out;
-        } a2{C, n, nb, d_hist, d_bases, d_recs};
-        void* args[] = {&a2.C, &a2.n, &a2.nb, &a2.hist, &a2.bases, &a2.out};
+          const int64_t* bases; uint32_t* selcnt; uint64_t* out;
+        } a2{C, n, nb, d_hist, d_bases, d_selcnt, d_recs};
+        void* args[] = {&a2.C, &a2.n, &a2.nb, &a2.hist, &a2.bases,
+                        &a2.selcnt, &a2.out};
         hipModuleLaunchKernel(f_use, grid, 1, 1, sthreads, 1, 1,
                               smem, c->stream, args, nullptr);
       }
@@ -2824,23 +2908,27 @@ static int groupby_partition_impl(
       if (code32) {
         struct {
           const uint32_t* rc; const uint64_t* rv; const int64_t* bases;
+          const int64_t* hist; const uint32_t* selcnt; int sgrid;
           unsigned long long* counter; uint64_t* tc; uint64_t* tv;
           unsigned long long* tg; int* ovf;
-        } a3{d_rcodes, d_rvals, d_bases, d_counter, d_tmp_codes,
-             (uint64_t*)d_tmp_vals, d_tmp_gcnt, d_ovf};
-        void* args[] = {&a3.rc, &a3.rv, &a3.bases, &a3.counter, &a3.tc,
-                        &a3.tv, &a3.tg, &a3.ovf};
+        } a3{d_rcodes, d_rvals, d_bases, d_hist, d_selcnt, grid, d_counter,
+             d_tmp_codes, (uint64_t*)d_tmp_vals, d_tmp_gcnt, d_ovf};
+        void* args[] = {&a3.rc, &a3.rv, &a3.bases, &a3.hist, &a3.selcnt,
+                        &a3.sgrid, &a3.counter, &a3.tc, &a3.tv, &a3.tg,
+                        &a3.ovf};
         hipModuleLaunchKernel(f_aggr, nb, 1, 1, athreads, 1, 1,
                               (unsigned)lds_bytes, c->stream, args, nullptr);
       } else {
         struct {
-          const uint64_t* recs; const int64_t* bases;
+          const uint64_t* recs; const int64_t* bases; const int64_t* hist;
+          const uint32_t* selcnt; int sgrid;
           unsigned long long* counter; uint64_t* tc; uint64_t* tv;
           unsigned long long* tg; int* ovf;
-        } a3{d_recs, d_bases, d_counter, d_tmp_codes,
-             (uint64_t*)d_tmp_vals, d_tmp_gcnt, d_ovf};
-        void* args[] = {&a3.recs, &a3.bases, &a3.counter, &a3.tc, &a3.tv,
-                        &a3.tg, &a3.ovf};
+        } a3{d_recs, d_bases, d_hist, d_selcnt, grid, d_counter,
+             d_tmp_codes, (uint64_t*)d_tmp_vals, d_tmp_gcnt, d_ovf};
+        void* args[] = {&a3.recs, &a3.bases, &a3.hist, &a3.selcnt,
+                        &a3.sgrid, &a3.counter, &a3.tc, &a3.tv, &a3.tg,
+                        &a3.ovf};
         hipModuleLaunchKernel(f_aggr, nb, 1, 1, athreads, 1, 1,
                               (unsigned)lds_bytes, c->stream, args, nullptr);
       }
