@@ -1678,3 +1678,52 @@ def test_device_window_ordered_at_scale(ctx):
                                rtol=1e-9)
     np.testing.assert_allclose(out["rs"].astype(np.float64),
                                exp["rs"].to_numpy(), rtol=1e-9)
+
+
+def test_rccl_exchange_world1(ctx):
+    """Execute the REAL RCCL branch of the distributed exchange at
+    world_size=1 (gpurun is single-GPU): validates the all_to_all_single /
+    all_gather call shapes, dtype staging and validity shipping that the
+    driver's 8-GPU scaling run will exercise at N>1."""
+    import os
+    import torch
+    import torch.distributed as dist
+
+    from dask_sql_amd.distributed import (allgather_device_columns,
+                                          exchange_buckets,
+                                          shuffle_device_columns)
+    if dist.is_initialized():
+        pytest.skip("process group already up")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29571")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("WORLD_SIZE", "1")
+    torch.cuda.set_device(0)
+    dist.init_process_group(backend="nccl", rank=0, world_size=1)
+    try:
+        runtime = ctx._get_runtime()
+        n = 100_000
+        rng = np.random.default_rng(61)
+        key = runtime.upload_column(
+            rng.integers(0, 1000, n).astype(np.int64))
+        vals = pd.array(rng.random(n))
+        varr = np.asarray(vals, dtype=np.float64)
+        validity = (rng.random(n) > 0.1).astype(np.uint8)
+        pay = runtime.upload_column(varr, validity)
+        # RCCL all_to_all_single path (world 1: identity exchange)
+        rkey, rvals, splits = shuffle_device_columns(runtime, key, [pay])
+        assert splits == [n]
+        a, _ = rkey.to_numpy()
+        kh, _ = key.to_numpy()
+        assert sorted(a.tolist()) == sorted(kh.tolist())
+        rv, rvalid = rvals[0].to_numpy()
+        assert rvalid is not None and rvalid.sum() == validity.sum()
+        # broadcast-join replication path
+        out = allgather_device_columns(runtime, [key, pay], n)
+        assert out[0].len == n and out[1].validity
+        # raw exchange: one tensor, nccl backend
+        t = torch.arange(1000, dtype=torch.int64, device="cuda:0")
+        recv, osp = exchange_buckets([t], [1000])
+        assert osp == [1000] and recv[0].sum().item() == t.sum().item()
+    finally:
+        dist.destroy_process_group()
