@@ -33,7 +33,7 @@ _DSX_TO_NP = {
 }
 _DSX_SIZE = {I64: 8, F64: 8, I32: 4, F32: 4, I8: 1, BOOL8: 1}
 
-MAX_PROG = 48
+MAX_PROG = 120
 MAX_COLS = 16
 MAX_AGGS = 16
 

@@ -82,7 +82,7 @@ typedef struct DsxInstr {
   int64_t imm;      /* literal bits */
 } DsxInstr;
 
-#define DSX_MAX_PROG 48
+#define DSX_MAX_PROG 120
 #define DSX_MAX_COLS 16
 #define DSX_MAX_AGGS 16
 #define DSX_MAX_KEYS 4
