@@ -521,3 +521,46 @@ def test_from_arrow_ingest(tmp_path):
     assert cols["b"].dtype == rt.BOOL8
     assert cols["i"].arr[-1] == n - 1
     assert abs(cols["f"].arr[7] - 1.0) < 1e-12
+
+
+def test_udf_agg_plan_rewrite():
+    """Registered aggregate UDFs parse as plain calls but must plan as
+    Aggregate nodes (builder._rewrite_udf_aggs; reference
+    register_aggregation routes through the Aggregate rel)."""
+    c = Context()
+    c.create_table("t", pd.DataFrame({"k": [1, 2], "b": [1.0, 2.0]}))
+
+    class A:
+        def __init__(self):
+            self.chunk = lambda s: s.sum()
+            self.agg = lambda s: s.sum()
+
+    c.register_aggregation(A(), "fagg", [("x", np.float64)], np.float64)
+    rel = c._get_ral("SELECT k, FAGG(b) AS f FROM t GROUP BY k")
+    node = rel
+    while node.get_current_node_type() != "Aggregate":
+        node = node.get_inputs()[0]
+    agg = node.aggregate()
+    names = [agg.getAggregationFuncName(call)
+             for call in agg.getNamedAggCalls()]
+    assert "udf:fagg" in names
+
+
+def test_udf_scalar_plan_typing():
+    c = Context()
+    c.create_table("t", pd.DataFrame({"a": [1.0, 2.0]}))
+
+    def f(x):
+        return x ** 2
+
+    c.register_function(f, "f", [("x", np.float64)], np.float64)
+    rel = c._get_ral("SELECT F(a) AS y FROM t")
+    proj = rel.projection().getNamedProjects()
+    expr, name = proj[0]
+    assert name == "y"
+    assert expr.getOperatorName() == "UDF:f"
+    assert expr.getType().getSqlType() == "DOUBLE"
+    # plan cache keyed on schema version: registering bumps it
+    v0 = c._schema_version
+    c.register_function(f, "g", [("x", np.float64)], np.float64)
+    assert c._schema_version > v0
