@@ -237,7 +237,8 @@ class Builder:
         return LogicalPlan("Join", [plan, subplan], RelDataType(lhs_fields),
                            JoinNode(jt, cond))
 
-    def _decorrelate_exists(self, sub, lead_items=None):
+    def _decorrelate_exists(self, sub, lead_items=None, tail_items=None,
+                            group=False, key_prefix=""):
         """Split the EXISTS subquery's WHERE into local conjuncts and
         equality correlations on OUTER columns (qualified names not bound
         by the sub's own FROM). Returns (subplan selecting the inner keys,
@@ -284,9 +285,12 @@ class Builder:
         for cj in local:
             where = cj if where is None else ("call", "AND", [where, cj])
         items = list(lead_items or [])
-        items += [(k, f"ck{i}") for i, k in enumerate(inner_keys)]
+        items += [(k, f"{key_prefix}ck{i}") for i, k in
+                  enumerate(inner_keys)]
+        items += list(tail_items or [])
         s2 = SelectStmt(items=items, from_tables=sub.from_tables,
-                        joins=sub.joins, where=where)
+                        joins=sub.joins, where=where,
+                        group_by=list(inner_keys) if group else [])
         return self.build_stmt(s2), outer_keys
 
     @staticmethod
@@ -459,6 +463,21 @@ class Builder:
         _pushdown = bool(_config.get("sql.predicate_pushdown", True))
         lhs_preserved = lhs_preserved and _pushdown
 
+        def _has_ssub(ast):
+            if not isinstance(ast, tuple):
+                return False
+            if ast[0] == "scalar_sub":
+                return True
+            if ast[0] == "call":
+                return any(_has_ssub(a) for a in ast[2])
+            if ast[0] == "cast":
+                return _has_ssub(ast[1])
+            if ast[0] == "case":
+                return any(_has_ssub(x) for c, v in ast[1]
+                           for x in (c, v)) or (
+                    ast[2] is not None and _has_ssub(ast[2]))
+            return False
+
         # 1. scans (+ pushed-down single-table filters, à la PushDownFilter)
         def scan_with_filters(tr: TableRef, push: bool = True) -> LogicalPlan:
             plan = self._scan(tr)
@@ -467,7 +486,9 @@ class Builder:
             quals = {(tr.alias or tr.name).lower()}
             conds = []
             for i, cj in enumerate(where_conjuncts):
-                if used[i] or self._has_agg(cj):
+                if used[i] or self._has_agg(cj) or _has_ssub(cj):
+                    # scalar subqueries resolve above the joins (correlated
+                    # ones LEFT-join a grouped subplan) — never pushed
                     continue
                 refs = self._tables_of(cj)
                 if not refs:
@@ -530,7 +551,7 @@ class Builder:
             conds = []
             if lhs_preserved:
                 for i, cj in enumerate(where_conjuncts):
-                    if used[i] or self._has_agg(cj):
+                    if used[i] or self._has_agg(cj) or _has_ssub(cj):
                         continue
                     refs = self._tables_of(cj)
                     if not refs:
@@ -554,6 +575,74 @@ class Builder:
             ) and not any(t in ("RIGHT", "FULL") for t in join_types[k + 1:])
             rhs = scan_with_filters(jc.table, push=rhs_push)
             plan = join_plans(plan, rhs, jc.join_type, [], jc.on)
+
+        # correlated scalar subqueries with equality correlation (e.g.
+        # `(SELECT MAX(x) FROM u WHERE u.k = t.k)`): rewrite to a GROUPED
+        # subplan LEFT-JOINed on the correlation keys — DataFusion's
+        # scalar-subquery decorrelation on the reference side (round-1
+        # raised on these). The sub's single item must be an aggregate.
+        n_user_fields = len(plan.getRowType().getFieldList())
+        ssub_n = [0]
+
+        def _pull_ssubs(ast):
+            nonlocal plan
+            if not isinstance(ast, tuple):
+                return ast
+            if ast[0] == "scalar_sub":
+                sub = ast[1]
+                if (len(sub.items) == 1
+                        and isinstance(sub.items[0][0], tuple)
+                        and sub.items[0][0][0] == "agg"
+                        and not sub.group_by):
+                    idx = ssub_n[0]
+                    valname = f"__ssub{idx}"
+                    try:
+                        subplan, outer_keys = self._decorrelate_exists(
+                            sub, tail_items=[(sub.items[0][0], valname)],
+                            group=True, key_prefix=f"__ssub{idx}_")
+                    except NotImplementedError:
+                        return ast
+                    if subplan is None:
+                        return ast  # uncorrelated: eager scalar later
+                    ssub_n[0] += 1
+                    sfields = subplan.getRowType().getFieldList()
+                    lhs_fields = plan.getRowType().getFieldList()
+                    tmp = LogicalPlan(
+                        "__combined__", [],
+                        RelDataType(lhs_fields + sfields), None)
+                    cond = None
+                    for i, ka in enumerate(outer_keys):
+                        eq = Call("=", [
+                            self._resolve(ka, tmp),
+                            InputRef(len(lhs_fields) + i,
+                                     sfields[i].getType())],
+                            SqlType("BOOLEAN"))
+                        cond = eq if cond is None else Call(
+                            "AND", [cond, eq], SqlType("BOOLEAN"))
+                    plan = LogicalPlan(
+                        "Join", [plan, subplan],
+                        RelDataType(lhs_fields + sfields),
+                        JoinNode("LEFT", cond))
+                    return ("col", None, valname)
+                return ast
+            if ast[0] == "call":
+                return (ast[0], ast[1], [_pull_ssubs(a) for a in ast[2]])
+            if ast[0] == "cast":
+                return (ast[0], _pull_ssubs(ast[1]), ast[2])
+            if ast[0] == "case":
+                return (ast[0],
+                        [(_pull_ssubs(c), _pull_ssubs(v))
+                         for c, v in ast[1]],
+                        _pull_ssubs(ast[2]) if ast[2] is not None else None)
+            if ast[0] == "agg":
+                return (ast[0], ast[1], [_pull_ssubs(a) for a in ast[2]],
+                        ast[3], ast[4])
+            return ast
+
+        stmt.items = [(_pull_ssubs(e), a) for e, a in stmt.items]
+        for i, cj in enumerate(where_conjuncts):
+            if not used[i]:
+                where_conjuncts[i] = _pull_ssubs(cj)
 
         # EXISTS: equality-correlated → SEMI/ANTI join over the DISTINCT
         # correlation keys (DataFusion's decorrelation on the reference
@@ -605,7 +694,7 @@ class Builder:
         items = []
         for e, alias in stmt.items:
             if e == ("star",):
-                for f in plan.getRowType().getFieldList():
+                for f in plan.getRowType().getFieldList()[:n_user_fields]:
                     items.append((("col", f.qualifier, f.getName()), None))
             else:
                 items.append((e, alias))

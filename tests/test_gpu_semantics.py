@@ -1743,3 +1743,24 @@ def test_like_wide_dictionary(ctx):
     out = c.sql("SELECT COUNT(*) AS c FROM tl WHERE s LIKE 'w%'").compute()
     exp = int((codes < 30).sum())
     assert int(out["c"].iloc[0]) == exp
+
+
+def test_correlated_scalar_subquery_exec(ctx):
+    """Correlated scalar subqueries (equality correlation) execute via the
+    decorrelated grouped LEFT join; missing keys yield NULL (SQL scalar
+    subquery of zero rows)."""
+    from dask_sql_amd.context import Context
+    c = Context()
+    c.create_table("t", pd.DataFrame({"k": np.array([1, 2, 3], np.int64),
+                                      "x": [1.0, 2.0, 3.0]}))
+    c.create_table("u", pd.DataFrame({"k": np.array([1, 1, 2], np.int64),
+                                      "y": [5.0, 7.0, 9.0]}))
+    out = c.sql("SELECT t.k, (SELECT MAX(u.y) FROM u WHERE u.k = t.k) AS m "
+                "FROM t").compute()
+    out = out.sort_values("k").reset_index(drop=True)
+    assert out["m"].tolist()[:2] == [7.0, 9.0]
+    assert pd.isna(out["m"].iloc[2])
+    out2 = c.sql("SELECT t.k FROM t WHERE t.x < (SELECT AVG(u.y) FROM u "
+                 "WHERE u.k = t.k)").compute()
+    # k=1: 1.0 < 6.0 T; k=2: 2.0 < 9.0 T; k=3: NULL comparison -> excluded
+    assert sorted(out2["k"].astype(np.int64).tolist()) == [1, 2]

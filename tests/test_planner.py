@@ -564,3 +564,24 @@ def test_udf_scalar_plan_typing():
     v0 = c._schema_version
     c.register_function(f, "g", [("x", np.float64)], np.float64)
     assert c._schema_version > v0
+
+
+def test_correlated_scalar_subquery_plan():
+    """Equality-correlated scalar subqueries decorrelate into a grouped
+    subplan LEFT-joined on the correlation keys (DataFusion's rewrite;
+    round-1 raised on these)."""
+    c = Context()
+    c.create_table("t", pd.DataFrame({"k": [1, 2], "x": [1.0, 2.0]}))
+    c.create_table("u", pd.DataFrame({"k": [1, 1, 2], "y": [5.0, 7.0, 9.0]}))
+    rel = c._get_ral(
+        "SELECT t.k, t.x, (SELECT MAX(u.y) FROM u WHERE u.k = t.k) AS m "
+        "FROM t")
+    txt = rel.explain()
+    assert "Join" in txt and "Aggregate" in txt
+    # the output row type keeps exactly the user columns
+    assert rel.getRowType().getFieldNames() == ["k", "x", "m"]
+    # star expansion must NOT leak the internal __ssub columns
+    rel2 = c._get_ral(
+        "SELECT * FROM t WHERE t.x < (SELECT AVG(u.y) FROM u "
+        "WHERE u.k = t.k)")
+    assert rel2.getRowType().getFieldNames() == ["k", "x"]
