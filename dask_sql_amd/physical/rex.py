@@ -473,10 +473,11 @@ class RexCompiler:
         """predicate: column's dict code ∈ codes (OR-chain of EQ; NULL→NULL
         via EQ validity). Empty set → always-FALSE via a never-present
         code."""
-        if len(codes) > 32:
-            # 32 codes ≈ 100 VM instructions of the 120-slot program
-            # budget (DSX_MAX_PROG); larger dictionaries need a LUT-gather
-            # predicate column (round-3)
+        if len(codes) > 24:
+            # 24 codes = 95 VM instructions, leaving headroom in the
+            # 120-slot program budget (DSX_MAX_PROG) for the surrounding
+            # predicate; larger dictionaries need a LUT-gather predicate
+            # column (round-3)
             raise RexCompileError(
                 f"predicate matches {len(codes)} dictionary entries "
                 "(> VM program budget)")

@@ -1730,18 +1730,19 @@ def test_rccl_exchange_world1(ctx):
 
 
 def test_like_wide_dictionary(ctx):
-    """LIKE matching up to 32 dictionary entries (VM budget lifted from 10
-    to 32 with the 120-slot program; DESIGN known-limitations)."""
+    """LIKE matching up to 24 dictionary entries (VM budget lifted from 10
+    to 24 with the 120-slot program; DESIGN known-limitations)."""
     from dask_sql_amd.context import Context
-    words = [f"w{i:02d}x" for i in range(30)] + ["zzz", "yyy"]
+    words = [f"w{i:02d}x" for i in range(22)] + ["zzz", "yyy"]
     rng = np.random.default_rng(71)
     codes = rng.integers(0, len(words), 100_000).astype(np.int8)
     df = pd.DataFrame({"s": pd.Categorical.from_codes(codes, words),
                        "v": np.ones(100_000, dtype=np.int64)})
     c = Context()
     c.create_table("tl", df)
-    out = c.sql("SELECT COUNT(*) AS c FROM tl WHERE s LIKE 'w%'").compute()
-    exp = int((codes < 30).sum())
+    out = c.sql("SELECT COUNT(*) AS c FROM tl "
+                "WHERE s LIKE 'w%' AND v > 0").compute()
+    exp = int((codes < 22).sum())
     assert int(out["c"].iloc[0]) == exp
 
 
