@@ -91,6 +91,9 @@ struct DsxCtx {
                                      // pointer-reuse hazard)
   };
   std::vector<GbHist> gb_hist_cache;
+  bool gb_hist_cache_on = true;  // off for externally-backed key columns
+                                 // (torch tensors recycle pointers outside
+                                 // the pool's eviction hook)
 };
 
 static void gb_hist_evict_ptr(DsxCtx* c, void* p);
@@ -2771,8 +2774,9 @@ static int groupby_partition_impl(
                   ((uint64_t)K.k[j].range << 8));
     }
     DsxCtx::GbHist* hit = nullptr;
-    for (auto& h : c->gb_hist_cache)
-      if (h.sig == hsig) hit = &h;
+    if (c->gb_hist_cache_on)
+      for (auto& h : c->gb_hist_cache)
+        if (h.sig == hsig) hit = &h;
     if (hit != nullptr) {
       d_hist = (int64_t*)hit->hist;
       d_bases = (int64_t*)hit->bases;
@@ -2798,12 +2802,20 @@ static int groupby_partition_impl(
                          c->stream, d_hist, grid, nb, d_totals);
       hipLaunchKernelGGL(k_gbpart_bases, dim3(1), dim3(1024), 0, c->stream,
                          d_totals, nb, d_bases);
-      if (c->gb_hist_cache.size() >= 6) {
-        pool_release(c, c->gb_hist_cache.front().hist);
-        pool_release(c, c->gb_hist_cache.front().bases);
-        c->gb_hist_cache.erase(c->gb_hist_cache.begin());
+      if (c->gb_hist_cache_on) {
+        if (c->gb_hist_cache.size() >= 6) {
+          pool_release(c, c->gb_hist_cache.front().hist);
+          pool_release(c, c->gb_hist_cache.front().bases);
+          c->gb_hist_cache.erase(c->gb_hist_cache.begin());
+        }
+        c->gb_hist_cache.push_back(ch);
+      } else {
+        // uncached: hand the buffers back once the call completes — they
+        // are read by the aggregate later on the same stream, so release
+        // into the stream-ordered pool is safe
+        pool_release(c, ch.hist);
+        pool_release(c, ch.bases);
       }
-      c->gb_hist_cache.push_back(ch);
     }
     if (hit == nullptr && direct_shift >= 0) {
       // balance guard: a skewed key distribution overloads range buckets
@@ -2993,6 +3005,13 @@ static int groupby_partition(DsxCtx* c, ColsArg& C, int64_t n, KeyArg& K,
   return groupby_partition_impl(c, C, n, K, P, A, aggs_arr, progs, lens,
                                 naggs, g_est, key_space, out_codes, out_vals,
                                 out_counts, out_groups, fell_back, -1, &redo);
+}
+
+/* Toggle the per-table histogram cache (off for externally-backed key
+ * columns whose pointers recycle outside the pool's eviction hook). */
+extern "C" int dsx_gb_hist_cache_enable(DsxCtx* c, int enable) {
+  c->gb_hist_cache_on = enable != 0;
+  return 0;
 }
 
 extern "C" int dsx_hash_groupby(DsxCtx* c, const DsxColumn* cols, int ncols,

@@ -675,6 +675,23 @@ class Runtime:
         self.lib.dsx_hash_table_free(table)
 
     def hash_groupby(self, cols, n, keyspecs, pred_prog, agg_specs):
+        # the C-side hist cache keys on device pointers with pool-free
+        # eviction; externally-backed key columns (torch/RCCL staging)
+        # recycle pointers outside that hook — disable for them
+        unsafe = any(
+            getattr(cols[spec[0]], "_keep_alive", None) is not None
+            and not getattr(cols[spec[0]], "_owner", True)
+            for spec in keyspecs)
+        if unsafe:
+            self.lib.dsx_gb_hist_cache_enable(self.ctx, 0)
+        try:
+            return self._hash_groupby(cols, n, keyspecs, pred_prog,
+                                      agg_specs)
+        finally:
+            if unsafe:
+                self.lib.dsx_gb_hist_cache_enable(self.ctx, 1)
+
+    def _hash_groupby(self, cols, n, keyspecs, pred_prog, agg_specs):
         """keyspecs: list of (col_idx, min, range, nullable) — key pack fused
         in-kernel. agg_specs: list of (agg_op, prog). Returns device pointers
         (out_codes, out_vals [naggs][G], out_counts [naggs][G], n_groups)."""

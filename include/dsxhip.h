@@ -311,6 +311,10 @@ typedef struct DsxAggSpec {
   DsxInstr prog[DSX_MAX_PROG];   /* input expression, fused into the kernel */
 } DsxAggSpec;
 
+/* Toggle the per-table groupby histogram cache; the Python layer turns it
+ * off for externally-backed (e.g. torch/RCCL staging) key columns. */
+int dsx_gb_hist_cache_enable(DsxCtx* ctx, int enable);
+
 /* replaces `df.groupby(by, dropna=False).agg(...)`
  * (dask_sql/physical/rel/logical/aggregate.py:575-581) with the WHERE
  * predicate fused in (filter.py:20-45 fused into the same scan — SURVEY §3
