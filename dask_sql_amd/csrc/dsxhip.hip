@@ -3417,10 +3417,13 @@ extern "C" int dsx_jit_selftest(void) {
   A.op[1] = DSX_AGG_COUNT;
   A.never_null[0] = A.never_null[1] = 1;
   int32_t val_of[2] = {0, -1};
+  ProgArg P0{};  // predicate-free shape: the contiguous aggregate variant
   for (int tile : {0, 4096}) {
     for (int code32 : {0, 1}) {
      for (int ds : {-1, 11}) {
-      std::string src = jit_gbpart_source(C, K, P, aggs, A, val_of, 2, 1,
+      for (int pf : {0, 1}) {
+      std::string src = jit_gbpart_source(C, K, pf ? P0 : P, aggs, A,
+                                          val_of, 2, 1,
                                           ds >= 0 ? (1 << ds) : 2048,
                                           code32 != 0, tile, ds);
       if (src.empty()) {
@@ -3447,6 +3450,7 @@ extern "C" int dsx_jit_selftest(void) {
         return 3;
       }
       hiprtcDestroyProgram(&prog);
+      }
      }
     }
   }

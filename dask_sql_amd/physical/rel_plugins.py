@@ -1915,8 +1915,10 @@ class DaskSortPlugin(BaseRelPlugin):
     class_name = "Sort"
 
     def convert(self, rel, context):
+        import os as _os
         (dc,) = self.assert_inputs(rel, 1, context)
-        if isinstance(dc, DataContainer) and dc.table.num_rows >= 65536:
+        min_dev = int(_os.environ.get("DSX_SORT_MIN", 65536))
+        if isinstance(dc, DataContainer) and dc.table.num_rows >= min_dev:
             out = self._device_sort(context, dc, rel.sort().getCollation())
             if out is not None:
                 return out

@@ -1765,3 +1765,42 @@ def test_correlated_scalar_subquery_exec(ctx):
                  "WHERE u.k = t.k)").compute()
     # k=1: 1.0 < 6.0 T; k=2: 2.0 < 9.0 T; k=3: NULL comparison -> excluded
     assert sorted(out2["k"].astype(np.int64).tolist()) == [1, 2]
+
+
+def test_sort_reference_cases_device(ctx, monkeypatch):
+    """The reference's test_sort.py:16-65 ORDER BY variants (mixed
+    ASC/DESC, multi-key, strings :280-294) through the DEVICE sort
+    (DSX_SORT_MIN=1 forces it at any size), compared against pandas
+    sort_values exactly as the reference's assert_eq does."""
+    from dask_sql_amd.context import Context
+    monkeypatch.setenv("DSX_SORT_MIN", "1")
+    np.random.seed(42)
+    user_table_1 = pd.DataFrame({"user_id": [2, 1, 2, 3],
+                                 "b": [3, 3, 1, 3]})
+    df700 = pd.DataFrame(
+        {"a": [1.0] * 100 + [2.0] * 200 + [3.0] * 400,
+         "b": 10 * np.random.rand(700)})
+    c = Context()
+    c.create_table("user_table_1", user_table_1)
+    c.create_table("df", df700)
+    for sql, by, asc in [
+        ("SELECT * FROM user_table_1 ORDER BY b, user_id DESC",
+         ["b", "user_id"], [True, False]),
+        ("SELECT * FROM df ORDER BY b DESC, a DESC", ["b", "a"],
+         [False, False]),
+        ("SELECT * FROM df ORDER BY a DESC, b", ["a", "b"], [False, True]),
+        ("SELECT * FROM df ORDER BY b, a", ["b", "a"], [True, True]),
+    ]:
+        got = ctx and c.sql(sql).compute().reset_index(drop=True)
+        tbl = user_table_1 if "user_table_1" in sql else df700
+        exp = tbl.sort_values(by, ascending=asc).reset_index(drop=True)
+        for col in exp.columns:
+            assert np.allclose(got[col].to_numpy(dtype=np.float64),
+                               exp[col].to_numpy(dtype=np.float64)), (sql,
+                                                                      col)
+    # strings sort by STRING rank through the dictionary LUT
+    st = pd.DataFrame({"a": ["zzhsd", "öfjdf", "baba"]})
+    c.create_table("string_table", st)
+    got = c.sql("SELECT * FROM string_table ORDER BY a").compute()
+    exp = st.sort_values("a").reset_index(drop=True)
+    assert got["a"].tolist() == exp["a"].tolist()
