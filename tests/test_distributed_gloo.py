@@ -314,3 +314,30 @@ def test_broadcast_exchange_gloo():
 
 def test_broadcast_exchange_gloo_world3():
     _spawn(_run_broadcast_exchange, world=3)
+
+
+def test_q1_merge_partials_edges():
+    """q1_merge_partials: None entries (non-zero ranks), single partial
+    identity, and empty-group alignment."""
+    import sys
+    sys.path.insert(0, str(REPO))
+    from datagen import gen_lineitem_q1
+    from dask_sql_amd.distributed import q1_merge_partials
+    from oracle.tpch import oracle_q1
+
+    li = gen_lineitem_q1(n=50_000, seed=13)
+    full = oracle_q1(li)
+    # identity: merging one partial (+ Nones) reproduces it
+    m = q1_merge_partials([full, None, None])
+    m = m.sort_values(["l_returnflag", "l_linestatus"]).reset_index(drop=True)
+    assert m["count_order"].tolist() == full["count_order"].tolist()
+    np.testing.assert_allclose(m["avg_disc"], full["avg_disc"], rtol=1e-12)
+    # a rank whose slice lacks some groups entirely
+    p1 = oracle_q1(li.iloc[:1000].reset_index(drop=True))
+    p2 = oracle_q1(li.iloc[1000:].reset_index(drop=True))
+    m2 = q1_merge_partials([p1, p2]).sort_values(
+        ["l_returnflag", "l_linestatus"]).reset_index(drop=True)
+    assert m2["count_order"].tolist() == full["count_order"].tolist()
+    np.testing.assert_allclose(m2["sum_charge"], full["sum_charge"],
+                               rtol=1e-9)
+    np.testing.assert_allclose(m2["avg_qty"], full["avg_qty"], rtol=1e-9)
