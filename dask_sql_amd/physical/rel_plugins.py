@@ -401,6 +401,19 @@ class DaskJoinPlugin(BaseRelPlugin):
             lhs_on, rhs_on, residual = self._split_join_condition(
                 condition, n_lhs_cols)
 
+        if (join_type == "leftanti" and getattr(join, "null_aware", False)
+                and rhs_on):
+            # NOT IN three-valued logic: any NULL in the subquery output
+            # makes `x NOT IN (...)` non-TRUE for every row → empty result
+            rcols0 = dc_rhs.backend_cols()
+            for ri in rhs_on:
+                col = rcols0[ri]
+                if col.validity:
+                    _, _, nn = _minmax_cached(runtime, col)
+                    if nn < dc_rhs.table.num_rows:
+                        return self._empty_output(rel, runtime, dc_lhs,
+                                                  cc_lhs)
+
         # materialize only columns the plan consumes (output_indices from the
         # pruning pass) plus residual-referenced temporaries
         force_l = join_type in ("outer", "right")
@@ -520,6 +533,24 @@ class DaskJoinPlugin(BaseRelPlugin):
         return DataContainer(DeviceTable(out_cols, num_rows=n_out), cc)
 
     # -- helpers ------------------------------------------------------------
+    def _empty_output(self, rel, runtime, dc_lhs, cc_lhs):
+        row_type = rel.getRowType()
+        field_names = [str(f) for f in row_type.getFieldNames()]
+        empty = runtime.empty_column(0, rt.I32)
+        out_cols = {}
+        mapping = {}
+        lcols = dc_lhs.backend_cols()
+        for i, name in enumerate(field_names):
+            src = lcols[i] if i < len(lcols) else lcols[0]
+            g = runtime.gather(src, empty.data, 0)
+            if getattr(src, "dictionary", None) is not None:
+                g.dictionary = src.dictionary
+            backend = f"j{i}__{name}"
+            out_cols[backend] = g
+            mapping[name] = backend
+        cc = ColumnContainer(field_names, mapping)
+        return DataContainer(DeviceTable(out_cols, num_rows=0), cc)
+
     def _anti_residual(self, runtime, dc_lhs, dc_rhs, lhs_on, rhs_on,
                        residual, cc_lhs, cc_rhs, n_lhs_cols):
         """LEFT ANTI with a residual condition: inner pairs, residual filter

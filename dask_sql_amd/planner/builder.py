@@ -211,7 +211,8 @@ class Builder:
             raise ValueError("aggregate in non-aggregate position")
         raise ValueError(f"cannot resolve {ast!r}")
 
-    def _semi_anti_join(self, plan, subplan, outer_asts, negated):
+    def _semi_anti_join(self, plan, subplan, outer_asts, negated,
+                        null_aware=False):
         """SEMI/ANTI join `plan` against the DISTINCT subplan output on
         positional key equalities (outer_asts resolve against the combined
         row; sub keys are the subplan's columns in order)."""
@@ -234,8 +235,14 @@ class Builder:
             cond = eq if cond is None else Call("AND", [cond, eq],
                                                 SqlType("BOOLEAN"))
         jt = "LEFTANTI" if negated else "LEFTSEMI"
+        node = JoinNode(jt, cond)
+        # SQL NOT IN: a NULL anywhere in the subquery output makes the
+        # predicate non-TRUE for EVERY row (three-valued logic) — the
+        # plugin returns an empty result when the build side holds NULL
+        # keys. Closes the r1 documented divergence.
+        node.null_aware = null_aware
         return LogicalPlan("Join", [plan, subplan], RelDataType(lhs_fields),
-                           JoinNode(jt, cond))
+                           node)
 
     def _decorrelate_exists(self, sub, lead_items=None, tail_items=None,
                             group=False, key_prefix=""):
@@ -677,7 +684,8 @@ class Builder:
                 subplan = self.build_stmt(substmt)
                 outer_keys = []
             plan = self._semi_anti_join(plan, subplan,
-                                        [e_ast] + outer_keys, negated)
+                                        [e_ast] + outer_keys, negated,
+                                        null_aware=negated)
 
         # leftover WHERE conjuncts → Filter (incl. scalar TRUE/FALSE)
         leftovers = [cj for i, cj in enumerate(where_conjuncts) if not used[i]

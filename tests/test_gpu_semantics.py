@@ -1804,3 +1804,24 @@ def test_sort_reference_cases_device(ctx, monkeypatch):
     got = c.sql("SELECT * FROM string_table ORDER BY a").compute()
     exp = st.sort_values("a").reset_index(drop=True)
     assert got["a"].tolist() == exp["a"].tolist()
+
+
+def test_not_in_with_nulls(ctx):
+    """SQL three-valued NOT IN: a NULL in the subquery output makes the
+    predicate non-TRUE for every row → empty result (r1 documented this
+    as a divergence; r2 implements the null-aware anti join)."""
+    from dask_sql_amd.context import Context
+    c = Context()
+    c.create_table("big", pd.DataFrame({"k": np.arange(10, dtype=np.int64)}))
+    c.create_table("small", pd.DataFrame(
+        {"id": pd.array([1, 2, None], dtype="Int64")}))
+    out = c.sql("SELECT k FROM big WHERE k NOT IN (SELECT id FROM small)"
+                ).compute()
+    assert len(out) == 0
+    # without NULLs the anti join behaves as before
+    c.create_table("small2", pd.DataFrame(
+        {"id": pd.array([1, 2], dtype="Int64")}))
+    out2 = c.sql("SELECT k FROM big WHERE k NOT IN (SELECT id FROM small2)"
+                 ).compute()
+    assert sorted(out2["k"].astype(np.int64).tolist()) == \
+        [0, 3, 4, 5, 6, 7, 8, 9]
