@@ -536,12 +536,17 @@ class DaskJoinPlugin(BaseRelPlugin):
     def _empty_output(self, rel, runtime, dc_lhs, cc_lhs):
         row_type = rel.getRowType()
         field_names = [str(f) for f in row_type.getFieldNames()]
+        join = rel.join()
+        out_idx = join.output_indices \
+            if getattr(join, "output_indices", None) is not None \
+            else list(range(len(field_names)))
         empty = runtime.empty_column(0, rt.I32)
         out_cols = {}
         mapping = {}
         lcols = dc_lhs.backend_cols()
-        for i, name in enumerate(field_names):
-            src = lcols[i] if i < len(lcols) else lcols[0]
+        for name, oi in zip(field_names, out_idx):
+            src = lcols[oi] if oi < len(lcols) else lcols[0]
+            i = oi
             g = runtime.gather(src, empty.data, 0)
             if getattr(src, "dictionary", None) is not None:
                 g.dictionary = src.dictionary
