@@ -436,3 +436,15 @@ def test_q1_real_text_parity(ctx):
     for col in ["sum_qty", "sum_base_price", "sum_disc_price", "sum_charge",
                 "avg_qty", "avg_price", "avg_disc"]:
         assert np.allclose(out[col], exp[col], rtol=REL_TOL), col
+
+
+def test_group_by_case_reference(c):
+    """reference test_groupby.py:155-171: expression GROUP BY key
+    (user_id + 1) with SUM(CASE WHEN b = 3 THEN 1 END) — expected frame
+    A=[2,3,4], S=[1,1,1]."""
+    out = c.sql('SELECT user_id + 1 AS "A", '
+                'SUM(CASE WHEN b = 3 THEN 1 END) AS "S" '
+                "FROM user_table_1 GROUP BY user_id + 1").compute()
+    out = out.sort_values("A").reset_index(drop=True)
+    assert out["A"].astype(np.int64).tolist() == [2, 3, 4]
+    assert out["S"].astype(np.float64).tolist() == [1.0, 1.0, 1.0]
