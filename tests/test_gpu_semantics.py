@@ -1825,3 +1825,38 @@ def test_not_in_with_nulls(ctx):
                  ).compute()
     assert sorted(out2["k"].astype(np.int64).tolist()) == \
         [0, 3, 4, 5, 6, 7, 8, 9]
+
+
+def test_fallback_knobs_subprocess(ctx):
+    """The interpreter/static fallbacks (DSX_DISABLE_JIT — static radix
+    and partition-groupby kernels, VM predicates) must produce the same
+    results as the JIT default. Env is latched at first library use, so
+    each knob runs in a subprocess (tests/knob_check.py)."""
+    import json
+    import os
+    import subprocess
+    import sys
+    from tests.conftest import REPO
+
+    def run(env_extra):
+        env = dict(os.environ)
+        env.update(env_extra)
+        r = subprocess.run(
+            [sys.executable, str(REPO / "tests" / "knob_check.py"),
+             str(REPO)],
+            capture_output=True, text=True, timeout=180, env=env)
+        assert r.returncode == 0, r.stderr[-2000:]
+        return json.loads(r.stdout.strip().splitlines()[-1])
+
+    base = run({"DSX_RADIX_MIN_BUILD": "1000",
+                "DSX_RADIX_MIN_PROBE": "1000"})
+    nojit = run({"DSX_DISABLE_JIT": "1", "DSX_RADIX_MIN_BUILD": "1000",
+                 "DSX_RADIX_MIN_PROBE": "1000"})
+    nopart = run({"DSX_DISABLE_PART": "1"})
+    for k in base:
+        if k == "g_sum":
+            assert abs(base[k] - nojit[k]) < 1e-6 * abs(base[k])
+            assert abs(base[k] - nopart[k]) < 1e-6 * abs(base[k])
+        else:
+            assert base[k] == nojit[k], (k, base[k], nojit[k])
+            assert base[k] == nopart[k], (k, base[k], nopart[k])
