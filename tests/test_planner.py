@@ -585,3 +585,29 @@ def test_correlated_scalar_subquery_plan():
         "SELECT * FROM t WHERE t.x < (SELECT AVG(u.y) FROM u "
         "WHERE u.k = t.k)")
     assert rel2.getRowType().getFieldNames() == ["k", "x"]
+
+
+def test_from_arrow_multichunk(tmp_path):
+    """_from_arrow with multi-chunk columns (row-group-split parquet) and
+    large_string dictionary encoding."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    import dask_sql_amd.runtime as rt
+    from dask_sql_amd.context import _from_arrow
+
+    n = 5000
+    t = pa.table({
+        "i": pa.array(range(n), type=pa.int64()),
+        "s": pa.array((["A", "B"][x % 2] for x in range(n)),
+                      type=pa.large_string()),
+    })
+    f = tmp_path / "m.parquet"
+    pq.write_table(t, f, row_group_size=700)  # 8 row groups → chunks
+    rd = pq.read_table(f)
+    assert rd.column("i").num_chunks > 1
+    cols = _from_arrow(rd)
+    assert cols["i"].arr[-1] == n - 1 and cols["i"].dtype == rt.I64
+    assert cols["s"].dictionary is not None
+    dec = [cols["s"].dictionary[c] for c in cols["s"].arr[:4]]
+    assert dec == ["A", "B", "A", "B"]
