@@ -311,6 +311,15 @@ PLAN_BATTERY = [
     "SELECT 1 + 1 AS two",
     "SELECT a FROM t ORDER BY a DESC NULLS LAST LIMIT 10 OFFSET 5",
     "SELECT a, SUM(b) AS s FROM t GROUP BY a ORDER BY SUM(b) DESC LIMIT 5",
+    # round-2 constructs
+    "SELECT a, (SELECT MAX(u.d) FROM u WHERE u.c = t.a) AS m FROM t",
+    "SELECT * FROM t WHERE b < (SELECT AVG(u.d) FROM u WHERE u.c = t.a)",
+    "SELECT a FROM t WHERE a NOT IN (SELECT c FROM u)",
+    "SELECT a + 1 AS g, SUM(CASE WHEN b > 1 THEN 1 END) FROM t "
+    "GROUP BY a + 1",
+    "SELECT a FROM t ORDER BY a DESC, b",
+    "SELECT RANK() OVER (PARTITION BY a ORDER BY b) AS r, "
+    "DENSE_RANK() OVER (ORDER BY b) AS d FROM t",
 ]
 
 
