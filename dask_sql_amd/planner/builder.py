@@ -750,7 +750,34 @@ class Builder:
             plan = LogicalPlan("Filter", [plan], plan.getRowType(),
                                FilterNode(cond))
 
-        # 4. final projection
+        # 4. final projection (+ HIDDEN sort columns: ORDER BY may name
+        # source columns that are not in the SELECT list — the reference
+        # sorts before the final projection drops them (DataFusion plans
+        # Sort over a projection containing the keys). We append them as
+        # hidden output columns and strip them after Sort/Limit.)
+        hidden = []
+        if stmt.order_by and not has_agg and not stmt.distinct:
+            def _matches_item(e):
+                if e[0] == "lit" and isinstance(e[1], int):
+                    return True
+                for it, alias in items:
+                    if it == e:
+                        return True
+                    if e[0] == "col":
+                        name = e[2].lower()
+                        if alias is not None and alias.lower() == name:
+                            return True
+                        if alias is None and it[0] == "col"                                 and it[2].lower() == name:
+                            return True
+                return False
+
+            for e, _asc, _nf in stmt.order_by:
+                if isinstance(e, tuple) and e[0] == "col"                         and not _matches_item(e)                         and all(e != h for h in hidden):
+                    hidden.append(e)
+        n_vis = len(items)
+        if hidden:
+            items = items + [(h, f"__sort_h{i}")
+                             for i, h in enumerate(hidden)]
         plan = self._build_projection(plan, items)
 
         # 5. DISTINCT
@@ -761,17 +788,27 @@ class Builder:
                                  distinct_columns=[f.getName() for f in fields])
             plan = LogicalPlan("Distinct", [plan], plan.getRowType(), node)
 
-        # 6. ORDER BY / LIMIT
+        # 6. ORDER BY / LIMIT (then strip any hidden sort columns —
+        # Sort→Limit order keeps the device top-k fusion applicable)
         if stmt.order_by:
             keys = []
             for e, asc, nf in stmt.order_by:
-                idx = self._find_output(e, stmt, plan)
+                if e in hidden:
+                    idx = n_vis + hidden.index(e)
+                else:
+                    idx = self._find_output(e, stmt, plan)
                 keys.append((idx, asc, nf))
             plan = LogicalPlan("Sort", [plan], plan.getRowType(),
                                SortNode(keys))
         if stmt.limit is not None or stmt.offset:
             plan = LogicalPlan("Limit", [plan], plan.getRowType(),
                                LimitNode(stmt.limit, stmt.offset))
+        if hidden:
+            fields = plan.getRowType().getFieldList()[:n_vis]
+            named = [(InputRef(i, f.getType()), f.getName())
+                     for i, f in enumerate(fields)]
+            plan = LogicalPlan("Projection", [plan], RelDataType(fields),
+                               ProjectionNode(named))
         return plan
 
     # ------------------------------------------------------------ union

@@ -1860,3 +1860,21 @@ def test_fallback_knobs_subprocess(ctx):
         else:
             assert base[k] == nojit[k], (k, base[k], nojit[k])
             assert base[k] == nopart[k], (k, base[k], nopart[k])
+
+
+def test_order_by_hidden_column_exec(ctx):
+    """ORDER BY a non-selected column (reference supports this; sorts then
+    drops the key)."""
+    from dask_sql_amd.context import Context
+    rng = np.random.default_rng(81)
+    n = 10_000
+    df = pd.DataFrame({"a": rng.integers(0, 1000, n),
+                       "b": rng.permutation(n)})
+    c = Context()
+    c.create_table("t", df)
+    out = c.sql("SELECT a FROM t ORDER BY b").compute()
+    exp = df.sort_values("b")["a"].reset_index(drop=True)
+    assert out["a"].astype(np.int64).tolist() == exp.tolist()
+    out2 = c.sql("SELECT a FROM t ORDER BY b DESC LIMIT 5").compute()
+    exp2 = df.sort_values("b", ascending=False)["a"].head(5)
+    assert out2["a"].astype(np.int64).tolist() == exp2.tolist()

@@ -620,3 +620,23 @@ def test_from_arrow_multichunk(tmp_path):
     assert cols["s"].dictionary is not None
     dec = [cols["s"].dictionary[c] for c in cols["s"].arr[:4]]
     assert dec == ["A", "B", "A", "B"]
+
+
+def test_order_by_hidden_column_plan():
+    """ORDER BY on a column not in the SELECT list: hidden sort column
+    appended, sorted, then stripped (DataFusion plans Sort below the final
+    projection)."""
+    c = Context()
+    c.create_table("t", pd.DataFrame({"a": [1, 2], "b": [0.5, 1.5]}))
+    rel = c._get_ral("SELECT a FROM t ORDER BY b DESC")
+    assert rel.getRowType().getFieldNames() == ["a"]
+    assert rel.get_current_node_type() == "Projection"
+    srt = rel.get_inputs()[0]
+    assert srt.get_current_node_type() == "Sort"
+    assert srt.getRowType().getFieldNames() == ["a", "__sort_h0"]
+    (idx, asc, _nf), = srt.sort().getCollation()
+    assert idx == 1 and asc is False
+    # with LIMIT the fusion shape Sort→Limit is preserved under the strip
+    rel2 = c._get_ral("SELECT a FROM t ORDER BY b LIMIT 1")
+    assert rel2.get_current_node_type() == "Projection"
+    assert rel2.get_inputs()[0].get_current_node_type() == "Limit"
