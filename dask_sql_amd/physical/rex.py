@@ -204,22 +204,51 @@ def dict_int_fn(expr, dicts):
             return None
         i, f = sub
         return i, (lambda s, f=f: len(f(s)))
+    if op == "POSITION" and len(expr.getOperands()) in (2, 3):
+        # POSITION(needle IN hay [FROM start]) — 1-based, 0 when absent
+        # (reference rex/core/call.py PositionOperation)
+        ops_ = expr.getOperands()
+        sub = dict_string_fn(ops_[0], dicts)
+        if sub is None or not (isinstance(ops_[1], Literal)
+                               and isinstance(ops_[1].getValue(), str)):
+            return None
+        start = _lit_int(ops_[2]) if len(ops_) > 2 else 1
+        if start is None:
+            return None
+        i, f = sub
+        needle = ops_[1].getValue()
+        return i, (lambda s, f=f, n=needle, st=start:
+                   f(s).find(n, max(st - 1, 0)) + 1)
     return None
 
 
-def _like_regex(pattern: str):
-    """SQL LIKE pattern → compiled regex: % = any run, _ = any single char,
-    everything else literal (reference rex/core/call.py LIKE lowering)."""
+def _like_regex(pattern: str, escape=None, mode="LIKE"):
+    """SQL LIKE/ILIKE/SIMILAR pattern → compiled regex. LIKE: % = any run,
+    _ = any single char, everything else literal; an ESCAPE char makes the
+    following char literal. ILIKE adds IGNORECASE. SIMILAR TO additionally
+    passes the SQL:1999 regex metacharacters ( ) [ ] | * + ? { } through
+    (reference rex/core/call.py LIKE/SIMILAR lowering)."""
     import re
+    similar_meta = set("()[]|*+?{}-^,")
     out = []
-    for ch in pattern:
+    i = 0
+    while i < len(pattern):
+        ch = pattern[i]
+        if escape and ch == escape and i + 1 < len(pattern):
+            out.append(re.escape(pattern[i + 1]))
+            i += 2
+            continue
         if ch == "%":
             out.append(".*")
         elif ch == "_":
             out.append(".")
+        elif mode == "SIMILAR" and ch in similar_meta:
+            out.append(ch)
         else:
             out.append(re.escape(ch))
-    return re.compile("".join(out), re.DOTALL)
+        i += 1
+    flags = re.DOTALL | (re.IGNORECASE if mode == "ILIKE" else 0)
+    return re.compile("".join(out), flags)
 
 
 class RexCompiler:
@@ -336,8 +365,8 @@ class RexCompiler:
         if op == "CASE":
             # operands: cond1, val1, cond2, val2, ..., else
             return self._compile_case(ops)
-        if op == "LIKE":
-            return self._compile_like(ops)
+        if op in ("LIKE", "ILIKE", "SIMILAR"):
+            return self._compile_like(ops, op)
         if op == "COALESCE":
             # right-fold of SELECT(arg IS NOT NULL, arg, rest)
             # (rex/core/call.py CoalesceOperation)
@@ -448,15 +477,121 @@ class RexCompiler:
                 self._emit(OP_DIV_I64)
             self._emit(_dx[op])
             return KI
+        if op == "EXTRACT_DATE" or op.startswith("FLOOR_TO_") \
+                or op.startswith("CEIL_TO_"):
+            return self._compile_dt_trunc(op, ops)
         raise RexCompileError(f"operator {op} not supported on GPU path")
 
-    def _compile_like(self, ops):
-        """LIKE on a dict-encoded column: the SQL pattern (%/_ wildcards,
-        reference rex/core/call.py SargPythonImplementation / re-based LIKE
-        lowering) is matched against the (small, host-resident) dictionary
-        once at compile time; the kernel-side predicate is an OR-chain of
-        integer code equalities, so NULL → NULL falls out of EQ validity."""
-        col, pat = ops
+    _DAY_NS = 86_400_000_000_000
+
+    def _emit_days(self, ast, is_ts):
+        """Push the operand's floor-div days-since-epoch (exact for
+        pre-1970: x - floormod(x, day) is a day multiple, so trunc div is
+        floor div)."""
+        k = self.compile(ast)
+        if k != KI:
+            raise RexCompileError("datetime op needs DATE/TIMESTAMP")
+        if is_ts:
+            self.compile(ast)
+            self._emit(OP_LIT_I64, 0, self._DAY_NS)
+            self._emit(OP_FLOORMOD_I64)
+            self._emit(OP_SUB_I64)
+            self._emit(OP_LIT_I64, 0, self._DAY_NS)
+            self._emit(OP_DIV_I64)
+
+    def _compile_dt_trunc(self, op, ops):
+        """FLOOR/CEIL(x TO unit) and EXTRACT(DATE FROM x) on the VM's
+        existing integer ops (reference rex/core/call.py CeilFloorDatetime
+        / ExtractOperation date truncation). Month/year starts come from
+        the epoch-day civil-calendar ops (OP_DAY / OP_YEAR) plus the
+        Gregorian leap-count identity; all intermediate divisions see
+        positive years, so trunc division is floor division."""
+        t0 = getattr(ops[0], "getType", lambda: None)()
+        is_ts = t0 is not None and t0.getSqlType() == "TIMESTAMP"
+        x = ops[0]
+        sub_ns = {"HOUR": 3_600_000_000_000, "MINUTE": 60_000_000_000,
+                  "SECOND": 1_000_000_000}
+        if op == "EXTRACT_DATE":
+            self._emit_days(x, is_ts)
+            return KI
+        kind, unit = op.split("_TO_")
+        if unit == "DAY" and not is_ts:
+            k = self.compile(x)  # a DATE is already day-aligned
+            if k != KI:
+                raise RexCompileError("FLOOR TO needs DATE/TIMESTAMP")
+            return KI
+        if unit in sub_ns or (unit == "DAY" and is_ts):
+            if not is_ts:
+                raise RexCompileError(f"{op} needs a TIMESTAMP operand")
+            u = self._DAY_NS if unit == "DAY" else sub_ns[unit]
+            if kind == "FLOOR":
+                self.compile(x)
+                self.compile(x)
+                self._emit(OP_LIT_I64, 0, u)
+                self._emit(OP_FLOORMOD_I64)
+                self._emit(OP_SUB_I64)
+            else:  # CEIL: x + floormod(-x, u)
+                self.compile(x)
+                self._emit(OP_LIT_I64, 0, 0)
+                self.compile(x)
+                self._emit(OP_SUB_I64)
+                self._emit(OP_LIT_I64, 0, u)
+                self._emit(OP_FLOORMOD_I64)
+                self._emit(OP_ADD_I64)
+            return KI
+        if kind == "CEIL":
+            raise RexCompileError(f"CEIL TO {unit} not supported on GPU "
+                                  "path (calendar ceil)")
+        if unit == "MONTH":
+            # month_start_days = days - (dayofmonth(days) - 1)
+            self._emit_days(x, is_ts)
+            self._emit_days(x, is_ts)
+            self._emit(OP_DAY)
+            self._emit(OP_LIT_I64, 0, 1)
+            self._emit(OP_SUB_I64)
+            self._emit(OP_SUB_I64)
+        elif unit == "YEAR":
+            # jan1_days = 365*y + (y-1)/4 - (y-1)/100 + (y-1)/400 - 719527
+            def emit_y():
+                self._emit_days(x, is_ts)
+                self._emit(OP_YEAR)
+
+            def emit_leap_term(div):
+                emit_y()
+                self._emit(OP_LIT_I64, 0, 1)
+                self._emit(OP_SUB_I64)
+                self._emit(OP_LIT_I64, 0, div)
+                self._emit(OP_DIV_I64)
+
+            emit_y()
+            self._emit(OP_LIT_I64, 0, 365)
+            self._emit(OP_MUL_I64)
+            emit_leap_term(4)
+            self._emit(OP_ADD_I64)
+            emit_leap_term(100)
+            self._emit(OP_SUB_I64)
+            emit_leap_term(400)
+            self._emit(OP_ADD_I64)
+            self._emit(OP_LIT_I64, 0, 719_527)
+            self._emit(OP_SUB_I64)
+        else:
+            raise RexCompileError(f"FLOOR TO {unit} not supported")
+        if is_ts:
+            self._emit(OP_LIT_I64, 0, self._DAY_NS)
+            self._emit(OP_MUL_I64)
+        return KI
+
+    def _compile_like(self, ops, mode="LIKE"):
+        """LIKE/ILIKE/SIMILAR on a dict-encoded column: the SQL pattern
+        (%/_ wildcards, reference rex/core/call.py SargPythonImplementation
+        / re-based LIKE lowering) is matched against the (small,
+        host-resident) dictionary once at compile time; the kernel-side
+        predicate is an OR-chain of integer code equalities, so NULL → NULL
+        falls out of EQ validity. A third operand is the ESCAPE char."""
+        col, pat = ops[0], ops[1]
+        esc = None
+        if len(ops) > 2 and isinstance(ops[2], Literal):
+            esc = ops[2].getValue()
         fn = dict_string_fn(col, self.dicts)
         if fn is None or not (isinstance(pat, Literal)
                               and isinstance(pat.getValue(), str)):
@@ -464,7 +599,7 @@ class RexCompiler:
                 "LIKE needs <dict string expr> LIKE '<pattern>'")
         ci, f = fn
         d = self.dicts[ci]
-        rx = _like_regex(pat.getValue())
+        rx = _like_regex(pat.getValue(), esc, mode)
         matched = [i for i, s in enumerate(d)
                    if s is not None and rx.fullmatch(f(s))]
         return self._emit_code_in(ci, matched)
@@ -573,7 +708,8 @@ class RexCompiler:
         if isinstance(expr, Call):
             op = expr.getOperatorName()
             if op in _CMP or op in ("AND", "OR", "NOT", "IS NULL",
-                                    "IS NOT NULL", "LIKE"):
+                                    "IS NOT NULL", "LIKE", "ILIKE",
+                                    "SIMILAR"):
                 return KB
             if op in _ARITH:
                 ka = self._peek_kind(expr.getOperands()[0])
@@ -586,7 +722,9 @@ class RexCompiler:
             if op == "CASE":
                 return self._peek_kind(expr.getOperands()[1])
             if op in ("MOD", "EXTRACT_YEAR", "EXTRACT_MONTH", "EXTRACT_DAY",
-                      "YEAR", "MONTH", "DAY", "DAYOFMONTH"):
+                      "YEAR", "MONTH", "DAY", "DAYOFMONTH", "EXTRACT_DATE") \
+                    or op.startswith("FLOOR_TO_") \
+                    or op.startswith("CEIL_TO_"):
                 return KI
             if op == "COALESCE":
                 kids = [self._peek_kind(o) for o in expr.getOperands()]

@@ -49,6 +49,7 @@ class UnionStmt:
     order_by: list = field(default_factory=list)
     limit: int | None = None
     offset: int = 0
+    ops: list | None = None  # per step: UNION | INTERSECT | EXCEPT
 
 
 KEYWORDS = {
@@ -65,7 +66,7 @@ WINDOW_FUNCS = {"ROW_NUMBER", "RANK", "DENSE_RANK", "SUM", "COUNT", "AVG",
 
 _TOKEN_RE = re.compile(
     r"""
-    (?P<ws>\s+)
+    (?P<ws>\s+|--[^\n]*|/\*[\s\S]*?\*/)
   | (?P<num>\d+\.\d*(?:[eE][+-]?\d+)?|\.\d+(?:[eE][+-]?\d+)?|\d+(?:[eE][+-]?\d+)?)
   | (?P<str>'(?:[^']|'')*')
   | (?P<qid>"[^"]+")
@@ -103,6 +104,10 @@ def tokenize(sql: str):
     return toks
 
 
+# words that terminate an implicit-alias position (reserved in standard SQL
+# but lexed as identifiers here)
+_NON_ALIAS = {"INTERSECT", "EXCEPT"}
+
 AGG_FUNCS = {"SUM", "COUNT", "AVG", "MIN", "MAX", "ANY_VALUE", "STDDEV",
              "STDDEV_POP", "STDDEV_SAMP", "VAR_SAMP", "VAR_POP", "VARIANCE",
              "SINGLE_VALUE", "EVERY", "BOOL_AND", "BOOL_OR"}
@@ -112,6 +117,7 @@ class Parser:
     def __init__(self, sql: str):
         self.toks = tokenize(sql)
         self.i = 0
+        self.ctes = {}  # lowercase name -> SelectStmt/UnionStmt (WITH ...)
 
     # -- token helpers ------------------------------------------------------
     def peek(self):
@@ -134,6 +140,19 @@ class Parser:
         if t != ("kw", kw):
             raise ValueError(f"expected {kw}, got {t}")
 
+    def _accept_word(self, w):
+        """Accept a non-reserved word (ILIKE/SIMILAR/TO/ESCAPE ...) given as
+        an identifier or keyword token, case-insensitively."""
+        t = self.peek()
+        if t[0] in ("id", "kw") and str(t[1]).upper() == w:
+            self.next()
+            return True
+        return False
+
+    def _expect_word(self, w):
+        if not self._accept_word(w):
+            raise ValueError(f"expected {w}, got {self.peek()}")
+
     def accept_op(self, *ops):
         t = self.peek()
         if t[0] == "op" and t[1] in ops:
@@ -148,24 +167,65 @@ class Parser:
 
     # -- entry --------------------------------------------------------------
     def parse(self):
-        stmt = self.select_stmt()
-        branches = [stmt]
-        alls = []
-        while self.accept_kw("UNION"):
-            alls.append(bool(self.accept_kw("ALL")))
-            branches.append(self.select_stmt())
+        # WITH name AS (select) [, name2 AS (...)] <stmt> — CTEs register as
+        # named derived tables; every FROM reference inlines a deep copy of
+        # the definition (the reference hands CTEs to DataFusion, which
+        # inlines them the same way for non-recursive WITH).
+        if self._accept_word("WITH"):
+            if self._accept_word("RECURSIVE"):
+                raise ValueError("WITH RECURSIVE not supported")
+            while True:
+                cname = self._name()
+                self.expect_kw("AS")
+                self.expect_op("(")
+                sub = self._set_tail(self.select_stmt())
+                self.expect_op(")")
+                self.ctes[cname.lower()] = sub
+                if not self.accept_op(","):
+                    break
+        stmt = self._set_tail(self.select_stmt())
         if self.peek()[0] != "eof":
             raise ValueError(f"trailing tokens: {self.peek()}")
+        if not isinstance(stmt, UnionStmt):
+            return stmt
+        # trailing ORDER BY / LIMIT bind to the whole set expression, not
+        # the last branch (standard SQL)
+        last = stmt.branches[-1]
+        while isinstance(last, UnionStmt):
+            last = last.branches[-1]
+        stmt.order_by, stmt.limit, stmt.offset = (last.order_by, last.limit,
+                                                  last.offset)
+        last.order_by, last.limit, last.offset = [], None, 0
+        return stmt
+
+    def _set_tail(self, first):
+        """UNION [ALL] / INTERSECT / EXCEPT chain after a select;
+        INTERSECT binds tighter than UNION/EXCEPT (standard SQL)."""
+        def chain(left):
+            while self._accept_word("INTERSECT"):
+                if self.accept_kw("ALL"):
+                    raise ValueError("INTERSECT ALL not supported")
+                left = UnionStmt([left, self.select_stmt()], alls=[False],
+                                 ops=["INTERSECT"])
+            return left
+
+        stmt = chain(first)
+        branches, alls, ops = [stmt], [], []
+        while True:
+            if self.accept_kw("UNION"):
+                op, allf = "UNION", bool(self.accept_kw("ALL"))
+            elif self._accept_word("EXCEPT"):
+                if self.accept_kw("ALL"):
+                    raise ValueError("EXCEPT ALL not supported")
+                op, allf = "EXCEPT", False
+            else:
+                break
+            branches.append(chain(self.select_stmt()))
+            alls.append(allf)
+            ops.append(op)
         if len(branches) == 1:
             return stmt
-        # trailing ORDER BY / LIMIT bind to the whole union, not the last
-        # branch (standard SQL)
-        last = branches[-1]
-        u = UnionStmt(branches=branches, alls=alls,
-                      order_by=last.order_by, limit=last.limit,
-                      offset=last.offset)
-        last.order_by, last.limit, last.offset = [], None, 0
-        return u
+        return UnionStmt(branches, alls, ops=ops)
 
     def select_stmt(self) -> SelectStmt:
         self.expect_kw("SELECT")
@@ -180,8 +240,14 @@ class Parser:
                 e = self.expr()
                 alias = None
                 if self.accept_kw("AS"):
-                    alias = self._name()
-                elif self.peek()[0] == "id":
+                    # after an explicit AS any word may serve as the alias,
+                    # including reserved ones (`... AS date`)
+                    t = self.next()
+                    if t[0] not in ("id", "kw"):
+                        raise ValueError(f"expected alias after AS, got {t}")
+                    alias = t[1] if t[0] == "id" else t[1].lower()
+                elif self.peek()[0] == "id" \
+                        and self.peek()[1].upper() not in _NON_ALIAS:
                     alias = self._name()
                 s.items.append((e, alias))
             if not self.accept_op(","):
@@ -267,28 +333,29 @@ class Parser:
         if self.peek() == ("op", "("):
             # derived table: FROM (SELECT ...) [AS] alias
             self.next()
-            sub = self.select_stmt()
-            branches, alls = [sub], []
-            while self.accept_kw("UNION"):
-                alls.append(bool(self.accept_kw("ALL")))
-                branches.append(self.select_stmt())
-            if len(branches) > 1:
-                sub = UnionStmt(branches=branches, alls=alls)
+            sub = self._set_tail(self.select_stmt())
             self.expect_op(")")
             alias = None
             if self.accept_kw("AS"):
                 alias = self._name()
-            elif self.peek()[0] == "id":
+            elif self.peek()[0] == "id" \
+                    and self.peek()[1].upper() not in _NON_ALIAS:
                 alias = self._name()
             if alias is None:
-                raise ValueError("derived table needs an alias")
+                self._dt_n = getattr(self, "_dt_n", 0) + 1
+                alias = f"_dt{self._dt_n}"
             return TableRef(name=None, alias=alias, subquery=sub)
         name = self._name()
         alias = None
         if self.accept_kw("AS"):
             alias = self._name()
-        elif self.peek()[0] == "id":
+        elif self.peek()[0] == "id" \
+                and self.peek()[1].upper() not in _NON_ALIAS:
             alias = self._name()
+        if name.lower() in self.ctes:
+            import copy
+            return TableRef(name=None, alias=alias or name,
+                            subquery=copy.deepcopy(self.ctes[name.lower()]))
         return TableRef(name, alias)
 
     def _name(self) -> str:
@@ -328,6 +395,20 @@ class Parser:
                 continue
             if self.accept_kw("IS"):
                 neg = self.accept_kw("NOT") is not None
+                if self._accept_word("UNKNOWN"):
+                    # IS [NOT] UNKNOWN on a boolean = IS [NOT] NULL
+                    e = ("call", "IS NOT NULL" if neg else "IS NULL", [e])
+                    continue
+                tv = self.accept_kw("TRUE", "FALSE")
+                if tv is not None:
+                    # x IS [NOT] TRUE/FALSE: NULL counts as "not true" and
+                    # "not false" (three-valued IS, reference
+                    # rex/core/call.py IsTrue/IsFalse lowering)
+                    want = ("call", "AND", [("call", "IS NOT NULL", [e]),
+                                            e if tv == "TRUE"
+                                            else ("call", "NOT", [e])])
+                    e = ("call", "NOT", [want]) if neg else want
+                    continue
                 self.expect_kw("NULL")
                 e = ("call", "IS NOT NULL" if neg else "IS NULL", [e])
                 continue
@@ -339,16 +420,28 @@ class Parser:
                      [("call", ">=", [e, lo]), ("call", "<=", [e, hi])])
                 continue
             if self.accept_kw("LIKE"):
-                pat = self.add_expr()
-                e = ("call", "LIKE", [e, pat])
+                e = self._like_tail(e, "LIKE")
+                continue
+            if self._accept_word("ILIKE"):
+                e = self._like_tail(e, "ILIKE")
+                continue
+            if self._accept_word("SIMILAR"):
+                self._expect_word("TO")
+                e = self._like_tail(e, "SIMILAR")
                 continue
             if self.accept_kw("NOT"):
                 if self.accept_kw("IN"):
                     e = ("call", "NOT", [self._in_list(e)])
                     continue
                 if self.accept_kw("LIKE"):
-                    pat = self.add_expr()
-                    e = ("call", "NOT", [("call", "LIKE", [e, pat])])
+                    e = ("call", "NOT", [self._like_tail(e, "LIKE")])
+                    continue
+                if self._accept_word("ILIKE"):
+                    e = ("call", "NOT", [self._like_tail(e, "ILIKE")])
+                    continue
+                if self._accept_word("SIMILAR"):
+                    self._expect_word("TO")
+                    e = ("call", "NOT", [self._like_tail(e, "SIMILAR")])
                     continue
                 if self.accept_kw("BETWEEN"):
                     lo = self.add_expr()
@@ -358,12 +451,26 @@ class Parser:
                          [("call", "AND",
                            [("call", ">=", [e, lo]), ("call", "<=", [e, hi])])])
                     continue
-                raise ValueError("expected IN or BETWEEN after NOT")
+                raise ValueError(
+                    "expected IN/BETWEEN/LIKE/ILIKE/SIMILAR after NOT")
             if self.accept_kw("IN"):
                 e = self._in_list(e)
                 continue
             break
         return e
+
+    def _like_tail(self, e, op):
+        """LIKE/ILIKE/SIMILAR TO pattern [ESCAPE '<c>'] — the escape char
+        rides as a third literal operand (reference rex/core/call.py LIKE
+        lowering takes an escape argument)."""
+        pat = self.add_expr()
+        args = [e, pat]
+        if self._accept_word("ESCAPE"):
+            esc = self.next()
+            if esc[0] != "str":
+                raise ValueError("ESCAPE needs a string literal")
+            args.append(("lit", esc[1], "VARCHAR"))
+        return ("call", op, args)
 
     def _in_list(self, e):
         self.expect_op("(")
@@ -441,29 +548,35 @@ class Parser:
             if t[1] == "NULL":
                 self.next()
                 return ("lit", None, "NULL")
-            if t[1] == "DATE":
+            if t[1] in ("DATE", "TIMESTAMP"):
+                # DATE '2026-01-01' is a literal; a bare DATE/TIMESTAMP not
+                # followed by a string is a column reference (a column
+                # aliased `AS date` being read back)
                 self.next()
-                s = self.next()
-                if s[0] != "str":
-                    raise ValueError("DATE needs a string literal")
-                return ("lit", s[1], "DATE")
-            if t[1] == "TIMESTAMP":
-                self.next()
-                s = self.next()
-                if s[0] != "str":
-                    raise ValueError("TIMESTAMP needs a string literal")
-                return ("lit", s[1], "TIMESTAMP")
+                if self.peek()[0] == "str":
+                    return ("lit", self.next()[1], t[1])
+                return self._colref_tail(t[1].lower())
             if t[1] == "INTERVAL":
-                # INTERVAL '<n>' DAY|WEEK|MONTH|YEAR (TPC-H date arithmetic)
+                # INTERVAL '<n>' DAY|WEEK|MONTH|YEAR, or the single-string
+                # form INTERVAL '5 days' (reference test_rex.py interval
+                # arithmetic)
                 self.next()
                 s = self.next()
                 if s[0] not in ("str", "num"):
                     raise ValueError("INTERVAL needs a quantity literal")
+                txt = str(s[1]).strip()
+                parts = txt.split()
+                units = ("DAY", "WEEK", "MONTH", "YEAR", "QUARTER",
+                         "HOUR", "MINUTE", "SECOND", "MILLISECOND",
+                         "MICROSECOND")
+                if s[0] == "str" and len(parts) == 2                         and parts[1].upper().rstrip("S") in units:
+                    return ("interval", int(parts[0]),
+                            parts[1].upper().rstrip("S"))
                 u = self.next()
                 unit = str(u[1]).upper().rstrip("S")
-                if unit not in ("DAY", "WEEK", "MONTH", "YEAR"):
+                if unit not in units:
                     raise ValueError(f"INTERVAL unit {u[1]!r} not supported")
-                return ("interval", int(str(s[1]).strip()), unit)
+                return ("interval", int(txt), unit)
             if t[1] == "EXISTS":
                 self.next()
                 self.expect_op("(")
@@ -525,7 +638,7 @@ class Parser:
                 ft = self.next()
                 field = ft[1].upper()
                 if field not in ("YEAR", "MONTH", "DAY", "HOUR", "MINUTE",
-                                 "SECOND"):
+                                 "SECOND", "DATE"):
                     raise ValueError(f"EXTRACT({field}) not supported")
                 self.expect_kw("FROM")
                 e = self.expr()
@@ -533,11 +646,14 @@ class Parser:
                 return ("call", f"EXTRACT_{field}", [e])
             if self.peek() == ("op", "("):
                 return self._func_call(name)
-            if self.accept_op("."):
-                col = self._name()
-                return ("col", name, col)
-            return ("col", None, name)
+            return self._colref_tail(name)
         raise ValueError(f"unexpected token {t}")
+
+    def _colref_tail(self, name):
+        if self.accept_op("."):
+            col = self._name()
+            return ("col", name, col)
+        return ("col", None, name)
 
     def _case(self):
         self.expect_kw("CASE")
@@ -568,9 +684,72 @@ class Parser:
                 self.next()
         return name
 
+    _TS_UNITS = ("YEAR", "QUARTER", "MONTH", "WEEK", "DAY", "HOUR",
+                 "MINUTE", "SECOND", "MILLISECOND", "MICROSECOND")
+
     def _func_call(self, name):
         fname = name.upper()
         self.expect_op("(")
+        if fname in ("TIMESTAMPADD", "TIMESTAMPDIFF"):
+            # TIMESTAMPADD(unit, n, ts) — the unit rides as a bare word
+            # (reference rex/core/call.py DatetimeSubOperation /
+            # TimeStampAddOperation)
+            ut = self.next()
+            unit = str(ut[1]).upper()
+            if unit not in self._TS_UNITS:
+                raise ValueError(f"{fname} unit {ut[1]!r} not supported")
+            self.expect_op(",")
+            a1 = self.expr()
+            self.expect_op(",")
+            a2 = self.expr()
+            self.expect_op(")")
+            if fname == "TIMESTAMPDIFF":
+                return ("call", "TIMESTAMPDIFF",
+                        [("lit", unit, "VARCHAR"), a1, a2])
+            # fold a literal count into an interval so the existing
+            # date±INTERVAL machinery does the arithmetic
+            neg = False
+            while isinstance(a1, tuple) and a1[0] == "call"                     and a1[1] == "NEG":
+                neg = not neg
+                a1 = a1[2][0]
+            if not (isinstance(a1, tuple) and a1[0] == "lit"
+                    and isinstance(a1[1], int)):
+                raise ValueError(f"{fname} count must be an integer literal")
+            n = -a1[1] if neg else a1[1]
+            return ("call", "+", [a2, ("interval", n, unit)])
+        if fname == "POSITION":
+            # POSITION(needle IN hay [FROM start]) — Calcite form
+            # needle parses below the IN-postfix level so `'a' IN a` is
+            # read as the POSITION separator, not a membership test
+            needle = self.add_expr()
+            if self.accept_kw("IN"):
+                hay = self.add_expr()
+                args = [hay, needle]
+                if self._accept_word("FROM"):
+                    args.append(self.expr())
+                self.expect_op(")")
+                return ("call", "POSITION", args)
+            args = [needle]
+            while self.accept_op(","):
+                args.append(self.expr())
+            self.expect_op(")")
+            return ("call", "POSITION", args)
+        if fname in ("FLOOR", "CEIL", "CEILING"):
+            # FLOOR(x TO DAY) — datetime truncation form
+            e = self.expr()
+            if self._accept_word("TO"):
+                ut = self.next()
+                unit = str(ut[1]).upper()
+                if unit not in self._TS_UNITS:
+                    raise ValueError(f"{fname}(.. TO {ut[1]!r}) unsupported")
+                self.expect_op(")")
+                base = "CEIL" if fname in ("CEIL", "CEILING") else "FLOOR"
+                return ("call", f"{base}_TO_{unit}", [e])
+            args = [e]
+            while self.accept_op(","):
+                args.append(self.expr())
+            self.expect_op(")")
+            return ("call", "CEIL" if fname == "CEILING" else fname, args)
         distinct = False
         args = []
         if self.accept_op("*"):

@@ -1,0 +1,103 @@
+"""Extract SQL strings from the reference's integration tests and try to
+PLAN each against approximated fixtures. Reports planning failures by
+category — a coverage probe, not a test (OUT-scope constructs expected
+to fail)."""
+import re
+import sys
+from pathlib import Path
+
+sys.path.insert(0, "/root/repo")
+import numpy as np
+import pandas as pd
+
+from dask_sql_amd.context import Context
+
+REF = Path("/root/reference/tests/integration")
+FILES = ["test_select.py", "test_filter.py", "test_groupby.py",
+         "test_join.py", "test_sort.py", "test_union.py", "test_rex.py",
+         "test_complex.py", "test_distributeby.py"]
+
+c = Context()
+np.random.seed(42)
+# fixtures.py approximations
+c.create_table("df", pd.DataFrame(
+    {"a": [1.0] * 100 + [2.0] * 200 + [3.0] * 400,
+     "b": 10 * np.random.rand(700),
+     # extra columns several test files add to their local `df` frames
+     "c": np.arange(700, dtype="int64"),
+     "d": pd.date_range("2021-01-01", periods=700, freq="7h"),
+     "t": pd.date_range("2020-06-01", periods=700, freq="13h"),
+     "x": np.random.rand(700),
+     "y": np.random.rand(700)}))
+c.create_table("df_simple", pd.DataFrame(
+    {"a": [1, 2, 3], "b": [1.1, 2.2, 3.3]}))
+c.create_table("df_wide", pd.DataFrame(
+    {"a": [1, 2], "b": [3, 4], "c": [5, 6]}))
+c.create_table("user_table_1", pd.DataFrame(
+    {"user_id": [2, 1, 2, 3], "b": [3, 3, 1, 3]}))
+c.create_table("user_table_2", pd.DataFrame(
+    {"user_id": [1, 1, 2, 4], "c": [1, 2, 3, 4]}))
+c.create_table("long_table", pd.DataFrame({"a": [0] * 100}))
+c.create_table("user_table_nan", pd.DataFrame(
+    {"c": pd.array([3, pd.NA, 1], dtype="Int8")}))
+c.create_table("user_table_inf", pd.DataFrame(
+    {"c": [3, float("inf"), 1]}))
+c.create_table("string_table", pd.DataFrame(
+    {"a": ["a normal string", "%_%", "^|()-*[]$"]}))
+c.create_table("datetime_table", pd.DataFrame({
+    "timezone": pd.date_range("2014-08-01", periods=3, freq="h"),
+    "no_timezone": pd.date_range("2014-08-01", periods=3, freq="h"),
+    "utc_timezone": pd.date_range("2014-08-01", periods=3, freq="h"),
+}))
+c.create_table("df1", pd.DataFrame(
+    {"a": [1, 2, 3], "b": [4, 5, 6], "id": [1, 2, 3]}))
+c.create_table("df2", pd.DataFrame(
+    {"a": [1, 2, 3], "b": [4, 5, 6], "id": [1, 2, 3]}))
+c.create_table("df_1", pd.DataFrame({"id": [1, 2, 3]}))
+c.create_table("df_2", pd.DataFrame({"id": [2, 3, 4]}))
+c.create_table("dates", pd.DataFrame(
+    {"d": pd.date_range("2021-01-01", periods=5, freq="D")}))
+c.create_table("datetime_test", pd.DataFrame(
+    {"a": pd.date_range("2021-01-01", periods=5, freq="D"),
+     "b": np.arange(5)}))
+c.create_table("gpu_df", pd.DataFrame({"a": [1.0], "b": [1.0]}))
+c.create_table("gpu_user_table_1", pd.DataFrame(
+    {"user_id": [2], "b": [3]}))
+c.create_table("gpu_long_table", pd.DataFrame({"a": [0]}))
+c.create_table("gpu_string_table", pd.DataFrame({"a": ["x"]}))
+c.create_table("gpu_datetime_table", pd.DataFrame(
+    {"timezone": pd.date_range("2014-08-01", periods=3, freq="h")}))
+
+sql_rx = re.compile(r'(?:c|context)\.sql\(\s*(?:f?"""(.*?)"""|f?"([^"]+)")',
+                    re.S)
+ok = bad = skipped = 0
+fails = {}
+for fn in FILES:
+    text = (REF / fn).read_text()
+    for m in sql_rx.finditer(text):
+        q = (m.group(1) or m.group(2)).strip()
+        if not q.upper().startswith(("SELECT", "WITH")):
+            skipped += 1
+            continue
+        if "{" in q:  # f-string templates — substitute common params
+            q2 = (q.replace("{input_table_1}", "user_table_1")
+                   .replace("{input_df}", "df")
+                   .replace("{input_table}", "user_table_1")
+                   .replace("{table}", "user_table_1")
+                   .replace("{gpu_t}", "df"))
+            if "{" in q2:
+                skipped += 1
+                continue
+            q = q2
+        try:
+            c._get_ral(q)
+            ok += 1
+        except Exception as e:
+            bad += 1
+            key = f"{type(e).__name__}: {str(e)[:90]}"
+            fails.setdefault(key, []).append((fn, q[:100].replace("\n", " ")))
+print(f"planned OK: {ok}  failed: {bad}  skipped(non-select/f-str): {skipped}")
+for k, v in sorted(fails.items(), key=lambda kv: -len(kv[1]))[:25]:
+    print(f"\n[{len(v)}x] {k}")
+    for fn, q in v[:2]:
+        print(f"    {fn}: {q}")

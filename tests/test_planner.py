@@ -320,6 +320,21 @@ PLAN_BATTERY = [
     "SELECT a FROM t ORDER BY a DESC, b",
     "SELECT RANK() OVER (PARTITION BY a ORDER BY b) AS r, "
     "DENSE_RANK() OVER (ORDER BY b) AS d FROM t",
+    # round-2 late additions: CTEs, set ops, IS TRUE, comments, aliases
+    "WITH big AS (SELECT a, b FROM t WHERE a > 1) "
+    "SELECT a, COUNT(*) AS n FROM big GROUP BY a",
+    "WITH x AS (SELECT a FROM t), y AS (SELECT c AS a FROM u) "
+    "SELECT x.a FROM x JOIN y ON x.a = y.a",
+    "SELECT a FROM t INTERSECT SELECT c FROM u",
+    "SELECT a FROM t EXCEPT SELECT c FROM u",
+    "SELECT a FROM t UNION SELECT c FROM u INTERSECT SELECT a FROM t",
+    "SELECT a FROM t INTERSECT SELECT c FROM u ORDER BY a LIMIT 2",
+    "SELECT a + 1 AS date FROM t WHERE a > 0",
+    "SELECT a IS TRUE AS x, a IS NOT TRUE AS y, a IS FALSE AS z, "
+    "a IS NOT UNKNOWN AS w FROM t",
+    "SELECT a -- line comment\n FROM t /* block\n comment */ WHERE a > 0",
+    "SELECT * FROM (SELECT a FROM t LIMIT 2) JOIN u ON a = c",
+    "SELECT TIMESTAMPDIFF(DAY, a, b) FROM t",
 ]
 
 
@@ -352,6 +367,17 @@ def test_plan_battery_dates():
         "SELECT TRIM('x' FROM s), INITCAP(s), REPLACE(s, 'x', 'y'), "
         "CHAR_LENGTH(s) FROM t",
         "SELECT v FROM t WHERE s LIKE 'x%' AND UPPER(s) = 'X'",
+        "SELECT v FROM t WHERE s ILIKE 'X%' OR s NOT ILIKE '%y'",
+        "SELECT v FROM t WHERE s SIMILAR TO '(x|y)%'",
+        "SELECT v FROM t WHERE s LIKE 'x!%%' ESCAPE '!'",
+        "SELECT v FROM t WHERE d >= d - INTERVAL '90 days'",
+        "SELECT TIMESTAMPADD(DAY, 5, ts) AS ts2 FROM t",
+        "SELECT TIMESTAMPADD(HOUR, -3, ts) AS ts3 FROM t",
+        "SELECT FLOOR(ts TO DAY), CEIL(ts TO HOUR), FLOOR(ts TO YEAR), "
+        "FLOOR(d TO MONTH) FROM t",
+        "SELECT EXTRACT(DATE FROM ts) AS dt FROM t",
+        "SELECT POSITION('x' IN s) AS p, POSITION('x' IN s FROM 2) FROM t",
+        "SELECT ts + INTERVAL '1' HOUR, ts - INTERVAL '30' MINUTE FROM t",
     ]:
         c.explain(q)
 
