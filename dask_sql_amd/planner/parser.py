@@ -252,6 +252,8 @@ class Parser:
                 s.items.append((e, alias))
             if not self.accept_op(","):
                 break
+            if self.peek() == ("kw", "FROM"):
+                break  # tolerated trailing comma (reference test_rex.py)
         if self.accept_kw("FROM"):
             s.from_tables.append(self.table_ref())
             while True:
@@ -791,6 +793,35 @@ class Parser:
                         break
             self.expect_op(")")
             return ("window", fname.lower(), args, tuple(part), tuple(order))
+        if fname in ("REGR_COUNT", "REGR_SXX", "REGR_SYY", "COVAR_POP",
+                     "COVAR_SAMP") and len(args) == 2:
+            # bivariate aggregates over non-NULL PAIRS, rewritten onto the
+            # existing SUM/COUNT machinery (reference rel/custom/wrappers +
+            # datafusion REGR lowering): REGR_COUNT = pairwise COUNT,
+            # REGR_SXX/SYY = S_vv - S_v^2/N, COVAR = (S_yx - S_y S_x/N)/N
+            y, x = args
+            pair = ("call", "AND", [("call", "IS NOT NULL", [y]),
+                                    ("call", "IS NOT NULL", [x])])
+            filt = pair if filter_expr is None else                 ("call", "AND", [filter_expr, pair])
+            cnt = ("agg", "count", [("star",)], False, filt)
+            if fname == "REGR_COUNT":
+                return cnt
+            nf = ("cast", cnt, "DOUBLE")
+
+            def sm(e):
+                return ("agg", "sum", [e], False, filt)
+
+            if fname in ("REGR_SXX", "REGR_SYY"):
+                v = x if fname == "REGR_SXX" else y
+                return ("call", "-",
+                        [sm(("call", "*", [v, v])),
+                         ("call", "/",
+                          [("call", "*", [sm(v), sm(v)]), nf])])
+            cov = ("call", "-",
+                   [sm(("call", "*", [y, x])),
+                    ("call", "/", [("call", "*", [sm(y), sm(x)]), nf])])
+            den = nf if fname == "COVAR_POP" else                 ("call", "-", [nf, ("lit", 1, "BIGINT")])
+            return ("call", "/", [cov, den])
         if fname in AGG_FUNCS:
             return ("agg", fname.lower(), args, distinct, filter_expr)
         return ("call", fname, args)

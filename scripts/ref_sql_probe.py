@@ -60,6 +60,34 @@ c.create_table("dates", pd.DataFrame(
 c.create_table("datetime_test", pd.DataFrame(
     {"a": pd.date_range("2021-01-01", periods=5, freq="D"),
      "b": np.arange(5)}))
+# dask.datasets.timeseries schema: id/name/x/y
+c.create_table("timeseries", pd.DataFrame(
+    {"id": np.arange(30), "name": pd.Series(["Alice", "Bob", "Xavier"] * 10
+                                            ).astype("category"),
+     "x": np.random.rand(30) * 2 - 1, "y": np.random.rand(30) * 2 - 1}))
+c.create_table("department_table", pd.DataFrame(
+    {"department_name": ["English", "Math", "Science"]}))
+c.create_table("string_table2", pd.DataFrame(
+    {"b": pd.Series(["a", "b", None]).astype("category")}))
+c.create_table("d_table", pd.DataFrame(
+    {"d_date": pd.to_datetime(["2023-07-01", "2023-07-05"]),
+     "x": [1, 2]}))
+c.create_table("sales", pd.DataFrame(
+    {"sales_hdemo_sk": [1], "sales_page_sk": [1], "sold_time_sk": [1]}))
+c.create_table("demos", pd.DataFrame(
+    {"demo_sku": [1], "hd_dep_count": [1]}))
+c.create_table("site_page", pd.DataFrame(
+    {"site_page_sk": [1], "site_char_count": [1]}))
+c.create_table("t_dim", pd.DataFrame({"t_time_sk": [1], "t_hour": [1]}))
+c.create_table("many_partitions", pd.DataFrame(
+    {"a": [1, 2], "b": [3, 4], "c": [5, 6]}))
+c.create_table("parquet_ddf", pd.DataFrame(
+    {"a": [1, 2, 3], "b": [0, 1, 2],
+     "c": pd.Series(["A"] * 3).astype("category"),
+     "d": pd.to_datetime(["2013-08-01 23:00:00"] * 3),
+     "index": [0, 1, 2]}))
+c.create_table("my_csv_table", pd.DataFrame(
+    {"a": [1, 2], "b": [1.0, 2.0], "c": [2, 3]}))
 c.create_table("gpu_df", pd.DataFrame({"a": [1.0], "b": [1.0]}))
 c.create_table("gpu_user_table_1", pd.DataFrame(
     {"user_id": [2], "b": [3]}))

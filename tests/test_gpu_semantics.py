@@ -2009,3 +2009,32 @@ def test_interval_string_and_comments(ctx):
                   "WHERE d < DATE '2021-01-01' + INTERVAL '5 days'"
                   ).compute()
     assert sorted(out["v"].astype(np.int64).tolist()) == [0]
+
+
+def test_regr_covar_family(ctx):
+    rng = np.random.default_rng(71)
+    n = 4000
+    df = pd.DataFrame({
+        "name": pd.Series(rng.choice(["a", "b", "c"], n)).astype("category"),
+        "x": np.round(rng.random(n), 3), "y": np.round(rng.random(n), 3)})
+    df.loc[rng.random(n) < 0.1, "x"] = np.nan
+    df.loc[rng.random(n) < 0.1, "y"] = np.nan
+    ctx.create_table("trc", df)
+    out = ctx.sql(
+        "SELECT name, REGR_COUNT(y, x) AS n, REGR_SXX(y, x) AS sxx, "
+        "REGR_SYY(y, x) AS syy, COVAR_POP(y, x) AS cp, "
+        "COVAR_SAMP(y, x) AS cs FROM trc GROUP BY name").compute()
+    out = out.sort_values("name").reset_index(drop=True)
+    pair = df.dropna(subset=["x", "y"])
+    for i, (g, sub) in enumerate(pair.groupby("name", observed=True)):
+        assert out.loc[i, "name"] == g
+        N = len(sub)
+        assert int(out.loc[i, "n"]) == N
+        sxx = ((sub.x - sub.x.mean()) ** 2).sum()
+        syy = ((sub.y - sub.y.mean()) ** 2).sum()
+        cp = ((sub.x - sub.x.mean()) * (sub.y - sub.y.mean())).sum() / N
+        cs = cp * N / (N - 1)
+        assert abs(out.loc[i, "sxx"] - sxx) < 1e-6 * max(1, abs(sxx))
+        assert abs(out.loc[i, "syy"] - syy) < 1e-6 * max(1, abs(syy))
+        assert abs(out.loc[i, "cp"] - cp) < 1e-6 * max(1, abs(cp))
+        assert abs(out.loc[i, "cs"] - cs) < 1e-6 * max(1, abs(cs))
