@@ -480,6 +480,8 @@ class RexCompiler:
         if op == "EXTRACT_DATE" or op.startswith("FLOOR_TO_") \
                 or op.startswith("CEIL_TO_"):
             return self._compile_dt_trunc(op, ops)
+        if op.startswith("EXTRACT_"):
+            return self._compile_extract_ext(op, ops)
         raise RexCompileError(f"operator {op} not supported on GPU path")
 
     _DAY_NS = 86_400_000_000_000
@@ -580,6 +582,92 @@ class RexCompiler:
             self._emit(OP_LIT_I64, 0, self._DAY_NS)
             self._emit(OP_MUL_I64)
         return KI
+
+    def _compile_extract_ext(self, op, ops):
+        """EXTRACT fields beyond Y/M/D/H/M/S, matching the REFERENCE's
+        date_part exactly (rex/core/call.py:917-960): CENTURY/DECADE/
+        MILLENNIUM = trunc(year/unit); DOW = (pandas dayofweek+1)%7, i.e.
+        Sunday=0 ((days+4) mod 7 — epoch day 0 was a Thursday); DOY = days
+        since Jan 1 + 1; QUARTER = (month+2)/3; MICROSECOND = sub-second
+        microseconds; MILLISECOND = 1000*microsecond (the reference's own
+        convention, kept for parity)."""
+        t0 = getattr(ops[0], "getType", lambda: None)()
+        is_ts = t0 is not None and t0.getSqlType() == "TIMESTAMP"
+        x = ops[0]
+        field = op[len("EXTRACT_"):]
+
+        def emit_y():
+            self._emit_days(x, is_ts)
+            self._emit(OP_YEAR)
+
+        if field in ("CENTURY", "DECADE", "MILLENNIUM"):
+            emit_y()
+            self._emit(OP_LIT_I64, 0,
+                       {"CENTURY": 100, "DECADE": 10,
+                        "MILLENNIUM": 1000}[field])
+            self._emit(OP_DIV_I64)
+            return KI
+        if field == "QUARTER":
+            self._emit_days(x, is_ts)
+            self._emit(OP_MONTH)
+            self._emit(OP_LIT_I64, 0, 2)
+            self._emit(OP_ADD_I64)
+            self._emit(OP_LIT_I64, 0, 3)
+            self._emit(OP_DIV_I64)
+            return KI
+        if field == "DOW":
+            self._emit_days(x, is_ts)
+            self._emit(OP_LIT_I64, 0, 4)
+            self._emit(OP_ADD_I64)
+            self._emit(OP_LIT_I64, 0, 7)
+            self._emit(OP_FLOORMOD_I64)
+            return KI
+        if field == "DOY":
+            # days - jan1_days + 1 (same leap-count identity as
+            # FLOOR_TO_YEAR)
+            self._emit_days(x, is_ts)
+            emit_y()
+            self._emit(OP_LIT_I64, 0, 365)
+            self._emit(OP_MUL_I64)
+
+            def leap(div):
+                emit_y()
+                self._emit(OP_LIT_I64, 0, 1)
+                self._emit(OP_SUB_I64)
+                self._emit(OP_LIT_I64, 0, div)
+                self._emit(OP_DIV_I64)
+
+            leap(4)
+            self._emit(OP_ADD_I64)
+            leap(100)
+            self._emit(OP_SUB_I64)
+            leap(400)
+            self._emit(OP_ADD_I64)
+            self._emit(OP_LIT_I64, 0, 719_527)
+            self._emit(OP_SUB_I64)
+            self._emit(OP_SUB_I64)
+            self._emit(OP_LIT_I64, 0, 1)
+            self._emit(OP_ADD_I64)
+            return KI
+        if field in ("MICROSECOND", "MILLISECOND"):
+            if not is_ts:
+                # a DATE has no sub-second part; 0*x keeps x's validity
+                k = self.compile(x)
+                if k != KI:
+                    raise RexCompileError(f"{op} needs DATE/TIMESTAMP")
+                self._emit(OP_LIT_I64, 0, 0)
+                self._emit(OP_MUL_I64)
+                return KI
+            self.compile(x)
+            self._emit(OP_LIT_I64, 0, 1_000_000_000)
+            self._emit(OP_FLOORMOD_I64)
+            self._emit(OP_LIT_I64, 0, 1000)
+            self._emit(OP_DIV_I64)
+            if field == "MILLISECOND":
+                self._emit(OP_LIT_I64, 0, 1000)
+                self._emit(OP_MUL_I64)
+            return KI
+        raise RexCompileError(f"operator {op} not supported on GPU path")
 
     def _compile_like(self, ops, mode="LIKE"):
         """LIKE/ILIKE/SIMILAR on a dict-encoded column: the SQL pattern
@@ -722,7 +810,8 @@ class RexCompiler:
             if op == "CASE":
                 return self._peek_kind(expr.getOperands()[1])
             if op in ("MOD", "EXTRACT_YEAR", "EXTRACT_MONTH", "EXTRACT_DAY",
-                      "YEAR", "MONTH", "DAY", "DAYOFMONTH", "EXTRACT_DATE") \
+                      "YEAR", "MONTH", "DAY", "DAYOFMONTH") \
+                    or op.startswith("EXTRACT_") \
                     or op.startswith("FLOOR_TO_") \
                     or op.startswith("CEIL_TO_"):
                 return KI

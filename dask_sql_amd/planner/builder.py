@@ -189,12 +189,11 @@ class Builder:
             elif op in ("FLOOR", "CEIL", "CEILING", "ROUND", "EXP", "LN",
                         "LOG", "POWER", "POW", "SQRT"):
                 ty = "DOUBLE"
-            elif op in ("EXTRACT_YEAR", "EXTRACT_MONTH", "EXTRACT_DAY",
-                        "EXTRACT_HOUR", "EXTRACT_MINUTE", "EXTRACT_SECOND",
-                        "MOD", "YEAR", "MONTH", "DAY", "DAYOFMONTH"):
-                ty = "BIGINT"
             elif op == "EXTRACT_DATE":
                 ty = "DATE"
+            elif op.startswith("EXTRACT_") or op in (
+                    "MOD", "YEAR", "MONTH", "DAY", "DAYOFMONTH"):
+                ty = "BIGINT"
             elif op.startswith("FLOOR_TO_") or op.startswith("CEIL_TO_"):
                 ty = _expr_type(ops[0])
             elif op == "TIMESTAMPDIFF":

@@ -639,8 +639,16 @@ class Parser:
                 self.next()
                 ft = self.next()
                 field = ft[1].upper()
+                if field == "CENTURIES":
+                    field = "CENTURY"
+                elif field.endswith("S") and field != "DOW":
+                    field = field[:-1]
+                if field in ("MILLENIUM", "MILLENNIUM"):
+                    field = "MILLENNIUM"  # reference accepts the misspelling
                 if field not in ("YEAR", "MONTH", "DAY", "HOUR", "MINUTE",
-                                 "SECOND", "DATE"):
+                                 "SECOND", "DATE", "CENTURY", "DECADE",
+                                 "MILLENNIUM", "DOW", "DOY", "QUARTER",
+                                 "MICROSECOND", "MILLISECOND"):
                     raise ValueError(f"EXTRACT({field}) not supported")
                 self.expect_kw("FROM")
                 e = self.expr()

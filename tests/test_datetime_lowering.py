@@ -153,3 +153,41 @@ def test_calendar_ceil_rejected():
         _compile("CEIL_TO_MONTH", "TIMESTAMP")
     with pytest.raises(R.RexCompileError):
         _compile("CEIL_TO_YEAR", "TIMESTAMP")
+
+
+def test_extract_extended_vs_reference_semantics():
+    """CENTURY/DECADE/MILLENNIUM/DOW/DOY/QUARTER/MICROSECOND/MILLISECOND
+    against the reference's date_part definitions (rex/core/call.py:917):
+    CENTURY = trunc(year/100), DOW = (pandas dayofweek+1)%7,
+    MILLISECOND = 1000*microsecond (reference convention)."""
+    for s in TS_SAMPLES:
+        ts = pd.Timestamp(s)
+        cases = {
+            "EXTRACT_CENTURY": int(ts.year / 100),
+            "EXTRACT_DECADE": int(ts.year / 10),
+            "EXTRACT_MILLENNIUM": int(ts.year / 1000),
+            "EXTRACT_DOW": (ts.dayofweek + 1) % 7,
+            "EXTRACT_DOY": ts.dayofyear,
+            "EXTRACT_QUARTER": ts.quarter,
+            "EXTRACT_MICROSECOND": ts.microsecond,
+            "EXTRACT_MILLISECOND": 1000 * ts.microsecond,
+        }
+        for op_name, want in cases.items():
+            prog = _compile(op_name, "TIMESTAMP")
+            assert interp_prog(prog, ts.value) == want, (op_name, s)
+
+
+def test_extract_extended_on_date():
+    for s in ("2021-02-17", "2000-02-29", "1971-01-01"):
+        ts = pd.Timestamp(s)
+        days = (ts - pd.Timestamp(0)).days
+        for op_name, want in {
+            "EXTRACT_DOW": (ts.dayofweek + 1) % 7,
+            "EXTRACT_DOY": ts.dayofyear,
+            "EXTRACT_QUARTER": ts.quarter,
+            "EXTRACT_CENTURY": int(ts.year / 100),
+            "EXTRACT_MICROSECOND": 0,
+            "EXTRACT_MILLISECOND": 0,
+        }.items():
+            prog = _compile(op_name, "DATE")
+            assert interp_prog(prog, days) == want, (op_name, s)

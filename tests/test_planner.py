@@ -379,6 +379,9 @@ def test_plan_battery_dates():
         "SELECT FLOOR(ts TO DAY), CEIL(ts TO HOUR), FLOOR(ts TO YEAR), "
         "FLOOR(d TO MONTH) FROM t",
         "SELECT EXTRACT(DATE FROM ts) AS dt FROM t",
+        "SELECT EXTRACT(CENTURY FROM ts), EXTRACT(DOW FROM d), "
+        "EXTRACT(DOY FROM ts), EXTRACT(QUARTER FROM d), "
+        "EXTRACT(MILLISECOND FROM ts), EXTRACT(DECADE FROM ts) FROM t",
         "SELECT POSITION('x' IN s) AS p, POSITION('x' IN s FROM 2) FROM t",
         "SELECT ts + INTERVAL '1' HOUR, ts - INTERVAL '30' MINUTE FROM t",
     ]:
