@@ -720,19 +720,27 @@ class RexCompiler:
     def _compile_case(self, ops):
         # rightmost-else first; build nested SELECTs. Postfix SELECT pops
         # (cond, a, b). Emit conds/vals in order with SELECT folds from the
-        # back: CASE c1 v1 c2 v2 e == SELECT(c1, v1, SELECT(c2, v2, e))
+        # back: CASE c1 v1 c2 v2 e == SELECT(c1, v1, SELECT(c2, v2, e)).
+        # Branch kinds are UNIFIED to float if any branch is float —
+        # SELECT moves raw slots, so mixed int/float branches would
+        # otherwise reinterpret bits downstream (same pre-scan COALESCE
+        # does).
         *pairs, els = ops
         assert len(pairs) % 2 == 0
+        vals = [pairs[i + 1] for i in range(0, len(pairs), 2)] + [els]
+        target = KF if any(self._peek_kind(v) == KF for v in vals) else None
 
         def emit_chain(i):
             if i >= len(pairs):
                 k = self.compile(els)
+                if target == KF:
+                    k = self._to_f(k)
                 return k
             self.compile(pairs[i])        # cond
             kv = self.compile(pairs[i + 1])  # val
-            ke = emit_chain(i + 2)
-            if kv == KF or ke == KF:
-                pass  # mixed kinds: caller ensured same family (planner types)
+            if target == KF:
+                kv = self._to_f(kv)
+            emit_chain(i + 2)
             self._emit(OP_SELECT)
             return kv
 
@@ -808,7 +816,15 @@ class RexCompiler:
             if op in ("NEG", "ABS"):
                 return self._peek_kind(expr.getOperands()[0])
             if op == "CASE":
-                return self._peek_kind(expr.getOperands()[1])
+                # must match _compile_case's target rule: float if ANY
+                # branch value (or the else) is float
+                ops_ = expr.getOperands()
+                vals = [ops_[i + 1] for i in range(0, len(ops_) - 1, 2)]
+                if len(ops_) % 2 == 1:
+                    vals.append(ops_[-1])
+                if any(self._peek_kind(v) == KF for v in vals):
+                    return KF
+                return self._peek_kind(ops_[1])
             if op in ("MOD", "EXTRACT_YEAR", "EXTRACT_MONTH", "EXTRACT_DAY",
                       "YEAR", "MONTH", "DAY", "DAYOFMONTH") \
                     or op.startswith("EXTRACT_") \

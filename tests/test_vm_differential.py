@@ -1,0 +1,270 @@
+"""Randomized differential test of the rex compiler: compile random typed
+expression ASTs, run them through the faithful CPU model of the device VM
+(tests/vm_interp.py), and compare against a direct SQL-semantics evaluation
+of the same AST (null propagation, Kleene AND/OR, floor-MOD — the reference
+rex/core/call.py behaviors). Catches compile-time lowering bugs (operand
+order, validity propagation, conversions) without a GPU; the device VM
+itself is pinned by the gpu-marked parity suites."""
+import math
+
+import numpy as np
+import pytest
+
+import types
+
+from dask_sql_amd import runtime as rt
+from dask_sql_amd.physical import rex as R
+from dask_sql_amd.planner.plan import Call, InputRef, Literal, SqlType
+from tests.vm_interp import interp, _i64, _tdiv
+
+
+class Skip(Exception):
+    """Input hits a documented divergence (int division by zero, float→int
+    of non-finite) — not compared."""
+
+
+# one compiler instance only for static kind queries (mirrors the typing
+# the compiled program actually uses for int-vs-float division)
+def _pk():
+    return R.RexCompiler(_dev_cols())._peek_kind
+
+
+# ---- direct SQL-semantics evaluator over the AST --------------------------
+def ev(e, cols, row, pk=None):
+    """Returns Python value or None (SQL NULL)."""
+    if pk is None:
+        pk = _pk()
+    if isinstance(e, InputRef):
+        vals, valid = cols[e.getIndex()]
+        if valid is not None and not valid[row]:
+            return None
+        v = vals[row]
+        return float(v) if "float" in str(getattr(v, "dtype", "")) \
+            else (float(v) if isinstance(v, float) else int(v))
+    if isinstance(e, Literal):
+        return e.getValue()
+    op = e.getOperatorName()
+    ops = e.getOperands()
+    if op == "AND":
+        a, b = ev(ops[0], cols, row, pk), ev(ops[1], cols, row, pk)
+        if a is False or b is False:
+            return False
+        if a is None or b is None:
+            return None
+        return True
+    if op == "OR":
+        a, b = ev(ops[0], cols, row, pk), ev(ops[1], cols, row, pk)
+        if a is True or b is True:
+            return True
+        if a is None or b is None:
+            return None
+        return False
+    if op == "NOT":
+        a = ev(ops[0], cols, row, pk)
+        return None if a is None else (not a)
+    if op == "IS NULL":
+        return ev(ops[0], cols, row, pk) is None
+    if op == "IS NOT NULL":
+        return ev(ops[0], cols, row, pk) is not None
+    if op == "CASE":
+        n = len(ops)
+        i = 0
+        while i + 1 < n:
+            if ev(ops[i], cols, row, pk) is True:
+                return ev(ops[i + 1], cols, row, pk)
+            i += 2
+        return ev(ops[-1], cols, row, pk) if n % 2 == 1 else None
+    if op == "COALESCE":
+        out = None
+        for o in ops:
+            out = ev(o, cols, row, pk)
+            if out is not None:
+                return out
+        return out
+    if op == "NULLIF":
+        a, b = ev(ops[0], cols, row, pk), ev(ops[1], cols, row, pk)
+        if a is None:
+            return None
+        if b is not None and a == b:
+            return None
+        return a
+    if op == "CAST":
+        a = ev(ops[0], cols, row, pk)
+        if a is None:
+            return None
+        ty = e.getType().getSqlType()
+        if ty in ("BIGINT", "INTEGER"):
+            if isinstance(a, float):
+                if not math.isfinite(a):
+                    raise Skip()
+                return _i64(int(a))
+            return int(a)
+        if ty == "DOUBLE":
+            return float(a)
+        raise Skip()
+    if op in ("ABS",):
+        a = ev(ops[0], cols, row, pk)
+        return None if a is None else abs(a)
+    if op == "NEG":
+        a = ev(ops[0], cols, row, pk)
+        if a is None:
+            return None
+        return -float(a) if pk(ops[0]) == R.KF else _i64(-int(a))
+    if op == "MOD":
+        a, b = ev(ops[0], cols, row, pk), ev(ops[1], cols, row, pk)
+        if a is None or b is None:
+            return None
+        if b == 0:
+            raise Skip()
+        return int(a) % int(b)  # python % IS floor-mod (operator.mod ref)
+    if op in ("+", "-", "*", "/"):
+        a, b = ev(ops[0], cols, row, pk), ev(ops[1], cols, row, pk)
+        if a is None or b is None:
+            return None
+        # the compiled program picks int vs float by STATIC kind
+        fl = R.KF in (pk(ops[0]), pk(ops[1]))
+        if fl:
+            a, b = float(a), float(b)
+        if op == "+":
+            r = a + b
+        elif op == "-":
+            r = a - b
+        elif op == "*":
+            r = a * b
+        else:
+            if not fl:
+                if b == 0:
+                    raise Skip()
+                return _i64(_tdiv(int(a), int(b)))
+            if float(b) == 0.0:
+                raise Skip()  # inf/nan sign subtleties not under test
+            r = float(a) / float(b)
+        return r if fl else _i64(r)
+    if op in ("=", "<>", "<", "<=", ">", ">="):
+        a, b = ev(ops[0], cols, row, pk), ev(ops[1], cols, row, pk)
+        if a is None or b is None:
+            return None
+        return {"=": a == b, "<>": a != b, "<": a < b, "<=": a <= b,
+                ">": a > b, ">=": a >= b}[op]
+    raise AssertionError(f"ev: op {op} not modeled")
+
+
+# ---- random typed AST generator -------------------------------------------
+B, I, F = "BOOLEAN", "BIGINT", "DOUBLE"
+
+
+def gen(rng, kind, depth):
+    """Random AST of SQL type `kind` over 4 columns:
+    0 int dense, 1 int nullable, 2 float dense, 3 float nullable."""
+    def lit_num():
+        if rng.random() < 0.5:
+            return Literal(int(rng.integers(-50, 50)), SqlType(I))
+        return Literal(round(float(rng.uniform(-50, 50)), 3), SqlType(F))
+
+    if kind == B:
+        r = rng.random()
+        if depth <= 0 or r < 0.15:
+            a, b = gen(rng, "NUM", 0), gen(rng, "NUM", 0)
+            cmp_op = rng.choice(["=", "<>", "<", "<=", ">", ">="])
+            return Call(str(cmp_op), [a, b], SqlType(B))
+        if r < 0.45:
+            return Call(str(rng.choice(["AND", "OR"])),
+                        [gen(rng, B, depth - 1), gen(rng, B, depth - 1)],
+                        SqlType(B))
+        if r < 0.6:
+            return Call("NOT", [gen(rng, B, depth - 1)], SqlType(B))
+        if r < 0.75:
+            return Call(str(rng.choice(["IS NULL", "IS NOT NULL"])),
+                        [gen(rng, "NUM", depth - 1)], SqlType(B))
+        a, b = gen(rng, "NUM", depth - 1), gen(rng, "NUM", depth - 1)
+        cmp_op = rng.choice(["=", "<>", "<", "<=", ">", ">="])
+        return Call(str(cmp_op), [a, b], SqlType(B))
+    # numeric
+    r = rng.random()
+    if depth <= 0 or r < 0.25:
+        c = rng.random()
+        if c < 0.2:
+            return lit_num()
+        if c < 0.25:
+            return Literal(None, SqlType("NULL"))
+        i = int(rng.integers(0, 4))
+        return InputRef(i, SqlType(I if i < 2 else F))
+    if r < 0.55:
+        op = str(rng.choice(["+", "-", "*"]))
+        return Call(op, [gen(rng, "NUM", depth - 1),
+                         gen(rng, "NUM", depth - 1)], SqlType(F))
+    if r < 0.62:
+        return Call("/", [gen(rng, "NUM", depth - 1),
+                          gen(rng, "NUM", depth - 1)], SqlType(F))
+    if r < 0.72:
+        return Call("CASE", [gen(rng, B, depth - 1),
+                             gen(rng, "NUM", depth - 1),
+                             gen(rng, "NUM", depth - 1)], SqlType(F))
+    if r < 0.8:
+        return Call("COALESCE", [gen(rng, "NUM", depth - 1),
+                                 gen(rng, "NUM", depth - 1)], SqlType(F))
+    if r < 0.85:
+        return Call("NULLIF", [gen(rng, "NUM", depth - 1),
+                               gen(rng, "NUM", depth - 1)], SqlType(F))
+    if r < 0.9:
+        return Call("ABS", [gen(rng, "NUM", depth - 1)], SqlType(F))
+    if r < 0.95:
+        return Call("NEG", [gen(rng, "NUM", depth - 1)], SqlType(F))
+    ty = I if rng.random() < 0.5 else F
+    return Call("CAST", [gen(rng, "NUM", depth - 1)], SqlType(ty))
+
+
+def _make_cols(rng, n):
+    iv = rng.integers(-40, 40, n).astype(np.int64)
+    inul = rng.integers(-40, 40, n).astype(np.int64)
+    ival = (rng.random(n) > 0.25).astype(np.uint8)
+    fv = np.round(rng.uniform(-40, 40, n), 3)
+    fnul = np.round(rng.uniform(-40, 40, n), 3)
+    fval = (rng.random(n) > 0.25).astype(np.uint8)
+    return [(iv, None), (inul, ival), (fv, None), (fnul, fval)]
+
+
+def _dev_cols():
+    return [types.SimpleNamespace(dtype=rt.I64),
+            types.SimpleNamespace(dtype=rt.I64),
+            types.SimpleNamespace(dtype=rt.F64),
+            types.SimpleNamespace(dtype=rt.F64)]
+
+
+@pytest.mark.parametrize("seed", range(8))
+def test_vm_differential(seed):
+    rng = np.random.default_rng(1234 + seed)
+    n = 40
+    cols = _make_cols(rng, n)
+    compared = 0
+    for k in range(60):
+        kind = B if k % 2 == 0 else "NUM"
+        e = gen(rng, kind, 4)
+        c = R.RexCompiler(_dev_cols())
+        try:
+            rk = c.compile(e)
+        except R.RexCompileError:
+            continue
+        for row in range(n):
+            try:
+                want = ev(e, cols, row)
+            except Skip:
+                continue
+            got, ok = interp(c.prog, cols, row)
+            compared += 1
+            if want is None:
+                assert not ok, (e, row, got)
+                continue
+            assert ok, (e, row, want)
+            if isinstance(want, bool):
+                assert (got != 0) == want, (e, row, got, want)
+            elif rk == R.KF or isinstance(want, float):
+                w = float(want)
+                if math.isnan(w):
+                    assert math.isnan(float(got)), (e, row, got)
+                else:
+                    assert abs(float(got) - w) <= 1e-9 * max(1.0, abs(w)), \
+                        (e, row, got, want)
+            else:
+                assert int(got) == int(want), (e, row, got, want)
+    assert compared > 500  # the sweep must actually exercise cases
