@@ -156,6 +156,36 @@ class Builder:
             fn = self.catalog.functions.get(op.lower())
             if fn is not None:
                 return Call(f"UDF:{op.lower()}", ops, SqlType(fn[1]))
+            if op == "TO_TIMESTAMP":
+                # reference rex/core/call.py ToTimestampOperation: string →
+                # strptime (host fold); numeric → seconds since epoch; an
+                # existing DATE/TIMESTAMP passes through (promoted to ns)
+                x = ops[0]
+                fmt = "%Y-%m-%d %H:%M:%S"
+                if len(ops) > 1 and isinstance(ops[1], Literal):
+                    fmt = str(ops[1].getValue()).replace('"', "")
+                    fmt = fmt.replace("'", "")
+                if isinstance(x, Literal) \
+                        and isinstance(x.getValue(), str):
+                    from datetime import datetime as _dt
+                    ns = int(np.datetime64(_dt.strptime(x.getValue(), fmt),
+                                           "ns").astype("int64"))
+                    return Literal(ns, SqlType("TIMESTAMP"))
+                tx = _expr_type(x)
+                if tx == "TIMESTAMP":
+                    return x
+                if len(ops) > 1 and fmt != "%Y-%m-%d %H:%M:%S":
+                    raise NotImplementedError(
+                        "Integer input does not accept a format argument")
+                scale = 86_400_000_000_000 if tx == "DATE" \
+                    else 1_000_000_000
+                mul = Call("*", [x, Literal(scale, SqlType("BIGINT"))],
+                           SqlType("TIMESTAMP"))
+                if tx in ("DOUBLE", "FLOAT", "DECIMAL"):
+                    # fractional seconds survive the ns multiply; the CAST
+                    # truncates to integer ns for the TIMESTAMP column
+                    return Call("CAST", [mul], SqlType("TIMESTAMP"))
+                return mul
             if op in ("+", "-") and any(
                     isinstance(o, Literal)
                     and o.getType().getSqlType() == "INTERVAL" for o in ops):

@@ -384,6 +384,9 @@ def test_plan_battery_dates():
         "EXTRACT(MILLISECOND FROM ts), EXTRACT(DECADE FROM ts) FROM t",
         "SELECT POSITION('x' IN s) AS p, POSITION('x' IN s FROM 2) FROM t",
         "SELECT ts + INTERVAL '1' HOUR, ts - INTERVAL '30' MINUTE FROM t",
+        "SELECT to_timestamp(v) AS t1, "
+        "to_timestamp('2021-03-02 10:00:00') AS t2, "
+        "TIMESTAMPADD(DAY, 2, to_timestamp(v)) AS t3 FROM t",
     ]:
         c.explain(q)
 
