@@ -440,7 +440,11 @@ class RexCompiler:
             return KF
         if op == "MOD":
             # reference evaluates operator.mod on pandas = FLOOR-mod
-            # (MOD(-5,3) = 1), not C truncated remainder
+            # (MOD(-5,3) = 1), not C truncated remainder.
+            # Integer-only: FLOORMOD reads raw int slots, a float operand
+            # would be reinterpreted — refuse loudly.
+            if KF in (self._peek_kind(ops[0]), self._peek_kind(ops[1])):
+                raise RexCompileError("MOD needs integer operands")
             self.compile(ops[0])
             self.compile(ops[1])
             self._emit(OP_FLOORMOD_I64)

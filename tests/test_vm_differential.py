@@ -105,6 +105,53 @@ def ev(e, cols, row, pk=None):
     if op in ("ABS",):
         a = ev(ops[0], cols, row, pk)
         return None if a is None else abs(a)
+    if op in ("FLOOR", "CEIL", "SQRT", "EXP", "LN"):
+        a = ev(ops[0], cols, row, pk)
+        if a is None:
+            return None
+        f = float(a)
+        if op == "FLOOR":
+            return float(math.floor(f))
+        if op == "CEIL":
+            return float(math.ceil(f))
+        if op == "SQRT":
+            if f < 0:
+                raise Skip()
+            return math.sqrt(f)
+        if op == "EXP":
+            try:
+                return math.exp(f)
+            except OverflowError:
+                return math.inf
+        if f <= 0:
+            raise Skip()
+        return math.log(f)
+    if op == "ROUND":
+        a = ev(ops[0], cols, row, pk)
+        if a is None:
+            return None
+        d = ops[1].getValue() if len(ops) > 1 else 0
+        # numpy-style ties-to-even via the same scale trick the VM uses
+        p10 = 10.0 ** int(d)
+        scaled = float(a) * p10
+        fl = math.floor(scaled)
+        fr = scaled - fl
+        if fr > 0.5:
+            r = fl + 1
+        elif fr < 0.5:
+            r = fl
+        else:
+            r = fl if fl % 2 == 0 else fl + 1
+        return r / p10
+    if op == "POWER":
+        a = ev(ops[0], cols, row, pk)
+        b = ev(ops[1], cols, row, pk)
+        if a is None or b is None:
+            return None
+        try:
+            return math.pow(float(a), float(b))
+        except (OverflowError, ValueError):
+            raise Skip()
     if op == "NEG":
         a = ev(ops[0], cols, row, pk)
         if a is None:
@@ -206,10 +253,20 @@ def gen(rng, kind, depth):
     if r < 0.85:
         return Call("NULLIF", [gen(rng, "NUM", depth - 1),
                                gen(rng, "NUM", depth - 1)], SqlType(F))
-    if r < 0.9:
+    if r < 0.88:
         return Call("ABS", [gen(rng, "NUM", depth - 1)], SqlType(F))
-    if r < 0.95:
+    if r < 0.9:
         return Call("NEG", [gen(rng, "NUM", depth - 1)], SqlType(F))
+    if r < 0.92:
+        fn = str(rng.choice(["FLOOR", "CEIL", "SQRT", "EXP", "LN"]))
+        return Call(fn, [gen(rng, "NUM", depth - 1)], SqlType(F))
+    if r < 0.94:
+        d = int(rng.integers(0, 3))
+        return Call("ROUND", [gen(rng, "NUM", depth - 1),
+                              Literal(d, SqlType(I))], SqlType(F))
+    if r < 0.95:
+        return Call("MOD", [gen(rng, "NUM", depth - 1),
+                            gen(rng, "NUM", depth - 1)], SqlType(I))
     ty = I if rng.random() < 0.5 else F
     return Call("CAST", [gen(rng, "NUM", depth - 1)], SqlType(ty))
 

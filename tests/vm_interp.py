@@ -137,7 +137,15 @@ def interp(prog, cols, row):
         elif op == R.OP_F64_TO_I64:
             a, av = pop1()
             f = float(a)
-            push(0 if math.isnan(f) else _i64(int(f)), av)
+            if math.isnan(f):
+                v = 0
+            elif f >= 2.0 ** 63:
+                v = (1 << 63) - 1   # hardware cvt saturates
+            elif f <= -(2.0 ** 63):
+                v = -(1 << 63)
+            else:
+                v = _i64(int(f))
+            push(v, av)
         elif op == R.OP_BITS_F64:
             a, av = pop1()
             push(struct.unpack("<d", struct.pack("<q", int(a)))[0], av)
