@@ -3379,6 +3379,34 @@ void jit_cache_destroy(DsxCtx* c) {
 #include "window.inc"
 
 // ---------------------------------------------------------------------------
+// dsx_jit_expr_source — TEST INFRASTRUCTURE: emit the JIT expression
+// evaluator's generated C for a program (the exact text the kernels
+// compile via hipRTC) into buf, so the CPU differential suite can compile
+// it with gcc and pin codegen semantics against the interpreter VM without
+// a GPU. Host-only: no HIP calls. Returns 1 if the result kind is double,
+// 0 for i64, <0 on error.
+extern "C" int dsx_jit_expr_source(const DsxInstr* prog, int prog_len,
+                                   const int32_t* dtypes,
+                                   const uint8_t* has_validity, int ncols,
+                                   char* buf, int64_t cap) {
+  if (prog_len <= 0 || prog_len > DSX_MAX_PROG || ncols < 0 ||
+      ncols > DSX_MAX_COLS)
+    return -3;
+  ColsArg C{};
+  C.ncols = ncols;
+  for (int i = 0; i < ncols; i++) {
+    C.dtype[i] = dtypes[i];
+    C.validity[i] = has_validity[i] ? (const uint8_t*)1 : nullptr;
+  }
+  std::ostringstream os;
+  char k = 'l';
+  if (!jit_emit_fn(os, "j_expr", prog, prog_len, &C, &k)) return -1;
+  std::string src = os.str();
+  if ((int64_t)src.size() + 1 > cap) return -2;
+  memcpy(buf, src.c_str(), src.size() + 1);
+  return k == 'd' ? 1 : 0;
+}
+
 // dsx_jit_selftest — hiprtc-compile a representative C2-shaped partition
 // groupby source (incl. j_scatter_staged) WITHOUT a GPU. Test harness only:
 // catches JIT codegen syntax breakage in the CPU container instead of a

@@ -353,6 +353,14 @@ int dsx_partition(DsxCtx* ctx, const uint64_t* codes, const uint8_t* validity,
                   int64_t n, int nbuckets, uint32_t* out_sel,
                   int64_t* out_offsets);
 
+/* TEST INFRASTRUCTURE: emit the hipRTC expression-evaluator source the
+ * JIT generates for `prog` (host-only, no GPU) so CPU tests can compile it
+ * with gcc and differential-test codegen semantics. Returns 1 when the
+ * expression kind is double, 0 for i64, <0 on error. */
+int dsx_jit_expr_source(const DsxInstr* prog, int prog_len,
+                        const int32_t* dtypes, const uint8_t* has_validity,
+                        int ncols, char* buf, int64_t cap);
+
 #ifdef __cplusplus
 }
 #endif
