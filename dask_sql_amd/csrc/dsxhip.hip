@@ -3407,6 +3407,34 @@ extern "C" int dsx_jit_expr_source(const DsxInstr* prog, int prog_len,
   return k == 'd' ? 1 : 0;
 }
 
+// dsx_jit_pack_source — TEST INFRASTRUCTURE (like dsx_jit_expr_source):
+// emit the JIT key-pack source for a key spec so CPU tests can gcc-compile
+// it and differential-test it against the pack_key spec. Host-only.
+// Returns 0, or <0 on error.
+extern "C" int dsx_jit_pack_source(const DsxKeySpec* keys, int nkeys,
+                                   const int32_t* dtypes,
+                                   const uint8_t* has_validity, int ncols,
+                                   char* buf, int64_t cap) {
+  if (nkeys <= 0 || nkeys > DSX_MAX_KEYS || ncols < 0 ||
+      ncols > DSX_MAX_COLS)
+    return -3;
+  KeyArg K{};
+  K.nkeys = nkeys;
+  for (int j = 0; j < nkeys; j++) K.k[j] = keys[j];
+  ColsArg C{};
+  C.ncols = ncols;
+  for (int i = 0; i < ncols; i++) {
+    C.dtype[i] = dtypes[i];
+    C.validity[i] = has_validity[i] ? (const uint8_t*)1 : nullptr;
+  }
+  std::ostringstream os;
+  jit_emit_pack(os, K, &C);
+  std::string src = os.str();
+  if ((int64_t)src.size() + 1 > cap) return -2;
+  memcpy(buf, src.c_str(), src.size() + 1);
+  return 0;
+}
+
 // dsx_jit_selftest — hiprtc-compile a representative C2-shaped partition
 // groupby source (incl. j_scatter_staged) WITHOUT a GPU. Test harness only:
 // catches JIT codegen syntax breakage in the CPU container instead of a

@@ -361,6 +361,12 @@ int dsx_jit_expr_source(const DsxInstr* prog, int prog_len,
                         const int32_t* dtypes, const uint8_t* has_validity,
                         int ncols, char* buf, int64_t cap);
 
+/* TEST INFRASTRUCTURE: emit the JIT key-pack source for a key spec
+ * (host-only, no GPU) for the CPU pack-semantics differential tests. */
+int dsx_jit_pack_source(const DsxKeySpec* keys, int nkeys,
+                        const int32_t* dtypes, const uint8_t* has_validity,
+                        int ncols, char* buf, int64_t cap);
+
 #ifdef __cplusplus
 }
 #endif
