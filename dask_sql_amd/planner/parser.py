@@ -277,6 +277,16 @@ class Parser:
                 s.group_by.append(self.expr())
         if self.accept_kw("HAVING"):
             s.having = self.expr()
+        if self._accept_word("DISTRIBUTE"):
+            # DISTRIBUTE BY <cols>: a partitioning hint in the reference
+            # (rel/custom/distributeby.py → dask shuffle). Row content is
+            # unchanged; in one process the shuffle is the identity, and the
+            # RCCL exchange path partitions explicitly (distributed.py) —
+            # parse and accept, keep the plan unchanged.
+            self.expect_kw("BY")
+            self.expr()
+            while self.accept_op(","):
+                self.expr()
         if self.accept_kw("ORDER"):
             self.expect_kw("BY")
             while True:
@@ -597,6 +607,23 @@ class Parser:
                 return ("cast", e, ty)
         if t[0] == "id":
             name = self._name()
+            if name.upper() == "DECIMAL" and self.peek()[0] == "str":
+                # DECIMAL '100.2' typed literal (reference maps DECIMAL to
+                # float64, mappings.py SqlTypeName.DECIMAL)
+                return ("lit", float(self.next()[1]), "DOUBLE")
+            if name.upper() == "TIME" and self.peek()[0] == "str":
+                # TIME 'HH:MM:SS[.fff]' → epoch-day timestamp (reference
+                # test_rex.py maps TIME to 1970-01-01 + time-of-day)
+                txt = self.next()[1].strip()
+                hh, mm, ss = txt.split(":")
+                ns = (int(hh) * 3600 + int(mm) * 60) * 1_000_000_000
+                if "." in ss:
+                    sec, frac = ss.split(".")
+                    ns += int(sec) * 1_000_000_000
+                    ns += int((frac + "0" * 9)[:9])
+                else:
+                    ns += int(ss) * 1_000_000_000
+                return ("lit", ns, "TIMESTAMP")
             if name.upper() == "TRIM" and self.peek() == ("op", "("):
                 # TRIM([LEADING|TRAILING|BOTH] ['ch'] FROM x) | TRIM(x)
                 self.next()

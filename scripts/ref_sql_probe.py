@@ -50,9 +50,11 @@ c.create_table("datetime_table", pd.DataFrame({
     "utc_timezone": pd.date_range("2014-08-01", periods=3, freq="h"),
 }))
 c.create_table("df1", pd.DataFrame(
-    {"a": [1, 2, 3], "b": [4, 5, 6], "id": [1, 2, 3]}))
+    {"id": [1, 2], "a": [1, 2], "b": pd.Series(["w", "x"]
+                                               ).astype("category")}))
 c.create_table("df2", pd.DataFrame(
-    {"a": [1, 2, 3], "b": [4, 5, 6], "id": [1, 2, 3]}))
+    {"id": [1, 2], "c": [2, 3], "d": pd.Series(["h", "i"]
+                                               ).astype("category")}))
 c.create_table("df_1", pd.DataFrame({"id": [1, 2, 3]}))
 c.create_table("df_2", pd.DataFrame({"id": [2, 3, 4]}))
 c.create_table("dates", pd.DataFrame(

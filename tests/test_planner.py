@@ -338,6 +338,9 @@ PLAN_BATTERY = [
     "SELECT a, REGR_COUNT(b, a) AS n, REGR_SXX(b, a) AS sxx, "
     "COVAR_POP(b, a) AS cp, COVAR_SAMP(b, a) AS cs FROM t GROUP BY a",
     "SELECT a, b, FROM t",
+    "SELECT * FROM t DISTRIBUTE BY a",
+    "SELECT a FROM t WHERE CAST(b AS DECIMAL) < DECIMAL '100.2'",
+    "SELECT TIME '08:08:00.091' AS tm FROM t",
 ]
 
 
