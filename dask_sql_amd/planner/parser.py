@@ -106,7 +106,7 @@ def tokenize(sql: str):
 
 # words that terminate an implicit-alias position (reserved in standard SQL
 # but lexed as identifiers here)
-_NON_ALIAS = {"INTERSECT", "EXCEPT"}
+_NON_ALIAS = {"INTERSECT", "EXCEPT", "DISTRIBUTE"}
 
 AGG_FUNCS = {"SUM", "COUNT", "AVG", "MIN", "MAX", "ANY_VALUE", "STDDEV",
              "STDDEV_POP", "STDDEV_SAMP", "VAR_SAMP", "VAR_POP", "VARIANCE",
