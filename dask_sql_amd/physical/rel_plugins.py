@@ -435,14 +435,16 @@ class DaskJoinPlugin(BaseRelPlugin):
         # RADIX PATH first: once the flat probe table
         # outgrows the XCD L2, partition both sides and probe LDS-resident
         # bucket tables instead (dsx_radix_join)
-        if (lhs_on and not residual and len(mat_idx) <= 16
+        null_eq = bool(getattr(join, "null_equal", False))
+        if (lhs_on and not residual and not null_eq and len(mat_idx) <= 16
                 and not _os.environ.get("DSX_DISABLE_RADIX")
                 and join_type in ("inner", "left", "right", "leftanti")):
             gathered, n_out = self._radix_join(
                 runtime, dc_lhs, dc_rhs, lhs_on, rhs_on, join_type,
                 combined, mat_idx, cc_lhs, cc_rhs)
         if (gathered is None
-                and lhs_on and not residual and len(mat_idx) <= 16
+                and lhs_on and not residual and not null_eq
+                and len(mat_idx) <= 16
                 and not _os.environ.get("DSX_DISABLE_JOINFUSE")
                 and join_type in ("inner", "left", "right", "leftanti")):
             gathered, n_out = self._equi_join_fused(
@@ -461,7 +463,8 @@ class DaskJoinPlugin(BaseRelPlugin):
                     residual = []
                 else:
                     pairs, n_out = self._equi_join(runtime, dc_lhs, dc_rhs,
-                                                   lhs_on, rhs_on, join_type)
+                                                   lhs_on, rhs_on, join_type,
+                                                   null_equal=null_eq)
             else:
                 pairs, n_out = self._cross_join(runtime, dc_lhs, dc_rhs,
                                                 join_type)
@@ -687,7 +690,8 @@ class DaskJoinPlugin(BaseRelPlugin):
             keep = vcol
         return codes, validity_ptr, keep, space
 
-    def _equi_join(self, runtime, dc_lhs, dc_rhs, lhs_on, rhs_on, join_type):
+    def _equi_join(self, runtime, dc_lhs, dc_rhs, lhs_on, rhs_on, join_type,
+                   null_equal=False):
         lcols = dc_lhs.backend_cols()
         rcols = dc_rhs.backend_cols()
         for i in lhs_on:
@@ -735,7 +739,10 @@ class DaskJoinPlugin(BaseRelPlugin):
         # merge how="outer" matches NA keys (join.py:202-213 drops NULL keys
         # only for inner/left/right/semi). ADVICE r1 (medium).
         null_flags = None
-        if join_type == "outer":
+        if join_type == "outer" or null_equal:
+            # null_equal: INTERSECT/EXCEPT semi/anti joins use DataFusion's
+            # null_equals_null — NULL keys pack into their own slot and
+            # match each other instead of being dropped
             pkc, bkc = probe_dc.backend_cols(), build_dc.backend_cols()
             null_flags = [
                 bool(pkc[pi].validity) or bool(bkc[bi].validity)

@@ -914,8 +914,12 @@ class Builder:
                     cond = eq if cond is None else Call(
                         "AND", [cond, eq], SqlType("BOOLEAN"))
                 jt = "LEFTSEMI" if op == "INTERSECT" else "LEFTANTI"
+                jnode = JoinNode(jt, cond)
+                # DataFusion rewrites Intersect/Except with
+                # null_equals_null=true: NULL keys compare equal
+                jnode.null_equal = True
                 plan = LogicalPlan("Join", [plan, rhs], RelDataType(lf),
-                                   JoinNode(jt, cond))
+                                   jnode)
                 gexprs = [InputRef(i, f.getType()) for i, f in enumerate(lf)]
                 node = AggregateNode(gexprs, [], distinct_node=True,
                                      distinct_columns=[f.getName()

@@ -1909,6 +1909,26 @@ def test_cte_multiple_refs(ctx):
         sorted(exp.tolist())
 
 
+def test_intersect_except_nulls(ctx):
+    # NULL keys compare EQUAL in set operations (DataFusion rewrites
+    # Intersect/Except with null_equals_null=true)
+    a = pd.DataFrame({"v": pd.array([1, None, 2], dtype="Int64")})
+    b = pd.DataFrame({"v": pd.array([None, 2, 5], dtype="Int64")})
+    ctx.create_table("tsn_a", a)
+    ctx.create_table("tsn_b", b)
+    out = ctx.sql("SELECT v FROM tsn_a INTERSECT SELECT v FROM tsn_b"
+                  ).compute()
+    got = sorted(out["v"].tolist(), key=lambda x: (x is not None
+                                                   and not pd.isna(x), x))
+    assert len(got) == 2  # NULL and 2
+    assert any(pd.isna(x) for x in got)
+    assert 2 in [x for x in got if not pd.isna(x)]
+    out = ctx.sql("SELECT v FROM tsn_a EXCEPT SELECT v FROM tsn_b"
+                  ).compute()
+    vals = out["v"].tolist()
+    assert len(vals) == 1 and vals[0] == 1  # NULL matched, 2 matched
+
+
 def test_intersect_except(ctx):
     a = pd.DataFrame({"v": np.array([1, 2, 2, 3, 4, 7], dtype=np.int64)})
     b = pd.DataFrame({"v": np.array([2, 3, 3, 5], dtype=np.int64)})
