@@ -121,12 +121,17 @@ class Parser:
 
     # -- token helpers ------------------------------------------------------
     def peek(self):
-        return self.toks[self.i]
+        return self.toks[min(self.i, len(self.toks) - 1)]
 
     def next(self):
+        if self.i >= len(self.toks):
+            raise ValueError("unexpected end of SQL")
         t = self.toks[self.i]
         self.i += 1
         return t
+
+    # peek() never walks past the trailing ("eof", "") sentinel
+
 
     def accept_kw(self, *kws):
         t = self.peek()

@@ -678,3 +678,16 @@ def test_order_by_hidden_column_plan():
     rel2 = c._get_ral("SELECT a FROM t ORDER BY b LIMIT 1")
     assert rel2.get_current_node_type() == "Projection"
     assert rel2.get_inputs()[0].get_current_node_type() == "Limit"
+
+
+def test_parser_truncation_errors_are_clean():
+    """Every truncated prefix of every battery query must fail with a
+    clean ValueError-family error, never IndexError/AttributeError (the
+    reference surfaces ParsingException the same way)."""
+    from dask_sql_amd.planner.parser import Parser
+    for q in PLAN_BATTERY:
+        for cut in range(len(q)):
+            try:
+                Parser(q[:cut]).parse()
+            except (ValueError, NotImplementedError, KeyError):
+                pass
