@@ -124,9 +124,10 @@ def harness():
     rng = np.random.default_rng(4321)
     n_rows = 40
     cols = _make_cols(rng, n_rows)
-    exprs = []
-    sources = []
-    kinds = []
+    # overwrite col 0 with day-scale / ns-scale values so the datetime
+    # programs below see plausible calendar inputs (dense int64 col)
+    cols[0] = (rng.integers(-30000, 60000, n_rows).astype(np.int64), None)
+    progs = []
     for k in range(120):
         e = gen(rng, "BOOLEAN" if k % 2 == 0 else "NUM", 4)
         c = R.RexCompiler(_dev_cols())
@@ -134,14 +135,29 @@ def harness():
             c.compile(e)
         except R.RexCompileError:
             continue
-        src, kind = _emit_source(lib, c.prog)
+        progs.append((e, c.prog))
+    # targeted datetime programs: jit_civil vs the VM model's civil math
+    import types
+    from dask_sql_amd.planner.plan import Call, InputRef, SqlType
+    for op_name in ("YEAR", "MONTH", "DAY", "EXTRACT_DOY", "EXTRACT_DOW",
+                    "EXTRACT_QUARTER", "FLOOR_TO_MONTH", "FLOOR_TO_YEAR",
+                    "EXTRACT_CENTURY"):
+        e = Call(op_name, [InputRef(0, SqlType("DATE"))], SqlType("DATE"))
+        c = R.RexCompiler(_dev_cols())
+        c.compile(e)
+        progs.append((e, c.prog))
+    exprs = []
+    sources = []
+    kinds = []
+    for e, prog in progs:
+        src, kind = _emit_source(lib, prog)
         if src is None:
             continue
         i = len(sources)
         sources.append(_devicify_casts(src).replace("j_expr",
                                                     f"j_expr_{i}"))
         kinds.append(kind)
-        exprs.append((e, c.prog))
+        exprs.append((e, prog))
 
     td = Path(tempfile.mkdtemp(prefix="dsx_jitdiff_"))
     calls = []
