@@ -190,3 +190,87 @@ def test_regr_covar_family(ctx):
         assert abs(out.loc[i, "syy"] - syy) < 1e-6 * max(1, abs(syy))
         assert abs(out.loc[i, "cp"] - cp) < 1e-6 * max(1, abs(cp))
         assert abs(out.loc[i, "cs"] - cs) < 1e-6 * max(1, abs(cs))
+
+
+# ---- reference integration cases ported verbatim ---------------------------
+def test_ref_sort_with_nan_matrix(ctx):
+    """reference test_sort.py:92 — full NULLS FIRST/LAST × ASC/DESC matrix
+    over a float key with NaN (float sort keys take the stable host path)."""
+    df = pd.DataFrame(
+        {"a": [1, 2, float("nan"), 2],
+         "b": [4, float("nan"), 5, float("inf")]})
+    ctx.create_table("zz_sortnan", df)
+    cases = [
+        ("ORDER BY a", [1, 2, 2, None], [4, None, float("inf"), 5]),
+        ("ORDER BY a NULLS FIRST", [None, 1, 2, 2],
+         [5, 4, None, float("inf")]),
+        ("ORDER BY a NULLS LAST", [1, 2, 2, None],
+         [4, None, float("inf"), 5]),
+        ("ORDER BY a DESC", [None, 2, 2, 1], [5, None, float("inf"), 4]),
+        ("ORDER BY a DESC NULLS LAST", [2, 2, 1, None],
+         [None, float("inf"), 4, 5]),
+    ]
+    for tail, ea, eb in cases:
+        out = ctx.sql(f"SELECT * FROM zz_sortnan {tail}").compute()
+        got_a = out["a"].tolist()
+        got_b = out["b"].tolist()
+        for g, w in zip(got_a, ea):
+            if w is None:
+                assert np.isnan(g), (tail, got_a)
+            else:
+                assert g == w, (tail, got_a)
+        for g, w in zip(got_b, eb):
+            if w is None:
+                assert np.isnan(g), (tail, got_b)
+            else:
+                assert g == w, (tail, got_b)
+
+
+def test_ref_sort_strings(ctx):
+    # reference test_sort.py:280
+    df = pd.DataFrame({"a": [1, 2, 3],
+                       "b": pd.Series(["zzhsd", "öfjdf", "baba"]
+                                      ).astype("category")})
+    ctx.create_table("zz_sortstr", df)
+    out = ctx.sql("SELECT * FROM zz_sortstr ORDER BY b").compute()
+    assert out["a"].astype(np.int64).tolist() == [3, 2, 1]
+
+
+def test_ref_sort_by_alias(ctx):
+    # reference test_sort.py:73 — ORDER BY a projected alias
+    df = pd.DataFrame({"user_id": [3, 1, 2], "b": [1.0, 2.0, 3.0]})
+    ctx.create_table("zz_alias", df)
+    out = ctx.sql("SELECT b AS my_column FROM zz_alias "
+                  "ORDER BY my_column DESC").compute()
+    assert out["my_column"].tolist() == [3.0, 2.0, 1.0]
+
+
+def test_ref_string_filter_like(ctx):
+    # reference test_filter.py:62 — LIKE with regex metachars in the data
+    df = pd.DataFrame({"a": pd.Series(
+        ["a normal string", "%_%", "^|()-*[]$"]).astype("category")})
+    ctx.create_table("zz_strf", df)
+    out = ctx.sql("SELECT * FROM zz_strf WHERE a LIKE '%n%'").compute()
+    assert out["a"].tolist() == ["a normal string"]
+    out = ctx.sql("SELECT * FROM zz_strf WHERE a LIKE '%|%'").compute()
+    assert out["a"].tolist() == ["^|()-*[]$"]
+
+
+def test_ref_filter_year(ctx):
+    # reference test_filter.py:130
+    df = pd.DataFrame({"dt": pd.to_datetime(
+        ["2021-01-02", "2022-03-04", "2023-05-06"]), "x": [1, 2, 3]})
+    ctx.create_table("zz_fy", df)
+    out = ctx.sql("SELECT x FROM zz_fy WHERE year(dt) < 2023").compute()
+    assert sorted(out["x"].astype(np.int64).tolist()) == [1, 2]
+
+
+def test_ref_filter_scalar_literals(ctx):
+    # reference test_filter.py:20 — constant predicates
+    df = _rand_frame(np.random.default_rng(5), 100, with_nulls=False)
+    ctx.create_table("zz_fs", df)
+    assert len(ctx.sql("SELECT * FROM zz_fs WHERE True").compute()) == 100
+    assert len(ctx.sql("SELECT * FROM zz_fs WHERE False").compute()) == 0
+    assert len(ctx.sql("SELECT * FROM zz_fs WHERE (1 = 1)").compute()) \
+        == 100
+    assert len(ctx.sql("SELECT * FROM zz_fs WHERE (1 = 0)").compute()) == 0
