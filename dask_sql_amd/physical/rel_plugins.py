@@ -1036,6 +1036,11 @@ class DaskJoinPlugin(BaseRelPlugin):
         return lhs_idx, rhs_idx, residual
 
 
+# aggregates the device kernels do not carry; the reference itself runs
+# them as pandas reductions (custom dask Aggregations)
+_HOST_ONLY_AGGS = {"bit_and", "bit_or", "bit_xor"}
+
+
 class DaskAggregatePlugin(BaseRelPlugin):
     """reference rel/logical/aggregate.py:91-589.
 
@@ -1059,6 +1064,8 @@ class DaskAggregatePlugin(BaseRelPlugin):
     # sample forms (pandas default), ddof=0 for *_POP.
     STD_FINS = {"std_samp", "std_pop", "var_samp", "var_pop"}
 
+    _HOST_ONLY_AGGS = frozenset()  # set below the class (shared)
+
     def _host_udf_aggregate(self, rel, agg, context):
         """Registered aggregate UDFs run as Python on host frames — the
         reference's own execution model for register_aggregation
@@ -1080,10 +1087,37 @@ class DaskAggregatePlugin(BaseRelPlugin):
         gb_keys = keys if keys else ["__const_1__"]
         field_names = [str(f) for f in rel.getRowType().getFieldNames()]
         results = {}
+        def _bitred(np_op):
+            def f(sg):
+                return sg.agg(lambda s: np_op.reduce(
+                    s.dropna().astype(np.int64))
+                    if s.notna().any() else None)
+            return f
+
         builtin = {"sum": lambda sg: sg.sum(min_count=1),
                    "count": lambda sg: sg.count(),
                    "avg": lambda sg: sg.mean(),
-                   "min": lambda sg: sg.min(), "max": lambda sg: sg.max()}
+                   "min": lambda sg: sg.min(), "max": lambda sg: sg.max(),
+                   "bit_and": _bitred(np.bitwise_and),
+                   "bit_or": _bitred(np.bitwise_or),
+                   "bit_xor": _bitred(np.bitwise_xor),
+                   "every": lambda sg: sg.agg(
+                       lambda s: bool(s.dropna().astype(bool).all())
+                       if s.notna().any() else None),
+                   "bool_and": lambda sg: sg.agg(
+                       lambda s: bool(s.dropna().astype(bool).all())
+                       if s.notna().any() else None),
+                   "bool_or": lambda sg: sg.agg(
+                       lambda s: bool(s.dropna().astype(bool).any())
+                       if s.notna().any() else None),
+                   "single_value": lambda sg: sg.first(),
+                   "any_value": lambda sg: sg.first(),
+                   "stddev": lambda sg: sg.std(),
+                   "stddev_samp": lambda sg: sg.std(),
+                   "stddev_pop": lambda sg: sg.std(ddof=0),
+                   "var_samp": lambda sg: sg.var(),
+                   "variance": lambda sg: sg.var(),
+                   "var_pop": lambda sg: sg.var(ddof=0)}
         for pos, call in enumerate(agg.getNamedAggCalls()):
             func = agg.getAggregationFuncName(call).lower()
             args = agg.getArgs(call)
@@ -1144,10 +1178,16 @@ class DaskAggregatePlugin(BaseRelPlugin):
         # and the projection arithmetic folded into each agg program. This is
         # what replaces the reference's filter→assign→groupby pass chain with
         # a single HBM scan.
-        if context.catalog.aggregations and not agg.isDistinctNode():
+        if not agg.isDistinctNode():
             calls = agg.getNamedAggCalls()
             if any(agg.getAggregationFuncName(c).lower().startswith("udf:")
+                   or agg.getAggregationFuncName(c).lower() in _HOST_ONLY_AGGS
                    for c in calls):
+                # registered UDF aggregates, and the bitwise reductions the
+                # device kernels don't carry — the reference computes BOTH
+                # as custom dask Aggregations on pandas (rel/custom/
+                # wrappers ReduceAggregation bit_and/bit_or), so host
+                # evaluation IS the reference execution model
                 return self._host_udf_aggregate(rel, agg, context)
 
         import os

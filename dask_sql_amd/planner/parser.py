@@ -110,7 +110,8 @@ _NON_ALIAS = {"INTERSECT", "EXCEPT", "DISTRIBUTE"}
 
 AGG_FUNCS = {"SUM", "COUNT", "AVG", "MIN", "MAX", "ANY_VALUE", "STDDEV",
              "STDDEV_POP", "STDDEV_SAMP", "VAR_SAMP", "VAR_POP", "VARIANCE",
-             "SINGLE_VALUE", "EVERY", "BOOL_AND", "BOOL_OR"}
+             "SINGLE_VALUE", "EVERY", "BOOL_AND", "BOOL_OR",
+             "BIT_AND", "BIT_OR", "BIT_XOR"}
 
 
 class Parser:

@@ -274,3 +274,44 @@ def test_ref_filter_scalar_literals(ctx):
     assert len(ctx.sql("SELECT * FROM zz_fs WHERE (1 = 1)").compute()) \
         == 100
     assert len(ctx.sql("SELECT * FROM zz_fs WHERE (1 = 0)").compute()) == 0
+
+
+def test_ref_aggregations_bit_every_single(ctx):
+    # reference test_groupby.py:205 — EVERY/BIT_AND/BIT_OR/MIN/
+    # SINGLE_VALUE/AVG over user_table_1/2 with the exact expected frames
+    u1 = pd.DataFrame({"user_id": [2, 1, 2, 3], "b": [3, 3, 1, 3]})
+    u2 = pd.DataFrame({"user_id": [1, 1, 2, 4], "c": [1, 2, 3, 4]})
+    ctx.create_table("zz_u1", u1)
+    ctx.create_table("zz_u2", u2)
+    out = ctx.sql(
+        "SELECT user_id, EVERY(b = 3) AS e, BIT_AND(b) AS b, "
+        "BIT_OR(b) AS bb, MIN(b) AS m, SINGLE_VALUE(b) AS s, AVG(b) AS a "
+        "FROM zz_u1 GROUP BY user_id").compute()
+    out = out.sort_values("user_id").reset_index(drop=True)
+    assert out["user_id"].astype(np.int64).tolist() == [1, 2, 3]
+    assert out["e"].astype(bool).tolist() == [True, False, True]
+    assert out["b"].astype(np.int64).tolist() == [3, 1, 3]
+    assert out["bb"].astype(np.int64).tolist() == [3, 3, 3]
+    assert out["m"].astype(np.int64).tolist() == [3, 1, 3]
+    assert out["a"].astype(float).tolist() == [3.0, 2.0, 3.0]
+    out = ctx.sql(
+        "SELECT user_id, EVERY(c = 3) AS e, BIT_AND(c) AS b, "
+        "BIT_OR(c) AS bb, MIN(c) AS m, AVG(c) AS a "
+        "FROM zz_u2 GROUP BY user_id").compute()
+    out = out.sort_values("user_id").reset_index(drop=True)
+    assert out["user_id"].astype(np.int64).tolist() == [1, 2, 4]
+    assert out["e"].astype(bool).tolist() == [False, True, False]
+    assert out["b"].astype(np.int64).tolist() == [0, 3, 4]
+    assert out["bb"].astype(np.int64).tolist() == [3, 3, 4]
+    assert out["a"].astype(float).tolist() == [1.5, 3.0, 4.0]
+
+
+def test_ref_groupby_string_minmax(ctx):
+    # reference test_groupby.py:205 tail — MIN/MAX over a string column
+    df = pd.DataFrame({"a": pd.Series(
+        ["a normal string", "%_%", "^|()-*[]$"]).astype("category")})
+    ctx.create_table("zz_sminmax", df)
+    out = ctx.sql('SELECT MAX(a) AS "max", MIN(a) AS "min" '
+                  "FROM zz_sminmax").compute()
+    assert out["max"].tolist() == ["a normal string"]
+    assert out["min"].tolist() == ["%_%"]
