@@ -1180,9 +1180,24 @@ class DaskAggregatePlugin(BaseRelPlugin):
         # a single HBM scan.
         if not agg.isDistinctNode():
             calls = agg.getNamedAggCalls()
-            if any(agg.getAggregationFuncName(c).lower().startswith("udf:")
-                   or agg.getAggregationFuncName(c).lower() in _HOST_ONLY_AGGS
-                   for c in calls):
+
+            def _is_host_only(c):
+                fn = agg.getAggregationFuncName(c).lower()
+                if fn.startswith("udf:") or fn in _HOST_ONLY_AGGS:
+                    return True
+                if fn in ("min", "max", "single_value", "any_value"):
+                    # string (dict-encoded) MIN/MAX must compare by STRING
+                    # order, not appearance-order codes — pandas does this
+                    # on host exactly like the reference
+                    args = agg.getArgs(c)
+                    if args and isinstance(args[0], InputRef):
+                        in_fields = rel.get_inputs()[0].getRowType()                             .getFieldList()
+                        i = args[0].getIndex()
+                        if i < len(in_fields) and in_fields[i].getType()                                 .getSqlType() == "VARCHAR":
+                            return True
+                return False
+
+            if any(_is_host_only(c) for c in calls):
                 # registered UDF aggregates, and the bitwise reductions the
                 # device kernels don't carry — the reference computes BOTH
                 # as custom dask Aggregations on pandas (rel/custom/
