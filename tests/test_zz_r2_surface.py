@@ -315,3 +315,37 @@ def test_ref_groupby_string_minmax(ctx):
                   "FROM zz_sminmax").compute()
     assert out["max"].tolist() == ["a normal string"]
     assert out["min"].tolist() == ["%_%"]
+
+
+def test_ref_union_cases(ctx):
+    # reference test_union.py — UNION dedups, UNION ALL concatenates,
+    # mixed-alias branches align positionally
+    rng = np.random.default_rng(9)
+    df = pd.DataFrame({"a": [1.0] * 10 + [2.0] * 20 + [3.0] * 40,
+                       "b": np.round(10 * rng.random(70), 3)})
+    lt = pd.DataFrame({"a": [0] * 10 + [1] * 11 + [2] * 13})
+    ctx.create_table("zz_udf", df)
+    ctx.create_table("zz_ult", lt)
+    out = ctx.sql("SELECT * FROM zz_udf UNION SELECT * FROM zz_udf "
+                  "UNION SELECT * FROM zz_udf").compute()
+    exp = df.drop_duplicates()
+    assert len(out) == len(exp)
+    out = ctx.sql("SELECT * FROM zz_udf UNION ALL SELECT * FROM zz_udf "
+                  "UNION ALL SELECT * FROM zz_udf").compute()
+    assert len(out) == 3 * len(df)
+    out = ctx.sql('SELECT a AS "I", b AS "II" FROM zz_udf UNION ALL '
+                  'SELECT a AS "I", a AS "II" FROM zz_ult').compute()
+    exp_i = sorted(df["a"].tolist() + [float(x) for x in lt["a"]])
+    assert sorted(out["I"].astype(float).tolist()) == exp_i
+    assert len(out.columns) == 2
+
+
+def test_ref_cross_join(ctx):
+    # reference test_join.py cross join via comma-FROM with TRUE condition
+    d1 = pd.DataFrame({"a": [1, 2, 3]})
+    d2 = pd.DataFrame({"b": [10, 20]})
+    ctx.create_table("zz_c1", d1)
+    ctx.create_table("zz_c2", d2)
+    out = ctx.sql("SELECT * FROM zz_c1 CROSS JOIN zz_c2").compute()
+    assert len(out) == 6
+    assert sorted(out["a"].astype(np.int64).tolist()) == [1, 1, 2, 2, 3, 3]
