@@ -690,21 +690,88 @@ class DaskJoinPlugin(BaseRelPlugin):
             keep = vcol
         return codes, validity_ptr, keep, space
 
+    def _densify_join_float_key(self, runtime, lcol, rcol):
+        """Float equi-join keys: give every distinct f64 bit-pattern ONE
+        dense integer id, consistent across BOTH sides, by running the
+        bits-mode groupby over the concatenation and joining the ids back
+        per row (same primitive chain as _densify_float_keys). NaN keys
+        share an id — pandas merge matches NaN with NaN; NULL rows keep
+        the original validity so the normal NULL-key drop applies.
+        Returns (lid, rid, n_distinct)."""
+        n_l, n_r = lcol.len, rcol.len
+        cat = runtime.concat_columns([lcol, rcol], rt.F64)
+        oc, ov, on_, G = runtime.hash_groupby(
+            [cat], n_l + n_r, [(0, 0, 0, True, 1)], None, [])
+        runtime._free(ov)
+        runtime._free(on_)
+        bcol = rt.DeviceColumn(runtime, oc, None, max(G, 1), rt.I64,
+                               owner=True)
+        pcodes, _ = runtime.keypack([cat], [(0, 0, 0, True, 1)], n_l + n_r)
+        table = runtime.hash_build(bcol, None, code_max=0)
+        try:
+            p, b, cnt = runtime.hash_probe(table, pcodes, rt.JOIN_INNER,
+                                           None)
+            psel = runtime.wrap_sel(p, cnt)
+            bsel = runtime.wrap_sel(b, cnt)
+            assert cnt == n_l + n_r, (cnt, n_l + n_r)
+            fid = runtime.scatter_rows(bsel, psel.data, cnt, n_l + n_r,
+                                       with_validity=False)
+        finally:
+            runtime.hash_table_free(table)
+        keep = (fid, lcol, rcol)
+        lid = rt.DeviceColumn(runtime, fid.data, lcol.validity, n_l,
+                              rt.I64, owner=False, keep_alive=keep)
+        rid = rt.DeviceColumn(runtime, fid.data + 8 * n_l, rcol.validity,
+                              n_r, rt.I64, owner=False, keep_alive=keep)
+        return lid, rid, G
+
     def _equi_join(self, runtime, dc_lhs, dc_rhs, lhs_on, rhs_on, join_type,
                    null_equal=False):
         lcols = dc_lhs.backend_cols()
         rcols = dc_rhs.backend_cols()
-        for i in lhs_on:
-            if lcols[i].dtype not in _INT_KINDS:
-                raise RexCompileError("non-integer join keys (round-2)")
-        for i in rhs_on:
-            if rcols[i].dtype not in _INT_KINDS:
-                raise RexCompileError("non-integer join keys (round-2)")
+        for li, ri in zip(lhs_on, rhs_on):
+            lk, rk = lcols[li].dtype, rcols[ri].dtype
+            if (lk in _INT_KINDS) != (rk in _INT_KINDS):
+                raise RexCompileError("join key type mismatch (int vs "
+                                      "float) — cast one side first")
+            if lk == rt.F32 or rk == rt.F32:
+                raise RexCompileError("float32 join keys: cast to DOUBLE")
+            if lk not in _INT_KINDS and lk != rt.F64:
+                raise RexCompileError("unsupported join key dtype")
         dc_rhs, rcols = self._reconcile_dict_keys(runtime, dc_lhs, dc_rhs,
                                                   lhs_on, rhs_on)
+        # f64 keys densify to consistent integer ids (single OR composite)
+        float_pairs = [j for j, (li, ri) in enumerate(zip(lhs_on, rhs_on))
+                       if lcols[li].dtype == rt.F64]
+        dense_ranges = {}
+        if float_pairs:
+            lcols = list(lcols)
+            rcols = list(rcols)
+            lhs_on = list(lhs_on)
+            rhs_on = list(rhs_on)
+            for j in float_pairs:
+                lid, rid, G = self._densify_join_float_key(
+                    runtime, lcols[lhs_on[j]], rcols[rhs_on[j]])
+                lcols.append(lid)
+                rcols.append(rid)
+                lhs_on[j] = len(lcols) - 1
+                rhs_on[j] = len(rcols) - 1
+                dense_ranges[j] = (0, max(G, 1))
+            import types as _types
+            dc_lhs = _types.SimpleNamespace(
+                backend_cols=lambda c=lcols: c,
+                table=_types.SimpleNamespace(
+                    num_rows=dc_lhs.table.num_rows))
+            dc_rhs = _types.SimpleNamespace(
+                backend_cols=lambda c=rcols: c,
+                table=_types.SimpleNamespace(
+                    num_rows=dc_rhs.table.num_rows))
         # combined ranges over both sides so codes are comparable
         ranges = []
-        for li, ri in zip(lhs_on, rhs_on):
+        for j, (li, ri) in enumerate(zip(lhs_on, rhs_on)):
+            if j in dense_ranges:
+                ranges.append(dense_ranges[j])
+                continue
             lmn, lmx, lnn = _minmax_cached(runtime, lcols[li])
             rmn, rmx, rnn = _minmax_cached(runtime, rcols[ri])
             if lnn == 0 and rnn == 0:
@@ -808,10 +875,10 @@ class DaskJoinPlugin(BaseRelPlugin):
         rcols = dc_rhs.backend_cols()
         for i in lhs_on:
             if lcols[i].dtype not in _INT_KINDS:
-                raise RexCompileError("non-integer join keys (round-2)")
+                return None, 0  # float keys: pair path densifies them
         for i in rhs_on:
             if rcols[i].dtype not in _INT_KINDS:
-                raise RexCompileError("non-integer join keys (round-2)")
+                return None, 0
         kdc_rhs, krcols = self._reconcile_dict_keys(runtime, dc_lhs, dc_rhs,
                                                     lhs_on, rhs_on)
         ranges = []
@@ -908,10 +975,10 @@ class DaskJoinPlugin(BaseRelPlugin):
         rcols = dc_rhs.backend_cols()
         for i in lhs_on:
             if lcols[i].dtype not in _INT_KINDS:
-                raise RexCompileError("non-integer join keys (round-2)")
+                return None, 0  # float keys: pair path densifies them
         for i in rhs_on:
             if rcols[i].dtype not in _INT_KINDS:
-                raise RexCompileError("non-integer join keys (round-2)")
+                return None, 0
         kdc_rhs, krcols = self._reconcile_dict_keys(runtime, dc_lhs, dc_rhs,
                                                     lhs_on, rhs_on)
         ranges = []

@@ -349,3 +349,45 @@ def test_ref_cross_join(ctx):
     out = ctx.sql("SELECT * FROM zz_c1 CROSS JOIN zz_c2").compute()
     assert len(out) == 6
     assert sorted(out["a"].astype(np.int64).tolist()) == [1, 1, 2, 2, 3, 3]
+
+
+def test_float_key_join(ctx):
+    # f64 equi-join keys densify to consistent ids (bits-mode groupby over
+    # the concatenated pair); NaN matches NaN like pandas merge, NULL drops
+    lhs = pd.DataFrame({"x": [1.5, 2.5, np.nan, -0.0, 7.25],
+                        "lv": np.arange(5, dtype=np.int64)})
+    rhs = pd.DataFrame({"x": [2.5, np.nan, 0.0, 9.0],
+                        "rv": np.arange(4, dtype=np.int64)})
+    ctx.create_table("zz_fl", lhs)
+    ctx.create_table("zz_fr", rhs)
+    out = ctx.sql("SELECT l.lv, r.rv FROM zz_fl l JOIN zz_fr r "
+                  "ON l.x = r.x").compute()
+    got = sorted(zip(out["lv"].astype(np.int64),
+                     out["rv"].astype(np.int64)))
+    # pandas merge semantics: 2.5↔2.5, NaN↔NaN, -0.0↔0.0
+    assert got == [(1, 0), (2, 1), (3, 2)], got
+
+
+def test_ref_complex_query(ctx):
+    """reference test_complex.py:4 — self-join of a timeseries on
+    (name, MAX(x)) — a composite join key with a FLOAT member."""
+    rng = np.random.default_rng(13)
+    n = 5000
+    df = pd.DataFrame({
+        "name": pd.Series(rng.choice(["Alice", "Bob", "Xavier"], n)
+                          ).astype("category"),
+        "id": rng.integers(0, 100, n).astype(np.int64),
+        "x": np.round(rng.random(n) * 2 - 1, 6)})
+    ctx.create_table("zz_timeseries", df)
+    result = ctx.sql(
+        "SELECT lhs.name, lhs.id, lhs.x FROM zz_timeseries AS lhs "
+        "JOIN (SELECT name AS max_name, MAX(x) AS max_x "
+        "      FROM zz_timeseries GROUP BY name) AS rhs "
+        "ON lhs.name = rhs.max_name AND lhs.x = rhs.max_x").compute()
+    assert len(result) > 0
+    exp = df.merge(
+        df.groupby("name", observed=True)["x"].max().reset_index()
+          .rename(columns={"name": "max_name", "x": "max_x"}),
+        left_on=["name", "x"], right_on=["max_name", "max_x"])
+    assert len(result) == len(exp)
+    assert sorted(result["x"].tolist()) == sorted(exp["x"].tolist())
