@@ -545,9 +545,84 @@ class RexCompiler:
                 self._emit(OP_FLOORMOD_I64)
                 self._emit(OP_ADD_I64)
             return KI
-        if kind == "CEIL":
-            raise RexCompileError(f"CEIL TO {unit} not supported on GPU "
-                                  "path (calendar ceil)")
+        if kind == "CEIL" and unit in ("MONTH", "YEAR"):
+            # CEIL(x) = start of the period FOLLOWING (x - 1 tick): a
+            # boundary x decrements into the previous period (whose next
+            # start is x itself), any other x stays in its own period.
+            # Avoids a SELECT and its double re-emission — the stack VM
+            # has no DUP, every reuse re-emits its subsequence, and the
+            # program must stay inside DSX_MAX_PROG.
+            def emit_pred_days():
+                # day count of (x - 1ns) / (x - 1 day)
+                def v():
+                    k = self.compile(x)
+                    if k != KI:
+                        raise RexCompileError(
+                            "CEIL TO needs DATE/TIMESTAMP")
+                    self._emit(OP_LIT_I64, 0, 1)
+                    self._emit(OP_SUB_I64)
+
+                if not is_ts:
+                    v()
+                    return
+                v()
+                v()
+                self._emit(OP_LIT_I64, 0, self._DAY_NS)
+                self._emit(OP_FLOORMOD_I64)
+                self._emit(OP_SUB_I64)
+                self._emit(OP_LIT_I64, 0, self._DAY_NS)
+                self._emit(OP_DIV_I64)
+
+            if unit == "MONTH":
+                def bumped():
+                    # month start of (x-1) plus 31 — always inside (or on
+                    # the first day of) the FOLLOWING month
+                    emit_pred_days()
+                    emit_pred_days()
+                    self._emit(OP_DAY)
+                    self._emit(OP_LIT_I64, 0, 1)
+                    self._emit(OP_SUB_I64)
+                    self._emit(OP_SUB_I64)
+                    self._emit(OP_LIT_I64, 0, 31)
+                    self._emit(OP_ADD_I64)
+
+                bumped()
+                bumped()
+                self._emit(OP_DAY)
+                self._emit(OP_LIT_I64, 0, 1)
+                self._emit(OP_SUB_I64)
+                self._emit(OP_SUB_I64)
+            else:
+                # jan1_days(Y) with Y = YEAR(x-1) + 1 — the same Gregorian
+                # identity FLOOR_TO_YEAR uses, evaluated for the next year
+                def emit_y1():
+                    emit_pred_days()
+                    self._emit(OP_YEAR)
+                    self._emit(OP_LIT_I64, 0, 1)
+                    self._emit(OP_ADD_I64)
+
+                def leap1(div):
+                    emit_y1()
+                    self._emit(OP_LIT_I64, 0, 1)
+                    self._emit(OP_SUB_I64)
+                    self._emit(OP_LIT_I64, 0, div)
+                    self._emit(OP_DIV_I64)
+
+                emit_y1()
+                self._emit(OP_LIT_I64, 0, 365)
+                self._emit(OP_MUL_I64)
+                leap1(4)
+                self._emit(OP_ADD_I64)
+                leap1(100)
+                self._emit(OP_SUB_I64)
+                leap1(400)
+                self._emit(OP_ADD_I64)
+                self._emit(OP_LIT_I64, 0, 719_527)
+                self._emit(OP_SUB_I64)
+            if is_ts:
+                self._emit(OP_LIT_I64, 0, self._DAY_NS)
+                self._emit(OP_MUL_I64)
+            return KI
         if unit == "MONTH":
             # month_start_days = days - (dayofmonth(days) - 1)
             self._emit_days(x, is_ts)

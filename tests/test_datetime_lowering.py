@@ -148,11 +148,39 @@ def test_extract_date_vs_pandas():
         assert interp_prog(prog, ts.value) == want, s
 
 
-def test_calendar_ceil_rejected():
-    with pytest.raises(R.RexCompileError):
-        _compile("CEIL_TO_MONTH", "TIMESTAMP")
-    with pytest.raises(R.RexCompileError):
-        _compile("CEIL_TO_YEAR", "TIMESTAMP")
+def test_ceil_month_year_vs_pandas():
+    """Calendar CEIL = next-period start of (x - 1 tick): boundary inputs
+    are fixed points, everything else rounds up (full interpreter — the
+    lowering uses EQ-free decrement arithmetic)."""
+    from tests.vm_interp import interp
+
+    def run(prog, v):
+        got, ok = interp(prog, [(np.array([v], dtype=np.int64), None)], 0)
+        assert ok
+        return got
+
+    pm = _compile("CEIL_TO_MONTH", "TIMESTAMP")
+    py = _compile("CEIL_TO_YEAR", "TIMESTAMP")
+    for s in TS_SAMPLES:
+        ts = pd.Timestamp(s)
+        wm = ts if ts == ts.to_period("M").start_time \
+            else (ts.to_period("M") + 1).start_time
+        wy = ts if ts == ts.to_period("Y").start_time \
+            else (ts.to_period("Y") + 1).start_time
+        assert run(pm, ts.value) == wm.value, ("M", s)
+        assert run(py, ts.value) == wy.value, ("Y", s)
+    pmd = _compile("CEIL_TO_MONTH", "DATE")
+    pyd = _compile("CEIL_TO_YEAR", "DATE")
+    for s in ("2021-02-17", "2021-02-01", "2020-12-31", "1969-07-20",
+              "2000-02-29"):
+        ts = pd.Timestamp(s)
+        days = (ts - pd.Timestamp(0)).days
+        wm = ts if ts == ts.to_period("M").start_time \
+            else (ts.to_period("M") + 1).start_time
+        wy = ts if ts == ts.to_period("Y").start_time \
+            else (ts.to_period("Y") + 1).start_time
+        assert run(pmd, days) == (wm - pd.Timestamp(0)).days, ("Md", s)
+        assert run(pyd, days) == (wy - pd.Timestamp(0)).days, ("Yd", s)
 
 
 def test_extract_extended_vs_reference_semantics():

@@ -380,7 +380,7 @@ def test_plan_battery_dates():
         "SELECT TIMESTAMPADD(DAY, 5, ts) AS ts2 FROM t",
         "SELECT TIMESTAMPADD(HOUR, -3, ts) AS ts3 FROM t",
         "SELECT FLOOR(ts TO DAY), CEIL(ts TO HOUR), FLOOR(ts TO YEAR), "
-        "FLOOR(d TO MONTH) FROM t",
+        "FLOOR(d TO MONTH), CEIL(ts TO MONTH), CEIL(d TO YEAR) FROM t",
         "SELECT EXTRACT(DATE FROM ts) AS dt FROM t",
         "SELECT EXTRACT(CENTURY FROM ts), EXTRACT(DOW FROM d), "
         "EXTRACT(DOY FROM ts), EXTRACT(QUARTER FROM d), "
