@@ -104,6 +104,28 @@ def dict_string_fn(expr, dicts):
             return i, g
         if op == "INITCAP" and len(ops_) == 1:
             return i, (lambda s, f=f: f(s).title())
+        if op == "OVERLAY" and len(ops_) in (3, 4):
+            # OVERLAY(x PLACING y FROM n [FOR m]) — reference
+            # rex/core/call.py OverlayOperation (1-based start; start<=0
+            # clamps to 0; splice length defaults to len(y))
+            if not (isinstance(ops_[1], Literal)
+                    and isinstance(ops_[1].getValue(), str)):
+                return None
+            start = _lit_int(ops_[2])
+            if start is None:
+                return None
+            length = _lit_int(ops_[3]) if len(ops_) == 4 else None
+            if len(ops_) == 4 and length is None:
+                return None
+            repl = ops_[1].getValue()
+
+            def ov(s, f=f, repl=repl, start=start, length=length):
+                t = f(s)
+                st = 0 if start <= 0 else start - 1
+                ln = len(repl) if length is None else length
+                return t[:st] + repl + t[st + ln:]
+
+            return i, ov
         if op == "TRIM" and len(ops_) == 3:
             if not all(isinstance(o, Literal) for o in ops_[1:]):
                 return None

@@ -211,7 +211,7 @@ class Builder:
             elif op == "NEG":
                 ty = _expr_type(ops[0])
             elif op in ("UPPER", "LOWER", "SUBSTRING", "SUBSTR", "CONCAT",
-                        "TRIM", "REPLACE", "INITCAP"):
+                        "TRIM", "REPLACE", "INITCAP", "OVERLAY"):
                 ty = "VARCHAR"
             elif op in ("CHAR_LENGTH", "CHARACTER_LENGTH", "LENGTH",
                         "POSITION"):

@@ -777,6 +777,17 @@ class Parser:
                 args.append(self.expr())
             self.expect_op(")")
             return ("call", "POSITION", args)
+        if fname == "OVERLAY":
+            # OVERLAY(x PLACING y FROM n [FOR m]) — Calcite form
+            x_ = self.expr()
+            self._expect_word("PLACING")
+            y_ = self.add_expr()
+            self._expect_word("FROM")
+            args = [x_, y_, self.add_expr()]
+            if self._accept_word("FOR"):
+                args.append(self.add_expr())
+            self.expect_op(")")
+            return ("call", "OVERLAY", args)
         if fname in ("FLOOR", "CEIL", "CEILING"):
             # FLOOR(x TO DAY) — datetime truncation form
             e = self.expr()

@@ -386,6 +386,7 @@ def test_plan_battery_dates():
         "EXTRACT(DOY FROM ts), EXTRACT(QUARTER FROM d), "
         "EXTRACT(MILLISECOND FROM ts), EXTRACT(DECADE FROM ts) FROM t",
         "SELECT POSITION('x' IN s) AS p, POSITION('x' IN s FROM 2) FROM t",
+        "SELECT OVERLAY(s PLACING 'XX' FROM 2 FOR 3) AS o FROM t",
         "SELECT ts + INTERVAL '1' HOUR, ts - INTERVAL '30' MINUTE FROM t",
         "SELECT to_timestamp(v) AS t1, "
         "to_timestamp('2021-03-02 10:00:00') AS t2, "

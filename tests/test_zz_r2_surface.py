@@ -391,3 +391,23 @@ def test_ref_complex_query(ctx):
         left_on=["name", "x"], right_on=["max_name", "max_x"])
     assert len(result) == len(exp)
     assert sorted(result["x"].tolist()) == sorted(exp["x"].tolist())
+
+
+def test_ref_overlay(ctx):
+    # reference test_rex.py OVERLAY rows (the exact expected strings from
+    # OverlayOperation semantics)
+    df = pd.DataFrame({"a": pd.Series(["a normal string"]
+                                      ).astype("category")})
+    ctx.create_table("zz_ov", df)
+    out = ctx.sql(
+        "SELECT OVERLAY(a PLACING 'XXX' FROM -1) AS l, "
+        "OVERLAY(a PLACING 'XXX' FROM 2 FOR 4) AS m, "
+        "OVERLAY(a PLACING 'XXX' FROM 2 FOR 1) AS n FROM zz_ov").compute()
+    s = "a normal string"
+    def ov(s, repl, start, length=None):
+        st = 0 if start <= 0 else start - 1
+        ln = len(repl) if length is None else length
+        return s[:st] + repl + s[st + ln:]
+    assert out["l"].tolist() == [ov(s, "XXX", -1)]
+    assert out["m"].tolist() == [ov(s, "XXX", 2, 4)]
+    assert out["n"].tolist() == [ov(s, "XXX", 2, 1)]
