@@ -281,8 +281,11 @@ class Builder:
         subplan = LogicalPlan("Distinct", [subplan], subplan.getRowType(),
                               node)
         lhs_fields = plan.getRowType().getFieldList()
-        tmp = LogicalPlan("__combined__", [],
-                          RelDataType(lhs_fields + sfields), None)
+        # outer key asts resolve in the OUTER scope only — a subplan output
+        # sharing the column name (IN over a grouped projection of the same
+        # table) must not make it ambiguous
+        tmp = LogicalPlan("__combined__", [], RelDataType(lhs_fields),
+                          None)
         cond = None
         for i, ast in enumerate(outer_asts):
             eq = Call("=", [self._resolve(ast, tmp),

@@ -411,3 +411,108 @@ def test_ref_overlay(ctx):
     assert out["l"].tolist() == [ov(s, "XXX", -1)]
     assert out["m"].tolist() == [ov(s, "XXX", 2, 4)]
     assert out["n"].tolist() == [ov(s, "XXX", 2, 1)]
+
+
+# ---- TPC-H shapes unlocked by round-2 planner work -------------------------
+@pytest.fixture(scope="module")
+def tpch2(ctx):
+    rng = np.random.default_rng(21)
+    n = 30_000
+    lineitem = pd.DataFrame({
+        "l_suppkey": rng.integers(0, 200, n).astype(np.int64),
+        "l_partkey": rng.integers(0, 400, n).astype(np.int64),
+        "l_quantity": rng.integers(1, 50, n).astype(np.int64),
+        "l_extendedprice": np.round(rng.random(n) * 1000, 2),
+        "l_discount": np.round(rng.random(n) * 0.1, 2),
+        "l_shipdate": pd.to_datetime("1995-01-01")
+        + pd.to_timedelta(rng.integers(0, 700, n), unit="D"),
+    })
+    supplier = pd.DataFrame({
+        "s_suppkey": np.arange(200, dtype=np.int64),
+        "s_nationkey": rng.integers(0, 25, 200).astype(np.int64)})
+    part = pd.DataFrame({
+        "p_partkey": np.arange(400, dtype=np.int64),
+        "p_brand": pd.Series(rng.choice(["Brand#1", "Brand#2", "Brand#3"],
+                                        400)).astype("category"),
+        "p_size": rng.integers(1, 50, 400).astype(np.int64)})
+    ctx.create_table("t2_lineitem", lineitem)
+    ctx.create_table("t2_supplier", supplier)
+    ctx.create_table("t2_part", part)
+    return lineitem, supplier, part
+
+
+def test_q15_shape_cte_scalar_max(ctx, tpch2):
+    """Q15: CTE revenue view + uncorrelated scalar MAX subquery over it."""
+    lineitem, supplier, part = tpch2
+    out = ctx.sql(
+        "WITH revenue AS ("
+        "  SELECT l_suppkey AS supplier_no, "
+        "         SUM(l_extendedprice * (1 - l_discount)) AS total_rev "
+        "  FROM t2_lineitem "
+        "  WHERE l_shipdate >= DATE '1995-06-01' "
+        "    AND l_shipdate < DATE '1995-06-01' + INTERVAL '3' MONTH "
+        "  GROUP BY l_suppkey) "
+        "SELECT s_suppkey, total_rev FROM t2_supplier "
+        "JOIN revenue ON s_suppkey = supplier_no "
+        "WHERE total_rev = (SELECT MAX(total_rev) FROM revenue)").compute()
+    li = lineitem[(lineitem.l_shipdate >= "1995-06-01")
+                  & (lineitem.l_shipdate < "1995-09-01")]
+    rev = li.assign(r=li.l_extendedprice * (1 - li.l_discount)) \
+        .groupby("l_suppkey")["r"].sum()
+    best = rev.max()
+    exp = rev[np.isclose(rev, best)]
+    assert len(out) == len(exp)
+    assert sorted(out["s_suppkey"].astype(np.int64).tolist()) == \
+        sorted(exp.index.tolist())
+
+
+def test_q16_shape_not_in_count_distinct(ctx, tpch2):
+    """Q16: NOT IN subquery + COUNT(DISTINCT) grouped by part attrs."""
+    lineitem, supplier, part = tpch2
+    out = ctx.sql(
+        "SELECT p_brand, COUNT(DISTINCT l_suppkey) AS supplier_cnt "
+        "FROM t2_lineitem JOIN t2_part ON l_partkey = p_partkey "
+        "WHERE p_size < 25 AND l_suppkey NOT IN "
+        "  (SELECT s_suppkey FROM t2_supplier WHERE s_nationkey = 7) "
+        "GROUP BY p_brand").compute()
+    bad = set(supplier[supplier.s_nationkey == 7].s_suppkey)
+    j = lineitem.merge(part, left_on="l_partkey", right_on="p_partkey")
+    j = j[(j.p_size < 25) & ~j.l_suppkey.isin(bad)]
+    exp = j.groupby("p_brand", observed=True)["l_suppkey"].nunique()
+    got = dict(zip(out["p_brand"], out["supplier_cnt"].astype(np.int64)))
+    assert got == exp.to_dict()
+
+
+def test_q19_shape_or_groups(ctx, tpch2):
+    """Q19: disjunction of conjunct groups over join output."""
+    lineitem, supplier, part = tpch2
+    out = ctx.sql(
+        "SELECT SUM(l_extendedprice * (1 - l_discount)) AS revenue "
+        "FROM t2_lineitem JOIN t2_part ON p_partkey = l_partkey "
+        "WHERE (p_brand = 'Brand#1' AND l_quantity BETWEEN 1 AND 11) "
+        "   OR (p_brand = 'Brand#2' AND l_quantity BETWEEN 10 AND 20) "
+        "   OR (p_brand = 'Brand#3' AND l_quantity BETWEEN 20 AND 30)"
+        ).compute()
+    j = lineitem.merge(part, left_on="l_partkey", right_on="p_partkey")
+    m = ((j.p_brand == "Brand#1") & j.l_quantity.between(1, 11)) \
+        | ((j.p_brand == "Brand#2") & j.l_quantity.between(10, 20)) \
+        | ((j.p_brand == "Brand#3") & j.l_quantity.between(20, 30))
+    exp = (j[m].l_extendedprice * (1 - j[m].l_discount)).sum()
+    assert abs(float(out["revenue"][0]) - exp) < 1e-6 * max(1.0, abs(exp))
+
+
+def test_q18_shape_in_grouped_having(ctx, tpch2):
+    """Q18: IN over a grouped HAVING subquery."""
+    lineitem, supplier, part = tpch2
+    out = ctx.sql(
+        "SELECT l_suppkey, SUM(l_quantity) AS tq FROM t2_lineitem "
+        "WHERE l_suppkey IN (SELECT l_suppkey FROM t2_lineitem "
+        "                    GROUP BY l_suppkey HAVING SUM(l_quantity) > "
+        "                    4000) "
+        "GROUP BY l_suppkey").compute()
+    tq = lineitem.groupby("l_suppkey")["l_quantity"].sum()
+    big = tq[tq > 4000]
+    assert len(out) == len(big)
+    got = dict(zip(out["l_suppkey"].astype(np.int64),
+                   out["tq"].astype(np.int64)))
+    assert got == {int(k): int(v) for k, v in big.items()}
