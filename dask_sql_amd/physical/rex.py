@@ -226,6 +226,23 @@ def dict_int_fn(expr, dicts):
             return None
         i, f = sub
         return i, (lambda s, f=f: len(f(s)))
+    if op == "TO_TIMESTAMP" and len(expr.getOperands()) == 2:
+        # string → timestamp: strptime over the dictionary (reference
+        # rex/core/call.py ToTimestampOperation string case)
+        ops_ = expr.getOperands()
+        sub = dict_string_fn(ops_[0], dicts)
+        if sub is None or not isinstance(ops_[1], Literal):
+            return None
+        fmt = str(ops_[1].getValue())
+        i, f = sub
+
+        def ts_of(s, f=f, fmt=fmt):
+            import numpy as _np
+            from datetime import datetime as _dt
+            return int(_np.datetime64(_dt.strptime(f(s), fmt),
+                                      "ns").astype("int64"))
+
+        return i, ts_of
     if op == "POSITION" and len(expr.getOperands()) in (2, 3):
         # POSITION(needle IN hay [FROM start]) — 1-based, 0 when absent
         # (reference rex/core/call.py PositionOperation)

@@ -174,6 +174,12 @@ class Builder:
                 tx = _expr_type(x)
                 if tx == "TIMESTAMP":
                     return x
+                if tx == "VARCHAR":
+                    # string column: strptime runs once per dictionary
+                    # entry (physical/rex.py dict_int_fn TO_TIMESTAMP)
+                    return Call("TO_TIMESTAMP",
+                                [x, Literal(fmt, SqlType("VARCHAR"))],
+                                SqlType("TIMESTAMP"))
                 if len(ops) > 1 and fmt != "%Y-%m-%d %H:%M:%S":
                     raise NotImplementedError(
                         "Integer input does not accept a format argument")

@@ -777,6 +777,17 @@ class Parser:
                 args.append(self.expr())
             self.expect_op(")")
             return ("call", "POSITION", args)
+        if fname == "TO_TIMESTAMP":
+            # second argument is a strptime format; the reference writes it
+            # double-quoted, which lexes as an identifier — accept both
+            args = [self.expr()]
+            if self.accept_op(","):
+                t = self.next()
+                if t[0] not in ("str", "id"):
+                    raise ValueError("to_timestamp format must be a string")
+                args.append(("lit", t[1], "VARCHAR"))
+            self.expect_op(")")
+            return ("call", "TO_TIMESTAMP", args)
         if fname == "OVERLAY":
             # OVERLAY(x PLACING y FROM n [FOR m]) — Calcite form
             x_ = self.expr()

@@ -61,6 +61,7 @@ c.create_table("dates", pd.DataFrame(
     {"d": pd.date_range("2021-01-01", periods=5, freq="D")}))
 c.create_table("datetime_test", pd.DataFrame(
     {"a": pd.date_range("2021-01-01", periods=5, freq="D"),
+     "dt": pd.date_range("2014-01-01", periods=5, freq="250D"),
      "b": np.arange(5)}))
 # dask.datasets.timeseries schema: id/name/x/y
 c.create_table("timeseries", pd.DataFrame(

@@ -516,3 +516,22 @@ def test_q18_shape_in_grouped_having(ctx, tpch2):
     got = dict(zip(out["l_suppkey"].astype(np.int64),
                    out["tq"].astype(np.int64)))
     assert got == {int(k): int(v) for k, v in big.items()}
+
+
+def test_to_timestamp_exec(ctx):
+    # reference ToTimestampOperation: integer seconds, string strptime
+    df = pd.DataFrame({
+        "secs": np.array([0, 86_400, 1_600_000_000], dtype=np.int64),
+        "txt": pd.Series(["01/02/2021", "03/15/2020", "12/31/1999"]
+                         ).astype("category"),
+        "v": np.arange(3, dtype=np.int64)})
+    ctx.create_table("zz_tts", df)
+    out = ctx.sql('SELECT to_timestamp(secs) AS a, '
+                  'to_timestamp(txt, "%m/%d/%Y") AS b, v FROM zz_tts'
+                  ).compute()
+    out = out.sort_values("v").reset_index(drop=True)
+    assert (pd.to_datetime(out["a"])
+            == pd.to_datetime(df["secs"], unit="s")).all()
+    assert (pd.to_datetime(out["b"])
+            == pd.to_datetime(df["txt"].astype(str),
+                              format="%m/%d/%Y")).all()
