@@ -602,6 +602,42 @@ __device__ __forceinline__ bool vm_eval(const DsxInstr* prog, int len, const Col
         res.i = -a.i;
         k.set(sp - 1, res, av);
         break;
+      case DSX_OP_SIN_F64:
+        UN();
+        res.f = sin(a.f);
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_COS_F64:
+        UN();
+        res.f = cos(a.f);
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_TAN_F64:
+        UN();
+        res.f = tan(a.f);
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_ASIN_F64:
+        UN();
+        res.f = asin(a.f);
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_ACOS_F64:
+        UN();
+        res.f = acos(a.f);
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_ATAN_F64:
+        UN();
+        res.f = atan(a.f);
+        k.set(sp - 1, res, av);
+        break;
+      case DSX_OP_ATAN2_F64:
+        k.get(--sp, b, bv);
+        k.get(sp - 1, a, av);
+        res.f = atan2(a.f, b.f);
+        k.set(sp - 1, res, av && bv);
+        break;
 #undef UN
       case DSX_OP_SELECT: {
         // (cond, a, b): cond true→a, false/NULL→b (CASE WHEN semantics)

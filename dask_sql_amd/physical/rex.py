@@ -25,6 +25,8 @@ OP_ABS_I64, OP_ABS_F64, OP_FLOOR_F64, OP_CEIL_F64 = 64, 65, 66, 67
 OP_RINT_F64, OP_EXP_F64, OP_LN_F64, OP_POW_F64 = 68, 69, 70, 71
 OP_YEAR, OP_MONTH, OP_DAY = 72, 73, 74
 OP_FLOORMOD_I64 = 75
+OP_SIN_F64, OP_COS_F64, OP_TAN_F64 = 76, 77, 78
+OP_ASIN_F64, OP_ACOS_F64, OP_ATAN_F64, OP_ATAN2_F64 = 79, 80, 81, 82
 
 # VM value kinds
 KI, KF, KB = "i", "f", "b"  # int64-like, float64, boolean
@@ -453,7 +455,10 @@ class RexCompiler:
             return k
         _f1 = {"FLOOR": OP_FLOOR_F64, "CEIL": OP_CEIL_F64, "CEILING":
                OP_CEIL_F64, "EXP": OP_EXP_F64, "LN": OP_LN_F64,
-               "LOG": OP_LN_F64, "SQRT": OP_SQRT_F64}
+               "LOG": OP_LN_F64, "SQRT": OP_SQRT_F64,
+               "SIN": OP_SIN_F64, "COS": OP_COS_F64, "TAN": OP_TAN_F64,
+               "ASIN": OP_ASIN_F64, "ACOS": OP_ACOS_F64,
+               "ATAN": OP_ATAN_F64}
         if op in _f1:
             self._to_f(self.compile(ops[0]))
             self._emit(_f1[op])
@@ -476,6 +481,19 @@ class RexCompiler:
             self._to_f(self.compile(ops[0]))
             self._to_f(self.compile(ops[1]))
             self._emit(OP_POW_F64)
+            return KF
+        if op == "ATAN2":
+            self._to_f(self.compile(ops[0]))
+            self._to_f(self.compile(ops[1]))
+            self._emit(OP_ATAN2_F64)
+            return KF
+        if op == "COT":
+            # reference: da.cos/da.sin ratio; 1/tan matches to f64 ulp
+            # except exactly at poles, where both are huge finite values
+            self._emit(OP_LIT_F64, 0, 1.0)
+            self._to_f(self.compile(ops[0]))
+            self._emit(OP_TAN_F64)
+            self._emit(OP_DIV_F64)
             return KF
         if op == "MOD":
             # reference evaluates operator.mod on pandas = FLOOR-mod

@@ -192,11 +192,64 @@ class Builder:
                     # truncates to integer ns for the TIMESTAMP column
                     return Call("CAST", [mul], SqlType("TIMESTAMP"))
                 return mul
+            if op in ("DEGREES", "RADIANS", "LOG10", "CBRT", "SIGN",
+                      "TRUNCATE") and len(ops) == 1:
+                # rewrites onto existing scalar ops (reference
+                # rex/core/call.py da.degrees/radians/log10/cbrt/sign/trunc)
+                import math as _m
+                x = ops[0]
+                D = SqlType("DOUBLE")
+                if op == "DEGREES":
+                    return Call("*", [x, Literal(180.0 / _m.pi, D)], D)
+                if op == "RADIANS":
+                    return Call("*", [x, Literal(_m.pi / 180.0, D)], D)
+                if op == "LOG10":
+                    return Call("*", [Call("LN", [x], D),
+                                      Literal(1.0 / _m.log(10.0), D)], D)
+                zero = Literal(0, SqlType("BIGINT"))
+                gt = Call(">", [x, zero], SqlType("BOOLEAN"))
+                lt = Call("<", [x, zero], SqlType("BOOLEAN"))
+                sgn_i = Call("-", [gt, lt], SqlType("BIGINT"))
+                is_f = _expr_type(x) in ("DOUBLE", "FLOAT", "DECIMAL")
+                if op == "SIGN":
+                    if not is_f:
+                        return sgn_i
+                    # np.sign(NaN) = NaN
+                    nan_guard = Call("<>", [x, x], SqlType("BOOLEAN"))
+                    return Call("CASE", [nan_guard, x,
+                                         Call("CAST", [sgn_i], D)], D)
+                if op == "CBRT":
+                    root = Call("POWER",
+                                [Call("ABS", [x], D),
+                                 Literal(1.0 / 3.0, D)], D)
+                    return Call("*", [sgn_i, root], D)
+                # TRUNCATE: toward zero (da.trunc)
+                ge = Call(">=", [x, zero], SqlType("BOOLEAN"))
+                return Call("CASE", [ge, Call("FLOOR", [x], D),
+                                     Call("CEIL", [x], D)], D)
             if op in ("+", "-") and any(
                     isinstance(o, Literal)
                     and o.getType().getSqlType() == "INTERVAL" for o in ops):
                 return self._date_interval(op, ops)
             if op in ("=", "<>", "<", "<=", ">", ">=") and len(ops) == 2:
+                ta, tb = _expr_type(ops[0]), _expr_type(ops[1])
+                # DATE/TIMESTAMP vs string literal: coerce the literal to
+                # the temporal side (reference lets pandas compare
+                # datetime64 against a parseable string)
+                for i_, (tx_, ty_) in ((0, (ta, tb)), (1, (tb, ta))):
+                    o = ops[i_]
+                    if ty_ in ("DATE", "TIMESTAMP") and tx_ == "VARCHAR" \
+                            and isinstance(o, Literal) \
+                            and isinstance(o.getValue(), str):
+                        try:
+                            if ty_ == "DATE":
+                                v = _date_to_days(o.getValue())
+                            else:
+                                v = int(np.datetime64(o.getValue(), "ns")
+                                        .astype("int64"))
+                        except Exception:
+                            continue
+                        ops[i_] = Literal(v, SqlType(ty_))
                 ta, tb = _expr_type(ops[0]), _expr_type(ops[1])
                 if {ta, tb} == {"TIMESTAMP", "DATE"}:
                     # promote the DATE side to ns so the compare is exact
@@ -223,7 +276,8 @@ class Builder:
                         "POSITION"):
                 ty = "BIGINT"
             elif op in ("FLOOR", "CEIL", "CEILING", "ROUND", "EXP", "LN",
-                        "LOG", "POWER", "POW", "SQRT"):
+                        "LOG", "POWER", "POW", "SQRT", "SIN", "COS", "TAN",
+                        "ASIN", "ACOS", "ATAN", "ATAN2", "COT"):
                 ty = "DOUBLE"
             elif op == "EXTRACT_DATE":
                 ty = "DATE"

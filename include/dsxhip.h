@@ -74,6 +74,11 @@ enum DsxOp {
   DSX_OP_FLOORMOD_I64 = 75, /* Python/pandas floor-mod: MOD(-5,3) = 1
                                (reference evaluates operator.mod on pandas,
                                rex/core/call.py:1047-1156) */
+  /* trigonometry (rex/core/call.py TrigonometricOperations — da.sin etc.);
+     unary ops take/return f64, ATAN2 is binary */
+  DSX_OP_SIN_F64 = 76, DSX_OP_COS_F64 = 77, DSX_OP_TAN_F64 = 78,
+  DSX_OP_ASIN_F64 = 79, DSX_OP_ACOS_F64 = 80, DSX_OP_ATAN_F64 = 81,
+  DSX_OP_ATAN2_F64 = 82,
 };
 
 typedef struct DsxInstr {

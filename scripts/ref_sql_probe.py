@@ -121,12 +121,75 @@ for fn in FILES:
                 continue
             q = q2
         try:
-            c._get_ral(q)
+            rel = c._get_ral(q)
             ok += 1
         except Exception as e:
             bad += 1
             key = f"{type(e).__name__}: {str(e)[:90]}"
             fails.setdefault(key, []).append((fn, q[:100].replace("\n", " ")))
+            continue
+        if "--compile" in sys.argv:
+            # rex-compile every Projection/Filter/Join expression in the
+            # plan against fake device columns of the planned dtypes —
+            # catches EXECUTION-time compile gaps without a GPU. String
+            # exprs legitimately route through the dict LUT paths
+            # (dict_string_fn/dict_int_fn) instead of the VM; exprs whose
+            # compile fails but whose dict path succeeds are NOT gaps.
+            import types as _t
+
+            from dask_sql_amd import runtime as rt
+            from dask_sql_amd.physical import rex as R
+
+            _SQL_DT = {"BIGINT": rt.I64, "INTEGER": rt.I64,
+                       "SMALLINT": rt.I64, "TINYINT": rt.I64,
+                       "DATE": rt.I32, "TIMESTAMP": rt.I64,
+                       "VARCHAR": rt.I32, "DOUBLE": rt.F64,
+                       "FLOAT": rt.F64, "DECIMAL": rt.F64,
+                       "BOOLEAN": rt.BOOL8, "NULL": rt.I64}
+
+            def cols_of(node):
+                fs = node.getRowType().getFieldList()
+                cols = [_t.SimpleNamespace(
+                    dtype=_SQL_DT.get(f.getType().getSqlType(), rt.I64))
+                    for f in fs]
+                dicts = [["aa", "bb", "cc"]
+                         if f.getType().getSqlType() == "VARCHAR" else None
+                         for f in fs]
+                return cols, dicts
+
+            def walk(node):
+                t = node.get_current_node_type()
+                exprs = []
+                if t == "Projection":
+                    exprs = [e for e, _ in
+                             node.projection().getNamedProjects()]
+                elif t == "Filter":
+                    exprs = [node.filter().getCondition()]
+                for ins in node.get_inputs():
+                    walk(ins)
+                if not exprs or not node.get_inputs():
+                    return
+                cols, dicts = cols_of(node.get_inputs()[0])
+                for e in exprs:
+                    try:
+                        R.RexCompiler(cols, dicts).compile(e)
+                    except R.RexCompileError as ex:
+                        if R.dict_string_fn(e, dicts) is not None:
+                            continue
+                        if R.dict_int_fn(e, dicts) is not None:
+                            continue
+                        key = f"COMPILE {str(ex)[:80]}"
+                        fails.setdefault(key, []).append(
+                            (fn, q[:100].replace("\n", " ")))
+                    except Exception as ex:
+                        key = f"COMPILE-{type(ex).__name__}: {str(ex)[:70]}"
+                        fails.setdefault(key, []).append(
+                            (fn, q[:100].replace("\n", " ")))
+            try:
+                walk(rel)
+            except Exception as ex:
+                fails.setdefault(f"WALK-{type(ex).__name__}: "
+                                 f"{str(ex)[:70]}", []).append((fn, q[:80]))
 print(f"planned OK: {ok}  failed: {bad}  skipped(non-select/f-str): {skipped}")
 for k, v in sorted(fails.items(), key=lambda kv: -len(kv[1]))[:25]:
     print(f"\n[{len(v)}x] {k}")

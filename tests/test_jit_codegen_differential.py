@@ -44,6 +44,15 @@ static inline double __longlong_as_double(i64 x) {
 static inline double __ocml_exp_f64(double x) { return exp(x); }
 static inline double __ocml_log_f64(double x) { return log(x); }
 static inline double __ocml_pow_f64(double a, double b) { return pow(a, b); }
+static inline double __ocml_sin_f64(double x) { return sin(x); }
+static inline double __ocml_cos_f64(double x) { return cos(x); }
+static inline double __ocml_tan_f64(double x) { return tan(x); }
+static inline double __ocml_asin_f64(double x) { return asin(x); }
+static inline double __ocml_acos_f64(double x) { return acos(x); }
+static inline double __ocml_atan_f64(double x) { return atan(x); }
+static inline double __ocml_atan2_f64(double a, double b) {
+  return atan2(a, b);
+}
 static inline i64 jit_absl(i64 v) { return v < 0 ? -v : v; }
 /* AMD v_cvt f64->i64 saturates and maps NaN to 0; x86 cvttsd2si gives
    INT64_MIN — route every (i64) cast through this to match the device */

@@ -105,6 +105,18 @@ def ev(e, cols, row, pk=None):
     if op in ("ABS",):
         a = ev(ops[0], cols, row, pk)
         return None if a is None else abs(a)
+    if op in ("SIN", "COS", "TAN", "ATAN"):
+        a = ev(ops[0], cols, row, pk)
+        if a is None:
+            return None
+        return {"SIN": math.sin, "COS": math.cos, "TAN": math.tan,
+                "ATAN": math.atan}[op](float(a))
+    if op == "ATAN2":
+        a = ev(ops[0], cols, row, pk)
+        b = ev(ops[1], cols, row, pk)
+        if a is None or b is None:
+            return None
+        return math.atan2(float(a), float(b))
     if op in ("FLOOR", "CEIL", "SQRT", "EXP", "LN"):
         a = ev(ops[0], cols, row, pk)
         if a is None:
@@ -258,7 +270,8 @@ def gen(rng, kind, depth):
     if r < 0.9:
         return Call("NEG", [gen(rng, "NUM", depth - 1)], SqlType(F))
     if r < 0.92:
-        fn = str(rng.choice(["FLOOR", "CEIL", "SQRT", "EXP", "LN"]))
+        fn = str(rng.choice(["FLOOR", "CEIL", "SQRT", "EXP", "LN",
+                             "SIN", "COS", "TAN", "ATAN"]))
         return Call(fn, [gen(rng, "NUM", depth - 1)], SqlType(F))
     if r < 0.94:
         d = int(rng.integers(0, 3))

@@ -535,3 +535,33 @@ def test_to_timestamp_exec(ctx):
     assert (pd.to_datetime(out["b"])
             == pd.to_datetime(df["txt"].astype(str),
                               format="%m/%d/%Y")).all()
+
+
+def test_trig_math_family(ctx):
+    # reference rex/core/call.py trigonometry + da.degrees/log10/cbrt/
+    # sign/trunc — vs numpy on the same values
+    x = np.array([0.5, -1.2, 0.0, 2.9, -0.001], dtype=np.float64)
+    a = np.array([3, -4, 0, 7, -1], dtype=np.int64)
+    df = pd.DataFrame({"x": x, "a": a, "v": np.arange(5, dtype=np.int64)})
+    ctx.create_table("zz_trig", df)
+    out = ctx.sql(
+        "SELECT SIN(x) AS s, COS(x) AS c, TAN(x) AS t, ATAN(x) AS at, "
+        "ATAN2(x, 1 + 0 * v) AS a2, COT(x + 2) AS ct, DEGREES(x) AS dg, "
+        "RADIANS(x) AS rd, LOG10(ABS(x) + 1) AS lg, CBRT(x) AS cb, "
+        "SIGN(x) AS sgf, SIGN(a) AS sgi, TRUNCATE(x * 3) AS tr, v "
+        "FROM zz_trig").compute()
+    out = out.sort_values("v").reset_index(drop=True)
+    np.testing.assert_allclose(out["s"], np.sin(x), rtol=1e-12)
+    np.testing.assert_allclose(out["c"], np.cos(x), rtol=1e-12)
+    np.testing.assert_allclose(out["t"], np.tan(x), rtol=1e-12)
+    np.testing.assert_allclose(out["at"], np.arctan(x), rtol=1e-12)
+    np.testing.assert_allclose(out["a2"], np.arctan2(x, 1.0), rtol=1e-12)
+    np.testing.assert_allclose(out["ct"], 1 / np.tan(x + 2), rtol=1e-12)
+    np.testing.assert_allclose(out["dg"], np.degrees(x), rtol=1e-12)
+    np.testing.assert_allclose(out["rd"], np.radians(x), rtol=1e-12)
+    np.testing.assert_allclose(out["lg"], np.log10(np.abs(x) + 1),
+                               rtol=1e-12)
+    np.testing.assert_allclose(out["cb"], np.cbrt(x), rtol=1e-12)
+    np.testing.assert_allclose(out["sgf"], np.sign(x))
+    assert out["sgi"].astype(np.int64).tolist() == [1, -1, 0, 1, -1]
+    np.testing.assert_allclose(out["tr"], np.trunc(x * 3))

@@ -210,6 +210,29 @@ def interp(prog, cols, row):
             except (OverflowError, ValueError):
                 r = math.nan
             push(r, av and bv)
+        elif op == R.OP_SIN_F64:
+            a, av = pop1()
+            push(math.sin(float(a)), av)
+        elif op == R.OP_COS_F64:
+            a, av = pop1()
+            push(math.cos(float(a)), av)
+        elif op == R.OP_TAN_F64:
+            a, av = pop1()
+            push(math.tan(float(a)), av)
+        elif op == R.OP_ASIN_F64:
+            a, av = pop1()
+            f = float(a)
+            push(math.asin(f) if -1.0 <= f <= 1.0 else math.nan, av)
+        elif op == R.OP_ACOS_F64:
+            a, av = pop1()
+            f = float(a)
+            push(math.acos(f) if -1.0 <= f <= 1.0 else math.nan, av)
+        elif op == R.OP_ATAN_F64:
+            a, av = pop1()
+            push(math.atan(float(a)), av)
+        elif op == R.OP_ATAN2_F64:
+            a, av, b, bv = pop2()
+            push(math.atan2(float(a), float(b)), av and bv)
         elif op == R.OP_YEAR:
             a, av = pop1()
             push(_civil(a)[0], av)
