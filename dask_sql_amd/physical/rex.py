@@ -498,10 +498,18 @@ class RexCompiler:
         if op == "MOD":
             # reference evaluates operator.mod on pandas = FLOOR-mod
             # (MOD(-5,3) = 1), not C truncated remainder.
-            # Integer-only: FLOORMOD reads raw int slots, a float operand
-            # would be reinterpreted — refuse loudly.
+            # Floats: a - floor(a/b)*b (python float % semantics; b=0
+            # yields NaN like pandas)
             if KF in (self._peek_kind(ops[0]), self._peek_kind(ops[1])):
-                raise RexCompileError("MOD needs integer operands")
+                self._to_f(self.compile(ops[0]))
+                self._to_f(self.compile(ops[0]))
+                self._to_f(self.compile(ops[1]))
+                self._emit(OP_DIV_F64)
+                self._emit(OP_FLOOR_F64)
+                self._to_f(self.compile(ops[1]))
+                self._emit(OP_MUL_F64)
+                self._emit(OP_SUB_F64)
+                return KF
             self.compile(ops[0])
             self.compile(ops[1])
             self._emit(OP_FLOORMOD_I64)
@@ -961,7 +969,11 @@ class RexCompiler:
                 if any(self._peek_kind(v) == KF for v in vals):
                     return KF
                 return self._peek_kind(ops_[1])
-            if op in ("MOD", "EXTRACT_YEAR", "EXTRACT_MONTH", "EXTRACT_DAY",
+            if op == "MOD":
+                ka = self._peek_kind(expr.getOperands()[0])
+                kb = self._peek_kind(expr.getOperands()[1])
+                return KF if KF in (ka, kb) else KI
+            if op in ("EXTRACT_YEAR", "EXTRACT_MONTH", "EXTRACT_DAY",
                       "YEAR", "MONTH", "DAY", "DAYOFMONTH") \
                     or op.startswith("EXTRACT_") \
                     or op.startswith("FLOOR_TO_") \

@@ -281,8 +281,13 @@ class Builder:
                 ty = "DOUBLE"
             elif op == "EXTRACT_DATE":
                 ty = "DATE"
+            elif op == "MOD":
+                ta = _expr_type(ops[0])
+                tb = _expr_type(ops[1])
+                ty = "DOUBLE" if "DOUBLE" in (ta, tb) or "FLOAT" in (
+                    ta, tb) or "DECIMAL" in (ta, tb) else "BIGINT"
             elif op.startswith("EXTRACT_") or op in (
-                    "MOD", "YEAR", "MONTH", "DAY", "DAYOFMONTH"):
+                    "YEAR", "MONTH", "DAY", "DAYOFMONTH"):
                 ty = "BIGINT"
             elif op.startswith("FLOOR_TO_") or op.startswith("CEIL_TO_"):
                 ty = _expr_type(ops[0])

@@ -885,6 +885,8 @@ class Parser:
                     ("call", "/", [("call", "*", [sm(y), sm(x)]), nf])])
             den = nf if fname == "COVAR_POP" else                 ("call", "-", [nf, ("lit", 1, "BIGINT")])
             return ("call", "/", [cov, den])
+        if fname == "MEAN":  # reference accepts MEAN as an AVG alias
+            return ("agg", "avg", args, distinct, filter_expr)
         if fname in AGG_FUNCS:
             return ("agg", fname.lower(), args, distinct, filter_expr)
         return ("call", fname, args)

@@ -173,6 +173,12 @@ def ev(e, cols, row, pk=None):
         a, b = ev(ops[0], cols, row, pk), ev(ops[1], cols, row, pk)
         if a is None or b is None:
             return None
+        if R.KF in (pk(ops[0]), pk(ops[1])):
+            if float(b) == 0.0:
+                raise Skip()  # NaN on the VM, ZeroDivision here
+            # exactly the VM's float floor-mod formula (python's own %
+            # applies an fmod correction that differs by ulps)
+            return float(a) - math.floor(float(a) / float(b)) * float(b)
         if b == 0:
             raise Skip()
         return int(a) % int(b)  # python % IS floor-mod (operator.mod ref)

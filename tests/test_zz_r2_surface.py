@@ -565,3 +565,17 @@ def test_trig_math_family(ctx):
     np.testing.assert_allclose(out["sgf"], np.sign(x))
     assert out["sgi"].astype(np.int64).tolist() == [1, -1, 0, 1, -1]
     np.testing.assert_allclose(out["tr"], np.trunc(x * 3))
+
+
+def test_float_mod_and_mean(ctx):
+    # reference MOD on floats = operator.mod (floor semantics); MEAN = AVG
+    df = pd.DataFrame({"b": [7.5, -7.5, 2.3], "a": [3, 3, 3],
+                       "v": np.arange(3, dtype=np.int64)})
+    ctx.create_table("zz_fm", df)
+    out = ctx.sql("SELECT MOD(b, 4) AS m, v FROM zz_fm").compute()
+    out = out.sort_values("v").reset_index(drop=True)
+    np.testing.assert_allclose(out["m"], df["b"] % 4, rtol=1e-12)
+    out = ctx.sql("SELECT MEAN(b) AS mb, MEAN(a) AS ma FROM zz_fm"
+                  ).compute()
+    assert abs(float(out["mb"][0]) - df["b"].mean()) < 1e-12
+    assert abs(float(out["ma"][0]) - 3.0) < 1e-12
