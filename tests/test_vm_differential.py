@@ -26,7 +26,17 @@ class Skip(Exception):
 # one compiler instance only for static kind queries (mirrors the typing
 # the compiled program actually uses for int-vs-float division)
 def _pk():
-    return R.RexCompiler(_dev_cols())._peek_kind
+    raw = R.RexCompiler(_dev_cols())._peek_kind
+    memo = {}
+
+    def pk(e):
+        k = memo.get(id(e))
+        if k is None:
+            k = raw(e)
+            memo[id(e)] = k
+        return k
+
+    return pk
 
 
 # ---- direct SQL-semantics evaluator over the AST --------------------------
@@ -321,9 +331,10 @@ def test_vm_differential(seed):
             rk = c.compile(e)
         except R.RexCompileError:
             continue
+        pk = _pk()
         for row in range(n):
             try:
-                want = ev(e, cols, row)
+                want = ev(e, cols, row, pk)
             except Skip:
                 continue
             got, ok = interp(c.prog, cols, row)
