@@ -42,6 +42,27 @@ def _common_type(a: str, b: str) -> str:
     return a if ra >= rb else b
 
 
+def _add_months_host(is_date):
+    """pandas DateOffset month/year arithmetic over raw day/ns ints (the
+    UDF slow path hands operands over as host Series)."""
+    def f(vals, months):
+        import pandas as pd
+        m = int(months.iloc[0])
+        if is_date:
+            base = pd.to_datetime(vals, unit="D")
+            out = pd.Series(base + pd.DateOffset(months=m))
+            res = ((out - pd.Timestamp(0)).dt.days).astype("float64")
+            return res  # NaT → NaN → validity
+        base = pd.to_datetime(vals)
+        out = pd.Series(base + pd.DateOffset(months=m))
+        if out.isna().any():
+            res = out.to_numpy("datetime64[ns]").view("int64")                 .astype("float64")
+            res[out.isna().to_numpy()] = np.nan
+            return pd.Series(res)
+        return pd.Series(out.to_numpy("datetime64[ns]").view("int64"))
+    return f
+
+
 class Catalog:
     """What the builder needs from the Context's schema: table → fields,
     plus the registered scalar/aggregate UDFs (reference
@@ -52,6 +73,16 @@ class Catalog:
         self.tables: dict[str, list[tuple[str, str]]] = {}
         self.functions: dict[str, tuple] = {}     # name → (f, ret_sql, row_udf)
         self.aggregations: dict[str, tuple] = {}  # name → (obj, ret_sql)
+        # hidden builtins: calendar month arithmetic on COLUMNS executes
+        # host-side exactly as the reference does (pandas Timestamp +
+        # DateOffset, rex/core/call.py datetime ops) through the normal
+        # UDF machinery
+        self.functions["__add_months_ts__"] = (
+            _add_months_host(False), "TIMESTAMP", False,
+            [("x", "TIMESTAMP"), ("m", "BIGINT")])
+        self.functions["__add_months_date__"] = (
+            _add_months_host(True), "DATE", False,
+            [("x", "DATE"), ("m", "BIGINT")])
 
     def add(self, name, fields):
         self.tables[name.lower()] = fields
@@ -464,8 +495,13 @@ class Builder:
             return Call(op, [other, Literal(step, SqlType("BIGINT"))],
                         SqlType(out_t))
         if not isinstance(other, Literal):
-            raise NotImplementedError(
-                "MONTH/YEAR interval arithmetic on a column")
+            # column operand: exact calendar arithmetic on the host UDF
+            # path (reference pandas + DateOffset)
+            months = sign * n_ * (12 if unit == "YEAR" else 1)
+            name = "__add_months_ts__" if is_ts else "__add_months_date__"
+            return Call(f"UDF:{name}",
+                        [other, Literal(months, SqlType("BIGINT"))],
+                        SqlType("TIMESTAMP" if is_ts else "DATE"))
         if is_ts:
             import pandas as pd
             months = sign * n_ * (12 if unit == "YEAR" else 1)

@@ -579,3 +579,34 @@ def test_float_mod_and_mean(ctx):
                   ).compute()
     assert abs(float(out["mb"][0]) - df["b"].mean()) < 1e-12
     assert abs(float(out["ma"][0]) - 3.0) < 1e-12
+
+
+def test_month_interval_on_column(ctx):
+    # calendar month/year arithmetic on a COLUMN runs the host UDF path
+    # (pandas DateOffset — end-of-month clamping included)
+    ts = pd.to_datetime(["2021-01-31 10:00:00", "2020-02-29 23:00:00",
+                         "2019-12-15 00:30:00"])
+    df = pd.DataFrame({"t": ts, "v": np.arange(3, dtype=np.int64)})
+    ctx.create_table("zz_mi", df)
+    out = ctx.sql("SELECT t + INTERVAL '1' MONTH AS a, "
+                  "t - INTERVAL '2' MONTH AS b, "
+                  "TIMESTAMPADD(YEAR, 1, t) AS y, v FROM zz_mi").compute()
+    out = out.sort_values("v").reset_index(drop=True)
+    s = pd.Series(ts)
+    assert (pd.to_datetime(out["a"])
+            == s + pd.DateOffset(months=1)).all()
+    assert (pd.to_datetime(out["b"])
+            == s - pd.DateOffset(months=2)).all()
+    assert (pd.to_datetime(out["y"])
+            == s + pd.DateOffset(months=12)).all()
+
+
+def test_month_interval_on_date_column(ctx):
+    d = pd.to_datetime(["2021-01-31", "2020-02-29", "2019-12-15"])
+    df = pd.DataFrame({"d": d, "v": np.arange(3, dtype=np.int64)})
+    ctx.create_table("zz_mid", df)
+    out = ctx.sql("SELECT d + INTERVAL '1' MONTH AS a, v FROM zz_mid"
+                  ).compute()
+    out = out.sort_values("v").reset_index(drop=True)
+    assert (pd.to_datetime(out["a"])
+            == pd.Series(d) + pd.DateOffset(months=1)).all()
