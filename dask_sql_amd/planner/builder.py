@@ -63,6 +63,18 @@ def _add_months_host(is_date):
     return f
 
 
+def _isoweek_host(is_date):
+    """EXTRACT(WEEK) = isocalendar().week (reference date_part WEEK)."""
+    def f(vals):
+        import pandas as pd
+        base = pd.to_datetime(vals, unit="D") if is_date \
+            else pd.to_datetime(vals)
+        wk = pd.Series(base).dt.isocalendar().week.astype("float64")
+        wk[pd.Series(base).isna().to_numpy()] = np.nan
+        return wk
+    return f
+
+
 class Catalog:
     """What the builder needs from the Context's schema: table → fields,
     plus the registered scalar/aggregate UDFs (reference
@@ -83,6 +95,10 @@ class Catalog:
         self.functions["__add_months_date__"] = (
             _add_months_host(True), "DATE", False,
             [("x", "DATE"), ("m", "BIGINT")])
+        self.functions["__isoweek_ts__"] = (
+            _isoweek_host(False), "BIGINT", False, [("x", "TIMESTAMP")])
+        self.functions["__isoweek_date__"] = (
+            _isoweek_host(True), "BIGINT", False, [("x", "DATE")])
 
     def add(self, name, fields):
         self.tables[name.lower()] = fields
@@ -187,6 +203,11 @@ class Builder:
             fn = self.catalog.functions.get(op.lower())
             if fn is not None:
                 return Call(f"UDF:{op.lower()}", ops, SqlType(fn[1]))
+            if op == "EXTRACT_WEEK":
+                x = ops[0]
+                name = "__isoweek_ts__" if _expr_type(x) == "TIMESTAMP" \
+                    else "__isoweek_date__"
+                return Call(f"UDF:{name}", [x], SqlType("BIGINT"))
             if op == "TO_TIMESTAMP":
                 # reference rex/core/call.py ToTimestampOperation: string →
                 # strptime (host fold); numeric → seconds since epoch; an

@@ -681,7 +681,7 @@ class Parser:
                 if field not in ("YEAR", "MONTH", "DAY", "HOUR", "MINUTE",
                                  "SECOND", "DATE", "CENTURY", "DECADE",
                                  "MILLENNIUM", "DOW", "DOY", "QUARTER",
-                                 "MICROSECOND", "MILLISECOND"):
+                                 "MICROSECOND", "MILLISECOND", "WEEK"):
                     raise ValueError(f"EXTRACT({field}) not supported")
                 self.expect_kw("FROM")
                 e = self.expr()

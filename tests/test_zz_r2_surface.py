@@ -610,3 +610,15 @@ def test_month_interval_on_date_column(ctx):
     out = out.sort_values("v").reset_index(drop=True)
     assert (pd.to_datetime(out["a"])
             == pd.Series(d) + pd.DateOffset(months=1)).all()
+
+
+def test_extract_week_exec(ctx):
+    d = pd.to_datetime(["2021-01-04", "2021-01-03", "2020-12-31",
+                        "2019-12-30"])
+    df = pd.DataFrame({"d": d, "v": np.arange(4, dtype=np.int64)})
+    ctx.create_table("zz_wk", df)
+    out = ctx.sql("SELECT EXTRACT(WEEK FROM d) AS w, v FROM zz_wk"
+                  ).compute()
+    out = out.sort_values("v").reset_index(drop=True)
+    want = pd.Series(d).dt.isocalendar().week.tolist()
+    assert out["w"].astype(np.int64).tolist() == [int(x) for x in want]
