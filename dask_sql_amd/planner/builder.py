@@ -167,6 +167,12 @@ class Builder:
             if not matches:
                 raise KeyError(f"column {q + '.' if q else ''}{n} not found in "
                                f"{[f.getQualifiedName() for f in fields]}")
+            if len(matches) > 1:
+                # names differing only in case (quoted identifiers): an
+                # EXACT-case match wins before declaring ambiguity
+                exact = [i for i in matches if fields[i].getName() == n]
+                if len(exact) == 1:
+                    matches = exact
             if len(matches) > 1 and q is None:
                 raise KeyError(f"ambiguous column {n}")
             i = matches[0]

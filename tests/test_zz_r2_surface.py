@@ -622,3 +622,17 @@ def test_extract_week_exec(ctx):
     out = out.sort_values("v").reset_index(drop=True)
     want = pd.Series(d).dt.isocalendar().week.tolist()
     assert out["w"].astype(np.int64).tolist() == [int(x) for x in want]
+
+
+def test_case_sensitive_quoted_aliases(ctx):
+    # reference test_select.py casing test: quoted aliases differing only
+    # in case resolve by exact match
+    df = pd.DataFrame({"a": np.array([5, 6], dtype=np.int64),
+                       "b": np.array([10, 20], dtype=np.int64)})
+    ctx.create_table("zz_case", df)
+    out = ctx.sql('SELECT "AAA", "aaa", "aAa" FROM '
+                  '(SELECT a - 1 AS "aAa", 2*b AS "aaa", a + b AS "AAA" '
+                  ' FROM zz_case) x').compute()
+    assert out["AAA"].astype(np.int64).tolist() == [15, 26]
+    assert out["aaa"].astype(np.int64).tolist() == [20, 40]
+    assert out["aAa"].astype(np.int64).tolist() == [4, 5]
