@@ -2402,6 +2402,19 @@ class DaskWindowPlugin(BaseRelPlugin):
     _AGGS = {"sum", "count", "avg", "min", "max"}
 
     def _one(self, runtime, cols, n, spec):
+        if getattr(spec, "frame", None) is None \
+                and spec.func == "last_value" and spec.order_idx \
+                and spec.arg_idx is not None:
+            # ordered default frame ends AT the current row, and the
+            # reference's row-based expanding window makes LAST_VALUE the
+            # current row's value (window.py LastValueOperation over
+            # expanding) — a straight copy of the operand
+            return runtime.concat_columns([cols[spec.arg_idx]],
+                                          cols[spec.arg_idx].dtype)
+        if getattr(spec, "frame", None) is not None:
+            # explicit ROWS frames run the reference's own rolling pandas
+            # computation host-side (window.py:145-198 map_on_each_group)
+            return self._host_ordered(runtime, cols, n, spec)
         device_ok = (spec.func in self._AGGS and not spec.order_idx
                      and all(cols[i].dtype in _INT_KINDS
                              for i in spec.part_idx))
@@ -2411,6 +2424,61 @@ class DaskWindowPlugin(BaseRelPlugin):
         if col is not None:
             return col
         return self._host_ordered(runtime, cols, n, spec)
+
+    def _frame_apply(self, df, grp, pnames, spec):
+        """Explicit ROWS frame over sorted partitions — the reference's
+        map_on_each_group expanding/rolling/Indexer chain restated
+        (window.py:145-198)."""
+        import pandas as pd
+        fk, lo, hi = spec.frame
+        f = spec.func
+        src = "v" if spec.arg_idx is not None else "_one_"
+        if src == "_one_":
+            df["_one_"] = 1.0
+            grp = df.groupby(pnames, dropna=False, sort=False)
+        if lo[0] == "unbounded_preceding" and hi[0] == "current":
+            roll = grp[src].expanding(min_periods=1)
+        elif lo[0] == "preceding" and hi[0] == "current":
+            roll = grp[src].rolling(window=lo[1] + 1, min_periods=1)
+        else:
+            from pandas.api.indexers import BaseIndexer
+
+            lo_off = {"unbounded_preceding": None, "current": 0,
+                      "preceding": -lo[1] if lo[1] is not None else None,
+                      "following": lo[1]}[lo[0]]
+            hi_off = {"unbounded_following": None, "current": 0,
+                      "preceding": -hi[1] if hi[1] is not None else None,
+                      "following": hi[1]}[hi[0]]
+
+            class _Ix(BaseIndexer):
+                def get_window_bounds(self, num_values=0, min_periods=None,
+                                      center=None, closed=None, step=None):
+                    i = np.arange(num_values, dtype=np.int64)
+                    start = np.zeros(num_values, dtype=np.int64)                         if lo_off is None                         else np.clip(i + lo_off, 0, num_values)
+                    end = np.full(num_values, num_values, dtype=np.int64)                         if hi_off is None                         else np.clip(i + hi_off + 1, 0, num_values)
+                    return start, np.maximum(start, end)
+
+            roll = grp[src].rolling(window=_Ix(), min_periods=1)
+        if f == "sum":
+            res = roll.sum()
+        elif f == "count":
+            res = roll.count()
+        elif f == "avg":
+            res = roll.mean()
+        elif f == "min":
+            res = roll.min()
+        elif f == "max":
+            res = roll.max()
+        elif f == "first_value":
+            res = roll.apply(lambda x: x.iloc[0], raw=False)
+        elif f == "last_value":
+            res = roll.apply(lambda x: x.iloc[-1], raw=False)
+        else:
+            raise RexCompileError(
+                f"window function {f} with an explicit frame")
+        # rolling output is ordered group-by-group exactly like the sorted
+        # frame — realign positionally
+        return pd.Series(res.to_numpy(), index=df.index)
 
     def _device_ordered(self, runtime, cols, n, spec):
         """Device ordered window frames: sort permutation
@@ -2641,8 +2709,14 @@ class DaskWindowPlugin(BaseRelPlugin):
         df = df.sort_values(pnames, na_position="last", kind="mergesort")
         grp = df.groupby(pnames, dropna=False, sort=False)
         f = spec.func
-        if f == "row_number":
+        if getattr(spec, "frame", None) is not None:
+            res = self._frame_apply(df, grp, pnames, spec)
+        elif f == "row_number":
             res = grp.cumcount() + 1
+        elif f == "last_value":
+            # default frame: ordered → the current row (reference expanding
+            # tail); unordered → whole partition → partition tail
+            res = df["v"] if onames else grp["v"].transform("last")
         elif f == "first_value":
             # default frame starts at the partition head (test_over.py:90)
             res = grp["v"].transform("first")

@@ -1126,7 +1126,9 @@ class Builder:
         specs = []
         win_out = {}
         for ast in win_asts:
-            _, func, args, part, order = ast
+            _, func, args, part, order, frame = ast
+            if func == "single_value":
+                func = "first_value"  # reference maps both to FirstValue
             arg_idx = None
             arg_t = None
             if args and args[0] != ("star",):
@@ -1134,9 +1136,11 @@ class Builder:
                 arg_t = pre_fields[arg_idx].getType().getSqlType()
             part_idx = [idx_of(p) for p in part]
             order_idx = [(idx_of(o), desc) for o, desc in order]
-            if func in ("row_number", "rank", "dense_rank", "lag",
-                        "lead", "first_value") and not order_idx:
+            if func in ("rank", "dense_rank", "lag", "lead") \
+                    and not order_idx:
                 raise ValueError(f"{func.upper()} requires ORDER BY in OVER")
+            # ROW_NUMBER / FIRST_VALUE / LAST_VALUE without ORDER BY run in
+            # input order (reference window.py row_number = range(1..n))
             offset, default = 1, None
             if func in ("lag", "lead"):
 
@@ -1160,10 +1164,16 @@ class Builder:
                 ty = "DOUBLE" if _is_float(arg_t or "BIGINT") else "BIGINT"
             else:  # min/max/lag/lead/first_value keep the arg type
                 ty = arg_t or "BIGINT"
+            if frame is not None:
+                fk, lo, hi = frame
+                if fk == "range":
+                    raise NotImplementedError(
+                        "RANGE frames with offsets (ROWS frames and the "
+                        "default RANGE frame are supported)")
             name = f"w{len(specs)}__{func}"
             specs.append(WindowSpec(func, arg_idx, part_idx, order_idx,
                                     name, SqlType(ty), offset=offset,
-                                    default=default))
+                                    default=default, frame=frame))
             win_out[repr(ast)] = name
         if len(pre_named) > len(in_fields):
             plan = LogicalPlan("Projection", [plan], RelDataType(pre_fields),

@@ -62,7 +62,8 @@ KEYWORDS = {
 }
 
 WINDOW_FUNCS = {"ROW_NUMBER", "RANK", "DENSE_RANK", "SUM", "COUNT", "AVG",
-                "MIN", "MAX", "LAG", "LEAD", "FIRST_VALUE"}
+                "MIN", "MAX", "LAG", "LEAD", "FIRST_VALUE", "LAST_VALUE",
+                "SINGLE_VALUE"}
 
 _TOKEN_RE = re.compile(
     r"""
@@ -854,8 +855,47 @@ class Parser:
                     order.append((e, desc))
                     if not self.accept_op(","):
                         break
+            frame = None
+            fk = "ROWS" if self._accept_word("ROWS") else (
+                "RANGE" if self._accept_word("RANGE") else None)
+            if fk is not None:
+                def bound():
+                    if self._accept_word("UNBOUNDED"):
+                        side = "preceding" \
+                            if self._accept_word("PRECEDING") else \
+                            ("following" if self._accept_word("FOLLOWING")
+                             else None)
+                        if side is None:
+                            raise ValueError("UNBOUNDED needs PRECEDING/"
+                                             "FOLLOWING")
+                        return ("unbounded_" + side, None)
+                    if self._accept_word("CURRENT"):
+                        self._expect_word("ROW")
+                        return ("current", 0)
+                    t = self.next()
+                    if t[0] != "num":
+                        raise ValueError(f"bad frame bound {t}")
+                    k = int(t[1])
+                    if self._accept_word("PRECEDING"):
+                        return ("preceding", k)
+                    self._expect_word("FOLLOWING")
+                    return ("following", k)
+
+                if self.accept_kw("BETWEEN"):
+                    lo = bound()
+                    self.expect_kw("AND")
+                    hi = bound()
+                else:
+                    lo = bound()
+                    hi = ("current", 0)
+                frame = (fk.lower(), lo, hi)
+                # the RANGE spelling of the default frame IS the default
+                if frame == ("range", ("unbounded_preceding", None),
+                             ("current", 0)):
+                    frame = None
             self.expect_op(")")
-            return ("window", fname.lower(), args, tuple(part), tuple(order))
+            return ("window", fname.lower(), args, tuple(part),
+                    tuple(order), frame)
         if fname in ("REGR_COUNT", "REGR_SXX", "REGR_SYY", "COVAR_POP",
                      "COVAR_SAMP") and len(args) == 2:
             # bivariate aggregates over non-NULL PAIRS, rewritten onto the

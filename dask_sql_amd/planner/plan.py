@@ -332,7 +332,7 @@ class WindowSpec:
 
     def __init__(self, func: str, arg_idx, part_idx: list,
                  order_idx: list, out_name: str, out_type,
-                 offset: int = 1, default=None):
+                 offset: int = 1, default=None, frame=None):
         self.func = func                  # row_number|rank|...|lag|lead
         self.arg_idx = arg_idx            # int | None (ranking / COUNT(*))
         self.part_idx = list(part_idx)
@@ -341,6 +341,10 @@ class WindowSpec:
         self.out_type = out_type
         self.offset = offset              # LAG/LEAD row offset
         self.default = default            # LAG/LEAD boundary default
+        # explicit frame ("rows"|"range", lo, hi) with bounds
+        # ("unbounded_preceding"|"current"|"preceding"|"following", k);
+        # None = the reference's defaults (window.py:280-300)
+        self.frame = frame
 
 
 class WindowNode:

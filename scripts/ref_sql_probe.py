@@ -15,7 +15,9 @@ from dask_sql_amd.context import Context
 REF = Path("/root/reference/tests/integration")
 FILES = ["test_select.py", "test_filter.py", "test_groupby.py",
          "test_join.py", "test_sort.py", "test_union.py", "test_rex.py",
-         "test_complex.py", "test_distributeby.py"]
+         "test_complex.py", "test_distributeby.py",
+         "test_over.py",
+         "test_function.py"]
 
 c = Context()
 np.random.seed(42)

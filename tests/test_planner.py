@@ -405,8 +405,11 @@ def test_plan_errors_are_loud():
         c.explain("SELECT missing FROM t")
     with pytest.raises(KeyError):
         c.explain("SELECT a FROM missing_table")
+    # ROW_NUMBER without ORDER BY is legal (input order, like the
+    # reference); RANK still needs an ordering
+    c.explain("SELECT ROW_NUMBER() OVER (PARTITION BY a) FROM t")
     with pytest.raises(ValueError):
-        c.explain("SELECT ROW_NUMBER() OVER (PARTITION BY a) FROM t")
+        c.explain("SELECT RANK() OVER (PARTITION BY a) FROM t")
     with pytest.raises((ValueError, NotImplementedError)):
         c.explain("SELECT a FROM t WHERE a IN (SELECT a, b FROM t)")
 
@@ -692,3 +695,45 @@ def test_parser_truncation_errors_are_clean():
                 Parser(q[:cut]).parse()
             except (ValueError, NotImplementedError, KeyError):
                 pass
+
+
+def test_window_frame_apply_cpu():
+    """_frame_apply is pure pandas — verify the ROWS-frame computation
+    against hand-rolled windows (reference map_on_each_group semantics)."""
+    import types
+
+    import numpy as np
+    import pandas as pd
+
+    from dask_sql_amd.physical.rel_plugins import DaskWindowPlugin
+
+    plug = DaskWindowPlugin.__new__(DaskWindowPlugin)
+    df = pd.DataFrame({"p0": [0, 0, 0, 0, 1, 1, 1],
+                       "v": [1.0, 2.0, 3.0, 4.0, 10.0, 20.0, 30.0]})
+    grp = df.groupby(["p0"], dropna=False, sort=False)
+
+    def run(func, frame):
+        spec = types.SimpleNamespace(func=func, frame=frame, arg_idx=0)
+        return plug._frame_apply(df.copy(), grp, ["p0"], spec).tolist()
+
+    # 2 PRECEDING..CURRENT sums
+    got = run("sum", ("rows", ("preceding", 2), ("current", 0)))
+    assert got == [1, 3, 6, 9, 10, 30, 60]
+    # 1 PRECEDING..1 FOLLOWING
+    got = run("sum", ("rows", ("preceding", 1), ("following", 1)))
+    assert got == [3, 6, 9, 7, 30, 60, 50]
+    # UNBOUNDED..CURRENT max
+    got = run("max", ("rows", ("unbounded_preceding", None),
+                      ("current", 0)))
+    assert got == [1, 2, 3, 4, 10, 20, 30]
+    # CURRENT..UNBOUNDED FOLLOWING first_value = current
+    got = run("first_value", ("rows", ("current", 0),
+                              ("unbounded_following", None)))
+    assert got == [1, 2, 3, 4, 10, 20, 30]
+    # UNBOUNDED..UNBOUNDED last_value = partition tail
+    got = run("last_value", ("rows", ("unbounded_preceding", None),
+                             ("unbounded_following", None)))
+    assert got == [4, 4, 4, 4, 30, 30, 30]
+    # counts over 1 PRECEDING..CURRENT
+    got = run("count", ("rows", ("preceding", 1), ("current", 0)))
+    assert got == [1, 2, 2, 2, 1, 2, 2]

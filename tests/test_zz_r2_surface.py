@@ -636,3 +636,38 @@ def test_case_sensitive_quoted_aliases(ctx):
     assert out["AAA"].astype(np.int64).tolist() == [15, 26]
     assert out["aaa"].astype(np.int64).tolist() == [20, 40]
     assert out["aAa"].astype(np.int64).tolist() == [4, 5]
+
+
+def test_window_rows_frames(ctx):
+    # reference test_over.py ROWS BETWEEN cases — host rolling path
+    df = pd.DataFrame({"a": np.array([1, 2, 3, 4, 5], dtype=np.int64),
+                       "u": np.array([0, 0, 0, 1, 1], dtype=np.int64)})
+    ctx.create_table("zz_wf", df)
+    out = ctx.sql(
+        'SELECT a, '
+        'SUM(a) OVER (ORDER BY a ROWS BETWEEN 2 PRECEDING AND CURRENT '
+        'ROW) AS o1, '
+        'SUM(a) OVER (PARTITION BY u ORDER BY a ROWS BETWEEN 1 PRECEDING '
+        'AND 1 FOLLOWING) AS o2, '
+        'COUNT(a) OVER (ORDER BY a ROWS BETWEEN 1 PRECEDING AND CURRENT '
+        'ROW) AS o3 FROM zz_wf').compute()
+    out = out.sort_values("a").reset_index(drop=True)
+    assert out["o1"].astype(np.int64).tolist() == [1, 3, 6, 9, 12]
+    assert out["o2"].astype(np.int64).tolist() == [3, 6, 5, 9, 9]
+    assert out["o3"].astype(np.int64).tolist() == [1, 2, 2, 2, 2]
+
+
+def test_window_last_value_and_rowid_order(ctx):
+    df = pd.DataFrame({"u": np.array([1, 1, 2, 2], dtype=np.int64),
+                       "b": np.array([5.0, 7.0, 1.0, 3.0])})
+    ctx.create_table("zz_lv", df)
+    out = ctx.sql("SELECT u, b, "
+                  "LAST_VALUE(b) OVER (PARTITION BY u ORDER BY b) AS lv, "
+                  "SINGLE_VALUE(b) OVER (PARTITION BY u ORDER BY b) AS sv,"
+                  " ROW_NUMBER() OVER (PARTITION BY u) AS rn "
+                  "FROM zz_lv").compute()
+    out = out.sort_values(["u", "b"]).reset_index(drop=True)
+    # ordered default frame: LAST_VALUE = current row, SINGLE/FIRST = head
+    assert out["lv"].tolist() == [5.0, 7.0, 1.0, 3.0]
+    assert out["sv"].tolist() == [5.0, 5.0, 1.0, 1.0]
+    assert sorted(out[out.u == 1]["rn"].astype(np.int64).tolist()) == [1, 2]
