@@ -752,14 +752,27 @@ class Parser:
                         [("lit", unit, "VARCHAR"), a1, a2])
             # fold a literal count into an interval so the existing
             # date±INTERVAL machinery does the arithmetic
-            neg = False
-            while isinstance(a1, tuple) and a1[0] == "call"                     and a1[1] == "NEG":
-                neg = not neg
-                a1 = a1[2][0]
-            if not (isinstance(a1, tuple) and a1[0] == "lit"
-                    and isinstance(a1[1], int)):
-                raise ValueError(f"{fname} count must be an integer literal")
-            n = -a1[1] if neg else a1[1]
+            def _fold_int(a):
+                if not isinstance(a, tuple):
+                    return None
+                if a[0] == "lit" and isinstance(a[1], int):
+                    return a[1]
+                if a[0] == "call" and a[1] == "NEG":
+                    v = _fold_int(a[2][0])
+                    return None if v is None else -v
+                if a[0] == "call" and a[1] in ("+", "-", "*") \
+                        and len(a[2]) == 2:
+                    x_, y_ = _fold_int(a[2][0]), _fold_int(a[2][1])
+                    if x_ is None or y_ is None:
+                        return None
+                    return {"+": x_ + y_, "-": x_ - y_,
+                            "*": x_ * y_}[a[1]]
+                return None
+
+            n = _fold_int(a1)
+            if n is None:
+                raise ValueError(
+                    f"{fname} count must be an integer literal")
             return ("call", "+", [a2, ("interval", n, unit)])
         if fname == "POSITION":
             # POSITION(needle IN hay [FROM start]) — Calcite form
