@@ -2402,6 +2402,10 @@ class DaskWindowPlugin(BaseRelPlugin):
     _AGGS = {"sum", "count", "avg", "min", "max"}
 
     def _one(self, runtime, cols, n, spec):
+        if any(nf is True for nf in getattr(spec, "order_nf", [])):
+            # explicit NULLS FIRST on a window order key: host sort path
+            # honors per-key placement
+            return self._host_ordered(runtime, cols, n, spec)
         if getattr(spec, "frame", None) is None \
                 and spec.func == "last_value" and spec.order_idx \
                 and spec.arg_idx is not None:
@@ -2704,8 +2708,18 @@ class DaskWindowPlugin(BaseRelPlugin):
             df["p0"] = 0
             pnames = ["p0"]
         if onames:
-            df = df.sort_values(onames, ascending=asc, na_position="last",
-                                kind="mergesort")
+            nfs = list(getattr(spec, "order_nf", [])) or [None] * len(
+                onames)
+            if any(nf is True for nf in nfs):
+                # mixed per-key NULL placement: successive stable sorts,
+                # last key first
+                for name_, a_, nf in reversed(list(zip(onames, asc, nfs))):
+                    df = df.sort_values(
+                        name_, ascending=a_, kind="mergesort",
+                        na_position="first" if nf else "last")
+            else:
+                df = df.sort_values(onames, ascending=asc,
+                                    na_position="last", kind="mergesort")
         df = df.sort_values(pnames, na_position="last", kind="mergesort")
         grp = df.groupby(pnames, dropna=False, sort=False)
         f = spec.func

@@ -1135,7 +1135,8 @@ class Builder:
                 arg_idx = idx_of(args[0])
                 arg_t = pre_fields[arg_idx].getType().getSqlType()
             part_idx = [idx_of(p) for p in part]
-            order_idx = [(idx_of(o), desc) for o, desc in order]
+            order_idx = [(idx_of(o[0]), o[1]) for o in order]
+            order_nf = [o[2] if len(o) > 2 else None for o in order]
             if func in ("rank", "dense_rank", "lag", "lead") \
                     and not order_idx:
                 raise ValueError(f"{func.upper()} requires ORDER BY in OVER")
@@ -1171,9 +1172,12 @@ class Builder:
                         "RANGE frames with offsets (ROWS frames and the "
                         "default RANGE frame are supported)")
             name = f"w{len(specs)}__{func}"
-            specs.append(WindowSpec(func, arg_idx, part_idx, order_idx,
-                                    name, SqlType(ty), offset=offset,
-                                    default=default, frame=frame))
+            spec = WindowSpec(func, arg_idx, part_idx, order_idx,
+                              name, SqlType(ty), offset=offset,
+                              default=default, frame=frame)
+            spec.order_nf = order_nf  # per-key NULLS FIRST/LAST (None =
+            # engine default: nulls sort last)
+            specs.append(spec)
             win_out[repr(ast)] = name
         if len(pre_named) > len(in_fields):
             plan = LogicalPlan("Projection", [plan], RelDataType(pre_fields),

@@ -865,7 +865,12 @@ class Parser:
                     desc = bool(self.accept_kw("DESC"))
                     if not desc:
                         self.accept_kw("ASC")
-                    order.append((e, desc))
+                    nf = None  # engine default (NULLS LAST placement)
+                    if self.accept_kw("NULLS"):
+                        nf = self.accept_kw("FIRST") is not None
+                        if not nf:
+                            self.expect_kw("LAST")
+                    order.append((e, desc, nf))
                     if not self.accept_op(","):
                         break
             frame = None

@@ -737,3 +737,21 @@ def test_window_frame_apply_cpu():
     # counts over 1 PRECEDING..CURRENT
     got = run("count", ("rows", ("preceding", 1), ("current", 0)))
     assert got == [1, 2, 2, 2, 1, 2, 2]
+
+
+def test_sqlite_corpus_plans_cpu():
+    """Every query of the sqlite-differential corpus must PLAN on CPU
+    (execution runs on the GPU tier, tests/test_zz_sqlite_compat.py)."""
+    import pandas as pd
+
+    import tests.test_zz_sqlite_compat as m
+    from dask_sql_amd.context import Context
+
+    c = Context()
+    m.PLAN_ONLY = True
+    try:
+        for name in dir(m):
+            if name.startswith("test_sqlc_"):
+                getattr(m, name)(c)
+    finally:
+        m.PLAN_ONLY = False
