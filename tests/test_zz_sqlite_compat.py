@@ -264,3 +264,59 @@ def test_sqlc_integration_1(ctx):
         SELECT a, b, n, ROW_NUMBER() OVER (PARTITION BY b ORDER BY a,
             n) AS r
         FROM cte WHERE b IS NOT NULL ORDER BY b, r""", a=a)
+
+
+def test_sqlc_window_min_max(ctx):
+    for func in ["MIN", "MAX"]:
+        a = rand_df(100, a=float, b=(int, 50), c=(str, 50))
+        eq_sqlite(ctx, f"""
+            SELECT a,b,
+                {func}(b) OVER () AS a1,
+                {func}(b) OVER (PARTITION BY c) AS a2,
+                {func}(b+a) OVER (PARTITION BY c,b) AS a3,
+                {func}(b+a) OVER (PARTITION BY b ORDER BY a
+                    ROWS BETWEEN UNBOUNDED PRECEDING AND CURRENT ROW)
+                    AS a4,
+                {func}(b+a) OVER (PARTITION BY b ORDER BY a DESC
+                    ROWS BETWEEN 2 PRECEDING AND CURRENT ROW) AS a5
+            FROM a
+            ORDER BY a NULLS FIRST, b NULLS FIRST, c NULLS FIRST""",
+            a=a)
+        eq_sqlite(ctx, f"""
+            SELECT a,b,
+                {func}(b) OVER (ORDER BY a DESC
+                    ROWS BETWEEN 2 PRECEDING AND 1 PRECEDING) AS a6,
+                {func}(b) OVER (ORDER BY a DESC
+                    ROWS BETWEEN 2 PRECEDING AND 1 FOLLOWING) AS a7,
+                {func}(b) OVER (ORDER BY a DESC
+                    ROWS BETWEEN 2 PRECEDING AND UNBOUNDED FOLLOWING)
+                    AS a8
+            FROM a
+            ORDER BY a NULLS FIRST, b NULLS FIRST, c NULLS FIRST""",
+            a=a)
+
+
+def test_sqlc_window_count_frames(ctx):
+    a = rand_df(100, a=float, b=(int, 50), c=(str, 50))
+    eq_sqlite(ctx, """
+        SELECT a,b,
+            COUNT(b) OVER (PARTITION BY c) AS a2,
+            COUNT(b) OVER (PARTITION BY b ORDER BY a
+                ROWS BETWEEN UNBOUNDED PRECEDING AND CURRENT ROW) AS a4,
+            COUNT(b) OVER (ORDER BY a DESC
+                ROWS BETWEEN 2 PRECEDING AND 1 FOLLOWING) AS a7
+        FROM a
+        ORDER BY a NULLS FIRST, b NULLS FIRST, c NULLS FIRST""", a=a)
+
+
+def test_sqlc_window_sum_avg_partition(ctx):
+    a = rand_df(100, a=float, b=(int, 50), c=(str, 50))
+    eq_sqlite(ctx, """
+        SELECT a,b,
+            SUM(b) OVER () AS a1,
+            AVG(b) OVER (PARTITION BY c) AS a2,
+            SUM(b+a) OVER (PARTITION BY c,b) AS a3,
+            AVG(b+a) OVER (PARTITION BY b ORDER BY a
+                ROWS BETWEEN UNBOUNDED PRECEDING AND CURRENT ROW) AS a4
+        FROM a
+        ORDER BY a NULLS FIRST, b NULLS FIRST, c NULLS FIRST""", a=a)
