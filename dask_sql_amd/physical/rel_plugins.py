@@ -2413,8 +2413,11 @@ class DaskWindowPlugin(BaseRelPlugin):
             # reference's row-based expanding window makes LAST_VALUE the
             # current row's value (window.py LastValueOperation over
             # expanding) — a straight copy of the operand
-            return runtime.concat_columns([cols[spec.arg_idx]],
-                                          cols[spec.arg_idx].dtype)
+            src = cols[spec.arg_idx]
+            out = runtime.concat_columns([src], src.dtype)
+            if getattr(src, "dictionary", None) is not None:
+                out.dictionary = src.dictionary
+            return out
         if getattr(spec, "frame", None) is not None:
             # explicit ROWS frames run the reference's own rolling pandas
             # computation host-side (window.py:145-198 map_on_each_group)
