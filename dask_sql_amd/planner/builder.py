@@ -439,15 +439,39 @@ class Builder:
                      if (t.alias or t.name)}
         sub_quals |= {(j.table.alias or j.table.name).lower()
                       for j in sub.joins if (j.table.alias or j.table.name)}
+        # column names BOUND by the sub's own FROM (registered tables; a
+        # derived-table member makes the set unknowable → conservative:
+        # unqualified names stay local, only qualified correlation works)
+        local_cols = set()
+        local_known = True
+        for t in list(sub.from_tables) + [j.table for j in sub.joins]:
+            if getattr(t, "subquery", None) is not None or t.name is None:
+                local_known = False
+                continue
+            try:
+                local_cols |= {n.lower() for n, _ in
+                               self.catalog.get(t.name)}
+            except KeyError:
+                local_known = False
         conjs = self._conjuncts(sub.where)
 
         def outer_col(x):
-            return (isinstance(x, tuple) and x[0] == "col"
-                    and x[1] is not None and x[1].lower() not in sub_quals)
+            if not (isinstance(x, tuple) and x[0] == "col"):
+                return False
+            if x[1] is not None:
+                return x[1].lower() not in sub_quals
+            # unqualified outer reference (TPC-H Q2/Q20 style): a name the
+            # sub's own tables do not bind
+            return local_known and x[2].lower() not in local_cols
 
         def has_outer(ast):
-            return any(q is not None and q.lower() not in sub_quals
-                       for q, _ in self._tables_of(ast))
+            for q, n in self._tables_of(ast):
+                if q is not None and q.lower() not in sub_quals:
+                    return True
+                if q is None and local_known \
+                        and n.lower() not in local_cols:
+                    return True
+            return False
 
         local, inner_keys, outer_keys = [], [], []
         for cj in conjs:
@@ -800,9 +824,21 @@ class Builder:
                 return ast
             if ast[0] == "scalar_sub":
                 sub = ast[1]
+
+                def _has_agg(a):
+                    if not isinstance(a, tuple):
+                        return False
+                    if a[0] == "agg":
+                        return True
+                    if a[0] == "call":
+                        return any(_has_agg(x) for x in a[2])
+                    if a[0] == "cast":
+                        return _has_agg(a[1])
+                    return False
+
                 if (len(sub.items) == 1
                         and isinstance(sub.items[0][0], tuple)
-                        and sub.items[0][0][0] == "agg"
+                        and _has_agg(sub.items[0][0])
                         and not sub.group_by):
                     idx = ssub_n[0]
                     valname = f"__ssub{idx}"

@@ -671,3 +671,88 @@ def test_window_last_value_and_rowid_order(ctx):
     assert out["lv"].tolist() == [5.0, 7.0, 1.0, 3.0]
     assert out["sv"].tolist() == [5.0, 5.0, 1.0, 1.0]
     assert sorted(out[out.u == 1]["rn"].astype(np.int64).tolist()) == [1, 2]
+
+
+def test_q2_shape_correlated_min(ctx):
+    """Q2: correlated scalar MIN subquery with an UNQUALIFIED outer
+    reference, over a 4-table join."""
+    rng = np.random.default_rng(41)
+    part = pd.DataFrame({"p_partkey": np.arange(50, dtype=np.int64),
+                         "p_size": rng.integers(1, 10, 50).astype(np.int64)})
+    supplier = pd.DataFrame({
+        "s_suppkey": np.arange(20, dtype=np.int64),
+        "s_nationkey": rng.integers(0, 5, 20).astype(np.int64),
+        "s_acctbal": np.round(rng.random(20) * 1000, 2)})
+    partsupp = pd.DataFrame({
+        "ps_partkey": rng.integers(0, 50, 300).astype(np.int64),
+        "ps_suppkey": rng.integers(0, 20, 300).astype(np.int64),
+        "ps_supplycost": np.round(rng.random(300) * 100, 2)})
+    nation = pd.DataFrame({"n_nationkey": np.arange(5, dtype=np.int64),
+                           "n_regionkey": np.arange(5, dtype=np.int64) % 2})
+    ctx.create_table("q2_part", part)
+    ctx.create_table("q2_supplier", supplier)
+    ctx.create_table("q2_partsupp", partsupp)
+    ctx.create_table("q2_nation", nation)
+    out = ctx.sql("""
+        SELECT s_acctbal, s_suppkey, p_partkey
+        FROM q2_part, q2_supplier, q2_partsupp, q2_nation
+        WHERE p_partkey = ps_partkey AND s_suppkey = ps_suppkey
+          AND p_size = 5 AND s_nationkey = n_nationkey
+          AND ps_supplycost = (SELECT MIN(ps_supplycost) FROM q2_partsupp
+                               WHERE p_partkey = ps_partkey)
+        ORDER BY s_acctbal DESC, p_partkey LIMIT 10""").compute()
+    mn = partsupp.groupby("ps_partkey")["ps_supplycost"].min()
+    j = partsupp.merge(part[part.p_size == 5], left_on="ps_partkey",
+                       right_on="p_partkey")
+    j = j[np.isclose(j.ps_supplycost,
+                     mn.reindex(j.ps_partkey).to_numpy())]
+    j = j.merge(supplier, left_on="ps_suppkey", right_on="s_suppkey") \
+         .merge(nation, left_on="s_nationkey", right_on="n_nationkey")
+    exp = j.sort_values(["s_acctbal", "p_partkey"],
+                        ascending=[False, True]).head(10)
+    assert len(out) == len(exp)
+    np.testing.assert_allclose(out["s_acctbal"].to_numpy(np.float64),
+                               exp["s_acctbal"].to_numpy(), rtol=1e-9)
+
+
+def test_q20_shape_nested_in_correlated(ctx):
+    """Q20: IN subquery containing another IN and a correlated scalar
+    expression-over-SUM subquery."""
+    rng = np.random.default_rng(43)
+    part = pd.DataFrame({"p_partkey": np.arange(40, dtype=np.int64),
+                         "p_size": rng.integers(1, 20, 40).astype(np.int64)})
+    partsupp = pd.DataFrame({
+        "ps_partkey": rng.integers(0, 40, 200).astype(np.int64),
+        "ps_suppkey": rng.integers(0, 15, 200).astype(np.int64),
+        "ps_availqty": rng.integers(1, 100, 200).astype(np.int64)})
+    lineitem = pd.DataFrame({
+        "l_partkey": rng.integers(0, 40, 1000).astype(np.int64),
+        "l_suppkey": rng.integers(0, 15, 1000).astype(np.int64),
+        "l_quantity": rng.integers(1, 50, 1000).astype(np.int64)})
+    supplier = pd.DataFrame({"s_suppkey": np.arange(15, dtype=np.int64)})
+    ctx.create_table("q20_part", part)
+    ctx.create_table("q20_partsupp", partsupp)
+    ctx.create_table("q20_lineitem", lineitem)
+    ctx.create_table("q20_supplier", supplier)
+    out = ctx.sql("""
+        SELECT s_suppkey FROM q20_supplier
+        WHERE s_suppkey IN (
+          SELECT ps_suppkey FROM q20_partsupp
+          WHERE ps_partkey IN (SELECT p_partkey FROM q20_part
+                               WHERE p_size < 10)
+            AND ps_availqty > (SELECT 0.5 * SUM(l_quantity)
+                               FROM q20_lineitem
+                               WHERE l_partkey = ps_partkey
+                                 AND l_suppkey = ps_suppkey)
+        )""").compute()
+    small = set(part[part.p_size < 10].p_partkey)
+    half = lineitem.groupby(["l_partkey", "l_suppkey"])["l_quantity"] \
+        .sum() * 0.5
+    ps = partsupp[partsupp.ps_partkey.isin(small)]
+    keep = []
+    for r in ps.itertuples(index=False):
+        h = half.get((r.ps_partkey, r.ps_suppkey))
+        if h is not None and r.ps_availqty > h:
+            keep.append(r.ps_suppkey)
+    exp = sorted(set(keep))
+    assert sorted(out["s_suppkey"].astype(np.int64).tolist()) == exp
