@@ -756,3 +756,16 @@ def test_q20_shape_nested_in_correlated(ctx):
             keep.append(r.ps_suppkey)
     exp = sorted(set(keep))
     assert sorted(out["s_suppkey"].astype(np.int64).tolist()) == exp
+
+
+def test_numeric_leading_column_name(ctx):
+    # reference test_compatibility.py:1078 — quoted "1b" column
+    df = pd.DataFrame({"a": np.arange(5, dtype=np.int64),
+                       "1b": np.arange(5, dtype=np.int64)})
+    ctx.create_table("zz_numcol", df)
+    out = ctx.sql('SELECT "1b" AS x FROM zz_numcol').compute()
+    assert out["x"].astype(np.int64).tolist() == [0, 1, 2, 3, 4]
+    out = ctx.sql('SELECT (CASE WHEN "1b"=1 THEN 0 END) AS x '
+                  "FROM zz_numcol").compute()
+    vals = out["x"].tolist()
+    assert vals[1] == 0 and sum(pd.isna(v) for v in vals) == 4
