@@ -2423,6 +2423,7 @@ class DaskWindowPlugin(BaseRelPlugin):
             # computation host-side (window.py:145-198 map_on_each_group)
             return self._host_ordered(runtime, cols, n, spec)
         device_ok = (spec.func in self._AGGS and not spec.order_idx
+                     and spec.part_idx  # OVER () = whole-table agg → host
                      and all(cols[i].dtype in _INT_KINDS
                              for i in spec.part_idx))
         if device_ok:
