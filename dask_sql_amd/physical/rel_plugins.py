@@ -2759,6 +2759,18 @@ class DaskWindowPlugin(BaseRelPlugin):
                 tid = tie.ngroup()
                 df["_tid"] = tid
                 res = tid - grp["_tid"].transform("first") + 1
+        elif not onames and f in ("sum", "count", "avg", "min", "max"):
+            # no ORDER BY: the default frame is the WHOLE partition
+            # (reference window.py:280-300 unbounded..unbounded)
+            if f == "count" and spec.arg_idx is None:
+                res = grp["p0" if "p0" in df.columns else pnames[0]] \
+                    .transform("size")
+            elif f == "count":
+                res = grp["v"].transform("count")
+            else:
+                res = grp["v"].transform(
+                    {"sum": "sum", "avg": "mean", "min": "min",
+                     "max": "max"}[f])
         else:
             # running aggregates, default RANGE UNBOUNDED..CURRENT frame:
             # cumulative then broadcast the tie-group's last value (peers)
