@@ -137,7 +137,7 @@ def harness():
     # programs below see plausible calendar inputs (dense int64 col)
     cols[0] = (rng.integers(-30000, 60000, n_rows).astype(np.int64), None)
     progs = []
-    for k in range(120):
+    for k in range(180):
         e = gen(rng, "BOOLEAN" if k % 2 == 0 else "NUM", 4)
         c = R.RexCompiler(_dev_cols())
         try:

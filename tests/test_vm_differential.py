@@ -317,7 +317,7 @@ def _dev_cols():
             types.SimpleNamespace(dtype=rt.F64)]
 
 
-@pytest.mark.parametrize("seed", range(8))
+@pytest.mark.parametrize("seed", range(14))
 def test_vm_differential(seed):
     rng = np.random.default_rng(1234 + seed)
     n = 40
