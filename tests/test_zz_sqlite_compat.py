@@ -352,3 +352,25 @@ def test_sqlc_float_count_distinct_last(ctx):
     eq_sqlite(ctx, "SELECT COUNT(DISTINCT e) AS cd_e FROM a", a=a)
     eq_sqlite(ctx, "SELECT a, COUNT(DISTINCT e) AS cd FROM a GROUP BY a",
               a=a)
+
+
+def test_sqlc_agg_min_max(ctx):
+    a = rand_df(100, a=(int, 50), b=(str, 50), c=(int, 30), d=(str, 40),
+                e=(float, 40), g=(datetime, 40))
+    eq_sqlite(ctx, """
+        SELECT
+            MIN(a) AS min_a, MAX(a) AS max_a,
+            MIN(b) AS min_b, MAX(b) AS max_b,
+            MIN(c) AS min_c, MAX(c) AS max_c,
+            MIN(e) AS min_e, MAX(e) AS max_e,
+            MIN(g) AS min_g, MAX(g) AS max_g,
+            MIN(a+e) AS mix_1, MIN(a)+MIN(e) AS mix_2
+        FROM a""", a=a)
+    eq_sqlite(ctx, """
+        SELECT a, b, a+1 AS c,
+            MIN(c) AS min_c, MAX(c) AS max_c,
+            MIN(d) AS min_d, MAX(d) AS max_d,
+            MIN(e) AS min_e, MAX(e) AS max_e,
+            MIN(a+e) AS mix_1, MIN(a)+MIN(e) AS mix_2
+        FROM a GROUP BY a, b
+        ORDER BY a NULLS FIRST, b NULLS FIRST""", a=a)
