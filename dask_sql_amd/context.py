@@ -48,6 +48,16 @@ def _from_pandas(df: pd.DataFrame, date_columns=(), dictionaries=None):
             dictionary = list(s.cat.categories)
             if validity.all():
                 validity = None
+        elif isinstance(dt, pd.StringDtype):
+            # pandas "string" dtype → dictionary encoding like object
+            # strings (factorize maps pd.NA to code -1)
+            codes, uniques = pd.factorize(s)
+            validity = (codes >= 0).astype(np.uint8)
+            if validity.all():
+                validity = None
+            arr = np.where(codes >= 0, codes, 0).astype(np.int32)
+            dtype, sqlt = rt.I32, "VARCHAR"
+            dictionary = list(uniques)
         elif pd.api.types.is_extension_array_dtype(dt):
             # nullable Int*/UInt*/Float*/boolean
             mask = s.isna().to_numpy()
