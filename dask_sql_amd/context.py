@@ -22,12 +22,14 @@ register_defaults()
 
 
 class _HostColumn:
-    def __init__(self, arr, validity, sql_type, dtype, dictionary=None):
+    def __init__(self, arr, validity, sql_type, dtype, dictionary=None,
+                 tz=None):
         self.arr = arr
         self.validity = validity
         self.sql_type = sql_type
         self.dtype = dtype
         self.dictionary = dictionary
+        self.tz = tz  # original timezone of a tz-aware datetime column
 
 
 def _from_pandas(df: pd.DataFrame, date_columns=(), dictionaries=None):
@@ -48,6 +50,19 @@ def _from_pandas(df: pd.DataFrame, date_columns=(), dictionaries=None):
             dictionary = list(s.cat.categories)
             if validity.all():
                 validity = None
+        elif isinstance(dt, pd.DatetimeTZDtype):
+            # tz-aware datetimes: store UTC ns, reattach the zone at
+            # materialization (reference keeps the dtype through pandas)
+            nat = s.isna().to_numpy()
+            validity = (~nat).astype(np.uint8) if nat.any() else None
+            ns = s.dt.tz_convert("UTC").dt.tz_localize(None) \
+                .to_numpy().astype("datetime64[ns]").astype(np.int64)
+            arr = np.where(nat, 0, ns)
+            dtype, sqlt = rt.I64, "TIMESTAMP"
+            col = _HostColumn(np.ascontiguousarray(arr), validity, sqlt,
+                              dtype, None, tz=str(dt.tz))
+            cols[str(name)] = col
+            continue
         elif isinstance(dt, pd.StringDtype):
             # pandas "string" dtype → dictionary encoding like object
             # strings (factorize maps pd.NA to code -1)
@@ -205,6 +220,8 @@ class RegisteredTable:
                 col = runtime.upload_column(h.arr, h.validity, h.dtype)
                 if h.dictionary is not None:
                     col.dictionary = h.dictionary
+                if getattr(h, "tz", None) is not None:
+                    col.tz = h.tz
                 cols[n] = col
             self.device_table = DeviceTable(cols)
         return self.device_table

@@ -60,6 +60,9 @@ def _convert(arr, valid, col, sql_t):
                                      errors="coerce"))
         if valid is not None:
             s[~valid] = pd.NaT
+        tz = getattr(col, "tz", None)
+        if tz is not None:
+            s = s.dt.tz_localize("UTC").dt.tz_convert(tz)
         return s
     if valid is not None and not valid.all():
         # NULL-bearing numeric → float64 with NaN (pandas upcast semantics)

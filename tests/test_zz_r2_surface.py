@@ -771,3 +771,21 @@ def test_numeric_leading_column_name(ctx):
                   "FROM zz_numcol").compute()
     vals = out["x"].tolist()
     assert vals[1] == 0 and sum(pd.isna(v) for v in vals) == 4
+
+
+def test_timezone_roundtrip(ctx):
+    # reference test_select.py:116 — tz-aware columns survive SELECT
+    df = pd.DataFrame({
+        "timezone": pd.date_range("2014-08-01 09:00", freq="8h",
+                                  periods=6, tz="Europe/Berlin"),
+        "no_timezone": pd.date_range("2014-08-01 09:00", freq="8h",
+                                     periods=6),
+        "utc_timezone": pd.date_range("2014-08-01 09:00", freq="8h",
+                                      periods=6, tz="UTC")})
+    ctx.create_table("zz_tz", df)
+    out = ctx.sql("SELECT * FROM zz_tz").compute()
+    assert (pd.to_datetime(out["timezone"]) == df["timezone"]).all()
+    assert (pd.to_datetime(out["no_timezone"])
+            == df["no_timezone"]).all()
+    assert (pd.to_datetime(out["utc_timezone"])
+            == df["utc_timezone"]).all()
