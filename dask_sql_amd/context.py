@@ -76,7 +76,8 @@ def _from_pandas(df: pd.DataFrame, date_columns=(), dictionaries=None):
         elif pd.api.types.is_extension_array_dtype(dt):
             # nullable Int*/UInt*/Float*/boolean
             mask = s.isna().to_numpy()
-            base = s.fillna(0).to_numpy()
+            base = s.fillna(False if dt == pd.BooleanDtype() else 0) \
+                .to_numpy()
             if base.dtype == object:
                 base = base.astype(np.float64)
             validity = (~mask).astype(np.uint8) if mask.any() else None
