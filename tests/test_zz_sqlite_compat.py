@@ -374,3 +374,25 @@ def test_sqlc_agg_min_max(ctx):
             MIN(a+e) AS mix_1, MIN(a)+MIN(e) AS mix_2
         FROM a GROUP BY a, b
         ORDER BY a NULLS FIRST, b NULLS FIRST""", a=a)
+
+
+def test_sqlc_window_lead_lag_partition(ctx):
+    a = rand_df(100, a=float, b=(int, 50), c=(str, 50))
+    eq_sqlite(ctx, """
+        SELECT
+            LAG(b, 1) OVER (PARTITION BY c ORDER BY a) AS l1,
+            LEAD(b, 1) OVER (PARTITION BY c ORDER BY a) AS l2,
+            a, c FROM a
+        ORDER BY a NULLS FIRST, c NULLS FIRST""", a=a)
+
+
+def test_sqlc_window_ranks_partition(ctx):
+    a = rand_df(100, a=int, b=(float, 50), c=(str, 50))
+    eq_sqlite(ctx, """
+        SELECT a, b,
+            RANK() OVER (PARTITION BY a ORDER BY b NULLS LAST, c
+                NULLS LAST) AS r,
+            DENSE_RANK() OVER (PARTITION BY a ORDER BY b NULLS LAST, c
+                NULLS LAST) AS d
+        FROM a
+        ORDER BY a, b NULLS LAST, c NULLS LAST""", a=a)
