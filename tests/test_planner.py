@@ -755,3 +755,57 @@ def test_sqlite_corpus_plans_cpu():
                 getattr(m, name)(c)
     finally:
         m.PLAN_ONLY = False
+
+
+def test_ingest_materialize_roundtrip_cpu():
+    """_from_pandas → materialize._convert is the identity for every
+    supported input dtype (CPU halves of the device round trip: the arrays
+    _convert sees are exactly what upload/download would carry)."""
+    import types
+
+    import numpy as np
+    import pandas as pd
+
+    from dask_sql_amd.context import _from_pandas
+    from dask_sql_amd.materialize import _convert
+
+    df = pd.DataFrame({
+        "i64": np.array([1, -2, 3], dtype=np.int64),
+        "i16": np.array([1, 2, 3], dtype=np.int16),
+        "u8": np.array([0, 255, 7], dtype=np.uint8),
+        "f64": [1.5, np.nan, -2.25],
+        "f32": np.array([1.5, 2.5, 3.5], dtype=np.float32),
+        "b": [True, False, True],
+        "bn": pd.array([True, None, False], dtype="boolean"),
+        "in64": pd.array([5, None, -7], dtype="Int64"),
+        "s": ["x", None, "y"],
+        "cat": pd.Series(["a", "b", "a"]).astype("category"),
+        "sd": pd.array(["p", None, "q"], dtype="string"),
+        "d": pd.to_datetime(["2021-01-01", "1969-12-31", "2100-06-01"]),
+        "ts": pd.to_datetime(["2021-01-01 10:11:12", "1969-12-31 23:00:00",
+                              "2100-06-01 00:00:01"]),
+        "tz": pd.date_range("2014-08-01 09:00", periods=3, freq="8h",
+                            tz="Europe/Berlin"),
+    })
+    cols = _from_pandas(df)
+    for name, h in cols.items():
+        stub = types.SimpleNamespace(dictionary=h.dictionary,
+                                     tz=getattr(h, "tz", None))
+        valid = h.validity.astype(bool) if h.validity is not None else None
+        out = _convert(np.asarray(h.arr), valid, stub, h.sql_type)
+        orig = df[name]
+        if name == "tz":
+            assert (pd.to_datetime(out) == orig).all(), name
+            continue
+        if str(orig.dtype).startswith("datetime"):
+            assert (pd.to_datetime(out) == orig).all(), name
+            continue
+        o = pd.Series(out).reset_index(drop=True)
+        e = orig.reset_index(drop=True)
+        for a, b in zip(o, e):
+            if pd.isna(b):
+                assert pd.isna(a), (name, a, b)
+            elif isinstance(b, str):
+                assert str(a) == b, (name, a, b)
+            else:
+                assert float(a) == float(b), (name, a, b)
