@@ -943,6 +943,18 @@ class Builder:
             if e == ("star",):
                 for f in plan.getRowType().getFieldList()[:n_user_fields]:
                     items.append((("col", f.qualifier, f.getName()), None))
+            elif isinstance(e, tuple) and e[0] == "qstar":
+                # t.* — the named table/alias's columns only
+                q = e[1].lower()
+                hit = False
+                for f in plan.getRowType().getFieldList()[:n_user_fields]:
+                    if (f.qualifier or "").lower() == q:
+                        items.append((("col", f.qualifier, f.getName()),
+                                      None))
+                        hit = True
+                if not hit:
+                    raise KeyError(f"{e[1]}.*: no columns for that "
+                                   "qualifier in scope")
             else:
                 items.append((e, alias))
 

@@ -695,6 +695,8 @@ class Parser:
 
     def _colref_tail(self, name):
         if self.accept_op("."):
+            if self.accept_op("*"):
+                return ("qstar", name)  # t.* — qualified wildcard
             col = self._name()
             return ("col", name, col)
         return ("col", None, name)

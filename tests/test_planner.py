@@ -335,6 +335,8 @@ PLAN_BATTERY = [
     "SELECT a -- line comment\n FROM t /* block\n comment */ WHERE a > 0",
     "SELECT * FROM (SELECT a FROM t LIMIT 2) JOIN u ON a = c",
     "SELECT TIMESTAMPDIFF(DAY, a, b) FROM t",
+    "SELECT t.* FROM t",
+    "SELECT u.c, t.* FROM t JOIN u ON t.a = u.c",
     "SELECT a, REGR_COUNT(b, a) AS n, REGR_SXX(b, a) AS sxx, "
     "COVAR_POP(b, a) AS cp, COVAR_SAMP(b, a) AS cs FROM t GROUP BY a",
     "SELECT a, b, FROM t",

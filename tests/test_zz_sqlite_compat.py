@@ -130,6 +130,8 @@ def test_sqlc_basic_select(ctx):
     eq_sqlite(ctx, "SELECT * FROM a", a=df)
     eq_sqlite(ctx, "SELECT * FROM a AS x", a=df)
     eq_sqlite(ctx, "SELECT b AS bb, a+1-2*3.0/4 AS cc FROM a", a=df)
+    eq_sqlite(ctx, "SELECT b AS bb, a+1-2*3.0/4 AS cc, x.* FROM a AS x",
+              a=df)
 
 
 def test_sqlc_case_when(ctx):
