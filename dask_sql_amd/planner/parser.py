@@ -190,7 +190,7 @@ class Parser:
                 self.ctes[cname.lower()] = sub
                 if not self.accept_op(","):
                     break
-        stmt = self._set_tail(self.select_stmt())
+        stmt = self._set_tail(self._select_branch())
         if self.peek()[0] != "eof":
             raise ValueError(f"trailing tokens: {self.peek()}")
         if not isinstance(stmt, UnionStmt):
@@ -205,6 +205,20 @@ class Parser:
         last.order_by, last.limit, last.offset = [], None, 0
         return stmt
 
+    def _select_branch(self):
+        """One branch of a set expression: SELECT ... or a parenthesized
+        select / set expression."""
+        if self.peek() == ("op", "("):
+            save = self.i
+            self.next()
+            if self.peek() == ("kw", "SELECT") or self.peek() == ("op",
+                                                                  "("):
+                sub = self._set_tail(self._select_branch())
+                self.expect_op(")")
+                return sub
+            self.i = save  # a parenthesized expression, not a select
+        return self.select_stmt()
+
     def _set_tail(self, first):
         """UNION [ALL] / INTERSECT / EXCEPT chain after a select;
         INTERSECT binds tighter than UNION/EXCEPT (standard SQL)."""
@@ -212,8 +226,8 @@ class Parser:
             while self._accept_word("INTERSECT"):
                 if self.accept_kw("ALL"):
                     raise ValueError("INTERSECT ALL not supported")
-                left = UnionStmt([left, self.select_stmt()], alls=[False],
-                                 ops=["INTERSECT"])
+                left = UnionStmt([left, self._select_branch()],
+                                 alls=[False], ops=["INTERSECT"])
             return left
 
         stmt = chain(first)
@@ -227,7 +241,7 @@ class Parser:
                 op, allf = "EXCEPT", False
             else:
                 break
-            branches.append(chain(self.select_stmt()))
+            branches.append(chain(self._select_branch()))
             alls.append(allf)
             ops.append(op)
         if len(branches) == 1:
@@ -312,7 +326,10 @@ class Parser:
                 if not self.accept_op(","):
                     break
         if self.accept_kw("LIMIT"):
-            s.limit = int(self.next()[1])
+            if self.accept_kw("ALL"):
+                pass  # LIMIT ALL = unlimited
+            else:
+                s.limit = int(self.next()[1])
         if self.accept_kw("OFFSET"):
             s.offset = int(self.next()[1])
         return s
