@@ -514,6 +514,12 @@ class Context:
         """reference context.py:819 _get_ral (planner entry). Plans are
         immutable → cached per (sql, schema version)."""
         from dask_sql_amd import config
+        up = sql.upper()
+        if "CURRENT_" in up or "LOCALTIME" in up or "NOW(" in up \
+                or "RAND" in up:
+            # now()-style and random expressions fold at build — never
+            # serve them from the plan cache
+            return Builder(self.catalog, self.schema_name).build(sql)
         key = (sql, self._schema_version, config.plan_fingerprint())
         plan = self._plan_cache.get(key)
         if plan is None:

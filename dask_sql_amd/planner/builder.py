@@ -75,6 +75,29 @@ def _isoweek_host(is_date):
     return f
 
 
+def _rand_host(integer):
+    """RAND[_INTEGER] — reference call.py RandOperation (numpy RNG; a
+    literal seed makes it deterministic)."""
+    def f(*args):
+        import numpy as np_
+        import pandas as pd_
+        if integer:
+            high = args[0]
+            seed = args[1] if len(args) > 1 else None
+        else:
+            high = None
+            seed = args[0] if args else None
+        n = len(args[0])
+        sv = None
+        if seed is not None and len(seed) and not pd_.isna(seed.iloc[0]):
+            sv = int(seed.iloc[0])
+        rng = np_.random.RandomState(sv)
+        if integer:
+            return pd_.Series(rng.randint(0, int(high.iloc[0]), n))
+        return pd_.Series(rng.rand(n))
+    return f
+
+
 class Catalog:
     """What the builder needs from the Context's schema: table → fields,
     plus the registered scalar/aggregate UDFs (reference
@@ -99,6 +122,11 @@ class Catalog:
             _isoweek_host(False), "BIGINT", False, [("x", "TIMESTAMP")])
         self.functions["__isoweek_date__"] = (
             _isoweek_host(True), "BIGINT", False, [("x", "DATE")])
+        self.functions["__rand__"] = (
+            _rand_host(False), "DOUBLE", False, [("seed", "BIGINT")])
+        self.functions["__rand_integer__"] = (
+            _rand_host(True), "BIGINT", False,
+            [("high", "BIGINT"), ("seed", "BIGINT")])
 
     def add(self, name, fields):
         self.tables[name.lower()] = fields
@@ -209,6 +237,28 @@ class Builder:
             fn = self.catalog.functions.get(op.lower())
             if fn is not None:
                 return Call(f"UDF:{op.lower()}", ops, SqlType(fn[1]))
+            if op == "CURRENT_TIMESTAMP" and not ops:
+                # evaluated when the plan is BUILT (the reference evaluates
+                # pd.Timestamp.now() when the rex runs — once per query
+                # either way); _get_ral skips the plan cache for these
+                import pandas as _pd
+                return Literal(int(_pd.Timestamp.now().value),
+                               SqlType("TIMESTAMP"))
+            if op in ("RAND", "RANDOM", "RAND_INTEGER"):
+                # reference call.py RandOperation/RandIntegerOperation —
+                # host builtins (Python randomness IS the reference's
+                # execution model); optional leading seed argument
+                if op == "RAND_INTEGER":
+                    if len(ops) == 2:
+                        seed, high = ops
+                    else:
+                        seed, high = None, ops[0]
+                    args = [high] + ([seed] if seed is not None else [])
+                    return Call("UDF:__rand_integer__", args,
+                                SqlType("BIGINT"))
+                args = list(ops) if ops else [Literal(None,
+                                                       SqlType("NULL"))]
+                return Call("UDF:__rand__", args, SqlType("DOUBLE"))
             if op == "EXTRACT_WEEK":
                 x = ops[0]
                 name = "__isoweek_ts__" if _expr_type(x) == "TIMESTAMP" \

@@ -631,6 +631,15 @@ class Parser:
                 return ("cast", e, ty)
         if t[0] == "id":
             name = self._name()
+            if name.upper() in ("CURRENT_TIMESTAMP", "CURRENT_DATE",
+                                "CURRENT_TIME", "LOCALTIME",
+                                "LOCALTIMESTAMP", "NOW"):
+                # parenthesis-less niladic datetime functions (reference
+                # call.py maps all of them to pd.Timestamp.now())
+                if self.peek() == ("op", "("):
+                    self.next()
+                    self.expect_op(")")
+                return ("call", "CURRENT_TIMESTAMP", [])
             if name.upper() == "DECIMAL" and self.peek()[0] == "str":
                 # DECIMAL '100.2' typed literal (reference maps DECIMAL to
                 # float64, mappings.py SqlTypeName.DECIMAL)
@@ -793,6 +802,20 @@ class Parser:
                 raise ValueError(
                     f"{fname} count must be an integer literal")
             return ("call", "+", [a2, ("interval", n, unit)])
+        if fname in ("LTRIM", "RTRIM", "BTRIM"):
+            # mode-fixed trims (reference call.py TrimOperation variants)
+            e = self.expr()
+            ch = " "
+            if self.accept_op(","):
+                t = self.next()
+                if t[0] != "str":
+                    raise ValueError(f"{fname} trim set must be a string")
+                ch = t[1]
+            self.expect_op(")")
+            mode = {"LTRIM": "LEADING", "RTRIM": "TRAILING",
+                    "BTRIM": "BOTH"}[fname]
+            return ("call", "TRIM", [e, ("lit", mode, "VARCHAR"),
+                                     ("lit", ch, "VARCHAR")])
         if fname == "POSITION":
             # POSITION(needle IN hay [FROM start]) — Calcite form
             # needle parses below the IN-postfix level so `'a' IN a` is

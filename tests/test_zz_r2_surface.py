@@ -789,3 +789,33 @@ def test_timezone_roundtrip(ctx):
             == df["no_timezone"]).all()
     assert (pd.to_datetime(out["utc_timezone"])
             == df["utc_timezone"]).all()
+
+
+def test_trim_variants_and_now(ctx):
+    df = pd.DataFrame({"s": pd.Series(["  pad  ", "xxvxx"]
+                                      ).astype("category"),
+                       "v": np.arange(2, dtype=np.int64)})
+    ctx.create_table("zz_trimv", df)
+    out = ctx.sql("SELECT LTRIM(s) AS l, RTRIM(s) AS r, BTRIM(s) AS b, "
+                  "BTRIM(s, 'x') AS bx, v FROM zz_trimv").compute()
+    out = out.sort_values("v").reset_index(drop=True)
+    assert out["l"].tolist() == ["pad  ", "xxvxx"]
+    assert out["r"].tolist() == ["  pad", "xxvxx"]
+    assert out["b"].tolist() == ["pad", "xxvxx"]
+    assert out["bx"].tolist() == ["  pad  ", "v"]
+    out = ctx.sql("SELECT CURRENT_TIMESTAMP AS n, v FROM zz_trimv"
+                  ).compute()
+    now = pd.Timestamp.now()
+    got = pd.to_datetime(out["n"]).iloc[0]
+    assert abs((now - got).total_seconds()) < 3600
+
+
+def test_rand_exec(ctx):
+    df = pd.DataFrame({"a": np.arange(100, dtype=np.int64)})
+    ctx.create_table("zz_rand", df)
+    out = ctx.sql("SELECT RAND(0) AS r, RAND_INTEGER(0, 10) AS ri, a "
+                  "FROM zz_rand").compute()
+    r = out["r"].astype(float)
+    assert ((r >= 0) & (r < 1)).all() and r.nunique() > 50
+    ri = out["ri"].astype(np.int64)
+    assert ((ri >= 0) & (ri < 10)).all()
