@@ -549,6 +549,8 @@ class RexCompiler:
         if op == "EXTRACT_DATE" or op.startswith("FLOOR_TO_") \
                 or op.startswith("CEIL_TO_"):
             return self._compile_dt_trunc(op, ops)
+        if op == "LAST_DAY":
+            return self._compile_last_day(ops)
         if op.startswith("EXTRACT_"):
             return self._compile_extract_ext(op, ops)
         raise RexCompileError(f"operator {op} not supported on GPU path")
@@ -725,6 +727,57 @@ class RexCompiler:
         if is_ts:
             self._emit(OP_LIT_I64, 0, self._DAY_NS)
             self._emit(OP_MUL_I64)
+        return KI
+
+    def _compile_last_day(self, ops):
+        """LAST_DAY(x) = x + MonthEnd(1) (reference call.py last_day):
+        month end of x, ROLLING to the next month's end when x is already
+        on one (pandas anchor semantics). Day identity:
+        ld_days = MF(MF(days+1) + 31) - 1 with MF(g) = g - (DOM(g)-1);
+        time of day is preserved."""
+        t0 = getattr(ops[0], "getType", lambda: None)()
+        is_ts = t0 is not None and t0.getSqlType() == "TIMESTAMP"
+        x = ops[0]
+
+        def emit_days1():
+            # day count of x, plus one
+            self._emit_days(x, is_ts)
+            self._emit(OP_LIT_I64, 0, 1)
+            self._emit(OP_ADD_I64)
+
+        def emit_mf_days1():
+            # month start of (days+1)
+            emit_days1()
+            emit_days1()
+            self._emit(OP_DAY)
+            self._emit(OP_LIT_I64, 0, 1)
+            self._emit(OP_SUB_I64)
+            self._emit(OP_SUB_I64)
+
+        def emit_bumped():
+            emit_mf_days1()
+            self._emit(OP_LIT_I64, 0, 31)
+            self._emit(OP_ADD_I64)
+
+        if is_ts:
+            # time of day rides on top of the day result
+            k = self.compile(x)
+            if k != KI:
+                raise RexCompileError("LAST_DAY needs DATE/TIMESTAMP")
+            self._emit(OP_LIT_I64, 0, self._DAY_NS)
+            self._emit(OP_FLOORMOD_I64)
+        emit_bumped()
+        emit_bumped()
+        self._emit(OP_DAY)
+        self._emit(OP_LIT_I64, 0, 1)
+        self._emit(OP_SUB_I64)
+        self._emit(OP_SUB_I64)
+        self._emit(OP_LIT_I64, 0, 1)
+        self._emit(OP_SUB_I64)
+        if is_ts:
+            self._emit(OP_LIT_I64, 0, self._DAY_NS)
+            self._emit(OP_MUL_I64)
+            self._emit(OP_ADD_I64)
         return KI
 
     def _compile_extract_ext(self, op, ops):
@@ -977,7 +1030,7 @@ class RexCompiler:
                       "YEAR", "MONTH", "DAY", "DAYOFMONTH") \
                     or op.startswith("EXTRACT_") \
                     or op.startswith("FLOOR_TO_") \
-                    or op.startswith("CEIL_TO_"):
+                    or op.startswith("CEIL_TO_") or op == "LAST_DAY":
                 return KI
             if op == "COALESCE":
                 kids = [self._peek_kind(o) for o in expr.getOperands()]

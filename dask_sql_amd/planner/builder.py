@@ -397,7 +397,8 @@ class Builder:
             elif op.startswith("EXTRACT_") or op in (
                     "YEAR", "MONTH", "DAY", "DAYOFMONTH"):
                 ty = "BIGINT"
-            elif op.startswith("FLOOR_TO_") or op.startswith("CEIL_TO_"):
+            elif op.startswith("FLOOR_TO_") or op.startswith("CEIL_TO_") \
+                    or op == "LAST_DAY":
                 ty = _expr_type(ops[0])
             elif op == "TIMESTAMPDIFF":
                 # TIMESTAMPDIFF(unit, a, b) = truncated count of whole

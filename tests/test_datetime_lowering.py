@@ -219,3 +219,28 @@ def test_extract_extended_on_date():
         }.items():
             prog = _compile(op_name, "DATE")
             assert interp_prog(prog, days) == want, (op_name, s)
+
+
+def test_last_day_vs_pandas():
+    """LAST_DAY = x + MonthEnd(1), incl. the anchor ROLL when x is already
+    a month end (reference call.py last_day)."""
+    from pandas.tseries.offsets import MonthEnd
+
+    from tests.vm_interp import interp
+
+    def run(prog, v):
+        got, ok = interp(prog, [(np.array([v], dtype=np.int64), None)], 0)
+        assert ok
+        return got
+
+    pt = _compile("LAST_DAY", "TIMESTAMP")
+    for s in TS_SAMPLES:
+        ts = pd.Timestamp(s)
+        want = ts + MonthEnd(1)
+        assert run(pt, ts.value) == want.value, s
+    pdt = _compile("LAST_DAY", "DATE")
+    for s in ("2021-01-15", "2021-01-31", "2020-02-29", "1969-12-15"):
+        ts = pd.Timestamp(s)
+        days = (ts - pd.Timestamp(0)).days
+        want = (ts + MonthEnd(1) - pd.Timestamp(0)).days
+        assert run(pdt, days) == want, s

@@ -384,6 +384,7 @@ def test_plan_battery_dates():
         "SELECT FLOOR(ts TO DAY), CEIL(ts TO HOUR), FLOOR(ts TO YEAR), "
         "FLOOR(d TO MONTH), CEIL(ts TO MONTH), CEIL(d TO YEAR) FROM t",
         "SELECT EXTRACT(DATE FROM ts) AS dt FROM t",
+        "SELECT LAST_DAY(ts) AS ld, LAST_DAY(d) AS ldd FROM t",
         "SELECT EXTRACT(CENTURY FROM ts), EXTRACT(DOW FROM d), "
         "EXTRACT(DOY FROM ts), EXTRACT(QUARTER FROM d), "
         "EXTRACT(MILLISECOND FROM ts), EXTRACT(DECADE FROM ts) FROM t",
