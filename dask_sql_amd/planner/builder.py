@@ -259,6 +259,18 @@ class Builder:
                 args = list(ops) if ops else [Literal(None,
                                                        SqlType("NULL"))]
                 return Call("UDF:__rand__", args, SqlType("DOUBLE"))
+            if op in ("DATEPART", "DATE_PART") and len(ops) == 2 \
+                    and isinstance(ops[0], Literal):
+                # DATEPART('field', x) — function form of EXTRACT
+                # (reference call.py datepart → ExtractOperation)
+                field = str(ops[0].getValue()).upper()
+                if field.endswith("S") and field not in ("DOW",):
+                    field = field[:-1]
+                ex = Call(f"EXTRACT_{field}", [ops[1]], SqlType("BIGINT"))
+                ops = [ops[1]]
+                op = f"EXTRACT_{field}"
+                return Call(op, ops, SqlType(
+                    "DATE" if op == "EXTRACT_DATE" else "BIGINT"))
             if op == "EXTRACT_WEEK":
                 x = ops[0]
                 name = "__isoweek_ts__" if _expr_type(x) == "TIMESTAMP" \
