@@ -264,13 +264,12 @@ class Builder:
                 # DATEPART('field', x) — function form of EXTRACT
                 # (reference call.py datepart → ExtractOperation)
                 field = str(ops[0].getValue()).upper()
-                if field.endswith("S") and field not in ("DOW",):
+                if field.endswith("S"):
                     field = field[:-1]
-                ex = Call(f"EXTRACT_{field}", [ops[1]], SqlType("BIGINT"))
+                # fall through: EXTRACT_WEEK and the generic typing below
+                # treat this exactly like EXTRACT(field FROM x)
                 ops = [ops[1]]
                 op = f"EXTRACT_{field}"
-                return Call(op, ops, SqlType(
-                    "DATE" if op == "EXTRACT_DATE" else "BIGINT"))
             if op == "EXTRACT_WEEK":
                 x = ops[0]
                 name = "__isoweek_ts__" if _expr_type(x) == "TIMESTAMP" \
