@@ -819,3 +819,18 @@ def test_rand_exec(ctx):
     assert ((r >= 0) & (r < 1)).all() and r.nunique() > 50
     ri = out["ri"].astype(np.int64)
     assert ((ri >= 0) & (ri < 10)).all()
+
+
+def test_last_day_and_datepart_exec(ctx):
+    from pandas.tseries.offsets import MonthEnd
+    ts = pd.to_datetime(["2021-01-15 10:30:00", "2021-01-31 23:59:59",
+                         "2020-02-29 00:00:00"])
+    df = pd.DataFrame({"t": ts, "v": np.arange(3, dtype=np.int64)})
+    ctx.create_table("zz_ld", df)
+    out = ctx.sql("SELECT LAST_DAY(t) AS ld, DATEPART('YEAR', t) AS y, "
+                  "DATEPART('week', t) AS w, v FROM zz_ld").compute()
+    out = out.sort_values("v").reset_index(drop=True)
+    assert (pd.to_datetime(out["ld"]) == pd.Series(ts) + MonthEnd(1)).all()
+    assert out["y"].astype(np.int64).tolist() == [2021, 2021, 2020]
+    want_w = pd.Series(ts).dt.isocalendar().week.tolist()
+    assert out["w"].astype(np.int64).tolist() == [int(x) for x in want_w]
