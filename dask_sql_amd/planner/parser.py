@@ -985,8 +985,15 @@ class Parser:
                     ("call", "/", [("call", "*", [sm(y), sm(x)]), nf])])
             den = nf if fname == "COVAR_POP" else                 ("call", "-", [nf, ("lit", 1, "BIGINT")])
             return ("call", "/", [cov, den])
-        if fname == "MEAN":  # reference accepts MEAN as an AVG alias
-            return ("agg", "avg", args, distinct, filter_expr)
+        _AGG_ALIASES = {"MEAN": "avg", "STD": "stddev",
+                        "STDDEVPOP": "stddev_pop",
+                        "STDDEVSAMP": "stddev_samp",
+                        "VARIANCE_POP": "var_pop",
+                        "VARIANCEPOP": "var_pop",
+                        "FIRST": "single_value"}
+        if fname in _AGG_ALIASES:  # reference AGGREGATION_MAPPING aliases
+            return ("agg", _AGG_ALIASES[fname], args, distinct,
+                    filter_expr)
         if fname in AGG_FUNCS:
             return ("agg", fname.lower(), args, distinct, filter_expr)
         return ("call", fname, args)
