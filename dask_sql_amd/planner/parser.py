@@ -584,6 +584,12 @@ class Parser:
             if t[1] == "NULL":
                 self.next()
                 return ("lit", None, "NULL")
+            if t[1] in ("FIRST", "LAST") \
+                    and self.toks[self.i + 1] == ("op", "("):
+                # FIRST(x) aggregate — the word is reserved for NULLS
+                # FIRST/LAST but callable as a function name
+                self.next()
+                return self._func_call(t[1])
             if t[1] in ("DATE", "TIMESTAMP"):
                 # DATE '2026-01-01' is a literal; a bare DATE/TIMESTAMP not
                 # followed by a string is a column reference (a column
