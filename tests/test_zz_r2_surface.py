@@ -97,7 +97,7 @@ def test_datetime_trunc_exec(ctx):
     ts = pd.to_datetime([
         "2021-02-01 13:45:12.345", "2020-02-29 23:59:59.999",
         "1969-07-20 20:17:40", "2000-12-31 00:00:00",
-        "2021-01-01 00:00:00"])
+        "2021-01-01 00:00:00"], format="mixed")
     df = pd.DataFrame({"ts": ts, "v": np.arange(5, dtype=np.int64)})
     ctx.create_table("tdt", df)
     out = ctx.sql("SELECT FLOOR(ts TO DAY) AS fd, CEIL(ts TO HOUR) AS ch, "
