@@ -398,3 +398,22 @@ def test_sqlc_window_ranks_partition(ctx):
                 NULLS LAST) AS d
         FROM a
         ORDER BY a, b NULLS LAST, c NULLS LAST""", a=a)
+
+
+def test_sqlc_agg_count_group_expr(ctx):
+    a = rand_df(100, a=int, b=(str, 50), c=(int, 30), d=(str, 40),
+                e=(float, 40))
+    eq_sqlite(ctx, """
+        SELECT a, b, a+1 AS c, COUNT(c) AS cnt_c, COUNT(d) AS cnt_d
+        FROM a GROUP BY a, b""", a=a)
+    eq_sqlite(ctx, """
+        SELECT b, COUNT(DISTINCT a) AS cnt_a, COUNT(DISTINCT e) AS cnt_e
+        FROM a GROUP BY b""", a=a)
+
+
+def test_sqlc_order_by_no_limit(ctx):
+    a = rand_df(100, a=(int, 20), b=(str, 30))
+    eq_sqlite(ctx, "SELECT * FROM a ORDER BY a NULLS LAST, b NULLS LAST",
+              a=a)
+    eq_sqlite(ctx, "SELECT b, a FROM a ORDER BY b DESC NULLS LAST, "
+                   "a ASC NULLS FIRST", a=a)
