@@ -812,3 +812,22 @@ def test_ingest_materialize_roundtrip_cpu():
                 assert str(a) == b, (name, a, b)
             else:
                 assert float(a) == float(b), (name, a, b)
+
+
+def test_plan_audit_gpu_suites_cpu():
+    """The GPU suites' registrations + first queries must PLAN on CPU
+    (scripts/plan_audit_gpu_tests.py in a subprocess — it patches
+    Context.sql class-wide). The small allowed-failure budget covers the
+    tests that construct a live runtime before their first query."""
+    import re
+    import subprocess
+    import sys
+
+    r = subprocess.run([sys.executable, "scripts/plan_audit_gpu_tests.py"],
+                       capture_output=True, text=True, timeout=600,
+                       cwd="/root/repo")
+    m = re.search(r"TOTAL planned-ok (\d+), not-auditable (\d+), "
+                  r"FAILED (\d+)", r.stdout)
+    assert m, r.stdout[-2000:]
+    assert int(m.group(1)) >= 150, r.stdout[-2000:]
+    assert int(m.group(3)) <= 8, r.stdout[-2000:]
