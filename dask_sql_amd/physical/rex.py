@@ -583,7 +583,8 @@ class RexCompiler:
         is_ts = t0 is not None and t0.getSqlType() == "TIMESTAMP"
         x = ops[0]
         sub_ns = {"HOUR": 3_600_000_000_000, "MINUTE": 60_000_000_000,
-                  "SECOND": 1_000_000_000}
+                  "SECOND": 1_000_000_000, "MILLISECOND": 1_000_000,
+                  "MICROSECOND": 1_000}
         if op == "EXTRACT_DATE":
             self._emit_days(x, is_ts)
             return KI

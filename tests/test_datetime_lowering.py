@@ -244,3 +244,44 @@ def test_last_day_vs_pandas():
         days = (ts - pd.Timestamp(0)).days
         want = (ts + MonthEnd(1) - pd.Timestamp(0)).days
         assert run(pdt, days) == want, s
+
+
+def test_reference_unit_golden_vectors():
+    """The reference's own unit expectations (tests/unit/test_call.py:202-
+    238, datetime(2021,10,3,15,53,42,47)) — EXTRACT fields and sub-day
+    FLOOR/CEIL including MILLISECOND."""
+    import datetime as dtm
+
+    from tests.vm_interp import interp
+
+    def run(op_name, ty, v):
+        prog = _compile(op_name, ty)
+        got, ok = interp(prog, [(np.array([v], dtype=np.int64), None)], 0)
+        assert ok
+        return got
+
+    ts = pd.Timestamp(dtm.datetime(2021, 10, 3, 15, 53, 42, 47))
+    v = ts.value
+    assert run("EXTRACT_CENTURY", "TIMESTAMP", v) == 20
+    assert run("EXTRACT_DECADE", "TIMESTAMP", v) == 202
+    assert run("EXTRACT_DOW", "TIMESTAMP", v) == 0
+    assert run("EXTRACT_DOY", "TIMESTAMP", v) == 276
+    assert run("EXTRACT_MILLENNIUM", "TIMESTAMP", v) == 2
+    assert run("EXTRACT_MICROSECOND", "TIMESTAMP", v) == 47
+    assert run("EXTRACT_MILLISECOND", "TIMESTAMP", v) == 47000
+    assert run("EXTRACT_QUARTER", "TIMESTAMP", v) == 4
+    d = dtm.datetime
+    assert run("CEIL_TO_DAY", "TIMESTAMP", v) == \
+        pd.Timestamp(d(2021, 10, 4)).value
+    assert run("CEIL_TO_HOUR", "TIMESTAMP", v) == \
+        pd.Timestamp(d(2021, 10, 3, 16)).value
+    assert run("CEIL_TO_MINUTE", "TIMESTAMP", v) == \
+        pd.Timestamp(d(2021, 10, 3, 15, 54)).value
+    assert run("CEIL_TO_SECOND", "TIMESTAMP", v) == \
+        pd.Timestamp(d(2021, 10, 3, 15, 53, 43)).value
+    assert run("CEIL_TO_MILLISECOND", "TIMESTAMP", v) == \
+        pd.Timestamp(d(2021, 10, 3, 15, 53, 42, 1000)).value
+    assert run("FLOOR_TO_DAY", "TIMESTAMP", v) == \
+        pd.Timestamp(d(2021, 10, 3)).value
+    assert run("FLOOR_TO_MILLISECOND", "TIMESTAMP", v) == \
+        pd.Timestamp(d(2021, 10, 3, 15, 53, 42)).value

@@ -831,3 +831,49 @@ def test_plan_audit_gpu_suites_cpu():
     assert m, r.stdout[-2000:]
     assert int(m.group(1)) >= 150, r.stdout[-2000:]
     assert int(m.group(3)) <= 8, r.stdout[-2000:]
+
+
+def test_reference_string_op_unit_vectors():
+    """reference tests/unit/test_call.py:178-199 string-op expectations,
+    evaluated through the dictionary-function compilers the engine uses."""
+    from dask_sql_amd.physical.rex import dict_int_fn, dict_string_fn
+    from dask_sql_amd.planner.plan import Call, InputRef, Literal, SqlType
+
+    a = "a normal string"
+    dicts = [[a]]
+
+    def sfn(op, *lits):
+        e = Call(op, [InputRef(0, SqlType("VARCHAR"))]
+                 + [Literal(v, SqlType("VARCHAR" if isinstance(v, str)
+                                       else "BIGINT")) for v in lits],
+                 SqlType("VARCHAR"))
+        i, f = dict_string_fn(e, dicts)
+        return f(a)
+
+    def ifn(op, *lits):
+        e = Call(op, [InputRef(0, SqlType("VARCHAR"))]
+                 + [Literal(v, SqlType("VARCHAR" if isinstance(v, str)
+                                       else "BIGINT")) for v in lits],
+                 SqlType("BIGINT"))
+        i, f = dict_int_fn(e, dicts)
+        return f(a)
+
+    assert ifn("CHAR_LENGTH") == 15
+    assert sfn("UPPER") == "A NORMAL STRING"
+    assert sfn("LOWER") == "a normal string"
+    # POSITION(needle IN hay FROM start) — operand order (hay, needle[, n])
+    assert ifn("POSITION", "a", 4) == 7
+    assert ifn("POSITION", "ZL") == 0
+    assert sfn("TRIM", "BOTH", "a") == " normal string"
+    assert sfn("TRIM", "LEADING", "a") == " normal string"
+    assert sfn("TRIM", "TRAILING", "a") == "a normal string"
+    assert sfn("OVERLAY", "XXX", 2) == "aXXXrmal string"
+    assert sfn("OVERLAY", "XXX", 2, 4) == "aXXXmal string"
+    assert sfn("OVERLAY", "XXX", 2, 1) == "aXXXnormal string"
+    assert sfn("SUBSTRING", -1) == "a normal string"
+    assert sfn("SUBSTRING", 10) == "string"
+    assert sfn("SUBSTRING", 2) == " normal string"
+    assert sfn("SUBSTRING", 2, 2) == " n"
+    assert sfn("INITCAP") == "A Normal String"
+    assert sfn("REPLACE", "nor", "") == "a mal string"
+    assert sfn("REPLACE", "normal", "new") == "a new string"
