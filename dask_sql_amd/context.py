@@ -425,6 +425,16 @@ class Context:
                                   list(parameters or []))
         self._schema_version += 1
 
+    def alter_table(self, old_table_name, new_table_name,
+                    schema_name=None):
+        """reference context.py:599 — rename a registered table."""
+        key = old_table_name.lower()
+        t = self.tables.pop(key)  # KeyError like the reference
+        self.tables[new_table_name.lower()] = t
+        self.catalog.drop(key)
+        self.catalog.add(new_table_name, [(n, ty) for n, ty in t.fields()])
+        self._schema_version += 1
+
     def drop_table(self, table_name: str):
         self.tables.pop(table_name.lower(), None)
         self.catalog.drop(table_name)
@@ -465,6 +475,19 @@ class Context:
                 })
             from dask_sql_amd.datacontainer import HostDataContainer
             return ResultFrame(HostDataContainer(pdf), None, self)
+        # ALTER TABLE [IF EXISTS] x RENAME TO y (reference DDL alter.py)
+        alm = re.match(r"\s*ALTER\s+TABLE\s+(IF\s+EXISTS\s+)?(\w+)\s+"
+                       r"RENAME\s+TO\s+(\w+)\s*;?\s*$", sql,
+                       re.IGNORECASE)
+        if alm:
+            from dask_sql_amd.datacontainer import HostDataContainer
+            try:
+                self.alter_table(alm.group(2), alm.group(3))
+            except KeyError:
+                if not alm.group(1):
+                    raise
+            return ResultFrame(HostDataContainer(pd.DataFrame()), None,
+                               self)
         # DROP TABLE (reference rel/custom/drop.py)
         dm = re.match(r"\s*DROP\s+TABLE\s+(?:IF\s+EXISTS\s+)?(\w+)\s*;?\s*$",
                       sql, re.IGNORECASE)
