@@ -877,3 +877,19 @@ def test_reference_string_op_unit_vectors():
     assert sfn("INITCAP") == "A Normal String"
     assert sfn("REPLACE", "nor", "") == "a mal string"
     assert sfn("REPLACE", "normal", "new") == "a new string"
+
+
+def test_alter_table_rename():
+    """reference tests/unit/test_context.py:301 — ALTER TABLE RENAME."""
+    import pandas as pd
+    import pytest
+
+    from dask_sql_amd.context import Context
+    c = Context()
+    c.create_table("maths", pd.DataFrame({"a": [1]}))
+    c.sql("ALTER TABLE maths RENAME TO physics")
+    assert "physics" in c.tables and "maths" not in c.tables
+    c.explain("SELECT a FROM physics")
+    with pytest.raises(KeyError):
+        c.sql("ALTER TABLE four_legs RENAME TO two_legs")
+    c.sql("ALTER TABLE IF EXISTS alien RENAME TO humans")
