@@ -489,6 +489,14 @@ class Context:
             return ResultFrame(HostDataContainer(pd.DataFrame()), None,
                                self)
         # DROP TABLE (reference rel/custom/drop.py)
+        dvm = re.match(r"\s*DROP\s+VIEW\s+(?:IF\s+EXISTS\s+)?(\w+)"
+                       r"\s*;?\s*$", sql, re.IGNORECASE)
+        if dvm:
+            from dask_sql_amd.datacontainer import HostDataContainer
+            self.catalog.views.pop(dvm.group(1).lower(), None)
+            self._schema_version += 1
+            return ResultFrame(HostDataContainer(pd.DataFrame()), None,
+                               self)
         dm = re.match(r"\s*DROP\s+TABLE\s+(?:IF\s+EXISTS\s+)?(\w+)\s*;?\s*$",
                       sql, re.IGNORECASE)
         if dm:
@@ -516,6 +524,19 @@ class Context:
             out = pd.concat([stats, extra])
             from dask_sql_amd.datacontainer import HostDataContainer
             return ResultFrame(HostDataContainer(out), None, self)
+        # CREATE [OR REPLACE] VIEW <name> AS <select> — a view re-plans
+        # its SELECT at every use (reference rel/custom/
+        # create_memory_table.py CreateView, persist=False)
+        vm = re.match(r"\s*CREATE\s+(?:OR\s+REPLACE\s+)?VIEW\s+(\w+)"
+                      r"\s+AS\s*\(?\s*(SELECT.*?)\)?\s*;?\s*$", sql,
+                      re.IGNORECASE | re.DOTALL)
+        if vm:
+            from dask_sql_amd.datacontainer import HostDataContainer
+            self._get_ral(vm.group(2))  # validate eagerly
+            self.catalog.views[vm.group(1).lower()] = vm.group(2)
+            self._schema_version += 1
+            return ResultFrame(HostDataContainer(pd.DataFrame()), None,
+                               self)
         # CREATE TABLE <name> AS <select> (reference DDL create_table.py)
         m = re.match(r"\s*CREATE\s+(?:OR\s+REPLACE\s+)?TABLE\s+(\w+)\s+AS\s*"
                      r"\(?\s*(SELECT.*?)\)?\s*;?\s*$", sql,

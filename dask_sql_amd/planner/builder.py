@@ -106,6 +106,8 @@ class Catalog:
 
     def __init__(self):
         self.tables: dict[str, list[tuple[str, str]]] = {}
+        self.views: dict[str, str] = {}  # name → view SQL (re-planned
+        # at every use, like the reference's CreateView persist=False)
         self.functions: dict[str, tuple] = {}     # name → (f, ret_sql, row_udf)
         self.aggregations: dict[str, tuple] = {}  # name → (obj, ret_sql)
         # hidden builtins: calendar month arithmetic on COLUMNS executes
@@ -159,6 +161,13 @@ class Builder:
 
     # ----------------------------------------------------------------- scans
     def _scan(self, tr: TableRef) -> LogicalPlan:
+        if getattr(tr, "subquery", None) is None and tr.name \
+                and tr.name.lower() in self.catalog.views:
+            # view reference: inline the stored SELECT as a derived table
+            from dask_sql_amd.planner.parser import parse_sql as _ps
+            sub = _ps(self.catalog.views[tr.name.lower()])
+            tr = TableRef(name=None, alias=tr.alias or tr.name,
+                          subquery=sub)
         if getattr(tr, "subquery", None) is not None:
             # derived table: build the sub-select, requalify its output with
             # the alias (reference: DataFusion subquery alias rel)

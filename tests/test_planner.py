@@ -893,3 +893,22 @@ def test_alter_table_rename():
     with pytest.raises(KeyError):
         c.sql("ALTER TABLE four_legs RENAME TO two_legs")
     c.sql("ALTER TABLE IF EXISTS alien RENAME TO humans")
+
+
+def test_create_view():
+    """CREATE VIEW re-plans its SELECT per use (reference CreateView,
+    persist=False); DROP VIEW unregisters."""
+    import pandas as pd
+    import pytest
+
+    from dask_sql_amd.context import Context
+    c = Context()
+    c.create_table("t", pd.DataFrame({"a": [1, 2, 3], "b": [1.0, 2.0, 3.0]}))
+    c.sql("CREATE VIEW big AS SELECT a, b FROM t WHERE a > 1")
+    c.explain("SELECT a, COUNT(*) AS n FROM big GROUP BY a")
+    c.explain("SELECT x.a FROM big x JOIN big y ON x.a = y.a")
+    c.sql("CREATE OR REPLACE VIEW big AS SELECT a FROM t")
+    c.explain("SELECT a FROM big")
+    c.sql("DROP VIEW big")
+    with pytest.raises(KeyError):
+        c.explain("SELECT a FROM big")

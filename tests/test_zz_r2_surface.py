@@ -834,3 +834,15 @@ def test_last_day_and_datepart_exec(ctx):
     assert out["y"].astype(np.int64).tolist() == [2021, 2021, 2020]
     want_w = pd.Series(ts).dt.isocalendar().week.tolist()
     assert out["w"].astype(np.int64).tolist() == [int(x) for x in want_w]
+
+
+def test_create_view_exec(ctx):
+    df = pd.DataFrame({"a": np.arange(10, dtype=np.int64),
+                       "b": np.arange(10, dtype=np.int64) % 3})
+    ctx.create_table("zz_vt", df)
+    ctx.sql("CREATE VIEW zz_vv AS SELECT a, b FROM zz_vt WHERE a >= 4")
+    out = ctx.sql("SELECT b, COUNT(*) AS n FROM zz_vv GROUP BY b").compute()
+    exp = df[df.a >= 4].groupby("b").size()
+    got = dict(zip(out["b"].astype(np.int64), out["n"].astype(np.int64)))
+    assert got == exp.to_dict()
+    ctx.sql("DROP VIEW zz_vv")
