@@ -537,6 +537,28 @@ class Context:
             self._schema_version += 1
             return ResultFrame(HostDataContainer(pd.DataFrame()), None,
                                self)
+        # CREATE TABLE <name> WITH (location=..., format=..., ...) —
+        # reference rel/custom/create_table.py (test_create.py:14-40)
+        cwm = re.match(r"\s*CREATE\s+(?:OR\s+REPLACE\s+)?TABLE\s+(\w+)"
+                       r"\s+WITH\s*\((.*)\)\s*;?\s*$", sql,
+                       re.IGNORECASE | re.DOTALL)
+        if cwm:
+            from dask_sql_amd.datacontainer import HostDataContainer
+            kv = {}
+            for m_ in re.finditer(r"(\w+)\s*=\s*(?:'([^']*)'|(\w+))",
+                                  cwm.group(2)):
+                kv[m_.group(1).lower()] = m_.group(2) \
+                    if m_.group(2) is not None else m_.group(3)
+            loc = kv.get("location")
+            if loc is None:
+                raise ValueError(
+                    "CREATE TABLE ... WITH needs a location")
+            self.create_table(cwm.group(1), loc,
+                              format=kv.get("format"),
+                              persist=str(kv.get("persist", "false")
+                                          ).lower() == "true")
+            return ResultFrame(HostDataContainer(pd.DataFrame()), None,
+                               self)
         # CREATE TABLE <name> AS <select> (reference DDL create_table.py)
         m = re.match(r"\s*CREATE\s+(?:OR\s+REPLACE\s+)?TABLE\s+(\w+)\s+AS\s*"
                      r"\(?\s*(SELECT.*?)\)?\s*;?\s*$", sql,

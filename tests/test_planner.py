@@ -912,3 +912,28 @@ def test_create_view():
     c.sql("DROP VIEW big")
     with pytest.raises(KeyError):
         c.explain("SELECT a FROM big")
+
+
+def test_create_table_with_location():
+    """reference test_create.py:14-40 — CREATE TABLE ... WITH (location,
+    format) registers from a file path (parquet via arrow, csv via
+    pandas)."""
+    import os
+    import tempfile
+
+    import pandas as pd
+
+    from dask_sql_amd.context import Context
+    df = pd.DataFrame({"a": [1, 2], "b": [0.5, 1.5]})
+    d = tempfile.mkdtemp()
+    pqf = os.path.join(d, "x.parquet")
+    csvf = os.path.join(d, "y.csv")
+    df.to_parquet(pqf)
+    df.to_csv(csvf, index=False)
+    c = Context()
+    c.sql(f"CREATE TABLE tp WITH (location = '{pqf}', "
+          f"format = 'parquet')")
+    c.sql(f"CREATE TABLE tc WITH (location = '{csvf}', format = 'csv', "
+          f"gpu = False)")
+    c.explain("SELECT a, b FROM tp")
+    c.explain("SELECT a FROM tc")
