@@ -442,7 +442,17 @@ class Context:
 
     # -- reference context.py:482 sql --------------------------------------
     def sql(self, sql: str, return_futures: bool = True,
-            config_options=None) -> ResultFrame:
+            config_options=None, dataframes=None) -> ResultFrame:
+        if dataframes:
+            # reference context.py sql(dataframes=...): register inline
+            for name, frame in dataframes.items():
+                self.create_table(name, frame)
+        em = re.match(r"\s*EXPLAIN\s+(.*)$", sql,
+                      re.IGNORECASE | re.DOTALL)
+        if em:
+            # EXPLAIN <select> returns the plan STRING
+            # (reference test_explain.py:13-23)
+            return self.explain(em.group(1))
         if config_options:
             # per-query overrides, scoped like the reference's
             # dask.config.set(config_options) (context.py:519); unknown or

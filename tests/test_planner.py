@@ -937,3 +937,18 @@ def test_create_table_with_location():
           f"gpu = False)")
     c.explain("SELECT a, b FROM tp")
     c.explain("SELECT a FROM tc")
+
+
+def test_explain_statement_and_inline_dataframes():
+    """reference test_explain.py:13-23 — EXPLAIN returns the plan string;
+    sql(dataframes=...) registers frames inline."""
+    import pandas as pd
+
+    from dask_sql_amd.context import Context
+    c = Context()
+    c.create_table("df", pd.DataFrame({"a": [1, 2, 3]}))
+    s = c.sql("EXPLAIN SELECT * FROM df")
+    assert isinstance(s, str) and "Projection" in s
+    s2 = c.sql("EXPLAIN SELECT MIN(a) AS a_min FROM other_df GROUP BY a",
+               dataframes={"other_df": pd.DataFrame({"a": [1]})})
+    assert isinstance(s2, str) and "a_min" in s2 or "MIN" in s2
