@@ -441,12 +441,19 @@ class Context:
         self._schema_version += 1
 
     # -- reference context.py:482 sql --------------------------------------
-    def sql(self, sql: str, return_futures: bool = True,
-            config_options=None, dataframes=None) -> ResultFrame:
+    def sql(self, sql, return_futures: bool = True,
+            config_options=None, dataframes=None, gpu: bool = True):
+        # `gpu` is accepted for signature parity (context.py:482-489);
+        # every table is device-resident here
         if dataframes:
             # reference context.py sql(dataframes=...): register inline
             for name, frame in dataframes.items():
                 self.create_table(name, frame)
+        if not isinstance(sql, str):
+            # a LogicalPlan built earlier (reference accepts those too)
+            dc = RelConverter.convert(sql, context=self)
+            res = ResultFrame(dc, sql, self)
+            return res.compute() if not return_futures else res
         em = re.match(r"\s*EXPLAIN\s+(.*)$", sql,
                       re.IGNORECASE | re.DOTALL)
         if em:
@@ -580,7 +587,8 @@ class Context:
         rel = self._get_ral(sql)
         logger.debug("plan:\n%s", rel.explain())
         dc = RelConverter.convert(rel, context=self)
-        return ResultFrame(dc, rel, self)
+        res = ResultFrame(dc, rel, self)
+        return res.compute() if not return_futures else res
 
     def explain(self, sql: str) -> str:
         return self._get_ral(sql).explain()
