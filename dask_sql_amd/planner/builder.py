@@ -1574,6 +1574,18 @@ class Builder:
             for i, f in enumerate(fields):
                 if f.getName().lower() == ast[2].lower():
                     return i
+        if ast[0] == "col" and ast[1] is not None:
+            # qualified ORDER BY key (ORDER BY t.c) against the output:
+            # match name + qualifier, else the name alone when unique
+            q, n = ast[1].lower(), ast[2].lower()
+            hits = [i for i, f in enumerate(fields)
+                    if f.getName().lower() == n
+                    and (f.qualifier or "").lower() == q]
+            if not hits:
+                hits = [i for i, f in enumerate(fields)
+                        if f.getName().lower() == n]
+            if len(hits) == 1:
+                return hits[0]
         # structural match against select items (ORDER BY SUM(v) where the
         # same aggregate appears in the SELECT list, possibly aliased)
         for i, (e, alias) in enumerate(stmt.items):

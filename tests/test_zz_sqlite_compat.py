@@ -417,3 +417,40 @@ def test_sqlc_order_by_no_limit(ctx):
               a=a)
     eq_sqlite(ctx, "SELECT b, a FROM a ORDER BY b DESC NULLS LAST, "
                    "a ASC NULLS FIRST", a=a)
+
+
+def test_sqlc_join_multi(ctx):
+    a = rand_df(100, a=(int, 40), b=(str, 40), c=(float, 40))
+    b = rand_df(80, d=(float, 10), a=(int, 10), b=(str, 10))
+    c = rand_df(80, dd=(float, 10), a=(int, 10), b=(str, 10))
+    eq_sqlite(ctx, """
+        SELECT a.*,d,dd FROM a
+            INNER JOIN b ON a.a=b.a AND a.b=b.b
+            INNER JOIN c ON a.a=c.a AND c.b=b.b
+        ORDER BY a.a NULLS FIRST, a.b NULLS FIRST, a.c NULLS FIRST,
+            dd NULLS FIRST, d NULLS FIRST""", a=a, b=b, c=c)
+
+
+def test_sqlc_nested_window_filter(ctx):
+    a = rand_df(100, a=float, b=(int, 50), c=(str, 50))
+    eq_sqlite(ctx, """
+        SELECT * FROM (
+        SELECT *,
+            ROW_NUMBER() OVER (PARTITION BY c ORDER BY b NULLS FIRST,
+                a ASC NULLS LAST) AS r
+        FROM a)
+        WHERE r=1
+        ORDER BY a NULLS LAST, b NULLS LAST, c NULLS LAST""", a=a)
+
+
+def test_sqlc_union_three(ctx):
+    a = rand_df(30, b=(int, 10), c=(str, 10))
+    b = rand_df(80, b=(int, 50), c=(str, 50))
+    c = rand_df(100, b=(int, 50), c=(str, 50))
+    eq_sqlite(ctx, """
+        SELECT * FROM a UNION SELECT * FROM b UNION SELECT * FROM c
+        ORDER BY b NULLS FIRST, c NULLS FIRST""", a=a, b=b, c=c)
+    eq_sqlite(ctx, """
+        SELECT * FROM a UNION ALL SELECT * FROM b
+            UNION ALL SELECT * FROM c
+        ORDER BY b NULLS FIRST, c NULLS FIRST""", a=a, b=b, c=c)
