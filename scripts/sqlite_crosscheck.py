@@ -51,7 +51,8 @@ DIALECT = ("syntax error", "no such function", "no such column", "near",
 SEMANTIC_SKIP = {"test_float_key_join", "test_float_mod_and_mean",
                  "test_ref_sort_with_nan_matrix",
                  "test_case_sensitive_quoted_aliases",
-                 "test_datetime_trunc_exec"}
+                 "test_datetime_trunc_exec",
+                 "test_create_view_exec"}  # DDL — read_sql can't run it
 
 
 def main():
