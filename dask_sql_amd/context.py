@@ -470,11 +470,13 @@ class Context:
         # SHOW SCHEMAS / TABLES / COLUMNS (reference rel/custom/*.py,
         # expected frames pinned by tests/integration/test_show.py)
         sm = re.match(r'\s*SHOW\s+(SCHEMAS|TABLES|COLUMNS)'
-                      r'(?:\s+FROM\s+([\w".]+))?\s*;?\s*$', sql,
+                      r'(?:\s+FROM\s+([\w".]+))?'
+                      r"(?:\s+LIKE\s+'([^']*)')?\s*;?\s*$", sql,
                       re.IGNORECASE)
         if sm:
             what = sm.group(1).upper()
             arg = (sm.group(2) or "").replace('"', "")
+            like = sm.group(3)
             if what == "SCHEMAS":
                 pdf = pd.DataFrame({"Schema": [self.schema_name,
                                                "information_schema"]})
@@ -490,6 +492,14 @@ class Context:
                     "Extra": [""] * len(sqlt_low),
                     "Comment": [""] * len(sqlt_low),
                 })
+            if like is not None:
+                # SHOW ... LIKE '<pattern>' filters the first column by
+                # the LIKE pattern (reference show_schemas.py)
+                from dask_sql_amd.physical.rex import _like_regex
+                rx = _like_regex(like)
+                first = pdf.columns[0]
+                pdf = pdf[[bool(rx.fullmatch(str(v)))
+                           for v in pdf[first]]].reset_index(drop=True)
             from dask_sql_amd.datacontainer import HostDataContainer
             return ResultFrame(HostDataContainer(pdf), None, self)
         # ALTER TABLE [IF EXISTS] x RENAME TO y (reference DDL alter.py)
