@@ -10,7 +10,9 @@ import pytest
 from tests.conftest import assert_frame_close
 from tests.test_gpu_semantics import ctx, _rand_frame  # noqa: F401
 
-pytestmark = pytest.mark.gpu
+pytestmark = [pytest.mark.gpu, pytest.mark.timeout(180)]
+# the timeout bounds a hang in these UNVALIDATED paths to a test failure
+# instead of a dead GPU box (pytest-timeout is in the image)
 
 
 def test_cte_end_to_end(ctx):

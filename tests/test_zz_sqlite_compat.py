@@ -14,7 +14,9 @@ import pytest
 
 from tests.test_gpu_semantics import ctx  # noqa: F401
 
-pytestmark = pytest.mark.gpu
+pytestmark = [pytest.mark.gpu, pytest.mark.timeout(180)]
+# the timeout bounds a hang in these UNVALIDATED paths to a test failure
+# instead of a dead GPU box (pytest-timeout is in the image)
 
 
 def rand_df(size, **kwargs):
