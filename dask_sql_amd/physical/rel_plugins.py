@@ -1873,17 +1873,60 @@ class DaskAggregatePlugin(BaseRelPlugin):
         if any(len(k) > 4 and k[4] == 1 for k in keyspecs):
             raise RexCompileError(
                 "DISTINCT aggregate with a float GROUP BY key")
+        float_ids = {}
         for call in calls:
             args = agg.getArgs(call)
             if not args or not isinstance(args[0], InputRef):
                 raise RexCompileError("DISTINCT agg needs a plain column")
-            if cols[args[0].getIndex()].dtype not in _INT_KINDS:
-                raise RexCompileError("DISTINCT agg on non-integer (round-2)")
+            ci = args[0].getIndex()
+            if cols[ci].dtype == rt.F64:
+                fname = agg.getAggregationFuncName(call).lower()
+                if fname != "count":
+                    raise RexCompileError(
+                        "SUM/AVG DISTINCT on a float column (COUNT "
+                        "DISTINCT densifies; value-carrying distincts "
+                        "are round-3)")
+                # COUNT(DISTINCT f64): only DISTINCTNESS matters — give
+                # each bit-canonical value a dense id (same primitive
+                # chain as _densify_float_keys) and count ids
+                if ci not in float_ids:
+                    col = cols[ci]
+                    n_ = col.len
+                    oc, ov, on_, G = runtime.hash_groupby(
+                        [col], n_, [(0, 0, 0, True, 1)], None, [])
+                    runtime._free(ov)
+                    runtime._free(on_)
+                    bcol = rt.DeviceColumn(runtime, oc, None, max(G, 1),
+                                           rt.I64, owner=True)
+                    pcodes, _ = runtime.keypack([col], [(0, 0, 0, True, 1)],
+                                                n_)
+                    table = runtime.hash_build(bcol, None, code_max=0)
+                    try:
+                        p_, b_, cnt = runtime.hash_probe(
+                            table, pcodes, rt.JOIN_INNER, None)
+                        psel = runtime.wrap_sel(p_, cnt)
+                        bsel = runtime.wrap_sel(b_, cnt)
+                        fid = runtime.scatter_rows(bsel, psel.data, cnt,
+                                                   n_,
+                                                   with_validity=False)
+                    finally:
+                        runtime.hash_table_free(table)
+                    idcol = rt.DeviceColumn(runtime, fid.data,
+                                            col.validity, n_, rt.I64,
+                                            owner=False,
+                                            keep_alive=(fid, col))
+                    cols = list(cols)
+                    cols.append(idcol)
+                    float_ids[ci] = len(cols) - 1
+            elif cols[ci].dtype not in _INT_KINDS:
+                raise RexCompileError("DISTINCT agg on non-integer key "
+                                      "type")
         # distinct over (group keys + arg col): one kernel with extended keys
         out = []
         codes_ref = None
         for call in calls:
             ai = agg.getArgs(call)[0].getIndex()
+            ai = float_ids.get(ai, ai)
             mn, mx, nn = _minmax_cached(runtime, cols[ai])
             if nn == 0:
                 mn, mx = 0, 0
