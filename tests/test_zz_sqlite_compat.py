@@ -456,3 +456,17 @@ def test_sqlc_union_three(ctx):
         SELECT * FROM a UNION ALL SELECT * FROM b
             UNION ALL SELECT * FROM c
         ORDER BY b NULLS FIRST, c NULLS FIRST""", a=a, b=b, c=c)
+
+
+def test_sqlc_window_sum_avg_irregular(ctx):
+    a = rand_df(100, a=float, b=(int, 50), c=(str, 50))
+    eq_sqlite(ctx, """
+        SELECT a,b,
+            SUM(b) OVER (ORDER BY a DESC
+                ROWS BETWEEN 2 PRECEDING AND 1 PRECEDING) AS s6,
+            AVG(b) OVER (ORDER BY a DESC
+                ROWS BETWEEN 2 PRECEDING AND 1 FOLLOWING) AS a7,
+            SUM(b) OVER (PARTITION BY c ORDER BY a
+                ROWS BETWEEN 2 PRECEDING AND UNBOUNDED FOLLOWING) AS s8
+        FROM a
+        ORDER BY a NULLS FIRST, b NULLS FIRST, c NULLS FIRST""", a=a)
