@@ -952,3 +952,18 @@ def test_explain_statement_and_inline_dataframes():
     s2 = c.sql("EXPLAIN SELECT MIN(a) AS a_min FROM other_df GROUP BY a",
                dataframes={"other_df": pd.DataFrame({"a": [1]})})
     assert isinstance(s2, str) and "a_min" in s2 or "MIN" in s2
+
+
+def test_show_like_filtering():
+    """reference test_show.py — SHOW ... LIKE filters the first column."""
+    import pandas as pd
+
+    from dask_sql_amd.context import Context
+    c = Context()
+    c.create_table("table", pd.DataFrame({"x": [1]}))
+    out = c.sql("SHOW SCHEMAS LIKE 'information_schema'").compute()
+    assert out["Schema"].tolist() == ["information_schema"]
+    out = c.sql('SHOW TABLES FROM "root"').compute()
+    assert out["Table"].tolist() == ["table"]
+    out = c.sql("SHOW TABLES LIKE 'no_such%'").compute()
+    assert out["Table"].tolist() == []
