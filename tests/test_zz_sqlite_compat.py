@@ -129,6 +129,8 @@ def _compare(ours, theirs, ordered):
 # ---- the corpus (SQL fixtures from test_compatibility.py) ------------------
 def test_sqlc_basic_select(ctx):
     df = rand_df(5, a=(int, 2), b=(str, 3), c=(float, 4))
+    eq_sqlite(ctx, "SELECT 1 AS a, 1.5 AS b, 'x' AS c")
+    eq_sqlite(ctx, "SELECT 1+2 AS a, 1.5*3 AS b, 'x' AS c")
     eq_sqlite(ctx, "SELECT * FROM a", a=df)
     eq_sqlite(ctx, "SELECT * FROM a AS x", a=df)
     eq_sqlite(ctx, "SELECT b AS bb, a+1-2*3.0/4 AS cc FROM a", a=df)
