@@ -472,3 +472,15 @@ def test_sqlc_window_sum_avg_irregular(ctx):
                 ROWS BETWEEN 2 PRECEDING AND UNBOUNDED FOLLOWING) AS s8
         FROM a
         ORDER BY a NULLS FIRST, b NULLS FIRST, c NULLS FIRST""", a=a)
+
+
+def test_sqlc_agg_count_full(ctx):
+    a = rand_df(100, a=(int, 50), b=(str, 50), c=(int, 30), d=(str, 40),
+                e=(float, 40))
+    eq_sqlite(ctx, """
+        SELECT a, b, a+1 AS c,
+            COUNT(c) AS c_c, COUNT(DISTINCT c) AS cd_c,
+            COUNT(d) AS c_d, COUNT(DISTINCT d) AS cd_d,
+            COUNT(e) AS c_e, COUNT(DISTINCT a) AS cd_e
+        FROM a GROUP BY a, b
+        ORDER BY a NULLS FIRST, b NULLS FIRST""", a=a)
