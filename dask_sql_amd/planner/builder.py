@@ -273,8 +273,12 @@ class Builder:
                 # DATEPART('field', x) — function form of EXTRACT
                 # (reference call.py datepart → ExtractOperation)
                 field = str(ops[0].getValue()).upper()
-                if field.endswith("S"):
+                if field == "CENTURIES":
+                    field = "CENTURY"
+                elif field.endswith("S"):
                     field = field[:-1]
+                if field == "MILLENIUM":
+                    field = "MILLENNIUM"
                 # fall through: EXTRACT_WEEK and the generic typing below
                 # treat this exactly like EXTRACT(field FROM x)
                 ops = [ops[1]]
