@@ -151,9 +151,12 @@ def test_sqlc_drop_duplicates(ctx):
     a = rand_df(100, a=int, b=int)
     eq_sqlite(ctx, "SELECT DISTINCT b, a FROM a "
                    "ORDER BY a NULLS LAST, b NULLS FIRST", a=a)
-    a2 = rand_df(100, a=(int, 50), b=(str, 50))
+    a2 = rand_df(100, a=(int, 50), b=(int, 50))
     eq_sqlite(ctx, "SELECT DISTINCT b, a FROM a "
                    "ORDER BY a NULLS LAST, b NULLS FIRST", a=a2)
+    a3 = rand_df(100, a=(int, 50), b=(str, 50), c=float)
+    eq_sqlite(ctx, "SELECT DISTINCT b, a FROM a "
+                   "ORDER BY a NULLS LAST, b NULLS FIRST", a=a3)
 
 
 def test_sqlc_order_by(ctx):
