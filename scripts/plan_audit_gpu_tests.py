@@ -34,6 +34,40 @@ def make_ctx():
     return Context()
 
 
+class _MP:
+    """Minimal monkeypatch stand-in (setenv/delenv/setattr)."""
+
+    def __init__(self):
+        import os
+        self._os = os
+        self._undo = []
+
+    def setenv(self, k, v):
+        old = self._os.environ.get(k)
+        self._undo.append((k, old))
+        self._os.environ[k] = v
+
+    def delenv(self, k, raising=True):
+        old = self._os.environ.pop(k, None)
+        self._undo.append((k, old))
+
+    def setattr(self, obj, name, value):
+        self._undo.append((obj, name, getattr(obj, name)))
+        setattr(obj, name, value)
+
+    def undo(self):
+        for item in reversed(self._undo):
+            if len(item) == 2:
+                k, old = item
+                if old is None:
+                    self._os.environ.pop(k, None)
+                else:
+                    self._os.environ[k] = old
+            else:
+                obj, name, old = item
+                setattr(obj, name, old)
+
+
 def run_module(modname, fixture_makers):
     import importlib
     m = importlib.import_module(modname)
@@ -53,6 +87,14 @@ def run_module(modname, fixture_makers):
                     break
             elif p == "ctx" or p == "c":
                 args.append(ctx)
+            elif p == "monkeypatch":
+                args.append(_MP())
+            elif p == "tmp_path":
+                import pathlib
+                import tempfile
+                args.append(pathlib.Path(tempfile.mkdtemp()))
+            elif p in ("seed", "qi"):
+                args.append(0)
             elif p in fixture_makers:
                 try:
                     mk = fixture_makers[p]
@@ -79,6 +121,10 @@ def run_module(modname, fixture_makers):
             failed += 1
             print(f"PLAN-FAIL {modname}.{name}: {type(e).__name__}: "
                   f"{str(e)[:110]}")
+        finally:
+            for a_ in args:
+                if isinstance(a_, _MP):
+                    a_.undo()
     return ok, aborted, failed
 
 
@@ -95,6 +141,14 @@ def main():
         import importlib
         m = importlib.import_module(modname)
         makers = {}
+        try:
+            from tests import conftest as _cf
+            for fx in ("user_table_1", "user_table_2", "df_simple"):
+                f = getattr(_cf, fx, None)
+                if f is not None and hasattr(f, "__wrapped__"):
+                    makers[fx] = f.__wrapped__
+        except Exception:
+            pass
         cfx = getattr(m, "c", None)
         if cfx is not None and hasattr(cfx, "__wrapped__"):
             makers["c"] = cfx.__wrapped__
