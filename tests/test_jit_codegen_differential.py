@@ -111,10 +111,10 @@ def _devicify_casts(src):
 
 def _emit_source(lib, prog_tuple):
     arr, n = rt.Runtime.make_prog(prog_tuple)
-    dt = (ct.c_int32 * 4)(rt.I64, rt.I64, rt.F64, rt.F64)
-    hv = (ct.c_uint8 * 4)(0, 1, 0, 1)
+    dt = (ct.c_int32 * 5)(rt.I64, rt.I64, rt.F64, rt.F64, rt.I64)
+    hv = (ct.c_uint8 * 5)(0, 1, 0, 1, 0)
     buf = ct.create_string_buffer(1 << 16)
-    rc = lib.dsx_jit_expr_source(arr, n, dt, hv, 4, buf, len(buf))
+    rc = lib.dsx_jit_expr_source(arr, n, dt, hv, 5, buf, len(buf))
     if rc < 0:
         return None, None
     return buf.value.decode(), ("d" if rc == 1 else "l")
@@ -178,19 +178,21 @@ def harness():
     main_src = HARNESS_PRELUDE + "\n".join(sources) + f"""
 int main(int argc, char** argv) {{
   int idx = atoi(argv[1]);
-  static i64 ic0[{n_rows}], ic1[{n_rows}];
+  static i64 ic0[{n_rows}], ic1[{n_rows}], dc4[{n_rows}];
   static double fc2[{n_rows}], fc3[{n_rows}];
   static u8 v1[{n_rows}], v3[{n_rows}];
   FILE* f = fopen(argv[2], "rb");
   fread(ic0, 8, {n_rows}, f); fread(ic1, 8, {n_rows}, f);
   fread(fc2, 8, {n_rows}, f); fread(fc3, 8, {n_rows}, f);
+  fread(dc4, 8, {n_rows}, f);
   fread(v1, 1, {n_rows}, f); fread(v3, 1, {n_rows}, f);
   fclose(f);
   struct ColsArg C;
-  C.ncols = 4;
+  C.ncols = 5;
   C.data[0] = ic0; C.data[1] = ic1; C.data[2] = fc2; C.data[3] = fc3;
+  C.data[4] = dc4;
   C.validity[0] = 0; C.validity[1] = v1; C.validity[2] = 0;
-  C.validity[3] = v3;
+  C.validity[3] = v3; C.validity[4] = 0;
   for (i64 r = 0; r < {n_rows}; r++) {{
     switch (idx) {{
 {chr(10).join(calls)}
@@ -213,6 +215,7 @@ int main(int argc, char** argv) {{
         f.write(cols[1][0].astype("<i8").tobytes())
         f.write(cols[2][0].astype("<f8").tobytes())
         f.write(cols[3][0].astype("<f8").tobytes())
+        f.write(cols[4][0].astype("<i8").tobytes())
         f.write(cols[1][1].astype("u1").tobytes())
         f.write(cols[3][1].astype("u1").tobytes())
 

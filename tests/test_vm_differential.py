@@ -115,6 +115,13 @@ def ev(e, cols, row, pk=None):
     if op in ("ABS",):
         a = ev(ops[0], cols, row, pk)
         return None if a is None else abs(a)
+    if op in ("YEAR", "MONTH", "DAY"):
+        a = ev(ops[0], cols, row, pk)
+        if a is None:
+            return None
+        from tests.vm_interp import _civil
+        idx = {"YEAR": 0, "MONTH": 1, "DAY": 2}[op]
+        return _civil(int(a))[idx]
     if op in ("SIN", "COS", "TAN", "ATAN"):
         a = ev(ops[0], cols, row, pk)
         if a is None:
@@ -294,6 +301,10 @@ def gen(rng, kind, depth):
         return Call("ROUND", [gen(rng, "NUM", depth - 1),
                               Literal(d, SqlType(I))], SqlType(F))
     if r < 0.95:
+        if rng.random() < 0.5:
+            # civil-calendar op over the day-scale int column
+            fn = str(rng.choice(["YEAR", "MONTH", "DAY"]))
+            return Call(fn, [InputRef(4, SqlType("DATE"))], SqlType(I))
         return Call("MOD", [gen(rng, "NUM", depth - 1),
                             gen(rng, "NUM", depth - 1)], SqlType(I))
     ty = I if rng.random() < 0.5 else F
@@ -302,19 +313,22 @@ def gen(rng, kind, depth):
 
 def _make_cols(rng, n):
     iv = rng.integers(-40, 40, n).astype(np.int64)
+    dv = rng.integers(-30000, 60000, n).astype(np.int64)  # day counts
     inul = rng.integers(-40, 40, n).astype(np.int64)
     ival = (rng.random(n) > 0.25).astype(np.uint8)
     fv = np.round(rng.uniform(-40, 40, n), 3)
     fnul = np.round(rng.uniform(-40, 40, n), 3)
     fval = (rng.random(n) > 0.25).astype(np.uint8)
-    return [(iv, None), (inul, ival), (fv, None), (fnul, fval)]
+    return [(iv, None), (inul, ival), (fv, None), (fnul, fval),
+            (dv, None)]
 
 
 def _dev_cols():
     return [types.SimpleNamespace(dtype=rt.I64),
             types.SimpleNamespace(dtype=rt.I64),
             types.SimpleNamespace(dtype=rt.F64),
-            types.SimpleNamespace(dtype=rt.F64)]
+            types.SimpleNamespace(dtype=rt.F64),
+            types.SimpleNamespace(dtype=rt.I64)]
 
 
 @pytest.mark.parametrize("seed", range(14))
