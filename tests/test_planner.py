@@ -967,3 +967,35 @@ def test_show_like_filtering():
     assert out["Table"].tolist() == ["table"]
     out = c.sql("SHOW TABLES LIKE 'no_such%'").compute()
     assert out["Table"].tolist() == []
+
+
+def test_hidden_host_builtins_cpu():
+    """The hidden UDF builtins (month arithmetic, isoweek, rand) are pure
+    pandas — pin them directly."""
+    import numpy as np
+    import pandas as pd
+
+    from dask_sql_amd.planner.builder import (_add_months_host,
+                                              _isoweek_host, _rand_host)
+
+    ts = pd.to_datetime(["2021-01-31 10:00:00", "2020-02-29 23:00:00"])
+    vals = pd.Series(ts.view("int64"))
+    out = _add_months_host(False)(vals, pd.Series([1]))
+    want = pd.Series(ts) + pd.DateOffset(months=1)
+    assert (pd.to_datetime(out.astype("int64")) == want).all()
+
+    days = pd.Series(((pd.Series(ts).dt.normalize()
+                       - pd.Timestamp(0)).dt.days).astype("int64"))
+    out = _add_months_host(True)(days, pd.Series([2]))
+    want_d = (pd.Series(ts).dt.normalize() + pd.DateOffset(months=2)
+              - pd.Timestamp(0)).dt.days
+    assert out.astype("int64").tolist() == want_d.tolist()
+
+    wk = _isoweek_host(False)(vals)
+    want_w = pd.Series(ts).dt.isocalendar().week
+    assert wk.astype("int64").tolist() == [int(x) for x in want_w]
+
+    r = _rand_host(False)(pd.Series([0, 0, 0]))
+    assert len(r) == 3 and ((r >= 0) & (r < 1)).all()
+    ri = _rand_host(True)(pd.Series([10] * 5), pd.Series([0] * 5))
+    assert len(ri) == 5 and ((ri >= 0) & (ri < 10)).all()
